@@ -1,0 +1,92 @@
+"""Engine configuration.
+
+The flag surface mirrors what the reference's controller passes to its
+delegated engine via container args (reference
+docs/.../core-design.md:88-209): --model, --tensor-parallel-size,
+--kv-transfer-config {kv_connector, kv_role}, --max-model-len, ports.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from typing import Optional
+
+
+@dataclasses.dataclass
+class ModelConfig:
+    """Architecture hyperparameters (decoder-only, Llama/Qwen family)."""
+
+    name: str = "Qwen3-8B"
+    hidden_size: int = 4096
+    num_layers: int = 36
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    intermediate_size: int = 12288
+    vocab_size: int = 151936
+    rope_theta: float = 1_000_000.0
+    rms_norm_eps: float = 1e-6
+    max_position_embeddings: int = 40960
+    qk_norm: bool = True              # Qwen3 per-head q/k RMSNorm
+    tie_word_embeddings: bool = False
+    dtype: str = "bfloat16"
+
+    @property
+    def q_size(self) -> int:
+        return self.num_heads * self.head_dim
+
+    @property
+    def kv_size(self) -> int:
+        return self.num_kv_heads * self.head_dim
+
+
+@dataclasses.dataclass
+class CacheConfig:
+    block_size: int = 16
+    num_gpu_blocks: Optional[int] = None   # None = derive from gpu_memory_utilization
+    gpu_memory_utilization: float = 0.85
+    enable_prefix_caching: bool = True
+
+
+@dataclasses.dataclass
+class SchedulerConfig:
+    max_num_seqs: int = 256                # max sequences resident per step
+    max_num_batched_tokens: int = 8192     # per-step token budget
+    max_model_len: int = 8192
+
+
+@dataclasses.dataclass
+class ParallelConfig:
+    tensor_parallel_size: int = 1
+    pipeline_parallel_size: int = 1
+    data_parallel_size: int = 1
+    rank: int = 0
+    world_size: int = 1
+    distributed_backend: str = "nccl"      # "nccl" is RCCL on ROCm; tests use "gloo"
+
+
+@dataclasses.dataclass
+class KVTransferConfig:
+    """PD-disaggregation connector config.
+
+    Mirrors the reference's --kv-transfer-config
+    '{"kv_connector":"PyNcclConnector","kv_role":"kv_producer"|"kv_consumer"}'
+    (reference docs/.../core-design.md:88-111). Our connector moves packed
+    KV blocks with RCCL send/recv over xGMI.
+    """
+
+    kv_connector: Optional[str] = None     # e.g. "RcclConnector"
+    kv_role: Optional[str] = None          # "kv_producer" | "kv_consumer"
+    kv_rank: int = 0
+    kv_world_size: int = 2
+
+
+@dataclasses.dataclass
+class EngineConfig:
+    model: ModelConfig = dataclasses.field(default_factory=ModelConfig)
+    cache: CacheConfig = dataclasses.field(default_factory=CacheConfig)
+    scheduler: SchedulerConfig = dataclasses.field(default_factory=SchedulerConfig)
+    parallel: ParallelConfig = dataclasses.field(default_factory=ParallelConfig)
+    kv_transfer: KVTransferConfig = dataclasses.field(default_factory=KVTransferConfig)
+    seed: int = 0
+    enforce_eager: bool = False            # True disables hipGraph decode capture

@@ -1,0 +1,194 @@
+"""Op dispatch: CDNA4 HIP kernels on GPU, PyTorch fp32 reference on CPU.
+
+Policy (required by the build contract): on a GPU box the HIP extension is
+the ONLY execution path — if ``_C.so`` is missing or fails to import, any
+GPU-tensor call raises immediately instead of silently falling back to
+eager PyTorch. CPU tensors (CI containers without a GPU) use the reference
+implementations from ``fusioninfer_amd.ops.reference``.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from fusioninfer_amd.ops import reference as ref
+
+_C = None
+_C_IMPORT_ERROR: Optional[BaseException] = None
+try:
+    import importlib
+
+    _C = importlib.import_module("fusioninfer_amd.ops._C")
+except Exception as e:  # pragma: no cover - exercised only when build broken
+    _C_IMPORT_ERROR = e
+
+
+def has_native() -> bool:
+    return _C is not None
+
+
+def _require_native():
+    if _C is None:
+        raise RuntimeError(
+            "fusioninfer_amd HIP extension (_C.so) is not available but a GPU "
+            "tensor was passed — refusing to fall back to eager PyTorch on "
+            "the GPU path. Build it with `python -m fusioninfer_amd.ops.build`."
+        ) from _C_IMPORT_ERROR
+
+
+# ---------------------------------------------------------------- norm ops
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        _require_native()
+        out = torch.empty_like(x)
+        _C.rms_norm(out, x, weight, eps)
+        return out
+    return ref.rms_norm(x, weight, eps)
+
+
+def fused_add_rms_norm(x, residual, weight, eps: float):
+    """In-place on GPU: residual += x; x = rmsnorm(residual). Returns (x, residual)."""
+    if x.is_cuda:
+        _require_native()
+        _C.fused_add_rms_norm(x, residual, weight, eps)
+        return x, residual
+    return ref.fused_add_rms_norm(x, residual, weight, eps)
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        _require_native()
+        out = torch.empty(
+            (*x.shape[:-1], x.shape[-1] // 2), dtype=x.dtype, device=x.device
+        )
+        _C.silu_and_mul(out, x)
+        return out
+    return ref.silu_and_mul(x)
+
+
+# ---------------------------------------------------------------- rope
+
+def rope_qk_norm_(
+    q: torch.Tensor,          # [T, Hq*D] (row-strided slice OK)
+    k: torch.Tensor,          # [T, Hk*D]
+    positions: torch.Tensor,  # [T] int32
+    cos_sin: torch.Tensor,    # [P, D] fp32
+    num_q_heads: int,
+    num_kv_heads: int,
+    head_dim: int,
+    q_weight: Optional[torch.Tensor] = None,
+    k_weight: Optional[torch.Tensor] = None,
+    eps: float = 1e-6,
+):
+    """Fused optional per-head RMSNorm + NeoX RoPE, in-place. Returns (q, k)."""
+    if q.is_cuda:
+        _require_native()
+        _C.rope_qk_norm(
+            q, k, q_weight, k_weight, cos_sin, positions,
+            num_q_heads, num_kv_heads, head_dim, eps,
+        )
+        return q, k
+    qv = q.view(-1, num_q_heads, head_dim)
+    kv = k.view(-1, num_kv_heads, head_dim)
+    qo, ko = ref.rope_qk_norm(qv, kv, positions.long(), cos_sin, q_weight, k_weight, eps)
+    q.copy_(qo.reshape(q.shape))
+    k.copy_(ko.reshape(k.shape))
+    return q, k
+
+
+# ---------------------------------------------------------------- kv cache
+
+def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping):
+    if k.is_cuda:
+        _require_native()
+        _C.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        return
+    H, D = k_cache.shape[1], k_cache.shape[3]
+    ref.reshape_and_cache(
+        k.view(-1, H, D), v.view(-1, H, D), k_cache, v_cache, slot_mapping.long()
+    )
+
+
+def gather_kv_blocks(k_cache, v_cache, block_ids) -> torch.Tensor:
+    if k_cache.is_cuda:
+        _require_native()
+        n = int(block_ids.numel())
+        staging = torch.empty(
+            (2, n, *k_cache.shape[1:]), dtype=k_cache.dtype, device=k_cache.device
+        )
+        _C.gather_kv_blocks(staging, k_cache, v_cache, block_ids)
+        return staging
+    return ref.gather_kv_blocks(k_cache, v_cache, block_ids.long())
+
+
+def scatter_kv_blocks(staging, k_cache, v_cache, block_ids):
+    if k_cache.is_cuda:
+        _require_native()
+        _C.scatter_kv_blocks(staging, k_cache, v_cache, block_ids)
+        return
+    ref.scatter_kv_blocks(staging, k_cache, v_cache, block_ids.long())
+
+
+# ---------------------------------------------------------------- attention
+
+def paged_attention_decode(
+    q: torch.Tensor,            # [S, Hq, D]
+    k_cache: torch.Tensor,      # [B, Hk, bs, D]
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,  # [S, max_blocks] int32
+    seq_lens: torch.Tensor,      # [S] int32
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        _require_native()
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        _C.paged_attention_decode(out, q, k_cache, v_cache, block_tables, seq_lens, scale)
+        return out
+    return ref.paged_attention_decode(q, k_cache, v_cache, block_tables, seq_lens, scale)
+
+
+def build_prefill_tiles(seq_lens, device=None):
+    """Host-side tile table for the prefill kernel: 64 q-rows per workgroup.
+
+    seq_lens: list[int]. Returns (tile_seq, tile_row0) int32 tensors.
+    """
+    tile_seq, tile_row0 = [], []
+    for s, L in enumerate(seq_lens):
+        for r0 in range(0, L, 64):
+            tile_seq.append(s)
+            tile_row0.append(r0)
+    return (
+        torch.tensor(tile_seq, dtype=torch.int32, device=device),
+        torch.tensor(tile_row0, dtype=torch.int32, device=device),
+    )
+
+
+def prefill_attention(
+    q: torch.Tensor,           # [T, Hq, D]
+    k: torch.Tensor,           # [T, Hk, D]
+    v: torch.Tensor,           # [T, Hk, D]
+    cu_seqlens: torch.Tensor,  # [S+1] int32
+    scale: Optional[float] = None,
+    tile_seq: Optional[torch.Tensor] = None,
+    tile_row0: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        _require_native()
+        if tile_seq is None:
+            lens = (cu_seqlens[1:] - cu_seqlens[:-1]).tolist()
+            tile_seq, tile_row0 = build_prefill_tiles(lens, device=q.device)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        _C.prefill_attention(out, q, k, v, tile_seq, tile_row0, cu_seqlens, scale)
+        return out
+    return ref.prefill_attention(q, k, v, cu_seqlens, scale, causal=True)
+
+
+compute_cos_sin_cache = ref.compute_cos_sin_cache

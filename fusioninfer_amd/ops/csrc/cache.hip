@@ -1,0 +1,106 @@
+// Paged-KV cache kernels for CDNA4:
+//  * reshape_and_cache — scatter freshly-computed K/V rows into the paged
+//    cache ([num_blocks, Hk, block_size, D] bf16) by slot_mapping.
+//  * gather/scatter_kv_blocks — pack cache blocks into ONE contiguous
+//    staging buffer (and back) so the PD prefiller→decoder handoff is a
+//    single large RCCL send over an xGMI link (SURVEY.md §2.3 "KV-block
+//    pack/unpack for PD transfer"; §5.8 — pack into large contiguous
+//    sends to hit link peak).
+
+#include "common.h"
+
+namespace fi {
+
+__global__ void reshape_and_cache_kernel(
+    const u16* __restrict__ k,   // [T] rows of Hk*D, stride k_stride
+    const u16* __restrict__ v,
+    u16* __restrict__ k_cache,   // [B, Hk, bs, D]
+    u16* __restrict__ v_cache,
+    const int* __restrict__ slot_mapping,  // [T]
+    const int64_t k_stride, const int64_t v_stride,
+    const int num_tokens, const int kv_heads, const int block_size,
+    const int head_dim) {
+  const int vec_per_tok = kv_heads * head_dim / 8;
+  const int64_t total = static_cast<int64_t>(num_tokens) * vec_per_tok;
+  for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    const int t = static_cast<int>(idx / vec_per_tok);
+    const int c = static_cast<int>(idx % vec_per_tok) * 8;  // elem in [0,Hk*D)
+    const int slot = slot_mapping[t];
+    if (slot < 0) continue;  // padding token
+    const int blk = slot / block_size;
+    const int off = slot % block_size;
+    const int h = c / head_dim;
+    const int d = c % head_dim;
+    const int64_t dst =
+        ((static_cast<int64_t>(blk) * kv_heads + h) * block_size + off) *
+            head_dim + d;
+    *reinterpret_cast<bf16x8*>(k_cache + dst) =
+        *reinterpret_cast<const bf16x8*>(k + t * k_stride + c);
+    *reinterpret_cast<bf16x8*>(v_cache + dst) =
+        *reinterpret_cast<const bf16x8*>(v + t * v_stride + c);
+  }
+}
+
+void launch_reshape_and_cache(const u16* k, const u16* v, u16* k_cache,
+                              u16* v_cache, const int* slot_mapping,
+                              int64_t k_stride, int64_t v_stride, int tokens,
+                              int kv_heads, int block_size, int head_dim,
+                              hipStream_t stream) {
+  const int64_t total = static_cast<int64_t>(tokens) * kv_heads * head_dim / 8;
+  const int block = 256;
+  const int grid = static_cast<int>(std::min<int64_t>((total + block - 1) / block, (int64_t)2048));
+  hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(grid), dim3(block), 0,
+                     stream, k, v, k_cache, v_cache, slot_mapping, k_stride,
+                     v_stride, tokens, kv_heads, block_size, head_dim);
+}
+
+// staging: [2, n, Hk, bs, D]; GATHER ? cache->staging : staging->cache.
+template <bool GATHER>
+__global__ void kv_block_copy_kernel(u16* __restrict__ staging,
+                                     u16* __restrict__ k_cache,
+                                     u16* __restrict__ v_cache,
+                                     const int* __restrict__ block_ids,
+                                     const int num_blocks,
+                                     const int64_t block_elems) {  // Hk*bs*D
+  const int64_t vec_per_block = block_elems / 8;
+  const int64_t total = num_blocks * vec_per_block;
+  for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += static_cast<int64_t>(gridDim.x) * blockDim.x) {
+    const int i = static_cast<int>(idx / vec_per_block);
+    const int64_t c = (idx % vec_per_block) * 8;
+    const int64_t cache_off = static_cast<int64_t>(block_ids[i]) * block_elems + c;
+    const int64_t stage_off = static_cast<int64_t>(i) * block_elems + c;
+    if (GATHER) {
+      *reinterpret_cast<bf16x8*>(staging + stage_off) =
+          *reinterpret_cast<const bf16x8*>(k_cache + cache_off);
+      *reinterpret_cast<bf16x8*>(staging + num_blocks * block_elems + stage_off) =
+          *reinterpret_cast<const bf16x8*>(v_cache + cache_off);
+    } else {
+      *reinterpret_cast<bf16x8*>(k_cache + cache_off) =
+          *reinterpret_cast<const bf16x8*>(staging + stage_off);
+      *reinterpret_cast<bf16x8*>(v_cache + cache_off) =
+          *reinterpret_cast<const bf16x8*>(
+              staging + num_blocks * block_elems + stage_off);
+    }
+  }
+}
+
+template <bool GATHER>
+void launch_kv_block_copy(u16* staging, u16* k_cache, u16* v_cache,
+                          const int* block_ids, int num_blocks,
+                          int64_t block_elems, hipStream_t stream) {
+  const int64_t total = num_blocks * (block_elems / 8);
+  const int block = 256;
+  const int grid = static_cast<int>(std::min<int64_t>((total + block - 1) / block, (int64_t)4096));
+  hipLaunchKernelGGL((kv_block_copy_kernel<GATHER>), dim3(grid), dim3(block),
+                     0, stream, staging, k_cache, v_cache, block_ids,
+                     num_blocks, block_elems);
+}
+
+template void launch_kv_block_copy<true>(u16*, u16*, u16*, const int*, int,
+                                         int64_t, hipStream_t);
+template void launch_kv_block_copy<false>(u16*, u16*, u16*, const int*, int,
+                                          int64_t, hipStream_t);
+
+}  // namespace fi

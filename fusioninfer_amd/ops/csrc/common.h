@@ -1,0 +1,85 @@
+// Common device helpers for FusionInfer-AMD CDNA4 (gfx950) kernels.
+//
+// Design notes (see /opt/skills/guides/cdna_hip_programming.md):
+//  * wave64 everywhere; block sizes are multiples of 64.
+//  * bf16 loads are ALWAYS vectorized (short4/short8 reinterpret, G13).
+//  * bf16<->f32 via bit ops (RNE) — matches hardware v_cvt rounding.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+#include <algorithm>
+
+#define FI_DEV __device__ __forceinline__
+
+namespace fi {
+
+constexpr int kWaveSize = 64;
+
+using u16 = unsigned short;
+using u32 = unsigned int;
+
+// 16-byte vector of 8 bf16 values (raw bits).
+struct alignas(16) bf16x8 {
+  u16 h[8];
+};
+// 8-byte vector of 4 bf16 values.
+struct alignas(8) bf16x4 {
+  u16 h[4];
+};
+
+FI_DEV float bf16_to_f32(u16 h) {
+  union {
+    u32 u;
+    float f;
+  } x;
+  x.u = static_cast<u32>(h) << 16;
+  return x.f;
+}
+
+FI_DEV u16 f32_to_bf16(float f) {
+  union {
+    u32 u;
+    float f;
+  } x;
+  x.f = f;
+  // round-to-nearest-even; NaN-safe (NaN stays NaN: mantissa MSB kept below)
+  u32 rounding_bias = 0x7fff + ((x.u >> 16) & 1);
+  if (x.u << 1 > 0xff000000u) {  // NaN
+    return static_cast<u16>((x.u >> 16) | 0x0040);
+  }
+  return static_cast<u16>((x.u + rounding_bias) >> 16);
+}
+
+// Wave-wide reductions (64 lanes).
+FI_DEV float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+FI_DEV float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+// Block-wide sum over NWAVES waves (each wave passes its wave-reduced value
+// with lane 0 valid). `scratch` must hold NWAVES floats.
+template <int NWAVES>
+FI_DEV float block_reduce_sum(float wave_val, float* scratch) {
+  const int wave = threadIdx.x / kWaveSize;
+  const int lane = threadIdx.x % kWaveSize;
+  if (lane == 0) scratch[wave] = wave_val;
+  __syncthreads();
+  float total = 0.f;
+#pragma unroll
+  for (int i = 0; i < NWAVES; ++i) total += scratch[i];
+  return total;
+}
+
+__host__ __device__ __forceinline__ constexpr int ceil_div(int a, int b) {
+  return (a + b - 1) / b;
+}
+
+}  // namespace fi

@@ -1,0 +1,211 @@
+// PyTorch bindings for the FusionInfer-AMD CDNA4 kernel library.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+namespace fi {
+
+template <bool FUSED_ADD>
+void launch_rms_norm(u16*, const u16*, u16*, const u16*, float, int, int,
+                     hipStream_t);
+void launch_silu_and_mul(u16*, const u16*, int64_t, int, hipStream_t);
+void launch_rope_qk_norm(u16*, u16*, int64_t, int64_t, const u16*, const u16*,
+                         const float*, const int*, int, int, int, int, float,
+                         hipStream_t);
+void launch_reshape_and_cache(const u16*, const u16*, u16*, u16*, const int*,
+                              int64_t, int64_t, int, int, int, int,
+                              hipStream_t);
+template <bool GATHER>
+void launch_kv_block_copy(u16*, u16*, u16*, const int*, int, int64_t,
+                          hipStream_t);
+void launch_paged_attn_decode(u16*, const u16*, const u16*, const u16*,
+                              const int*, const int*, int, int64_t, int, int,
+                              int, int, float, hipStream_t);
+void launch_prefill_attn(u16*, const u16*, const u16*, const u16*, const int*,
+                         const int*, const int*, int, int64_t, int64_t,
+                         int64_t, int, int, int, float, hipStream_t);
+
+}  // namespace fi
+
+namespace {
+
+using fi::u16;
+
+#define CHECK_BF16_CUDA(t)                                     \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");            \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16")
+
+u16* bf16_ptr(at::Tensor& t) { return reinterpret_cast<u16*>(t.data_ptr()); }
+const u16* bf16_cptr(const at::Tensor& t) {
+  return reinterpret_cast<const u16*>(t.data_ptr());
+}
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void rms_norm(at::Tensor out, at::Tensor input, at::Tensor weight, double eps) {
+  CHECK_BF16_CUDA(out);
+  CHECK_BF16_CUDA(input);
+  TORCH_CHECK(input.is_contiguous() && out.is_contiguous());
+  const int hidden = input.size(-1);
+  const int tokens = input.numel() / hidden;
+  fi::launch_rms_norm<false>(bf16_ptr(out), bf16_cptr(input), nullptr,
+                             bf16_cptr(weight), static_cast<float>(eps),
+                             tokens, hidden, current_stream());
+}
+
+void fused_add_rms_norm(at::Tensor input, at::Tensor residual,
+                        at::Tensor weight, double eps) {
+  CHECK_BF16_CUDA(input);
+  CHECK_BF16_CUDA(residual);
+  TORCH_CHECK(input.is_contiguous() && residual.is_contiguous());
+  const int hidden = input.size(-1);
+  const int tokens = input.numel() / hidden;
+  // in-place: residual += input; input = rmsnorm(residual)
+  fi::launch_rms_norm<true>(bf16_ptr(input), bf16_cptr(input),
+                            bf16_ptr(residual), bf16_cptr(weight),
+                            static_cast<float>(eps), tokens, hidden,
+                            current_stream());
+}
+
+void silu_and_mul(at::Tensor out, at::Tensor input) {
+  CHECK_BF16_CUDA(out);
+  CHECK_BF16_CUDA(input);
+  TORCH_CHECK(input.is_contiguous() && out.is_contiguous());
+  const int inter = out.size(-1);
+  TORCH_CHECK(input.size(-1) == 2 * inter);
+  TORCH_CHECK(inter % 8 == 0);
+  fi::launch_silu_and_mul(bf16_ptr(out), bf16_cptr(input),
+                          input.numel() / (2 * inter), inter,
+                          current_stream());
+}
+
+void rope_qk_norm(at::Tensor q, at::Tensor k,
+                  c10::optional<at::Tensor> q_weight,
+                  c10::optional<at::Tensor> k_weight, at::Tensor cos_sin,
+                  at::Tensor positions, int64_t num_q_heads,
+                  int64_t num_kv_heads, int64_t head_dim, double eps) {
+  CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(k);
+  TORCH_CHECK(cos_sin.scalar_type() == at::kFloat && cos_sin.is_cuda());
+  TORCH_CHECK(positions.scalar_type() == at::kInt && positions.is_cuda());
+  // q/k may be row-strided slices of the fused qkv projection output
+  TORCH_CHECK(q.dim() == 2 && k.dim() == 2);
+  TORCH_CHECK(q.stride(1) == 1 && k.stride(1) == 1);
+  const int tokens = q.size(0);
+  fi::launch_rope_qk_norm(
+      bf16_ptr(q), bf16_ptr(k), q.stride(0), k.stride(0),
+      q_weight ? bf16_cptr(*q_weight) : nullptr,
+      k_weight ? bf16_cptr(*k_weight) : nullptr, cos_sin.data_ptr<float>(),
+      positions.data_ptr<int>(), tokens, num_q_heads, num_kv_heads, head_dim,
+      static_cast<float>(eps), current_stream());
+}
+
+void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
+                       at::Tensor v_cache, at::Tensor slot_mapping) {
+  CHECK_BF16_CUDA(k);
+  CHECK_BF16_CUDA(v);
+  CHECK_BF16_CUDA(k_cache);
+  CHECK_BF16_CUDA(v_cache);
+  TORCH_CHECK(slot_mapping.scalar_type() == at::kInt);
+  TORCH_CHECK(k.dim() == 2 && v.dim() == 2);  // [T, Hk*D] (maybe strided rows)
+  TORCH_CHECK(k.stride(1) == 1 && v.stride(1) == 1);
+  const int kv_heads = k_cache.size(1);
+  const int block_size = k_cache.size(2);
+  const int head_dim = k_cache.size(3);
+  fi::launch_reshape_and_cache(
+      bf16_cptr(k), bf16_cptr(v), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      slot_mapping.data_ptr<int>(), k.stride(0), v.stride(0), k.size(0),
+      kv_heads, block_size, head_dim, current_stream());
+}
+
+void gather_kv_blocks(at::Tensor staging, at::Tensor k_cache,
+                      at::Tensor v_cache, at::Tensor block_ids) {
+  CHECK_BF16_CUDA(staging);
+  TORCH_CHECK(block_ids.scalar_type() == at::kInt);
+  const int n = block_ids.size(0);
+  const int64_t block_elems =
+      k_cache.size(1) * k_cache.size(2) * k_cache.size(3);
+  fi::launch_kv_block_copy<true>(bf16_ptr(staging), bf16_ptr(k_cache),
+                                 bf16_ptr(v_cache), block_ids.data_ptr<int>(),
+                                 n, block_elems, current_stream());
+}
+
+void scatter_kv_blocks(at::Tensor staging, at::Tensor k_cache,
+                       at::Tensor v_cache, at::Tensor block_ids) {
+  CHECK_BF16_CUDA(staging);
+  TORCH_CHECK(block_ids.scalar_type() == at::kInt);
+  const int n = block_ids.size(0);
+  const int64_t block_elems =
+      k_cache.size(1) * k_cache.size(2) * k_cache.size(3);
+  fi::launch_kv_block_copy<false>(bf16_ptr(staging), bf16_ptr(k_cache),
+                                  bf16_ptr(v_cache), block_ids.data_ptr<int>(),
+                                  n, block_elems, current_stream());
+}
+
+void paged_attention_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
+                            at::Tensor v_cache, at::Tensor block_tables,
+                            at::Tensor seq_lens, double scale) {
+  CHECK_BF16_CUDA(out);
+  CHECK_BF16_CUDA(q);
+  TORCH_CHECK(block_tables.scalar_type() == at::kInt &&
+              block_tables.is_contiguous());
+  TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
+  TORCH_CHECK(q.dim() == 3);  // [S, Hq, D], rows may be strided
+  const int num_seqs = q.size(0);
+  const int num_heads = q.size(1);
+  const int head_dim = q.size(2);
+  const int num_kv_heads = k_cache.size(1);
+  TORCH_CHECK(k_cache.size(2) == 16, "cache block_size must be 16");
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == head_dim);
+  fi::launch_paged_attn_decode(
+      bf16_ptr(out), bf16_cptr(q), bf16_cptr(k_cache), bf16_cptr(v_cache),
+      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), num_seqs,
+      q.stride(0), block_tables.size(1), num_kv_heads, head_dim,
+      num_heads / num_kv_heads, static_cast<float>(scale), current_stream());
+}
+
+void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
+                       at::Tensor v, at::Tensor tile_seq, at::Tensor tile_row0,
+                       at::Tensor cu_seqlens, double scale) {
+  CHECK_BF16_CUDA(out);
+  CHECK_BF16_CUDA(q);
+  TORCH_CHECK(tile_seq.scalar_type() == at::kInt &&
+              tile_row0.scalar_type() == at::kInt &&
+              cu_seqlens.scalar_type() == at::kInt);
+  TORCH_CHECK(q.dim() == 3 && k.dim() == 3 && v.dim() == 3);
+  const int num_q_heads = q.size(1);
+  const int head_dim = q.size(2);
+  const int num_kv_heads = k.size(1);
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == head_dim);
+  TORCH_CHECK(k.stride(2) == 1 && k.stride(1) == head_dim);
+  TORCH_CHECK(v.stride(2) == 1 && v.stride(1) == head_dim);
+  fi::launch_prefill_attn(
+      bf16_ptr(out), bf16_cptr(q), bf16_cptr(k), bf16_cptr(v),
+      tile_seq.data_ptr<int>(), tile_row0.data_ptr<int>(),
+      cu_seqlens.data_ptr<int>(), tile_seq.size(0), q.stride(0), k.stride(0),
+      v.stride(0), num_q_heads, num_kv_heads, head_dim,
+      static_cast<float>(scale), current_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm, "RMSNorm (bf16, fp32 accum)");
+  m.def("fused_add_rms_norm", &fused_add_rms_norm,
+        "in-place residual add + RMSNorm");
+  m.def("silu_and_mul", &silu_and_mul, "fused SiLU-mul");
+  m.def("rope_qk_norm", &rope_qk_norm,
+        "fused per-head qk RMSNorm + NeoX RoPE (in-place)");
+  m.def("reshape_and_cache", &reshape_and_cache, "scatter K/V into paged cache");
+  m.def("gather_kv_blocks", &gather_kv_blocks, "pack KV blocks for PD send");
+  m.def("scatter_kv_blocks", &scatter_kv_blocks, "unpack KV blocks from PD recv");
+  m.def("paged_attention_decode", &paged_attention_decode,
+        "paged decode attention");
+  m.def("prefill_attention", &prefill_attention, "varlen causal MFMA prefill");
+}

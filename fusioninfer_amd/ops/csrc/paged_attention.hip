@@ -1,0 +1,248 @@
+// Paged-attention DECODE kernel for CDNA4 (gfx950).
+//
+// One-token attention per sequence against the paged KV cache
+// ([num_blocks, Hk, 16, D] bf16). Memory-bound: the job is to stream each
+// sequence's K/V exactly once at near-HBM rate while reusing every K/V read
+// across the GQA group (G q-heads per kv-head).
+//
+// Structure: one 4-wave workgroup per (seq, kv_head). Waves process
+// interleaved 16-token chunks (1 chunk == 1 cache block, so a chunk's K/V
+// are contiguous). Per chunk, two lane layouts:
+//   phase A (scores): lane = token*4 + i, lane reads K[token][i*32 .. +32]
+//     as 4x bf16x8 (64 B); dot with q (fp32, staged in LDS); 2-level
+//     shfl_xor reduce; online softmax state (m,l per G) kept wave-uniform
+//     in registers; probs parked in per-wave LDS.
+//   phase B (PV): lane = dim pair (2 elems), per-token V rows read 256 B
+//     coalesced; fp32 accumulator acc[G][2] per lane.
+// Final cross-wave flash-merge through LDS.
+//
+// Capability parity: the paged-attention decode the reference delegates to
+// its vLLM containers (SURVEY.md §2.3 "Paged-attention decode kernel").
+
+#include "common.h"
+
+namespace fi {
+
+constexpr int kBlockSz = 16;   // cache block size (tokens)
+constexpr int kNWaves = 4;
+constexpr float kNegInf = -1e30f;
+
+template <int D, int G>
+__global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
+    u16* __restrict__ out,            // [S, Hq, D]
+    const u16* __restrict__ q,        // [S] rows, stride q_stride, Hq*D elems
+    const u16* __restrict__ k_cache,  // [B, Hk, 16, D]
+    const u16* __restrict__ v_cache,
+    const int* __restrict__ block_tables,  // [S, max_blocks]
+    const int* __restrict__ seq_lens,      // [S]
+    const int64_t q_stride, const int max_blocks, const int num_kv_heads,
+    const float scale) {
+  const int seq = blockIdx.x;
+  const int kv_head = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int wave = tid / kWaveSize;
+  const int lane = tid % kWaveSize;
+
+  const int ctx = seq_lens[seq];
+  const int num_chunks = (ctx + kBlockSz - 1) / kBlockSz;
+
+  __shared__ float q_lds[G][D];
+  __shared__ float p_lds[kNWaves][kBlockSz][G];
+  __shared__ float merge_m[kNWaves][G];
+  __shared__ float merge_l[kNWaves][G];
+  __shared__ float merge_acc[kNWaves][G][D];
+
+  // stage q (G heads) into LDS as fp32
+  {
+    const u16* q_row = q + seq * q_stride +
+                       static_cast<int64_t>(kv_head) * G * D;
+    for (int e = tid; e < G * D; e += kNWaves * kWaveSize)
+      q_lds[e / D][e % D] = bf16_to_f32(q_row[e]) * scale;
+  }
+  __syncthreads();
+
+  // online-softmax state, wave-uniform (every lane holds the same copy)
+  float m[G], l[G], acc[G][2];
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    m[g] = kNegInf;
+    l[g] = 0.f;
+    acc[g][0] = acc[g][1] = 0.f;
+  }
+
+  const int tok = lane / 4;          // phase-A token within chunk
+  const int quad = lane % 4;         // phase-A dim quarter
+  constexpr int DPQ = D / 4;         // dims per phase-A lane
+
+  for (int chunk = wave; chunk < num_chunks; chunk += kNWaves) {
+    const int block_id = block_tables[seq * max_blocks + chunk];
+    const int64_t kv_base =
+        ((static_cast<int64_t>(block_id) * num_kv_heads + kv_head) * kBlockSz) * D;
+    const int token_pos = chunk * kBlockSz + tok;
+
+    // ---- phase A: scores for 16 tokens x G heads ----
+    float s[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) s[g] = 0.f;
+    {
+      const u16* k_row = k_cache + kv_base + tok * D + quad * DPQ;
+      float kf[DPQ];
+#pragma unroll
+      for (int j8 = 0; j8 < DPQ / 8; ++j8) {
+        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k_row + j8 * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kf[j8 * 8 + j] = bf16_to_f32(kv8.h[j]);
+      }
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        float acc_s = 0.f;
+#pragma unroll
+        for (int j4 = 0; j4 < DPQ / 4; ++j4) {
+          // float4 -> ds_read_b128 (4x fewer LDS cycles than scalar reads)
+          const float4 qv = *reinterpret_cast<const float4*>(
+              &q_lds[g][quad * DPQ + j4 * 4]);
+          acc_s = fmaf(kf[j4 * 4 + 0], qv.x, acc_s);
+          acc_s = fmaf(kf[j4 * 4 + 1], qv.y, acc_s);
+          acc_s = fmaf(kf[j4 * 4 + 2], qv.z, acc_s);
+          acc_s = fmaf(kf[j4 * 4 + 3], qv.w, acc_s);
+        }
+        s[g] = acc_s;
+      }
+    }
+    // reduce over the 4 dim-quarters (lanes 4t..4t+3)
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      s[g] += __shfl_xor(s[g], 1, 64);
+      s[g] += __shfl_xor(s[g], 2, 64);
+      if (token_pos >= ctx) s[g] = kNegInf;
+    }
+
+    // chunk max over tokens (xor 4..32 spans the 16 token groups)
+    float m_new[G], alpha[G];
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      float cm = s[g];
+#pragma unroll
+      for (int off = 4; off < 64; off <<= 1)
+        cm = fmaxf(cm, __shfl_xor(cm, off, 64));
+      m_new[g] = fmaxf(m[g], cm);
+      alpha[g] = __expf(m[g] - m_new[g]);  // exp(-inf-(-inf)) guarded below
+      if (m[g] <= kNegInf && m_new[g] <= kNegInf) alpha[g] = 0.f;
+      m[g] = m_new[g];
+    }
+
+    // probs + row-sum; every token is replicated on 4 lanes -> scale by 1/4
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      float p = (s[g] <= kNegInf) ? 0.f : __expf(s[g] - m[g]);
+      if (quad == 0) p_lds[wave][tok][g] = p;
+      float psum = p;
+#pragma unroll
+      for (int off = 1; off < 64; off <<= 1) psum += __shfl_xor(psum, off, 64);
+      l[g] = l[g] * alpha[g] + psum * 0.25f;
+    }
+
+    // ---- phase B: PV accumulate; lane covers dims {2l, 2l+1} ----
+    const int nvalid = min(ctx - chunk * kBlockSz, kBlockSz);
+    const u16* v_rows = v_cache + kv_base;
+    if (lane < D / 2) {
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        acc[g][0] *= alpha[g];
+        acc[g][1] *= alpha[g];
+      }
+      for (int t = 0; t < nvalid; ++t) {
+        const u32 vbits = *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
+        const float v0 = bf16_to_f32(static_cast<u16>(vbits & 0xffff));
+        const float v1 = bf16_to_f32(static_cast<u16>(vbits >> 16));
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const float p = p_lds[wave][t][g];
+          acc[g][0] = fmaf(p, v0, acc[g][0]);
+          acc[g][1] = fmaf(p, v1, acc[g][1]);
+        }
+      }
+    } else {
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        acc[g][0] *= alpha[g];
+        acc[g][1] *= alpha[g];
+      }
+    }
+  }
+
+  // ---- cross-wave flash merge ----
+  if (lane < G) {
+    merge_m[wave][lane] = m[lane];
+    merge_l[wave][lane] = l[lane];
+  }
+  if (lane < D / 2) {
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      merge_acc[wave][g][2 * lane] = acc[g][0];
+      merge_acc[wave][g][2 * lane + 1] = acc[g][1];
+    }
+  }
+  __syncthreads();
+
+  // each wave merges a share of the G heads
+  constexpr int kGPerWave = (G + kNWaves - 1) / kNWaves;
+#pragma unroll
+  for (int gi = 0; gi < kGPerWave; ++gi) {
+    const int g = wave * kGPerWave + gi;
+    if (g >= G || lane >= D / 2) continue;
+    float gm = kNegInf;
+#pragma unroll
+    for (int w = 0; w < kNWaves; ++w) gm = fmaxf(gm, merge_m[w][g]);
+    float L = 0.f, o0 = 0.f, o1 = 0.f;
+#pragma unroll
+    for (int w = 0; w < kNWaves; ++w) {
+      const float mw = merge_m[w][g];
+      const float f = (mw <= kNegInf) ? 0.f : __expf(mw - gm);
+      L += merge_l[w][g] * f;
+      o0 = fmaf(merge_acc[w][g][2 * lane], f, o0);
+      o1 = fmaf(merge_acc[w][g][2 * lane + 1], f, o1);
+    }
+    const float inv = 1.f / L;
+    u16* o_row = out + (static_cast<int64_t>(seq) * num_kv_heads * G +
+                        kv_head * G + g) * D;
+    const u32 obits = static_cast<u32>(f32_to_bf16(o0 * inv)) |
+                      (static_cast<u32>(f32_to_bf16(o1 * inv)) << 16);
+    *reinterpret_cast<u32*>(o_row + 2 * lane) = obits;
+  }
+}
+
+void launch_paged_attn_decode(u16* out, const u16* q, const u16* k_cache,
+                              const u16* v_cache, const int* block_tables,
+                              const int* seq_lens, int num_seqs,
+                              int64_t q_stride, int max_blocks,
+                              int num_kv_heads, int head_dim, int group,
+                              float scale, hipStream_t stream) {
+  dim3 grid(num_seqs, num_kv_heads), block(kNWaves * kWaveSize);
+#define FI_LAUNCH(DD, GG)                                                    \
+  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG>), grid, block, 0,     \
+                     stream, out, q, k_cache, v_cache, block_tables,         \
+                     seq_lens, q_stride, max_blocks, num_kv_heads, scale)
+  if (head_dim == 128) {
+    switch (group) {
+      case 1: FI_LAUNCH(128, 1); break;
+      case 2: FI_LAUNCH(128, 2); break;
+      case 4: FI_LAUNCH(128, 4); break;
+      case 8: FI_LAUNCH(128, 8); break;
+      default: abort();
+    }
+  } else if (head_dim == 64) {
+    switch (group) {
+      case 1: FI_LAUNCH(64, 1); break;
+      case 2: FI_LAUNCH(64, 2); break;
+      case 4: FI_LAUNCH(64, 4); break;
+      case 8: FI_LAUNCH(64, 8); break;
+      default: abort();
+    }
+  } else {
+    abort();
+  }
+#undef FI_LAUNCH
+}
+
+}  // namespace fi

@@ -1,0 +1,190 @@
+"""Pure-PyTorch fp32 reference implementations of every HIP op.
+
+These are (a) the numerics oracle the GPU kernels are tested against and
+(b) the CPU execution path for tests in GPU-less environments. They are
+intentionally simple and allocate freely — never used on the GPU hot path
+(ops/__init__.py raises if the HIP extension is missing on a GPU box).
+"""
+
+from __future__ import annotations
+
+import torch
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    """RMSNorm over the last dim. Computes in fp32, returns x.dtype."""
+    xf = x.float()
+    var = xf.pow(2).mean(dim=-1, keepdim=True)
+    out = xf * torch.rsqrt(var + eps) * weight.float()
+    return out.to(x.dtype)
+
+
+def fused_add_rms_norm(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
+):
+    """residual = residual + x; out = rms_norm(residual). Returns (out, residual)."""
+    new_residual = (residual.float() + x.float()).to(x.dtype)
+    return rms_norm(new_residual, weight, eps), new_residual
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    """x: [..., 2*I] (gate | up) -> silu(gate) * up, in fp32."""
+    gate, up = x.float().chunk(2, dim=-1)
+    return (torch.nn.functional.silu(gate) * up).to(x.dtype)
+
+
+def compute_cos_sin_cache(
+    head_dim: int, max_positions: int, theta: float, device=None
+) -> torch.Tensor:
+    """[max_positions, head_dim] fp32; first half cos, second half sin (NeoX)."""
+    inv_freq = 1.0 / (
+        theta ** (torch.arange(0, head_dim, 2, dtype=torch.float32, device=device) / head_dim)
+    )
+    t = torch.arange(max_positions, dtype=torch.float32, device=device)
+    freqs = torch.outer(t, inv_freq)  # [P, D/2]
+    return torch.cat([freqs.cos(), freqs.sin()], dim=-1)
+
+
+def _apply_rope_neox(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+    """x: [T, H, D]; cos/sin: [T, D/2]. Rotate-half (NeoX) style, fp32."""
+    d2 = x.shape[-1] // 2
+    x1, x2 = x[..., :d2], x[..., d2:]
+    cos = cos.unsqueeze(1)
+    sin = sin.unsqueeze(1)
+    return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], dim=-1)
+
+
+def rope_qk_norm(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    positions: torch.Tensor,
+    cos_sin: torch.Tensor,
+    q_weight=None,
+    k_weight=None,
+    eps: float = 1e-6,
+):
+    """Optional per-head RMSNorm (Qwen3 qk-norm) then NeoX RoPE.
+
+    q: [T, Hq, D], k: [T, Hk, D], positions: [T] int. Returns new (q, k).
+    """
+    dtype = q.dtype
+    qf, kf = q.float(), k.float()
+    if q_weight is not None:
+        qf = rms_norm(qf, q_weight, eps)
+    if k_weight is not None:
+        kf = rms_norm(kf, k_weight, eps)
+    d2 = q.shape[-1] // 2
+    cs = cos_sin[positions]  # [T, D]
+    cos, sin = cs[:, :d2], cs[:, d2:]
+    qf = _apply_rope_neox(qf.float(), cos, sin)
+    kf = _apply_rope_neox(kf.float(), cos, sin)
+    return qf.to(dtype), kf.to(dtype)
+
+
+def reshape_and_cache(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+):
+    """k/v: [T, Hk, D]; caches: [num_blocks, Hk, block_size, D]; slots: [T]."""
+    block_size = k_cache.shape[2]
+    blk = torch.div(slot_mapping, block_size, rounding_mode="floor")
+    off = slot_mapping % block_size
+    k_cache[blk, :, off] = k.to(k_cache.dtype)
+    v_cache[blk, :, off] = v.to(v_cache.dtype)
+
+
+def paged_attention_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    """One-token attention per sequence against the paged cache.
+
+    q: [S, Hq, D]; caches: [B, Hk, bs, D]; block_tables: [S, max_blocks];
+    seq_lens: [S] (context length INCLUDING the current token). fp32 math.
+    """
+    num_seqs, num_heads, head_dim = q.shape
+    num_kv_heads = k_cache.shape[1]
+    bs = k_cache.shape[2]
+    group = num_heads // num_kv_heads
+    out = torch.empty_like(q, dtype=torch.float32)
+    for s in range(num_seqs):
+        ctx = int(seq_lens[s])
+        nblk = (ctx + bs - 1) // bs
+        blocks = block_tables[s, :nblk].long()
+        # [Hk, nblk*bs, D]
+        keys = k_cache[blocks].permute(1, 0, 2, 3).reshape(num_kv_heads, -1, head_dim)
+        vals = v_cache[blocks].permute(1, 0, 2, 3).reshape(num_kv_heads, -1, head_dim)
+        keys = keys[:, :ctx].float()
+        vals = vals[:, :ctx].float()
+        qh = q[s].float()  # [Hq, D]
+        for h in range(num_heads):
+            kv_h = h // group
+            scores = keys[kv_h] @ qh[h] * scale  # [ctx]
+            p = torch.softmax(scores, dim=-1)
+            out[s, h] = p @ vals[kv_h]
+    return out.to(q.dtype)
+
+
+def prefill_attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cu_seqlens: torch.Tensor,
+    scale: float,
+    causal: bool = True,
+) -> torch.Tensor:
+    """Varlen causal attention, dense (non-paged) K/V.
+
+    q: [T, Hq, D]; k/v: [T, Hk, D]; cu_seqlens: [num_seqs+1] int32.
+    """
+    T, num_heads, head_dim = q.shape
+    num_kv_heads = k.shape[1]
+    group = num_heads // num_kv_heads
+    out = torch.empty_like(q, dtype=torch.float32)
+    for i in range(cu_seqlens.numel() - 1):
+        s, e = int(cu_seqlens[i]), int(cu_seqlens[i + 1])
+        L = e - s
+        qf = q[s:e].float()  # [L, Hq, D]
+        kf = k[s:e].float()
+        vf = v[s:e].float()
+        for h in range(num_heads):
+            kv_h = h // group
+            scores = qf[:, h] @ kf[:, kv_h].T * scale  # [L, L]
+            if causal:
+                mask = torch.triu(
+                    torch.ones(L, L, dtype=torch.bool, device=q.device), diagonal=1
+                )
+                scores = scores.masked_fill(mask, float("-inf"))
+            p = torch.softmax(scores, dim=-1)
+            out[s:e, h] = p @ vf[:, kv_h]
+    return out.to(q.dtype)
+
+
+def gather_kv_blocks(
+    k_cache: torch.Tensor, v_cache: torch.Tensor, block_ids: torch.Tensor
+) -> torch.Tensor:
+    """Pack the given cache blocks into one contiguous staging buffer.
+
+    Returns [2, nblocks, Hk, bs, D] (K plane then V plane) — a single large
+    contiguous tensor so the PD handoff is one RCCL send per batch of blocks.
+    """
+    idx = block_ids.long()
+    return torch.stack([k_cache[idx], v_cache[idx]], dim=0).contiguous()
+
+
+def scatter_kv_blocks(
+    staging: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_ids: torch.Tensor,
+):
+    idx = block_ids.long()
+    k_cache[idx] = staging[0].to(k_cache.dtype)
+    v_cache[idx] = staging[1].to(v_cache.dtype)

@@ -1,0 +1,187 @@
+"""GPU numerics tests: every HIP kernel vs the PyTorch fp32 reference.
+
+Mirrors the reference's test strategy tier 1 (table-driven unit tests,
+SURVEY.md §4) applied to the data plane: each op is compared against
+ops/reference.py on random inputs.
+"""
+
+import math
+
+import pytest
+import torch
+
+import fusioninfer_amd.ops as ops
+from fusioninfer_amd.ops import reference as ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def setup_module():
+    assert ops.has_native(), "HIP extension must be present on a GPU box"
+    torch.manual_seed(0)
+
+
+def assert_close_bf16(actual, expected_f32, atol=2e-2, rtol=2e-2):
+    torch.testing.assert_close(
+        actual.float(), expected_f32.float(), atol=atol, rtol=rtol
+    )
+
+
+@pytest.mark.parametrize("tokens,hidden", [(1, 1024), (17, 4096), (256, 4096), (33, 8192)])
+def test_rms_norm(tokens, hidden):
+    x = torch.randn(tokens, hidden, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(hidden, dtype=torch.bfloat16, device=DEV)
+    out = ops.rms_norm(x, w, 1e-6)
+    expected = ref.rms_norm(x.cpu(), w.cpu(), 1e-6)
+    assert_close_bf16(out.cpu(), expected)
+
+
+@pytest.mark.parametrize("tokens,hidden", [(16, 4096), (130, 4096)])
+def test_fused_add_rms_norm(tokens, hidden):
+    x = torch.randn(tokens, hidden, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn(tokens, hidden, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(hidden, dtype=torch.bfloat16, device=DEV)
+    x_ref, res_ref = ref.fused_add_rms_norm(x.cpu(), res.cpu(), w.cpu(), 1e-6)
+    out, new_res = ops.fused_add_rms_norm(x, res, w, 1e-6)
+    assert_close_bf16(new_res.cpu(), res_ref)
+    assert_close_bf16(out.cpu(), x_ref)
+
+
+@pytest.mark.parametrize("tokens,inter", [(7, 12288), (64, 512)])
+def test_silu_and_mul(tokens, inter):
+    x = torch.randn(tokens, 2 * inter, dtype=torch.bfloat16, device=DEV)
+    out = ops.silu_and_mul(x)
+    expected = ref.silu_and_mul(x.cpu())
+    assert_close_bf16(out.cpu(), expected)
+
+
+@pytest.mark.parametrize("qk_norm", [True, False])
+@pytest.mark.parametrize("head_dim", [64, 128])
+def test_rope_qk_norm(qk_norm, head_dim):
+    T, Hq, Hk = 33, 8, 2
+    torch.manual_seed(1)
+    # row-strided slices of a fused qkv buffer, like the real model
+    qkv = torch.randn(T, (Hq + 2 * Hk) * head_dim, dtype=torch.bfloat16, device=DEV)
+    q = qkv[:, : Hq * head_dim]
+    k = qkv[:, Hq * head_dim : (Hq + Hk) * head_dim]
+    positions = torch.randint(0, 500, (T,), dtype=torch.int32, device=DEV)
+    cos_sin = ops.compute_cos_sin_cache(head_dim, 512, 10000.0).to(DEV)
+    qw = torch.randn(head_dim, dtype=torch.bfloat16, device=DEV) if qk_norm else None
+    kw = torch.randn(head_dim, dtype=torch.bfloat16, device=DEV) if qk_norm else None
+
+    q_ref, k_ref = ref.rope_qk_norm(
+        q.cpu().view(T, Hq, head_dim),
+        k.cpu().view(T, Hk, head_dim),
+        positions.cpu().long(),
+        cos_sin.cpu(),
+        qw.cpu() if qw is not None else None,
+        kw.cpu() if kw is not None else None,
+        1e-6,
+    )
+    ops.rope_qk_norm_(q, k, positions, cos_sin, Hq, Hk, head_dim, qw, kw, 1e-6)
+    assert_close_bf16(q.cpu().view(T, Hq, head_dim), q_ref)
+    assert_close_bf16(k.cpu().view(T, Hk, head_dim), k_ref)
+
+
+def test_reshape_and_cache():
+    T, Hk, D, bs, nblocks = 37, 4, 128, 16, 12
+    k = torch.randn(T, Hk * D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, Hk * D, dtype=torch.bfloat16, device=DEV)
+    k_cache = torch.zeros(nblocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV)
+    v_cache = torch.zeros_like(k_cache)
+    slots = torch.randperm(nblocks * bs, device=DEV)[:T].to(torch.int32)
+    k_ref = k_cache.cpu().clone()
+    v_ref = v_cache.cpu().clone()
+    ref.reshape_and_cache(
+        k.cpu().view(T, Hk, D), v.cpu().view(T, Hk, D), k_ref, v_ref, slots.cpu().long()
+    )
+    ops.reshape_and_cache(k, v, k_cache, v_cache, slots)
+    torch.testing.assert_close(k_cache.cpu(), k_ref)
+    torch.testing.assert_close(v_cache.cpu(), v_ref)
+
+
+def test_kv_block_gather_scatter_roundtrip():
+    Hk, D, bs, nblocks = 2, 128, 16, 20
+    k_cache = torch.randn(nblocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV)
+    v_cache = torch.randn_like(k_cache)
+    ids = torch.tensor([3, 7, 1, 19], dtype=torch.int32, device=DEV)
+    staging = ops.gather_kv_blocks(k_cache, v_cache, ids)
+    expected = ref.gather_kv_blocks(k_cache.cpu(), v_cache.cpu(), ids.cpu().long())
+    torch.testing.assert_close(staging.cpu(), expected)
+
+    # scatter into a fresh cache at different ids
+    k2 = torch.zeros_like(k_cache)
+    v2 = torch.zeros_like(v_cache)
+    ids2 = torch.tensor([0, 2, 4, 6], dtype=torch.int32, device=DEV)
+    ops.scatter_kv_blocks(staging, k2, v2, ids2)
+    torch.testing.assert_close(k2[ids2.long()].cpu(), k_cache[ids.long()].cpu())
+    torch.testing.assert_close(v2[ids2.long()].cpu(), v_cache[ids.long()].cpu())
+
+
+@pytest.mark.parametrize("group", [1, 4, 8])
+@pytest.mark.parametrize(
+    "seq_lens", [[1], [16], [1, 5, 16, 17, 255, 1023]]
+)
+def test_paged_attention_decode(group, seq_lens):
+    torch.manual_seed(2)
+    Hk, D, bs = 2, 128, 16
+    Hq = Hk * group
+    S = len(seq_lens)
+    max_blocks = (max(seq_lens) + bs - 1) // bs
+    total_blocks = sum((L + bs - 1) // bs for L in seq_lens) + 2
+    q = torch.randn(S, Hq, D, dtype=torch.bfloat16, device=DEV)
+    k_cache = torch.randn(total_blocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV)
+    v_cache = torch.randn_like(k_cache)
+    # assign blocks sequentially
+    bt = torch.zeros(S, max_blocks, dtype=torch.int32, device=DEV)
+    nxt = 1
+    for s, L in enumerate(seq_lens):
+        n = (L + bs - 1) // bs
+        bt[s, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    lens = torch.tensor(seq_lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention_decode(q, k_cache, v_cache, bt, lens, scale)
+    expected = ref.paged_attention_decode(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), lens.cpu(), scale
+    )
+    assert_close_bf16(out.cpu(), expected, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("head_dim", [64, 128])
+@pytest.mark.parametrize(
+    "seq_lens,hq,hk",
+    [([64], 4, 4), ([128], 8, 2), ([1, 33, 64, 100, 257], 8, 2)],
+)
+def test_prefill_attention(head_dim, seq_lens, hq, hk):
+    torch.manual_seed(3)
+    T = sum(seq_lens)
+    q = torch.randn(T, hq, head_dim, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, hk, head_dim, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, hk, head_dim, dtype=torch.bfloat16, device=DEV)
+    cu = torch.tensor(
+        [0] + list(torch.tensor(seq_lens).cumsum(0)), dtype=torch.int32, device=DEV
+    )
+    scale = 1.0 / math.sqrt(head_dim)
+    out = ops.prefill_attention(q, k, v, cu, scale)
+    expected = ref.prefill_attention(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), scale)
+    assert_close_bf16(out.cpu(), expected, atol=3e-2, rtol=3e-2)
+
+
+def test_prefill_attention_spiked_scores():
+    """Force large per-tile max jumps so online-softmax rescale paths fire
+    (guide §5.4 rule 26: bounded random data never exercises them)."""
+    torch.manual_seed(4)
+    L, hq, hk, D = 256, 2, 2, 128
+    q = torch.randn(L, hq, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(L, hk, D, dtype=torch.bfloat16, device=DEV) * 0.1
+    # spike some K rows late in the sequence: max jumps at a late tile
+    k[200] = q[250, 0].sign() * 3.0
+    k[77] = q[100, 1].sign() * 2.0
+    v = torch.randn(L, hk, D, dtype=torch.bfloat16, device=DEV)
+    cu = torch.tensor([0, L], dtype=torch.int32, device=DEV)
+    out = ops.prefill_attention(q, k, v, cu)
+    expected = ref.prefill_attention(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), 1.0 / math.sqrt(D))
+    assert_close_bf16(out.cpu(), expected, atol=3e-2, rtol=3e-2)

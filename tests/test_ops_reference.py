@@ -1,0 +1,73 @@
+"""CPU sanity tests for the reference ops (the oracle the HIP kernels
+are checked against) — validated against independent torch formulations."""
+
+import math
+
+import torch
+
+from fusioninfer_amd.ops import reference as ref
+
+
+def test_rms_norm_matches_formula():
+    torch.manual_seed(0)
+    x = torch.randn(5, 64)
+    w = torch.randn(64)
+    out = ref.rms_norm(x, w, 1e-6)
+    expected = x / (x.pow(2).mean(-1, keepdim=True) + 1e-6).sqrt() * w
+    torch.testing.assert_close(out, expected, atol=1e-5, rtol=1e-5)
+
+
+def test_silu_and_mul():
+    x = torch.randn(3, 32)
+    out = ref.silu_and_mul(x)
+    g, u = x.chunk(2, -1)
+    torch.testing.assert_close(out, torch.nn.functional.silu(g) * u)
+
+
+def test_prefill_attention_vs_sdpa():
+    torch.manual_seed(1)
+    L, H, D = 37, 4, 32
+    q = torch.randn(L, H, D)
+    k = torch.randn(L, H, D)
+    v = torch.randn(L, H, D)
+    cu = torch.tensor([0, L], dtype=torch.int32)
+    out = ref.prefill_attention(q, k, v, cu, 1.0 / math.sqrt(D))
+    expected = torch.nn.functional.scaled_dot_product_attention(
+        q.transpose(0, 1), k.transpose(0, 1), v.transpose(0, 1), is_causal=True
+    ).transpose(0, 1)
+    torch.testing.assert_close(out, expected, atol=1e-4, rtol=1e-4)
+
+
+def test_decode_consistent_with_prefill_last_row():
+    """Decoding the last token against cached K/V must equal the last row of
+    full prefill attention."""
+    torch.manual_seed(2)
+    L, Hq, Hk, D, bs = 33, 4, 2, 32, 16
+    q = torch.randn(L, Hq, D)
+    k = torch.randn(L, Hk, D)
+    v = torch.randn(L, Hk, D)
+    cu = torch.tensor([0, L], dtype=torch.int32)
+    full = ref.prefill_attention(q, k, v, cu, 1.0 / math.sqrt(D))
+
+    nblk = (L + bs - 1) // bs
+    k_cache = torch.zeros(nblk + 1, Hk, bs, D)
+    v_cache = torch.zeros_like(k_cache)
+    slots = torch.arange(L)
+    ref.reshape_and_cache(k, v, k_cache, v_cache, slots)
+    bt = torch.arange(nblk, dtype=torch.int32).unsqueeze(0)
+    lens = torch.tensor([L], dtype=torch.int32)
+    dec = ref.paged_attention_decode(
+        q[-1:], k_cache, v_cache, bt, lens, 1.0 / math.sqrt(D)
+    )
+    torch.testing.assert_close(dec[0], full[-1], atol=1e-4, rtol=1e-4)
+
+
+def test_kv_pack_unpack_roundtrip():
+    k_cache = torch.randn(8, 2, 16, 32)
+    v_cache = torch.randn(8, 2, 16, 32)
+    ids = torch.tensor([5, 0, 3])
+    staging = ref.gather_kv_blocks(k_cache, v_cache, ids)
+    k2, v2 = torch.zeros_like(k_cache), torch.zeros_like(v_cache)
+    ref.scatter_kv_blocks(staging, k2, v2, ids)
+    torch.testing.assert_close(k2[ids], k_cache[ids])
+    torch.testing.assert_close(v2[ids], v_cache[ids])
