@@ -1,0 +1,131 @@
+"""Paged-KV block manager: allocation, ref-counting, prefix-cache reuse.
+
+Owns the mapping sequence -> physical cache blocks. Hash-based prefix
+caching mirrors the reuse the reference's EPP prefix-cache scorer assumes
+exists engine-side (reference pkg/router/strategy.go:51-68): full prompt
+blocks are content-hashed (chained) and reusable across sequences.
+
+NOTE v1: block reuse is wired, but the attention path does not yet read
+cached context during prefill, so the engine only activates reuse when
+`enable_prefix_caching` is on AND the runner supports cached context.
+"""
+
+from __future__ import annotations
+
+from collections import OrderedDict
+from typing import Dict, List, Optional
+
+from fusioninfer_amd.engine.sequence import Sequence
+
+
+class BlockManager:
+    def __init__(self, num_blocks: int, block_size: int,
+                 enable_prefix_caching: bool = False):
+        self.num_blocks = num_blocks
+        self.block_size = block_size
+        self.enable_prefix_caching = enable_prefix_caching
+        self.free_blocks: List[int] = list(range(num_blocks))
+        self.ref_count: Dict[int, int] = {}
+        # content hash -> block id, for FULL blocks only
+        self.hash_to_block: Dict[int, int] = {}
+        self.block_hash: Dict[int, int] = {}
+        # blocks with ref 0 that still hold reusable content (LRU)
+        self.cached_free: "OrderedDict[int, None]" = OrderedDict()
+
+    # ------------------------------------------------------------ helpers
+    def num_free(self) -> int:
+        return len(self.free_blocks) + len(self.cached_free)
+
+    def _pop_free_block(self) -> int:
+        if self.free_blocks:
+            return self.free_blocks.pop()
+        # evict LRU cached block
+        blk, _ = self.cached_free.popitem(last=False)
+        h = self.block_hash.pop(blk, None)
+        if h is not None:
+            self.hash_to_block.pop(h, None)
+        return blk
+
+    def blocks_needed(self, num_tokens: int) -> int:
+        return (num_tokens + self.block_size - 1) // self.block_size
+
+    # ------------------------------------------------------------ alloc
+    def can_allocate(self, num_tokens: int) -> bool:
+        return self.num_free() >= self.blocks_needed(num_tokens)
+
+    def allocate(self, seq: Sequence) -> None:
+        """Allocate blocks for the whole prompt; reuse prefix-cache hits."""
+        assert not seq.block_ids
+        n = self.blocks_needed(seq.num_prompt_tokens)
+        cached = 0
+        if self.enable_prefix_caching:
+            h = 0
+            for b in range(n):
+                start, end = b * self.block_size, (b + 1) * self.block_size
+                if end > seq.num_prompt_tokens:
+                    break  # partial last block never cache-hits
+                h = hash((h, tuple(seq.prompt_token_ids[start:end])))
+                blk = self.hash_to_block.get(h)
+                if blk is None:
+                    break
+                self._take(blk)
+                seq.block_ids.append(blk)
+                cached += self.block_size
+        seq.num_cached_tokens = cached
+        while len(seq.block_ids) < n:
+            blk = self._pop_free_block()
+            self.ref_count[blk] = 1
+            seq.block_ids.append(blk)
+        if self.enable_prefix_caching:
+            self._register_hashes(seq)
+
+    def _take(self, blk: int) -> None:
+        if blk in self.cached_free:
+            del self.cached_free[blk]
+            self.ref_count[blk] = 1
+        else:
+            self.ref_count[blk] += 1
+
+    def _register_hashes(self, seq: Sequence) -> None:
+        h = 0
+        for b, blk in enumerate(seq.block_ids):
+            end = (b + 1) * self.block_size
+            if end > seq.num_prompt_tokens:
+                break
+            start = b * self.block_size
+            h = hash((h, tuple(seq.prompt_token_ids[start:end])))
+            if blk not in self.block_hash:
+                self.block_hash[blk] = h
+                self.hash_to_block.setdefault(h, blk)
+
+    def _blocks_missing(self, seq: Sequence) -> int:
+        """Blocks needed so the NEXT decode position (num_tokens-1) has a slot."""
+        position = seq.num_tokens - 1
+        have = len(seq.block_ids)
+        need = position // self.block_size + 1
+        return max(need - have, 0)
+
+    def can_append_slot(self, seq: Sequence) -> bool:
+        return self.num_free() >= self._blocks_missing(seq)
+
+    def append_slot(self, seq: Sequence) -> None:
+        """Ensure a block exists for position seq.num_tokens - 1."""
+        for _ in range(self._blocks_missing(seq)):
+            blk = self._pop_free_block()
+            self.ref_count[blk] = 1
+            seq.block_ids.append(blk)
+
+    def free(self, seq: Sequence) -> None:
+        for blk in seq.block_ids:
+            self.ref_count[blk] -= 1
+            if self.ref_count[blk] == 0:
+                del self.ref_count[blk]
+                if self.enable_prefix_caching and blk in self.block_hash:
+                    self.cached_free[blk] = None
+                else:
+                    self.free_blocks.append(blk)
+        seq.block_ids = []
+
+    def slot_for(self, seq: Sequence, position: int) -> int:
+        blk = seq.block_ids[position // self.block_size]
+        return blk * self.block_size + position % self.block_size

@@ -1,0 +1,131 @@
+"""LLMEngine: the per-GPU serving engine (continuous batching over paged KV).
+
+One engine instance per GPU process (TP ranks share one logical engine:
+rank 0 drives scheduling; all ranks execute the same batch — v1 runs
+TP via symmetric SPMD stepping, see distributed/parallel_state.py).
+"""
+
+from __future__ import annotations
+
+import itertools
+import time
+from typing import Dict, List, Optional
+
+import torch
+
+from fusioninfer_amd.config import EngineConfig
+from fusioninfer_amd.engine.block_manager import BlockManager
+from fusioninfer_amd.engine.model_runner import ModelRunner
+from fusioninfer_amd.engine.sampler import Sampler
+from fusioninfer_amd.engine.scheduler import ScheduledBatch, Scheduler
+from fusioninfer_amd.engine.sequence import SamplingParams, Sequence, SeqStatus
+
+
+class RequestOutput:
+    def __init__(self, seq: Sequence):
+        self.request_id = seq.seq_id
+        self.prompt_token_ids = seq.prompt_token_ids
+        self.output_token_ids = list(seq.output_token_ids)
+        self.finished = seq.is_finished()
+        self.ttft = seq.ttft
+        self.arrival_time = seq.arrival_time
+        self.finish_time = seq.finish_time
+
+
+class LLMEngine:
+    def __init__(self, cfg: EngineConfig, device: Optional[str] = None):
+        from fusioninfer_amd.distributed import parallel_state as ps
+
+        ps.ensure_single_process()
+        self.cfg = cfg
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = device
+        self.runner = ModelRunner(cfg, device)
+        self.runner.allocate_kv_caches()
+        self.block_manager = BlockManager(
+            self.runner.num_gpu_blocks,
+            cfg.cache.block_size,
+            enable_prefix_caching=False,  # v1: see block_manager.py note
+        )
+        self.scheduler = Scheduler(cfg.scheduler, self.block_manager)
+        self.sampler = Sampler(cfg.seed, device)
+        self.runner.capture_decode_graphs()
+        self._req_counter = itertools.count()
+        self.seqs: Dict[str, Sequence] = {}
+        # running stats for metrics / EPP scorers
+        self.num_finished = 0
+        self.num_generated_tokens = 0
+        self.num_prefilled_tokens = 0
+
+    # ------------------------------------------------------------ requests
+    def add_request(
+        self,
+        prompt_token_ids: List[int],
+        sampling: Optional[SamplingParams] = None,
+        request_id: Optional[str] = None,
+    ) -> str:
+        if request_id is None:
+            request_id = f"req-{next(self._req_counter)}"
+        seq = Sequence(request_id, prompt_token_ids, sampling or SamplingParams())
+        assert seq.num_prompt_tokens <= self.cfg.scheduler.max_model_len
+        self.seqs[request_id] = seq
+        self.scheduler.add(seq)
+        return request_id
+
+    def has_unfinished(self) -> bool:
+        return self.scheduler.has_work()
+
+    # ------------------------------------------------------------ metrics
+    def gpu_cache_usage(self) -> float:
+        total = self.block_manager.num_blocks
+        return 1.0 - self.block_manager.num_free() / max(total, 1)
+
+    def num_waiting(self) -> int:
+        return self.scheduler.num_waiting
+
+    def num_running(self) -> int:
+        return self.scheduler.num_running
+
+    # ------------------------------------------------------------ stepping
+    def step(self) -> List[RequestOutput]:
+        """One engine iteration. Returns outputs for sequences that produced
+        a token this step (finished ones flagged)."""
+        batch = self.scheduler.schedule()
+        if batch.is_empty:
+            return []
+        if batch.is_prefill:
+            seqs = batch.prefill_seqs
+            logits = self.runner.execute_prefill(seqs, self.block_manager)
+            self.num_prefilled_tokens += sum(s.num_prompt_tokens for s in seqs)
+        else:
+            seqs = batch.decode_seqs
+            logits = self.runner.execute_decode(seqs, self.block_manager)
+        next_tokens = self.sampler.sample(logits.float(), seqs)
+        next_tokens = next_tokens.tolist()
+
+        outputs: List[RequestOutput] = []
+        for seq, tok in zip(seqs, next_tokens):
+            seq.append_token(int(tok))
+            self.num_generated_tokens += 1
+            if seq.check_stop():
+                seq.finish_time = time.monotonic()
+                self.scheduler.finish(seq)
+                self.num_finished += 1
+                del self.seqs[seq.seq_id]
+            outputs.append(RequestOutput(seq))
+        return outputs
+
+    # ------------------------------------------------------------ sync API
+    def generate(
+        self,
+        prompts: List[List[int]],
+        sampling: Optional[SamplingParams] = None,
+    ) -> List[RequestOutput]:
+        ids = [self.add_request(p, sampling) for p in prompts]
+        done: Dict[str, RequestOutput] = {}
+        while self.has_unfinished() and len(done) < len(ids):
+            for out in self.step():
+                if out.finished:
+                    done[out.request_id] = out
+        return [done[i] for i in ids]

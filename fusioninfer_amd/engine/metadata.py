@@ -1,0 +1,30 @@
+"""Per-step batch metadata handed from the scheduler to the model forward.
+
+Token layout in a step's flat batch: [prefill tokens ... | decode tokens].
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from typing import Optional
+
+import torch
+
+
+@dataclasses.dataclass
+class AttnMetadata:
+    num_prefill_tokens: int
+    num_decode_tokens: int
+    positions: torch.Tensor            # [T] int32, all tokens
+    slot_mapping: torch.Tensor         # [T] int32, cache slot per token
+    # prefill segment
+    cu_seqlens: Optional[torch.Tensor] = None    # [P+1] int32
+    tile_seq: Optional[torch.Tensor] = None      # kernel tile table
+    tile_row0: Optional[torch.Tensor] = None
+    # decode segment
+    block_tables: Optional[torch.Tensor] = None  # [Dq, max_blocks] int32
+    seq_lens: Optional[torch.Tensor] = None      # [Dq] int32 (ctx incl. current)
+
+    @property
+    def num_tokens(self) -> int:
+        return self.num_prefill_tokens + self.num_decode_tokens

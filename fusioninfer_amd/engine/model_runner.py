@@ -1,0 +1,219 @@
+"""Model runner: batch preparation + forward execution + hipGraph decode.
+
+Decode steps have static shapes (pure-decode batches, fixed-width block
+tables), so they are captured once per batch-size bucket into hipGraphs
+(torch.cuda.CUDAGraph == hipGraph on ROCm) and replayed — removing
+launch overhead from the latency-critical decode loop (guide: "capture
+launch-bound inner loops in hipGraphs"). Prefill runs eager.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from fusioninfer_amd.config import EngineConfig
+from fusioninfer_amd.engine.block_manager import BlockManager
+from fusioninfer_amd.engine.metadata import AttnMetadata
+from fusioninfer_amd.engine.sequence import Sequence
+from fusioninfer_amd.models.model import CausalLM
+import fusioninfer_amd.ops as ops_mod
+
+_DECODE_BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256]
+
+
+class ModelRunner:
+    def __init__(self, cfg: EngineConfig, device: str):
+        self.cfg = cfg
+        self.device = torch.device(device)
+        self.is_cuda = self.device.type == "cuda"
+        mc = cfg.model
+        torch.manual_seed(cfg.seed)
+        with torch.device(self.device):
+            self.model = CausalLM(mc).eval()
+        self.block_size = cfg.cache.block_size
+        self.max_blocks_per_seq = (
+            cfg.scheduler.max_model_len + self.block_size - 1
+        ) // self.block_size
+        self.kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = []
+        self.num_gpu_blocks = 0
+        self._graphs: Dict[int, Tuple[object, torch.Tensor]] = {}
+        self._graph_pool = None
+        self._static: Dict[str, torch.Tensor] = {}
+
+    # ------------------------------------------------------------ kv cache
+    def kv_block_bytes(self) -> int:
+        mc = self.cfg.model
+        import fusioninfer_amd.distributed.parallel_state as ps
+
+        kv_heads = mc.num_kv_heads // ps.tp_world_size()
+        return mc.num_layers * 2 * kv_heads * self.block_size * mc.head_dim * 2
+
+    def profile_num_blocks(self) -> int:
+        if self.cfg.cache.num_gpu_blocks is not None:
+            return self.cfg.cache.num_gpu_blocks
+        assert self.is_cuda
+        free_b, total_b = torch.cuda.mem_get_info(self.device)
+        usable = int(
+            total_b * self.cfg.cache.gpu_memory_utilization
+            - (total_b - free_b)
+        )
+        # headroom for activations / graphs
+        usable -= 4 << 30
+        n = max(usable // self.kv_block_bytes(), 16)
+        return int(n)
+
+    def allocate_kv_caches(self) -> None:
+        import fusioninfer_amd.distributed.parallel_state as ps
+
+        mc = self.cfg.model
+        self.num_gpu_blocks = self.profile_num_blocks()
+        kv_heads = mc.num_kv_heads // ps.tp_world_size()
+        shape = (self.num_gpu_blocks, kv_heads, self.block_size, mc.head_dim)
+        self.kv_caches = [
+            (
+                torch.zeros(shape, dtype=torch.bfloat16, device=self.device),
+                torch.zeros(shape, dtype=torch.bfloat16, device=self.device),
+            )
+            for _ in range(mc.num_layers)
+        ]
+
+    # ------------------------------------------------------------ prefill
+    def _prepare_prefill(self, seqs: List[Sequence], bm: BlockManager):
+        input_ids: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+        cu = [0]
+        lens = []
+        for seq in seqs:
+            toks = seq.all_token_ids
+            L = len(toks)
+            input_ids.extend(toks)
+            positions.extend(range(L))
+            slots.extend(bm.slot_for(seq, p) for p in range(L))
+            cu.append(cu[-1] + L)
+            lens.append(L)
+        dev = self.device
+        tile_seq, tile_row0 = ops_mod.build_prefill_tiles(lens, device=dev)
+        meta = AttnMetadata(
+            num_prefill_tokens=cu[-1],
+            num_decode_tokens=0,
+            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+            slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
+            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+            tile_seq=tile_seq,
+            tile_row0=tile_row0,
+        )
+        ids = torch.tensor(input_ids, dtype=torch.long, device=dev)
+        logits_idx = torch.tensor(
+            [c - 1 for c in cu[1:]], dtype=torch.long, device=dev
+        )
+        return ids, meta, logits_idx
+
+    def execute_prefill(self, seqs: List[Sequence], bm: BlockManager):
+        ids, meta, logits_idx = self._prepare_prefill(seqs, bm)
+        with torch.no_grad():
+            hidden = self.model(ids, meta, self.kv_caches)
+            logits = self.model.compute_logits(hidden[logits_idx])
+        return logits
+
+    # ------------------------------------------------------------ decode
+    def _alloc_static(self, max_bs: int) -> None:
+        dev = self.device
+        self._static = {
+            "ids": torch.zeros(max_bs, dtype=torch.long, device=dev),
+            "positions": torch.zeros(max_bs, dtype=torch.int32, device=dev),
+            "slots": torch.full((max_bs,), -1, dtype=torch.int32, device=dev),
+            "block_tables": torch.zeros(
+                (max_bs, self.max_blocks_per_seq), dtype=torch.int32, device=dev
+            ),
+            "seq_lens": torch.ones(max_bs, dtype=torch.int32, device=dev),
+        }
+
+    def _decode_forward(self, bs: int) -> torch.Tensor:
+        s = self._static
+        meta = AttnMetadata(
+            num_prefill_tokens=0,
+            num_decode_tokens=bs,
+            positions=s["positions"][:bs],
+            slot_mapping=s["slots"][:bs],
+            block_tables=s["block_tables"][:bs],
+            seq_lens=s["seq_lens"][:bs],
+        )
+        hidden = self.model(s["ids"][:bs], meta, self.kv_caches)
+        return self.model.compute_logits(hidden)
+
+    def capture_decode_graphs(self) -> None:
+        if not self.is_cuda or self.cfg.enforce_eager:
+            return
+        max_bs = min(self.cfg.scheduler.max_num_seqs, _DECODE_BUCKETS[-1])
+        buckets = [b for b in _DECODE_BUCKETS if b <= max_bs]
+        if not self._static:
+            self._alloc_static(max(buckets))
+        torch.cuda.synchronize()
+        with torch.no_grad():
+            for bs in reversed(buckets):  # largest first shares the pool best
+                self._decode_forward(bs)  # warm-up (workspaces, autotuners)
+                torch.cuda.synchronize()
+                g = torch.cuda.CUDAGraph()
+                if self._graph_pool is None:
+                    with torch.cuda.graph(g):
+                        out = self._decode_forward(bs)
+                    self._graph_pool = g.pool()
+                else:
+                    with torch.cuda.graph(g, pool=self._graph_pool):
+                        out = self._decode_forward(bs)
+                self._graphs[bs] = (g, out)
+        torch.cuda.synchronize()
+
+    def _fill_decode_inputs(self, seqs: List[Sequence], bm: BlockManager,
+                            bs: int) -> None:
+        n = len(seqs)
+        ids = [s.all_token_ids[-1] for s in seqs]
+        pos = [s.num_tokens - 1 for s in seqs]
+        slots = [bm.slot_for(s, s.num_tokens - 1) for s in seqs]
+        lens = [s.num_tokens for s in seqs]
+        st = self._static
+        dev = self.device
+        st["ids"][:n].copy_(
+            torch.tensor(ids, dtype=torch.long), non_blocking=True
+        )
+        st["positions"][:n].copy_(
+            torch.tensor(pos, dtype=torch.int32), non_blocking=True
+        )
+        st["slots"][:n].copy_(
+            torch.tensor(slots, dtype=torch.int32), non_blocking=True
+        )
+        st["seq_lens"][:n].copy_(
+            torch.tensor(lens, dtype=torch.int32), non_blocking=True
+        )
+        bt = torch.zeros((n, self.max_blocks_per_seq), dtype=torch.int32)
+        for i, s in enumerate(seqs):
+            bt[i, : len(s.block_ids)] = torch.tensor(
+                s.block_ids, dtype=torch.int32
+            )
+        st["block_tables"][:n].copy_(bt.to(dev), non_blocking=True)
+        if bs > n:  # padding rows: decode block 0, pos 0, len 1, slot -1
+            st["ids"][n:bs].zero_()
+            st["positions"][n:bs].zero_()
+            st["slots"][n:bs].fill_(-1)
+            st["seq_lens"][n:bs].fill_(1)
+            st["block_tables"][n:bs].zero_()
+
+    def execute_decode(self, seqs: List[Sequence], bm: BlockManager):
+        n = len(seqs)
+        if not self._static:
+            self._alloc_static(max(self.cfg.scheduler.max_num_seqs, n))
+        bucket = next((b for b in _DECODE_BUCKETS if b >= n), None)
+        use_graph = (
+            bucket is not None and bucket in self._graphs and self.is_cuda
+        )
+        bs = bucket if use_graph else n
+        self._fill_decode_inputs(seqs, bm, bs)
+        if use_graph:
+            g, out = self._graphs[bucket]
+            g.replay()
+            return out[:n]
+        with torch.no_grad():
+            return self._decode_forward(n)[:n]

@@ -1,0 +1,117 @@
+"""Continuous-batching scheduler (token-level, vLLM-class policy).
+
+Step policy: prefill-priority. When prompts are waiting and blocks/budget
+allow, the step is a pure-prefill batch (bounded by max_num_batched_tokens
+and max_num_seqs); otherwise it is a pure-decode batch over all RUNNING
+sequences. Pure-decode steps have static shapes so the runner replays
+hipGraphs for them. Preemption-by-recompute frees the newest sequence
+when the cache runs out of blocks mid-decode.
+
+Capability parity: the "continuous-batching scheduler + paged KV cache"
+the reference delegates to vLLM and whose kv-util / queue-depth metrics
+its EPP scorers consume (reference pkg/router/
+strategy.go:70-98).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from collections import deque
+from typing import Deque, List
+
+from fusioninfer_amd.config import SchedulerConfig
+from fusioninfer_amd.engine.block_manager import BlockManager
+from fusioninfer_amd.engine.sequence import Sequence, SeqStatus
+
+
+@dataclasses.dataclass
+class ScheduledBatch:
+    prefill_seqs: List[Sequence]
+    decode_seqs: List[Sequence]
+    preempted: List[Sequence]
+
+    @property
+    def is_empty(self) -> bool:
+        return not self.prefill_seqs and not self.decode_seqs
+
+    @property
+    def is_prefill(self) -> bool:
+        return bool(self.prefill_seqs)
+
+
+class Scheduler:
+    def __init__(self, cfg: SchedulerConfig, block_manager: BlockManager):
+        self.cfg = cfg
+        self.bm = block_manager
+        self.waiting: Deque[Sequence] = deque()
+        self.running: List[Sequence] = []
+
+    # ----------------------------------------------------------- queue ops
+    def add(self, seq: Sequence) -> None:
+        self.waiting.append(seq)
+
+    @property
+    def num_waiting(self) -> int:
+        return len(self.waiting)
+
+    @property
+    def num_running(self) -> int:
+        return len(self.running)
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    # ----------------------------------------------------------- schedule
+    def schedule(self) -> ScheduledBatch:
+        preempted: List[Sequence] = []
+
+        # try to build a prefill batch
+        prefill: List[Sequence] = []
+        budget = self.cfg.max_num_batched_tokens
+        while (
+            self.waiting
+            and len(self.running) + len(prefill) < self.cfg.max_num_seqs
+        ):
+            seq = self.waiting[0]
+            if seq.num_prompt_tokens > budget:
+                break
+            if not self.bm.can_allocate(seq.num_prompt_tokens):
+                break
+            self.waiting.popleft()
+            self.bm.allocate(seq)
+            seq.status = SeqStatus.RUNNING
+            prefill.append(seq)
+            budget -= seq.num_prompt_tokens
+        if prefill:
+            self.running.extend(prefill)
+            return ScheduledBatch(prefill, [], preempted)
+
+        # decode step: every running sequence generates one token
+        decode: List[Sequence] = []
+        for seq in list(self.running):
+            if seq.status != SeqStatus.RUNNING:
+                continue  # preempted earlier in this same pass
+            if not self.bm.can_append_slot(seq):
+                victim = self._preempt_newest()
+                preempted.append(victim)
+                if victim is seq:
+                    continue
+            self.bm.append_slot(seq)
+            decode.append(seq)
+        return ScheduledBatch([], decode, preempted)
+
+    def _preempt_newest(self) -> Sequence:
+        victim = self.running.pop()  # newest
+        self.bm.free(victim)
+        victim.status = SeqStatus.PREEMPTED
+        # recompute: generated tokens become part of the prompt
+        victim.prompt_token_ids.extend(victim.output_token_ids)
+        victim.output_token_ids = []
+        self.waiting.appendleft(victim)
+        return victim
+
+    # ----------------------------------------------------------- finish
+    def finish(self, seq: Sequence) -> None:
+        seq.status = SeqStatus.FINISHED
+        self.bm.free(seq)
+        self.running.remove(seq)

@@ -1,0 +1,86 @@
+"""Request/sequence state for the continuous-batching engine."""
+
+from __future__ import annotations
+
+import enum
+import time
+from typing import List, Optional
+
+
+class SeqStatus(enum.Enum):
+    WAITING = "waiting"
+    RUNNING = "running"
+    PREEMPTED = "preempted"
+    FINISHED = "finished"
+
+
+class SamplingParams:
+    def __init__(
+        self,
+        max_tokens: int = 128,
+        temperature: float = 0.0,
+        top_p: float = 1.0,
+        ignore_eos: bool = True,
+        stop_token_ids: Optional[List[int]] = None,
+    ):
+        self.max_tokens = max_tokens
+        self.temperature = temperature
+        self.top_p = top_p
+        self.ignore_eos = ignore_eos
+        self.stop_token_ids = stop_token_ids or []
+
+
+class Sequence:
+    """One request = one sequence (no beam/parallel sampling in v1)."""
+
+    def __init__(self, seq_id: str, prompt_token_ids: List[int],
+                 sampling: SamplingParams):
+        self.seq_id = seq_id
+        self.prompt_token_ids = list(prompt_token_ids)
+        self.output_token_ids: List[int] = []
+        self.sampling = sampling
+        self.status = SeqStatus.WAITING
+        self.block_ids: List[int] = []
+        self.arrival_time = time.monotonic()
+        self.first_token_time: Optional[float] = None
+        self.finish_time: Optional[float] = None
+        # number of prompt tokens whose KV was satisfied by prefix cache
+        self.num_cached_tokens = 0
+
+    @property
+    def num_prompt_tokens(self) -> int:
+        return len(self.prompt_token_ids)
+
+    @property
+    def num_tokens(self) -> int:
+        return len(self.prompt_token_ids) + len(self.output_token_ids)
+
+    @property
+    def all_token_ids(self) -> List[int]:
+        return self.prompt_token_ids + self.output_token_ids
+
+    def append_token(self, token_id: int) -> None:
+        if self.first_token_time is None:
+            self.first_token_time = time.monotonic()
+        self.output_token_ids.append(token_id)
+
+    def is_finished(self) -> bool:
+        return self.status == SeqStatus.FINISHED
+
+    def check_stop(self) -> bool:
+        s = self.sampling
+        if len(self.output_token_ids) >= s.max_tokens:
+            return True
+        if (
+            not s.ignore_eos
+            and self.output_token_ids
+            and self.output_token_ids[-1] in s.stop_token_ids
+        ):
+            return True
+        return False
+
+    @property
+    def ttft(self) -> Optional[float]:
+        if self.first_token_time is None:
+            return None
+        return self.first_token_time - self.arrival_time

@@ -1,0 +1,195 @@
+"""Decoder-only transformer (Llama/Qwen3 family) on the FusionInfer-AMD ops.
+
+Every hot op is a hand-written CDNA4 HIP kernel (fusioninfer_amd/ops);
+GEMMs are hipBLASLt via F.linear. TP sharding is Megatron-style: heads
+sharded for attention, intermediate sharded for the MLP, one RCCL
+all-reduce after o_proj and one after down_proj per layer.
+
+Capability parity: the model execution the reference delegates to
+vLLM images (`vllm serve --model Qwen/Qwen3-8B`, reference README.md:136-148).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+
+import fusioninfer_amd.ops as ops
+from fusioninfer_amd.config import ModelConfig
+from fusioninfer_amd.distributed import parallel_state as ps
+from fusioninfer_amd.distributed.layers import (
+    ColumnParallelLinear,
+    ReplicatedLinear,
+    RowParallelLinear,
+)
+from fusioninfer_amd.engine.metadata import AttnMetadata
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        tp = ps.tp_world_size()
+        assert cfg.num_heads % tp == 0 and cfg.num_kv_heads % tp == 0, (
+            "head counts must divide TP size"
+        )
+        self.num_heads = cfg.num_heads // tp
+        self.num_kv_heads = cfg.num_kv_heads // tp
+        self.head_dim = cfg.head_dim
+        self.q_size = self.num_heads * self.head_dim
+        self.kv_size = self.num_kv_heads * self.head_dim
+        self.scale = 1.0 / math.sqrt(self.head_dim)
+        self.eps = cfg.rms_norm_eps
+        hidden = cfg.hidden_size
+        self.qkv_proj = ColumnParallelLinear(
+            hidden, (cfg.num_heads + 2 * cfg.num_kv_heads) * cfg.head_dim
+        )
+        self.o_proj = RowParallelLinear(cfg.num_heads * cfg.head_dim, hidden)
+        if cfg.qk_norm:
+            self.q_norm_weight = nn.Parameter(
+                torch.ones(cfg.head_dim, dtype=torch.bfloat16), requires_grad=False
+            )
+            self.k_norm_weight = nn.Parameter(
+                torch.ones(cfg.head_dim, dtype=torch.bfloat16), requires_grad=False
+            )
+        else:
+            self.q_norm_weight = None
+            self.k_norm_weight = None
+
+    def forward(
+        self,
+        x: torch.Tensor,
+        meta: AttnMetadata,
+        kv_cache,  # (k_cache, v_cache) [B, Hk, bs, D]
+        cos_sin: torch.Tensor,
+    ) -> torch.Tensor:
+        qkv = self.qkv_proj(x)
+        q = qkv[:, : self.q_size]
+        k = qkv[:, self.q_size : self.q_size + self.kv_size]
+        v = qkv[:, self.q_size + self.kv_size :]
+        ops.rope_qk_norm_(
+            q, k, meta.positions, cos_sin,
+            self.num_heads, self.num_kv_heads, self.head_dim,
+            self.q_norm_weight, self.k_norm_weight, self.eps,
+        )
+        k_cache, v_cache = kv_cache
+        ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
+
+        T = x.shape[0]
+        np_, nd = meta.num_prefill_tokens, meta.num_decode_tokens
+        out = torch.empty(T, self.q_size, dtype=x.dtype, device=x.device)
+        if np_ > 0:
+            o = ops.prefill_attention(
+                q[:np_].view(np_, self.num_heads, self.head_dim),
+                k[:np_].reshape(np_, self.num_kv_heads, self.head_dim),
+                v[:np_].reshape(np_, self.num_kv_heads, self.head_dim),
+                meta.cu_seqlens,
+                self.scale,
+                tile_seq=meta.tile_seq,
+                tile_row0=meta.tile_row0,
+            )
+            out[:np_] = o.view(np_, self.q_size)
+        if nd > 0:
+            o = ops.paged_attention_decode(
+                q[np_:].view(nd, self.num_heads, self.head_dim),
+                k_cache,
+                v_cache,
+                meta.block_tables,
+                meta.seq_lens,
+                self.scale,
+            )
+            out[np_:] = o.view(nd, self.q_size)
+        return self.o_proj(out)
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.gate_up_proj = ColumnParallelLinear(
+            cfg.hidden_size, 2 * cfg.intermediate_size
+        )
+        self.down_proj = RowParallelLinear(cfg.intermediate_size, cfg.hidden_size)
+        self.inter_per_rank = cfg.intermediate_size // ps.tp_world_size()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        gu = self.gate_up_proj(x)
+        # column-parallel packs [gate_shard | up_shard] per rank already
+        return self.down_proj(ops.silu_and_mul(gu))
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.input_norm_weight = nn.Parameter(
+            torch.ones(cfg.hidden_size, dtype=torch.bfloat16), requires_grad=False
+        )
+        self.post_norm_weight = nn.Parameter(
+            torch.ones(cfg.hidden_size, dtype=torch.bfloat16), requires_grad=False
+        )
+        self.self_attn = Attention(cfg)
+        self.mlp = MLP(cfg)
+        self.eps = cfg.rms_norm_eps
+
+    def forward(self, hidden, residual, meta, kv_cache, cos_sin):
+        if residual is None:
+            residual = hidden.clone()
+            hidden = ops.rms_norm(hidden, self.input_norm_weight, self.eps)
+        else:
+            hidden, residual = ops.fused_add_rms_norm(
+                hidden, residual, self.input_norm_weight, self.eps
+            )
+        hidden = self.self_attn(hidden, meta, kv_cache, cos_sin)
+        hidden, residual = ops.fused_add_rms_norm(
+            hidden, residual, self.post_norm_weight, self.eps
+        )
+        hidden = self.mlp(hidden)
+        return hidden, residual
+
+
+class CausalLM(nn.Module):
+    """The full model: embedding -> N decoder layers -> final norm -> lm_head."""
+
+    def __init__(self, cfg: ModelConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed_tokens = nn.Parameter(
+            torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16)
+            .normal_(0.0, 0.02),
+            requires_grad=False,
+        )
+        self.layers = nn.ModuleList(
+            [DecoderLayer(cfg) for _ in range(cfg.num_layers)]
+        )
+        self.final_norm_weight = nn.Parameter(
+            torch.ones(cfg.hidden_size, dtype=torch.bfloat16), requires_grad=False
+        )
+        if cfg.tie_word_embeddings:
+            self.lm_head = None
+        else:
+            self.lm_head = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size)
+        cos_sin = ops.compute_cos_sin_cache(
+            cfg.head_dim, cfg.max_position_embeddings, cfg.rope_theta
+        )
+        self.register_buffer("cos_sin", cos_sin, persistent=False)
+
+    def forward(
+        self,
+        input_ids: torch.Tensor,     # [T] int
+        meta: AttnMetadata,
+        kv_caches,                   # list of (k_cache, v_cache) per layer
+    ) -> torch.Tensor:
+        hidden = self.embed_tokens[input_ids]
+        residual = None
+        for layer, kv in zip(self.layers, kv_caches):
+            hidden, residual = layer(hidden, residual, meta, kv, self.cos_sin)
+        hidden, _ = ops.fused_add_rms_norm(
+            hidden, residual, self.final_norm_weight, self.cfg.rms_norm_eps
+        )
+        return hidden
+
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        """hidden: [S, H] (already gathered to the sampled rows)."""
+        if self.lm_head is None:
+            return hidden @ self.embed_tokens.T
+        return self.lm_head(hidden)
