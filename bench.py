@@ -1,0 +1,221 @@
+#!/usr/bin/env python3
+"""FusionInfer-AMD flagship serving benchmark.
+
+Measures the BASELINE.json metric — goodput (completed requests/s) and p50
+TTFT for Qwen3-8B serving on MI355X — on synthetic requests with
+random-init weights (no network in this environment; BASELINE.md documents
+that the reference publishes no numbers of its own).
+
+One rank per GPU (DP replicas, weak scaling — the reference's `replicas`
+orchestration). Each rank runs a closed-loop load: target
+concurrency is kept topped up; a bench "step" is one engine iteration
+(one continuous-batching scheduler step + forward). Timed region:
+barrier + torch.cuda.synchronize on both sides of EXACTLY --steps steps.
+
+Usage:
+  python bench.py --gpus 1 --steps 64 --warmup 16
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=64)
+    p.add_argument("--warmup", type=int, default=16)
+    p.add_argument("--model", type=str, default="Qwen3-8B")
+    p.add_argument("--prompt-len", type=int, default=1024)
+    p.add_argument("--gen-len", type=int, default=128)
+    p.add_argument("--concurrency", type=int, default=256)
+    p.add_argument("--max-batched-tokens", type=int, default=8192)
+    p.add_argument("--enforce-eager", action="store_true")
+    p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree")
+    return p.parse_args()
+
+
+class ClosedLoopLoad:
+    """Keeps `concurrency` requests in flight; tracks completions + TTFT."""
+
+    def __init__(self, engine, prompt_len: int, gen_len: int, concurrency: int,
+                 vocab: int, seed: int):
+        from fusioninfer_amd.engine.sequence import SamplingParams
+
+        self.engine = engine
+        self.params = SamplingParams(max_tokens=gen_len, temperature=0.0)
+        self.prompt_len = prompt_len
+        self.concurrency = concurrency
+        self.vocab = vocab
+        self.rng = torch.Generator().manual_seed(seed)
+        self.completion_times = []
+        self.first_token_times = {}
+        self.ttfts = []
+
+    def _new_prompt(self):
+        return torch.randint(
+            0, self.vocab, (self.prompt_len,), generator=self.rng
+        ).tolist()
+
+    def top_up(self):
+        in_flight = self.engine.num_waiting() + self.engine.num_running()
+        for _ in range(self.concurrency - in_flight):
+            self.engine.add_request(self._new_prompt(), self.params)
+
+    def step(self):
+        outs = self.engine.step()
+        now = time.monotonic()
+        for o in outs:
+            if o.finished:
+                self.completion_times.append(now)
+                if o.ttft is not None:
+                    self.ttfts.append(o.ttft)
+        self.top_up()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_cuda = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if use_cuda else "cpu"
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+
+    import torch.distributed as dist
+
+    distributed = world > 1
+    if distributed:
+        dist.init_process_group(backend="nccl" if use_cuda else "gloo")
+
+    from fusioninfer_amd.config import (
+        CacheConfig,
+        EngineConfig,
+        ParallelConfig,
+        SchedulerConfig,
+    )
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.models.registry import get_model_config
+
+    mc = get_model_config(args.model)
+    max_len = args.prompt_len + args.gen_len + 64
+    cfg = EngineConfig(
+        model=mc,
+        cache=CacheConfig(
+            num_gpu_blocks=None if use_cuda else 4096,
+            gpu_memory_utilization=0.85,
+        ),
+        scheduler=SchedulerConfig(
+            max_num_seqs=args.concurrency,
+            max_num_batched_tokens=args.max_batched_tokens,
+            max_model_len=max_len,
+        ),
+        parallel=ParallelConfig(
+            tensor_parallel_size=args.tp, rank=rank, world_size=world
+        ),
+        seed=1234 + rank,
+        enforce_eager=args.enforce_eager,
+    )
+    engine = LLMEngine(cfg, device=device)
+    load = ClosedLoopLoad(
+        engine, args.prompt_len, args.gen_len, args.concurrency,
+        mc.vocab_size, seed=99 + rank,
+    )
+    load.top_up()
+
+    def barrier_sync():
+        if distributed:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    # Load initialization: ramp the closed loop to steady state (prefills
+    # admitted, first completions observed) so the timed window measures
+    # steady-state serving for ANY --steps/--warmup. Bounded.
+    ramp_limit = 3 * (args.prompt_len // 256 + args.gen_len) + 500
+    ramp = 0
+    while not load.completion_times and ramp < ramp_limit:
+        load.step()
+        ramp += 1
+
+    for _ in range(args.warmup):
+        load.step()
+
+    barrier_sync()
+    t0 = time.monotonic()
+    completed_before = len(load.completion_times)
+    tokens_before = engine.num_generated_tokens
+    ttft_mark = len(load.ttfts)
+    for _ in range(args.steps):
+        load.step()
+    barrier_sync()
+    elapsed = time.monotonic() - t0
+
+    completed = len(load.completion_times) - completed_before
+    gen_tokens = engine.num_generated_tokens - tokens_before
+    ttfts = load.ttfts[ttft_mark:] or load.ttfts
+
+    if distributed:
+        t = torch.tensor(
+            [elapsed], dtype=torch.float64,
+            device=device if use_cuda else "cpu",
+        )
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        c = torch.tensor(
+            [completed, gen_tokens], dtype=torch.float64,
+            device=device if use_cuda else "cpu",
+        )
+        dist.all_reduce(c)
+        completed, gen_tokens = float(c[0].item()), float(c[1].item())
+        all_ttfts = [None] * world
+        dist.all_gather_object(all_ttfts, ttfts)
+        ttfts = [x for lst in all_ttfts for x in lst]
+
+    goodput = completed / elapsed
+    p50_ttft_ms = (
+        statistics.median(ttfts) * 1000.0 if ttfts else float("nan")
+    )
+    if rank == 0:
+        result = {
+            "metric": "goodput req/s (Qwen3-8B serving, closed-loop)",
+            "value": round(goodput, 3),
+            "unit": "req/s",
+            "n_gpus": world if distributed else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "p50_ttft_ms": round(p50_ttft_ms, 1),
+            "tokens_per_s": round(gen_tokens / elapsed, 1),
+            "config": {
+                "model": args.model,
+                "global_batch": args.concurrency * (world if distributed else 1),
+                "seq_len": args.prompt_len + args.gen_len,
+                "prompt_len": args.prompt_len,
+                "gen_len": args.gen_len,
+                "parallelism": f"dp{world if distributed else args.gpus}"
+                + (f"-tp{args.tp}" if args.tp > 1 else ""),
+            },
+        }
+        print(json.dumps(result))
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
