@@ -49,10 +49,8 @@ class ClosedLoopLoad:
 
     def __init__(self, engine, prompt_len: int, gen_len: int, concurrency: int,
                  vocab: int, seed: int):
-        from fusioninfer_amd.engine.sequence import SamplingParams
-
         self.engine = engine
-        self.params = SamplingParams(max_tokens=gen_len, temperature=0.0)
+        self.gen_len = gen_len
         self.prompt_len = prompt_len
         self.concurrency = concurrency
         self.vocab = vocab
@@ -66,10 +64,19 @@ class ClosedLoopLoad:
             0, self.vocab, (self.prompt_len,), generator=self.rng
         ).tolist()
 
+    def _new_params(self):
+        from fusioninfer_amd.engine.sequence import SamplingParams
+
+        # gen lengths uniform in [gen_len/2, 3*gen_len/2] (mean = gen_len):
+        # staggers completions so goodput is measurable in short windows
+        lo, hi = max(self.gen_len // 2, 1), self.gen_len + self.gen_len // 2
+        n = int(torch.randint(lo, hi + 1, (1,), generator=self.rng))
+        return SamplingParams(max_tokens=n, temperature=0.0)
+
     def top_up(self):
         in_flight = self.engine.num_waiting() + self.engine.num_running()
         for _ in range(self.concurrency - in_flight):
-            self.engine.add_request(self._new_prompt(), self.params)
+            self.engine.add_request(self._new_prompt(), self._new_params())
 
     def step(self):
         outs = self.engine.step()
@@ -142,9 +149,10 @@ def main():
     # Load initialization: ramp the closed loop to steady state (prefills
     # admitted, first completions observed) so the timed window measures
     # steady-state serving for ANY --steps/--warmup. Bounded.
-    ramp_limit = 3 * (args.prompt_len // 256 + args.gen_len) + 500
+    ramp_limit = 6 * (args.prompt_len // 256 + args.gen_len) + 500
+    ramp_target = max(args.concurrency // 8, 4)
     ramp = 0
-    while not load.completion_times and ramp < ramp_limit:
+    while len(load.completion_times) < ramp_target and ramp < ramp_limit:
         load.step()
         ramp += 1
 
