@@ -1,0 +1,233 @@
+"""InferenceService reconciler.
+
+Single-pass reconcile mirroring the reference controller
+(pkg/controller/inferenceservice_controller.go:66-156):
+fetch -> init condition -> PodGroup -> per-role LWS (+ orphan cleanup) ->
+router stack -> in-memory status aggregation -> conditions -> ONE status
+update. Every create-or-update is gated on the spec-hash label
+(:191-197, :255-262, :406-410), so metadata-only changes are no-ops.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional
+
+from fusioninfer_amd.controlplane import conditions as cond
+from fusioninfer_amd.controlplane import router as router_mod
+from fusioninfer_amd.controlplane import scheduling as sched
+from fusioninfer_amd.controlplane import workload
+from fusioninfer_amd.controlplane.api import (
+    PHASE_DEPLOYING,
+    PHASE_PENDING,
+    PHASE_RUNNING,
+    InferenceService,
+    Role,
+)
+from fusioninfer_amd.controlplane.fake import FakeClient, NotFoundError
+from fusioninfer_amd.controlplane.workload import (
+    LABEL_ROLE,
+    LABEL_SERVICE,
+    LABEL_SPEC_HASH,
+)
+
+
+class InferenceServiceReconciler:
+    def __init__(self, client: FakeClient):
+        self.client = client
+
+    # --------------------------------------------------------------- util
+    def _create_or_update(self, desired: Dict[str, Any]) -> bool:
+        """Create if missing; update only when the spec-hash label changed.
+        Returns True if a write happened."""
+        md = desired["metadata"]
+        existing = self.client.try_get(
+            desired["kind"], md["name"], md.get("namespace", "default")
+        )
+        if existing is None:
+            self.client.create(desired)
+            return True
+        old_hash = existing["metadata"].get("labels", {}).get(LABEL_SPEC_HASH)
+        new_hash = md.get("labels", {}).get(LABEL_SPEC_HASH)
+        if new_hash is not None and old_hash == new_hash:
+            return False
+        desired = dict(desired)
+        desired["metadata"] = dict(md)
+        desired["metadata"]["resourceVersion"] = existing["metadata"].get(
+            "resourceVersion"
+        )
+        self.client.update(desired)
+        return True
+
+    # ---------------------------------------------------------- reconcile
+    def reconcile(self, name: str, namespace: str = "default") -> Optional[Dict]:
+        obj = self.client.try_get("InferenceService", name, namespace)
+        if obj is None:
+            return None  # deleted; children are garbage-collected via owner refs
+        svc = InferenceService.from_dict(obj)
+        status = dict(svc.status)
+
+        if not cond.has_condition(status, cond.COND_INITIALIZED):
+            cond.set_init_condition(status, svc.generation)
+
+        errors: List[str] = []
+        try:
+            self._reconcile_podgroup(svc)
+        except Exception as e:  # pragma: no cover
+            errors.append(f"podgroup: {e}")
+
+        for role in svc.worker_roles():
+            try:
+                self._reconcile_lws(svc, role)
+                self._cleanup_orphan_lws(svc, role)
+            except Exception as e:
+                errors.append(f"lws[{role.component_type}]: {e}")
+
+        for role in svc.router_roles():
+            try:
+                self._reconcile_router(svc, role)
+            except Exception as e:
+                errors.append(f"router: {e}")
+
+        components = self._aggregate_status(svc)
+        status["components"] = components
+
+        if errors:
+            cond.set_failed_condition(status, svc.generation, "; ".join(errors))
+        elif self._all_ready(svc, components):
+            cond.set_active_condition(status, svc.generation)
+        else:
+            cond.set_processing_condition(status, svc.generation)
+
+        obj["status"] = status
+        self.client.update_status(obj)
+        return status
+
+    # ----------------------------------------------------------- podgroup
+    def _reconcile_podgroup(self, svc: InferenceService) -> None:
+        if not sched.needs_gang_scheduling(svc):
+            return
+        pg = sched.build_podgroup(svc)
+        self._set_owner(pg, svc)
+        self._create_or_update(pg)
+
+    # ---------------------------------------------------------------- lws
+    def _reconcile_lws(self, svc: InferenceService, role: Role) -> None:
+        gang = sched.needs_gang_scheduling_for_role(svc, role)
+        for i in range(role.replicas):
+            lws = workload.build_lws(
+                svc,
+                role,
+                i,
+                gang_scheduled=gang,
+                podgroup_name=sched.podgroup_name(svc) if gang else None,
+                task_name=sched.task_name(role, i),
+            )
+            self._set_owner(lws, svc)
+            self._create_or_update(lws)
+
+    def _cleanup_orphan_lws(self, svc: InferenceService, role: Role) -> None:
+        """Delete per-replica LWS beyond the desired replica count
+        (reference :275-310)."""
+        expected = set(workload.list_expected_lws_names(svc, role))
+        existing = self.client.list(
+            "LeaderWorkerSet",
+            svc.namespace,
+            label_selector={
+                LABEL_SERVICE: svc.name,
+                LABEL_ROLE: role.component_type,
+            },
+        )
+        for obj in existing:
+            n = obj["metadata"]["name"]
+            if n not in expected:
+                self.client.delete("LeaderWorkerSet", n, svc.namespace)
+
+    # -------------------------------------------------------------- router
+    def _reconcile_router(self, svc: InferenceService, role: Role) -> None:
+        """8 sequential create-or-updates (reference :321-358)."""
+        for builder in (
+            router_mod.build_epp_service_account,
+            router_mod.build_epp_role,
+            router_mod.build_epp_role_binding,
+        ):
+            obj = builder(svc)
+            self._set_owner(obj, svc)
+            self._create_or_update_unhashed(obj)
+        for obj in (
+            router_mod.build_epp_configmap(svc, role),
+            router_mod.build_epp_deployment(svc),
+            router_mod.build_epp_service(svc),
+            router_mod.build_inference_pool(svc),
+            router_mod.build_httproute(svc, role),
+        ):
+            self._set_owner(obj, svc)
+            self._create_or_update(obj)
+
+    def _create_or_update_unhashed(self, desired: Dict[str, Any]) -> None:
+        md = desired["metadata"]
+        if self.client.try_get(
+            desired["kind"], md["name"], md.get("namespace", "default")
+        ) is None:
+            self.client.create(desired)
+
+    # -------------------------------------------------------------- status
+    def _aggregate_status(self, svc: InferenceService) -> Dict[str, Any]:
+        """Per-role aggregation over per-replica LWS (reference :639-686)."""
+        components: Dict[str, Any] = {}
+        for role in svc.worker_roles():
+            nodes = role.node_count()
+            ready_replicas = 0
+            ready_pods = 0
+            found = 0
+            for i in range(role.replicas):
+                lws = self.client.try_get(
+                    "LeaderWorkerSet",
+                    workload.generate_lws_name(svc.name, role, i),
+                    svc.namespace,
+                )
+                if lws is None:
+                    continue
+                found += 1
+                st = lws.get("status") or {}
+                rr = int(st.get("readyReplicas", 0) or 0)
+                if rr >= 1:
+                    ready_replicas += 1
+                    ready_pods += nodes
+            if ready_replicas == role.replicas and role.replicas > 0:
+                phase = PHASE_RUNNING
+            elif ready_replicas > 0 or found > 0:
+                phase = PHASE_DEPLOYING
+            else:
+                phase = PHASE_PENDING
+            components[role.component_type] = {
+                "componentType": role.component_type,
+                "replicas": role.replicas,
+                "readyReplicas": ready_replicas,
+                "readyPods": ready_pods,
+                "phase": phase,
+            }
+        return components
+
+    @staticmethod
+    def _all_ready(svc: InferenceService, components: Dict[str, Any]) -> bool:
+        workers = svc.worker_roles()
+        if not workers:
+            return False
+        return all(
+            components.get(r.component_type, {}).get("phase") == PHASE_RUNNING
+            for r in workers
+        )
+
+    # ---------------------------------------------------------------- misc
+    @staticmethod
+    def _set_owner(obj: Dict[str, Any], svc: InferenceService) -> None:
+        obj["metadata"].setdefault("ownerReferences", []).append(
+            {
+                "apiVersion": "fusioninfer.io/v1alpha1",
+                "kind": "InferenceService",
+                "name": svc.name,
+                "uid": svc.uid,
+                "controller": True,
+                "blockOwnerDeletion": True,
+            }
+        )
