@@ -1,0 +1,102 @@
+"""Engine server entrypoint: `python -m fusioninfer_amd.server ...`
+
+CLI surface mirrors the engine flags the reference controller passes via
+container args (--model / --tensor-parallel-size /
+--kv-transfer-config '{"kv_connector":...,"kv_role":...}' / --max-model-len
+/ port 8000), plus the rendezvous flags the control plane's LWS wrapper
+appends for multi-node roles (--nnodes/--node-rank/--nproc-per-node/
+--master-addr/--master-port, consuming LWS_LEADER_ADDRESS).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+
+
+def parse_args(argv=None):
+    p = argparse.ArgumentParser("fusioninfer-amd engine server")
+    p.add_argument("--model", default="Qwen3-8B")
+    p.add_argument("--host", default="0.0.0.0")
+    p.add_argument("--port", type=int, default=8000)
+    p.add_argument("--tensor-parallel-size", type=int, default=1)
+    p.add_argument("--pipeline-parallel-size", type=int, default=1)
+    p.add_argument("--data-parallel-size", type=int, default=1)
+    p.add_argument("--max-model-len", type=int, default=8192)
+    p.add_argument("--max-num-seqs", type=int, default=256)
+    p.add_argument("--max-num-batched-tokens", type=int, default=8192)
+    p.add_argument("--gpu-memory-utilization", type=float, default=0.85)
+    p.add_argument("--kv-transfer-config", type=str, default=None,
+                   help='JSON: {"kv_connector": "RcclConnector", '
+                        '"kv_role": "kv_producer"|"kv_consumer"}')
+    p.add_argument("--enforce-eager", action="store_true")
+    p.add_argument("--enable-prefix-caching", action="store_true")
+    # multi-node rendezvous flags injected by the LWS wrapper
+    p.add_argument("--nnodes", type=int, default=1)
+    p.add_argument("--node-rank", type=int, default=0)
+    p.add_argument("--nproc-per-node", type=int, default=1)
+    p.add_argument("--master-addr", default=None)
+    p.add_argument("--master-port", type=int, default=29500)
+    return p.parse_args(argv)
+
+
+def build_engine_config(args):
+    from fusioninfer_amd.config import (
+        CacheConfig,
+        EngineConfig,
+        KVTransferConfig,
+        ParallelConfig,
+        SchedulerConfig,
+    )
+    from fusioninfer_amd.models.registry import get_model_config
+
+    kvt = KVTransferConfig()
+    if args.kv_transfer_config:
+        raw = json.loads(args.kv_transfer_config)
+        kvt = KVTransferConfig(
+            kv_connector=raw.get("kv_connector"),
+            kv_role=raw.get("kv_role"),
+            kv_rank=int(raw.get("kv_rank", 0)),
+            kv_world_size=int(raw.get("kv_world_size", 2)),
+        )
+    return EngineConfig(
+        model=get_model_config(args.model),
+        cache=CacheConfig(
+            gpu_memory_utilization=args.gpu_memory_utilization,
+            enable_prefix_caching=args.enable_prefix_caching,
+        ),
+        scheduler=SchedulerConfig(
+            max_num_seqs=args.max_num_seqs,
+            max_num_batched_tokens=args.max_num_batched_tokens,
+            max_model_len=args.max_model_len,
+        ),
+        parallel=ParallelConfig(
+            tensor_parallel_size=args.tensor_parallel_size,
+            pipeline_parallel_size=args.pipeline_parallel_size,
+            data_parallel_size=args.data_parallel_size,
+        ),
+        kv_transfer=kvt,
+        enforce_eager=args.enforce_eager,
+    )
+
+
+def main(argv=None):
+    args = parse_args(argv)
+    if args.master_addr:
+        os.environ.setdefault("MASTER_ADDR", args.master_addr)
+        os.environ.setdefault("MASTER_PORT", str(args.master_port))
+
+    import uvicorn
+
+    from fusioninfer_amd.server.api_server import build_app
+    from fusioninfer_amd.server.serving import ServingEngine
+
+    cfg = build_engine_config(args)
+    serving = ServingEngine(cfg)
+    app = build_app(serving, args.model)
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

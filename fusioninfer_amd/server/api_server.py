@@ -1,0 +1,233 @@
+"""OpenAI-compatible HTTP server on :8000.
+
+Capability parity with the engine surface the reference's InferencePool /
+HTTPRoute target (`vllm serve` on port 8000; targetPort
+pinned at reference pkg/router/inferencepool.go:31-32): /v1/completions,
+/v1/chat/completions (streaming + non-streaming), /health, /v1/models,
+and Prometheus /metrics with the vLLM-compatible gauge names the EPP
+scorers scrape (vllm:gpu_cache_usage_perc, vllm:num_requests_waiting —
+reference pkg/router/strategy.go:70-98 consumes these).
+
+Tokenization: there is no network access for tokenizer files, so string
+prompts use a reversible byte-level fallback; OpenAI's prompt-as-token-ids
+form is supported natively and is what the benchmark and EPP flows use.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+from typing import Any, AsyncGenerator, Dict, List, Optional, Union
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
+
+from fusioninfer_amd.engine.sequence import SamplingParams
+from fusioninfer_amd.server.serving import ServingEngine
+
+# ----------------------------------------------------------- tokenization
+
+
+def encode_prompt(prompt: Union[str, List[int]], vocab_size: int) -> List[int]:
+    if isinstance(prompt, list):
+        return [int(t) for t in prompt]
+    # byte-level fallback: offset so ids stay in-vocab and reversible
+    return [min(b + 3, vocab_size - 1) for b in prompt.encode("utf-8")]
+
+
+def decode_tokens(token_ids: List[int]) -> str:
+    try:
+        return bytes(max(t - 3, 0) & 0xFF for t in token_ids).decode(
+            "utf-8", errors="replace"
+        )
+    except Exception:
+        return " ".join(str(t) for t in token_ids)
+
+
+def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
+    app = FastAPI(title="fusioninfer-amd")
+    vocab = serving.engine.cfg.model.vocab_size
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok"}
+
+    @app.get("/v1/models")
+    async def models():
+        return {
+            "object": "list",
+            "data": [
+                {"id": model_name, "object": "model", "owned_by": "fusioninfer-amd"}
+            ],
+        }
+
+    @app.get("/metrics")
+    async def metrics():
+        m = serving.metrics()
+        lines = []
+        for name, mtype in [
+            ("gpu_cache_usage_perc", "gauge"),
+            ("num_requests_waiting", "gauge"),
+            ("num_requests_running", "gauge"),
+            ("generation_tokens_total", "counter"),
+            ("prompt_tokens_total", "counter"),
+            ("request_success_total", "counter"),
+        ]:
+            lines.append(f"# TYPE vllm:{name} {mtype}")
+            lines.append(
+                f'vllm:{name}{{model_name="{model_name}"}} {m[name]}'
+            )
+        return PlainTextResponse("\n".join(lines) + "\n")
+
+    def _sampling_from(body: Dict[str, Any]) -> SamplingParams:
+        return SamplingParams(
+            max_tokens=int(body.get("max_tokens", 16)),
+            temperature=float(body.get("temperature", 1.0)),
+            top_p=float(body.get("top_p", 1.0)),
+            ignore_eos=bool(body.get("ignore_eos", False)),
+            stop_token_ids=body.get("stop_token_ids") or [],
+        )
+
+    async def _collect(q) -> List[int]:
+        loop = asyncio.get_event_loop()
+        toks: List[int] = []
+        while True:
+            tok, finished = await loop.run_in_executor(None, q.get)
+            if tok is not None:
+                toks.append(tok)
+            if finished:
+                return toks
+
+    async def _stream(q) -> AsyncGenerator:
+        loop = asyncio.get_event_loop()
+        while True:
+            tok, finished = await loop.run_in_executor(None, q.get)
+            yield tok, finished
+            if finished:
+                return
+
+    @app.post("/v1/completions")
+    async def completions(request: Request):
+        body = await request.json()
+        prompt = body.get("prompt", "")
+        prompt_ids = encode_prompt(prompt, vocab)
+        sampling = _sampling_from(body)
+        req_id, q = serving.submit(prompt_ids, sampling)
+        created = int(time.time())
+        cid = f"cmpl-{uuid.uuid4().hex[:16]}"
+
+        if body.get("stream"):
+            async def sse():
+                async for tok, finished in _stream(q):
+                    delta = decode_tokens([tok]) if tok is not None else ""
+                    chunk = {
+                        "id": cid,
+                        "object": "text_completion",
+                        "created": created,
+                        "model": model_name,
+                        "choices": [
+                            {
+                                "index": 0,
+                                "text": delta,
+                                "token_ids": [tok] if tok is not None else [],
+                                "finish_reason": "stop" if finished else None,
+                            }
+                        ],
+                    }
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
+        toks = await _collect(q)
+        return JSONResponse(
+            {
+                "id": cid,
+                "object": "text_completion",
+                "created": created,
+                "model": model_name,
+                "choices": [
+                    {
+                        "index": 0,
+                        "text": decode_tokens(toks),
+                        "token_ids": toks,
+                        "finish_reason": "stop",
+                    }
+                ],
+                "usage": {
+                    "prompt_tokens": len(prompt_ids),
+                    "completion_tokens": len(toks),
+                    "total_tokens": len(prompt_ids) + len(toks),
+                },
+            }
+        )
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(request: Request):
+        body = await request.json()
+        messages = body.get("messages", [])
+        text = "\n".join(
+            f"{m.get('role', 'user')}: {m.get('content', '')}" for m in messages
+        )
+        prompt_ids = encode_prompt(text, vocab)
+        sampling = _sampling_from(body)
+        req_id, q = serving.submit(prompt_ids, sampling)
+        created = int(time.time())
+        cid = f"chatcmpl-{uuid.uuid4().hex[:16]}"
+
+        if body.get("stream"):
+            async def sse():
+                first = True
+                async for tok, finished in _stream(q):
+                    delta: Dict[str, Any] = {}
+                    if first:
+                        delta["role"] = "assistant"
+                        first = False
+                    if tok is not None:
+                        delta["content"] = decode_tokens([tok])
+                    chunk = {
+                        "id": cid,
+                        "object": "chat.completion.chunk",
+                        "created": created,
+                        "model": model_name,
+                        "choices": [
+                            {
+                                "index": 0,
+                                "delta": delta,
+                                "finish_reason": "stop" if finished else None,
+                            }
+                        ],
+                    }
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(sse(), media_type="text/event-stream")
+
+        toks = await _collect(q)
+        return JSONResponse(
+            {
+                "id": cid,
+                "object": "chat.completion",
+                "created": created,
+                "model": model_name,
+                "choices": [
+                    {
+                        "index": 0,
+                        "message": {
+                            "role": "assistant",
+                            "content": decode_tokens(toks),
+                        },
+                        "finish_reason": "stop",
+                    }
+                ],
+                "usage": {
+                    "prompt_tokens": len(prompt_ids),
+                    "completion_tokens": len(toks),
+                    "total_tokens": len(prompt_ids) + len(toks),
+                },
+            }
+        )
+
+    return app

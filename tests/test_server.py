@@ -1,0 +1,140 @@
+"""OpenAI server tests (CPU, tiny model) via httpx ASGI transport."""
+
+import asyncio
+import json
+
+import httpx
+import pytest
+
+from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+from fusioninfer_amd.models.registry import get_model_config
+from fusioninfer_amd.server.api_server import build_app, decode_tokens, encode_prompt
+from fusioninfer_amd.server.serving import ServingEngine
+
+
+@pytest.fixture(scope="module")
+def serving():
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=128),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=512, max_model_len=256
+        ),
+    )
+    s = ServingEngine(cfg, device="cpu")
+    yield s
+    s.shutdown()
+
+
+@pytest.fixture(scope="module")
+def app(serving):
+    return build_app(serving, "tiny-qwen3")
+
+
+def _client(app):
+    return httpx.AsyncClient(
+        transport=httpx.ASGITransport(app=app), base_url="http://test"
+    )
+
+
+def test_tokenizer_roundtrip():
+    ids = encode_prompt("hello world", 1024)
+    assert decode_tokens(ids) == "hello world"
+    assert encode_prompt([1, 2, 3], 1024) == [1, 2, 3]
+
+
+def test_health_and_models(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.get("/health")
+            assert r.status_code == 200 and r.json()["status"] == "ok"
+            r = await c.get("/v1/models")
+            assert r.json()["data"][0]["id"] == "tiny-qwen3"
+
+    asyncio.run(run())
+
+
+def test_completions_token_ids(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": [5, 6, 7, 8] * 6, "max_tokens": 4,
+                      "temperature": 0, "ignore_eos": True},
+            )
+            body = r.json()
+            assert len(body["choices"][0]["token_ids"]) == 4
+            assert body["usage"]["prompt_tokens"] == 24
+            assert body["usage"]["completion_tokens"] == 4
+
+    asyncio.run(run())
+
+
+def test_completions_streaming(app):
+    async def run():
+        async with _client(app) as c:
+            toks = []
+            async with c.stream(
+                "POST",
+                "/v1/completions",
+                json={"prompt": "hi there", "max_tokens": 3, "stream": True,
+                      "temperature": 0, "ignore_eos": True},
+            ) as r:
+                async for line in r.aiter_lines():
+                    if line.startswith("data: ") and "[DONE]" not in line:
+                        chunk = json.loads(line[6:])
+                        toks.extend(chunk["choices"][0]["token_ids"])
+            assert len(toks) == 3
+
+    asyncio.run(run())
+
+
+def test_chat_completions(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/chat/completions",
+                json={
+                    "messages": [{"role": "user", "content": "hello"}],
+                    "max_tokens": 3,
+                    "temperature": 0,
+                    "ignore_eos": True,
+                },
+            )
+            body = r.json()
+            assert body["choices"][0]["message"]["role"] == "assistant"
+            assert body["usage"]["completion_tokens"] == 3
+
+    asyncio.run(run())
+
+
+def test_metrics_vllm_names(app, serving):
+    async def run():
+        async with _client(app) as c:
+            r = await c.get("/metrics")
+            text = r.text
+            # the metric names the EPP scorers scrape
+            assert "vllm:gpu_cache_usage_perc" in text
+            assert "vllm:num_requests_waiting" in text
+            assert "vllm:num_requests_running" in text
+
+    asyncio.run(run())
+
+
+def test_cli_config_surface():
+    from fusioninfer_amd.server.__main__ import build_engine_config, parse_args
+
+    args = parse_args(
+        [
+            "--model", "tiny-qwen3",
+            "--tensor-parallel-size", "2",
+            "--max-model-len", "1024",
+            "--kv-transfer-config",
+            '{"kv_connector": "RcclConnector", "kv_role": "kv_producer"}',
+        ]
+    )
+    cfg = build_engine_config(args)
+    assert cfg.parallel.tensor_parallel_size == 2
+    assert cfg.scheduler.max_model_len == 1024
+    assert cfg.kv_transfer.kv_connector == "RcclConnector"
+    assert cfg.kv_transfer.kv_role == "kv_producer"
