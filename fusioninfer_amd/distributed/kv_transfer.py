@@ -1,0 +1,95 @@
+"""PD-disaggregation KV connector: prefiller -> decoder KV handoff.
+
+MI355X-native equivalent of the reference's PyNcclConnector /
+NixlConnector flag surface (reference docs/.../core-design.md:88-111,
+SURVEY.md §2.3): the prefiller (kv_producer) packs a request's KV blocks
+from EVERY layer into ONE contiguous staging tensor (HIP gather kernel)
+and ships it with a single RCCL send — xGMI is point-to-point (7 links x
+~153 GB/s), so one large send per request hits link peak where many
+per-layer sends would not. The decoder (kv_consumer)
+receives and scatters into its own paged cache, then decodes.
+
+Wire protocol per request (tags are implicit via ordering on the p2p pair):
+  1. header  int64[4]: [num_blocks, prompt_len, first_token, reserved]
+  2. staging bf16 [layers, 2, num_blocks, Hk, bs, D]
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+import fusioninfer_amd.ops as ops
+
+KV_PRODUCER = "kv_producer"
+KV_CONSUMER = "kv_consumer"
+
+
+class RcclKVConnector:
+    """Point-to-point KV mover between a prefiller rank and a decoder rank."""
+
+    def __init__(
+        self,
+        role: str,
+        peer_rank: int,
+        group: Optional[dist.ProcessGroup] = None,
+        device: str = "cpu",
+    ):
+        assert role in (KV_PRODUCER, KV_CONSUMER), role
+        self.role = role
+        self.peer_rank = peer_rank
+        self.group = group
+        self.device = torch.device(device)
+
+    # ------------------------------------------------------------- producer
+    def send_kv(
+        self,
+        kv_caches: List[Tuple[torch.Tensor, torch.Tensor]],
+        block_ids: List[int],
+        prompt_len: int,
+        first_token: int,
+    ) -> None:
+        assert self.role == KV_PRODUCER
+        ids = torch.tensor(block_ids, dtype=torch.int32, device=self.device)
+        header = torch.tensor(
+            [len(block_ids), prompt_len, first_token, 0], dtype=torch.int64,
+            device=self.device,
+        )
+        dist.send(header, self.peer_rank, group=self.group)
+        # one contiguous staging tensor across all layers
+        per_layer = [
+            ops.gather_kv_blocks(kc, vc, ids) for kc, vc in kv_caches
+        ]
+        staging = torch.stack(per_layer, dim=0).contiguous()
+        dist.send(staging, self.peer_rank, group=self.group)
+
+    # ------------------------------------------------------------- consumer
+    def recv_kv(
+        self,
+        kv_caches: List[Tuple[torch.Tensor, torch.Tensor]],
+        allocate_blocks,  # callable(num_blocks) -> List[int]
+    ) -> Tuple[List[int], int, int]:
+        """Receives one request's KV; returns (block_ids, prompt_len,
+        first_token)."""
+        assert self.role == KV_CONSUMER
+        header = torch.zeros(4, dtype=torch.int64, device=self.device)
+        dist.recv(header, self.peer_rank, group=self.group)
+        num_blocks, prompt_len, first_token = (
+            int(header[0]), int(header[1]), int(header[2])
+        )
+        block_ids = allocate_blocks(num_blocks)
+        assert len(block_ids) == num_blocks
+        kc0, _ = kv_caches[0]
+        layers = len(kv_caches)
+        staging = torch.empty(
+            (layers, 2, num_blocks, *kc0.shape[1:]),
+            dtype=kc0.dtype,
+            device=self.device,
+        )
+        dist.recv(staging, self.peer_rank, group=self.group)
+        ids = torch.tensor(block_ids, dtype=torch.int32, device=self.device)
+        for layer, (kc, vc) in enumerate(kv_caches):
+            ops.scatter_kv_blocks(staging[layer], kc, vc, ids)
+        return block_ids, prompt_len, first_token
