@@ -89,5 +89,11 @@ def destroy() -> None:
 def tp_all_reduce(t: torch.Tensor) -> torch.Tensor:
     if _TP_WORLD == 1:
         return t
+    if t.device.type == "cpu" and t.dtype == torch.bfloat16:
+        # gloo (CPU test path) lacks bf16 reduction; RCCL path is native bf16
+        f = t.float()
+        dist.all_reduce(f, group=_TP_GROUP)
+        t.copy_(f.to(t.dtype))
+        return t
     dist.all_reduce(t, group=_TP_GROUP)
     return t

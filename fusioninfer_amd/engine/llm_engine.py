@@ -53,6 +53,8 @@ class LLMEngine:
         self.runner.capture_decode_graphs()
         self._req_counter = itertools.count()
         self.seqs: Dict[str, Sequence] = {}
+        self._held: Dict[str, Sequence] = {}
+        self._import_holder = None
         # running stats for metrics / EPP scorers
         self.num_finished = 0
         self.num_generated_tokens = 0
@@ -75,6 +77,63 @@ class LLMEngine:
 
     def has_unfinished(self) -> bool:
         return self.scheduler.has_work()
+
+    # ------------------------------------------------------- PD interfaces
+    def prefill_export(self, prompt_token_ids: List[int]):
+        """PD producer: prefill one request, sample its first token, and
+        return (first_token, block_ids) with the blocks HELD for KV export.
+        Call release_held() after the connector has shipped the blocks."""
+        from fusioninfer_amd.engine.sequence import SamplingParams
+
+        req_id = self.add_request(
+            prompt_token_ids, SamplingParams(max_tokens=1, temperature=0.0)
+        )
+        seq = self.seqs[req_id]
+        seq.hold_blocks = True
+        while req_id in self.seqs:
+            self.step()
+        assert len(seq.output_token_ids) == 1
+        self._held[req_id] = seq
+        return req_id, seq.output_token_ids[0], seq.block_ids
+
+    def release_held(self, req_id: str) -> None:
+        seq = self._held.pop(req_id)
+        self.block_manager.free(seq)
+
+    def allocate_import_blocks(self, num_blocks: int) -> List[int]:
+        """PD consumer: reserve blocks the connector will scatter KV into."""
+        from fusioninfer_amd.engine.sequence import SamplingParams, Sequence
+
+        assert self.block_manager.num_free() >= num_blocks
+        holder = Sequence("_import", [0] * (num_blocks * self.cfg.cache.block_size),
+                          SamplingParams())
+        self.block_manager.allocate(holder)
+        self._import_holder = holder
+        return holder.block_ids
+
+    def add_imported_request(
+        self,
+        prompt_len: int,
+        first_token: int,
+        sampling: Optional["SamplingParams"] = None,
+        request_id: Optional[str] = None,
+    ) -> str:
+        """PD consumer: register a request whose prompt KV was imported into
+        the blocks reserved by allocate_import_blocks(). The sequence joins
+        the RUNNING set directly (its prefill happened on the prefiller)."""
+        from fusioninfer_amd.engine.sequence import SamplingParams, Sequence, SeqStatus
+
+        holder = self._import_holder
+        self._import_holder = None
+        if request_id is None:
+            request_id = f"req-{next(self._req_counter)}"
+        seq = Sequence(request_id, [0] * prompt_len, sampling or SamplingParams())
+        seq.block_ids = holder.block_ids
+        seq.append_token(first_token)
+        seq.status = SeqStatus.RUNNING
+        self.seqs[request_id] = seq
+        self.scheduler.running.append(seq)
+        return request_id
 
     # ------------------------------------------------------------ metrics
     def gpu_cache_usage(self) -> float:

@@ -113,5 +113,6 @@ class Scheduler:
     # ----------------------------------------------------------- finish
     def finish(self, seq: Sequence) -> None:
         seq.status = SeqStatus.FINISHED
-        self.bm.free(seq)
+        if not seq.hold_blocks:
+            self.bm.free(seq)
         self.running.remove(seq)

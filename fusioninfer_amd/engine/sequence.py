@@ -46,6 +46,8 @@ class Sequence:
         self.finish_time: Optional[float] = None
         # number of prompt tokens whose KV was satisfied by prefix cache
         self.num_cached_tokens = 0
+        # PD producer: keep cache blocks alive after finish for KV export
+        self.hold_blocks = False
 
     @property
     def num_prompt_tokens(self) -> int:

@@ -20,7 +20,7 @@ import fusioninfer_amd.ops as ops
 from fusioninfer_amd.config import ModelConfig
 from fusioninfer_amd.distributed import parallel_state as ps
 from fusioninfer_amd.distributed.layers import (
-    ColumnParallelLinear,
+    MergedColumnParallelLinear,
     ReplicatedLinear,
     RowParallelLinear,
 )
@@ -42,8 +42,13 @@ class Attention(nn.Module):
         self.scale = 1.0 / math.sqrt(self.head_dim)
         self.eps = cfg.rms_norm_eps
         hidden = cfg.hidden_size
-        self.qkv_proj = ColumnParallelLinear(
-            hidden, (cfg.num_heads + 2 * cfg.num_kv_heads) * cfg.head_dim
+        self.qkv_proj = MergedColumnParallelLinear(
+            hidden,
+            [
+                cfg.num_heads * cfg.head_dim,
+                cfg.num_kv_heads * cfg.head_dim,
+                cfg.num_kv_heads * cfg.head_dim,
+            ],
         )
         self.o_proj = RowParallelLinear(cfg.num_heads * cfg.head_dim, hidden)
         if cfg.qk_norm:
@@ -106,8 +111,8 @@ class Attention(nn.Module):
 class MLP(nn.Module):
     def __init__(self, cfg: ModelConfig):
         super().__init__()
-        self.gate_up_proj = ColumnParallelLinear(
-            cfg.hidden_size, 2 * cfg.intermediate_size
+        self.gate_up_proj = MergedColumnParallelLinear(
+            cfg.hidden_size, [cfg.intermediate_size, cfg.intermediate_size]
         )
         self.down_proj = RowParallelLinear(cfg.intermediate_size, cfg.hidden_size)
         self.inter_per_rank = cfg.intermediate_size // ps.tp_world_size()

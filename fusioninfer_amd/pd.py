@@ -1,0 +1,74 @@
+"""PD-disaggregated serving roles: prefiller and decoder workers.
+
+Wires the engine's PD interfaces to the RCCL KV connector
+(distributed/kv_transfer.py). The reference expresses these roles as
+`kv_role: kv_producer / kv_consumer` engine flags the controller injects
+(reference docs/.../core-design.md:88-111) — same surface
+here via --kv-transfer-config.
+
+Flow per request (reference's PD data path):
+  EPP prefill profile -> prefiller pod: prefill, sample first token,
+  pack+send KV (one RCCL send over an xGMI link) ->
+  EPP decode profile -> decoder pod: recv+scatter KV, decode loop.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+from fusioninfer_amd.distributed.kv_transfer import (
+    KV_CONSUMER,
+    KV_PRODUCER,
+    RcclKVConnector,
+)
+from fusioninfer_amd.engine.llm_engine import LLMEngine
+from fusioninfer_amd.engine.sequence import SamplingParams
+
+
+class PDPrefiller:
+    def __init__(self, engine: LLMEngine, connector: RcclKVConnector):
+        assert connector.role == KV_PRODUCER
+        self.engine = engine
+        self.connector = connector
+
+    def process(self, prompt_token_ids: List[int]) -> int:
+        """Prefill a prompt, ship its KV to the decoder, return the first
+        sampled token (also shipped in the header)."""
+        req_id, first_token, block_ids = self.engine.prefill_export(
+            prompt_token_ids
+        )
+        self.connector.send_kv(
+            self.engine.runner.kv_caches,
+            block_ids,
+            len(prompt_token_ids),
+            first_token,
+        )
+        self.engine.release_held(req_id)
+        return first_token
+
+
+class PDDecoder:
+    def __init__(self, engine: LLMEngine, connector: RcclKVConnector):
+        assert connector.role == KV_CONSUMER
+        self.engine = engine
+        self.connector = connector
+
+    def accept(self, sampling: Optional[SamplingParams] = None) -> str:
+        """Receive one request's KV from the prefiller and admit it into the
+        decode loop. Returns the request id."""
+        _, prompt_len, first_token = self.connector.recv_kv(
+            self.engine.runner.kv_caches,
+            self.engine.allocate_import_blocks,
+        )
+        return self.engine.add_imported_request(
+            prompt_len, first_token, sampling
+        )
+
+    def decode_all(self):
+        """Run the decode loop until all admitted requests finish."""
+        results = {}
+        while self.engine.has_unfinished():
+            for out in self.engine.step():
+                if out.finished:
+                    results[out.request_id] = out
+        return results
