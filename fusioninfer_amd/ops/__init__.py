@@ -153,14 +153,18 @@ def paged_attention_decode(
     return ref.paged_attention_decode(q, k_cache, v_cache, block_tables, seq_lens, scale)
 
 
+PREFILL_TILE_ROWS = 256  # q rows per workgroup (8 waves x 32); kernel contract
+
+
 def build_prefill_tiles(seq_lens, device=None):
-    """Host-side tile table for the prefill kernel: 64 q-rows per workgroup.
+    """Host-side tile table for the prefill kernel: one workgroup per
+    PREFILL_TILE_ROWS q-rows of each sequence.
 
     seq_lens: list[int]. Returns (tile_seq, tile_row0) int32 tensors.
     """
     tile_seq, tile_row0 = [], []
     for s, L in enumerate(seq_lens):
-        for r0 in range(0, L, 64):
+        for r0 in range(0, L, PREFILL_TILE_ROWS):
             tile_seq.append(s)
             tile_row0.append(r0)
     return (

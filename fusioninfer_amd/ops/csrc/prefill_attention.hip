@@ -1,19 +1,20 @@
-// Varlen causal prefill attention (flash-style, MFMA) for CDNA4 (gfx950).
+// Varlen causal prefill attention (flash-style, MFMA) for CDNA4 (gfx950) — v2.
 //
-// out[t, h, :] = softmax(Q K^T * scale, causal) V over each sequence's own
-// rows, GQA (G q-heads per kv-head), bf16 in / fp32 accumulate.
+// out[t, h, :] = softmax(Q K^T * scale, causal) V, GQA, bf16 in / fp32 accum.
 //
-// v1 structure (correctness-first; the optimization ladder in
-// cdna_hip_programming.md §B raises this later):
-//  * workgroup = 4 independent waves; each wave owns a 16-row Q tile of one
-//    (seq, q-head) and runs its own online-softmax loop over 32-token KV
-//    tiles — no cross-wave barriers in the main loop.
-//  * QK^T: mfma_f32_16x16x32_bf16. With row-major Q and K, BOTH fragments
-//    are contiguous bf16x8 loads: A[i][k] = Q[row i][k-chunk], and
-//    B[k][j] = K^T[k][j] = K[row j][k-chunk] (the "B^T input" form).
-//  * P goes through LDS (bf16) so the PV A-fragment is a ds_read_b128;
-//    V^T is staged per-wave into LDS so the PV B-fragment is contiguous.
-//  * online softmax per 16x32 S-tile; C-fragment mapping (guide §3):
+// v2 structure (the 8-wave ladder of cdna_hip_programming.md §B):
+//  * workgroup = 8 waves; wave w owns q rows [row0 + 32w, +32) of one
+//    (seq, q-head) — a 256-row Q block per workgroup.
+//  * KV tiles of 64 tokens are staged ONCE per workgroup into shared LDS
+//    (cooperative coalesced loads, barrier-synced): an 8x cut in global
+//    K/V traffic vs per-wave reads (v1's top cost: 26% of bench GPU time).
+//  * K tile [64][256B] is XOR-swizzled (byte ^= (row&15)<<4) so the QK^T
+//    B-fragment ds_read_b128 is bank-conflict-free (guide T2/G4);
+//    V is staged TRANSPOSED ([128][128B] rows, byte ^= (row&7)<<4) so the
+//    PV B-fragment is a contiguous swizzled b128 read.
+//  * QK^T and PV both mfma_f32_16x16x32_bf16; P goes through per-wave
+//    swizzled LDS (bf16) to become the PV A-fragment.
+//  * online softmax per 16-row fragment block; C-fragment mapping:
 //    col = lane&15, row = (lane>>4)*4 + reg.
 //
 // Capability parity: the paged-attention prefill the reference delegates to
@@ -26,13 +27,22 @@ namespace fi {
 typedef __attribute__((ext_vector_type(8))) short short8;
 typedef __attribute__((ext_vector_type(4))) float floatx4;
 
-constexpr int kQTile = 16;    // q rows per wave
-constexpr int kKVTile = 32;   // kv tokens per tile (= one MFMA K depth)
-constexpr int kPFWaves = 4;
+constexpr int kWaves = 8;
+constexpr int kQPerWave = 32;                 // q rows per wave (2 MFMA blocks)
+constexpr int kQBlock = kWaves * kQPerWave;   // 256 q rows per workgroup
+constexpr int kKVTile = 64;                   // kv tokens per LDS tile
 constexpr float kPNegInf = -1e30f;
 
+// byte-address XOR swizzles, bijective within a ROWB-byte row (guide T2/G4):
+// spread a 16-lane group's distinct-row b128 reads over ROWB/16 bank slots.
+template <int ROWB>
+FI_DEV int swz(int row, int byte_in_row) {
+  constexpr int kMask = (ROWB / 16 > 16 ? 16 : ROWB / 16) - 1;
+  return row * ROWB + (byte_in_row ^ ((row & kMask) << 4));
+}
+
 template <int D>
-__global__ __launch_bounds__(kPFWaves * kWaveSize) void prefill_attn_kernel(
+__global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     u16* __restrict__ out,        // [T, Hq, D]
     const u16* __restrict__ q,    // [T] rows, stride q_stride
     const u16* __restrict__ k,    // [T] rows, stride k_stride
@@ -42,171 +52,212 @@ __global__ __launch_bounds__(kPFWaves * kWaveSize) void prefill_attn_kernel(
     const int* __restrict__ cu_seqlens,  // [nseqs+1]
     const int64_t q_stride, const int64_t k_stride, const int64_t v_stride,
     const int num_q_heads, const int num_kv_heads, const float scale) {
-  constexpr int KB = D / 32;   // MFMA k-chunks over the head dim
-  constexpr int CB = D / 16;   // output col blocks
+  constexpr int KB = D / 32;   // QK^T k-chunks over the head dim
+  constexpr int CB = D / 16;   // PV output col blocks
+  constexpr int kKRowB = D * 2;        // K row bytes (D=128 -> 256)
+  constexpr int kVTRowB = kKVTile * 2; // V^T row bytes (64 kv -> 128)
 
   const int head = blockIdx.y;
   const int kv_head = head / (num_q_heads / num_kv_heads);
-  const int wave = threadIdx.x / kWaveSize;
-  const int lane = threadIdx.x % kWaveSize;
-  const int col = lane & 15;       // fragment col / B row-token
-  const int hi = lane >> 4;        // fragment 8-chunk index (0..3)
+  const int tid = threadIdx.x;
+  const int wave = tid / kWaveSize;
+  const int lane = tid % kWaveSize;
+  const int col = lane & 15;
+  const int hi = lane >> 4;
 
   const int seq = tile_seq[blockIdx.x];
   const int seq_start = cu_seqlens[seq];
   const int seq_len = cu_seqlens[seq + 1] - seq_start;
-  const int row0 = tile_row0[blockIdx.x] + wave * kQTile;  // within seq
-  if (row0 >= seq_len) return;
+  const int wg_row0 = tile_row0[blockIdx.x];
+  const int row0 = wg_row0 + wave * kQPerWave;      // this wave's first q row
+  const bool active = row0 < seq_len;
 
-  // LDS: per-wave V^T tile + per-wave P tile
-  __shared__ u16 vt_lds[kPFWaves][D][kKVTile];        // V^T (dim-major)
-  __shared__ u16 p_lds[kPFWaves][kQTile][kKVTile];
+  // LDS: K tile + V^T tile (workgroup-shared) + P tiles (per-wave)
+  __shared__ u16 k_lds[kKVTile * D];            // swizzled rows of 256 B
+  __shared__ u16 vt_lds[D * kKVTile];           // transposed, swizzled 128-B rows
+  __shared__ u16 p_lds[kWaves][kQPerWave * kKVTile];  // swizzled 128-B rows
 
-  // ---- load Q fragments (A): lane holds Q[row0+col][kb*32 + hi*8 .. +8]
-  short8 a_q[KB];
-  {
-    const int qr = min(row0 + col, seq_len - 1);  // clamp; masked later
-    const u16* qrow = q + (seq_start + qr) * q_stride +
-                      static_cast<int64_t>(head) * D;
+  // ---- Q fragments: a_q[rb][kb], lane holds Q[row0+rb*16+col][kb*32+hi*8..]
+  short8 a_q[2][KB];
+  if (active) {
 #pragma unroll
-    for (int kb = 0; kb < KB; ++kb)
-      a_q[kb] = *reinterpret_cast<const short8*>(qrow + kb * 32 + hi * 8);
+    for (int rb = 0; rb < 2; ++rb) {
+      const int qr = min(row0 + rb * 16 + col, seq_len - 1);
+      const u16* qrow = q + (seq_start + qr) * q_stride +
+                        static_cast<int64_t>(head) * D;
+#pragma unroll
+      for (int kb = 0; kb < KB; ++kb)
+        a_q[rb][kb] = *reinterpret_cast<const short8*>(qrow + kb * 32 + hi * 8);
+    }
   }
 
-  float m[4], l[4];
-  floatx4 o_acc[CB];
+  float m[2][4], l[2][4];
+  floatx4 o_acc[2][CB];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    m[r] = kPNegInf;
-    l[r] = 0.f;
+  for (int rb = 0; rb < 2; ++rb) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m[rb][r] = kPNegInf;
+      l[rb][r] = 0.f;
+    }
+#pragma unroll
+    for (int cb = 0; cb < CB; ++cb) o_acc[rb][cb] = {0.f, 0.f, 0.f, 0.f};
   }
-#pragma unroll
-  for (int cb = 0; cb < CB; ++cb) o_acc[cb] = {0.f, 0.f, 0.f, 0.f};
 
-  const int q_max = min(row0 + kQTile, seq_len) - 1;   // last valid q row
-  const int num_kv_tiles = q_max / kKVTile + 1;
+  // kv range: the whole workgroup iterates to its max causal row
+  const int wg_q_max = min(wg_row0 + kQBlock, seq_len) - 1;
+  const int num_kv_tiles = wg_q_max / kKVTile + 1;
+  const int my_q_max = min(row0 + kQPerWave, seq_len) - 1;  // per-wave
 
   for (int t = 0; t < num_kv_tiles; ++t) {
     const int kv0 = t * kKVTile;
-    const int kv_valid = min(seq_len - kv0, kKVTile);
 
-    // ---- QK^T: S[16 x 32] = Q[16 x D] K^T ----
-    floatx4 s_acc[2];
-    s_acc[0] = {0.f, 0.f, 0.f, 0.f};
-    s_acc[1] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-    for (int ch = 0; ch < 2; ++ch) {
-      const int kt = min(kv0 + ch * 16 + col, seq_len - 1);
-      const u16* krow = k + (seq_start + kt) * k_stride +
-                        static_cast<int64_t>(kv_head) * D;
-#pragma unroll
-      for (int kb = 0; kb < KB; ++kb) {
-        const short8 b_k =
-            *reinterpret_cast<const short8*>(krow + kb * 32 + hi * 8);
-        s_acc[ch] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[kb], b_k,
-                                                            s_acc[ch], 0, 0, 0);
-      }
-    }
-
-    // ---- stage V^T into LDS (each lane: one V row, 8 dims per store loop)
+    // ---- cooperative staging: K [64][D] swizzled; V^T [D][64] swizzled ----
     {
-      // 64 lanes cover 32 rows x (D/8 col-chunks per half): lane maps to
-      // row = lane % 32, chunk = lane / 32, strided by 2 chunks.
-      const int vrow = lane & 31;
-      const int kt = min(kv0 + vrow, seq_len - 1);
-      const u16* vr = v + (seq_start + kt) * v_stride +
-                      static_cast<int64_t>(kv_head) * D;
-      for (int c8 = lane >> 5; c8 < D / 8; c8 += 2) {
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(vr + c8 * 8);
+      // 512 threads: kv = tid/8 (64 rows), c16 = tid%8 (8 x 16-elem chunks)
+      const int kv_r = tid >> 3;
+      const int c16 = tid & 7;
+      const int src = min(kv0 + kv_r, seq_len - 1);
+      const u16* krow = k + (seq_start + src) * k_stride +
+                        static_cast<int64_t>(kv_head) * D + c16 * 16;
+      const u16* vrow = v + (seq_start + src) * v_stride +
+                        static_cast<int64_t>(kv_head) * D + c16 * 16;
+      bf16x8 k0 = *reinterpret_cast<const bf16x8*>(krow);
+      bf16x8 k1 = *reinterpret_cast<const bf16x8*>(krow + 8);
+      bf16x8 v0 = *reinterpret_cast<const bf16x8*>(vrow);
+      bf16x8 v1 = *reinterpret_cast<const bf16x8*>(vrow + 8);
+      char* kbase = reinterpret_cast<char*>(k_lds);
+      *reinterpret_cast<bf16x8*>(kbase + swz<kKRowB>(kv_r, c16 * 32)) = k0;
+      *reinterpret_cast<bf16x8*>(kbase + swz<kKRowB>(kv_r, c16 * 32 + 16)) = k1;
+      // V transpose: element (kv_r, d) -> vt row d, col kv_r
+      char* vbase = reinterpret_cast<char*>(vt_lds);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) vt_lds[wave][c8 * 8 + j][vrow] = vv.h[j];
+      for (int j = 0; j < 8; ++j) {
+        const int d0 = c16 * 16 + j;
+        const int d1 = d0 + 8;
+        *reinterpret_cast<u16*>(
+            vbase + swz<kVTRowB>(d0, kv_r * 2)) = v0.h[j];
+        *reinterpret_cast<u16*>(
+            vbase + swz<kVTRowB>(d1, kv_r * 2)) = v1.h[j];
       }
     }
+    __syncthreads();
 
-    // ---- mask + online softmax on the 16x32 S tile ----
-    float p[2][4];
+    const bool compute = active && kv0 <= my_q_max;
+    if (compute) {
+      const char* kbase = reinterpret_cast<const char*>(k_lds);
+      const char* vbase = reinterpret_cast<const char*>(vt_lds);
+      char* pbase = reinterpret_cast<char*>(p_lds[wave]);
+
 #pragma unroll
-    for (int ch = 0; ch < 2; ++ch) {
+      for (int rb = 0; rb < 2; ++rb) {
+        // ---- QK^T: S[16 x 64] ----
+        floatx4 s_acc[4];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int q_pos = row0 + hi * 4 + r;
-        const int kv_pos = kv0 + ch * 16 + col;
-        float sv = s_acc[ch][r] * scale;
-        if (kv_pos > q_pos || q_pos >= seq_len || kv_pos >= seq_len)
-          sv = kPNegInf;
-        p[ch][r] = sv;
+        for (int cb4 = 0; cb4 < 4; ++cb4) {
+          s_acc[cb4] = {0.f, 0.f, 0.f, 0.f};
+          const int krow_idx = cb4 * 16 + col;  // kv token within tile
+#pragma unroll
+          for (int kb = 0; kb < KB; ++kb) {
+            const short8 b_k = *reinterpret_cast<const short8*>(
+                kbase + swz<kKRowB>(krow_idx, kb * 32 * 2 + hi * 16));
+            s_acc[cb4] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_q[rb][kb], b_k, s_acc[cb4], 0, 0, 0);
+          }
+        }
+
+        // ---- mask + online softmax over the 16x64 block ----
+        float p[4][4];
+        const int q_base = row0 + rb * 16 + hi * 4;
+#pragma unroll
+        for (int cb4 = 0; cb4 < 4; ++cb4) {
+          const int kv_pos = kv0 + cb4 * 16 + col;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int q_pos = q_base + r;
+            float sv = s_acc[cb4][r] * scale;
+            if (kv_pos > q_pos || q_pos >= seq_len || kv_pos >= seq_len)
+              sv = kPNegInf;
+            p[cb4][r] = sv;
+          }
+        }
+        float alpha[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float rm = fmaxf(fmaxf(p[0][r], p[1][r]), fmaxf(p[2][r], p[3][r]));
+#pragma unroll
+          for (int off = 1; off < 16; off <<= 1)
+            rm = fmaxf(rm, __shfl_xor(rm, off, 64));
+          const float m_new = fmaxf(m[rb][r], rm);
+          if (m_new <= kPNegInf) {  // row fully masked so far
+            alpha[r] = 0.f;
+#pragma unroll
+            for (int cb4 = 0; cb4 < 4; ++cb4) p[cb4][r] = 0.f;
+            continue;
+          }
+          alpha[r] = (m[rb][r] <= kPNegInf) ? 0.f : __expf(m[rb][r] - m_new);
+          m[rb][r] = m_new;
+          float rs = 0.f;
+#pragma unroll
+          for (int cb4 = 0; cb4 < 4; ++cb4) {
+            p[cb4][r] =
+                (p[cb4][r] <= kPNegInf) ? 0.f : __expf(p[cb4][r] - m_new);
+            rs += p[cb4][r];
+          }
+#pragma unroll
+          for (int off = 1; off < 16; off <<= 1) rs += __shfl_xor(rs, off, 64);
+          l[rb][r] = l[rb][r] * alpha[r] + rs;
+        }
+
+        // ---- P -> per-wave LDS (bf16, swizzled 128-B rows) ----
+#pragma unroll
+        for (int cb4 = 0; cb4 < 4; ++cb4)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            *reinterpret_cast<u16*>(
+                pbase + swz<kVTRowB>(rb * 16 + hi * 4 + r,
+                                        (cb4 * 16 + col) * 2)) =
+                f32_to_bf16(p[cb4][r]);
+
+        // ---- PV: O[16 x D] += P[16 x 64] V[64 x D] ----
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {  // kv 64 = 2 MFMA k-depths
+          const short8 a_p = *reinterpret_cast<const short8*>(
+              pbase + swz<kVTRowB>(rb * 16 + col, kc * 64 + hi * 16));
+#pragma unroll
+          for (int cb = 0; cb < CB; ++cb) {
+            const short8 b_v = *reinterpret_cast<const short8*>(
+                vbase + swz<kVTRowB>(cb * 16 + col, kc * 64 + hi * 16));
+            floatx4 prev = o_acc[rb][cb];
+            if (kc == 0) {
+#pragma unroll
+              for (int r = 0; r < 4; ++r) prev[r] *= alpha[r];
+            }
+            o_acc[rb][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_p, b_v, prev, 0, 0, 0);
+          }
+        }
       }
     }
-    float alpha[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float rm = fmaxf(p[0][r], p[1][r]);
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        rm = fmaxf(rm, __shfl_xor(rm, off, 64));
-      const float m_new = fmaxf(m[r], rm);
-      alpha[r] = (m[r] <= kPNegInf) ? 0.f : __expf(m[r] - m_new);
-      if (m_new <= kPNegInf) {  // fully-masked row so far
-        alpha[r] = 0.f;
-        m[r] = m_new;
-        p[0][r] = 0.f;
-        p[1][r] = 0.f;
-        continue;
-      }
-      m[r] = m_new;
-      float rs = 0.f;
-#pragma unroll
-      for (int ch = 0; ch < 2; ++ch) {
-        p[ch][r] = (p[ch][r] <= kPNegInf) ? 0.f : __expf(p[ch][r] - m_new);
-        rs += p[ch][r];
-      }
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1) rs += __shfl_xor(rs, off, 64);
-      l[r] = l[r] * alpha[r] + rs;
-    }
-
-    // ---- P -> LDS (bf16) ----
-#pragma unroll
-    for (int ch = 0; ch < 2; ++ch)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        p_lds[wave][hi * 4 + r][ch * 16 + col] = f32_to_bf16(p[ch][r]);
-
-    // zero the tail of V^T's kv rows so garbage never enters PV
-    if (kv_valid < kKVTile) {
-      const int vrow = lane & 31;
-      if (vrow >= kv_valid)
-        for (int c8 = lane >> 5; c8 < D / 8; c8 += 2)
-#pragma unroll
-          for (int j = 0; j < 8; ++j) vt_lds[wave][c8 * 8 + j][vrow] = 0;
-    }
-
-    // ---- PV: O[16 x D] += P[16 x 32] V[32 x D] ----
-    // A-frag: lane holds P[col][hi*8 .. +8] (contiguous in p_lds)
-    const short8 a_p = *reinterpret_cast<const short8*>(&p_lds[wave][col][hi * 8]);
-#pragma unroll
-    for (int cb = 0; cb < CB; ++cb) {
-      // B-frag: lane holds V^T[cb*16 + col][hi*8 .. +8] (contiguous)
-      const short8 b_v =
-          *reinterpret_cast<const short8*>(&vt_lds[wave][cb * 16 + col][hi * 8]);
-      floatx4 prev = o_acc[cb];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) prev[r] *= alpha[r];
-      o_acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, b_v, prev, 0, 0, 0);
-    }
+    __syncthreads();
   }
 
   // ---- epilogue: normalize and store ----
+  if (active) {
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int q_pos = row0 + hi * 4 + r;
-    if (q_pos >= seq_len) continue;
-    const float inv = (l[r] > 0.f) ? 1.f / l[r] : 0.f;
-    u16* orow = out + (static_cast<int64_t>(seq_start + q_pos) * num_q_heads +
-                       head) * D;
+    for (int rb = 0; rb < 2; ++rb) {
 #pragma unroll
-    for (int cb = 0; cb < CB; ++cb)
-      orow[cb * 16 + col] = f32_to_bf16(o_acc[cb][r] * inv);
+      for (int r = 0; r < 4; ++r) {
+        const int q_pos = row0 + rb * 16 + hi * 4 + r;
+        if (q_pos >= seq_len) continue;
+        const float inv = (l[rb][r] > 0.f) ? 1.f / l[rb][r] : 0.f;
+        u16* orow = out + (static_cast<int64_t>(seq_start + q_pos) *
+                               num_q_heads + head) * D;
+#pragma unroll
+        for (int cb = 0; cb < CB; ++cb)
+          orow[cb * 16 + col] = f32_to_bf16(o_acc[rb][cb][r] * inv);
+      }
+    }
   }
 }
 
@@ -216,7 +267,7 @@ void launch_prefill_attn(u16* out, const u16* q, const u16* k, const u16* v,
                          int64_t k_stride, int64_t v_stride, int num_q_heads,
                          int num_kv_heads, int head_dim, float scale,
                          hipStream_t stream) {
-  dim3 grid(ntiles, num_q_heads), block(kPFWaves * kWaveSize);
+  dim3 grid(ntiles, num_q_heads), block(kWaves * kWaveSize);
   if (head_dim == 128) {
     hipLaunchKernelGGL((prefill_attn_kernel<128>), grid, block, 0, stream, out,
                        q, k, v, tile_seq, tile_row0, cu_seqlens, q_stride,
