@@ -135,6 +135,9 @@ def scatter_kv_blocks(staging, k_cache, v_cache, block_ids):
 
 # ---------------------------------------------------------------- attention
 
+DECODE_PARTITION_TOKENS = 512  # flash-decoding partition size (kernel contract)
+
+
 def paged_attention_decode(
     q: torch.Tensor,            # [S, Hq, D]
     k_cache: torch.Tensor,      # [B, Hk, bs, D]
@@ -147,8 +150,27 @@ def paged_attention_decode(
         scale = 1.0 / math.sqrt(q.shape[-1])
     if q.is_cuda:
         _require_native()
+        S, Hq, D = q.shape
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        _C.paged_attention_decode(out, q, k_cache, v_cache, block_tables, seq_lens, scale)
+        # partitions from the block-table width: static shape under
+        # hipGraph capture (seq_lens vary between replays, width doesn't)
+        max_tokens = int(block_tables.shape[1]) * int(k_cache.shape[2])
+        num_parts = max(
+            (max_tokens + DECODE_PARTITION_TOKENS - 1) // DECODE_PARTITION_TOKENS,
+            1,
+        )
+        ml_ws = acc_ws = None
+        if num_parts > 1:
+            ml_ws = torch.empty(
+                (S, Hq, num_parts, 2), dtype=torch.float32, device=q.device
+            )
+            acc_ws = torch.empty(
+                (S, Hq, num_parts, D), dtype=torch.float32, device=q.device
+            )
+        _C.paged_attention_decode(
+            out, q, k_cache, v_cache, block_tables, seq_lens,
+            ml_ws, acc_ws, num_parts, scale,
+        )
         return out
     return ref.paged_attention_decode(q, k_cache, v_cache, block_tables, seq_lens, scale)
 

@@ -22,9 +22,10 @@ void launch_reshape_and_cache(const u16*, const u16*, u16*, u16*, const int*,
 template <bool GATHER>
 void launch_kv_block_copy(u16*, u16*, u16*, const int*, int, int64_t,
                           hipStream_t);
-void launch_paged_attn_decode(u16*, const u16*, const u16*, const u16*,
-                              const int*, const int*, int, int64_t, int, int,
-                              int, int, float, hipStream_t);
+void launch_paged_attn_decode(u16*, float*, float*, const u16*, const u16*,
+                              const u16*, const int*, const int*, int,
+                              int64_t, int, int, int, int, int, float,
+                              hipStream_t);
 void launch_prefill_attn(u16*, const u16*, const u16*, const u16*, const int*,
                          const int*, const int*, int, int64_t, int64_t,
                          int64_t, int, int, int, float, hipStream_t);
@@ -150,7 +151,10 @@ void scatter_kv_blocks(at::Tensor staging, at::Tensor k_cache,
 
 void paged_attention_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                             at::Tensor v_cache, at::Tensor block_tables,
-                            at::Tensor seq_lens, double scale) {
+                            at::Tensor seq_lens,
+                            c10::optional<at::Tensor> ml_ws,
+                            c10::optional<at::Tensor> acc_ws,
+                            int64_t num_parts, double scale) {
   CHECK_BF16_CUDA(out);
   CHECK_BF16_CUDA(q);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt &&
@@ -163,11 +167,20 @@ void paged_attention_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
   const int num_kv_heads = k_cache.size(1);
   TORCH_CHECK(k_cache.size(2) == 16, "cache block_size must be 16");
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == head_dim);
+  float* ml = nullptr;
+  float* acc = nullptr;
+  if (num_parts > 1) {
+    TORCH_CHECK(ml_ws && acc_ws, "workspace required when num_parts > 1");
+    ml = ml_ws->data_ptr<float>();
+    acc = acc_ws->data_ptr<float>();
+  }
   fi::launch_paged_attn_decode(
-      bf16_ptr(out), bf16_cptr(q), bf16_cptr(k_cache), bf16_cptr(v_cache),
-      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(), num_seqs,
-      q.stride(0), block_tables.size(1), num_kv_heads, head_dim,
-      num_heads / num_kv_heads, static_cast<float>(scale), current_stream());
+      bf16_ptr(out), ml, acc, bf16_cptr(q), bf16_cptr(k_cache),
+      bf16_cptr(v_cache), block_tables.data_ptr<int>(),
+      seq_lens.data_ptr<int>(), num_seqs, q.stride(0), block_tables.size(1),
+      num_kv_heads, head_dim, num_heads / num_kv_heads,
+      static_cast<int>(num_parts), static_cast<float>(scale),
+      current_stream());
 }
 
 void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,

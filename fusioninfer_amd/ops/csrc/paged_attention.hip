@@ -1,20 +1,24 @@
-// Paged-attention DECODE kernel for CDNA4 (gfx950).
+// Paged-attention DECODE kernels for CDNA4 (gfx950) — v2 (flash-decoding).
 //
 // One-token attention per sequence against the paged KV cache
-// ([num_blocks, Hk, 16, D] bf16). Memory-bound: the job is to stream each
-// sequence's K/V exactly once at near-HBM rate while reusing every K/V read
-// across the GQA group (G q-heads per kv-head).
+// ([num_blocks, Hk, 16, D] bf16). Memory-bound: stream each sequence's K/V
+// once at near-HBM rate, reusing every K/V read across the GQA group
+// (G q-heads per kv-head).
 //
-// Structure: one 4-wave workgroup per (seq, kv_head). Waves process
-// interleaved 16-token chunks (1 chunk == 1 cache block, so a chunk's K/V
-// are contiguous). Per chunk, two lane layouts:
+// Grid: (num_seqs, num_kv_heads, num_partitions) — the context is split
+// into 512-token partitions (flash-decoding) so small decode batches still
+// fill 256 CUs; partial (m, l, acc) go to an fp32 workspace and a second
+// kernel merges partitions. num_partitions == 1 writes output directly.
+//
+// Within a workgroup: 4 waves process interleaved 16-token chunks (1 chunk
+// == 1 cache block, so a chunk's K/V are contiguous). Per chunk:
 //   phase A (scores): lane = token*4 + i, lane reads K[token][i*32 .. +32]
-//     as 4x bf16x8 (64 B); dot with q (fp32, staged in LDS); 2-level
-//     shfl_xor reduce; online softmax state (m,l per G) kept wave-uniform
-//     in registers; probs parked in per-wave LDS.
-//   phase B (PV): lane = dim pair (2 elems), per-token V rows read 256 B
-//     coalesced; fp32 accumulator acc[G][2] per lane.
-// Final cross-wave flash-merge through LDS.
+//     as 4x bf16x8 (64 B); dot with q (fp32 in LDS, read as float4 =
+//     ds_read_b128); 2-level shfl_xor reduce; online-softmax state (m, l
+//     per G) wave-uniform in registers; probs parked in per-wave LDS.
+//   phase B (PV): lane = dim pair, per-token V rows read 256 B coalesced;
+//     fp32 accumulator acc[G][2] per lane.
+// Cross-wave flash-merge through LDS ends the workgroup.
 //
 // Capability parity: the paged-attention decode the reference delegates to
 // its vLLM containers (SURVEY.md §2.3 "Paged-attention decode kernel").
@@ -23,13 +27,16 @@
 
 namespace fi {
 
-constexpr int kBlockSz = 16;   // cache block size (tokens)
+constexpr int kBlockSz = 16;     // cache block size (tokens)
 constexpr int kNWaves = 4;
+constexpr int kPartChunks = 32;  // 512 tokens per partition
 constexpr float kNegInf = -1e30f;
 
 template <int D, int G>
 __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
-    u16* __restrict__ out,            // [S, Hq, D]
+    u16* __restrict__ out,            // [S, Hq, D] (written when 1 partition)
+    float* __restrict__ ml_ws,        // [S, Hq, P, 2] (multi-partition)
+    float* __restrict__ acc_ws,       // [S, Hq, P, D]
     const u16* __restrict__ q,        // [S] rows, stride q_stride, Hq*D elems
     const u16* __restrict__ k_cache,  // [B, Hk, 16, D]
     const u16* __restrict__ v_cache,
@@ -39,12 +46,29 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     const float scale) {
   const int seq = blockIdx.x;
   const int kv_head = blockIdx.y;
+  const int part = blockIdx.z;
+  const int num_parts = gridDim.z;
   const int tid = threadIdx.x;
   const int wave = tid / kWaveSize;
   const int lane = tid % kWaveSize;
 
   const int ctx = seq_lens[seq];
   const int num_chunks = (ctx + kBlockSz - 1) / kBlockSz;
+  const int chunk_lo = part * kPartChunks;
+  const int chunk_hi = min(num_chunks, chunk_lo + kPartChunks);
+  const int num_heads = num_kv_heads * G;
+
+  // empty partition: publish "nothing" and leave
+  if (chunk_lo >= num_chunks) {
+    if (tid < G) {
+      const int h = kv_head * G + tid;
+      float* ml = ml_ws + ((static_cast<int64_t>(seq) * num_heads + h) *
+                               num_parts + part) * 2;
+      ml[0] = kNegInf;
+      ml[1] = 0.f;
+    }
+    return;
+  }
 
   __shared__ float q_lds[G][D];
   __shared__ float p_lds[kNWaves][kBlockSz][G];
@@ -52,7 +76,7 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   __shared__ float merge_l[kNWaves][G];
   __shared__ float merge_acc[kNWaves][G][D];
 
-  // stage q (G heads) into LDS as fp32
+  // stage q (G heads) into LDS as fp32 (pre-scaled)
   {
     const u16* q_row = q + seq * q_stride +
                        static_cast<int64_t>(kv_head) * G * D;
@@ -74,7 +98,7 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   const int quad = lane % 4;         // phase-A dim quarter
   constexpr int DPQ = D / 4;         // dims per phase-A lane
 
-  for (int chunk = wave; chunk < num_chunks; chunk += kNWaves) {
+  for (int chunk = chunk_lo + wave; chunk < chunk_hi; chunk += kNWaves) {
     const int block_id = block_tables[seq * max_blocks + chunk];
     const int64_t kv_base =
         ((static_cast<int64_t>(block_id) * num_kv_heads + kv_head) * kBlockSz) * D;
@@ -82,8 +106,6 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
 
     // ---- phase A: scores for 16 tokens x G heads ----
     float s[G];
-#pragma unroll
-    for (int g = 0; g < G; ++g) s[g] = 0.f;
     {
       const u16* k_row = k_cache + kv_base + tok * D + quad * DPQ;
       float kf[DPQ];
@@ -118,17 +140,17 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     }
 
     // chunk max over tokens (xor 4..32 spans the 16 token groups)
-    float m_new[G], alpha[G];
+    float alpha[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) {
       float cm = s[g];
 #pragma unroll
       for (int off = 4; off < 64; off <<= 1)
         cm = fmaxf(cm, __shfl_xor(cm, off, 64));
-      m_new[g] = fmaxf(m[g], cm);
-      alpha[g] = __expf(m[g] - m_new[g]);  // exp(-inf-(-inf)) guarded below
-      if (m[g] <= kNegInf && m_new[g] <= kNegInf) alpha[g] = 0.f;
-      m[g] = m_new[g];
+      const float m_new = fmaxf(m[g], cm);
+      alpha[g] = __expf(m[g] - m_new);
+      if (m[g] <= kNegInf && m_new <= kNegInf) alpha[g] = 0.f;
+      m[g] = m_new;
     }
 
     // probs + row-sum; every token is replicated on 4 lanes -> scale by 1/4
@@ -145,12 +167,12 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     // ---- phase B: PV accumulate; lane covers dims {2l, 2l+1} ----
     const int nvalid = min(ctx - chunk * kBlockSz, kBlockSz);
     const u16* v_rows = v_cache + kv_base;
-    if (lane < D / 2) {
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
-        acc[g][0] *= alpha[g];
-        acc[g][1] *= alpha[g];
-      }
+    for (int g = 0; g < G; ++g) {
+      acc[g][0] *= alpha[g];
+      acc[g][1] *= alpha[g];
+    }
+    if (lane < D / 2) {
       for (int t = 0; t < nvalid; ++t) {
         const u32 vbits = *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
         const float v0 = bf16_to_f32(static_cast<u16>(vbits & 0xffff));
@@ -161,12 +183,6 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
           acc[g][0] = fmaf(p, v0, acc[g][0]);
           acc[g][1] = fmaf(p, v1, acc[g][1]);
         }
-      }
-    } else {
-#pragma unroll
-      for (int g = 0; g < G; ++g) {
-        acc[g][0] *= alpha[g];
-        acc[g][1] *= alpha[g];
       }
     }
   }
@@ -203,26 +219,79 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       o0 = fmaf(merge_acc[w][g][2 * lane], f, o0);
       o1 = fmaf(merge_acc[w][g][2 * lane + 1], f, o1);
     }
-    const float inv = 1.f / L;
-    u16* o_row = out + (static_cast<int64_t>(seq) * num_kv_heads * G +
-                        kv_head * G + g) * D;
-    const u32 obits = static_cast<u32>(f32_to_bf16(o0 * inv)) |
-                      (static_cast<u32>(f32_to_bf16(o1 * inv)) << 16);
-    *reinterpret_cast<u32*>(o_row + 2 * lane) = obits;
+    const int h = kv_head * G + g;
+    if (num_parts == 1) {
+      const float inv = 1.f / L;
+      u16* o_row = out + (static_cast<int64_t>(seq) * num_heads + h) * D;
+      const u32 obits = static_cast<u32>(f32_to_bf16(o0 * inv)) |
+                        (static_cast<u32>(f32_to_bf16(o1 * inv)) << 16);
+      *reinterpret_cast<u32*>(o_row + 2 * lane) = obits;
+    } else {
+      const int64_t slot = (static_cast<int64_t>(seq) * num_heads + h) *
+                               gridDim.z + part;
+      if (lane == 0) {
+        ml_ws[slot * 2] = gm;
+        ml_ws[slot * 2 + 1] = L;
+      }
+      float2* arow = reinterpret_cast<float2*>(acc_ws + slot * D);
+      arow[lane] = make_float2(o0, o1);
+    }
   }
 }
 
-void launch_paged_attn_decode(u16* out, const u16* q, const u16* k_cache,
+// merge partials: grid (S, Hq), one wave; lane covers dims {2l, 2l+1}
+template <int D>
+__global__ __launch_bounds__(kWaveSize) void paged_attn_reduce_kernel(
+    u16* __restrict__ out,            // [S, Hq, D]
+    const float* __restrict__ ml_ws,  // [S, Hq, P, 2]
+    const float* __restrict__ acc_ws, // [S, Hq, P, D]
+    const int num_parts) {
+  const int seq = blockIdx.x;
+  const int h = blockIdx.y;
+  const int num_heads = gridDim.y;
+  const int lane = threadIdx.x;
+  if (lane >= D / 2) return;
+  const int64_t base = (static_cast<int64_t>(seq) * num_heads + h) * num_parts;
+
+  float gm = kNegInf;
+  for (int p = 0; p < num_parts; ++p)
+    gm = fmaxf(gm, ml_ws[(base + p) * 2]);
+  float L = 0.f, o0 = 0.f, o1 = 0.f;
+  for (int p = 0; p < num_parts; ++p) {
+    const float mp = ml_ws[(base + p) * 2];
+    if (mp <= kNegInf) continue;
+    const float f = __expf(mp - gm);
+    L += ml_ws[(base + p) * 2 + 1] * f;
+    const float2 a =
+        reinterpret_cast<const float2*>(acc_ws + (base + p) * D)[lane];
+    o0 = fmaf(a.x, f, o0);
+    o1 = fmaf(a.y, f, o1);
+  }
+  const float inv = 1.f / L;
+  u16* o_row = out + (static_cast<int64_t>(seq) * num_heads + h) * D;
+  const u32 obits = static_cast<u32>(f32_to_bf16(o0 * inv)) |
+                    (static_cast<u32>(f32_to_bf16(o1 * inv)) << 16);
+  *reinterpret_cast<u32*>(o_row + 2 * lane) = obits;
+}
+
+void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
+                              const u16* q, const u16* k_cache,
                               const u16* v_cache, const int* block_tables,
                               const int* seq_lens, int num_seqs,
                               int64_t q_stride, int max_blocks,
                               int num_kv_heads, int head_dim, int group,
-                              float scale, hipStream_t stream) {
-  dim3 grid(num_seqs, num_kv_heads), block(kNWaves * kWaveSize);
-#define FI_LAUNCH(DD, GG)                                                    \
-  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG>), grid, block, 0,     \
-                     stream, out, q, k_cache, v_cache, block_tables,         \
-                     seq_lens, q_stride, max_blocks, num_kv_heads, scale)
+                              int num_parts, float scale, hipStream_t stream) {
+  dim3 grid(num_seqs, num_kv_heads, num_parts), block(kNWaves * kWaveSize);
+#define FI_LAUNCH(DD, GG)                                                     \
+  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG>), grid, block, 0,      \
+                     stream, out, ml_ws, acc_ws, q, k_cache, v_cache,         \
+                     block_tables, seq_lens, q_stride, max_blocks,            \
+                     num_kv_heads, scale);                                    \
+  if (num_parts > 1) {                                                        \
+    dim3 rgrid(num_seqs, num_kv_heads * GG), rblock(kWaveSize);               \
+    hipLaunchKernelGGL((paged_attn_reduce_kernel<DD>), rgrid, rblock, 0,      \
+                       stream, out, ml_ws, acc_ws, num_parts);                \
+  }
   if (head_dim == 128) {
     switch (group) {
       case 1: FI_LAUNCH(128, 1); break;

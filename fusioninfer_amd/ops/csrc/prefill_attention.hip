@@ -114,31 +114,34 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
 
     // ---- cooperative staging: K [64][D] swizzled; V^T [D][64] swizzled ----
     {
-      // 512 threads: kv = tid/8 (64 rows), c16 = tid%8 (8 x 16-elem chunks)
-      const int kv_r = tid >> 3;
-      const int c16 = tid & 7;
-      const int src = min(kv0 + kv_r, seq_len - 1);
-      const u16* krow = k + (seq_start + src) * k_stride +
-                        static_cast<int64_t>(kv_head) * D + c16 * 16;
-      const u16* vrow = v + (seq_start + src) * v_stride +
-                        static_cast<int64_t>(kv_head) * D + c16 * 16;
-      bf16x8 k0 = *reinterpret_cast<const bf16x8*>(krow);
-      bf16x8 k1 = *reinterpret_cast<const bf16x8*>(krow + 8);
-      bf16x8 v0 = *reinterpret_cast<const bf16x8*>(vrow);
-      bf16x8 v1 = *reinterpret_cast<const bf16x8*>(vrow + 8);
-      char* kbase = reinterpret_cast<char*>(k_lds);
-      *reinterpret_cast<bf16x8*>(kbase + swz<kKRowB>(kv_r, c16 * 32)) = k0;
-      *reinterpret_cast<bf16x8*>(kbase + swz<kKRowB>(kv_r, c16 * 32 + 16)) = k1;
-      // V transpose: element (kv_r, d) -> vt row d, col kv_r
-      char* vbase = reinterpret_cast<char*>(vt_lds);
+      // one unit = 16 consecutive elements of one K/V row
+      constexpr int kChunks = D / 16;  // 16-elem chunks per row
+      for (int u = tid; u < kKVTile * kChunks; u += kWaves * kWaveSize) {
+        const int kv_r = u / kChunks;
+        const int c16 = u % kChunks;
+        const int src = min(kv0 + kv_r, seq_len - 1);
+        const u16* krow = k + (seq_start + src) * k_stride +
+                          static_cast<int64_t>(kv_head) * D + c16 * 16;
+        const u16* vrow = v + (seq_start + src) * v_stride +
+                          static_cast<int64_t>(kv_head) * D + c16 * 16;
+        bf16x8 k0 = *reinterpret_cast<const bf16x8*>(krow);
+        bf16x8 k1 = *reinterpret_cast<const bf16x8*>(krow + 8);
+        bf16x8 v0 = *reinterpret_cast<const bf16x8*>(vrow);
+        bf16x8 v1 = *reinterpret_cast<const bf16x8*>(vrow + 8);
+        char* kbase = reinterpret_cast<char*>(k_lds);
+        *reinterpret_cast<bf16x8*>(kbase + swz<kKRowB>(kv_r, c16 * 32)) = k0;
+        *reinterpret_cast<bf16x8*>(kbase + swz<kKRowB>(kv_r, c16 * 32 + 16)) = k1;
+        // V transpose: element (kv_r, d) -> vt row d, col kv_r
+        char* vbase = reinterpret_cast<char*>(vt_lds);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int d0 = c16 * 16 + j;
-        const int d1 = d0 + 8;
-        *reinterpret_cast<u16*>(
-            vbase + swz<kVTRowB>(d0, kv_r * 2)) = v0.h[j];
-        *reinterpret_cast<u16*>(
-            vbase + swz<kVTRowB>(d1, kv_r * 2)) = v1.h[j];
+        for (int j = 0; j < 8; ++j) {
+          const int d0 = c16 * 16 + j;
+          const int d1 = d0 + 8;
+          *reinterpret_cast<u16*>(
+              vbase + swz<kVTRowB>(d0, kv_r * 2)) = v0.h[j];
+          *reinterpret_cast<u16*>(
+              vbase + swz<kVTRowB>(d1, kv_r * 2)) = v1.h[j];
+        }
       }
     }
     __syncthreads();

@@ -5,8 +5,11 @@ Prints per-shape timing + achieved TFLOP/s (attention) or GB/s (memory ops).
 """
 
 import math
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
