@@ -72,10 +72,15 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
   const int row0 = wg_row0 + wave * kQPerWave;      // this wave's first q row
   const bool active = row0 < seq_len;
 
-  // LDS: K tile + V^T tile (workgroup-shared) + P tiles (per-wave)
-  __shared__ u16 k_lds[kKVTile * D];            // swizzled rows of 256 B
-  __shared__ u16 vt_lds[D * kKVTile];           // transposed, swizzled 128-B rows
-  __shared__ u16 p_lds[kWaves][kQPerWave * kKVTile];  // swizzled 128-B rows
+  // LDS: double-buffered K (glds target) + V^T tile + P tiles (per-wave).
+  // ONE __shared__ object: a second one makes hipcc drain vmcnt(0) before
+  // every ds_read beside a glds pipeline (guide §5 ".s-level traps" (a)).
+  __shared__ u16 smem[2 * kKVTile * D + D * kKVTile +
+                      kWaves * kQPerWave * kKVTile];
+  u16* k_lds0 = smem;                            // swizzled K rows, buffer 0
+  u16* k_lds1 = smem + kKVTile * D;              // buffer 1
+  u16* vt_lds = smem + 2 * kKVTile * D;          // transposed V, swizzled rows
+  u16* p_lds = vt_lds + D * kKVTile + wave * (kQPerWave * kKVTile);
 
   // ---- Q fragments: a_q[rb][kb], lane holds Q[row0+rb*16+col][kb*32+hi*8..]
   short8 a_q[2][KB];
@@ -109,53 +114,87 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
   const int num_kv_tiles = wg_q_max / kKVTile + 1;
   const int my_q_max = min(row0 + kQPerWave, seq_len) - 1;  // per-wave
 
+  // ---- staging helpers -------------------------------------------------
+  // K tile -> LDS by global_load_lds (direct DMA, no VGPR round trip).
+  // glds writes lane-linear (base + lane*16), so the T2 swizzle moves to
+  // the per-lane SOURCE address (guide §5.4 rule 21).
+  constexpr int kKTileBytes = kKVTile * kKRowB;
+  constexpr int kGldsPerWave = kKTileBytes / (kWaves * kWaveSize * 16);
+  const u16* k_head_base = k + static_cast<int64_t>(kv_head) * D;
+  auto stage_k_glds = [&](int t, u16* kbuf) {
+    const int kv0 = t * kKVTile;
+#pragma unroll
+    for (int i = 0; i < kGldsPerWave; ++i) {
+      const int base_off =
+          wave * (kKTileBytes / kWaves) + i * (kWaveSize * 16);
+      const int X = base_off + lane * 16;
+      const int row = X / kKRowB;
+      const int sbyte = X % kKRowB;
+      constexpr int kMask = (kKRowB / 16 > 16 ? 16 : kKRowB / 16) - 1;
+      const int byte = sbyte ^ ((row & kMask) << 4);
+      const int src_row = min(kv0 + row, seq_len - 1);
+      const u16* gsrc =
+          k_head_base + (seq_start + src_row) * k_stride + byte / 2;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gsrc,
+          (__attribute__((address_space(3))) void*)(
+              reinterpret_cast<char*>(kbuf) + base_off),
+          16, 0, 0);
+    }
+  };
+  // V: per-thread register load of one 16-elem row chunk (issued early so
+  // HBM latency hides under MFMA — guide T14), transposed into LDS later.
+  constexpr int kChunks = D / 16;
+  const int v_kv = tid % kKVTile;          // this thread's kv row
+  const int v_c16 = tid / kKVTile;         // 16-elem chunk (D=128: 0..7)
+  const u16* v_head_base = v + static_cast<int64_t>(kv_head) * D;
+  bf16x8 vreg0, vreg1;
+  auto vload = [&](int t) {
+    if (v_c16 >= kChunks) return;  // D=64: only 4 chunks per row
+    const int src = min(t * kKVTile + v_kv, seq_len - 1);
+    const u16* vrow =
+        v_head_base + (seq_start + src) * v_stride + v_c16 * 16;
+    vreg0 = *reinterpret_cast<const bf16x8*>(vrow);
+    vreg1 = *reinterpret_cast<const bf16x8*>(vrow + 8);
+  };
+  auto vwrite = [&]() {
+    if (v_c16 >= kChunks) return;
+    char* vbase = reinterpret_cast<char*>(vt_lds);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      *reinterpret_cast<u16*>(
+          vbase + swz<kVTRowB>(v_c16 * 16 + j, v_kv * 2)) = vreg0.h[j];
+      *reinterpret_cast<u16*>(
+          vbase + swz<kVTRowB>(v_c16 * 16 + j + 8, v_kv * 2)) = vreg1.h[j];
+    }
+  };
+
+  // ---- software-pipelined main loop -------------------------------------
+  // invariant at loop top: K(t) resident in kbuf(t&1) (glds issued and
+  // drained by the previous barrier), V(t) in registers.
+  stage_k_glds(0, k_lds0);
+  vload(0);
+  __syncthreads();  // drains the glds (vmcnt 0) and publishes K(0)
+
   for (int t = 0; t < num_kv_tiles; ++t) {
     const int kv0 = t * kKVTile;
+    u16* kbuf = (t & 1) ? k_lds1 : k_lds0;
+    u16* kbuf_next = (t & 1) ? k_lds0 : k_lds1;
 
-    // ---- cooperative staging: K [64][D] swizzled; V^T [D][64] swizzled ----
-    {
-      // one unit = 16 consecutive elements of one K/V row
-      constexpr int kChunks = D / 16;  // 16-elem chunks per row
-      for (int u = tid; u < kKVTile * kChunks; u += kWaves * kWaveSize) {
-        const int kv_r = u / kChunks;
-        const int c16 = u % kChunks;
-        const int src = min(kv0 + kv_r, seq_len - 1);
-        const u16* krow = k + (seq_start + src) * k_stride +
-                          static_cast<int64_t>(kv_head) * D + c16 * 16;
-        const u16* vrow = v + (seq_start + src) * v_stride +
-                          static_cast<int64_t>(kv_head) * D + c16 * 16;
-        bf16x8 k0 = *reinterpret_cast<const bf16x8*>(krow);
-        bf16x8 k1 = *reinterpret_cast<const bf16x8*>(krow + 8);
-        bf16x8 v0 = *reinterpret_cast<const bf16x8*>(vrow);
-        bf16x8 v1 = *reinterpret_cast<const bf16x8*>(vrow + 8);
-        char* kbase = reinterpret_cast<char*>(k_lds);
-        *reinterpret_cast<bf16x8*>(kbase + swz<kKRowB>(kv_r, c16 * 32)) = k0;
-        *reinterpret_cast<bf16x8*>(kbase + swz<kKRowB>(kv_r, c16 * 32 + 16)) = k1;
-        // V transpose: element (kv_r, d) -> vt row d, col kv_r
-        char* vbase = reinterpret_cast<char*>(vt_lds);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d0 = c16 * 16 + j;
-          const int d1 = d0 + 8;
-          *reinterpret_cast<u16*>(
-              vbase + swz<kVTRowB>(d0, kv_r * 2)) = v0.h[j];
-          *reinterpret_cast<u16*>(
-              vbase + swz<kVTRowB>(d1, kv_r * 2)) = v1.h[j];
-        }
-      }
-    }
-    __syncthreads();
+    vwrite();  // V(t) -> vt_lds (read after the mid barrier)
+    if (t + 1 < num_kv_tiles)
+      stage_k_glds(t + 1, kbuf_next);  // flies under QK^T(t)
 
     const bool compute = active && kv0 <= my_q_max;
     if (compute) {
-      const char* kbase = reinterpret_cast<const char*>(k_lds);
-      const char* vbase = reinterpret_cast<const char*>(vt_lds);
-      char* pbase = reinterpret_cast<char*>(p_lds[wave]);
+      const char* kbase = reinterpret_cast<const char*>(kbuf);
+      char* pbase = reinterpret_cast<char*>(p_lds);
 
 #pragma unroll
       for (int rb = 0; rb < 2; ++rb) {
         // ---- QK^T: S[16 x 64] ----
         floatx4 s_acc[4];
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int cb4 = 0; cb4 < 4; ++cb4) {
           s_acc[cb4] = {0.f, 0.f, 0.f, 0.f};
@@ -168,9 +207,9 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
                 a_q[rb][kb], b_k, s_acc[cb4], 0, 0, 0);
           }
         }
+        __builtin_amdgcn_s_setprio(0);
 
-        // ---- mask + online softmax over the 16x64 block ----
-        float p[4][4];
+        // ---- mask + online softmax over the 16x64 block (in s_acc) ----
         const int q_base = row0 + rb * 16 + hi * 4;
 #pragma unroll
         for (int cb4 = 0; cb4 < 4; ++cb4) {
@@ -181,47 +220,58 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
             float sv = s_acc[cb4][r] * scale;
             if (kv_pos > q_pos || q_pos >= seq_len || kv_pos >= seq_len)
               sv = kPNegInf;
-            p[cb4][r] = sv;
+            s_acc[cb4][r] = sv;
           }
         }
-        float alpha[4];
+        // o_acc is rescaled HERE (phase 1) so nothing but P crosses the
+        // barrier into the PV phase
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          float rm = fmaxf(fmaxf(p[0][r], p[1][r]), fmaxf(p[2][r], p[3][r]));
+          float rm = fmaxf(fmaxf(s_acc[0][r], s_acc[1][r]),
+                           fmaxf(s_acc[2][r], s_acc[3][r]));
 #pragma unroll
           for (int off = 1; off < 16; off <<= 1)
             rm = fmaxf(rm, __shfl_xor(rm, off, 64));
           const float m_new = fmaxf(m[rb][r], rm);
+          float alpha;
           if (m_new <= kPNegInf) {  // row fully masked so far
-            alpha[r] = 0.f;
-#pragma unroll
-            for (int cb4 = 0; cb4 < 4; ++cb4) p[cb4][r] = 0.f;
-            continue;
+            alpha = 0.f;
+          } else {
+            alpha = (m[rb][r] <= kPNegInf) ? 0.f : __expf(m[rb][r] - m_new);
+            m[rb][r] = m_new;
           }
-          alpha[r] = (m[rb][r] <= kPNegInf) ? 0.f : __expf(m[rb][r] - m_new);
-          m[rb][r] = m_new;
+#pragma unroll
+          for (int cb = 0; cb < CB; ++cb) o_acc[rb][cb][r] *= alpha;
           float rs = 0.f;
 #pragma unroll
           for (int cb4 = 0; cb4 < 4; ++cb4) {
-            p[cb4][r] =
-                (p[cb4][r] <= kPNegInf) ? 0.f : __expf(p[cb4][r] - m_new);
-            rs += p[cb4][r];
+            const float pv = (m_new <= kPNegInf || s_acc[cb4][r] <= kPNegInf)
+                                 ? 0.f
+                                 : __expf(s_acc[cb4][r] - m_new);
+            rs += pv;
+            *reinterpret_cast<u16*>(
+                pbase + swz<kVTRowB>(rb * 16 + hi * 4 + r,
+                                     (cb4 * 16 + col) * 2)) = f32_to_bf16(pv);
           }
 #pragma unroll
           for (int off = 1; off < 16; off <<= 1) rs += __shfl_xor(rs, off, 64);
-          l[rb][r] = l[rb][r] * alpha[r] + rs;
+          l[rb][r] = l[rb][r] * alpha + rs;
         }
+      }
+    }
 
-        // ---- P -> per-wave LDS (bf16, swizzled 128-B rows) ----
-#pragma unroll
-        for (int cb4 = 0; cb4 < 4; ++cb4)
-#pragma unroll
-          for (int r = 0; r < 4; ++r)
-            *reinterpret_cast<u16*>(
-                pbase + swz<kVTRowB>(rb * 16 + hi * 4 + r,
-                                        (cb4 * 16 + col) * 2)) =
-                f32_to_bf16(p[cb4][r]);
+    // publish vt_lds(t) to every wave; also drains the K(t+1) glds, whose
+    // latency QK^T just covered
+    __syncthreads();
+    if (t + 1 < num_kv_tiles)
+      vload(t + 1);  // flies under PV(t); consumed by vwrite next iteration
 
+    if (compute) {
+      const char* vbase = reinterpret_cast<const char*>(vt_lds);
+      char* pbase = reinterpret_cast<char*>(p_lds);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int rb = 0; rb < 2; ++rb) {
         // ---- PV: O[16 x D] += P[16 x 64] V[64 x D] ----
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {  // kv 64 = 2 MFMA k-depths
@@ -231,17 +281,14 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
           for (int cb = 0; cb < CB; ++cb) {
             const short8 b_v = *reinterpret_cast<const short8*>(
                 vbase + swz<kVTRowB>(cb * 16 + col, kc * 64 + hi * 16));
-            floatx4 prev = o_acc[rb][cb];
-            if (kc == 0) {
-#pragma unroll
-              for (int r = 0; r < 4; ++r) prev[r] *= alpha[r];
-            }
             o_acc[rb][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a_p, b_v, prev, 0, 0, 0);
+                a_p, b_v, o_acc[rb][cb], 0, 0, 0);
           }
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
+    // PV(t) reads of vt_lds done before vwrite(t+1) overwrites it
     __syncthreads();
   }
 
