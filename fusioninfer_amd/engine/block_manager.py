@@ -62,8 +62,11 @@ class BlockManager:
             h = 0
             for b in range(n):
                 start, end = b * self.block_size, (b + 1) * self.block_size
-                if end > seq.num_prompt_tokens:
-                    break  # partial last block never cache-hits
+                # `>=`: at least ONE prompt token must be recomputed so the
+                # sequence produces logits — and that token then lands in a
+                # FRESH block, so shared (cached) blocks are never written.
+                if end >= seq.num_prompt_tokens:
+                    break
                 h = hash((h, tuple(seq.prompt_token_ids[start:end])))
                 blk = self.hash_to_block.get(h)
                 if blk is None:

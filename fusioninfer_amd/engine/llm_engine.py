@@ -46,7 +46,7 @@ class LLMEngine:
         self.block_manager = BlockManager(
             self.runner.num_gpu_blocks,
             cfg.cache.block_size,
-            enable_prefix_caching=False,  # v1: see block_manager.py note
+            enable_prefix_caching=cfg.cache.enable_prefix_caching,
         )
         self.scheduler = Scheduler(cfg.scheduler, self.block_manager)
         self.sampler = Sampler(cfg.seed, device)

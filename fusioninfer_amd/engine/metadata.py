@@ -17,10 +17,13 @@ class AttnMetadata:
     num_decode_tokens: int
     positions: torch.Tensor            # [T] int32, all tokens
     slot_mapping: torch.Tensor         # [T] int32, cache slot per token
-    # prefill segment
-    cu_seqlens: Optional[torch.Tensor] = None    # [P+1] int32
+    # prefill segment (context attention over the paged cache: q rows are
+    # only the NEW tokens; cached prefix tokens are skipped entirely)
+    cu_seqlens: Optional[torch.Tensor] = None    # [P+1] int32, NEW tokens
     tile_seq: Optional[torch.Tensor] = None      # kernel tile table
     tile_row0: Optional[torch.Tensor] = None
+    prefill_block_tables: Optional[torch.Tensor] = None  # [P, max_blocks]
+    prefill_seq_lens_k: Optional[torch.Tensor] = None    # [P] total ctx
     # decode segment
     block_tables: Optional[torch.Tensor] = None  # [Dq, max_blocks] int32
     seq_lens: Optional[torch.Tensor] = None      # [Dq] int32 (ctx incl. current)

@@ -81,21 +81,33 @@ class ModelRunner:
 
     # ------------------------------------------------------------ prefill
     def _prepare_prefill(self, seqs: List[Sequence], bm: BlockManager):
+        """Builds the batch over each sequence's NEW tokens only — prefix-
+        cache hits (seq.num_cached_tokens) are skipped; attention runs over
+        the paged cache (context attention)."""
         input_ids: List[int] = []
         positions: List[int] = []
         slots: List[int] = []
         cu = [0]
-        lens = []
+        new_lens = []
+        total_lens = []
         for seq in seqs:
             toks = seq.all_token_ids
             L = len(toks)
-            input_ids.extend(toks)
-            positions.extend(range(L))
-            slots.extend(bm.slot_for(seq, p) for p in range(L))
-            cu.append(cu[-1] + L)
-            lens.append(L)
+            C = seq.num_cached_tokens
+            input_ids.extend(toks[C:])
+            positions.extend(range(C, L))
+            slots.extend(bm.slot_for(seq, p) for p in range(C, L))
+            cu.append(cu[-1] + (L - C))
+            new_lens.append(L - C)
+            total_lens.append(L)
         dev = self.device
-        tile_seq, tile_row0 = ops_mod.build_prefill_tiles(lens, device=dev)
+        tile_seq, tile_row0 = ops_mod.build_prefill_tiles(new_lens, device=dev)
+        max_blocks = max(len(s.block_ids) for s in seqs)
+        bt = torch.zeros((len(seqs), max_blocks), dtype=torch.int32)
+        for i, s in enumerate(seqs):
+            bt[i, : len(s.block_ids)] = torch.tensor(
+                s.block_ids, dtype=torch.int32
+            )
         meta = AttnMetadata(
             num_prefill_tokens=cu[-1],
             num_decode_tokens=0,
@@ -104,6 +116,10 @@ class ModelRunner:
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
             tile_seq=tile_seq,
             tile_row0=tile_row0,
+            prefill_block_tables=bt.to(dev),
+            prefill_seq_lens_k=torch.tensor(
+                total_lens, dtype=torch.int32, device=dev
+            ),
         )
         ids = torch.tensor(input_ids, dtype=torch.long, device=dev)
         logits_idx = torch.tensor(

@@ -85,11 +85,15 @@ class Attention(nn.Module):
         np_, nd = meta.num_prefill_tokens, meta.num_decode_tokens
         out = torch.empty(T, self.q_size, dtype=x.dtype, device=x.device)
         if np_ > 0:
-            o = ops.prefill_attention(
+            # context attention over the paged cache (the new tokens' K/V
+            # were just written above); cached prefixes are never recomputed
+            o = ops.prefill_attention_paged(
                 q[:np_].view(np_, self.num_heads, self.head_dim),
-                k[:np_].reshape(np_, self.num_kv_heads, self.head_dim),
-                v[:np_].reshape(np_, self.num_kv_heads, self.head_dim),
+                k_cache,
+                v_cache,
+                meta.prefill_block_tables,
                 meta.cu_seqlens,
+                meta.prefill_seq_lens_k,
                 self.scale,
                 tile_seq=meta.tile_seq,
                 tile_row0=meta.tile_row0,

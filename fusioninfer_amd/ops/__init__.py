@@ -217,4 +217,35 @@ def prefill_attention(
     return ref.prefill_attention(q, k, v, cu_seqlens, scale, causal=True)
 
 
+def prefill_attention_paged(
+    q: torch.Tensor,             # [Tnew, Hq, D] new tokens only
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,  # [S, max_blocks] int32
+    cu_seqlens_q: torch.Tensor,  # [S+1] int32 over new tokens
+    seq_lens_k: torch.Tensor,    # [S] int32 total context length
+    scale: Optional[float] = None,
+    tile_seq: Optional[torch.Tensor] = None,
+    tile_row0: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Context attention: new tokens attend over the paged cache (their own
+    K/V must already be written via reshape_and_cache)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        _require_native()
+        if tile_seq is None:
+            lens = (cu_seqlens_q[1:] - cu_seqlens_q[:-1]).tolist()
+            tile_seq, tile_row0 = build_prefill_tiles(lens, device=q.device)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        _C.prefill_attention_paged(
+            out, q, k_cache, v_cache, tile_seq, tile_row0, cu_seqlens_q,
+            block_tables, seq_lens_k, scale,
+        )
+        return out
+    return ref.prefill_attention_paged(
+        q, k_cache, v_cache, block_tables, cu_seqlens_q, seq_lens_k, scale
+    )
+
+
 compute_cos_sin_cache = ref.compute_cos_sin_cache

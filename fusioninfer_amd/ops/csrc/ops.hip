@@ -27,8 +27,9 @@ void launch_paged_attn_decode(u16*, float*, float*, const u16*, const u16*,
                               int64_t, int, int, int, int, int, float,
                               hipStream_t);
 void launch_prefill_attn(u16*, const u16*, const u16*, const u16*, const int*,
-                         const int*, const int*, int, int64_t, int64_t,
-                         int64_t, int, int, int, float, hipStream_t);
+                         const int*, const int*, const int*, const int*, int,
+                         int, int64_t, int64_t, int64_t, int, int, int, float,
+                         hipStream_t);
 
 }  // namespace fi
 
@@ -201,8 +202,37 @@ void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
   fi::launch_prefill_attn(
       bf16_ptr(out), bf16_cptr(q), bf16_cptr(k), bf16_cptr(v),
       tile_seq.data_ptr<int>(), tile_row0.data_ptr<int>(),
-      cu_seqlens.data_ptr<int>(), tile_seq.size(0), q.stride(0), k.stride(0),
-      v.stride(0), num_q_heads, num_kv_heads, head_dim,
+      cu_seqlens.data_ptr<int>(), nullptr, nullptr, 0, tile_seq.size(0),
+      q.stride(0), k.stride(0), v.stride(0), num_q_heads, num_kv_heads,
+      head_dim, static_cast<float>(scale), current_stream());
+}
+
+void prefill_attention_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
+                             at::Tensor v_cache, at::Tensor tile_seq,
+                             at::Tensor tile_row0, at::Tensor cu_seqlens,
+                             at::Tensor block_tables, at::Tensor seq_lens_k,
+                             double scale) {
+  CHECK_BF16_CUDA(out);
+  CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(k_cache);
+  TORCH_CHECK(tile_seq.scalar_type() == at::kInt &&
+              tile_row0.scalar_type() == at::kInt &&
+              cu_seqlens.scalar_type() == at::kInt &&
+              block_tables.scalar_type() == at::kInt &&
+              seq_lens_k.scalar_type() == at::kInt);
+  TORCH_CHECK(block_tables.is_contiguous());
+  TORCH_CHECK(q.dim() == 3);
+  const int num_q_heads = q.size(1);
+  const int head_dim = q.size(2);
+  const int num_kv_heads = k_cache.size(1);
+  TORCH_CHECK(k_cache.size(2) == 16, "cache block_size must be 16");
+  TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == head_dim);
+  fi::launch_prefill_attn(
+      bf16_ptr(out), bf16_cptr(q), bf16_cptr(k_cache), bf16_cptr(v_cache),
+      tile_seq.data_ptr<int>(), tile_row0.data_ptr<int>(),
+      cu_seqlens.data_ptr<int>(), block_tables.data_ptr<int>(),
+      seq_lens_k.data_ptr<int>(), block_tables.size(1), tile_seq.size(0),
+      q.stride(0), 0, 0, num_q_heads, num_kv_heads, head_dim,
       static_cast<float>(scale), current_stream());
 }
 
@@ -221,4 +251,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention_decode", &paged_attention_decode,
         "paged decode attention");
   m.def("prefill_attention", &prefill_attention, "varlen causal MFMA prefill");
+  m.def("prefill_attention_paged", &prefill_attention_paged,
+        "varlen causal MFMA prefill over the paged cache (context attention)");
 }

@@ -41,15 +41,21 @@ FI_DEV int swz(int row, int byte_in_row) {
   return row * ROWB + (byte_in_row ^ ((row & kMask) << 4));
 }
 
-template <int D>
+// PAGED: K/V come from the paged cache ([B, Hk, 16, D]) via block_tables,
+// and q rows are only the NEW tokens of each sequence (context attention —
+// prefix-cache hits / chunked prefill skip cached tokens).
+template <int D, bool PAGED>
 __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     u16* __restrict__ out,        // [T, Hq, D]
     const u16* __restrict__ q,    // [T] rows, stride q_stride
-    const u16* __restrict__ k,    // [T] rows, stride k_stride
-    const u16* __restrict__ v,    // [T] rows, stride v_stride
+    const u16* __restrict__ k,    // dense rows OR k_cache when PAGED
+    const u16* __restrict__ v,    // dense rows OR v_cache when PAGED
     const int* __restrict__ tile_seq,    // [ntiles] sequence index
-    const int* __restrict__ tile_row0,   // [ntiles] first q row (within seq)
-    const int* __restrict__ cu_seqlens,  // [nseqs+1]
+    const int* __restrict__ tile_row0,   // [ntiles] first NEW q row
+    const int* __restrict__ cu_seqlens,  // [nseqs+1] over NEW tokens
+    const int* __restrict__ block_tables,  // [S, max_blocks] (PAGED)
+    const int* __restrict__ seq_lens_k,    // [S] total ctx len (PAGED)
+    const int max_blocks,
     const int64_t q_stride, const int64_t k_stride, const int64_t v_stride,
     const int num_q_heads, const int num_kv_heads, const float scale) {
   constexpr int KB = D / 32;   // QK^T k-chunks over the head dim
@@ -67,10 +73,23 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
 
   const int seq = tile_seq[blockIdx.x];
   const int seq_start = cu_seqlens[seq];
-  const int seq_len = cu_seqlens[seq + 1] - seq_start;
+  const int seq_len = cu_seqlens[seq + 1] - seq_start;  // NEW q rows
+  const int k_len = PAGED ? seq_lens_k[seq] : seq_len;  // total kv rows
+  const int ctx_start = k_len - seq_len;  // cached tokens before q row 0
   const int wg_row0 = tile_row0[blockIdx.x];
   const int row0 = wg_row0 + wave * kQPerWave;      // this wave's first q row
   const bool active = row0 < seq_len;
+  const int* bt_row = PAGED ? block_tables + static_cast<int64_t>(seq) * max_blocks
+                            : nullptr;
+  // element offset of kv row r (absolute position) in the k/v source
+  auto kv_off = [&](int r, int64_t dense_stride) -> int64_t {
+    if (PAGED) {
+      const int blk = bt_row[r >> 4];
+      return ((static_cast<int64_t>(blk) * num_kv_heads + kv_head) * 16 +
+              (r & 15)) * D;
+    }
+    return (seq_start + r) * dense_stride + static_cast<int64_t>(kv_head) * D;
+  };
 
   // LDS: double-buffered K (glds target) + V^T tile + P tiles (per-wave).
   // ONE __shared__ object: a second one makes hipcc drain vmcnt(0) before
@@ -109,10 +128,10 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     for (int cb = 0; cb < CB; ++cb) o_acc[rb][cb] = {0.f, 0.f, 0.f, 0.f};
   }
 
-  // kv range: the whole workgroup iterates to its max causal row
-  const int wg_q_max = min(wg_row0 + kQBlock, seq_len) - 1;
+  // kv range: the whole workgroup iterates to its max causal ABSOLUTE row
+  const int wg_q_max = ctx_start + min(wg_row0 + kQBlock, seq_len) - 1;
   const int num_kv_tiles = wg_q_max / kKVTile + 1;
-  const int my_q_max = min(row0 + kQPerWave, seq_len) - 1;  // per-wave
+  const int my_q_max = ctx_start + min(row0 + kQPerWave, seq_len) - 1;
 
   // ---- staging helpers -------------------------------------------------
   // K tile -> LDS by global_load_lds (direct DMA, no VGPR round trip).
@@ -120,7 +139,6 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
   // the per-lane SOURCE address (guide §5.4 rule 21).
   constexpr int kKTileBytes = kKVTile * kKRowB;
   constexpr int kGldsPerWave = kKTileBytes / (kWaves * kWaveSize * 16);
-  const u16* k_head_base = k + static_cast<int64_t>(kv_head) * D;
   auto stage_k_glds = [&](int t, u16* kbuf) {
     const int kv0 = t * kKVTile;
 #pragma unroll
@@ -132,9 +150,8 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
       const int sbyte = X % kKRowB;
       constexpr int kMask = (kKRowB / 16 > 16 ? 16 : kKRowB / 16) - 1;
       const int byte = sbyte ^ ((row & kMask) << 4);
-      const int src_row = min(kv0 + row, seq_len - 1);
-      const u16* gsrc =
-          k_head_base + (seq_start + src_row) * k_stride + byte / 2;
+      const int src_row = min(kv0 + row, k_len - 1);
+      const u16* gsrc = k + kv_off(src_row, k_stride) + byte / 2;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)gsrc,
           (__attribute__((address_space(3))) void*)(
@@ -147,13 +164,11 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
   constexpr int kChunks = D / 16;
   const int v_kv = tid % kKVTile;          // this thread's kv row
   const int v_c16 = tid / kKVTile;         // 16-elem chunk (D=128: 0..7)
-  const u16* v_head_base = v + static_cast<int64_t>(kv_head) * D;
   bf16x8 vreg0, vreg1;
   auto vload = [&](int t) {
     if (v_c16 >= kChunks) return;  // D=64: only 4 chunks per row
-    const int src = min(t * kKVTile + v_kv, seq_len - 1);
-    const u16* vrow =
-        v_head_base + (seq_start + src) * v_stride + v_c16 * 16;
+    const int src = min(t * kKVTile + v_kv, k_len - 1);
+    const u16* vrow = v + kv_off(src, v_stride) + v_c16 * 16;
     vreg0 = *reinterpret_cast<const bf16x8*>(vrow);
     vreg1 = *reinterpret_cast<const bf16x8*>(vrow + 8);
   };
@@ -213,12 +228,12 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
         const int q_base = row0 + rb * 16 + hi * 4;
 #pragma unroll
         for (int cb4 = 0; cb4 < 4; ++cb4) {
-          const int kv_pos = kv0 + cb4 * 16 + col;
+          const int kv_pos = kv0 + cb4 * 16 + col;  // absolute position
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            const int q_pos = q_base + r;
+            const int q_pos = ctx_start + q_base + r;  // absolute position
             float sv = s_acc[cb4][r] * scale;
-            if (kv_pos > q_pos || q_pos >= seq_len || kv_pos >= seq_len)
+            if (kv_pos > q_pos || q_base + r >= seq_len || kv_pos >= k_len)
               sv = kPNegInf;
             s_acc[cb4][r] = sv;
           }
@@ -313,22 +328,26 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
 
 void launch_prefill_attn(u16* out, const u16* q, const u16* k, const u16* v,
                          const int* tile_seq, const int* tile_row0,
-                         const int* cu_seqlens, int ntiles, int64_t q_stride,
-                         int64_t k_stride, int64_t v_stride, int num_q_heads,
-                         int num_kv_heads, int head_dim, float scale,
-                         hipStream_t stream) {
+                         const int* cu_seqlens, const int* block_tables,
+                         const int* seq_lens_k, int max_blocks, int ntiles,
+                         int64_t q_stride, int64_t k_stride, int64_t v_stride,
+                         int num_q_heads, int num_kv_heads, int head_dim,
+                         float scale, hipStream_t stream) {
   dim3 grid(ntiles, num_q_heads), block(kWaves * kWaveSize);
+  const bool paged = block_tables != nullptr;
+#define FI_PF_LAUNCH(DD, PP)                                                  \
+  hipLaunchKernelGGL((prefill_attn_kernel<DD, PP>), grid, block, 0, stream,   \
+                     out, q, k, v, tile_seq, tile_row0, cu_seqlens,           \
+                     block_tables, seq_lens_k, max_blocks, q_stride,          \
+                     k_stride, v_stride, num_q_heads, num_kv_heads, scale)
   if (head_dim == 128) {
-    hipLaunchKernelGGL((prefill_attn_kernel<128>), grid, block, 0, stream, out,
-                       q, k, v, tile_seq, tile_row0, cu_seqlens, q_stride,
-                       k_stride, v_stride, num_q_heads, num_kv_heads, scale);
+    if (paged) FI_PF_LAUNCH(128, true); else FI_PF_LAUNCH(128, false);
   } else if (head_dim == 64) {
-    hipLaunchKernelGGL((prefill_attn_kernel<64>), grid, block, 0, stream, out,
-                       q, k, v, tile_seq, tile_row0, cu_seqlens, q_stride,
-                       k_stride, v_stride, num_q_heads, num_kv_heads, scale);
+    if (paged) FI_PF_LAUNCH(64, true); else FI_PF_LAUNCH(64, false);
   } else {
     abort();
   }
+#undef FI_PF_LAUNCH
 }
 
 }  // namespace fi

@@ -167,6 +167,47 @@ def prefill_attention(
     return out.to(q.dtype)
 
 
+def prefill_attention_paged(
+    q: torch.Tensor,            # [Tnew, Hq, D] (new tokens only)
+    k_cache: torch.Tensor,      # [B, Hk, bs, D]
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,  # [S, max_blocks]
+    cu_seqlens_q: torch.Tensor,  # [S+1] over NEW tokens
+    seq_lens_k: torch.Tensor,    # [S] total context length
+    scale: float,
+) -> torch.Tensor:
+    """Context attention: new tokens attend over the paged cache (which
+    already contains their own K/V plus any cached prefix)."""
+    Tn, num_heads, head_dim = q.shape
+    num_kv_heads = k_cache.shape[1]
+    bs = k_cache.shape[2]
+    group = num_heads // num_kv_heads
+    out = torch.empty_like(q, dtype=torch.float32)
+    for s in range(cu_seqlens_q.numel() - 1):
+        qs, qe = int(cu_seqlens_q[s]), int(cu_seqlens_q[s + 1])
+        q_len = qe - qs
+        k_len = int(seq_lens_k[s])
+        ctx_start = k_len - q_len
+        nblk = (k_len + bs - 1) // bs
+        blocks = block_tables[s, :nblk].long()
+        keys = k_cache[blocks].permute(1, 0, 2, 3).reshape(num_kv_heads, -1, head_dim)
+        vals = v_cache[blocks].permute(1, 0, 2, 3).reshape(num_kv_heads, -1, head_dim)
+        keys = keys[:, :k_len].float()
+        vals = vals[:, :k_len].float()
+        qf = q[qs:qe].float()  # [q_len, Hq, D]
+        # causal mask with context offset
+        kv_pos = torch.arange(k_len, device=q.device)
+        q_pos = ctx_start + torch.arange(q_len, device=q.device)
+        mask = kv_pos.unsqueeze(0) > q_pos.unsqueeze(1)  # [q_len, k_len]
+        for h in range(num_heads):
+            kv_h = h // group
+            scores = qf[:, h] @ keys[kv_h].T * scale
+            scores = scores.masked_fill(mask, float("-inf"))
+            p = torch.softmax(scores, dim=-1)
+            out[qs:qe, h] = p @ vals[kv_h]
+    return out.to(q.dtype)
+
+
 def gather_kv_blocks(
     k_cache: torch.Tensor, v_cache: torch.Tensor, block_ids: torch.Tensor
 ) -> torch.Tensor:

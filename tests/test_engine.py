@@ -114,3 +114,37 @@ def test_block_manager_accounting():
     assert len(s.block_ids) == 4
     bm.free(s)
     assert bm.num_free() == 8
+
+
+def test_prefix_caching_end_to_end():
+    """Second identical-prompt request reuses cached blocks and produces
+    identical greedy tokens (context-attention path)."""
+    torch.manual_seed(0)
+    eng = make_engine()
+    assert eng.block_manager.enable_prefix_caching
+    prompt = [7, 3, 9, 1] * 12  # 48 tokens = 3 full blocks
+    out1 = eng.generate([prompt], SamplingParams(max_tokens=5))[0]
+    out2 = eng.generate([prompt], SamplingParams(max_tokens=5))[0]
+    assert out1.output_token_ids == out2.output_token_ids
+    # the second request must have hit the prefix cache (2 full blocks;
+    # the 3rd is capped so >=1 token is recomputed)
+    # (seq objects are gone; verify via the hash table state)
+    assert len(eng.block_manager.hash_to_block) >= 2
+
+
+def test_prefix_caching_shared_prefix_divergent_tail():
+    torch.manual_seed(0)
+    eng = make_engine()
+    base = [5, 1, 2, 6] * 10          # 40 tokens, 2 full blocks cacheable
+    p1 = base + [11, 12, 13]
+    p2 = base + [21, 22, 23, 24]
+    r1 = eng.generate([p1], SamplingParams(max_tokens=4))[0]
+    # second shares the 2-block prefix but diverges after
+    r2 = eng.generate([p2], SamplingParams(max_tokens=4))[0]
+    # oracle without caching
+    eng2 = make_engine()
+    eng2.block_manager.enable_prefix_caching = False
+    e1 = eng2.generate([p1], SamplingParams(max_tokens=4))[0]
+    e2 = eng2.generate([p2], SamplingParams(max_tokens=4))[0]
+    assert r1.output_token_ids == e1.output_token_ids
+    assert r2.output_token_ids == e2.output_token_ids

@@ -185,3 +185,37 @@ def test_prefill_attention_spiked_scores():
     out = ops.prefill_attention(q, k, v, cu)
     expected = ref.prefill_attention(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), 1.0 / math.sqrt(D))
     assert_close_bf16(out.cpu(), expected, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize(
+    "ctx_lens,new_lens",
+    [([48], [48]), ([128], [32]), ([64, 100, 300], [64, 36, 44])],
+)
+def test_prefill_attention_paged(ctx_lens, new_lens):
+    """Context attention over the paged cache vs the fp32 reference."""
+    torch.manual_seed(5)
+    Hq, Hk, D, bs = 8, 2, 128, 16
+    S = len(ctx_lens)
+    Tn = sum(new_lens)
+    max_blocks = max((c + bs - 1) // bs for c in ctx_lens)
+    total_blocks = sum((c + bs - 1) // bs for c in ctx_lens) + 1
+    q = torch.randn(Tn, Hq, D, dtype=torch.bfloat16, device=DEV)
+    k_cache = torch.randn(total_blocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV)
+    v_cache = torch.randn_like(k_cache)
+    bt = torch.zeros(S, max_blocks, dtype=torch.int32, device=DEV)
+    nxt = 1
+    for s, c in enumerate(ctx_lens):
+        n = (c + bs - 1) // bs
+        bt[s, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    cu_q = torch.tensor(
+        [0] + list(torch.tensor(new_lens).cumsum(0)), dtype=torch.int32, device=DEV
+    )
+    lens_k = torch.tensor(ctx_lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.prefill_attention_paged(q, k_cache, v_cache, bt, cu_q, lens_k, scale)
+    expected = ref.prefill_attention_paged(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), cu_q.cpu(),
+        lens_k.cpu(), scale,
+    )
+    assert_close_bf16(out.cpu(), expected, atol=3e-2, rtol=3e-2)
