@@ -443,3 +443,36 @@ class TestReconcile:
         assert comp["readyReplicas"] == 1
         assert comp["readyPods"] == 3
         assert comp["phase"] == "Deploying"
+
+
+class TestCLIAndCRD:
+    def test_crd_renders(self):
+        from fusioninfer_amd.controlplane.crd import inference_service_crd
+
+        crd = inference_service_crd()
+        assert crd["metadata"]["name"] == "inferenceservices.fusioninfer.io"
+        v = crd["spec"]["versions"][0]
+        role = v["schema"]["openAPIV3Schema"]["properties"]["spec"][
+            "properties"]["roles"]["items"]
+        assert set(role["properties"]["componentType"]["enum"]) == {
+            "router", "prefiller", "decoder", "worker"
+        }
+        assert "status" in v["subresources"]
+
+    def test_render_cli_on_samples(self, tmp_path, capsys):
+        import glob
+
+        from fusioninfer_amd.controlplane.__main__ import main
+
+        for sample in sorted(glob.glob("config/samples/*.yaml")):
+            assert main(["render", sample]) == 0
+            out = capsys.readouterr().out
+            docs = [d for d in yaml.safe_load_all(out) if d]
+            kinds = {d["kind"] for d in docs}
+            assert "LeaderWorkerSet" in kinds
+            assert "InferencePool" in kinds
+            assert "HTTPRoute" in kinds
+            # every worker pod asks for amd.com/gpu, never nvidia.com/gpu
+            assert "nvidia.com/gpu" not in out
+            if "pd" in sample or "tp8" in sample:
+                assert "PodGroup" in kinds
