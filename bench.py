@@ -133,7 +133,22 @@ def main():
         seed=1234 + rank,
         enforce_eager=args.enforce_eager,
     )
+    # TP: ranks form DP groups of size tp; only group leaders drive load.
+    driver_group = None
+    if distributed and args.tp > 1:
+        driver_ranks = list(range(0, world, args.tp))
+        driver_group = dist.new_group(driver_ranks)
+
     engine = LLMEngine(cfg, device=device)
+
+    if args.tp > 1 and not engine.is_driver:
+        engine.worker_loop()   # released by stop_workers() after timing
+        _aggregate_and_report(args, rank, world, distributed, device,
+                              use_cuda, 0.0, 0, 0, [], is_driver=False)
+        if distributed:
+            dist.destroy_process_group()
+        return
+
     load = ClosedLoopLoad(
         engine, args.prompt_len, args.gen_len, args.concurrency,
         mc.vocab_size, seed=99 + rank,
@@ -142,7 +157,7 @@ def main():
 
     def barrier_sync():
         if distributed:
-            dist.barrier()
+            dist.barrier(group=driver_group)
         if use_cuda:
             torch.cuda.synchronize()
 
@@ -173,56 +188,62 @@ def main():
     gen_tokens = engine.num_generated_tokens - tokens_before
     ttfts = load.ttfts[ttft_mark:] or load.ttfts
 
+    engine.stop_workers()
+    _aggregate_and_report(args, rank, world, distributed, device, use_cuda,
+                          elapsed, completed, gen_tokens, ttfts,
+                          is_driver=True)
     if distributed:
-        t = torch.tensor(
-            [elapsed], dtype=torch.float64,
-            device=device if use_cuda else "cpu",
-        )
+        dist.destroy_process_group()
+
+
+def _aggregate_and_report(args, rank, world, distributed, device, use_cuda,
+                          elapsed, completed, gen_tokens, ttfts, is_driver):
+    if distributed:
+        dev = device if use_cuda else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
-        c = torch.tensor(
-            [completed, gen_tokens], dtype=torch.float64,
-            device=device if use_cuda else "cpu",
-        )
+        c = torch.tensor([completed, gen_tokens], dtype=torch.float64,
+                         device=dev)
         dist.all_reduce(c)
         completed, gen_tokens = float(c[0].item()), float(c[1].item())
         all_ttfts = [None] * world
         dist.all_gather_object(all_ttfts, ttfts)
         ttfts = [x for lst in all_ttfts for x in lst]
 
-    goodput = completed / elapsed
+    if rank != 0:
+        return
+    goodput = completed / elapsed if elapsed > 0 else 0.0
     p50_ttft_ms = (
         statistics.median(ttfts) * 1000.0 if ttfts else float("nan")
     )
-    if rank == 0:
-        result = {
-            "metric": "goodput req/s (Qwen3-8B serving, closed-loop)",
-            "value": round(goodput, 3),
-            "unit": "req/s",
-            "n_gpus": world if distributed else args.gpus,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
-            "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": None,
-            "dtype": "bf16",
-            "data": "synthetic",
-            "p50_ttft_ms": round(p50_ttft_ms, 1),
-            "tokens_per_s": round(gen_tokens / elapsed, 1),
-            "config": {
-                "model": args.model,
-                "global_batch": args.concurrency * (world if distributed else 1),
-                "seq_len": args.prompt_len + args.gen_len,
-                "prompt_len": args.prompt_len,
-                "gen_len": args.gen_len,
-                "parallelism": f"dp{world if distributed else args.gpus}"
-                + (f"-tp{args.tp}" if args.tp > 1 else ""),
-            },
-        }
-        print(json.dumps(result))
-    if distributed:
-        dist.destroy_process_group()
+    dp = (world // args.tp) if distributed else args.gpus
+    result = {
+        "metric": "goodput req/s (Qwen3-8B serving, closed-loop)",
+        "value": round(goodput, 3),
+        "unit": "req/s",
+        "n_gpus": world if distributed else args.gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16",
+        "data": "synthetic",
+        "p50_ttft_ms": round(p50_ttft_ms, 1),
+        "tokens_per_s": round(gen_tokens / elapsed, 1) if elapsed else 0.0,
+        "config": {
+            "model": args.model,
+            "global_batch": args.concurrency * dp,
+            "seq_len": args.prompt_len + args.gen_len,
+            "prompt_len": args.prompt_len,
+            "gen_len": args.gen_len,
+            "parallelism": f"dp{dp}"
+            + (f"-tp{args.tp}" if args.tp > 1 else ""),
+        },
+    }
+    print(json.dumps(result))
 
 
 if __name__ == "__main__":

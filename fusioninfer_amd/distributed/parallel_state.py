@@ -86,6 +86,24 @@ def destroy() -> None:
     _TP_GROUP, _TP_RANK, _TP_WORLD, _INITIALIZED = None, 0, 1, False
 
 
+def tp_broadcast_object(obj=None):
+    """Broadcast a picklable object from the TP group's first rank."""
+    if _TP_WORLD == 1:
+        return obj
+    lst = [obj]
+    src = (dist.get_rank() // _TP_WORLD) * _TP_WORLD
+    dist.broadcast_object_list(lst, src=src, group=_TP_GROUP)
+    return lst[0]
+
+
+def tp_all_reduce_min_int(value: int) -> int:
+    if _TP_WORLD == 1 or not dist.is_initialized():
+        return value
+    t = torch.tensor([value], dtype=torch.int64)
+    dist.all_reduce(t, op=dist.ReduceOp.MIN, group=_TP_GROUP)
+    return int(t.item())
+
+
 def tp_all_reduce(t: torch.Tensor) -> torch.Tensor:
     if _TP_WORLD == 1:
         return t

@@ -36,7 +36,15 @@ class LLMEngine:
     def __init__(self, cfg: EngineConfig, device: Optional[str] = None):
         from fusioninfer_amd.distributed import parallel_state as ps
 
-        ps.ensure_single_process()
+        if cfg.parallel.tensor_parallel_size > 1:
+            ps.init_distributed(
+                cfg.parallel.tensor_parallel_size,
+                backend=cfg.parallel.distributed_backend,
+            )
+        else:
+            ps.ensure_single_process()
+        self._ps = ps
+        self.is_driver = ps.tp_rank() == 0
         self.cfg = cfg
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
@@ -148,18 +156,30 @@ class LLMEngine:
 
     # ------------------------------------------------------------ stepping
     def step(self) -> List[RequestOutput]:
-        """One engine iteration. Returns outputs for sequences that produced
-        a token this step (finished ones flagged)."""
+        """One engine iteration (TP driver / single rank). Returns outputs
+        for sequences that produced a token this step (finished flagged).
+
+        TP > 1: rank 0 schedules and broadcasts the batch payload; worker
+        ranks (in worker_loop) execute the same forward so the per-layer
+        RCCL all-reduces line up."""
+        assert self.is_driver, "only TP rank 0 steps; others run worker_loop"
+        tp = self.cfg.parallel.tensor_parallel_size
         batch = self.scheduler.schedule()
         if batch.is_empty:
             return []
         if batch.is_prefill:
             seqs = batch.prefill_seqs
-            logits = self.runner.execute_prefill(seqs, self.block_manager)
+            payload = self.runner.build_prefill_payload(seqs, self.block_manager)
+            if tp > 1:
+                self._ps.tp_broadcast_object(payload)
+            logits = self.runner.run_prefill(payload)
             self.num_prefilled_tokens += sum(s.num_prompt_tokens for s in seqs)
         else:
             seqs = batch.decode_seqs
-            logits = self.runner.execute_decode(seqs, self.block_manager)
+            payload = self.runner.build_decode_payload(seqs, self.block_manager)
+            if tp > 1:
+                self._ps.tp_broadcast_object(payload)
+            logits = self.runner.run_decode(payload)
         next_tokens = self.sampler.sample(logits.float(), seqs)
         next_tokens = next_tokens.tolist()
 
@@ -174,6 +194,23 @@ class LLMEngine:
                 del self.seqs[seq.seq_id]
             outputs.append(RequestOutput(seq))
         return outputs
+
+    # --------------------------------------------------------- TP workers
+    def worker_loop(self) -> None:
+        """TP rank > 0: execute broadcast batches until the driver stops."""
+        assert not self.is_driver
+        while True:
+            payload = self._ps.tp_broadcast_object(None)
+            if payload is None or payload.get("kind") == "stop":
+                return
+            if payload["kind"] == "prefill":
+                self.runner.run_prefill(payload)
+            else:
+                self.runner.run_decode(payload)
+
+    def stop_workers(self) -> None:
+        if self.is_driver and self.cfg.parallel.tensor_parallel_size > 1:
+            self._ps.tp_broadcast_object({"kind": "stop"})
 
     # ------------------------------------------------------------ sync API
     def generate(

@@ -51,8 +51,10 @@ class ModelRunner:
         return mc.num_layers * 2 * kv_heads * self.block_size * mc.head_dim * 2
 
     def profile_num_blocks(self) -> int:
+        import fusioninfer_amd.distributed.parallel_state as ps
+
         if self.cfg.cache.num_gpu_blocks is not None:
-            return self.cfg.cache.num_gpu_blocks
+            return ps.tp_all_reduce_min_int(self.cfg.cache.num_gpu_blocks)
         assert self.is_cuda
         free_b, total_b = torch.cuda.mem_get_info(self.device)
         usable = int(
@@ -62,7 +64,8 @@ class ModelRunner:
         # headroom for activations / graphs
         usable -= 4 << 30
         n = max(usable // self.kv_block_bytes(), 16)
-        return int(n)
+        # TP ranks must agree on the cache size
+        return ps.tp_all_reduce_min_int(int(n))
 
     def allocate_kv_caches(self) -> None:
         import fusioninfer_amd.distributed.parallel_state as ps
@@ -80,10 +83,11 @@ class ModelRunner:
         ]
 
     # ------------------------------------------------------------ prefill
-    def _prepare_prefill(self, seqs: List[Sequence], bm: BlockManager):
-        """Builds the batch over each sequence's NEW tokens only — prefix-
+    def build_prefill_payload(self, seqs: List[Sequence], bm: BlockManager):
+        """Host-side batch over each sequence's NEW tokens only — prefix-
         cache hits (seq.num_cached_tokens) are skipped; attention runs over
-        the paged cache (context attention)."""
+        the paged cache (context attention). The payload is plain lists so
+        the TP driver can broadcast it to worker ranks."""
         input_ids: List[int] = []
         positions: List[int] = []
         slots: List[int] = []
@@ -100,39 +104,57 @@ class ModelRunner:
             cu.append(cu[-1] + (L - C))
             new_lens.append(L - C)
             total_lens.append(L)
-        dev = self.device
-        tile_seq, tile_row0 = ops_mod.build_prefill_tiles(new_lens, device=dev)
         max_blocks = max(len(s.block_ids) for s in seqs)
-        bt = torch.zeros((len(seqs), max_blocks), dtype=torch.int32)
-        for i, s in enumerate(seqs):
-            bt[i, : len(s.block_ids)] = torch.tensor(
-                s.block_ids, dtype=torch.int32
-            )
+        bt = [
+            s.block_ids + [0] * (max_blocks - len(s.block_ids)) for s in seqs
+        ]
+        return {
+            "kind": "prefill",
+            "ids": input_ids,
+            "positions": positions,
+            "slots": slots,
+            "cu": cu,
+            "new_lens": new_lens,
+            "total_lens": total_lens,
+            "bt": bt,
+        }
+
+    def run_prefill(self, payload) -> torch.Tensor:
+        dev = self.device
+        cu = payload["cu"]
+        tile_seq, tile_row0 = ops_mod.build_prefill_tiles(
+            payload["new_lens"], device=dev
+        )
         meta = AttnMetadata(
             num_prefill_tokens=cu[-1],
             num_decode_tokens=0,
-            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
-            slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
+            positions=torch.tensor(
+                payload["positions"], dtype=torch.int32, device=dev
+            ),
+            slot_mapping=torch.tensor(
+                payload["slots"], dtype=torch.int32, device=dev
+            ),
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
             tile_seq=tile_seq,
             tile_row0=tile_row0,
-            prefill_block_tables=bt.to(dev),
+            prefill_block_tables=torch.tensor(
+                payload["bt"], dtype=torch.int32, device=dev
+            ),
             prefill_seq_lens_k=torch.tensor(
-                total_lens, dtype=torch.int32, device=dev
+                payload["total_lens"], dtype=torch.int32, device=dev
             ),
         )
-        ids = torch.tensor(input_ids, dtype=torch.long, device=dev)
+        ids = torch.tensor(payload["ids"], dtype=torch.long, device=dev)
         logits_idx = torch.tensor(
             [c - 1 for c in cu[1:]], dtype=torch.long, device=dev
         )
-        return ids, meta, logits_idx
-
-    def execute_prefill(self, seqs: List[Sequence], bm: BlockManager):
-        ids, meta, logits_idx = self._prepare_prefill(seqs, bm)
         with torch.no_grad():
             hidden = self.model(ids, meta, self.kv_caches)
             logits = self.model.compute_logits(hidden[logits_idx])
         return logits
+
+    def execute_prefill(self, seqs: List[Sequence], bm: BlockManager):
+        return self.run_prefill(self.build_prefill_payload(seqs, bm))
 
     # ------------------------------------------------------------ decode
     def _alloc_static(self, max_bs: int) -> None:
@@ -183,32 +205,37 @@ class ModelRunner:
                 self._graphs[bs] = (g, out)
         torch.cuda.synchronize()
 
-    def _fill_decode_inputs(self, seqs: List[Sequence], bm: BlockManager,
-                            bs: int) -> None:
-        n = len(seqs)
-        ids = [s.all_token_ids[-1] for s in seqs]
-        pos = [s.num_tokens - 1 for s in seqs]
-        slots = [bm.slot_for(s, s.num_tokens - 1) for s in seqs]
-        lens = [s.num_tokens for s in seqs]
+    def build_decode_payload(self, seqs: List[Sequence], bm: BlockManager):
+        return {
+            "kind": "decode",
+            "ids": [s.all_token_ids[-1] for s in seqs],
+            "positions": [s.num_tokens - 1 for s in seqs],
+            "slots": [bm.slot_for(s, s.num_tokens - 1) for s in seqs],
+            "lens": [s.num_tokens for s in seqs],
+            "bt": [list(s.block_ids) for s in seqs],
+        }
+
+    def _fill_decode_inputs(self, payload, bs: int) -> None:
+        n = len(payload["ids"])
         st = self._static
         dev = self.device
         st["ids"][:n].copy_(
-            torch.tensor(ids, dtype=torch.long), non_blocking=True
+            torch.tensor(payload["ids"], dtype=torch.long), non_blocking=True
         )
         st["positions"][:n].copy_(
-            torch.tensor(pos, dtype=torch.int32), non_blocking=True
+            torch.tensor(payload["positions"], dtype=torch.int32),
+            non_blocking=True,
         )
         st["slots"][:n].copy_(
-            torch.tensor(slots, dtype=torch.int32), non_blocking=True
+            torch.tensor(payload["slots"], dtype=torch.int32),
+            non_blocking=True,
         )
         st["seq_lens"][:n].copy_(
-            torch.tensor(lens, dtype=torch.int32), non_blocking=True
+            torch.tensor(payload["lens"], dtype=torch.int32), non_blocking=True
         )
         bt = torch.zeros((n, self.max_blocks_per_seq), dtype=torch.int32)
-        for i, s in enumerate(seqs):
-            bt[i, : len(s.block_ids)] = torch.tensor(
-                s.block_ids, dtype=torch.int32
-            )
+        for i, ids in enumerate(payload["bt"]):
+            bt[i, : len(ids)] = torch.tensor(ids, dtype=torch.int32)
         st["block_tables"][:n].copy_(bt.to(dev), non_blocking=True)
         if bs > n:  # padding rows: decode block 0, pos 0, len 1, slot -1
             st["ids"][n:bs].zero_()
@@ -217,8 +244,8 @@ class ModelRunner:
             st["seq_lens"][n:bs].fill_(1)
             st["block_tables"][n:bs].zero_()
 
-    def execute_decode(self, seqs: List[Sequence], bm: BlockManager):
-        n = len(seqs)
+    def run_decode(self, payload) -> torch.Tensor:
+        n = len(payload["ids"])
         if not self._static:
             self._alloc_static(max(self.cfg.scheduler.max_num_seqs, n))
         bucket = next((b for b in _DECODE_BUCKETS if b >= n), None)
@@ -226,10 +253,13 @@ class ModelRunner:
             bucket is not None and bucket in self._graphs and self.is_cuda
         )
         bs = bucket if use_graph else n
-        self._fill_decode_inputs(seqs, bm, bs)
+        self._fill_decode_inputs(payload, bs)
         if use_graph:
             g, out = self._graphs[bucket]
             g.replay()
             return out[:n]
         with torch.no_grad():
             return self._decode_forward(n)[:n]
+
+    def execute_decode(self, seqs: List[Sequence], bm: BlockManager):
+        return self.run_decode(self.build_decode_payload(seqs, bm))

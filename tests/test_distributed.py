@@ -153,3 +153,49 @@ def test_tp2_matches_tp1_logits():
             p.join(timeout=240)
             assert p.exitcode == 0, f"rank exited {p.exitcode}"
         assert results[0] == "ok" and results[1] == "ok"
+
+
+# ------------------------------------------------------ TP engine (driver/worker)
+def _tp_engine_worker(rank, port, results):
+    from fusioninfer_amd.distributed import parallel_state as ps
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+
+    _init(rank, 2, port)
+    try:
+        cfg = _engine_cfg()
+        cfg.parallel.tensor_parallel_size = 2
+        cfg.parallel.distributed_backend = "gloo"
+        engine = LLMEngine(cfg, device="cpu")
+        if engine.is_driver:
+            outs = engine.generate(
+                [PROMPT, [9, 9, 2] * 8],
+                SamplingParams(max_tokens=4, temperature=0.0),
+            )
+            assert all(len(o.output_token_ids) == 4 for o in outs)
+            engine.stop_workers()
+        else:
+            engine.worker_loop()
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+        ps.destroy()
+
+
+def test_tp2_engine_driver_worker():
+    """Full engine under TP=2: driver schedules + broadcasts, worker executes
+    the same forwards (per-layer all-reduces line up); generation completes."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29617
+        procs = [
+            ctx.Process(target=_tp_engine_worker, args=(r, port, results))
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=240)
+            assert p.exitcode == 0, f"rank exited {p.exitcode}"
+        assert results[0] == "ok" and results[1] == "ok"
