@@ -139,6 +139,7 @@ class LLMEngine:
         seq.block_ids = holder.block_ids
         seq.append_token(first_token)
         seq.status = SeqStatus.RUNNING
+        seq.num_computed_tokens = prompt_len  # KV imported, nothing to prefill
         self.seqs[request_id] = seq
         self.scheduler.running.append(seq)
         return request_id
@@ -169,22 +170,35 @@ class LLMEngine:
             return []
         if batch.is_prefill:
             seqs = batch.prefill_seqs
-            payload = self.runner.build_prefill_payload(seqs, self.block_manager)
+            payload = self.runner.build_prefill_payload(
+                seqs, self.block_manager, batch.prefill_chunks
+            )
             if tp > 1:
                 self._ps.tp_broadcast_object(payload)
             logits = self.runner.run_prefill(payload)
-            self.num_prefilled_tokens += sum(s.num_prompt_tokens for s in seqs)
+            for seq, chunk in zip(seqs, batch.prefill_chunks):
+                seq.num_computed_tokens = (
+                    seq.num_computed_tokens or seq.num_cached_tokens
+                ) + chunk
+                self.num_prefilled_tokens += chunk
+            # only sequences whose prompt completed this step sample a token
+            sample_seqs = [
+                s for s, smp in zip(seqs, payload["sample"]) if smp
+            ]
         else:
             seqs = batch.decode_seqs
             payload = self.runner.build_decode_payload(seqs, self.block_manager)
             if tp > 1:
                 self._ps.tp_broadcast_object(payload)
             logits = self.runner.run_decode(payload)
-        next_tokens = self.sampler.sample(logits.float(), seqs)
+            sample_seqs = seqs
+        if not sample_seqs:
+            return []
+        next_tokens = self.sampler.sample(logits.float(), sample_seqs)
         next_tokens = next_tokens.tolist()
 
         outputs: List[RequestOutput] = []
-        for seq, tok in zip(seqs, next_tokens):
+        for seq, tok in zip(sample_seqs, next_tokens):
             seq.append_token(int(tok))
             self.num_generated_tokens += 1
             if seq.check_stop():

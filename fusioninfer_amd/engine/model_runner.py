@@ -83,27 +83,40 @@ class ModelRunner:
         ]
 
     # ------------------------------------------------------------ prefill
-    def build_prefill_payload(self, seqs: List[Sequence], bm: BlockManager):
-        """Host-side batch over each sequence's NEW tokens only — prefix-
-        cache hits (seq.num_cached_tokens) are skipped; attention runs over
-        the paged cache (context attention). The payload is plain lists so
-        the TP driver can broadcast it to worker ranks."""
+    def build_prefill_payload(
+        self,
+        seqs: List[Sequence],
+        bm: BlockManager,
+        chunks: Optional[List[int]] = None,
+    ):
+        """Host-side batch over each sequence's scheduled prompt chunk
+        [num_computed_tokens, +chunk) — prefix-cache hits and previously
+        processed chunks are skipped; attention runs over the paged cache
+        (context attention). The payload is plain lists so the TP driver
+        can broadcast it to worker ranks. `sample` marks sequences whose
+        chunk completes the prompt (only those produce a token)."""
+        if chunks is None:
+            chunks = [
+                s.num_prompt_tokens - s.num_computed_tokens for s in seqs
+            ]
         input_ids: List[int] = []
         positions: List[int] = []
         slots: List[int] = []
         cu = [0]
         new_lens = []
         total_lens = []
-        for seq in seqs:
+        sample = []
+        for seq, chunk in zip(seqs, chunks):
             toks = seq.all_token_ids
-            L = len(toks)
-            C = seq.num_cached_tokens
-            input_ids.extend(toks[C:])
-            positions.extend(range(C, L))
-            slots.extend(bm.slot_for(seq, p) for p in range(C, L))
-            cu.append(cu[-1] + (L - C))
-            new_lens.append(L - C)
-            total_lens.append(L)
+            C = seq.num_computed_tokens or seq.num_cached_tokens
+            end = C + chunk
+            input_ids.extend(toks[C:end])
+            positions.extend(range(C, end))
+            slots.extend(bm.slot_for(seq, p) for p in range(C, end))
+            cu.append(cu[-1] + chunk)
+            new_lens.append(chunk)
+            total_lens.append(end)
+            sample.append(end >= seq.num_prompt_tokens)
         max_blocks = max(len(s.block_ids) for s in seqs)
         bt = [
             s.block_ids + [0] * (max_blocks - len(s.block_ids)) for s in seqs
@@ -117,6 +130,7 @@ class ModelRunner:
             "new_lens": new_lens,
             "total_lens": total_lens,
             "bt": bt,
+            "sample": sample,
         }
 
     def run_prefill(self, payload) -> torch.Tensor:
@@ -145,11 +159,16 @@ class ModelRunner:
             ),
         )
         ids = torch.tensor(payload["ids"], dtype=torch.long, device=dev)
+        sample = payload.get("sample") or [True] * (len(cu) - 1)
         logits_idx = torch.tensor(
-            [c - 1 for c in cu[1:]], dtype=torch.long, device=dev
+            [c - 1 for c, smp in zip(cu[1:], sample) if smp],
+            dtype=torch.long,
+            device=dev,
         )
         with torch.no_grad():
             hidden = self.model(ids, meta, self.kv_caches)
+            if logits_idx.numel() == 0:
+                return hidden.new_empty((0, self.cfg.model.vocab_size))
             logits = self.model.compute_logits(hidden[logits_idx])
         return logits
 

@@ -27,6 +27,7 @@ from fusioninfer_amd.engine.sequence import Sequence, SeqStatus
 @dataclasses.dataclass
 class ScheduledBatch:
     prefill_seqs: List[Sequence]
+    prefill_chunks: List[int]          # tokens scheduled per prefill seq
     decode_seqs: List[Sequence]
     preempted: List[Sequence]
 
@@ -64,33 +65,49 @@ class Scheduler:
     # ----------------------------------------------------------- schedule
     def schedule(self) -> ScheduledBatch:
         preempted: List[Sequence] = []
-
-        # try to build a prefill batch
-        prefill: List[Sequence] = []
         budget = self.cfg.max_num_batched_tokens
+
+        # prefill batch: first continue partially-prefilled running seqs
+        # (chunked prefill), then admit waiting prompts up to the budget
+        prefill: List[Sequence] = []
+        chunks: List[int] = []
+        for seq in self.running:
+            if seq.status != SeqStatus.RUNNING:
+                continue
+            remaining = seq.num_prompt_tokens - seq.num_computed_tokens
+            if remaining <= 0 or budget <= 0:
+                continue
+            chunk = min(remaining, budget)
+            prefill.append(seq)
+            chunks.append(chunk)
+            budget -= chunk
         while (
-            self.waiting
-            and len(self.running) + len(prefill) < self.cfg.max_num_seqs
+            budget > 0
+            and self.waiting
+            and len(self.running) < self.cfg.max_num_seqs
         ):
             seq = self.waiting[0]
-            if seq.num_prompt_tokens > budget:
-                break
             if not self.bm.can_allocate(seq.num_prompt_tokens):
                 break
             self.waiting.popleft()
             self.bm.allocate(seq)
             seq.status = SeqStatus.RUNNING
+            seq.num_computed_tokens = seq.num_cached_tokens
+            chunk = min(seq.num_prompt_tokens - seq.num_computed_tokens, budget)
             prefill.append(seq)
-            budget -= seq.num_prompt_tokens
+            chunks.append(chunk)
+            budget -= chunk
+            self.running.append(seq)
         if prefill:
-            self.running.extend(prefill)
-            return ScheduledBatch(prefill, [], preempted)
+            return ScheduledBatch(prefill, chunks, [], preempted)
 
-        # decode step: every running sequence generates one token
+        # decode step: every fully-prefilled running sequence, one token each
         decode: List[Sequence] = []
         for seq in list(self.running):
             if seq.status != SeqStatus.RUNNING:
                 continue  # preempted earlier in this same pass
+            if seq.num_computed_tokens < seq.num_prompt_tokens:
+                continue  # still prefilling
             if not self.bm.can_append_slot(seq):
                 victim = self._preempt_newest()
                 preempted.append(victim)
@@ -98,7 +115,7 @@ class Scheduler:
                     continue
             self.bm.append_slot(seq)
             decode.append(seq)
-        return ScheduledBatch([], decode, preempted)
+        return ScheduledBatch([], [], decode, preempted)
 
     def _preempt_newest(self) -> Sequence:
         victim = self.running.pop()  # newest
@@ -107,6 +124,8 @@ class Scheduler:
         # recompute: generated tokens become part of the prompt
         victim.prompt_token_ids.extend(victim.output_token_ids)
         victim.output_token_ids = []
+        victim.num_computed_tokens = 0
+        victim.num_cached_tokens = 0
         self.waiting.appendleft(victim)
         return victim
 

@@ -46,6 +46,9 @@ class Sequence:
         self.finish_time: Optional[float] = None
         # number of prompt tokens whose KV was satisfied by prefix cache
         self.num_cached_tokens = 0
+        # prompt tokens processed so far (chunked prefill); set to
+        # num_cached_tokens at admission, advances per prefill chunk
+        self.num_computed_tokens = 0
         # PD producer: keep cache blocks alive after finish for KV export
         self.hold_blocks = False
 

@@ -148,3 +148,43 @@ def test_prefix_caching_shared_prefix_divergent_tail():
     e2 = eng2.generate([p2], SamplingParams(max_tokens=4))[0]
     assert r1.output_token_ids == e1.output_token_ids
     assert r2.output_token_ids == e2.output_token_ids
+
+
+def test_chunked_prefill_matches_unchunked():
+    """A prompt longer than max_num_batched_tokens is prefilled in chunks
+    (context attention over the cache) and yields the same greedy tokens."""
+    torch.manual_seed(0)
+    # budget 48 tokens per step, prompt 100 -> 3 chunks
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=256),
+        scheduler=SchedulerConfig(
+            max_num_seqs=4, max_num_batched_tokens=48, max_model_len=256
+        ),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    prompt = [3, 7, 2, 9, 4] * 20  # 100 tokens
+    expected = oracle_greedy(eng, prompt, 5)
+    out = eng.generate([prompt], SamplingParams(max_tokens=5))[0]
+    assert out.output_token_ids == expected
+
+
+def test_chunked_prefill_interleaves_decode():
+    """While one prompt is mid-chunk, already-running sequences keep
+    decoding between chunks (prefill-priority per step, chunk-bounded)."""
+    torch.manual_seed(0)
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=256),
+        scheduler=SchedulerConfig(
+            max_num_seqs=4, max_num_batched_tokens=32, max_model_len=256
+        ),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    a = eng.add_request([1, 2] * 8, SamplingParams(max_tokens=20))
+    for _ in range(3):
+        eng.step()
+    b = eng.add_request([4, 5, 6] * 30, SamplingParams(max_tokens=3))  # 90 toks
+    while eng.has_unfinished():
+        eng.step()
+    assert eng.num_finished == 2
