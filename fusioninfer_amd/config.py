@@ -30,6 +30,8 @@ class ModelConfig:
     qk_norm: bool = True              # Qwen3 per-head q/k RMSNorm
     tie_word_embeddings: bool = False
     dtype: str = "bfloat16"
+    # optional HF-layout safetensors checkpoint dir; None = random init
+    model_path: Optional[str] = None
 
     @property
     def q_size(self) -> int:

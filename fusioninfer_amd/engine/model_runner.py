@@ -32,6 +32,10 @@ class ModelRunner:
         torch.manual_seed(cfg.seed)
         with torch.device(self.device):
             self.model = CausalLM(mc).eval()
+        if mc.model_path:
+            from fusioninfer_amd.models.weight_loader import load_safetensors_dir
+
+            load_safetensors_dir(self.model, mc.model_path)
         self.block_size = cfg.cache.block_size
         self.max_blocks_per_seq = (
             cfg.scheduler.max_model_len + self.block_size - 1

@@ -1,0 +1,168 @@
+"""Checkpoint loading: HF-layout safetensors -> the FusionInfer-AMD model.
+
+The reference treats model weight materialization as design-doc-only
+(node caches / warm-up jobs); the engine itself must still
+be able to load real weights. This maps HuggingFace parameter names
+(Qwen3 / Llama layout) onto the fused, TP-sharded modules:
+
+  q_proj+k_proj+v_proj -> qkv_proj     (per-segment rank shard)
+  gate_proj+up_proj    -> gate_up_proj (per-segment rank shard)
+  o_proj / down_proj   -> row-parallel shard (input-dim slice)
+"""
+
+from __future__ import annotations
+
+import json
+import os
+from typing import Dict, Iterable, Tuple
+
+import torch
+
+from fusioninfer_amd.distributed import parallel_state as ps
+
+
+def _shard_rows(w: torch.Tensor, tp: int, rank: int) -> torch.Tensor:
+    per = w.shape[0] // tp
+    return w[rank * per : (rank + 1) * per]
+
+
+def _shard_cols(w: torch.Tensor, tp: int, rank: int) -> torch.Tensor:
+    per = w.shape[1] // tp
+    return w[:, rank * per : (rank + 1) * per]
+
+
+def load_hf_state_dict(
+    model, tensors: Iterable[Tuple[str, torch.Tensor]]
+) -> int:
+    """Load HF-named tensors into a CausalLM. Returns #tensors consumed."""
+    cfg = model.cfg
+    tp = ps.tp_world_size()
+    rank = ps.tp_rank()
+    q_size = cfg.num_heads * cfg.head_dim
+    kv_size = cfg.num_kv_heads * cfg.head_dim
+
+    # staging for fused weights: (layer, which) -> tensor
+    pending: Dict[Tuple[int, str], torch.Tensor] = {}
+    loaded = 0
+
+    def put(param: torch.nn.Parameter, value: torch.Tensor):
+        assert param.data.shape == value.shape, (param.data.shape, value.shape)
+        param.data.copy_(value.to(param.dtype))
+
+    def try_fuse_qkv(li: int):
+        keys = [(li, "q"), (li, "k"), (li, "v")]
+        if all(k in pending for k in keys):
+            qw = _shard_rows(pending.pop(keys[0]), tp, rank)
+            kw = _shard_rows(pending.pop(keys[1]), tp, rank)
+            vw = _shard_rows(pending.pop(keys[2]), tp, rank)
+            put(model.layers[li].self_attn.qkv_proj.weight,
+                torch.cat([qw, kw, vw], dim=0))
+
+    def try_fuse_gate_up(li: int):
+        keys = [(li, "gate"), (li, "up")]
+        if all(k in pending for k in keys):
+            gw = _shard_rows(pending.pop(keys[0]), tp, rank)
+            uw = _shard_rows(pending.pop(keys[1]), tp, rank)
+            put(model.layers[li].mlp.gate_up_proj.weight,
+                torch.cat([gw, uw], dim=0))
+
+    for name, w in tensors:
+        loaded += 1
+        if name == "model.embed_tokens.weight":
+            put(model.embed_tokens, w)
+        elif name == "model.norm.weight":
+            put(model.final_norm_weight, w)
+        elif name == "lm_head.weight":
+            if model.lm_head is not None:
+                put(model.lm_head.weight, w)
+        elif name.startswith("model.layers."):
+            parts = name.split(".")
+            li = int(parts[2])
+            rest = ".".join(parts[3:])
+            layer = model.layers[li]
+            if rest == "self_attn.q_proj.weight":
+                pending[(li, "q")] = w
+                try_fuse_qkv(li)
+            elif rest == "self_attn.k_proj.weight":
+                pending[(li, "k")] = w
+                try_fuse_qkv(li)
+            elif rest == "self_attn.v_proj.weight":
+                pending[(li, "v")] = w
+                try_fuse_qkv(li)
+            elif rest == "self_attn.o_proj.weight":
+                put(layer.self_attn.o_proj.weight, _shard_cols(w, tp, rank))
+            elif rest == "self_attn.q_norm.weight":
+                put(layer.self_attn.q_norm_weight, w)
+            elif rest == "self_attn.k_norm.weight":
+                put(layer.self_attn.k_norm_weight, w)
+            elif rest == "mlp.gate_proj.weight":
+                pending[(li, "gate")] = w
+                try_fuse_gate_up(li)
+            elif rest == "mlp.up_proj.weight":
+                pending[(li, "up")] = w
+                try_fuse_gate_up(li)
+            elif rest == "mlp.down_proj.weight":
+                put(layer.mlp.down_proj.weight, _shard_cols(w, tp, rank))
+            elif rest == "input_layernorm.weight":
+                put(layer.input_norm_weight, w)
+            elif rest == "post_attention_layernorm.weight":
+                put(layer.post_norm_weight, w)
+            else:
+                loaded -= 1  # unknown (bias-less models shouldn't hit this)
+        else:
+            loaded -= 1
+    assert not pending, f"unfused partial weights remain: {list(pending)}"
+    return loaded
+
+
+def load_safetensors_dir(model, path: str) -> int:
+    """Load all *.safetensors under `path` (HF checkpoint directory)."""
+    from safetensors.torch import safe_open
+
+    files = sorted(
+        os.path.join(path, f)
+        for f in os.listdir(path)
+        if f.endswith(".safetensors")
+    )
+    assert files, f"no safetensors files under {path}"
+
+    def tensor_iter():
+        for f in files:
+            with safe_open(f, framework="pt") as sf:
+                for name in sf.keys():
+                    yield name, sf.get_tensor(name)
+
+    return load_hf_state_dict(model, tensor_iter())
+
+
+def export_hf_state_dict(model) -> Dict[str, torch.Tensor]:
+    """Inverse mapping (TP=1 only) — used by tests and for producing
+    checkpoints from random-init models."""
+    assert ps.tp_world_size() == 1
+    cfg = model.cfg
+    q_size = cfg.num_heads * cfg.head_dim
+    kv_size = cfg.num_kv_heads * cfg.head_dim
+    out: Dict[str, torch.Tensor] = {
+        "model.embed_tokens.weight": model.embed_tokens.data.clone(),
+        "model.norm.weight": model.final_norm_weight.data.clone(),
+    }
+    if model.lm_head is not None:
+        out["lm_head.weight"] = model.lm_head.weight.data.clone()
+    for li, layer in enumerate(model.layers):
+        p = f"model.layers.{li}."
+        qkv = layer.self_attn.qkv_proj.weight.data
+        out[p + "self_attn.q_proj.weight"] = qkv[:q_size].clone()
+        out[p + "self_attn.k_proj.weight"] = qkv[q_size : q_size + kv_size].clone()
+        out[p + "self_attn.v_proj.weight"] = qkv[q_size + kv_size :].clone()
+        out[p + "self_attn.o_proj.weight"] = layer.self_attn.o_proj.weight.data.clone()
+        if layer.self_attn.q_norm_weight is not None:
+            out[p + "self_attn.q_norm.weight"] = layer.self_attn.q_norm_weight.data.clone()
+            out[p + "self_attn.k_norm.weight"] = layer.self_attn.k_norm_weight.data.clone()
+        gu = layer.mlp.gate_up_proj.weight.data
+        inter = gu.shape[0] // 2
+        out[p + "mlp.gate_proj.weight"] = gu[:inter].clone()
+        out[p + "mlp.up_proj.weight"] = gu[inter:].clone()
+        out[p + "mlp.down_proj.weight"] = layer.mlp.down_proj.weight.data.clone()
+        out[p + "input_layernorm.weight"] = layer.input_norm_weight.data.clone()
+        out[p + "post_attention_layernorm.weight"] = layer.post_norm_weight.data.clone()
+    return out

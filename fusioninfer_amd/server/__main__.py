@@ -18,6 +18,8 @@ import os
 def parse_args(argv=None):
     p = argparse.ArgumentParser("fusioninfer-amd engine server")
     p.add_argument("--model", default="Qwen3-8B")
+    p.add_argument("--model-path", default=None,
+                   help="HF-layout safetensors dir; default random init")
     p.add_argument("--host", default="0.0.0.0")
     p.add_argument("--port", type=int, default=8000)
     p.add_argument("--tensor-parallel-size", type=int, default=1)
@@ -60,8 +62,10 @@ def build_engine_config(args):
             kv_rank=int(raw.get("kv_rank", 0)),
             kv_world_size=int(raw.get("kv_world_size", 2)),
         )
+    mc = get_model_config(args.model)
+    mc.model_path = args.model_path
     return EngineConfig(
-        model=get_model_config(args.model),
+        model=mc,
         cache=CacheConfig(
             gpu_memory_utilization=args.gpu_memory_utilization,
             enable_prefix_caching=args.enable_prefix_caching,

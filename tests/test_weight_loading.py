@@ -1,0 +1,79 @@
+"""Checkpoint loading tests: HF-layout safetensors roundtrip."""
+
+import torch
+
+from fusioninfer_amd.distributed import parallel_state as ps
+from fusioninfer_amd.models.model import CausalLM
+from fusioninfer_amd.models.registry import get_model_config
+from fusioninfer_amd.models.weight_loader import (
+    export_hf_state_dict,
+    load_hf_state_dict,
+    load_safetensors_dir,
+)
+
+
+def _model(seed):
+    ps.ensure_single_process()
+    torch.manual_seed(seed)
+    return CausalLM(get_model_config("tiny-qwen3")).eval()
+
+
+def _forward_logits(model, tokens):
+    from fusioninfer_amd.engine.block_manager import BlockManager
+    from fusioninfer_amd.engine.metadata import AttnMetadata
+    import fusioninfer_amd.ops as ops
+
+    T = len(tokens)
+    nblk = (T + 15) // 16
+    kv = [
+        (torch.zeros(nblk + 1, 2, 16, 64, dtype=torch.bfloat16),
+         torch.zeros(nblk + 1, 2, 16, 64, dtype=torch.bfloat16))
+        for _ in range(model.cfg.num_layers)
+    ]
+    meta = AttnMetadata(
+        num_prefill_tokens=T,
+        num_decode_tokens=0,
+        positions=torch.arange(T, dtype=torch.int32),
+        slot_mapping=torch.arange(T, dtype=torch.int32),
+        cu_seqlens=torch.tensor([0, T], dtype=torch.int32),
+        prefill_block_tables=torch.arange(nblk, dtype=torch.int32).unsqueeze(0),
+        prefill_seq_lens_k=torch.tensor([T], dtype=torch.int32),
+    )
+    with torch.no_grad():
+        hidden = model(torch.tensor(tokens), meta, kv)
+        return model.compute_logits(hidden[-1:])
+
+
+def test_hf_state_dict_roundtrip():
+    src = _model(1)
+    dst = _model(2)
+    tokens = [3, 1, 4, 1, 5] * 6
+    before = _forward_logits(dst, tokens)
+    sd = export_hf_state_dict(src)
+    assert "model.layers.0.self_attn.q_norm.weight" in sd  # qwen3 qk-norm
+    n = load_hf_state_dict(dst, sd.items())
+    assert n == len(sd)
+    after = _forward_logits(dst, tokens)
+    expected = _forward_logits(src, tokens)
+    assert not torch.allclose(before.float(), expected.float())
+    torch.testing.assert_close(after.float(), expected.float())
+
+
+def test_safetensors_dir_loading(tmp_path):
+    from safetensors.torch import save_file
+
+    src = _model(3)
+    sd = export_hf_state_dict(src)
+    # split across two files like real HF checkpoints
+    keys = sorted(sd)
+    half = len(keys) // 2
+    save_file({k: sd[k] for k in keys[:half]}, str(tmp_path / "model-1.safetensors"))
+    save_file({k: sd[k] for k in keys[half:]}, str(tmp_path / "model-2.safetensors"))
+    dst = _model(4)
+    n = load_safetensors_dir(dst, str(tmp_path))
+    assert n == len(sd)
+    tokens = [9, 8, 7] * 5
+    torch.testing.assert_close(
+        _forward_logits(dst, tokens).float(),
+        _forward_logits(src, tokens).float(),
+    )
