@@ -168,30 +168,22 @@ class LLMEngine:
         batch = self.scheduler.schedule()
         if batch.is_empty:
             return []
-        if batch.is_prefill:
-            seqs = batch.prefill_seqs
-            payload = self.runner.build_prefill_payload(
-                seqs, self.block_manager, batch.prefill_chunks
-            )
-            if tp > 1:
-                self._ps.tp_broadcast_object(payload)
-            logits = self.runner.run_prefill(payload)
-            for seq, chunk in zip(seqs, batch.prefill_chunks):
-                seq.num_computed_tokens = (
-                    seq.num_computed_tokens or seq.num_cached_tokens
-                ) + chunk
-                self.num_prefilled_tokens += chunk
-            # only sequences whose prompt completed this step sample a token
-            sample_seqs = [
-                s for s, smp in zip(seqs, payload["sample"]) if smp
-            ]
-        else:
-            seqs = batch.decode_seqs
-            payload = self.runner.build_decode_payload(seqs, self.block_manager)
-            if tp > 1:
-                self._ps.tp_broadcast_object(payload)
-            logits = self.runner.run_decode(payload)
-            sample_seqs = seqs
+        payload = self.runner.build_batch_payload(
+            batch.prefill_seqs, batch.prefill_chunks, batch.decode_seqs,
+            self.block_manager,
+        )
+        if tp > 1:
+            self._ps.tp_broadcast_object(payload)
+        logits = self.runner.run_batch(payload)
+        for seq, chunk in zip(batch.prefill_seqs, batch.prefill_chunks):
+            seq.num_computed_tokens = (
+                seq.num_computed_tokens or seq.num_cached_tokens
+            ) + chunk
+            self.num_prefilled_tokens += chunk
+        # logits rows: completing prefill chunks first, then decode rows
+        sample_seqs = [
+            s for s, smp in zip(batch.prefill_seqs, payload["sample"]) if smp
+        ] + batch.decode_seqs
         if not sample_seqs:
             return []
         next_tokens = self.sampler.sample(logits.float(), sample_seqs)
@@ -217,7 +209,9 @@ class LLMEngine:
             payload = self._ps.tp_broadcast_object(None)
             if payload is None or payload.get("kind") == "stop":
                 return
-            if payload["kind"] == "prefill":
+            if payload["kind"] == "mixed":
+                self.runner.run_batch(payload)
+            elif payload["kind"] == "prefill":
                 self.runner.run_prefill(payload)
             else:
                 self.runner.run_decode(payload)

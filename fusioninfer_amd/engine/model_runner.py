@@ -179,6 +179,86 @@ class ModelRunner:
     def execute_prefill(self, seqs: List[Sequence], bm: BlockManager):
         return self.run_prefill(self.build_prefill_payload(seqs, bm))
 
+    # ------------------------------------------------------------- mixed
+    def build_batch_payload(
+        self,
+        prefill_seqs: List[Sequence],
+        chunks: List[int],
+        decode_seqs: List[Sequence],
+        bm: BlockManager,
+    ):
+        """One payload for a mixed step: [prefill chunk tokens | decode
+        tokens]. Either segment may be empty."""
+        payload = (
+            self.build_prefill_payload(prefill_seqs, bm, chunks)
+            if prefill_seqs
+            else {"kind": "prefill", "ids": [], "positions": [], "slots": [],
+                  "cu": [0], "new_lens": [], "total_lens": [], "bt": [],
+                  "sample": []}
+        )
+        payload["kind"] = "mixed"
+        d = (
+            self.build_decode_payload(decode_seqs, bm)
+            if decode_seqs
+            else {"ids": [], "positions": [], "slots": [], "lens": [], "bt": []}
+        )
+        payload["decode"] = d
+        return payload
+
+    def run_batch(self, payload) -> torch.Tensor:
+        """Mixed forward. Pure-decode payloads take the hipGraph path."""
+        d = payload["decode"]
+        if not payload["ids"]:
+            return self.run_decode(d)
+        if not d["ids"]:
+            return self.run_prefill(payload)
+        dev = self.device
+        cu = payload["cu"]
+        np_ = cu[-1]
+        nd = len(d["ids"])
+        tile_seq, tile_row0 = ops_mod.build_prefill_tiles(
+            payload["new_lens"], device=dev
+        )
+        max_blocks = max(len(b) for b in d["bt"])
+        dbt = torch.zeros((nd, max_blocks), dtype=torch.int32)
+        for i, ids in enumerate(d["bt"]):
+            dbt[i, : len(ids)] = torch.tensor(ids, dtype=torch.int32)
+        meta = AttnMetadata(
+            num_prefill_tokens=np_,
+            num_decode_tokens=nd,
+            positions=torch.tensor(
+                payload["positions"] + d["positions"],
+                dtype=torch.int32, device=dev,
+            ),
+            slot_mapping=torch.tensor(
+                payload["slots"] + d["slots"], dtype=torch.int32, device=dev
+            ),
+            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+            tile_seq=tile_seq,
+            tile_row0=tile_row0,
+            prefill_block_tables=torch.tensor(
+                payload["bt"], dtype=torch.int32, device=dev
+            ),
+            prefill_seq_lens_k=torch.tensor(
+                payload["total_lens"], dtype=torch.int32, device=dev
+            ),
+            block_tables=dbt.to(dev),
+            seq_lens=torch.tensor(d["lens"], dtype=torch.int32, device=dev),
+        )
+        ids = torch.tensor(
+            payload["ids"] + d["ids"], dtype=torch.long, device=dev
+        )
+        # logits rows: completing prefill chunks first, then every decode row
+        pf_idx = [
+            c - 1 for c, smp in zip(cu[1:], payload["sample"]) if smp
+        ]
+        logits_idx = torch.tensor(
+            pf_idx + list(range(np_, np_ + nd)), dtype=torch.long, device=dev
+        )
+        with torch.no_grad():
+            hidden = self.model(ids, meta, self.kv_caches)
+            return self.model.compute_logits(hidden[logits_idx])
+
     # ------------------------------------------------------------ decode
     def _alloc_static(self, max_bs: int) -> None:
         dev = self.device

@@ -64,11 +64,31 @@ class Scheduler:
 
     # ----------------------------------------------------------- schedule
     def schedule(self) -> ScheduledBatch:
+        """One MIXED batch per step (vLLM-v1-style unified scheduling):
+        decode tokens for every fully-prefilled running sequence PLUS
+        prompt chunks (continuing chunked prefills, then new admissions)
+        up to the shared max_num_batched_tokens budget. Decode-ready
+        sequences never stall behind prefills."""
         preempted: List[Sequence] = []
         budget = self.cfg.max_num_batched_tokens
 
-        # prefill batch: first continue partially-prefilled running seqs
-        # (chunked prefill), then admit waiting prompts up to the budget
+        # decode: every fully-prefilled running sequence, one token each
+        decode: List[Sequence] = []
+        for seq in list(self.running):
+            if seq.status != SeqStatus.RUNNING:
+                continue  # preempted earlier in this same pass
+            if seq.num_computed_tokens < seq.num_prompt_tokens:
+                continue  # still prefilling
+            if not self.bm.can_append_slot(seq):
+                victim = self._preempt_newest()
+                preempted.append(victim)
+                if victim is seq:
+                    continue
+            self.bm.append_slot(seq)
+            decode.append(seq)
+            budget -= 1
+
+        # prefill chunks: continue partially-prefilled running seqs first
         prefill: List[Sequence] = []
         chunks: List[int] = []
         for seq in self.running:
@@ -81,6 +101,7 @@ class Scheduler:
             prefill.append(seq)
             chunks.append(chunk)
             budget -= chunk
+        # then admit waiting prompts
         while (
             budget > 0
             and self.waiting
@@ -98,24 +119,7 @@ class Scheduler:
             chunks.append(chunk)
             budget -= chunk
             self.running.append(seq)
-        if prefill:
-            return ScheduledBatch(prefill, chunks, [], preempted)
-
-        # decode step: every fully-prefilled running sequence, one token each
-        decode: List[Sequence] = []
-        for seq in list(self.running):
-            if seq.status != SeqStatus.RUNNING:
-                continue  # preempted earlier in this same pass
-            if seq.num_computed_tokens < seq.num_prompt_tokens:
-                continue  # still prefilling
-            if not self.bm.can_append_slot(seq):
-                victim = self._preempt_newest()
-                preempted.append(victim)
-                if victim is seq:
-                    continue
-            self.bm.append_slot(seq)
-            decode.append(seq)
-        return ScheduledBatch([], [], decode, preempted)
+        return ScheduledBatch(prefill, chunks, decode, preempted)
 
     def _preempt_newest(self) -> Sequence:
         victim = self.running.pop()  # newest
