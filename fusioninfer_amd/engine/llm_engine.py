@@ -86,6 +86,24 @@ class LLMEngine:
     def has_unfinished(self) -> bool:
         return self.scheduler.has_work()
 
+    def abort_request(self, request_id: str) -> bool:
+        """Cancel a request wherever it is (waiting / running); frees its
+        cache blocks. Returns True if it was found."""
+        from fusioninfer_amd.engine.sequence import SeqStatus
+
+        seq = self.seqs.pop(request_id, None)
+        if seq is None:
+            return False
+        if seq in self.scheduler.waiting:
+            self.scheduler.waiting.remove(seq)
+        if seq in self.scheduler.running:
+            self.scheduler.running.remove(seq)
+            self.block_manager.free(seq)
+        elif seq.block_ids:
+            self.block_manager.free(seq)
+        seq.status = SeqStatus.FINISHED
+        return True
+
     # ------------------------------------------------------- PD interfaces
     def prefill_export(self, prompt_token_ids: List[int]):
         """PD producer: prefill one request, sample its first token, and

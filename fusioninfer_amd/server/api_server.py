@@ -120,24 +120,30 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
 
         if body.get("stream"):
             async def sse():
-                async for tok, finished in _stream(q):
-                    delta = decode_tokens([tok]) if tok is not None else ""
-                    chunk = {
-                        "id": cid,
-                        "object": "text_completion",
-                        "created": created,
-                        "model": model_name,
-                        "choices": [
-                            {
-                                "index": 0,
-                                "text": delta,
-                                "token_ids": [tok] if tok is not None else [],
-                                "finish_reason": "stop" if finished else None,
-                            }
-                        ],
-                    }
-                    yield f"data: {json.dumps(chunk)}\n\n"
-                yield "data: [DONE]\n\n"
+                done = False
+                try:
+                    async for tok, finished in _stream(q):
+                        done = finished
+                        delta = decode_tokens([tok]) if tok is not None else ""
+                        chunk = {
+                            "id": cid,
+                            "object": "text_completion",
+                            "created": created,
+                            "model": model_name,
+                            "choices": [
+                                {
+                                    "index": 0,
+                                    "text": delta,
+                                    "token_ids": [tok] if tok is not None else [],
+                                    "finish_reason": "stop" if finished else None,
+                                }
+                            ],
+                        }
+                        yield f"data: {json.dumps(chunk)}\n\n"
+                    yield "data: [DONE]\n\n"
+                finally:
+                    if not done:  # client disconnected mid-stream
+                        serving.abort(req_id)
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 

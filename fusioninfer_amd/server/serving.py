@@ -48,6 +48,11 @@ class ServingEngine:
             "request_success_total": float(e.num_finished),
         }
 
+    def abort(self, request_id: str) -> bool:
+        with self._lock:
+            self._streams.pop(request_id, None)
+            return self.engine.abort_request(request_id)
+
     def shutdown(self):
         self._stop = True
         self._work.set()

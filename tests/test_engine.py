@@ -188,3 +188,19 @@ def test_chunked_prefill_interleaves_decode():
     while eng.has_unfinished():
         eng.step()
     assert eng.num_finished == 2
+
+
+def test_abort_request():
+    torch.manual_seed(0)
+    eng = make_engine()
+    a = eng.add_request([1, 2, 3] * 10, SamplingParams(max_tokens=50))
+    b = eng.add_request([4, 5] * 10, SamplingParams(max_tokens=5))
+    for _ in range(3):
+        eng.step()
+    free_before = eng.block_manager.num_free()
+    assert eng.abort_request(a)
+    assert eng.block_manager.num_free() > free_before
+    assert not eng.abort_request(a)  # already gone
+    while eng.has_unfinished():
+        eng.step()
+    assert eng.num_finished == 1  # only b completed
