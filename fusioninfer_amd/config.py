@@ -55,6 +55,10 @@ class SchedulerConfig:
     max_num_seqs: int = 256                # max sequences resident per step
     max_num_batched_tokens: int = 8192     # per-step token budget
     max_model_len: int = 8192
+    # admission hysteresis: hold NEW prompts until this many prompt tokens
+    # are waiting (or nothing is decoding), so most steps stay pure-decode
+    # and take the hipGraph path; 0 = admit eagerly every step
+    prefill_admission_tokens: int = 4096
 
 
 @dataclasses.dataclass

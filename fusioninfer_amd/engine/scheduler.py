@@ -101,9 +101,19 @@ class Scheduler:
             prefill.append(seq)
             chunks.append(chunk)
             budget -= chunk
-        # then admit waiting prompts
+        # then admit waiting prompts — with hysteresis: when decodes are
+        # running, hold admissions until enough prompt tokens queue up so
+        # most steps stay pure-decode (hipGraph path)
+        waiting_tokens = sum(s.num_prompt_tokens for s in self.waiting)
+        threshold = min(self.cfg.prefill_admission_tokens, budget)
+        if not decode or prefill:
+            # idle decode path, or this step is mixed anyway: admit freely
+            admit = True
+        else:
+            admit = waiting_tokens >= threshold
         while (
-            budget > 0
+            admit
+            and budget > 0
             and self.waiting
             and len(self.running) < self.cfg.max_num_seqs
         ):
