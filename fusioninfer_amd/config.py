@@ -58,7 +58,7 @@ class SchedulerConfig:
     # admission hysteresis: hold NEW prompts until this many prompt tokens
     # are waiting (or nothing is decoding), so most steps stay pure-decode
     # and take the hipGraph path; 0 = admit eagerly every step
-    prefill_admission_tokens: int = 4096
+    prefill_admission_tokens: int = 8192
 
 
 @dataclasses.dataclass
