@@ -85,6 +85,14 @@ class RowParallelLinear(nn.Module):
         out = F.linear(x, self.weight)
         return ps.tp_all_reduce(out)
 
+    def forward_with_lora(self, x, lora, layer_idx, target):
+        """LoRA delta added BEFORE the TP all-reduce (A is input-sharded,
+        so per-rank deltas are partial sums)."""
+        out = F.linear(x, self.weight)
+        if lora is not None:
+            lora.apply(layer_idx, target, x, out)
+        return ps.tp_all_reduce(out)
+
 
 class ReplicatedLinear(nn.Module):
     def __init__(self, in_features: int, out_features: int, dtype=torch.bfloat16):

@@ -69,15 +69,35 @@ class LLMEngine:
         self.num_prefilled_tokens = 0
 
     # ------------------------------------------------------------ requests
+    def add_lora(self, name: str, rank: int = 16, alpha: float = 32.0,
+                 seed: Optional[int] = None):
+        """Register a LoRA adapter (random-init when seed given, zero-B
+        otherwise — zero-B adapters are numerically the base model)."""
+        from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
+
+        if self.runner.lora_registry is None:
+            self.runner.lora_registry = LoRARegistry()
+        adapter = LoRAAdapter(
+            name, rank, alpha, self.cfg.model, device=self.device, seed=seed
+        )
+        self.runner.lora_registry.add(adapter)
+        return adapter
+
+    def active_loras(self) -> List[str]:
+        reg = self.runner.lora_registry
+        return reg.names() if reg else []
+
     def add_request(
         self,
         prompt_token_ids: List[int],
         sampling: Optional[SamplingParams] = None,
         request_id: Optional[str] = None,
+        lora_name: Optional[str] = None,
     ) -> str:
         if request_id is None:
             request_id = f"req-{next(self._req_counter)}"
         seq = Sequence(request_id, prompt_token_ids, sampling or SamplingParams())
+        seq.lora_name = lora_name
         assert seq.num_prompt_tokens <= self.cfg.scheduler.max_model_len
         self.seqs[request_id] = seq
         self.scheduler.add(seq)

@@ -27,6 +27,8 @@ class AttnMetadata:
     # decode segment
     block_tables: Optional[torch.Tensor] = None  # [Dq, max_blocks] int32
     seq_lens: Optional[torch.Tensor] = None      # [Dq] int32 (ctx incl. current)
+    # LoRA groups for this step (fusioninfer_amd.lora.LoRABatch) or None
+    lora: object = None
 
     @property
     def num_tokens(self) -> int:

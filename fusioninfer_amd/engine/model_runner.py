@@ -45,6 +45,7 @@ class ModelRunner:
         self._graphs: Dict[int, Tuple[object, torch.Tensor]] = {}
         self._graph_pool = None
         self._static: Dict[str, torch.Tensor] = {}
+        self.lora_registry = None  # set by the engine when LoRA is enabled
 
     # ------------------------------------------------------------ kv cache
     def kv_block_bytes(self) -> int:
@@ -135,7 +136,38 @@ class ModelRunner:
             "total_lens": total_lens,
             "bt": bt,
             "sample": sample,
+            "lora_names": [s.lora_name for s in seqs],
         }
+
+    def _lora_batch(self, prefill_payload, decode_payload, np_=0):
+        """Group this step's token rows by adapter -> LoRABatch or None."""
+        if self.lora_registry is None:
+            return None
+        groups: Dict[str, List[int]] = {}
+        if prefill_payload is not None:
+            cu = prefill_payload["cu"]
+            for i, name in enumerate(prefill_payload.get("lora_names") or []):
+                if name:
+                    groups.setdefault(name, []).extend(
+                        range(cu[i], cu[i + 1])
+                    )
+        if decode_payload is not None:
+            for i, name in enumerate(decode_payload.get("lora_names") or []):
+                if name:
+                    groups.setdefault(name, []).append(np_ + i)
+        if not groups:
+            return None
+        from fusioninfer_amd.lora import LoRABatch
+
+        return LoRABatch(
+            [
+                (
+                    self.lora_registry.get(name),
+                    torch.tensor(rows, dtype=torch.long, device=self.device),
+                )
+                for name, rows in groups.items()
+            ]
+        )
 
     def run_prefill(self, payload) -> torch.Tensor:
         dev = self.device
@@ -162,6 +194,7 @@ class ModelRunner:
                 payload["total_lens"], dtype=torch.int32, device=dev
             ),
         )
+        meta.lora = self._lora_batch(payload, None)
         ids = torch.tensor(payload["ids"], dtype=torch.long, device=dev)
         sample = payload.get("sample") or [True] * (len(cu) - 1)
         logits_idx = torch.tensor(
@@ -245,6 +278,7 @@ class ModelRunner:
             block_tables=dbt.to(dev),
             seq_lens=torch.tensor(d["lens"], dtype=torch.int32, device=dev),
         )
+        meta.lora = self._lora_batch(payload, d, np_=np_)
         ids = torch.tensor(
             payload["ids"] + d["ids"], dtype=torch.long, device=dev
         )
@@ -272,11 +306,12 @@ class ModelRunner:
             "seq_lens": torch.ones(max_bs, dtype=torch.int32, device=dev),
         }
 
-    def _decode_forward(self, bs: int) -> torch.Tensor:
+    def _decode_forward(self, bs: int, lora=None) -> torch.Tensor:
         s = self._static
         meta = AttnMetadata(
             num_prefill_tokens=0,
             num_decode_tokens=bs,
+            lora=lora,
             positions=s["positions"][:bs],
             slot_mapping=s["slots"][:bs],
             block_tables=s["block_tables"][:bs],
@@ -316,6 +351,7 @@ class ModelRunner:
             "slots": [bm.slot_for(s, s.num_tokens - 1) for s in seqs],
             "lens": [s.num_tokens for s in seqs],
             "bt": [list(s.block_ids) for s in seqs],
+            "lora_names": [s.lora_name for s in seqs],
         }
 
     def _fill_decode_inputs(self, payload, bs: int) -> None:
@@ -351,9 +387,11 @@ class ModelRunner:
         n = len(payload["ids"])
         if not self._static:
             self._alloc_static(max(self.cfg.scheduler.max_num_seqs, n))
+        lora = self._lora_batch(None, payload, np_=0)
         bucket = next((b for b in _DECODE_BUCKETS if b >= n), None)
         use_graph = (
             bucket is not None and bucket in self._graphs and self.is_cuda
+            and lora is None  # LoRA steps run eager
         )
         bs = bucket if use_graph else n
         self._fill_decode_inputs(payload, bs)
@@ -362,7 +400,7 @@ class ModelRunner:
             g.replay()
             return out[:n]
         with torch.no_grad():
-            return self._decode_forward(n)[:n]
+            return self._decode_forward(n, lora=lora)[:n]
 
     def execute_decode(self, seqs: List[Sequence], bm: BlockManager):
         return self.run_decode(self.build_decode_payload(seqs, bm))

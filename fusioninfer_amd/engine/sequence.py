@@ -51,6 +51,8 @@ class Sequence:
         self.num_computed_tokens = 0
         # PD producer: keep cache blocks alive after finish for KV export
         self.hold_blocks = False
+        # LoRA adapter name (None = base model)
+        self.lora_name = None
 
     @property
     def num_prompt_tokens(self) -> int:

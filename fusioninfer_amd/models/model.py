@@ -28,8 +28,9 @@ from fusioninfer_amd.engine.metadata import AttnMetadata
 
 
 class Attention(nn.Module):
-    def __init__(self, cfg: ModelConfig):
+    def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
         super().__init__()
+        self.layer_idx = layer_idx
         tp = ps.tp_world_size()
         assert cfg.num_heads % tp == 0 and cfg.num_kv_heads % tp == 0, (
             "head counts must divide TP size"
@@ -70,6 +71,8 @@ class Attention(nn.Module):
         cos_sin: torch.Tensor,
     ) -> torch.Tensor:
         qkv = self.qkv_proj(x)
+        if meta.lora is not None:
+            meta.lora.apply(self.layer_idx, "qkv", x, qkv)
         q = qkv[:, : self.q_size]
         k = qkv[:, self.q_size : self.q_size + self.kv_size]
         v = qkv[:, self.q_size + self.kv_size :]
@@ -109,26 +112,36 @@ class Attention(nn.Module):
                 self.scale,
             )
             out[np_:] = o.view(nd, self.q_size)
+        if meta.lora is not None:
+            return self.o_proj.forward_with_lora(out, meta.lora,
+                                                 self.layer_idx, "o")
         return self.o_proj(out)
 
 
 class MLP(nn.Module):
-    def __init__(self, cfg: ModelConfig):
+    def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
         super().__init__()
+        self.layer_idx = layer_idx
         self.gate_up_proj = MergedColumnParallelLinear(
             cfg.hidden_size, [cfg.intermediate_size, cfg.intermediate_size]
         )
         self.down_proj = RowParallelLinear(cfg.intermediate_size, cfg.hidden_size)
         self.inter_per_rank = cfg.intermediate_size // ps.tp_world_size()
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
         gu = self.gate_up_proj(x)
-        # column-parallel packs [gate_shard | up_shard] per rank already
-        return self.down_proj(ops.silu_and_mul(gu))
+        if lora is not None:
+            lora.apply(self.layer_idx, "gate_up", x, gu)
+        act = ops.silu_and_mul(gu)
+        if lora is not None:
+            return self.down_proj.forward_with_lora(
+                act, lora, self.layer_idx, "down"
+            )
+        return self.down_proj(act)
 
 
 class DecoderLayer(nn.Module):
-    def __init__(self, cfg: ModelConfig):
+    def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
         super().__init__()
         self.input_norm_weight = nn.Parameter(
             torch.ones(cfg.hidden_size, dtype=torch.bfloat16), requires_grad=False
@@ -136,8 +149,8 @@ class DecoderLayer(nn.Module):
         self.post_norm_weight = nn.Parameter(
             torch.ones(cfg.hidden_size, dtype=torch.bfloat16), requires_grad=False
         )
-        self.self_attn = Attention(cfg)
-        self.mlp = MLP(cfg)
+        self.self_attn = Attention(cfg, layer_idx)
+        self.mlp = MLP(cfg, layer_idx)
         self.eps = cfg.rms_norm_eps
 
     def forward(self, hidden, residual, meta, kv_cache, cos_sin):
@@ -152,7 +165,7 @@ class DecoderLayer(nn.Module):
         hidden, residual = ops.fused_add_rms_norm(
             hidden, residual, self.post_norm_weight, self.eps
         )
-        hidden = self.mlp(hidden)
+        hidden = self.mlp(hidden, lora=meta.lora)
         return hidden, residual
 
 
@@ -168,7 +181,7 @@ class CausalLM(nn.Module):
             requires_grad=False,
         )
         self.layers = nn.ModuleList(
-            [DecoderLayer(cfg) for _ in range(cfg.num_layers)]
+            [DecoderLayer(cfg, i) for i in range(cfg.num_layers)]
         )
         self.final_norm_weight = nn.Parameter(
             torch.ones(cfg.hidden_size, dtype=torch.bfloat16), requires_grad=False

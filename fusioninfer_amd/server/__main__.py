@@ -33,6 +33,10 @@ def parse_args(argv=None):
                    help='JSON: {"kv_connector": "RcclConnector", '
                         '"kv_role": "kv_producer"|"kv_consumer"}')
     p.add_argument("--enforce-eager", action="store_true")
+    p.add_argument("--enable-lora", action="store_true")
+    p.add_argument("--max-loras", type=int, default=8)
+    p.add_argument("--lora-modules", nargs="*", default=[],
+                   help="adapters to register at startup: name[=rank]")
     p.add_argument("--enable-prefix-caching", action="store_true")
     # multi-node rendezvous flags injected by the LWS wrapper
     p.add_argument("--nnodes", type=int, default=1)
@@ -98,6 +102,10 @@ def main(argv=None):
 
     cfg = build_engine_config(args)
     serving = ServingEngine(cfg)
+    if args.enable_lora or args.lora_modules:
+        for spec in args.lora_modules:
+            name, _, rank = spec.partition("=")
+            serving.engine.add_lora(name, rank=int(rank) if rank else 16)
     app = build_app(serving, args.model)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 

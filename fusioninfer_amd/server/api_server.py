@@ -56,12 +56,15 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
 
     @app.get("/v1/models")
     async def models():
-        return {
-            "object": "list",
-            "data": [
-                {"id": model_name, "object": "model", "owned_by": "fusioninfer-amd"}
-            ],
-        }
+        data = [
+            {"id": model_name, "object": "model", "owned_by": "fusioninfer-amd"}
+        ]
+        for name in serving.engine.active_loras():
+            data.append(
+                {"id": name, "object": "model", "owned_by": "fusioninfer-amd",
+                 "parent": model_name}
+            )
+        return {"object": "list", "data": data}
 
     @app.get("/metrics")
     async def metrics():
@@ -114,7 +117,11 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         prompt = body.get("prompt", "")
         prompt_ids = encode_prompt(prompt, vocab)
         sampling = _sampling_from(body)
-        req_id, q = serving.submit(prompt_ids, sampling)
+        # OpenAI-style adapter selection: model == a registered LoRA name
+        lora = body.get("model")
+        if lora not in serving.engine.active_loras():
+            lora = None
+        req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
         created = int(time.time())
         cid = f"cmpl-{uuid.uuid4().hex[:16]}"
 
@@ -179,7 +186,10 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         )
         prompt_ids = encode_prompt(text, vocab)
         sampling = _sampling_from(body)
-        req_id, q = serving.submit(prompt_ids, sampling)
+        lora = body.get("model")
+        if lora not in serving.engine.active_loras():
+            lora = None
+        req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
         created = int(time.time())
         cid = f"chatcmpl-{uuid.uuid4().hex[:16]}"
 

@@ -28,11 +28,14 @@ class ServingEngine:
 
     # ------------------------------------------------------------- public
     def submit(
-        self, prompt_token_ids: List[int], sampling: SamplingParams
+        self, prompt_token_ids: List[int], sampling: SamplingParams,
+        lora_name=None,
     ) -> Tuple[str, "queue.Queue[Tuple[Optional[int], bool]]"]:
         q: "queue.Queue[Tuple[Optional[int], bool]]" = queue.Queue()
         with self._lock:
-            req_id = self.engine.add_request(prompt_token_ids, sampling)
+            req_id = self.engine.add_request(
+                prompt_token_ids, sampling, lora_name=lora_name
+            )
             self._streams[req_id] = q
         self._work.set()
         return req_id, q
