@@ -67,6 +67,10 @@ class LLMEngine:
         self.num_finished = 0
         self.num_generated_tokens = 0
         self.num_prefilled_tokens = 0
+        self.ttft_sum = 0.0
+        self.e2e_latency_sum = 0.0
+        self.step_time_sum = 0.0
+        self.num_steps = 0
 
     # ------------------------------------------------------------ requests
     def add_lora(self, name: str, rank: int = 16, alpha: float = 32.0,
@@ -202,6 +206,7 @@ class LLMEngine:
         ranks (in worker_loop) execute the same forward so the per-layer
         RCCL all-reduces line up."""
         assert self.is_driver, "only TP rank 0 steps; others run worker_loop"
+        step_t0 = time.monotonic()
         tp = self.cfg.parallel.tensor_parallel_size
         batch = self.scheduler.schedule()
         if batch.is_empty:
@@ -235,8 +240,13 @@ class LLMEngine:
                 seq.finish_time = time.monotonic()
                 self.scheduler.finish(seq)
                 self.num_finished += 1
+                if seq.ttft is not None:
+                    self.ttft_sum += seq.ttft
+                self.e2e_latency_sum += seq.finish_time - seq.arrival_time
                 del self.seqs[seq.seq_id]
             outputs.append(RequestOutput(seq))
+        self.step_time_sum += time.monotonic() - step_t0
+        self.num_steps += 1
         return outputs
 
     # --------------------------------------------------------- TP workers

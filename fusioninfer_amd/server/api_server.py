@@ -77,6 +77,12 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             ("generation_tokens_total", "counter"),
             ("prompt_tokens_total", "counter"),
             ("request_success_total", "counter"),
+            ("time_to_first_token_seconds_sum", "counter"),
+            ("time_to_first_token_seconds_count", "counter"),
+            ("e2e_request_latency_seconds_sum", "counter"),
+            ("e2e_request_latency_seconds_count", "counter"),
+            ("engine_step_seconds_sum", "counter"),
+            ("engine_step_seconds_count", "counter"),
         ]:
             lines.append(f"# TYPE vllm:{name} {mtype}")
             lines.append(

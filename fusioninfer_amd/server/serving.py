@@ -49,6 +49,12 @@ class ServingEngine:
             "generation_tokens_total": float(e.num_generated_tokens),
             "prompt_tokens_total": float(e.num_prefilled_tokens),
             "request_success_total": float(e.num_finished),
+            "time_to_first_token_seconds_sum": e.ttft_sum,
+            "time_to_first_token_seconds_count": float(e.num_finished),
+            "e2e_request_latency_seconds_sum": e.e2e_latency_sum,
+            "e2e_request_latency_seconds_count": float(e.num_finished),
+            "engine_step_seconds_sum": e.step_time_sum,
+            "engine_step_seconds_count": float(e.num_steps),
         }
 
     def abort(self, request_id: str) -> bool:
