@@ -27,15 +27,6 @@
 
 namespace fi {
 
-typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2v;
-
-FI_DEV float dot2_bf16(u32 a_pair, u32 b_pair, float c) {
-  // v_dot2_f32_bf16: c + a0*b0 + a1*b1 on packed bf16 pairs
-  return __builtin_amdgcn_fdot2_f32_bf16(
-      __builtin_bit_cast(bf16x2v, a_pair),
-      __builtin_bit_cast(bf16x2v, b_pair), c, false);
-}
-
 constexpr int kBlockSz = 16;     // cache block size (tokens)
 constexpr int kNWaves = 4;
 constexpr int kPartChunks = 32;  // 512 tokens per partition
@@ -79,18 +70,18 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     return;
   }
 
-  __shared__ u32 q_lds[G][D / 2];   // packed bf16 pairs (fed to v_dot2)
+  __shared__ float q_lds[G][D];
   __shared__ float p_lds[kNWaves][kBlockSz][G];
   __shared__ float merge_m[kNWaves][G];
   __shared__ float merge_l[kNWaves][G];
   __shared__ float merge_acc[kNWaves][G][D];
 
-  // stage q (G heads) into LDS as raw bf16 pairs; scale applied post-dot
+  // stage q (G heads) into LDS as fp32 (pre-scaled)
   {
-    const u32* q_row = reinterpret_cast<const u32*>(
-        q + seq * q_stride + static_cast<int64_t>(kv_head) * G * D);
-    for (int e = tid; e < G * D / 2; e += kNWaves * kWaveSize)
-      q_lds[e / (D / 2)][e % (D / 2)] = q_row[e];
+    const u16* q_row = q + seq * q_stride +
+                       static_cast<int64_t>(kv_head) * G * D;
+    for (int e = tid; e < G * D; e += kNWaves * kWaveSize)
+      q_lds[e / D][e % D] = bf16_to_f32(q_row[e]) * scale;
   }
   __syncthreads();
 
@@ -113,28 +104,31 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
         ((static_cast<int64_t>(block_id) * num_kv_heads + kv_head) * kBlockSz) * D;
     const int token_pos = chunk * kBlockSz + tok;
 
-    // ---- phase A: scores for 16 tokens x G heads (packed bf16 dot2) ----
+    // ---- phase A: scores for 16 tokens x G heads ----
     float s[G];
     {
       const u16* k_row = k_cache + kv_base + tok * D + quad * DPQ;
-      uint4 kq[DPQ / 8];   // raw bf16 pairs of K (16 B loads)
+      float kf[DPQ];
 #pragma unroll
-      for (int j8 = 0; j8 < DPQ / 8; ++j8)
-        kq[j8] = *reinterpret_cast<const uint4*>(k_row + j8 * 8);
+      for (int j8 = 0; j8 < DPQ / 8; ++j8) {
+        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k_row + j8 * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kf[j8 * 8 + j] = bf16_to_f32(kv8.h[j]);
+      }
 #pragma unroll
       for (int g = 0; g < G; ++g) {
         float acc_s = 0.f;
 #pragma unroll
-        for (int j8 = 0; j8 < DPQ / 8; ++j8) {
-          // uint4 -> ds_read_b128 (4 packed q pairs per read)
-          const uint4 qv = *reinterpret_cast<const uint4*>(
-              &q_lds[g][quad * (DPQ / 2) + j8 * 4]);
-          acc_s = dot2_bf16(kq[j8].x, qv.x, acc_s);
-          acc_s = dot2_bf16(kq[j8].y, qv.y, acc_s);
-          acc_s = dot2_bf16(kq[j8].z, qv.z, acc_s);
-          acc_s = dot2_bf16(kq[j8].w, qv.w, acc_s);
+        for (int j4 = 0; j4 < DPQ / 4; ++j4) {
+          // float4 -> ds_read_b128 (4x fewer LDS cycles than scalar reads)
+          const float4 qv = *reinterpret_cast<const float4*>(
+              &q_lds[g][quad * DPQ + j4 * 4]);
+          acc_s = fmaf(kf[j4 * 4 + 0], qv.x, acc_s);
+          acc_s = fmaf(kf[j4 * 4 + 1], qv.y, acc_s);
+          acc_s = fmaf(kf[j4 * 4 + 2], qv.z, acc_s);
+          acc_s = fmaf(kf[j4 * 4 + 3], qv.w, acc_s);
         }
-        s[g] = acc_s * scale;
+        s[g] = acc_s;
       }
     }
     // reduce over the 4 dim-quarters (lanes 4t..4t+3)

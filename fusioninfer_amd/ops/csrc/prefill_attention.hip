@@ -225,28 +225,21 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
         __builtin_amdgcn_s_setprio(0);
 
         // ---- mask + online softmax over the 16x64 block (in s_acc) ----
-        // scores stay RAW (scale > 0 preserves order); the scale folds
-        // into the exp2 argument: p = 2^((s - m) * scale * log2 e)
         const int q_base = row0 + rb * 16 + hi * 4;
-        const bool need_mask =
-            (kv0 + kKVTile - 1 > ctx_start + q_base)   // causal edge
-            || (kv0 + kKVTile > k_len)                  // ragged kv tail
-            || (row0 + kQPerWave > seq_len);            // invalid q rows
-        if (need_mask) {
 #pragma unroll
-          for (int cb4 = 0; cb4 < 4; ++cb4) {
-            const int kv_pos = kv0 + cb4 * 16 + col;  // absolute position
+        for (int cb4 = 0; cb4 < 4; ++cb4) {
+          const int kv_pos = kv0 + cb4 * 16 + col;  // absolute position
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              const int q_pos = ctx_start + q_base + r;
-              if (kv_pos > q_pos || q_base + r >= seq_len || kv_pos >= k_len)
-                s_acc[cb4][r] = kPNegInf;
-            }
+          for (int r = 0; r < 4; ++r) {
+            const int q_pos = ctx_start + q_base + r;  // absolute position
+            float sv = s_acc[cb4][r] * scale;
+            if (kv_pos > q_pos || q_base + r >= seq_len || kv_pos >= k_len)
+              sv = kPNegInf;
+            s_acc[cb4][r] = sv;
           }
         }
-        const float sl2e = scale * 1.44269504088896f;  // scale * log2(e)
-        bool grew = false;
-        float m_news[4], alphas[4];
+        // o_acc is rescaled HERE (phase 1) so nothing but P crosses the
+        // barrier into the PV phase
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           float rm = fmaxf(fmaxf(s_acc[0][r], s_acc[1][r]),
@@ -255,34 +248,21 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
           for (int off = 1; off < 16; off <<= 1)
             rm = fmaxf(rm, __shfl_xor(rm, off, 64));
           const float m_new = fmaxf(m[rb][r], rm);
-          m_news[r] = m_new;
+          float alpha;
           if (m_new <= kPNegInf) {  // row fully masked so far
-            alphas[r] = 0.f;
+            alpha = 0.f;
           } else {
-            alphas[r] = (m[rb][r] <= kPNegInf)
-                            ? 0.f
-                            : exp2f((m[rb][r] - m_new) * sl2e);
-            grew = grew || (m_new > m[rb][r]);
+            alpha = (m[rb][r] <= kPNegInf) ? 0.f : __expf(m[rb][r] - m_new);
             m[rb][r] = m_new;
           }
-        }
-        // skip the 64-multiply o rescale on tiles where no row max grew
-        // (the common interior case) — alpha == 1 exactly there
-        if (__any(grew)) {
 #pragma unroll
-          for (int r = 0; r < 4; ++r)
-#pragma unroll
-            for (int cb = 0; cb < CB; ++cb) o_acc[rb][cb][r] *= alphas[r];
-        }
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
+          for (int cb = 0; cb < CB; ++cb) o_acc[rb][cb][r] *= alpha;
           float rs = 0.f;
-          const float m_new = m_news[r];
 #pragma unroll
           for (int cb4 = 0; cb4 < 4; ++cb4) {
             const float pv = (m_new <= kPNegInf || s_acc[cb4][r] <= kPNegInf)
                                  ? 0.f
-                                 : exp2f((s_acc[cb4][r] - m_new) * sl2e);
+                                 : __expf(s_acc[cb4][r] - m_new);
             rs += pv;
             *reinterpret_cast<u16*>(
                 pbase + swz<kVTRowB>(rb * 16 + hi * 4 + r,
@@ -290,7 +270,7 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
           }
 #pragma unroll
           for (int off = 1; off < 16; off <<= 1) rs += __shfl_xor(rs, off, 64);
-          l[rb][r] = l[rb][r] * alphas[r] + rs;
+          l[rb][r] = l[rb][r] * alpha + rs;
         }
       }
     }
