@@ -41,6 +41,10 @@ def parse_args():
     p.add_argument("--max-batched-tokens", type=int, default=8192)
     p.add_argument("--enforce-eager", action="store_true")
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree")
+    p.add_argument("--shared-prefix-len", type=int, default=0,
+                   help="shared-prefix workload (BASELINE config #4): each "
+                        "prompt = one of 8 shared prefixes of this length + "
+                        "a unique tail; exercises prefix caching")
     return p.parse_args()
 
 
@@ -48,7 +52,7 @@ class ClosedLoopLoad:
     """Keeps `concurrency` requests in flight; tracks completions + TTFT."""
 
     def __init__(self, engine, prompt_len: int, gen_len: int, concurrency: int,
-                 vocab: int, seed: int):
+                 vocab: int, seed: int, shared_prefix_len: int = 0):
         self.engine = engine
         self.gen_len = gen_len
         self.prompt_len = prompt_len
@@ -58,8 +62,22 @@ class ClosedLoopLoad:
         self.completion_times = []
         self.first_token_times = {}
         self.ttfts = []
+        self.prefixes = []
+        if shared_prefix_len > 0:
+            self.prefixes = [
+                torch.randint(0, vocab, (shared_prefix_len,),
+                              generator=self.rng).tolist()
+                for _ in range(8)
+            ]
 
     def _new_prompt(self):
+        if self.prefixes:
+            i = int(torch.randint(0, len(self.prefixes), (1,),
+                                  generator=self.rng))
+            tail_len = max(self.prompt_len - len(self.prefixes[i]), 8)
+            tail = torch.randint(0, self.vocab, (tail_len,),
+                                 generator=self.rng).tolist()
+            return self.prefixes[i] + tail
         return torch.randint(
             0, self.vocab, (self.prompt_len,), generator=self.rng
         ).tolist()
@@ -152,6 +170,7 @@ def main():
     load = ClosedLoopLoad(
         engine, args.prompt_len, args.gen_len, args.concurrency,
         mc.vocab_size, seed=99 + rank,
+        shared_prefix_len=args.shared_prefix_len,
     )
     load.top_up()
 
