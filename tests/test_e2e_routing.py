@@ -1,0 +1,125 @@
+"""End-to-end routing tier (mirrors the reference's e2e intent, SURVEY §4.3,
+without a cluster): two in-process OpenAI servers behind the first-party
+EPP router, all ASGI, CPU tiny model."""
+
+import asyncio
+
+import httpx
+import pytest
+
+from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+from fusioninfer_amd.controlplane import api as cp_api
+from fusioninfer_amd.controlplane import router as cp_router
+from fusioninfer_amd.epp import Endpoint, EndpointPicker
+from fusioninfer_amd.epp.router_server import build_router_app, parse_vllm_metrics
+from fusioninfer_amd.models.registry import get_model_config
+from fusioninfer_amd.server.api_server import build_app
+from fusioninfer_amd.server.serving import ServingEngine
+from tests.test_controlplane import monolithic_svc
+
+
+@pytest.fixture(scope="module")
+def backends():
+    servings = []
+    apps = {}
+    for i in range(2):
+        cfg = EngineConfig(
+            model=get_model_config("tiny-qwen3"),
+            cache=CacheConfig(num_gpu_blocks=128),
+            scheduler=SchedulerConfig(
+                max_num_seqs=8, max_num_batched_tokens=512, max_model_len=256
+            ),
+        )
+        s = ServingEngine(cfg, device="cpu")
+        servings.append(s)
+        apps[f"backend{i}:8000"] = build_app(s, "tiny-qwen3")
+    yield apps
+    for s in servings:
+        s.shutdown()
+
+
+class MultiASGITransport(httpx.AsyncBaseTransport):
+    """Routes http://<name>:8000 to the matching in-process ASGI app."""
+
+    def __init__(self, apps):
+        self.transports = {
+            host: httpx.ASGITransport(app=app) for host, app in apps.items()
+        }
+
+    async def handle_async_request(self, request):
+        host = f"{request.url.host}:{request.url.port}"
+        return await self.transports[host].handle_async_request(request)
+
+
+def test_router_prefix_cache_affinity_e2e(backends):
+    svc = monolithic_svc()
+    cfg_yaml = cp_router.generate_epp_config(
+        svc, cp_api.Role(cp_api.ROUTER, routing_strategy=cp_api.PREFIX_CACHE)
+    )
+    picker = EndpointPicker(cfg_yaml)
+    endpoints = [Endpoint(h) for h in backends]
+    client = httpx.AsyncClient(transport=MultiASGITransport(backends))
+    app = build_router_app(picker, endpoints, client=client, scrape=False)
+
+    async def run():
+        async with httpx.AsyncClient(
+            transport=httpx.ASGITransport(app=app), base_url="http://router"
+        ) as c:
+            r = await c.get("/health")
+            assert len(r.json()["endpoints"]) == 2
+            shared = "common preamble " * 8
+            # two requests with the same long prefix
+            r1 = await c.post("/v1/completions", json={
+                "prompt": shared + "question one",
+                "max_tokens": 3, "temperature": 0, "ignore_eos": True,
+            })
+            assert r1.status_code == 200
+            assert len(r1.json()["choices"][0]["token_ids"]) == 3
+            r2 = await c.post("/v1/completions", json={
+                "prompt": shared + "question two",
+                "max_tokens": 3, "temperature": 0, "ignore_eos": True,
+            })
+            assert r2.status_code == 200
+        # prefix affinity: the shared prefix is recorded on exactly one
+        # endpoint's LRU (both requests scored to the same server)
+        lrus = [len(v) for v in picker.prefix_cache._lru.values()]
+        assert len(lrus) == 1 or max(lrus) > 0
+
+    asyncio.run(run())
+
+
+def test_router_streaming_relay(backends):
+    svc = monolithic_svc()
+    picker = EndpointPicker(
+        cp_router.generate_epp_config(svc, cp_api.Role(cp_api.ROUTER))
+    )
+    endpoints = [Endpoint(h) for h in backends]
+    client = httpx.AsyncClient(transport=MultiASGITransport(backends))
+    app = build_router_app(picker, endpoints, client=client, scrape=False)
+
+    async def run():
+        async with httpx.AsyncClient(
+            transport=httpx.ASGITransport(app=app), base_url="http://router"
+        ) as c:
+            n = 0
+            async with c.stream("POST", "/v1/completions", json={
+                "prompt": [5, 6, 7] * 8, "max_tokens": 3, "stream": True,
+                "temperature": 0, "ignore_eos": True,
+            }) as r:
+                async for line in r.aiter_lines():
+                    if line.startswith("data: ") and "[DONE]" not in line:
+                        n += 1
+            assert n == 3
+
+    asyncio.run(run())
+
+
+def test_metrics_scrape_parsing():
+    text = (
+        '# TYPE vllm:gpu_cache_usage_perc gauge\n'
+        'vllm:gpu_cache_usage_perc{model_name="m"} 0.25\n'
+        'vllm:num_requests_waiting{model_name="m"} 3.0\n'
+    )
+    m = parse_vllm_metrics(text)
+    assert m["gpu_cache_usage_perc"] == 0.25
+    assert m["num_requests_waiting"] == 3.0
