@@ -1,49 +1,60 @@
-// Varlen causal prefill attention (flash-style, MFMA) for CDNA4 (gfx950) — v2.
+// Varlen causal prefill / context attention (flash-style, MFMA) — v3.
 //
-// out[t, h, :] = softmax(Q K^T * scale, causal) V, GQA, bf16 in / fp32 accum.
+// out[t, h, :] = softmax(Q K^T * scale, causal) V, GQA, bf16 in / fp32
+// accumulate. PAGED variant reads K/V from the paged cache via block tables
+// (q rows are only each sequence's NEW tokens — prefix-cache hits and
+// chunked prefill skip cached tokens).
 //
-// v2 structure (the 8-wave ladder of cdna_hip_programming.md §B):
-//  * workgroup = 8 waves; wave w owns q rows [row0 + 32w, +32) of one
-//    (seq, q-head) — a 256-row Q block per workgroup.
-//  * KV tiles of 64 tokens are staged ONCE per workgroup into shared LDS
-//    (cooperative coalesced loads, barrier-synced): an 8x cut in global
-//    K/V traffic vs per-wave reads (v1's top cost: 26% of bench GPU time).
-//  * K tile [64][256B] is XOR-swizzled (byte ^= (row&15)<<4) so the QK^T
-//    B-fragment ds_read_b128 is bank-conflict-free (guide T2/G4);
-//    V is staged TRANSPOSED ([128][128B] rows, byte ^= (row&7)<<4) so the
-//    PV B-fragment is a contiguous swizzled b128 read.
-//  * QK^T and PV both mfma_f32_16x16x32_bf16; P goes through per-wave
-//    swizzled LDS (bf16) to become the PV A-fragment.
-//  * online softmax per 16-row fragment block; C-fragment mapping:
-//    col = lane&15, row = (lane>>4)*4 + reg.
+// v3 structure (the 8-wave 32x32 swapped-operand ladder of
+// cdna_hip_programming.md §B):
+//  * workgroup = 8 waves; wave w owns q rows [row0 + 32w, +32); 256 q rows
+//    per workgroup; KV tiles of 64 staged once per workgroup.
+//  * K by global_load_lds (double-buffered, T2 swizzle on the SOURCE
+//    address); V register-staged early (T14) and written TRANSPOSED
+//    (V^T, swizzled rows) so the PV B-fragment is a clean ds_read_b128.
+//  * SWAPPED QK^T: S^T = mfma_f32_32x32x16_bf16(A=K, B=Q) puts each q
+//    row's scores lane-local (C col = lane&31 = q row) — the online
+//    softmax is pure in-register VALU + ONE shfl_xor(32) per reduction
+//    (v2's cross-lane softmax + P-LDS round trip measured 14 VALU insts
+//    per MFMA; this removes both).
+//  * P -> PV A-fragments in-register: pack bf16 pairs + permlane32_swap
+//    half-exchange (guide T12); PV = mfma_32x32x16(A=P, B=V^T read).
+//  * C-fragment row map (32x32): row = (r&3) + 8*(r>>2) + 4*(lane>>5),
+//    col = lane&31.
 //
-// Capability parity: the paged-attention prefill the reference delegates to
-// vLLM (SURVEY.md §2.3 "Paged-attention prefill kernel").
+// Capability parity: the paged-attention prefill the reference delegates
+// to vLLM.
 
 #include "common.h"
 
 namespace fi {
 
 typedef __attribute__((ext_vector_type(8))) short short8;
-typedef __attribute__((ext_vector_type(4))) float floatx4;
+typedef __attribute__((ext_vector_type(16))) float floatx16;
 
 constexpr int kWaves = 8;
-constexpr int kQPerWave = 32;                 // q rows per wave (2 MFMA blocks)
+constexpr int kQPerWave = 32;                 // q rows per wave (1 MFMA block)
 constexpr int kQBlock = kWaves * kQPerWave;   // 256 q rows per workgroup
 constexpr int kKVTile = 64;                   // kv tokens per LDS tile
 constexpr float kPNegInf = -1e30f;
 
-// byte-address XOR swizzles, bijective within a ROWB-byte row (guide T2/G4):
-// spread a 16-lane group's distinct-row b128 reads over ROWB/16 bank slots.
+// byte-address XOR swizzles, bijective within a ROWB-byte row (guide T2/G4)
 template <int ROWB>
 FI_DEV int swz(int row, int byte_in_row) {
   constexpr int kMask = (ROWB / 16 > 16 ? 16 : ROWB / 16) - 1;
   return row * ROWB + (byte_in_row ^ ((row & kMask) << 4));
 }
 
-// PAGED: K/V come from the paged cache ([B, Hk, 16, D]) via block_tables,
-// and q rows are only the NEW tokens of each sequence (context attention —
-// prefix-cache hits / chunked prefill skip cached tokens).
+// C-fragment row for reg r in a 32x32 MFMA (hi1 = lane>>5)
+FI_DEV constexpr int crow(int r, int hi1) {
+  return (r & 3) + 8 * (r >> 2) + 4 * hi1;
+}
+
+FI_DEV u32 pack_bf16x2(float lo, float hi) {
+  return static_cast<u32>(f32_to_bf16(lo)) |
+         (static_cast<u32>(f32_to_bf16(hi)) << 16);
+}
+
 template <int D, bool PAGED>
 __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     u16* __restrict__ out,        // [T, Hq, D]
@@ -58,8 +69,8 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     const int max_blocks,
     const int64_t q_stride, const int64_t k_stride, const int64_t v_stride,
     const int num_q_heads, const int num_kv_heads, const float scale) {
-  constexpr int KB = D / 32;   // QK^T k-chunks over the head dim
-  constexpr int CB = D / 16;   // PV output col blocks
+  constexpr int KF = D / 16;           // QK^T 16-deep k-chunks over head dim
+  constexpr int CB = D / 32;           // PV output 32-col blocks
   constexpr int kKRowB = D * 2;        // K row bytes (D=128 -> 256)
   constexpr int kVTRowB = kKVTile * 2; // V^T row bytes (64 kv -> 128)
 
@@ -68,20 +79,19 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
   const int tid = threadIdx.x;
   const int wave = tid / kWaveSize;
   const int lane = tid % kWaveSize;
-  const int col = lane & 15;
-  const int hi = lane >> 4;
+  const int col = lane & 31;           // this lane's q row (within the wave)
+  const int hi1 = lane >> 5;
 
   const int seq = tile_seq[blockIdx.x];
   const int seq_start = cu_seqlens[seq];
   const int seq_len = cu_seqlens[seq + 1] - seq_start;  // NEW q rows
   const int k_len = PAGED ? seq_lens_k[seq] : seq_len;  // total kv rows
-  const int ctx_start = k_len - seq_len;  // cached tokens before q row 0
+  const int ctx_start = k_len - seq_len;
   const int wg_row0 = tile_row0[blockIdx.x];
-  const int row0 = wg_row0 + wave * kQPerWave;      // this wave's first q row
+  const int row0 = wg_row0 + wave * kQPerWave;
   const bool active = row0 < seq_len;
-  const int* bt_row = PAGED ? block_tables + static_cast<int64_t>(seq) * max_blocks
-                            : nullptr;
-  // element offset of kv row r (absolute position) in the k/v source
+  const int* bt_row = PAGED
+      ? block_tables + static_cast<int64_t>(seq) * max_blocks : nullptr;
   auto kv_off = [&](int r, int64_t dense_stride) -> int64_t {
     if (PAGED) {
       const int blk = bt_row[r >> 4];
@@ -91,60 +101,45 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     return (seq_start + r) * dense_stride + static_cast<int64_t>(kv_head) * D;
   };
 
-  // LDS: double-buffered K (glds target) + V^T tile + P tiles (per-wave).
-  // ONE __shared__ object: a second one makes hipcc drain vmcnt(0) before
-  // every ds_read beside a glds pipeline (guide §5 ".s-level traps" (a)).
-  __shared__ u16 smem[2 * kKVTile * D + D * kKVTile +
-                      kWaves * kQPerWave * kKVTile];
-  u16* k_lds0 = smem;                            // swizzled K rows, buffer 0
-  u16* k_lds1 = smem + kKVTile * D;              // buffer 1
-  u16* vt_lds = smem + 2 * kKVTile * D;          // transposed V, swizzled rows
-  u16* p_lds = vt_lds + D * kKVTile + wave * (kQPerWave * kKVTile);
+  // LDS: double-buffered K (glds target) + V^T tile. ONE __shared__ object
+  // (guide §5 ".s-level traps" (a)).
+  __shared__ u16 smem[2 * kKVTile * D + D * kKVTile];
+  u16* k_lds0 = smem;
+  u16* k_lds1 = smem + kKVTile * D;
+  u16* vt_lds = smem + 2 * kKVTile * D;
 
-  // ---- Q fragments: a_q[rb][kb], lane holds Q[row0+rb*16+col][kb*32+hi*8..]
-  short8 a_q[2][KB];
+  // ---- Q fragments (the QK^T B operand): lane holds
+  // Q[row0+col][f*16 + hi1*8 .. +8] for each 16-deep k-chunk f
+  short8 b_q[KF];
   if (active) {
+    const int qr = min(row0 + col, seq_len - 1);
+    const u16* qrow = q + (seq_start + qr) * q_stride +
+                      static_cast<int64_t>(head) * D;
 #pragma unroll
-    for (int rb = 0; rb < 2; ++rb) {
-      const int qr = min(row0 + rb * 16 + col, seq_len - 1);
-      const u16* qrow = q + (seq_start + qr) * q_stride +
-                        static_cast<int64_t>(head) * D;
-#pragma unroll
-      for (int kb = 0; kb < KB; ++kb)
-        a_q[rb][kb] = *reinterpret_cast<const short8*>(qrow + kb * 32 + hi * 8);
-    }
+    for (int f = 0; f < KF; ++f)
+      b_q[f] = *reinterpret_cast<const short8*>(qrow + f * 16 + hi1 * 8);
   }
 
-  float m[2][4], l[2][4];
-  floatx4 o_acc[2][CB];
+  // online-softmax state: ONE q row per lane (lanes j and j+32 share row j)
+  float m_row = kPNegInf, l_row = 0.f;
+  floatx16 o_acc[CB];
 #pragma unroll
-  for (int rb = 0; rb < 2; ++rb) {
+  for (int cb = 0; cb < CB; ++cb)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      m[rb][r] = kPNegInf;
-      l[rb][r] = 0.f;
-    }
-#pragma unroll
-    for (int cb = 0; cb < CB; ++cb) o_acc[rb][cb] = {0.f, 0.f, 0.f, 0.f};
-  }
+    for (int r = 0; r < 16; ++r) o_acc[cb][r] = 0.f;
 
-  // kv range: the whole workgroup iterates to its max causal ABSOLUTE row
   const int wg_q_max = ctx_start + min(wg_row0 + kQBlock, seq_len) - 1;
   const int num_kv_tiles = wg_q_max / kKVTile + 1;
   const int my_q_max = ctx_start + min(row0 + kQPerWave, seq_len) - 1;
 
-  // ---- staging helpers -------------------------------------------------
-  // K tile -> LDS by global_load_lds (direct DMA, no VGPR round trip).
-  // glds writes lane-linear (base + lane*16), so the T2 swizzle moves to
-  // the per-lane SOURCE address (guide §5.4 rule 21).
+  // ---- staging (same pipeline as v2.1) --------------------------------
   constexpr int kKTileBytes = kKVTile * kKRowB;
   constexpr int kGldsPerWave = kKTileBytes / (kWaves * kWaveSize * 16);
   auto stage_k_glds = [&](int t, u16* kbuf) {
     const int kv0 = t * kKVTile;
 #pragma unroll
     for (int i = 0; i < kGldsPerWave; ++i) {
-      const int base_off =
-          wave * (kKTileBytes / kWaves) + i * (kWaveSize * 16);
+      const int base_off = wave * (kKTileBytes / kWaves) + i * (kWaveSize * 16);
       const int X = base_off + lane * 16;
       const int row = X / kKRowB;
       const int sbyte = X % kKRowB;
@@ -159,14 +154,12 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
           16, 0, 0);
     }
   };
-  // V: per-thread register load of one 16-elem row chunk (issued early so
-  // HBM latency hides under MFMA — guide T14), transposed into LDS later.
   constexpr int kChunks = D / 16;
-  const int v_kv = tid % kKVTile;          // this thread's kv row
-  const int v_c16 = tid / kKVTile;         // 16-elem chunk (D=128: 0..7)
+  const int v_kv = tid % kKVTile;
+  const int v_c16 = tid / kKVTile;
   bf16x8 vreg0, vreg1;
   auto vload = [&](int t) {
-    if (v_c16 >= kChunks) return;  // D=64: only 4 chunks per row
+    if (v_c16 >= kChunks) return;
     const int src = min(t * kKVTile + v_kv, k_len - 1);
     const u16* vrow = v + kv_off(src, v_stride) + v_c16 * 16;
     vreg0 = *reinterpret_cast<const bf16x8*>(vrow);
@@ -184,121 +177,138 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     }
   };
 
-  // ---- software-pipelined main loop -------------------------------------
-  // invariant at loop top: K(t) resident in kbuf(t&1) (glds issued and
-  // drained by the previous barrier), V(t) in registers.
   stage_k_glds(0, k_lds0);
   vload(0);
-  __syncthreads();  // drains the glds (vmcnt 0) and publishes K(0)
+  __syncthreads();
 
   for (int t = 0; t < num_kv_tiles; ++t) {
     const int kv0 = t * kKVTile;
     u16* kbuf = (t & 1) ? k_lds1 : k_lds0;
     u16* kbuf_next = (t & 1) ? k_lds0 : k_lds1;
 
-    vwrite();  // V(t) -> vt_lds (read after the mid barrier)
-    if (t + 1 < num_kv_tiles)
-      stage_k_glds(t + 1, kbuf_next);  // flies under QK^T(t)
+    vwrite();
+    if (t + 1 < num_kv_tiles) stage_k_glds(t + 1, kbuf_next);
 
     const bool compute = active && kv0 <= my_q_max;
+    floatx16 s_acc[2];  // S^T for kv blocks [kv0, +32) and [kv0+32, +64)
     if (compute) {
       const char* kbase = reinterpret_cast<const char*>(kbuf);
-      char* pbase = reinterpret_cast<char*>(p_lds);
+#pragma unroll
+      for (int b = 0; b < 2; ++b)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) s_acc[b][r] = 0.f;
+      // ---- swapped QK^T: S^T[kv 64 x q 32] ----
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int f = 0; f < KF; ++f) {
+        const short8 a_k0 = *reinterpret_cast<const short8*>(
+            kbase + swz<kKRowB>(col, f * 32 + hi1 * 16));
+        const short8 a_k1 = *reinterpret_cast<const short8*>(
+            kbase + swz<kKRowB>(32 + col, f * 32 + hi1 * 16));
+        s_acc[0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a_k0, b_q[f], s_acc[0], 0, 0, 0);
+        s_acc[1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a_k1, b_q[f], s_acc[1], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
 
+      // ---- in-register online softmax (row = col, lane-local) ----
+      const int q_pos = ctx_start + row0 + col;     // this lane's q row
+      const bool row_valid = row0 + col < seq_len;
 #pragma unroll
-      for (int rb = 0; rb < 2; ++rb) {
-        // ---- QK^T: S[16 x 64] ----
-        floatx4 s_acc[4];
-        __builtin_amdgcn_s_setprio(1);
+      for (int b = 0; b < 2; ++b)
 #pragma unroll
-        for (int cb4 = 0; cb4 < 4; ++cb4) {
-          s_acc[cb4] = {0.f, 0.f, 0.f, 0.f};
-          const int krow_idx = cb4 * 16 + col;  // kv token within tile
-#pragma unroll
-          for (int kb = 0; kb < KB; ++kb) {
-            const short8 b_k = *reinterpret_cast<const short8*>(
-                kbase + swz<kKRowB>(krow_idx, kb * 32 * 2 + hi * 16));
-            s_acc[cb4] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a_q[rb][kb], b_k, s_acc[cb4], 0, 0, 0);
-          }
+        for (int r = 0; r < 16; ++r) {
+          const int kv_pos = kv0 + 32 * b + crow(r, hi1);
+          float sv = s_acc[b][r] * scale;
+          if (!row_valid || kv_pos > q_pos || kv_pos >= k_len)
+            sv = kPNegInf;
+          s_acc[b][r] = sv;
         }
-        __builtin_amdgcn_s_setprio(0);
+      float rm = kPNegInf;
+#pragma unroll
+      for (int b = 0; b < 2; ++b)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) rm = fmaxf(rm, s_acc[b][r]);
+      rm = fmaxf(rm, __shfl_xor(rm, 32, 64));
+      const float m_new = fmaxf(m_row, rm);
+      float alpha;
+      if (m_new <= kPNegInf) {
+        alpha = 0.f;  // row has seen no unmasked score yet (o is 0)
+      } else {
+        alpha = (m_row <= kPNegInf) ? 0.f : __expf(m_row - m_new);
+        m_row = m_new;
+      }
+      float rs = 0.f;
+#pragma unroll
+      for (int b = 0; b < 2; ++b)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float pv = (m_new <= kPNegInf || s_acc[b][r] <= kPNegInf)
+                               ? 0.f
+                               : __expf(s_acc[b][r] - m_new);
+          s_acc[b][r] = pv;  // reuse as P
+          rs += pv;
+        }
+      rs += __shfl_xor(rs, 32, 64);
+      l_row = l_row * alpha + rs;
 
-        // ---- mask + online softmax over the 16x64 block (in s_acc) ----
-        const int q_base = row0 + rb * 16 + hi * 4;
+      // rescale o: alpha is per-ROW (lane-local); o_acc rows follow the
+      // crow map, so gather the right alpha per reg via shfl
+      if (__any(alpha != 1.f)) {
 #pragma unroll
-        for (int cb4 = 0; cb4 < 4; ++cb4) {
-          const int kv_pos = kv0 + cb4 * 16 + col;  // absolute position
+        for (int r = 0; r < 16; ++r) {
+          const float a_r = __shfl(alpha, crow(r, hi1), 64);
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const int q_pos = ctx_start + q_base + r;  // absolute position
-            float sv = s_acc[cb4][r] * scale;
-            if (kv_pos > q_pos || q_base + r >= seq_len || kv_pos >= k_len)
-              sv = kPNegInf;
-            s_acc[cb4][r] = sv;
-          }
-        }
-        // o_acc is rescaled HERE (phase 1) so nothing but P crosses the
-        // barrier into the PV phase
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float rm = fmaxf(fmaxf(s_acc[0][r], s_acc[1][r]),
-                           fmaxf(s_acc[2][r], s_acc[3][r]));
-#pragma unroll
-          for (int off = 1; off < 16; off <<= 1)
-            rm = fmaxf(rm, __shfl_xor(rm, off, 64));
-          const float m_new = fmaxf(m[rb][r], rm);
-          float alpha;
-          if (m_new <= kPNegInf) {  // row fully masked so far
-            alpha = 0.f;
-          } else {
-            alpha = (m[rb][r] <= kPNegInf) ? 0.f : __expf(m[rb][r] - m_new);
-            m[rb][r] = m_new;
-          }
-#pragma unroll
-          for (int cb = 0; cb < CB; ++cb) o_acc[rb][cb][r] *= alpha;
-          float rs = 0.f;
-#pragma unroll
-          for (int cb4 = 0; cb4 < 4; ++cb4) {
-            const float pv = (m_new <= kPNegInf || s_acc[cb4][r] <= kPNegInf)
-                                 ? 0.f
-                                 : __expf(s_acc[cb4][r] - m_new);
-            rs += pv;
-            *reinterpret_cast<u16*>(
-                pbase + swz<kVTRowB>(rb * 16 + hi * 4 + r,
-                                     (cb4 * 16 + col) * 2)) = f32_to_bf16(pv);
-          }
-#pragma unroll
-          for (int off = 1; off < 16; off <<= 1) rs += __shfl_xor(rs, off, 64);
-          l[rb][r] = l[rb][r] * alpha + rs;
+          for (int cb = 0; cb < CB; ++cb) o_acc[cb][r] *= a_r;
         }
       }
     }
 
-    // publish vt_lds(t) to every wave; also drains the K(t+1) glds, whose
-    // latency QK^T just covered
+    // publish vt_lds(t); drains the K(t+1) glds QK^T just covered
     __syncthreads();
-    if (t + 1 < num_kv_tiles)
-      vload(t + 1);  // flies under PV(t); consumed by vwrite next iteration
+    if (t + 1 < num_kv_tiles) vload(t + 1);
 
     if (compute) {
       const char* vbase = reinterpret_cast<const char*>(vt_lds);
-      char* pbase = reinterpret_cast<char*>(p_lds);
+      // ---- P -> A-fragments in-register (pack pairs + half swap) ----
+      // A-frag for kv chunk [32b + 16c, +16): lane needs kv {hi1*8..+8} of
+      // that range; own regs hold kv {..}+4*hi1 interleaved, the partner
+      // half-wave holds the other 4-run — one permlane32_swap per pair
+      // merges them (guide T12).
+      short8 a_p[4];
+#pragma unroll
+      for (int b = 0; b < 2; ++b)
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          const int base = 8 * c;
+          const u32 u1 = pack_bf16x2(s_acc[b][base + 0], s_acc[b][base + 1]);
+          const u32 u2 = pack_bf16x2(s_acc[b][base + 2], s_acc[b][base + 3]);
+          const u32 v1 = pack_bf16x2(s_acc[b][base + 4], s_acc[b][base + 5]);
+          const u32 v2 = pack_bf16x2(s_acc[b][base + 6], s_acc[b][base + 7]);
+          const auto s1 = __builtin_amdgcn_permlane32_swap(
+              static_cast<int>(u1), static_cast<int>(v1), false, false);
+          const auto s2 = __builtin_amdgcn_permlane32_swap(
+              static_cast<int>(u2), static_cast<int>(v2), false, false);
+          typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
+          u32x4 w;
+          w.x = static_cast<u32>(s1[0]);
+          w.y = static_cast<u32>(s2[0]);
+          w.z = static_cast<u32>(s1[1]);
+          w.w = static_cast<u32>(s2[1]);
+          a_p[2 * b + c] = __builtin_bit_cast(short8, w);
+        }
+
+      // ---- PV: O[32 q x D] += P[32 x 64] V[64 x D] ----
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int rb = 0; rb < 2; ++rb) {
-        // ---- PV: O[16 x D] += P[16 x 64] V[64 x D] ----
+      for (int kc = 0; kc < 4; ++kc) {   // 4 x 16-deep kv chunks
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {  // kv 64 = 2 MFMA k-depths
-          const short8 a_p = *reinterpret_cast<const short8*>(
-              pbase + swz<kVTRowB>(rb * 16 + col, kc * 64 + hi * 16));
-#pragma unroll
-          for (int cb = 0; cb < CB; ++cb) {
-            const short8 b_v = *reinterpret_cast<const short8*>(
-                vbase + swz<kVTRowB>(cb * 16 + col, kc * 64 + hi * 16));
-            o_acc[rb][cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a_p, b_v, o_acc[rb][cb], 0, 0, 0);
-          }
+        for (int cb = 0; cb < CB; ++cb) {
+          const short8 b_v = *reinterpret_cast<const short8*>(
+              vbase + swz<kVTRowB>(cb * 32 + col, kc * 32 + hi1 * 16));
+          o_acc[cb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_p[kc], b_v, o_acc[cb], 0, 0, 0);
         }
       }
       __builtin_amdgcn_s_setprio(0);
@@ -307,21 +317,19 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     __syncthreads();
   }
 
-  // ---- epilogue: normalize and store ----
+  // ---- epilogue: normalize (l gathered per crow row) and store ----
   if (active) {
+    const float inv_own = (l_row > 0.f) ? 1.f / l_row : 0.f;
 #pragma unroll
-    for (int rb = 0; rb < 2; ++rb) {
+    for (int r = 0; r < 16; ++r) {
+      const int qr = row0 + crow(r, hi1);
+      if (qr >= seq_len) continue;
+      const float inv = __shfl(inv_own, crow(r, hi1), 64);
+      u16* orow = out + (static_cast<int64_t>(seq_start + qr) * num_q_heads +
+                         head) * D;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int q_pos = row0 + rb * 16 + hi * 4 + r;
-        if (q_pos >= seq_len) continue;
-        const float inv = (l[rb][r] > 0.f) ? 1.f / l[rb][r] : 0.f;
-        u16* orow = out + (static_cast<int64_t>(seq_start + q_pos) *
-                               num_q_heads + head) * D;
-#pragma unroll
-        for (int cb = 0; cb < CB; ++cb)
-          orow[cb * 16 + col] = f32_to_bf16(o_acc[rb][cb][r] * inv);
-      }
+      for (int cb = 0; cb < CB; ++cb)
+        orow[cb * 32 + col] = f32_to_bf16(o_acc[cb][r] * inv);
     }
   }
 }
