@@ -89,25 +89,77 @@ def build_engine_config(args):
     )
 
 
-def main(argv=None):
-    args = parse_args(argv)
-    if args.master_addr:
-        os.environ.setdefault("MASTER_ADDR", args.master_addr)
-        os.environ.setdefault("MASTER_PORT", str(args.master_port))
+def _rank_main(local_rank: int, args, nproc: int):
+    """One engine rank. Global rank 0 serves HTTP; others execute the TP
+    driver's broadcast batches (engine.worker_loop)."""
+    import torch
 
-    import uvicorn
+    rank = args.node_rank * nproc + local_rank
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(local_rank)
+    os.environ["WORLD_SIZE"] = str(args.nnodes * nproc)
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+    else:
+        device = "cpu"
 
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
     from fusioninfer_amd.server.api_server import build_app
     from fusioninfer_amd.server.serving import ServingEngine
 
     cfg = build_engine_config(args)
-    serving = ServingEngine(cfg)
+    if rank != 0:
+        engine = LLMEngine(cfg, device=device)
+        engine.worker_loop()
+        return
+    serving = ServingEngine(cfg, device=device)
     if args.enable_lora or args.lora_modules:
         for spec in args.lora_modules:
-            name, _, rank = spec.partition("=")
-            serving.engine.add_lora(name, rank=int(rank) if rank else 16)
+            name, _, r = spec.partition("=")
+            serving.engine.add_lora(name, rank=int(r) if r else 16)
+
+    import uvicorn
+
     app = build_app(serving, args.model)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+    serving.engine.stop_workers()
+
+
+def main(argv=None):
+    """Single- or multi-rank launch. For multi-node roles the control
+    plane's LWS wrapper appends --nnodes/--node-rank/--nproc-per-node/
+    --master-addr (from LWS_LEADER_ADDRESS) — this entrypoint then spawns
+    one engine process per local GPU and forms the RCCL process group
+    itself (the reference delegated this to Ray)."""
+    args = parse_args(argv)
+    if args.master_addr:
+        os.environ["MASTER_ADDR"] = args.master_addr
+        os.environ["MASTER_PORT"] = str(args.master_port)
+    else:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", str(args.master_port))
+
+    nproc = max(args.nproc_per_node, 1)
+    world = args.nnodes * nproc
+    if world <= 1:
+        _rank_main(0, args, nproc)
+        return
+    assert args.tensor_parallel_size == world, (
+        "multi-rank launch currently maps every rank into one TP group: "
+        f"--tensor-parallel-size {args.tensor_parallel_size} != {world}"
+    )
+    import torch.multiprocessing as mp
+
+    procs = []
+    ctx = mp.get_context("spawn")
+    for lr in range(1, nproc):
+        p = ctx.Process(target=_rank_main, args=(lr, args, nproc), daemon=True)
+        p.start()
+        procs.append(p)
+    _rank_main(0, args, nproc)
+    for p in procs:
+        p.join(timeout=10)
 
 
 if __name__ == "__main__":
