@@ -1,4 +1,4 @@
-"""Token sampling: greedy / temperature / top-p over the logits rows."""
+"""Token sampling: penalties, temperature, top-k/top-p, per-request seeds."""
 
 from __future__ import annotations
 
@@ -11,39 +11,90 @@ from fusioninfer_amd.engine.sequence import Sequence
 
 class Sampler:
     def __init__(self, seed: int = 0, device: str = "cpu"):
+        self.device = device
+        self.base_seed = seed
         self.generator = None
         if device != "cpu":
             self.generator = torch.Generator(device=device)
             self.generator.manual_seed(seed)
 
+    def _apply_penalties(self, logits: torch.Tensor, seqs: List[Sequence]):
+        """repetition / presence / frequency penalties on already-emitted
+        (prompt + output) tokens, OpenAI/vLLM semantics."""
+        for i, s in enumerate(seqs):
+            sp = s.sampling
+            if (
+                sp.repetition_penalty == 1.0
+                and sp.presence_penalty == 0.0
+                and sp.frequency_penalty == 0.0
+            ):
+                continue
+            token_ids = torch.tensor(
+                s.all_token_ids, dtype=torch.long, device=logits.device
+            )
+            uniq, counts = token_ids.unique(return_counts=True)
+            row = logits[i]
+            if sp.repetition_penalty != 1.0:
+                vals = row[uniq]
+                row[uniq] = torch.where(
+                    vals > 0,
+                    vals / sp.repetition_penalty,
+                    vals * sp.repetition_penalty,
+                )
+            if sp.presence_penalty != 0.0:
+                row[uniq] -= sp.presence_penalty
+            if sp.frequency_penalty != 0.0:
+                row[uniq] -= sp.frequency_penalty * counts.to(row.dtype)
+        return logits
+
     def sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> torch.Tensor:
-        """logits: [S, V] fp32/bf16; returns [S] int64 token ids (on device)."""
+        """logits: [S, V] fp32; returns [S] int64 token ids."""
+        logits = self._apply_penalties(logits, seqs)
         temps = torch.tensor(
             [s.sampling.temperature for s in seqs],
             dtype=torch.float32,
             device=logits.device,
         )
-        if bool((temps == 0).all()):
-            return logits.argmax(dim=-1)
-        logits = logits.float() / temps.clamp(min=1e-5).unsqueeze(1)
-        top_p = torch.tensor(
-            [s.sampling.top_p for s in seqs], dtype=torch.float32,
-            device=logits.device,
-        )
-        probs = torch.softmax(logits, dim=-1)
-        if bool((top_p < 1.0).any()):
-            sorted_probs, idx = probs.sort(dim=-1, descending=True)
-            cum = sorted_probs.cumsum(dim=-1)
-            mask = cum - sorted_probs > top_p.unsqueeze(1)
-            sorted_probs[mask] = 0.0
-            sorted_probs /= sorted_probs.sum(dim=-1, keepdim=True)
-            choice = torch.multinomial(
-                sorted_probs, 1, generator=self.generator
-            ).squeeze(1)
-            sampled = idx.gather(1, choice.unsqueeze(1)).squeeze(1)
-        else:
-            sampled = torch.multinomial(
-                probs, 1, generator=self.generator
-            ).squeeze(1)
         greedy = logits.argmax(dim=-1)
-        return torch.where(temps == 0, greedy, sampled)
+        if bool((temps == 0).all()):
+            return greedy
+        scaled = logits / temps.clamp(min=1e-5).unsqueeze(1)
+
+        # top-k then top-p filtering
+        for i, s in enumerate(seqs):
+            sp = s.sampling
+            if sp.temperature == 0:
+                continue
+            row = scaled[i]
+            if sp.top_k and sp.top_k > 0:
+                kth = torch.topk(row, min(sp.top_k, row.numel())).values[-1]
+                row[row < kth] = float("-inf")
+            if sp.top_p < 1.0:
+                sorted_logits, idx = row.sort(descending=True)
+                probs = torch.softmax(sorted_logits, dim=-1)
+                cum = probs.cumsum(dim=-1)
+                cut = cum - probs > sp.top_p
+                sorted_logits[cut] = float("-inf")
+                row.scatter_(0, idx, sorted_logits)
+
+        probs = torch.softmax(scaled, dim=-1)
+        sampled = torch.empty_like(greedy)
+        # group rows by generator: per-request seeds get their own draw
+        default_rows = []
+        for i, s in enumerate(seqs):
+            sp = s.sampling
+            if sp.temperature == 0:
+                sampled[i] = greedy[i]
+            elif sp.seed is not None:
+                g = torch.Generator(device=logits.device)
+                g.manual_seed(sp.seed + len(s.output_token_ids))
+                sampled[i] = torch.multinomial(probs[i], 1, generator=g)[0]
+            else:
+                default_rows.append(i)
+        if default_rows:
+            rows = torch.tensor(default_rows, device=logits.device)
+            draw = torch.multinomial(
+                probs[rows], 1, generator=self.generator
+            ).squeeze(1)
+            sampled[rows] = draw
+        return sampled

@@ -20,12 +20,22 @@ class SamplingParams:
         max_tokens: int = 128,
         temperature: float = 0.0,
         top_p: float = 1.0,
+        top_k: int = 0,
+        repetition_penalty: float = 1.0,
+        presence_penalty: float = 0.0,
+        frequency_penalty: float = 0.0,
+        seed: Optional[int] = None,
         ignore_eos: bool = True,
         stop_token_ids: Optional[List[int]] = None,
     ):
         self.max_tokens = max_tokens
         self.temperature = temperature
         self.top_p = top_p
+        self.top_k = top_k
+        self.repetition_penalty = repetition_penalty
+        self.presence_penalty = presence_penalty
+        self.frequency_penalty = frequency_penalty
+        self.seed = seed
         self.ignore_eos = ignore_eos
         self.stop_token_ids = stop_token_ids or []
 

@@ -97,21 +97,39 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   const int tok = lane / 4;          // phase-A token within chunk
   const int quad = lane % 4;         // phase-A dim quarter
   constexpr int DPQ = D / 4;         // dims per phase-A lane
+  constexpr int KQ4 = DPQ / 8;       // 16-B K loads per lane per chunk
+
+  // software-prefetch: this wave's NEXT chunk's K flies while the current
+  // chunk's softmax + PV run (the phases were serialized on K latency)
+  auto kv_base_of = [&](int chunk) -> int64_t {
+    const int block_id = block_tables[seq * max_blocks + chunk];
+    return ((static_cast<int64_t>(block_id) * num_kv_heads + kv_head) *
+            kBlockSz) * D;
+  };
+  uint4 kraw[KQ4];
+  auto load_k = [&](int64_t kv_base, uint4* dst) {
+    const u16* k_row = k_cache + kv_base + tok * D + quad * DPQ;
+#pragma unroll
+    for (int j8 = 0; j8 < KQ4; ++j8)
+      dst[j8] = *reinterpret_cast<const uint4*>(k_row + j8 * 8);
+  };
+  int64_t kv_base = 0;
+  if (chunk_lo + wave < chunk_hi) {
+    kv_base = kv_base_of(chunk_lo + wave);
+    load_k(kv_base, kraw);
+  }
 
   for (int chunk = chunk_lo + wave; chunk < chunk_hi; chunk += kNWaves) {
-    const int block_id = block_tables[seq * max_blocks + chunk];
-    const int64_t kv_base =
-        ((static_cast<int64_t>(block_id) * num_kv_heads + kv_head) * kBlockSz) * D;
     const int token_pos = chunk * kBlockSz + tok;
+    const int64_t kv_base_cur = kv_base;
 
     // ---- phase A: scores for 16 tokens x G heads ----
     float s[G];
     {
-      const u16* k_row = k_cache + kv_base + tok * D + quad * DPQ;
       float kf[DPQ];
 #pragma unroll
-      for (int j8 = 0; j8 < DPQ / 8; ++j8) {
-        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k_row + j8 * 8);
+      for (int j8 = 0; j8 < KQ4; ++j8) {
+        const bf16x8 kv8 = __builtin_bit_cast(bf16x8, kraw[j8]);
 #pragma unroll
         for (int j = 0; j < 8; ++j) kf[j8 * 8 + j] = bf16_to_f32(kv8.h[j]);
       }
@@ -130,6 +148,11 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
         }
         s[g] = acc_s;
       }
+    }
+    // issue next chunk's K now; it lands under softmax + PV
+    if (chunk + kNWaves < chunk_hi) {
+      kv_base = kv_base_of(chunk + kNWaves);
+      load_k(kv_base, kraw);
     }
     // reduce over the 4 dim-quarters (lanes 4t..4t+3)
 #pragma unroll
@@ -166,7 +189,7 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
 
     // ---- phase B: PV accumulate; lane covers dims {2l, 2l+1} ----
     const int nvalid = min(ctx - chunk * kBlockSz, kBlockSz);
-    const u16* v_rows = v_cache + kv_base;
+    const u16* v_rows = v_cache + kv_base_cur;
 #pragma unroll
     for (int g = 0; g < G; ++g) {
       acc[g][0] *= alpha[g];

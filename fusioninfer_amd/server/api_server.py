@@ -95,6 +95,11 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             max_tokens=int(body.get("max_tokens", 16)),
             temperature=float(body.get("temperature", 1.0)),
             top_p=float(body.get("top_p", 1.0)),
+            top_k=int(body.get("top_k", 0)),
+            repetition_penalty=float(body.get("repetition_penalty", 1.0)),
+            presence_penalty=float(body.get("presence_penalty", 0.0)),
+            frequency_penalty=float(body.get("frequency_penalty", 0.0)),
+            seed=body.get("seed"),
             ignore_eos=bool(body.get("ignore_eos", False)),
             stop_token_ids=body.get("stop_token_ids") or [],
         )
