@@ -27,6 +27,7 @@ import statistics
 import time
 
 import torch
+import torch.distributed as dist
 
 
 def parse_args():
@@ -116,8 +117,6 @@ def main():
     device = f"cuda:{local_rank}" if use_cuda else "cpu"
     if use_cuda:
         torch.cuda.set_device(local_rank)
-
-    import torch.distributed as dist
 
     distributed = world > 1
     if distributed:
