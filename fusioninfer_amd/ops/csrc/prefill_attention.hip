@@ -72,7 +72,10 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
   constexpr int KF = D / 16;           // QK^T 16-deep k-chunks over head dim
   constexpr int CB = D / 32;           // PV output 32-col blocks
   constexpr int kKRowB = D * 2;        // K row bytes (D=128 -> 256)
-  constexpr int kVTRowB = kKVTile * 2; // V^T row bytes (64 kv -> 128)
+  // V^T rows PADDED to 256 B (data is 128 B): lets the full (row&15)<<4
+  // swizzle fit in-row, making the PV B-fragment reads conflict-free
+  // (the 128-B row's (row&7) swizzle measured 2.6% bank-conflict cycles)
+  constexpr int kVTRowB = 256;
 
   const int head = blockIdx.y;
   const int kv_head = head / (num_q_heads / num_kv_heads);
@@ -103,7 +106,7 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
 
   // LDS: double-buffered K (glds target) + V^T tile. ONE __shared__ object
   // (guide §5 ".s-level traps" (a)).
-  __shared__ u16 smem[2 * kKVTile * D + D * kKVTile];
+  __shared__ u16 smem[2 * kKVTile * D + D * (kVTRowB / 2)];
   u16* k_lds0 = smem;
   u16* k_lds1 = smem + kKVTile * D;
   u16* vt_lds = smem + 2 * kKVTile * D;
@@ -177,6 +180,21 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     }
   };
 
+  // hoist the swizzled LDS read offsets (constant per lane) out of the
+  // tile loop — the swz() address math was otherwise recomputed per tile
+  int k_off[2][KF];
+  int v_off[4][CB];
+#pragma unroll
+  for (int f = 0; f < KF; ++f) {
+    k_off[0][f] = swz<kKRowB>(col, f * 32 + hi1 * 16);
+    k_off[1][f] = swz<kKRowB>(32 + col, f * 32 + hi1 * 16);
+  }
+#pragma unroll
+  for (int kc = 0; kc < 4; ++kc)
+#pragma unroll
+    for (int cb = 0; cb < CB; ++cb)
+      v_off[kc][cb] = swz<kVTRowB>(cb * 32 + col, kc * 32 + hi1 * 16);
+
   stage_k_glds(0, k_lds0);
   vload(0);
   __syncthreads();
@@ -201,10 +219,10 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int f = 0; f < KF; ++f) {
-        const short8 a_k0 = *reinterpret_cast<const short8*>(
-            kbase + swz<kKRowB>(col, f * 32 + hi1 * 16));
-        const short8 a_k1 = *reinterpret_cast<const short8*>(
-            kbase + swz<kKRowB>(32 + col, f * 32 + hi1 * 16));
+        const short8 a_k0 =
+            *reinterpret_cast<const short8*>(kbase + k_off[0][f]);
+        const short8 a_k1 =
+            *reinterpret_cast<const short8*>(kbase + k_off[1][f]);
         s_acc[0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             a_k0, b_q[f], s_acc[0], 0, 0, 0);
         s_acc[1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
@@ -215,16 +233,29 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
       // ---- in-register online softmax (row = col, lane-local) ----
       const int q_pos = ctx_start + row0 + col;     // this lane's q row
       const bool row_valid = row0 + col < seq_len;
+      // interior tiles (all kv strictly before every valid q row, no
+      // ragged tail, no invalid rows) skip the per-element masking
+      const bool need_mask =
+          (kv0 + kKVTile - 1 > ctx_start + row0)
+          || (kv0 + kKVTile > k_len)
+          || (row0 + kQPerWave > seq_len);
+      if (need_mask) {
 #pragma unroll
-      for (int b = 0; b < 2; ++b)
+        for (int b = 0; b < 2; ++b)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int kv_pos = kv0 + 32 * b + crow(r, hi1);
-          float sv = s_acc[b][r] * scale;
-          if (!row_valid || kv_pos > q_pos || kv_pos >= k_len)
-            sv = kPNegInf;
-          s_acc[b][r] = sv;
-        }
+          for (int r = 0; r < 16; ++r) {
+            const int kv_pos = kv0 + 32 * b + crow(r, hi1);
+            float sv = s_acc[b][r] * scale;
+            if (!row_valid || kv_pos > q_pos || kv_pos >= k_len)
+              sv = kPNegInf;
+            s_acc[b][r] = sv;
+          }
+      } else {
+#pragma unroll
+        for (int b = 0; b < 2; ++b)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) s_acc[b][r] *= scale;
+      }
       float rm = kPNegInf;
 #pragma unroll
       for (int b = 0; b < 2; ++b)
@@ -305,8 +336,8 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
       for (int kc = 0; kc < 4; ++kc) {   // 4 x 16-deep kv chunks
 #pragma unroll
         for (int cb = 0; cb < CB; ++cb) {
-          const short8 b_v = *reinterpret_cast<const short8*>(
-              vbase + swz<kVTRowB>(cb * 32 + col, kc * 32 + hi1 * 16));
+          const short8 b_v =
+              *reinterpret_cast<const short8*>(vbase + v_off[kc][cb]);
           o_acc[cb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a_p[kc], b_v, o_acc[cb], 0, 0, 0);
         }
