@@ -52,6 +52,10 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
 
     @app.get("/health")
     async def health():
+        if not serving.healthy:
+            return JSONResponse(
+                {"status": "unhealthy", "error": serving.last_error}, 503
+            )
         return {"status": "ok"}
 
     @app.get("/v1/models")

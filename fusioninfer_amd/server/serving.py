@@ -23,6 +23,8 @@ class ServingEngine:
         self._lock = threading.Lock()
         self._work = threading.Event()
         self._stop = False
+        self.healthy = True
+        self.last_error: str = ""
         self._thread = threading.Thread(target=self._loop, daemon=True)
         self._thread.start()
 
@@ -76,8 +78,17 @@ class ServingEngine:
                 self._work.clear()
                 self._work.wait(timeout=0.25)
                 continue
-            with self._lock:
-                outputs = self.engine.step()
+            try:
+                with self._lock:
+                    outputs = self.engine.step()
+            except Exception as e:  # engine fault: fail requests, go unhealthy
+                self.healthy = False
+                self.last_error = repr(e)
+                with self._lock:
+                    for q in self._streams.values():
+                        q.put((None, True))
+                    self._streams.clear()
+                return
             for out in outputs:
                 q = self._streams.get(out.request_id)
                 if q is None:
