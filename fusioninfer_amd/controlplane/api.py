@@ -168,3 +168,32 @@ class InferenceService:
 
     def spec_dict(self) -> Dict[str, Any]:
         return {"roles": [r.to_dict() for r in self.roles]}
+
+
+@dataclasses.dataclass
+class ModelLoader:
+    """Scaffolded CRD mirroring the reference's stub ModelLoader
+    (reference api/core/v1alpha1/modelloader_types.go:26-59 — a `Foo`
+    placeholder spec shipped unimplemented; its controller is a no-op,
+    modelloader_controller.go:49-55). Real weight materialization in this
+    build lives engine-side (fusioninfer_amd/models/weight_loader.py)."""
+
+    name: str
+    namespace: str = "default"
+    foo: Optional[str] = None
+
+    def to_dict(self) -> Dict[str, Any]:
+        return {
+            "apiVersion": API_VERSION,
+            "kind": "ModelLoader",
+            "metadata": {"name": self.name, "namespace": self.namespace},
+            "spec": {"foo": self.foo} if self.foo is not None else {},
+        }
+
+    @staticmethod
+    def from_dict(d: Dict[str, Any]) -> "ModelLoader":
+        return ModelLoader(
+            name=d["metadata"]["name"],
+            namespace=d["metadata"].get("namespace", "default"),
+            foo=d.get("spec", {}).get("foo"),
+        )

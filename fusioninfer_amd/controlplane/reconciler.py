@@ -231,3 +231,15 @@ class InferenceServiceReconciler:
                 "blockOwnerDeletion": True,
             }
         )
+
+
+class ModelLoaderReconciler:
+    """No-op reconcile loop, mirroring the reference's stub controller
+    (pkg/controller/modelloader_controller.go:49-55)."""
+
+    def __init__(self, client: FakeClient):
+        self.client = client
+
+    def reconcile(self, name: str, namespace: str = "default") -> None:
+        self.client.try_get("ModelLoader", name, namespace)
+        return None

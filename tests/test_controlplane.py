@@ -476,3 +476,17 @@ class TestCLIAndCRD:
             assert "nvidia.com/gpu" not in out
             if "pd" in sample or "tp8" in sample:
                 assert "PodGroup" in kinds
+
+
+class TestModelLoaderStub:
+    def test_roundtrip_and_noop_reconcile(self):
+        from fusioninfer_amd.controlplane.api import ModelLoader
+        from fusioninfer_amd.controlplane.reconciler import ModelLoaderReconciler
+
+        client = FakeClient()
+        ml = ModelLoader("demo", foo="bar")
+        client.create(ml.to_dict())
+        rec = ModelLoaderReconciler(client)
+        assert rec.reconcile("demo") is None
+        back = ModelLoader.from_dict(client.get("ModelLoader", "demo"))
+        assert back.foo == "bar"
