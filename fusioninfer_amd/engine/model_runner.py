@@ -1,10 +1,12 @@
 """Model runner: batch preparation + forward execution + hipGraph decode.
 
-Decode steps have static shapes (pure-decode batches, fixed-width block
-tables), so they are captured once per batch-size bucket into hipGraphs
-(torch.cuda.CUDAGraph == hipGraph on ROCm) and replayed — removing
-launch overhead from the latency-critical decode loop (guide: "capture
-launch-bound inner loops in hipGraphs"). Prefill runs eager.
+Pure-decode steps have static shapes (fixed-width block tables), so they
+are captured once per batch-size bucket into hipGraphs
+(torch.cuda.CUDAGraph == hipGraph on ROCm) and replayed — removing launch
+overhead from the latency-critical decode loop (guide: "capture
+launch-bound inner loops in hipGraphs"). Prefill and mixed steps run
+eager; batches with live LoRA adapters also run eager. Payloads are plain
+host lists so the TP driver can broadcast them to worker ranks.
 """
 
 from __future__ import annotations

@@ -1,11 +1,13 @@
-"""Continuous-batching scheduler (token-level, vLLM-class policy).
+"""Continuous-batching scheduler (token-level, unified mixed batches).
 
-Step policy: prefill-priority. When prompts are waiting and blocks/budget
-allow, the step is a pure-prefill batch (bounded by max_num_batched_tokens
-and max_num_seqs); otherwise it is a pure-decode batch over all RUNNING
-sequences. Pure-decode steps have static shapes so the runner replays
-hipGraphs for them. Preemption-by-recompute frees the newest sequence
-when the cache runs out of blocks mid-decode.
+Each step schedules ONE mixed batch: a decode token for every
+fully-prefilled running sequence PLUS prompt chunks (chunked prefill)
+under a shared max_num_batched_tokens budget — decode-ready sequences
+never stall behind prefills. New-prompt admission uses hysteresis
+(prefill_admission_tokens): prompts are held until enough tokens queue
+up, so steady-state steps stay pure-decode and take the hipGraph path.
+Preemption-by-recompute frees the newest sequence when the cache runs
+out of blocks mid-decode.
 
 Capability parity: the "continuous-batching scheduler + paged KV cache"
 the reference delegates to vLLM and whose kv-util / queue-depth metrics
