@@ -27,6 +27,7 @@ class RequestOutput:
         self.prompt_token_ids = seq.prompt_token_ids
         self.output_token_ids = list(seq.output_token_ids)
         self.finished = seq.is_finished()
+        self.logprobs = list(seq.logprobs)
         self.ttft = seq.ttft
         self.arrival_time = seq.arrival_time
         self.finish_time = seq.finish_time
@@ -229,8 +230,28 @@ class LLMEngine:
         ] + batch.decode_seqs
         if not sample_seqs:
             return []
-        next_tokens = self.sampler.sample(logits.float(), sample_seqs)
+        logits_f = logits.float()
+        next_tokens = self.sampler.sample(logits_f, sample_seqs)
         next_tokens = next_tokens.tolist()
+        # optional logprobs for requests that asked
+        lp_rows = [
+            i for i, s in enumerate(sample_seqs)
+            if s.sampling.logprobs is not None
+        ]
+        if lp_rows:
+            lp = torch.log_softmax(logits_f[lp_rows], dim=-1)
+            for j, i in enumerate(lp_rows):
+                s = sample_seqs[i]
+                k = max(int(s.sampling.logprobs), 0)
+                entry_top = {}
+                if k > 0:
+                    vals, idx = lp[j].topk(k)
+                    entry_top = {
+                        int(t): float(v) for t, v in zip(idx, vals)
+                    }
+                s.logprobs.append(
+                    (float(lp[j, next_tokens[i]]), entry_top)
+                )
 
         outputs: List[RequestOutput] = []
         for seq, tok in zip(sample_seqs, next_tokens):

@@ -25,6 +25,7 @@ class SamplingParams:
         presence_penalty: float = 0.0,
         frequency_penalty: float = 0.0,
         seed: Optional[int] = None,
+        logprobs: Optional[int] = None,
         ignore_eos: bool = True,
         stop_token_ids: Optional[List[int]] = None,
     ):
@@ -36,6 +37,7 @@ class SamplingParams:
         self.presence_penalty = presence_penalty
         self.frequency_penalty = frequency_penalty
         self.seed = seed
+        self.logprobs = logprobs
         self.ignore_eos = ignore_eos
         self.stop_token_ids = stop_token_ids or []
 
@@ -63,6 +65,9 @@ class Sequence:
         self.hold_blocks = False
         # LoRA adapter name (None = base model)
         self.lora_name = None
+        # per-output-token logprob entries when sampling.logprobs is set:
+        # [(logprob_of_sampled, {token_id: logprob, ...top-k}), ...]
+        self.logprobs = []
 
     @property
     def num_prompt_tokens(self) -> int:

@@ -104,6 +104,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             presence_penalty=float(body.get("presence_penalty", 0.0)),
             frequency_penalty=float(body.get("frequency_penalty", 0.0)),
             seed=body.get("seed"),
+            logprobs=body.get("logprobs"),
             ignore_eos=bool(body.get("ignore_eos", False)),
             stop_token_ids=body.get("stop_token_ids") or [],
         )

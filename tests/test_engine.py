@@ -204,3 +204,23 @@ def test_abort_request():
     while eng.has_unfinished():
         eng.step()
     assert eng.num_finished == 1  # only b completed
+
+
+def test_logprobs():
+    torch.manual_seed(0)
+    eng = make_engine()
+    rid = eng.add_request(
+        [3, 1, 4] * 8, SamplingParams(max_tokens=4, logprobs=3)
+    )
+    outs = {}
+    while eng.has_unfinished():
+        for o in eng.step():
+            if o.finished:
+                outs[o.request_id] = o
+    out = outs[rid]
+    assert len(out.logprobs) == 4
+    for sampled_lp, top in out.logprobs:
+        assert sampled_lp <= 0.0
+        assert len(top) == 3
+        # greedy sampling: the sampled token's logprob equals the max
+        assert abs(max(top.values()) - sampled_lp) < 1e-5
