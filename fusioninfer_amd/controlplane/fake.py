@@ -10,7 +10,6 @@ the survey calls out as a gap to close).
 from __future__ import annotations
 
 import copy
-import fnmatch
 import itertools
 from typing import Any, Dict, List, Optional, Tuple
 

@@ -23,7 +23,7 @@ from fusioninfer_amd.controlplane.api import (
     InferenceService,
     Role,
 )
-from fusioninfer_amd.controlplane.fake import FakeClient, NotFoundError
+from fusioninfer_amd.controlplane.fake import FakeClient
 from fusioninfer_amd.controlplane.workload import (
     LABEL_ROLE,
     LABEL_SERVICE,

@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import copy
 import os
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict
 
 import yaml
 
@@ -23,7 +23,6 @@ from fusioninfer_amd.controlplane.api import (
     QUEUE_SIZE,
     DECODER,
     PREFILLER,
-    WORKER,
     InferenceService,
     Role,
 )

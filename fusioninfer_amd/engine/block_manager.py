@@ -5,15 +5,15 @@ caching mirrors the reuse the reference's EPP prefix-cache scorer assumes
 exists engine-side (reference pkg/router/strategy.go:51-68): full prompt
 blocks are content-hashed (chained) and reusable across sequences.
 
-NOTE v1: block reuse is wired, but the attention path does not yet read
-cached context during prefill, so the engine only activates reuse when
-`enable_prefix_caching` is on AND the runner supports cached context.
+Cache hits are consumed by the context-attention prefill path: the runner
+skips cached tokens entirely and the paged prefill kernel attends the new
+tokens over the cached blocks (ops/csrc/prefill_attention.hip, PAGED).
 """
 
 from __future__ import annotations
 
 from collections import OrderedDict
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from fusioninfer_amd.engine.sequence import Sequence
 

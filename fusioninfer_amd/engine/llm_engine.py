@@ -17,7 +17,7 @@ from fusioninfer_amd.config import EngineConfig
 from fusioninfer_amd.engine.block_manager import BlockManager
 from fusioninfer_amd.engine.model_runner import ModelRunner
 from fusioninfer_amd.engine.sampler import Sampler
-from fusioninfer_amd.engine.scheduler import ScheduledBatch, Scheduler
+from fusioninfer_amd.engine.scheduler import Scheduler
 from fusioninfer_amd.engine.sequence import SamplingParams, Sequence, SeqStatus
 
 

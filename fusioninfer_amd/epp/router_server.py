@@ -15,7 +15,6 @@ from __future__ import annotations
 
 import asyncio
 import re
-import time
 from typing import Dict, List, Optional
 
 import httpx

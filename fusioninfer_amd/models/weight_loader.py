@@ -12,7 +12,6 @@ be able to load real weights. This maps HuggingFace parameter names
 
 from __future__ import annotations
 
-import json
 import os
 from typing import Dict, Iterable, Tuple
 

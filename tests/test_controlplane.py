@@ -7,8 +7,6 @@ httproute_test / hash_test); tier 2 mirrors its envtest integration suite
 no-op invariant and the readiness aggregation the reference left untested.
 """
 
-import copy
-
 import yaml
 
 from fusioninfer_amd.controlplane import api, router, scheduling, workload

@@ -10,7 +10,6 @@ construction (8-GPU runs are driver-side only):
 
 import os
 
-import pytest
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp

@@ -10,7 +10,6 @@ import torch
 from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
 from fusioninfer_amd.engine.block_manager import BlockManager
 from fusioninfer_amd.engine.llm_engine import LLMEngine
-from fusioninfer_amd.engine.metadata import AttnMetadata
 from fusioninfer_amd.engine.sequence import SamplingParams, Sequence
 from fusioninfer_amd.models.registry import get_model_config
 

@@ -2,7 +2,6 @@
 
 import torch
 
-from fusioninfer_amd.engine.llm_engine import LLMEngine
 from fusioninfer_amd.engine.sequence import SamplingParams
 from tests.test_engine import make_engine
 

@@ -19,9 +19,7 @@ def _model(seed):
 
 
 def _forward_logits(model, tokens):
-    from fusioninfer_amd.engine.block_manager import BlockManager
     from fusioninfer_amd.engine.metadata import AttnMetadata
-    import fusioninfer_amd.ops as ops
 
     T = len(tokens)
     nblk = (T + 15) // 16
