@@ -22,7 +22,7 @@ from fusioninfer_amd.engine.sequence import Sequence
 from fusioninfer_amd.models.model import CausalLM
 import fusioninfer_amd.ops as ops_mod
 
-_DECODE_BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256]
+_DECODE_BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 160, 192, 224, 256, 320, 384, 448, 512]
 
 
 class ModelRunner:
