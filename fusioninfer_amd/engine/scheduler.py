@@ -18,6 +18,7 @@ strategy.go:70-98).
 from __future__ import annotations
 
 import dataclasses
+import time
 from collections import deque
 from typing import Deque, List
 
@@ -113,6 +114,12 @@ class Scheduler:
             admit = True
         else:
             admit = waiting_tokens >= threshold
+            # aging escape: at low load the batching threshold may never be
+            # reached — never hold a prompt longer than 50 ms
+            if not admit and self.waiting:
+                admit = (
+                    time.monotonic() - self.waiting[0].arrival_time > 0.05
+                )
         while (
             admit
             and budget > 0

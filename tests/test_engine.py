@@ -223,3 +223,31 @@ def test_logprobs():
         assert len(top) == 3
         # greedy sampling: the sampled token's logprob equals the max
         assert abs(max(top.values()) - sampled_lp) < 1e-5
+
+
+def test_admission_aging_escape_under_low_load(monkeypatch):
+    """A waiting prompt must not starve behind long-running decodes when
+    the batching threshold is never reached (low-load latency bug)."""
+    import time as _time
+
+    torch.manual_seed(0)
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=256),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=4096, max_model_len=512,
+        ),
+    )
+    cfg.scheduler.prefill_admission_tokens = 4096
+    eng = LLMEngine(cfg, device="cpu")
+    # one long decode in flight
+    eng.add_request([1, 2, 3] * 5, SamplingParams(max_tokens=400))
+    for _ in range(3):
+        eng.step()
+    # a small prompt arrives: threshold (4096) will never accumulate
+    b = eng.add_request([9, 9] * 8, SamplingParams(max_tokens=2))
+    arrival = eng.seqs[b].arrival_time
+    # age it past the 50 ms escape
+    eng.seqs[b].arrival_time = arrival - 1.0
+    eng.step()
+    assert eng.seqs[b].num_computed_tokens > 0  # admitted, not starved
