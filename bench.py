@@ -41,6 +41,8 @@ def parse_args():
     p.add_argument("--concurrency", type=int, default=256)
     p.add_argument("--max-batched-tokens", type=int, default=8192)
     p.add_argument("--enforce-eager", action="store_true")
+    p.add_argument("--quantization", choices=["fp8"], default=None,
+                   help="opt-in fp8 serving mode (headline stays bf16)")
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree")
     p.add_argument("--shared-prefix-len", type=int, default=0,
                    help="shared-prefix workload (BASELINE config #4): each "
@@ -132,6 +134,7 @@ def main():
     from fusioninfer_amd.models.registry import get_model_config
 
     mc = get_model_config(args.model)
+    mc.quantization = args.quantization
     max_len = args.prompt_len + args.gen_len + 64
     cfg = EngineConfig(
         model=mc,
@@ -247,7 +250,7 @@ def _aggregate_and_report(args, rank, world, distributed, device, use_cuda,
         "higher_is_better": True,
         "scaling": "weak",
         "vs_baseline": None,
-        "dtype": "bf16",
+        "dtype": args.quantization or "bf16",
         "data": "synthetic",
         "p50_ttft_ms": round(p50_ttft_ms, 1),
         "tokens_per_s": round(gen_tokens / elapsed, 1) if elapsed else 0.0,

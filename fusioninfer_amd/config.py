@@ -32,6 +32,8 @@ class ModelConfig:
     dtype: str = "bfloat16"
     # optional HF-layout safetensors checkpoint dir; None = random init
     model_path: Optional[str] = None
+    # None = bf16; "fp8" = OCP e4m3 weights + dynamic per-token activations
+    quantization: Optional[str] = None
 
     @property
     def q_size(self) -> int:

@@ -38,6 +38,13 @@ class ModelRunner:
             from fusioninfer_amd.models.weight_loader import load_safetensors_dir
 
             load_safetensors_dir(self.model, mc.model_path)
+        if mc.quantization == "fp8":
+            from fusioninfer_amd.quantization import convert_linear_to_fp8
+
+            n = convert_linear_to_fp8(self.model)
+            assert n > 0
+        elif mc.quantization:
+            raise ValueError(f"unknown quantization {mc.quantization!r}")
         self.block_size = cfg.cache.block_size
         self.max_blocks_per_seq = (
             cfg.scheduler.max_model_len + self.block_size - 1

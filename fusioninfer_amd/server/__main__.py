@@ -33,6 +33,7 @@ def parse_args(argv=None):
                    help='JSON: {"kv_connector": "RcclConnector", '
                         '"kv_role": "kv_producer"|"kv_consumer"}')
     p.add_argument("--enforce-eager", action="store_true")
+    p.add_argument("--quantization", choices=["fp8"], default=None)
     p.add_argument("--enable-lora", action="store_true")
     p.add_argument("--max-loras", type=int, default=8)
     p.add_argument("--lora-modules", nargs="*", default=[],
@@ -68,6 +69,7 @@ def build_engine_config(args):
         )
     mc = get_model_config(args.model)
     mc.model_path = args.model_path
+    mc.quantization = args.quantization
     return EngineConfig(
         model=mc,
         cache=CacheConfig(
