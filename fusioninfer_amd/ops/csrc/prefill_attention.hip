@@ -37,6 +37,11 @@ constexpr int kQPerWave = 32;                 // q rows per wave (1 MFMA block)
 constexpr int kQBlock = kWaves * kQPerWave;   // 256 q rows per workgroup
 constexpr int kKVTile = 64;                   // kv tokens per LDS tile
 constexpr float kPNegInf = -1e30f;
+// defer-max threshold (guide T13): skip the O/l rescale while the tile max
+// grows by <= this much — P is then bounded by e^8, which f32 accumulation
+// absorbs (max-abs error ~3x vs THR=0 per the guide's measurement; the
+// spiked-scores test forces the rescale branch).
+constexpr float kDeferThr = 8.0f;
 
 // byte-address XOR swizzles, bijective within a ROWB-byte row (guide T2/G4)
 template <int ROWB>
@@ -262,23 +267,27 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) rm = fmaxf(rm, s_acc[b][r]);
       rm = fmaxf(rm, __shfl_xor(rm, 32, 64));
-      const float m_new = fmaxf(m_row, rm);
       float alpha;
-      if (m_new <= kPNegInf) {
-        alpha = 0.f;  // row has seen no unmasked score yet (o is 0)
+      if (m_row > kPNegInf && __all(rm - m_row <= kDeferThr)) {
+        alpha = 1.f;  // defer-max: keep the old running max, skip rescale
       } else {
-        alpha = (m_row <= kPNegInf) ? 0.f : __expf(m_row - m_new);
-        m_row = m_new;
+        const float m_new = fmaxf(m_row, rm);
+        if (m_new <= kPNegInf) {
+          alpha = 0.f;  // row has seen no unmasked score yet (o is 0)
+        } else {
+          alpha = (m_row <= kPNegInf) ? 0.f : __expf(m_row - m_new);
+          m_row = m_new;
+        }
       }
       float rs = 0.f;
 #pragma unroll
       for (int b = 0; b < 2; ++b)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const float pv = (m_new <= kPNegInf || s_acc[b][r] <= kPNegInf)
+          const float pv = (m_row <= kPNegInf || s_acc[b][r] <= kPNegInf)
                                ? 0.f
-                               : __expf(s_acc[b][r] - m_new);
-          s_acc[b][r] = pv;  // reuse as P
+                               : __expf(s_acc[b][r] - m_row);
+          s_acc[b][r] = pv;  // reuse as P (bounded by e^kDeferThr)
           rs += pv;
         }
       rs += __shfl_xor(rs, 32, 64);
