@@ -83,3 +83,38 @@ def test_gpu_graph_replay_consistent_with_eager():
     graphed = make_engine(False).generate(prompts, SamplingParams(max_tokens=8))
     for a, b in zip(eager, graphed):
         assert a.output_token_ids == b.output_token_ids
+
+
+def test_gpu_lora_request_differs_from_base():
+    torch.manual_seed(0)
+    eng = make_engine(True)
+    eng.add_lora("tuned", rank=8, seed=42)
+    prompt = list(range(20, 80))
+    base = eng.generate(prompt and [prompt], SamplingParams(max_tokens=5))[0]
+    req = eng.add_request(prompt, SamplingParams(max_tokens=5), lora_name="tuned")
+    outs = {}
+    while eng.has_unfinished():
+        for o in eng.step():
+            if o.finished:
+                outs[o.request_id] = o
+    assert outs[req].output_token_ids != base.output_token_ids
+
+
+def test_gpu_fp8_engine_generates():
+    from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+
+    torch.manual_seed(0)
+    mc = get_model_config("Qwen3-0.6B")
+    mc.num_layers = 4
+    mc.quantization = "fp8"
+    cfg = EngineConfig(
+        model=mc,
+        cache=CacheConfig(num_gpu_blocks=256),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=1024, max_model_len=256
+        ),
+        enforce_eager=True,
+    )
+    eng = LLMEngine(cfg, device="cuda:0")
+    out = eng.generate([[7, 8, 9] * 10], SamplingParams(max_tokens=4))[0]
+    assert len(out.output_token_ids) == 4
