@@ -69,7 +69,8 @@ class ModelRunner:
 
         if self.cfg.cache.num_gpu_blocks is not None:
             return ps.tp_all_reduce_min_int(self.cfg.cache.num_gpu_blocks)
-        assert self.is_cuda
+        if not self.is_cuda:  # CPU debug serving: small fixed pool
+            return 2048
         free_b, total_b = torch.cuda.mem_get_info(self.device)
         usable = int(
             total_b * self.cfg.cache.gpu_memory_utilization
