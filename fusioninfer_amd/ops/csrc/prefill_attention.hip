@@ -109,12 +109,15 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     return (seq_start + r) * dense_stride + static_cast<int64_t>(kv_head) * D;
   };
 
-  // LDS: double-buffered K (glds target) + V^T tile. ONE __shared__ object
+  // LDS: double-buffered K (glds target) + DOUBLE-buffered V^T so the
+  // tile loop needs only ONE barrier (PV reads vt[t&1] while the next
+  // tile's V lands in vt[(t+1)&1]). ONE __shared__ object
   // (guide §5 ".s-level traps" (a)).
-  __shared__ u16 smem[2 * kKVTile * D + D * (kVTRowB / 2)];
+  __shared__ u16 smem[2 * kKVTile * D + 2 * D * (kVTRowB / 2)];
   u16* k_lds0 = smem;
   u16* k_lds1 = smem + kKVTile * D;
-  u16* vt_lds = smem + 2 * kKVTile * D;
+  u16* vt_lds0 = smem + 2 * kKVTile * D;
+  u16* vt_lds1 = vt_lds0 + D * (kVTRowB / 2);
 
   // ---- Q fragments (the QK^T B operand): lane holds
   // Q[row0+col][f*16 + hi1*8 .. +8] for each 16-deep k-chunk f
@@ -173,9 +176,9 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     vreg0 = *reinterpret_cast<const bf16x8*>(vrow);
     vreg1 = *reinterpret_cast<const bf16x8*>(vrow + 8);
   };
-  auto vwrite = [&]() {
+  auto vwrite = [&](u16* vt_buf) {
     if (v_c16 >= kChunks) return;
-    char* vbase = reinterpret_cast<char*>(vt_lds);
+    char* vbase = reinterpret_cast<char*>(vt_buf);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       *reinterpret_cast<u16*>(
@@ -200,16 +203,21 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     for (int cb = 0; cb < CB; ++cb)
       v_off[kc][cb] = swz<kVTRowB>(cb * 32 + col, kc * 32 + hi1 * 16);
 
+  // prologue: K(0) in flight + V(0) staged and published, V(1) in flight
   stage_k_glds(0, k_lds0);
   vload(0);
-  __syncthreads();
+  __syncthreads();          // drains K(0) glds
+  vwrite(vt_lds0);
+  if (num_kv_tiles > 1) vload(1);
+  __syncthreads();          // publishes vt[0]
 
   for (int t = 0; t < num_kv_tiles; ++t) {
     const int kv0 = t * kKVTile;
     u16* kbuf = (t & 1) ? k_lds1 : k_lds0;
     u16* kbuf_next = (t & 1) ? k_lds0 : k_lds1;
+    const u16* vt_cur = (t & 1) ? vt_lds1 : vt_lds0;
+    u16* vt_next = (t & 1) ? vt_lds0 : vt_lds1;
 
-    vwrite();
     if (t + 1 < num_kv_tiles) stage_k_glds(t + 1, kbuf_next);
 
     const bool compute = active && kv0 <= my_q_max;
@@ -305,12 +313,8 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
       }
     }
 
-    // publish vt_lds(t); drains the K(t+1) glds QK^T just covered
-    __syncthreads();
-    if (t + 1 < num_kv_tiles) vload(t + 1);
-
     if (compute) {
-      const char* vbase = reinterpret_cast<const char*>(vt_lds);
+      const char* vbase = reinterpret_cast<const char*>(vt_cur);
       // ---- P -> A-fragments in-register (pack pairs + half swap) ----
       // A-frag for kv chunk [32b + 16c, +16): lane needs kv {hi1*8..+8} of
       // that range; own regs hold kv {..}+4*hi1 interleaved, the partner
@@ -353,7 +357,14 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
       }
       __builtin_amdgcn_s_setprio(0);
     }
-    // PV(t) reads of vt_lds done before vwrite(t+1) overwrites it
+    // stage V(t+1) (regs already resident) and prefetch V(t+2)
+    if (t + 1 < num_kv_tiles) {
+      vwrite(vt_next);
+      if (t + 2 < num_kv_tiles) vload(t + 2);
+    }
+    // ONE barrier: publishes vt[(t+1)&1]; all waves' PV(t) reads of
+    // vt[t&1] complete (it is rewritten only after the NEXT barrier);
+    // drains the K(t+1) glds that QK^T(t)+PV(t) covered
     __syncthreads();
   }
 
