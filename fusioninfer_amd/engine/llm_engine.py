@@ -103,7 +103,11 @@ class LLMEngine:
             request_id = f"req-{next(self._req_counter)}"
         seq = Sequence(request_id, prompt_token_ids, sampling or SamplingParams())
         seq.lora_name = lora_name
-        assert seq.num_prompt_tokens <= self.cfg.scheduler.max_model_len
+        if seq.num_prompt_tokens > self.cfg.scheduler.max_model_len:
+            raise ValueError(
+                f"prompt length {seq.num_prompt_tokens} exceeds "
+                f"max_model_len {self.cfg.scheduler.max_model_len}"
+            )
         self.seqs[request_id] = seq
         self.scheduler.add(seq)
         return request_id

@@ -137,7 +137,11 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         lora = body.get("model")
         if lora not in serving.engine.active_loras():
             lora = None
-        req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
+        try:
+            req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
+        except ValueError as e:
+            return JSONResponse({"error": {"message": str(e),
+                                           "type": "invalid_request_error"}}, 400)
         created = int(time.time())
         cid = f"cmpl-{uuid.uuid4().hex[:16]}"
 
@@ -205,7 +209,11 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         lora = body.get("model")
         if lora not in serving.engine.active_loras():
             lora = None
-        req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
+        try:
+            req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
+        except ValueError as e:
+            return JSONResponse({"error": {"message": str(e),
+                                           "type": "invalid_request_error"}}, 400)
         created = int(time.time())
         cid = f"chatcmpl-{uuid.uuid4().hex[:16]}"
 

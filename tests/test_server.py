@@ -138,3 +138,16 @@ def test_cli_config_surface():
     assert cfg.scheduler.max_model_len == 1024
     assert cfg.kv_transfer.kv_connector == "RcclConnector"
     assert cfg.kv_transfer.kv_role == "kv_producer"
+
+
+def test_too_long_prompt_returns_400(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": [1] * 10000, "max_tokens": 1},
+            )
+            assert r.status_code == 400
+            assert "max_model_len" in r.json()["error"]["message"]
+
+    asyncio.run(run())
