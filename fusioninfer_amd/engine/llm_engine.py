@@ -8,6 +8,7 @@ TP via symmetric SPMD stepping, see distributed/parallel_state.py).
 from __future__ import annotations
 
 import itertools
+import os
 import time
 from typing import Dict, List, Optional
 
@@ -64,6 +65,8 @@ class LLMEngine:
         self.seqs: Dict[str, Sequence] = {}
         self._held: Dict[str, Sequence] = {}
         self._import_holder = None
+        self._timing = os.environ.get("FI_STEP_TIMING") == "1"
+        self._tacc = {}
         # running stats for metrics / EPP scorers
         self.num_finished = 0
         self.num_generated_tokens = 0
@@ -214,15 +217,30 @@ class LLMEngine:
         step_t0 = time.monotonic()
         tp = self.cfg.parallel.tensor_parallel_size
         batch = self.scheduler.schedule()
+        t_sched = time.monotonic()
         if batch.is_empty:
             return []
         payload = self.runner.build_batch_payload(
             batch.prefill_seqs, batch.prefill_chunks, batch.decode_seqs,
             self.block_manager,
         )
+        t_payload = time.monotonic()
         if tp > 1:
             self._ps.tp_broadcast_object(payload)
         logits = self.runner.run_batch(payload)
+        if self._timing:
+            import torch as _t
+
+            if logits.is_cuda:
+                _t.cuda.synchronize()
+            t_fwd = time.monotonic()
+            kind = "P" if batch.prefill_seqs else "D"
+            acc = self._tacc.setdefault(kind, [0.0, 0.0, 0.0, 0.0, 0])
+            acc[0] += t_sched - step_t0
+            acc[1] += t_payload - t_sched
+            acc[2] += t_fwd - t_payload
+            acc[4] += 1
+            self._t_fwd_mark = t_fwd
         for seq, chunk in zip(batch.prefill_seqs, batch.prefill_chunks):
             seq.num_computed_tokens = (
                 seq.num_computed_tokens or seq.num_cached_tokens
@@ -270,6 +288,17 @@ class LLMEngine:
                 self.e2e_latency_sum += seq.finish_time - seq.arrival_time
                 del self.seqs[seq.seq_id]
             outputs.append(RequestOutput(seq))
+        if self._timing and hasattr(self, "_t_fwd_mark"):
+            kind = "P" if batch.prefill_seqs else "D"
+            self._tacc[kind][3] += time.monotonic() - self._t_fwd_mark
+            n = self._tacc[kind][4]
+            if n % 50 == 0:
+                a = self._tacc[kind]
+                print(
+                    f"[timing {kind}] n={n} sched={a[0]/n*1e3:.2f}ms "
+                    f"payload={a[1]/n*1e3:.2f}ms fwd={a[2]/n*1e3:.2f}ms "
+                    f"sample+book={a[3]/n*1e3:.2f}ms", flush=True,
+                )
         self.step_time_sum += time.monotonic() - step_t0
         self.num_steps += 1
         return outputs
