@@ -83,6 +83,14 @@ class LLMEngine:
         otherwise — zero-B adapters are numerically the base model)."""
         from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
 
+        # TP>1 would need the adapter registered on every rank before any
+        # batch references it (worker forwards must match the driver's) —
+        # adapter broadcast is round-2 work
+        assert self.cfg.parallel.tensor_parallel_size == 1, (
+            "LoRA with tensor parallelism needs rank-synchronized adapter "
+            "registration (not yet wired)"
+        )
+
         if self.runner.lora_registry is None:
             self.runner.lora_registry = LoRARegistry()
         adapter = LoRAAdapter(
