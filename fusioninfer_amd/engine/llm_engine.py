@@ -90,6 +90,11 @@ class LLMEngine:
             "LoRA with tensor parallelism needs rank-synchronized adapter "
             "registration (not yet wired)"
         )
+        # fp8 mode feeds (fp8, scale) tuples through the layers; LoRA's
+        # bf16 shrink/expand GEMMs would need the bf16 activations
+        assert self.cfg.model.quantization != "fp8", (
+            "LoRA is not supported with --quantization fp8"
+        )
 
         if self.runner.lora_registry is None:
             self.runner.lora_registry = LoRARegistry()

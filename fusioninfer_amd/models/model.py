@@ -65,7 +65,7 @@ class Attention(nn.Module):
 
     def forward(
         self,
-        x: torch.Tensor,
+        x,  # [T, H] bf16, or (x_fp8, scales) from a fused fp8 epilogue
         meta: AttnMetadata,
         kv_cache,  # (k_cache, v_cache) [B, Hk, bs, D]
         cos_sin: torch.Tensor,
@@ -84,9 +84,9 @@ class Attention(nn.Module):
         k_cache, v_cache = kv_cache
         ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
 
-        T = x.shape[0]
+        T = qkv.shape[0]
         np_, nd = meta.num_prefill_tokens, meta.num_decode_tokens
-        out = torch.empty(T, self.q_size, dtype=x.dtype, device=x.device)
+        out = torch.empty(T, self.q_size, dtype=qkv.dtype, device=qkv.device)
         if np_ > 0:
             # context attention over the paged cache (the new tokens' K/V
             # were just written above); cached prefixes are never recomputed
@@ -122,14 +122,18 @@ class MLP(nn.Module):
     def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
         super().__init__()
         self.layer_idx = layer_idx
+        self.fp8 = cfg.quantization == "fp8"
         self.gate_up_proj = MergedColumnParallelLinear(
             cfg.hidden_size, [cfg.intermediate_size, cfg.intermediate_size]
         )
         self.down_proj = RowParallelLinear(cfg.intermediate_size, cfg.hidden_size)
         self.inter_per_rank = cfg.intermediate_size // ps.tp_world_size()
 
-    def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
+    def forward(self, x, lora=None) -> torch.Tensor:
         gu = self.gate_up_proj(x)
+        if self.fp8:
+            # fused SwiGLU + fp8 row quant straight into the fp8 down_proj
+            return self.down_proj(ops.silu_and_mul_fp8(gu))
         if lora is not None:
             lora.apply(self.layer_idx, "gate_up", x, gu)
         act = ops.silu_and_mul(gu)
@@ -152,8 +156,26 @@ class DecoderLayer(nn.Module):
         self.self_attn = Attention(cfg, layer_idx)
         self.mlp = MLP(cfg, layer_idx)
         self.eps = cfg.rms_norm_eps
+        self.fp8 = cfg.quantization == "fp8"
 
     def forward(self, hidden, residual, meta, kv_cache, cos_sin):
+        if self.fp8:
+            # fused epilogues: the norms emit (fp8, per-row scale) directly
+            # into the fp8 GEMMs — the bf16 normalized row never hits HBM
+            if residual is None:
+                residual = hidden.clone()
+                x = ops.rms_norm_fp8(hidden, self.input_norm_weight, self.eps)
+            else:
+                f8, sc, residual = ops.fused_add_rms_norm_fp8(
+                    hidden, residual, self.input_norm_weight, self.eps
+                )
+                x = (f8, sc)
+            hidden = self.self_attn(x, meta, kv_cache, cos_sin)
+            f8, sc, residual = ops.fused_add_rms_norm_fp8(
+                hidden, residual, self.post_norm_weight, self.eps
+            )
+            hidden = self.mlp((f8, sc))
+            return hidden, residual
         if residual is None:
             residual = hidden.clone()
             hidden = ops.rms_norm(hidden, self.input_norm_weight, self.eps)

@@ -70,6 +70,60 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     return ref.silu_and_mul(x)
 
 
+# ------------------------------------------------------- fp8 fused epilogues
+#
+# The fp8 serving mode quantizes activations per-token for torch._scaled_mm;
+# done eagerly that costs more than the fp8 GEMM saves, so quantization is
+# fused into the producing kernel (norm / SwiGLU) or done in one pass
+# (quant_fp8_rows for the attention output).
+
+def rms_norm_fp8(x: torch.Tensor, weight: torch.Tensor, eps: float):
+    """Returns (x_fp8 [T, H] e4m3, scales [T] fp32)."""
+    if x.is_cuda:
+        _require_native()
+        out = torch.empty(x.shape, dtype=torch.float8_e4m3fn, device=x.device)
+        scales = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+        _C.rms_norm_fp8(out, scales, x, weight, eps)
+        return out, scales
+    return ref.rms_norm_fp8(x, weight, eps)
+
+
+def fused_add_rms_norm_fp8(x, residual, weight, eps: float):
+    """residual += x in place; returns (x_fp8, scales, residual)."""
+    if x.is_cuda:
+        _require_native()
+        out = torch.empty(x.shape, dtype=torch.float8_e4m3fn, device=x.device)
+        scales = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+        _C.fused_add_rms_norm_fp8(out, scales, x, residual, weight, eps)
+        return out, scales, residual
+    return ref.fused_add_rms_norm_fp8(x, residual, weight, eps)
+
+
+def silu_and_mul_fp8(x: torch.Tensor):
+    """x: [T, 2*I] -> (act_fp8 [T, I] e4m3, scales [T] fp32)."""
+    if x.is_cuda:
+        _require_native()
+        inter = x.shape[-1] // 2
+        out = torch.empty(
+            (*x.shape[:-1], inter), dtype=torch.float8_e4m3fn, device=x.device
+        )
+        scales = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+        _C.silu_and_mul_fp8(out, scales, x)
+        return out, scales
+    return ref.silu_and_mul_fp8(x)
+
+
+def quant_fp8_rows(x: torch.Tensor):
+    """Per-row dynamic fp8 quant: (x_fp8, scales [T] fp32)."""
+    if x.is_cuda:
+        _require_native()
+        out = torch.empty(x.shape, dtype=torch.float8_e4m3fn, device=x.device)
+        scales = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+        _C.quant_fp8_rows(out, scales, x)
+        return out, scales
+    return ref.quant_fp8_rows(x)
+
+
 # ---------------------------------------------------------------- rope
 
 def rope_qk_norm_(

@@ -21,6 +21,7 @@ SOURCES = [
     "ops.hip",
     "norm.hip",
     "activation.hip",
+    "quant.hip",
     "rope.hip",
     "cache.hip",
     "paged_attention.hip",

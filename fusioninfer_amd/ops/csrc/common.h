@@ -79,4 +79,31 @@ __host__ __device__ __forceinline__ constexpr int ceil_div(int a, int b) {
   return (a + b - 1) / b;
 }
 
+// pack 8 f32 into 8 OCP-e4m3 bytes (two u32 words) via v_cvt_pk_fp8_f32
+FI_DEV void pack_fp8x8(const float* v, unsigned int out[2]) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  unsigned int w0 = 0, w1 = 0;
+  w0 = __builtin_amdgcn_cvt_pk_fp8_f32(v[0], v[1], w0, false);
+  w0 = __builtin_amdgcn_cvt_pk_fp8_f32(v[2], v[3], w0, true);
+  w1 = __builtin_amdgcn_cvt_pk_fp8_f32(v[4], v[5], w1, false);
+  w1 = __builtin_amdgcn_cvt_pk_fp8_f32(v[6], v[7], w1, true);
+  out[0] = w0;
+  out[1] = w1;
+#else
+  out[0] = out[1] = 0;  // host pass never executes device code
+#endif
+}
+
+template <int NWAVES>
+FI_DEV float block_reduce_max(float wave_val, float* scratch) {
+  const int wave = threadIdx.x / kWaveSize;
+  const int lane = threadIdx.x % kWaveSize;
+  if (lane == 0) scratch[wave] = wave_val;
+  __syncthreads();
+  float m = scratch[0];
+#pragma unroll
+  for (int i = 1; i < NWAVES; ++i) m = fmaxf(m, scratch[i]);
+  return m;
+}
+
 }  // namespace fi

@@ -9,10 +9,13 @@
 
 namespace fi {
 
-template <bool FUSED_ADD>
-void launch_rms_norm(u16*, const u16*, u16*, const u16*, float, int, int,
-                     hipStream_t);
+template <bool FUSED_ADD, bool FP8_OUT>
+void launch_rms_norm(void*, float*, const u16*, u16*, const u16*, float, int,
+                     int, hipStream_t);
 void launch_silu_and_mul(u16*, const u16*, int64_t, int, hipStream_t);
+template <bool SILU_MUL>
+void launch_row_quant_fp8(unsigned char*, float*, const u16*, int, int,
+                          hipStream_t);
 void launch_rope_qk_norm(u16*, u16*, int64_t, int64_t, const u16*, const u16*,
                          const float*, const int*, int, int, int, int, float,
                          hipStream_t);
@@ -56,9 +59,10 @@ void rms_norm(at::Tensor out, at::Tensor input, at::Tensor weight, double eps) {
   TORCH_CHECK(input.is_contiguous() && out.is_contiguous());
   const int hidden = input.size(-1);
   const int tokens = input.numel() / hidden;
-  fi::launch_rms_norm<false>(bf16_ptr(out), bf16_cptr(input), nullptr,
-                             bf16_cptr(weight), static_cast<float>(eps),
-                             tokens, hidden, current_stream());
+  fi::launch_rms_norm<false, false>(bf16_ptr(out), nullptr, bf16_cptr(input),
+                                    nullptr, bf16_cptr(weight),
+                                    static_cast<float>(eps), tokens, hidden,
+                                    current_stream());
 }
 
 void fused_add_rms_norm(at::Tensor input, at::Tensor residual,
@@ -69,10 +73,69 @@ void fused_add_rms_norm(at::Tensor input, at::Tensor residual,
   const int hidden = input.size(-1);
   const int tokens = input.numel() / hidden;
   // in-place: residual += input; input = rmsnorm(residual)
-  fi::launch_rms_norm<true>(bf16_ptr(input), bf16_cptr(input),
-                            bf16_ptr(residual), bf16_cptr(weight),
-                            static_cast<float>(eps), tokens, hidden,
-                            current_stream());
+  fi::launch_rms_norm<true, false>(bf16_ptr(input), nullptr, bf16_cptr(input),
+                                   bf16_ptr(residual), bf16_cptr(weight),
+                                   static_cast<float>(eps), tokens, hidden,
+                                   current_stream());
+}
+
+#define CHECK_FP8_OUT(o, s)                                                  \
+  TORCH_CHECK((o).scalar_type() == at::kFloat8_e4m3fn && (o).is_cuda() &&    \
+              (o).is_contiguous());                                          \
+  TORCH_CHECK((s).scalar_type() == at::kFloat && (s).is_cuda())
+
+void rms_norm_fp8(at::Tensor out, at::Tensor out_scales, at::Tensor input,
+                  at::Tensor weight, double eps) {
+  CHECK_BF16_CUDA(input);
+  CHECK_FP8_OUT(out, out_scales);
+  TORCH_CHECK(input.is_contiguous());
+  const int hidden = input.size(-1);
+  const int tokens = input.numel() / hidden;
+  fi::launch_rms_norm<false, true>(
+      out.data_ptr(), out_scales.data_ptr<float>(), bf16_cptr(input), nullptr,
+      bf16_cptr(weight), static_cast<float>(eps), tokens, hidden,
+      current_stream());
+}
+
+void fused_add_rms_norm_fp8(at::Tensor out, at::Tensor out_scales,
+                            at::Tensor input, at::Tensor residual,
+                            at::Tensor weight, double eps) {
+  CHECK_BF16_CUDA(input);
+  CHECK_BF16_CUDA(residual);
+  CHECK_FP8_OUT(out, out_scales);
+  TORCH_CHECK(input.is_contiguous() && residual.is_contiguous());
+  const int hidden = input.size(-1);
+  const int tokens = input.numel() / hidden;
+  // residual += input (in place); out = fp8(rmsnorm(residual))
+  fi::launch_rms_norm<true, true>(
+      out.data_ptr(), out_scales.data_ptr<float>(), bf16_cptr(input),
+      bf16_ptr(residual), bf16_cptr(weight), static_cast<float>(eps), tokens,
+      hidden, current_stream());
+}
+
+void silu_and_mul_fp8(at::Tensor out, at::Tensor out_scales, at::Tensor input) {
+  CHECK_BF16_CUDA(input);
+  CHECK_FP8_OUT(out, out_scales);
+  TORCH_CHECK(input.is_contiguous());
+  const int inter = out.size(-1);
+  TORCH_CHECK(input.size(-1) == 2 * inter);
+  TORCH_CHECK(inter % 8 == 0);
+  fi::launch_row_quant_fp8<true>(
+      static_cast<unsigned char*>(out.data_ptr()),
+      out_scales.data_ptr<float>(), bf16_cptr(input),
+      input.numel() / (2 * inter), inter, current_stream());
+}
+
+void quant_fp8_rows(at::Tensor out, at::Tensor out_scales, at::Tensor input) {
+  CHECK_BF16_CUDA(input);
+  CHECK_FP8_OUT(out, out_scales);
+  TORCH_CHECK(input.is_contiguous());
+  const int cols = input.size(-1);
+  TORCH_CHECK(cols % 8 == 0);
+  fi::launch_row_quant_fp8<false>(
+      static_cast<unsigned char*>(out.data_ptr()),
+      out_scales.data_ptr<float>(), bf16_cptr(input), input.numel() / cols,
+      cols, current_stream());
 }
 
 void silu_and_mul(at::Tensor out, at::Tensor input) {
@@ -243,6 +306,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_add_rms_norm", &fused_add_rms_norm,
         "in-place residual add + RMSNorm");
   m.def("silu_and_mul", &silu_and_mul, "fused SiLU-mul");
+  m.def("rms_norm_fp8", &rms_norm_fp8, "RMSNorm with fused fp8 row quant");
+  m.def("fused_add_rms_norm_fp8", &fused_add_rms_norm_fp8,
+        "residual add + RMSNorm with fused fp8 row quant");
+  m.def("silu_and_mul_fp8", &silu_and_mul_fp8,
+        "fused SiLU-mul with fp8 row quant");
+  m.def("quant_fp8_rows", &quant_fp8_rows, "per-row dynamic fp8 quant");
   m.def("rope_qk_norm", &rope_qk_norm,
         "fused per-head qk RMSNorm + NeoX RoPE (in-place)");
   m.def("reshape_and_cache", &reshape_and_cache, "scatter K/V into paged cache");

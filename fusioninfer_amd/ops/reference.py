@@ -33,6 +33,31 @@ def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     return (torch.nn.functional.silu(gate) * up).to(x.dtype)
 
 
+def quant_fp8_rows(x: torch.Tensor):
+    """Per-row dynamic OCP-e4m3 quantization: (x_fp8 [T, C], scale [T] fp32)."""
+    xf = x.float()
+    amax = xf.abs().amax(dim=-1, keepdim=True).clamp(min=1e-8)
+    scale = amax / 448.0
+    x_fp8 = (xf / scale).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    return x_fp8, scale.squeeze(-1)
+
+
+def rms_norm_fp8(x: torch.Tensor, weight: torch.Tensor, eps: float):
+    return quant_fp8_rows(rms_norm(x, weight, eps))
+
+
+def fused_add_rms_norm_fp8(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float
+):
+    out, new_residual = fused_add_rms_norm(x, residual, weight, eps)
+    x_fp8, scale = quant_fp8_rows(out)
+    return x_fp8, scale, new_residual
+
+
+def silu_and_mul_fp8(x: torch.Tensor):
+    return quant_fp8_rows(silu_and_mul(x))
+
+
 def compute_cos_sin_cache(
     head_dim: int, max_positions: int, theta: float, device=None
 ) -> torch.Tensor:

@@ -40,10 +40,21 @@ def quantize_activation_fp8(x: torch.Tensor):
     return x_fp8, scale.squeeze(1)
 
 
-def fp8_linear(x: torch.Tensor, w_fp8: torch.Tensor, w_scale: torch.Tensor):
-    """y[T, O] = x @ W^T with fp8 operands, bf16 output."""
-    if x.is_cuda:
-        x_fp8, x_scale = quantize_activation_fp8(x)
+def fp8_linear(x, w_fp8: torch.Tensor, w_scale: torch.Tensor):
+    """y[T, O] = x @ W^T with fp8 operands, bf16 output.
+
+    `x` is either a bf16 tensor (quantized here, one fused HIP kernel on
+    GPU) or an (x_fp8, x_scale) tuple already produced by a fused epilogue
+    (rms_norm_fp8 / silu_and_mul_fp8 / quant_fp8_rows)."""
+    if isinstance(x, tuple):
+        x_fp8, x_scale = x
+    elif x.is_cuda:
+        from fusioninfer_amd import ops
+
+        x_fp8, x_scale = ops.quant_fp8_rows(x)
+    else:
+        x_fp8, x_scale = None, None  # CPU: keep the bf16 tensor
+    if x_fp8 is not None and x_fp8.is_cuda:
         return torch._scaled_mm(
             x_fp8,
             w_fp8.t(),  # [I, O], column-major view of the row-major weight
@@ -52,6 +63,8 @@ def fp8_linear(x: torch.Tensor, w_fp8: torch.Tensor, w_scale: torch.Tensor):
             out_dtype=torch.bfloat16,
         )
     # CPU reference path (tests): dequantize and matmul in fp32
+    if x_fp8 is not None:  # tuple input on CPU
+        x = x_fp8.float() * x_scale.unsqueeze(1).float()
     w = w_fp8.float() * w_scale.unsqueeze(1).float()
     return (x.float() @ w.t()).to(torch.bfloat16)
 

@@ -51,6 +51,64 @@ def test_engine_fp8_generates():
     assert len(out.output_token_ids) == 4
 
 
+def test_fused_fp8_epilogue_references_match_eager_quant():
+    """The fused CPU references (rms_norm_fp8 etc.) must agree with
+    quantize_activation_fp8 applied to the bf16 op output."""
+    import fusioninfer_amd.ops.reference as ref
+
+    torch.manual_seed(3)
+    x = torch.randn(8, 256, dtype=torch.bfloat16)
+    w = torch.rand(256, dtype=torch.bfloat16) + 0.5
+    out_bf16 = ref.rms_norm(x, w, 1e-6)
+    f8, sc = ref.rms_norm_fp8(x, w, 1e-6)
+    e8, es = quantize_activation_fp8(out_bf16)
+    assert torch.equal(f8.float(), e8.float())
+    assert torch.allclose(sc, es)
+
+    res = torch.randn(8, 256, dtype=torch.bfloat16)
+    out2, new_res = ref.fused_add_rms_norm(x, res.clone(), w, 1e-6)
+    f8, sc, r2 = ref.fused_add_rms_norm_fp8(x, res.clone(), w, 1e-6)
+    assert torch.equal(r2, new_res)
+    e8, _ = quantize_activation_fp8(out2)
+    assert torch.equal(f8.float(), e8.float())
+
+    gu = torch.randn(8, 512, dtype=torch.bfloat16)
+    f8, sc = ref.silu_and_mul_fp8(gu)
+    e8, es = quantize_activation_fp8(ref.silu_and_mul(gu))
+    assert torch.equal(f8.float(), e8.float())
+
+
+def test_fp8_linear_accepts_prequantized_tuple():
+    torch.manual_seed(4)
+    x = torch.randn(16, 128, dtype=torch.bfloat16)
+    w = torch.randn(64, 128, dtype=torch.bfloat16) * 0.05
+    w8, s = quantize_weight_fp8(w)
+    x8, xs = quantize_activation_fp8(x)
+    y_tuple = fp8_linear((x8, xs), w8, s).float()
+    y = x.float() @ w.float().t()
+    rel = (y_tuple - y).norm() / y.norm()
+    assert rel < 0.05
+
+
+def test_lora_rejected_in_fp8_mode():
+    from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.models.registry import get_model_config
+
+    mc = get_model_config("tiny-qwen3")
+    mc.quantization = "fp8"
+    cfg = EngineConfig(
+        model=mc,
+        cache=CacheConfig(num_gpu_blocks=32),
+        scheduler=SchedulerConfig(
+            max_num_seqs=2, max_num_batched_tokens=128, max_model_len=64
+        ),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    with pytest.raises(AssertionError):
+        eng.add_lora("a", rank=4)
+
+
 @pytest.mark.gpu
 def test_fp8_scaled_mm_gpu():
     torch.manual_seed(2)
