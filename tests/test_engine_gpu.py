@@ -100,6 +100,40 @@ def test_gpu_lora_request_differs_from_base():
     assert outs[req].output_token_ids != base.output_token_ids
 
 
+def test_gpu_fp8_logits_close_to_bf16():
+    """fp8 (fused-epilogue) forward vs bf16 forward with the SAME weights:
+    logits must agree to fp8 quantization error accumulated over 4 layers.
+    Catches wiring bugs (wrong scales, transposed operands) that a
+    generate-only smoke would miss."""
+    from fusioninfer_amd.engine.sequence import SamplingParams, Sequence
+    from fusioninfer_amd.engine.block_manager import BlockManager
+
+    def logits_for(quant):
+        torch.manual_seed(0)  # identical random-init weights both runs
+        mc = get_model_config("Qwen3-0.6B")
+        mc.num_layers = 4
+        mc.quantization = quant
+        cfg = EngineConfig(
+            model=mc,
+            cache=CacheConfig(num_gpu_blocks=256),
+            scheduler=SchedulerConfig(
+                max_num_seqs=8, max_num_batched_tokens=1024, max_model_len=256
+            ),
+            seed=7,
+            enforce_eager=True,
+        )
+        eng = LLMEngine(cfg, device="cuda:0")
+        bm = BlockManager(eng.runner.num_gpu_blocks, cfg.cache.block_size)
+        seq = Sequence("s", list(range(30, 130)), SamplingParams())
+        bm.allocate(seq)
+        return eng.runner.execute_prefill([seq], bm).float().cpu()
+
+    l_bf16 = logits_for(None)
+    l_fp8 = logits_for("fp8")
+    rel = (l_fp8 - l_bf16).norm() / l_bf16.norm()
+    assert rel.item() < 0.20, rel.item()
+
+
 def test_gpu_fp8_engine_generates():
     from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
 

@@ -57,15 +57,22 @@ def test_silu_and_mul(tokens, inter):
     assert_close_bf16(out.cpu(), expected)
 
 
-def _assert_fp8_close(got_f8, got_sc, exp_f8, exp_sc):
-    """fp8 kernel vs CPU reference: scales must match tightly; dequantized
-    values within one fp8 quantum (hardware cvt vs torch cast can differ
-    by 1 ulp on ties)."""
-    torch.testing.assert_close(got_sc.cpu(), exp_sc, atol=1e-5, rtol=1e-3)
+def _assert_fp8_close(got_f8, got_sc, exp_f8, exp_sc, exact=None):
+    """fp8 kernel vs CPU reference. The kernel quantizes from unrounded
+    f32 registers while the reference quantizes the bf16-rounded op
+    output, so scales agree only to bf16 precision and elements may land
+    one e4m3 quantum apart — compare dequantized values against the
+    reference DEQUANTIZED output within fp8 quantization error (e4m3: 3
+    mantissa bits, max rel step 2^-4; two independent quantizations of
+    near-identical data differ by <~1 quantum RMS)."""
+    torch.testing.assert_close(got_sc.cpu(), exp_sc, atol=1e-5, rtol=2e-2)
     got = got_f8.float().cpu() * got_sc.cpu().unsqueeze(-1)
     exp = exp_f8.float() * exp_sc.unsqueeze(-1)
     rel = (got - exp).norm() / exp.norm().clamp(min=1e-6)
-    assert rel.item() < 0.02, rel.item()
+    assert rel.item() < 0.05, rel.item()
+    if exact is not None:  # anchor to the unquantized fp32 op output
+        rel = (got - exact.float()).norm() / exact.float().norm().clamp(min=1e-6)
+        assert rel.item() < 0.06, rel.item()
 
 
 @pytest.mark.parametrize("tokens,hidden", [(1, 1024), (17, 4096), (64, 8192)])
@@ -74,7 +81,7 @@ def test_rms_norm_fp8(tokens, hidden):
     w = torch.randn(hidden, dtype=torch.bfloat16, device=DEV)
     f8, sc = ops.rms_norm_fp8(x, w, 1e-6)
     ef8, esc = ref.rms_norm_fp8(x.cpu(), w.cpu(), 1e-6)
-    _assert_fp8_close(f8, sc, ef8, esc)
+    _assert_fp8_close(f8, sc, ef8, esc, exact=ref.rms_norm(x.cpu(), w.cpu(), 1e-6))
 
 
 @pytest.mark.parametrize("tokens,hidden", [(16, 4096), (130, 4096)])
@@ -82,10 +89,11 @@ def test_fused_add_rms_norm_fp8(tokens, hidden):
     x = torch.randn(tokens, hidden, dtype=torch.bfloat16, device=DEV)
     res = torch.randn(tokens, hidden, dtype=torch.bfloat16, device=DEV)
     w = torch.randn(hidden, dtype=torch.bfloat16, device=DEV)
+    exact, _ = ref.fused_add_rms_norm(x.cpu(), res.cpu(), w.cpu(), 1e-6)
     ef8, esc, eres = ref.fused_add_rms_norm_fp8(x.cpu(), res.cpu(), w.cpu(), 1e-6)
     f8, sc, new_res = ops.fused_add_rms_norm_fp8(x, res, w, 1e-6)
     assert_close_bf16(new_res.cpu(), eres)
-    _assert_fp8_close(f8, sc, ef8, esc)
+    _assert_fp8_close(f8, sc, ef8, esc, exact=exact)
 
 
 @pytest.mark.parametrize("tokens,inter", [(7, 12288), (64, 512)])
@@ -93,7 +101,7 @@ def test_silu_and_mul_fp8(tokens, inter):
     x = torch.randn(tokens, 2 * inter, dtype=torch.bfloat16, device=DEV)
     f8, sc = ops.silu_and_mul_fp8(x)
     ef8, esc = ref.silu_and_mul_fp8(x.cpu())
-    _assert_fp8_close(f8, sc, ef8, esc)
+    _assert_fp8_close(f8, sc, ef8, esc, exact=ref.silu_and_mul(x.cpu()))
 
 
 @pytest.mark.parametrize("tokens,cols", [(1, 4096), (33, 2048), (256, 4096)])
@@ -101,7 +109,7 @@ def test_quant_fp8_rows(tokens, cols):
     x = torch.randn(tokens, cols, dtype=torch.bfloat16, device=DEV)
     f8, sc = ops.quant_fp8_rows(x)
     ef8, esc = ref.quant_fp8_rows(x.cpu())
-    _assert_fp8_close(f8, sc, ef8, esc)
+    _assert_fp8_close(f8, sc, ef8, esc, exact=x.cpu())
 
 
 @pytest.mark.parametrize("qk_norm", [True, False])
