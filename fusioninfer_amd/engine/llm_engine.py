@@ -182,20 +182,29 @@ class LLMEngine:
         self._import_holder = holder
         return holder.block_ids
 
+    def take_import_holder(self):
+        """PD consumer: detach the block-holder sequence created by the last
+        allocate_import_blocks() call so admission can happen later (the
+        HTTP decode request that claims this KV may not have arrived yet)."""
+        holder = self._import_holder
+        self._import_holder = None
+        return holder
+
     def add_imported_request(
         self,
         prompt_len: int,
         first_token: int,
         sampling: Optional["SamplingParams"] = None,
         request_id: Optional[str] = None,
+        holder=None,
     ) -> str:
         """PD consumer: register a request whose prompt KV was imported into
         the blocks reserved by allocate_import_blocks(). The sequence joins
         the RUNNING set directly (its prefill happened on the prefiller)."""
         from fusioninfer_amd.engine.sequence import SamplingParams, Sequence, SeqStatus
 
-        holder = self._import_holder
-        self._import_holder = None
+        if holder is None:
+            holder = self.take_import_holder()
         if request_id is None:
             request_id = f"req-{next(self._req_counter)}"
         seq = Sequence(request_id, [0] * prompt_len, sampling or SamplingParams())

@@ -97,6 +97,19 @@ def build_router_app(
         if pick.endpoint is None:
             return JSONResponse({"error": "no endpoint available"}, 503)
         headers = dict(pick.headers)
+        if pick.prefill_endpoint is not None:
+            # PD flow: prefill on the prefill-profile pick
+            # (prefiller ships KV to the decoder via the connector), then
+            # send the decode request claiming that KV batch by tag
+            pr = await http.post(
+                f"http://{pick.prefill_endpoint.address}/pd/prefill",
+                json={"prompt": token_ids},
+            )
+            if pr.status_code != 200:
+                return JSONResponse(
+                    {"error": f"prefill failed: {pr.text}"}, 502
+                )
+            headers["x-pd-tag"] = str(pr.json()["pd_tag"])
         url = f"http://{pick.endpoint.address}{path}"
         if body.get("stream"):
             async def relay():

@@ -25,6 +25,34 @@ from fusioninfer_amd.engine.llm_engine import LLMEngine
 from fusioninfer_amd.engine.sequence import SamplingParams
 
 
+def build_pd_connector(kvt, device: str = "cpu"):
+    """Create the KV connector for a PD server from its --kv-transfer-config
+    (reference surface: '{"kv_connector":"...","kv_role":"kv_producer"}').
+
+    The PD pair forms a dedicated 2-rank process group — prefiller rank 0,
+    decoder rank 1 by convention — over RCCL on GPU (one xGMI p2p pair) or
+    gloo on CPU. MASTER_ADDR/MASTER_PORT name the rendezvous (the control
+    plane's workload renderer points both pods at the same service).
+    PD currently composes with TP=1 servers (the pair group is the default
+    process group)."""
+    if kvt is None or not kvt.kv_connector:
+        return None
+    import torch
+    import torch.distributed as dist
+
+    role = kvt.kv_role
+    assert role in (KV_PRODUCER, KV_CONSUMER), role
+    backend = (
+        "nccl" if device.startswith("cuda") and torch.cuda.is_available()
+        else "gloo"
+    )
+    if not dist.is_initialized():
+        dist.init_process_group(
+            backend=backend, rank=kvt.kv_rank, world_size=kvt.kv_world_size
+        )
+    return RcclKVConnector(role, peer_rank=1 - kvt.kv_rank, device=device)
+
+
 class PDPrefiller:
     def __init__(self, engine: LLMEngine, connector: RcclKVConnector):
         assert connector.role == KV_PRODUCER
@@ -56,7 +84,7 @@ class PDDecoder:
     def accept(self, sampling: Optional[SamplingParams] = None) -> str:
         """Receive one request's KV from the prefiller and admit it into the
         decode loop. Returns the request id."""
-        _, prompt_len, first_token = self.connector.recv_kv(
+        _, prompt_len, first_token, _tag = self.connector.recv_kv(
             self.engine.runner.kv_caches,
             self.engine.allocate_import_blocks,
         )

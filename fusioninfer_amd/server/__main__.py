@@ -61,10 +61,12 @@ def build_engine_config(args):
     kvt = KVTransferConfig()
     if args.kv_transfer_config:
         raw = json.loads(args.kv_transfer_config)
+        role = raw.get("kv_role")
         kvt = KVTransferConfig(
             kv_connector=raw.get("kv_connector"),
-            kv_role=raw.get("kv_role"),
-            kv_rank=int(raw.get("kv_rank", 0)),
+            kv_role=role,
+            # convention: prefiller is rank 0 of the PD pair, decoder rank 1
+            kv_rank=int(raw.get("kv_rank", 0 if role == "kv_producer" else 1)),
             kv_world_size=int(raw.get("kv_world_size", 2)),
         )
     mc = get_model_config(args.model)
@@ -115,7 +117,12 @@ def _rank_main(local_rank: int, args, nproc: int):
         engine = LLMEngine(cfg, device=device)
         engine.worker_loop()
         return
-    serving = ServingEngine(cfg, device=device)
+    kv_connector = None
+    if cfg.kv_transfer.kv_connector:
+        from fusioninfer_amd.pd import build_pd_connector
+
+        kv_connector = build_pd_connector(cfg.kv_transfer, device)
+    serving = ServingEngine(cfg, device=device, kv_connector=kv_connector)
     if args.enable_lora or args.lora_modules:
         for spec in args.lora_modules:
             name, _, r = spec.partition("=")

@@ -127,6 +127,24 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             if finished:
                 return
 
+    @app.post("/pd/prefill")
+    async def pd_prefill(request: Request):
+        """PD producer: prefill the prompt, ship its KV to the decoder, and
+        return the pd_tag the router forwards to the decode endpoint
+        (x-pd-tag header) plus the prefiller-sampled first token."""
+        if serving.kv_connector is None:
+            return JSONResponse(
+                {"error": {"message": "server is not a PD prefiller",
+                           "type": "invalid_request_error"}}, 400)
+        body = await request.json()
+        prompt_ids = encode_prompt(body.get("prompt", ""), vocab)
+        loop = asyncio.get_event_loop()
+        tag, first_token = await loop.run_in_executor(
+            None, serving.prefill_via_pd, prompt_ids
+        )
+        return {"pd_tag": tag, "first_token": first_token,
+                "prompt_tokens": len(prompt_ids)}
+
     @app.post("/v1/completions")
     async def completions(request: Request):
         body = await request.json()
@@ -137,11 +155,23 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         lora = body.get("model")
         if lora not in serving.engine.active_loras():
             lora = None
+        pd_tag = request.headers.get("x-pd-tag")
         try:
-            req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
+            if pd_tag is not None:
+                # PD consumer: claim the KV the prefiller shipped under this
+                # tag; the request joins the decode loop directly
+                loop = asyncio.get_event_loop()
+                req_id, q = await loop.run_in_executor(
+                    None, serving.submit_imported, int(pd_tag), sampling
+                )
+            else:
+                req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
         except ValueError as e:
             return JSONResponse({"error": {"message": str(e),
                                            "type": "invalid_request_error"}}, 400)
+        except TimeoutError as e:
+            return JSONResponse({"error": {"message": str(e),
+                                           "type": "server_error"}}, 504)
         created = int(time.time())
         cid = f"cmpl-{uuid.uuid4().hex[:16]}"
 

@@ -7,17 +7,24 @@ per-request token deltas through thread-safe queues.
 
 from __future__ import annotations
 
+import itertools
 import queue
 import threading
 from typing import Dict, List, Optional, Tuple
 
 from fusioninfer_amd.config import EngineConfig
+from fusioninfer_amd.distributed.kv_transfer import KV_CONSUMER, KV_PRODUCER
 from fusioninfer_amd.engine.llm_engine import LLMEngine
 from fusioninfer_amd.engine.sequence import SamplingParams
 
 
 class ServingEngine:
-    def __init__(self, cfg: EngineConfig, device: Optional[str] = None):
+    def __init__(
+        self,
+        cfg: EngineConfig,
+        device: Optional[str] = None,
+        kv_connector=None,
+    ):
         self.engine = LLMEngine(cfg, device=device)
         self._streams: Dict[str, "queue.Queue[Tuple[Optional[int], bool]]"] = {}
         self._lock = threading.Lock()
@@ -25,6 +32,14 @@ class ServingEngine:
         self._stop = False
         self.healthy = True
         self.last_error: str = ""
+        # PD disaggregation: producer prefills+ships KV,
+        # consumer receives KV on a background thread and admits on claim
+        self.kv_connector = kv_connector
+        self._pd_tags = itertools.count(1)
+        self._pending_imports: Dict[int, tuple] = {}
+        self._pd_cond = threading.Condition()
+        if kv_connector is not None and kv_connector.role == KV_CONSUMER:
+            threading.Thread(target=self._pd_recv_loop, daemon=True).start()
         self._thread = threading.Thread(target=self._loop, daemon=True)
         self._thread.start()
 
@@ -39,6 +54,83 @@ class ServingEngine:
                 prompt_token_ids, sampling, lora_name=lora_name
             )
             self._streams[req_id] = q
+        self._work.set()
+        return req_id, q
+
+    # ---------------------------------------------------- PD disaggregation
+    def prefill_via_pd(self, prompt_token_ids: List[int]) -> Tuple[int, int]:
+        """Producer role: prefill a prompt, ship its KV to the decoder, and
+        return (pd_tag, first_token). The tag travels in the KV header and
+        back to the router, which hands it to the decode endpoint —
+        out-of-order decode claims still match their own KV."""
+        assert self.kv_connector is not None and \
+            self.kv_connector.role == KV_PRODUCER, "not a PD prefiller"
+        with self._lock:
+            req_id, first_token, block_ids = self.engine.prefill_export(
+                prompt_token_ids
+            )
+            tag = next(self._pd_tags)
+            self.kv_connector.send_kv(
+                self.engine.runner.kv_caches,
+                block_ids,
+                len(prompt_token_ids),
+                first_token,
+                tag,
+            )
+            self.engine.release_held(req_id)
+        return tag, first_token
+
+    def _pd_recv_loop(self):
+        def alloc(n):
+            with self._lock:
+                return self.engine.allocate_import_blocks(n)
+
+        while not self._stop:
+            try:
+                kwargs = {}
+                if hasattr(self.kv_connector, "channel"):  # in-memory: poll
+                    kwargs["timeout"] = 0.25
+                _, prompt_len, first_token, tag = self.kv_connector.recv_kv(
+                    self.engine.runner.kv_caches, alloc, **kwargs
+                )
+            except queue.Empty:
+                continue
+            except Exception as e:
+                if self._stop:
+                    return
+                self.healthy = False
+                self.last_error = repr(e)
+                return
+            with self._lock:
+                holder = self.engine.take_import_holder()
+            with self._pd_cond:
+                self._pending_imports[tag] = (holder, prompt_len, first_token)
+                self._pd_cond.notify_all()
+
+    def submit_imported(
+        self, pd_tag: int, sampling: SamplingParams, timeout: float = 60.0
+    ) -> Tuple[str, "queue.Queue[Tuple[Optional[int], bool]]"]:
+        """Consumer role: claim the KV batch tagged `pd_tag` (waiting for the
+        connector to deliver it if needed) and admit the request straight
+        into the decode loop."""
+        with self._pd_cond:
+            while pd_tag not in self._pending_imports:
+                if not self._pd_cond.wait(timeout=timeout):
+                    raise TimeoutError(f"KV for pd_tag={pd_tag} never arrived")
+            holder, prompt_len, first_token = self._pending_imports.pop(pd_tag)
+        q: "queue.Queue[Tuple[Optional[int], bool]]" = queue.Queue()
+        with self._lock:
+            req_id = self.engine.add_imported_request(
+                prompt_len, first_token, sampling, holder=holder
+            )
+            if sampling.max_tokens <= 1:
+                # first (prefiller-sampled) token already satisfies the
+                # budget; finish without a decode step
+                self.engine.abort_request(req_id)
+                q.put((first_token, True))
+                return req_id, q
+            self._streams[req_id] = q
+        q.put((first_token, False))
         self._work.set()
         return req_id, q
 

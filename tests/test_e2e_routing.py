@@ -114,6 +114,133 @@ def test_router_streaming_relay(backends):
     asyncio.run(run())
 
 
+def test_router_pd_disaggregation_e2e():
+    """Full PD flow over HTTP: router picks prefill+decode
+    endpoints via the pd-profile-handler config the control plane renders,
+    prefills on the prefiller (KV ships through the connector), decodes on
+    the decoder — and the tokens match a monolithic engine with the same
+    weights."""
+    import torch
+
+    from fusioninfer_amd.distributed.kv_transfer import make_inmemory_pair
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from tests.test_controlplane import pd_svc
+
+    def make_cfg():
+        return EngineConfig(
+            model=get_model_config("tiny-qwen3"),
+            cache=CacheConfig(num_gpu_blocks=128),
+            scheduler=SchedulerConfig(
+                max_num_seqs=8, max_num_batched_tokens=512, max_model_len=256
+            ),
+        )
+
+    prod, cons = make_inmemory_pair()
+    # identical random-init weights on every engine (same seed)
+    torch.manual_seed(11)
+    pre = ServingEngine(make_cfg(), device="cpu", kv_connector=prod)
+    torch.manual_seed(11)
+    dec = ServingEngine(make_cfg(), device="cpu", kv_connector=cons)
+    torch.manual_seed(11)
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+
+    mono = LLMEngine(make_cfg(), device="cpu")
+
+    apps = {
+        "prefiller:8000": build_app(pre, "tiny-qwen3"),
+        "decoder:8000": build_app(dec, "tiny-qwen3"),
+    }
+    svc = pd_svc()
+    router_role = next(r for r in svc.roles if r.component_type == cp_api.ROUTER)
+    picker = EndpointPicker(cp_router.generate_epp_config(svc, router_role))
+    assert picker.is_pd
+    endpoints = [
+        Endpoint("prefiller:8000",
+                 labels={"fusioninfer.io/component-type": "prefiller"}),
+        Endpoint("decoder:8000",
+                 labels={"fusioninfer.io/component-type": "decoder"}),
+    ]
+    client = httpx.AsyncClient(transport=MultiASGITransport(apps))
+    app = build_router_app(picker, endpoints, client=client, scrape=False)
+
+    prompt = [7, 11, 13, 17, 19] * 6
+    expected = mono.generate(
+        [prompt], SamplingParams(max_tokens=4, temperature=0.0, ignore_eos=True)
+    )[0].output_token_ids
+
+    async def run():
+        async with httpx.AsyncClient(
+            transport=httpx.ASGITransport(app=app), base_url="http://router"
+        ) as c:
+            r = await c.post("/v1/completions", json={
+                "prompt": prompt, "max_tokens": 4,
+                "temperature": 0, "ignore_eos": True,
+            })
+            assert r.status_code == 200, r.text
+            toks = r.json()["choices"][0]["token_ids"]
+            assert len(toks) == 4
+            assert toks == expected, (toks, expected)
+            # streaming decode through the same PD path
+            n = 0
+            async with c.stream("POST", "/v1/completions", json={
+                "prompt": prompt, "max_tokens": 3, "stream": True,
+                "temperature": 0, "ignore_eos": True,
+            }) as r2:
+                async for line in r2.aiter_lines():
+                    if line.startswith("data: ") and "[DONE]" not in line:
+                        n += 1
+            assert n == 3
+
+    try:
+        asyncio.run(run())
+        # division of labor: decoder never prefilled, prefiller never decoded
+        # (second identical prompt may recompute fewer tokens via prefix cache)
+        assert pre.engine.num_prefilled_tokens >= len(prompt)
+        assert dec.engine.num_prefilled_tokens == 0
+        assert pre.metrics()["generation_tokens_total"] == 2  # 1 per prefill
+        # all blocks released on both sides after completion
+        assert pre.engine.gpu_cache_usage() == 0.0
+        assert dec.engine.gpu_cache_usage() == 0.0
+    finally:
+        pre.shutdown()
+        dec.shutdown()
+
+
+def test_pd_decode_max_tokens_one():
+    """max_tokens=1 PD request finishes with the prefiller-sampled token and
+    frees its imported blocks without a decode step."""
+    import torch
+
+    from fusioninfer_amd.distributed.kv_transfer import make_inmemory_pair
+    from fusioninfer_amd.engine.sequence import SamplingParams
+
+    def make_cfg():
+        return EngineConfig(
+            model=get_model_config("tiny-qwen3"),
+            cache=CacheConfig(num_gpu_blocks=64),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=256, max_model_len=128
+            ),
+        )
+
+    prod, cons = make_inmemory_pair()
+    torch.manual_seed(3)
+    pre = ServingEngine(make_cfg(), device="cpu", kv_connector=prod)
+    torch.manual_seed(3)
+    dec = ServingEngine(make_cfg(), device="cpu", kv_connector=cons)
+    try:
+        tag, first = pre.prefill_via_pd([5, 6, 7] * 8)
+        _, q = dec.submit_imported(
+            tag, SamplingParams(max_tokens=1, temperature=0.0)
+        )
+        tok, finished = q.get(timeout=10)
+        assert tok == first and finished
+        assert dec.engine.gpu_cache_usage() == 0.0
+    finally:
+        pre.shutdown()
+        dec.shutdown()
+
+
 def test_metrics_scrape_parsing():
     text = (
         '# TYPE vllm:gpu_cache_usage_perc gauge\n'
