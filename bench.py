@@ -41,6 +41,8 @@ def parse_args():
     p.add_argument("--concurrency", type=int, default=256)
     p.add_argument("--max-batched-tokens", type=int, default=8192)
     p.add_argument("--enforce-eager", action="store_true")
+    p.add_argument("--kv-cache-dtype", choices=["auto", "fp8"], default="auto",
+                   help="fp8 = e4m3 KV cache (halves KV bytes)")
     p.add_argument("--quantization", choices=["fp8"], default=None,
                    help="opt-in fp8 serving mode (headline stays bf16)")
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree")
@@ -141,6 +143,7 @@ def main():
         cache=CacheConfig(
             num_gpu_blocks=None if use_cuda else 4096,
             gpu_memory_utilization=0.85,
+            kv_cache_dtype=args.kv_cache_dtype,
         ),
         scheduler=SchedulerConfig(
             max_num_seqs=args.concurrency,
@@ -250,7 +253,8 @@ def _aggregate_and_report(args, rank, world, distributed, device, use_cuda,
         "higher_is_better": True,
         "scaling": "weak",
         "vs_baseline": None,
-        "dtype": args.quantization or "bf16",
+        "dtype": (args.quantization or "bf16")
+        + ("+kv_fp8" if args.kv_cache_dtype == "fp8" else ""),
         "data": "synthetic",
         "p50_ttft_ms": round(p50_ttft_ms, 1),
         "tokens_per_s": round(gen_tokens / elapsed, 1) if elapsed else 0.0,

@@ -63,7 +63,8 @@ class ModelRunner:
         import fusioninfer_amd.distributed.parallel_state as ps
 
         kv_heads = mc.num_kv_heads // ps.tp_world_size()
-        return mc.num_layers * 2 * kv_heads * self.block_size * mc.head_dim * 2
+        esize = 1 if self.cfg.cache.kv_cache_dtype == "fp8" else 2
+        return mc.num_layers * 2 * kv_heads * self.block_size * mc.head_dim * esize
 
     def profile_num_blocks(self) -> int:
         import fusioninfer_amd.distributed.parallel_state as ps
@@ -90,10 +91,15 @@ class ModelRunner:
         self.num_gpu_blocks = self.profile_num_blocks()
         kv_heads = mc.num_kv_heads // ps.tp_world_size()
         shape = (self.num_gpu_blocks, kv_heads, self.block_size, mc.head_dim)
+        kv_dtype = (
+            torch.float8_e4m3fn
+            if self.cfg.cache.kv_cache_dtype == "fp8"
+            else torch.bfloat16
+        )
         self.kv_caches = [
             (
-                torch.zeros(shape, dtype=torch.bfloat16, device=self.device),
-                torch.zeros(shape, dtype=torch.bfloat16, device=self.device),
+                torch.zeros(shape, dtype=kv_dtype, device=self.device),
+                torch.zeros(shape, dtype=kv_dtype, device=self.device),
             )
             for _ in range(mc.num_layers)
         ]

@@ -11,11 +11,15 @@
 
 namespace fi {
 
+// FP8: the cache stores OCP e4m3 at scale 1.0 (K rows are qk-normed and
+// O(1); saturation at |448| is the only loss) — halves decode KV bytes
+// and doubles cache capacity. Conversion uses v_cvt_pk_fp8_f32.
+template <bool FP8>
 __global__ void reshape_and_cache_kernel(
-    const u16* __restrict__ k,   // [T] rows of Hk*D, stride k_stride
+    const u16* __restrict__ k,   // [T] rows of Hk*D, stride k_stride (bf16)
     const u16* __restrict__ v,
-    u16* __restrict__ k_cache,   // [B, Hk, bs, D]
-    u16* __restrict__ v_cache,
+    void* __restrict__ k_cache,  // [B, Hk, bs, D] bf16 (FP8: e4m3 bytes)
+    void* __restrict__ v_cache,
     const int* __restrict__ slot_mapping,  // [T]
     const int64_t k_stride, const int64_t v_stride,
     const int num_tokens, const int kv_heads, const int block_size,
@@ -35,24 +39,48 @@ __global__ void reshape_and_cache_kernel(
     const int64_t dst =
         ((static_cast<int64_t>(blk) * kv_heads + h) * block_size + off) *
             head_dim + d;
-    *reinterpret_cast<bf16x8*>(k_cache + dst) =
-        *reinterpret_cast<const bf16x8*>(k + t * k_stride + c);
-    *reinterpret_cast<bf16x8*>(v_cache + dst) =
-        *reinterpret_cast<const bf16x8*>(v + t * v_stride + c);
+    const bf16x8 kv = *reinterpret_cast<const bf16x8*>(k + t * k_stride + c);
+    const bf16x8 vv = *reinterpret_cast<const bf16x8*>(v + t * v_stride + c);
+    if (FP8) {
+      float kf[8], vf[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        kf[j] = bf16_to_f32(kv.h[j]);
+        vf[j] = bf16_to_f32(vv.h[j]);
+      }
+      u32 kp[2], vp[2];
+      pack_fp8x8(kf, kp);
+      pack_fp8x8(vf, vp);
+      *reinterpret_cast<uint2*>(static_cast<unsigned char*>(k_cache) + dst) =
+          make_uint2(kp[0], kp[1]);
+      *reinterpret_cast<uint2*>(static_cast<unsigned char*>(v_cache) + dst) =
+          make_uint2(vp[0], vp[1]);
+    } else {
+      *reinterpret_cast<bf16x8*>(static_cast<u16*>(k_cache) + dst) = kv;
+      *reinterpret_cast<bf16x8*>(static_cast<u16*>(v_cache) + dst) = vv;
+    }
   }
 }
 
-void launch_reshape_and_cache(const u16* k, const u16* v, u16* k_cache,
-                              u16* v_cache, const int* slot_mapping,
+void launch_reshape_and_cache(const u16* k, const u16* v, void* k_cache,
+                              void* v_cache, const int* slot_mapping,
                               int64_t k_stride, int64_t v_stride, int tokens,
                               int kv_heads, int block_size, int head_dim,
-                              hipStream_t stream) {
+                              bool fp8, hipStream_t stream) {
   const int64_t total = static_cast<int64_t>(tokens) * kv_heads * head_dim / 8;
   const int block = 256;
   const int grid = static_cast<int>(std::min<int64_t>((total + block - 1) / block, (int64_t)2048));
-  hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(grid), dim3(block), 0,
-                     stream, k, v, k_cache, v_cache, slot_mapping, k_stride,
-                     v_stride, tokens, kv_heads, block_size, head_dim);
+  if (fp8) {
+    hipLaunchKernelGGL((reshape_and_cache_kernel<true>), dim3(grid),
+                       dim3(block), 0, stream, k, v, k_cache, v_cache,
+                       slot_mapping, k_stride, v_stride, tokens, kv_heads,
+                       block_size, head_dim);
+  } else {
+    hipLaunchKernelGGL((reshape_and_cache_kernel<false>), dim3(grid),
+                       dim3(block), 0, stream, k, v, k_cache, v_cache,
+                       slot_mapping, k_stride, v_stride, tokens, kv_heads,
+                       block_size, head_dim);
+  }
 }
 
 // staging: [2, n, Hk, bs, D]; GATHER ? cache->staging : staging->cache.

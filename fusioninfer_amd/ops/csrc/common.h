@@ -94,6 +94,21 @@ FI_DEV void pack_fp8x8(const float* v, unsigned int out[2]) {
 #endif
 }
 
+// unpack 4 OCP-e4m3 bytes (one u32) -> 4 f32 via v_cvt_pk_f32_fp8
+FI_DEV void unpack_fp8x4(u32 w, float out[4]) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  typedef __attribute__((ext_vector_type(2))) float f32x2;
+  const f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+  const f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+  out[0] = lo[0];
+  out[1] = lo[1];
+  out[2] = hi[0];
+  out[3] = hi[1];
+#else
+  out[0] = out[1] = out[2] = out[3] = 0.f;
+#endif
+}
+
 template <int NWAVES>
 FI_DEV float block_reduce_max(float wave_val, float* scratch) {
   const int wave = threadIdx.x / kWaveSize;

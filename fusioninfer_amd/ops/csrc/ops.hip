@@ -19,20 +19,20 @@ void launch_row_quant_fp8(unsigned char*, float*, const u16*, int, int,
 void launch_rope_qk_norm(u16*, u16*, int64_t, int64_t, const u16*, const u16*,
                          const float*, const int*, int, int, int, int, float,
                          hipStream_t);
-void launch_reshape_and_cache(const u16*, const u16*, u16*, u16*, const int*,
-                              int64_t, int64_t, int, int, int, int,
-                              hipStream_t);
+void launch_reshape_and_cache(const u16*, const u16*, void*, void*,
+                              const int*, int64_t, int64_t, int, int, int,
+                              int, bool, hipStream_t);
 template <bool GATHER>
 void launch_kv_block_copy(u16*, u16*, u16*, const int*, int, int64_t,
                           hipStream_t);
-void launch_paged_attn_decode(u16*, float*, float*, const u16*, const u16*,
-                              const u16*, const int*, const int*, int,
-                              int64_t, int, int, int, int, int, float,
+void launch_paged_attn_decode(u16*, float*, float*, const u16*, const void*,
+                              const void*, const int*, const int*, int,
+                              int64_t, int, int, int, int, int, float, bool,
                               hipStream_t);
-void launch_prefill_attn(u16*, const u16*, const u16*, const u16*, const int*,
-                         const int*, const int*, const int*, const int*, int,
-                         int, int64_t, int64_t, int64_t, int, int, int, float,
-                         hipStream_t);
+void launch_prefill_attn(u16*, const u16*, const void*, const void*,
+                         const int*, const int*, const int*, const int*,
+                         const int*, int, int, int64_t, int64_t, int64_t, int,
+                         int, int, float, bool, hipStream_t);
 
 }  // namespace fi
 
@@ -51,6 +51,14 @@ const u16* bf16_cptr(const at::Tensor& t) {
 
 hipStream_t current_stream() {
   return c10::hip::getCurrentHIPStream().stream();
+}
+
+// KV caches are bf16 or fp8 (OCP e4m3, scale 1.0)
+bool cache_is_fp8(const at::Tensor& t) {
+  TORCH_CHECK(
+      t.scalar_type() == at::kBFloat16 || t.scalar_type() == at::kFloat8_e4m3fn,
+      "kv cache must be bf16 or float8_e4m3fn");
+  return t.scalar_type() == at::kFloat8_e4m3fn;
 }
 
 void rms_norm(at::Tensor out, at::Tensor input, at::Tensor weight, double eps) {
@@ -175,8 +183,7 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor slot_mapping) {
   CHECK_BF16_CUDA(k);
   CHECK_BF16_CUDA(v);
-  CHECK_BF16_CUDA(k_cache);
-  CHECK_BF16_CUDA(v_cache);
+  const bool fp8 = cache_is_fp8(k_cache);
   TORCH_CHECK(slot_mapping.scalar_type() == at::kInt);
   TORCH_CHECK(k.dim() == 2 && v.dim() == 2);  // [T, Hk*D] (maybe strided rows)
   TORCH_CHECK(k.stride(1) == 1 && v.stride(1) == 1);
@@ -184,33 +191,40 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
   const int block_size = k_cache.size(2);
   const int head_dim = k_cache.size(3);
   fi::launch_reshape_and_cache(
-      bf16_cptr(k), bf16_cptr(v), bf16_ptr(k_cache), bf16_ptr(v_cache),
+      bf16_cptr(k), bf16_cptr(v), k_cache.data_ptr(), v_cache.data_ptr(),
       slot_mapping.data_ptr<int>(), k.stride(0), v.stride(0), k.size(0),
-      kv_heads, block_size, head_dim, current_stream());
+      kv_heads, block_size, head_dim, fp8, current_stream());
 }
 
 void gather_kv_blocks(at::Tensor staging, at::Tensor k_cache,
                       at::Tensor v_cache, at::Tensor block_ids) {
-  CHECK_BF16_CUDA(staging);
+  TORCH_CHECK(staging.is_cuda() &&
+              staging.scalar_type() == k_cache.scalar_type());
   TORCH_CHECK(block_ids.scalar_type() == at::kInt);
   const int n = block_ids.size(0);
-  const int64_t block_elems =
-      k_cache.size(1) * k_cache.size(2) * k_cache.size(3);
-  fi::launch_kv_block_copy<true>(bf16_ptr(staging), bf16_ptr(k_cache),
-                                 bf16_ptr(v_cache), block_ids.data_ptr<int>(),
-                                 n, block_elems, current_stream());
+  // raw byte mover in u16 units (fp8 blocks are half the bytes)
+  const int64_t block_elems = k_cache.size(1) * k_cache.size(2) *
+      k_cache.size(3) * k_cache.element_size() / 2;
+  fi::launch_kv_block_copy<true>(
+      static_cast<fi::u16*>(staging.data_ptr()),
+      static_cast<fi::u16*>(k_cache.data_ptr()),
+      static_cast<fi::u16*>(v_cache.data_ptr()), block_ids.data_ptr<int>(), n,
+      block_elems, current_stream());
 }
 
 void scatter_kv_blocks(at::Tensor staging, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor block_ids) {
-  CHECK_BF16_CUDA(staging);
+  TORCH_CHECK(staging.is_cuda() &&
+              staging.scalar_type() == k_cache.scalar_type());
   TORCH_CHECK(block_ids.scalar_type() == at::kInt);
   const int n = block_ids.size(0);
-  const int64_t block_elems =
-      k_cache.size(1) * k_cache.size(2) * k_cache.size(3);
-  fi::launch_kv_block_copy<false>(bf16_ptr(staging), bf16_ptr(k_cache),
-                                  bf16_ptr(v_cache), block_ids.data_ptr<int>(),
-                                  n, block_elems, current_stream());
+  const int64_t block_elems = k_cache.size(1) * k_cache.size(2) *
+      k_cache.size(3) * k_cache.element_size() / 2;
+  fi::launch_kv_block_copy<false>(
+      static_cast<fi::u16*>(staging.data_ptr()),
+      static_cast<fi::u16*>(k_cache.data_ptr()),
+      static_cast<fi::u16*>(v_cache.data_ptr()), block_ids.data_ptr<int>(), n,
+      block_elems, current_stream());
 }
 
 void paged_attention_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
@@ -221,6 +235,7 @@ void paged_attention_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                             int64_t num_parts, double scale) {
   CHECK_BF16_CUDA(out);
   CHECK_BF16_CUDA(q);
+  const bool fp8 = cache_is_fp8(k_cache);
   TORCH_CHECK(block_tables.scalar_type() == at::kInt &&
               block_tables.is_contiguous());
   TORCH_CHECK(seq_lens.scalar_type() == at::kInt);
@@ -239,11 +254,11 @@ void paged_attention_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
     acc = acc_ws->data_ptr<float>();
   }
   fi::launch_paged_attn_decode(
-      bf16_ptr(out), ml, acc, bf16_cptr(q), bf16_cptr(k_cache),
-      bf16_cptr(v_cache), block_tables.data_ptr<int>(),
+      bf16_ptr(out), ml, acc, bf16_cptr(q), k_cache.data_ptr(),
+      v_cache.data_ptr(), block_tables.data_ptr<int>(),
       seq_lens.data_ptr<int>(), num_seqs, q.stride(0), block_tables.size(1),
       num_kv_heads, head_dim, num_heads / num_kv_heads,
-      static_cast<int>(num_parts), static_cast<float>(scale),
+      static_cast<int>(num_parts), static_cast<float>(scale), fp8,
       current_stream());
 }
 
@@ -267,7 +282,7 @@ void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
       tile_seq.data_ptr<int>(), tile_row0.data_ptr<int>(),
       cu_seqlens.data_ptr<int>(), nullptr, nullptr, 0, tile_seq.size(0),
       q.stride(0), k.stride(0), v.stride(0), num_q_heads, num_kv_heads,
-      head_dim, static_cast<float>(scale), current_stream());
+      head_dim, static_cast<float>(scale), false, current_stream());
 }
 
 void prefill_attention_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
@@ -277,7 +292,7 @@ void prefill_attention_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                              double scale) {
   CHECK_BF16_CUDA(out);
   CHECK_BF16_CUDA(q);
-  CHECK_BF16_CUDA(k_cache);
+  const bool fp8 = cache_is_fp8(k_cache);
   TORCH_CHECK(tile_seq.scalar_type() == at::kInt &&
               tile_row0.scalar_type() == at::kInt &&
               cu_seqlens.scalar_type() == at::kInt &&
@@ -291,12 +306,12 @@ void prefill_attention_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
   TORCH_CHECK(k_cache.size(2) == 16, "cache block_size must be 16");
   TORCH_CHECK(q.stride(2) == 1 && q.stride(1) == head_dim);
   fi::launch_prefill_attn(
-      bf16_ptr(out), bf16_cptr(q), bf16_cptr(k_cache), bf16_cptr(v_cache),
+      bf16_ptr(out), bf16_cptr(q), k_cache.data_ptr(), v_cache.data_ptr(),
       tile_seq.data_ptr<int>(), tile_row0.data_ptr<int>(),
       cu_seqlens.data_ptr<int>(), block_tables.data_ptr<int>(),
       seq_lens_k.data_ptr<int>(), block_tables.size(1), tile_seq.size(0),
       q.stride(0), 0, 0, num_q_heads, num_kv_heads, head_dim,
-      static_cast<float>(scale), current_stream());
+      static_cast<float>(scale), fp8, current_stream());
 }
 
 }  // namespace

@@ -25,6 +25,8 @@
 
 #include "common.h"
 
+#include <type_traits>
+
 namespace fi {
 
 constexpr int kBlockSz = 16;     // cache block size (tokens)
@@ -34,18 +36,23 @@ constexpr float kNegInf = -1e30f;
 // kNWaves is a launch-time choice: 8 waves when the grid underfills the
 // chip (small decode batches: measured +46% at batch<=8), 4 when it is
 // full (4-wave measured ~5% faster there).
-template <int D, int G, int kNWaves>
+// FP8: the cache holds OCP e4m3 at scale 1.0 — half the KV bytes of bf16
+// (this kernel is KV-bandwidth-bound); v_cvt_pk_f32_fp8 dequant on load.
+template <int D, int G, int kNWaves, bool FP8>
 __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     u16* __restrict__ out,            // [S, Hq, D] (written when 1 partition)
     float* __restrict__ ml_ws,        // [S, Hq, P, 2] (multi-partition)
     float* __restrict__ acc_ws,       // [S, Hq, P, D]
     const u16* __restrict__ q,        // [S] rows, stride q_stride, Hq*D elems
-    const u16* __restrict__ k_cache,  // [B, Hk, 16, D]
-    const u16* __restrict__ v_cache,
+    const void* __restrict__ k_cache_p,  // [B, Hk, 16, D] bf16 (FP8: e4m3)
+    const void* __restrict__ v_cache_p,
     const int* __restrict__ block_tables,  // [S, max_blocks]
     const int* __restrict__ seq_lens,      // [S]
     const int64_t q_stride, const int max_blocks, const int num_kv_heads,
     const float scale) {
+  using CT = typename std::conditional<FP8, unsigned char, u16>::type;
+  const CT* k_cache = static_cast<const CT*>(k_cache_p);
+  const CT* v_cache = static_cast<const CT*>(v_cache_p);
   const int seq = blockIdx.x;
   const int kv_head = blockIdx.y;
   const int part = blockIdx.z;
@@ -99,7 +106,7 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   const int tok = lane / 4;          // phase-A token within chunk
   const int quad = lane % 4;         // phase-A dim quarter
   constexpr int DPQ = D / 4;         // dims per phase-A lane
-  constexpr int KQ4 = DPQ / 8;       // 16-B K loads per lane per chunk
+  constexpr int KQ4 = DPQ * sizeof(CT) / 16;  // 16-B K loads per lane per chunk
 
   // software-prefetch: this wave's NEXT chunk's K flies while the current
   // chunk's softmax + PV run (the phases were serialized on K latency)
@@ -110,10 +117,10 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   };
   uint4 kraw[KQ4];
   auto load_k = [&](int64_t kv_base, uint4* dst) {
-    const u16* k_row = k_cache + kv_base + tok * D + quad * DPQ;
+    const CT* k_row = k_cache + kv_base + tok * D + quad * DPQ;
 #pragma unroll
     for (int j8 = 0; j8 < KQ4; ++j8)
-      dst[j8] = *reinterpret_cast<const uint4*>(k_row + j8 * 8);
+      dst[j8] = reinterpret_cast<const uint4*>(k_row)[j8];
   };
   int64_t kv_base = 0;
   if (chunk_lo + wave < chunk_hi) {
@@ -129,11 +136,17 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     float s[G];
     {
       float kf[DPQ];
+      if (FP8) {
+        const u32* kw = reinterpret_cast<const u32*>(kraw);
 #pragma unroll
-      for (int j8 = 0; j8 < KQ4; ++j8) {
-        const bf16x8 kv8 = __builtin_bit_cast(bf16x8, kraw[j8]);
+        for (int w = 0; w < DPQ / 4; ++w) unpack_fp8x4(kw[w], &kf[w * 4]);
+      } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) kf[j8 * 8 + j] = bf16_to_f32(kv8.h[j]);
+        for (int j8 = 0; j8 < KQ4; ++j8) {
+          const bf16x8 kv8 = __builtin_bit_cast(bf16x8, kraw[j8]);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) kf[j8 * 8 + j] = bf16_to_f32(kv8.h[j]);
+        }
       }
 #pragma unroll
       for (int g = 0; g < G; ++g) {
@@ -191,7 +204,7 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
 
     // ---- phase B: PV accumulate; lane covers dims {2l, 2l+1} ----
     const int nvalid = min(ctx - chunk * kBlockSz, kBlockSz);
-    const u16* v_rows = v_cache + kv_base_cur;
+    const CT* v_rows = v_cache + kv_base_cur;
 #pragma unroll
     for (int g = 0; g < G; ++g) {
       acc[g][0] *= alpha[g];
@@ -199,9 +212,19 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     }
     if (lane < D / 2) {
       for (int t = 0; t < nvalid; ++t) {
-        const u32 vbits = *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
-        const float v0 = bf16_to_f32(static_cast<u16>(vbits & 0xffff));
-        const float v1 = bf16_to_f32(static_cast<u16>(vbits >> 16));
+        float v0, v1;
+        if (FP8) {
+          const u16 vb = *reinterpret_cast<const u16*>(v_rows + t * D + 2 * lane);
+          float vf[4];
+          unpack_fp8x4(vb, vf);
+          v0 = vf[0];
+          v1 = vf[1];
+        } else {
+          const u32 vbits =
+              *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
+          v0 = bf16_to_f32(static_cast<u16>(vbits & 0xffff));
+          v1 = bf16_to_f32(static_cast<u16>(vbits >> 16));
+        }
 #pragma unroll
         for (int g = 0; g < G; ++g) {
           const float p = p_lds[wave][t][g];
@@ -300,26 +323,28 @@ __global__ __launch_bounds__(kWaveSize) void paged_attn_reduce_kernel(
 }
 
 void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
-                              const u16* q, const u16* k_cache,
-                              const u16* v_cache, const int* block_tables,
+                              const u16* q, const void* k_cache,
+                              const void* v_cache, const int* block_tables,
                               const int* seq_lens, int num_seqs,
                               int64_t q_stride, int max_blocks,
                               int num_kv_heads, int head_dim, int group,
-                              int num_parts, float scale, hipStream_t stream) {
+                              int num_parts, float scale, bool fp8,
+                              hipStream_t stream) {
   const bool wide = num_seqs * num_kv_heads * num_parts < 256;
   const int nwaves = wide ? 8 : 4;
   dim3 grid(num_seqs, num_kv_heads, num_parts), block(nwaves * kWaveSize);
+#define FI_LAUNCH_1(DD, GG, NW, F8)                                           \
+  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG, NW, F8>), grid, block, \
+                     0, stream, out, ml_ws, acc_ws, q, k_cache, v_cache,      \
+                     block_tables, seq_lens, q_stride, max_blocks,            \
+                     num_kv_heads, scale)
 #define FI_LAUNCH(DD, GG)                                                     \
-  if (wide) {                                                                 \
-    hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG, 8>), grid, block, 0, \
-                       stream, out, ml_ws, acc_ws, q, k_cache, v_cache,       \
-                       block_tables, seq_lens, q_stride, max_blocks,          \
-                       num_kv_heads, scale);                                  \
+  if (fp8) {                                                                  \
+    if (wide) { FI_LAUNCH_1(DD, GG, 8, true); }                               \
+    else      { FI_LAUNCH_1(DD, GG, 4, true); }                               \
   } else {                                                                    \
-    hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG, 4>), grid, block, 0, \
-                       stream, out, ml_ws, acc_ws, q, k_cache, v_cache,       \
-                       block_tables, seq_lens, q_stride, max_blocks,          \
-                       num_kv_heads, scale);                                  \
+    if (wide) { FI_LAUNCH_1(DD, GG, 8, false); }                              \
+    else      { FI_LAUNCH_1(DD, GG, 4, false); }                              \
   }                                                                           \
   if (num_parts > 1) {                                                        \
     dim3 rgrid(num_seqs, num_kv_heads * GG), rblock(kWaveSize);               \
@@ -346,6 +371,7 @@ void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
     abort();
   }
 #undef FI_LAUNCH
+#undef FI_LAUNCH_1
 }
 
 }  // namespace fi

@@ -27,6 +27,8 @@
 
 #include "common.h"
 
+#include <type_traits>
+
 namespace fi {
 
 typedef __attribute__((ext_vector_type(8))) short short8;
@@ -60,12 +62,17 @@ FI_DEV u32 pack_bf16x2(float lo, float hi) {
          (static_cast<u32>(f32_to_bf16(hi)) << 16);
 }
 
-template <int D, bool PAGED>
+// FP8 (PAGED only): the cache holds OCP e4m3 at scale 1.0. K cannot use
+// global_load_lds (the LDS tile must stay bf16 for the MFMA fragments), so
+// both K and V go through the V-style register-stage pipeline with a
+// v_cvt_pk_f32_fp8 dequant between load and ds_write — half the HBM bytes
+// per tile for ~8 extra VALU ops per 16 elements.
+template <int D, bool PAGED, bool FP8>
 __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     u16* __restrict__ out,        // [T, Hq, D]
     const u16* __restrict__ q,    // [T] rows, stride q_stride
-    const u16* __restrict__ k,    // dense rows OR k_cache when PAGED
-    const u16* __restrict__ v,    // dense rows OR v_cache when PAGED
+    const void* __restrict__ k_p, // dense rows OR k_cache when PAGED
+    const void* __restrict__ v_p, // dense rows OR v_cache when PAGED
     const int* __restrict__ tile_seq,    // [ntiles] sequence index
     const int* __restrict__ tile_row0,   // [ntiles] first NEW q row
     const int* __restrict__ cu_seqlens,  // [nseqs+1] over NEW tokens
@@ -74,6 +81,9 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     const int max_blocks,
     const int64_t q_stride, const int64_t k_stride, const int64_t v_stride,
     const int num_q_heads, const int num_kv_heads, const float scale) {
+  using CT = typename std::conditional<FP8, unsigned char, u16>::type;
+  const CT* k = static_cast<const CT*>(k_p);
+  const CT* v = static_cast<const CT*>(v_p);
   constexpr int KF = D / 16;           // QK^T 16-deep k-chunks over head dim
   constexpr int CB = D / 32;           // PV output 32-col blocks
   constexpr int kKRowB = D * 2;        // K row bytes (D=128 -> 256)
@@ -157,7 +167,7 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
       constexpr int kMask = (kKRowB / 16 > 16 ? 16 : kKRowB / 16) - 1;
       const int byte = sbyte ^ ((row & kMask) << 4);
       const int src_row = min(kv0 + row, k_len - 1);
-      const u16* gsrc = k + kv_off(src_row, k_stride) + byte / 2;
+      const CT* gsrc = k + kv_off(src_row, k_stride) + byte / sizeof(u16);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)gsrc,
           (__attribute__((address_space(3))) void*)(
@@ -168,13 +178,48 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
   constexpr int kChunks = D / 16;
   const int v_kv = tid % kKVTile;
   const int v_c16 = tid / kKVTile;
+  // 16 fp8 bytes (one uint4) -> two bf16x8
+  auto cvt_fp8x16 = [](uint4 raw, bf16x8& lo, bf16x8& hi) {
+    const u32 w[4] = {raw.x, raw.y, raw.z, raw.w};
+    float f[16];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) unpack_fp8x4(w[i], &f[i * 4]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      lo.h[j] = f32_to_bf16(f[j]);
+      hi.h[j] = f32_to_bf16(f[8 + j]);
+    }
+  };
   bf16x8 vreg0, vreg1;
   auto vload = [&](int t) {
     if (v_c16 >= kChunks) return;
     const int src = min(t * kKVTile + v_kv, k_len - 1);
-    const u16* vrow = v + kv_off(src, v_stride) + v_c16 * 16;
-    vreg0 = *reinterpret_cast<const bf16x8*>(vrow);
-    vreg1 = *reinterpret_cast<const bf16x8*>(vrow + 8);
+    const CT* vrow = v + kv_off(src, v_stride) + v_c16 * 16;
+    if (FP8) {
+      cvt_fp8x16(*reinterpret_cast<const uint4*>(vrow), vreg0, vreg1);
+    } else {
+      vreg0 = *reinterpret_cast<const bf16x8*>(vrow);
+      vreg1 = *reinterpret_cast<const bf16x8*>(
+          reinterpret_cast<const u16*>(vrow) + 8);
+    }
+  };
+  // FP8 K staging: same (row, 16-col chunk) ownership as V, but written
+  // UNtransposed into the K tile layout ([64 rows][D] bf16, row swizzle);
+  // each converted 16-elem group is one 16-B-aligned ds_write_b128 pair
+  bf16x8 kreg0, kreg1;
+  auto kload_fp8 = [&](int t) {
+    if (v_c16 >= kChunks) return;
+    const int src = min(t * kKVTile + v_kv, k_len - 1);
+    const CT* krow = k + kv_off(src, k_stride) + v_c16 * 16;
+    cvt_fp8x16(*reinterpret_cast<const uint4*>(krow), kreg0, kreg1);
+  };
+  auto kwrite_fp8 = [&](u16* kbuf) {
+    if (v_c16 >= kChunks) return;
+    char* kbase = reinterpret_cast<char*>(kbuf);
+    *reinterpret_cast<bf16x8*>(
+        kbase + swz<kKRowB>(v_kv, (v_c16 * 16) * 2)) = kreg0;
+    *reinterpret_cast<bf16x8*>(
+        kbase + swz<kKRowB>(v_kv, (v_c16 * 16 + 8) * 2)) = kreg1;
   };
   auto vwrite = [&](u16* vt_buf) {
     if (v_c16 >= kChunks) return;
@@ -204,12 +249,24 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
       v_off[kc][cb] = swz<kVTRowB>(cb * 32 + col, kc * 32 + hi1 * 16);
 
   // prologue: K(0) in flight + V(0) staged and published, V(1) in flight
-  stage_k_glds(0, k_lds0);
-  vload(0);
-  __syncthreads();          // drains K(0) glds
-  vwrite(vt_lds0);
-  if (num_kv_tiles > 1) vload(1);
-  __syncthreads();          // publishes vt[0]
+  if (FP8) {
+    kload_fp8(0);
+    vload(0);
+    kwrite_fp8(k_lds0);
+    vwrite(vt_lds0);
+    if (num_kv_tiles > 1) {
+      kload_fp8(1);
+      vload(1);
+    }
+    __syncthreads();        // publishes K(0) + vt[0]
+  } else {
+    stage_k_glds(0, k_lds0);
+    vload(0);
+    __syncthreads();        // drains K(0) glds
+    vwrite(vt_lds0);
+    if (num_kv_tiles > 1) vload(1);
+    __syncthreads();        // publishes vt[0]
+  }
 
   for (int t = 0; t < num_kv_tiles; ++t) {
     const int kv0 = t * kKVTile;
@@ -218,7 +275,7 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
     const u16* vt_cur = (t & 1) ? vt_lds1 : vt_lds0;
     u16* vt_next = (t & 1) ? vt_lds0 : vt_lds1;
 
-    if (t + 1 < num_kv_tiles) stage_k_glds(t + 1, kbuf_next);
+    if (!FP8 && t + 1 < num_kv_tiles) stage_k_glds(t + 1, kbuf_next);
 
     const bool compute = active && kv0 <= my_q_max;
     floatx16 s_acc[2];  // S^T for kv blocks [kv0, +32) and [kv0+32, +64)
@@ -357,10 +414,14 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
       }
       __builtin_amdgcn_s_setprio(0);
     }
-    // stage V(t+1) (regs already resident) and prefetch V(t+2)
+    // stage K/V(t+1) (regs already resident) and prefetch (t+2)
     if (t + 1 < num_kv_tiles) {
+      if (FP8) kwrite_fp8(kbuf_next);
       vwrite(vt_next);
-      if (t + 2 < num_kv_tiles) vload(t + 2);
+      if (t + 2 < num_kv_tiles) {
+        if (FP8) kload_fp8(t + 2);
+        vload(t + 2);
+      }
     }
     // ONE barrier: publishes vt[(t+1)&1]; all waves' PV(t) reads of
     // vt[t&1] complete (it is rewritten only after the NEXT barrier);
@@ -385,24 +446,35 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
   }
 }
 
-void launch_prefill_attn(u16* out, const u16* q, const u16* k, const u16* v,
+void launch_prefill_attn(u16* out, const u16* q, const void* k, const void* v,
                          const int* tile_seq, const int* tile_row0,
                          const int* cu_seqlens, const int* block_tables,
                          const int* seq_lens_k, int max_blocks, int ntiles,
                          int64_t q_stride, int64_t k_stride, int64_t v_stride,
                          int num_q_heads, int num_kv_heads, int head_dim,
-                         float scale, hipStream_t stream) {
+                         float scale, bool fp8, hipStream_t stream) {
   dim3 grid(ntiles, num_q_heads), block(kWaves * kWaveSize);
   const bool paged = block_tables != nullptr;
-#define FI_PF_LAUNCH(DD, PP)                                                  \
-  hipLaunchKernelGGL((prefill_attn_kernel<DD, PP>), grid, block, 0, stream,   \
-                     out, q, k, v, tile_seq, tile_row0, cu_seqlens,           \
+#define FI_PF_LAUNCH(DD, PP, F8)                                              \
+  hipLaunchKernelGGL((prefill_attn_kernel<DD, PP, F8>), grid, block, 0,       \
+                     stream, out, q, k, v, tile_seq, tile_row0, cu_seqlens,   \
                      block_tables, seq_lens_k, max_blocks, q_stride,          \
                      k_stride, v_stride, num_q_heads, num_kv_heads, scale)
+  if (fp8 && !paged) abort();  // fp8 KV is a cache format; dense is bf16
   if (head_dim == 128) {
-    if (paged) FI_PF_LAUNCH(128, true); else FI_PF_LAUNCH(128, false);
+    if (paged) {
+      if (fp8) FI_PF_LAUNCH(128, true, true);
+      else FI_PF_LAUNCH(128, true, false);
+    } else {
+      FI_PF_LAUNCH(128, false, false);
+    }
   } else if (head_dim == 64) {
-    if (paged) FI_PF_LAUNCH(64, true); else FI_PF_LAUNCH(64, false);
+    if (paged) {
+      if (fp8) FI_PF_LAUNCH(64, true, true);
+      else FI_PF_LAUNCH(64, true, false);
+    } else {
+      FI_PF_LAUNCH(64, false, false);
+    }
   } else {
     abort();
   }

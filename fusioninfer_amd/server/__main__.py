@@ -34,6 +34,7 @@ def parse_args(argv=None):
                         '"kv_role": "kv_producer"|"kv_consumer"}')
     p.add_argument("--enforce-eager", action="store_true")
     p.add_argument("--quantization", choices=["fp8"], default=None)
+    p.add_argument("--kv-cache-dtype", choices=["auto", "fp8"], default="auto")
     p.add_argument("--enable-lora", action="store_true")
     p.add_argument("--max-loras", type=int, default=8)
     p.add_argument("--lora-modules", nargs="*", default=[],
@@ -77,6 +78,7 @@ def build_engine_config(args):
         cache=CacheConfig(
             gpu_memory_utilization=args.gpu_memory_utilization,
             enable_prefix_caching=args.enable_prefix_caching,
+            kv_cache_dtype=args.kv_cache_dtype,
         ),
         scheduler=SchedulerConfig(
             max_num_seqs=args.max_num_seqs,

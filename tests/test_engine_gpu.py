@@ -100,6 +100,64 @@ def test_gpu_lora_request_differs_from_base():
     assert outs[req].output_token_ids != base.output_token_ids
 
 
+def test_gpu_fp8_kv_cache_logits_close_to_bf16():
+    """fp8 (e4m3, scale-1) KV cache vs bf16 KV with the SAME weights:
+    prefill logits over a 150-token context must agree to KV quantization
+    error (catches dequant/layout bugs in the fp8 paths of both attention
+    kernels and reshape_and_cache)."""
+    from fusioninfer_amd.engine.block_manager import BlockManager
+    from fusioninfer_amd.engine.sequence import SamplingParams, Sequence
+
+    def logits_for(kv_dtype):
+        torch.manual_seed(0)
+        mc = get_model_config("Qwen3-0.6B")
+        mc.num_layers = 4
+        cfg = EngineConfig(
+            model=mc,
+            cache=CacheConfig(num_gpu_blocks=256, kv_cache_dtype=kv_dtype),
+            scheduler=SchedulerConfig(
+                max_num_seqs=8, max_num_batched_tokens=1024, max_model_len=256
+            ),
+            seed=7,
+            enforce_eager=True,
+        )
+        eng = LLMEngine(cfg, device="cuda:0")
+        bm = BlockManager(eng.runner.num_gpu_blocks, cfg.cache.block_size)
+        seq = Sequence("s", list(range(30, 180)), SamplingParams())
+        bm.allocate(seq)
+        logits_p = eng.runner.execute_prefill([seq], bm).float().cpu()
+        # and one decode step over the fp8 cache
+        forced = int(logits_p.argmax(-1)[0])
+        seq.output_token_ids = [forced]
+        bm.append_slot(seq)
+        logits_d = eng.runner.execute_decode([seq], bm).float().cpu()
+        return logits_p, logits_d
+
+    p_bf16, d_bf16 = logits_for("auto")
+    p_fp8, d_fp8 = logits_for("fp8")
+    rel_p = (p_fp8 - p_bf16).norm() / p_bf16.norm()
+    rel_d = (d_fp8 - d_bf16).norm() / d_bf16.norm()
+    assert rel_p.item() < 0.15, rel_p.item()
+    assert rel_d.item() < 0.15, rel_d.item()
+
+
+def test_gpu_fp8_kv_engine_generates():
+    torch.manual_seed(0)
+    mc = get_model_config("Qwen3-0.6B")
+    mc.num_layers = 4
+    cfg = EngineConfig(
+        model=mc,
+        cache=CacheConfig(num_gpu_blocks=256, kv_cache_dtype="fp8"),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=1024, max_model_len=256
+        ),
+        enforce_eager=False,  # hipGraph decode path over the fp8 cache
+    )
+    eng = LLMEngine(cfg, device="cuda:0")
+    out = eng.generate([[7, 8, 9] * 20], SamplingParams(max_tokens=6))[0]
+    assert len(out.output_token_ids) == 6
+
+
 def test_gpu_fp8_logits_close_to_bf16():
     """fp8 (fused-epilogue) forward vs bf16 forward with the SAME weights:
     logits must agree to fp8 quantization error accumulated over 4 layers.

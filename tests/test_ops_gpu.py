@@ -274,3 +274,122 @@ def test_prefill_attention_paged(ctx_lens, new_lens):
         lens_k.cpu(), scale,
     )
     assert_close_bf16(out.cpu(), expected, atol=3e-2, rtol=3e-2)
+
+
+# ------------------------------------------------------------ fp8 KV cache
+
+def _fp8_cache(t):
+    return t.to(torch.float8_e4m3fn)
+
+
+def test_reshape_and_cache_fp8():
+    T, Hk, D, bs, nblocks = 37, 4, 128, 16, 12
+    k = torch.randn(T, Hk * D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, Hk * D, dtype=torch.bfloat16, device=DEV)
+    k_cache = torch.zeros(nblocks, Hk, bs, D, dtype=torch.float8_e4m3fn,
+                          device=DEV)
+    v_cache = torch.zeros_like(k_cache)
+    slots = torch.randperm(nblocks * bs, device=DEV)[:T].to(torch.int32)
+    k_ref = torch.zeros(nblocks, Hk, bs, D, dtype=torch.float8_e4m3fn)
+    v_ref = torch.zeros_like(k_ref)
+    ref.reshape_and_cache(
+        k.cpu().view(T, Hk, D), v.cpu().view(T, Hk, D), k_ref, v_ref,
+        slots.cpu().long()
+    )
+    ops.reshape_and_cache(k, v, k_cache, v_cache, slots)
+    # hardware cvt vs torch cast can differ 1 ulp on RNE ties -> compare
+    # dequantized with one-quantum tolerance
+    torch.testing.assert_close(
+        k_cache.float().cpu(), k_ref.float(), atol=0.07, rtol=0.07
+    )
+    torch.testing.assert_close(
+        v_cache.float().cpu(), v_ref.float(), atol=0.07, rtol=0.07
+    )
+
+
+@pytest.mark.parametrize("group", [1, 4])
+@pytest.mark.parametrize("seq_lens", [[1], [1, 5, 16, 17, 255, 1023]])
+def test_paged_attention_decode_fp8kv(group, seq_lens):
+    """Decode over an fp8 (e4m3, scale-1) cache: the kernel and the fp32
+    reference read the SAME quantized cache, so tolerances stay tight."""
+    torch.manual_seed(6)
+    Hk, D, bs = 2, 128, 16
+    Hq = Hk * group
+    S = len(seq_lens)
+    max_blocks = (max(seq_lens) + bs - 1) // bs
+    total_blocks = sum((L + bs - 1) // bs for L in seq_lens) + 2
+    q = torch.randn(S, Hq, D, dtype=torch.bfloat16, device=DEV)
+    k_cache = _fp8_cache(
+        torch.randn(total_blocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV))
+    v_cache = _fp8_cache(
+        torch.randn(total_blocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV))
+    bt = torch.zeros(S, max_blocks, dtype=torch.int32, device=DEV)
+    nxt = 1
+    for s, L in enumerate(seq_lens):
+        n = (L + bs - 1) // bs
+        bt[s, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    lens = torch.tensor(seq_lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention_decode(q, k_cache, v_cache, bt, lens, scale)
+    expected = ref.paged_attention_decode(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), lens.cpu(), scale
+    )
+    assert_close_bf16(out.cpu(), expected, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("head_dim", [64, 128])
+@pytest.mark.parametrize(
+    "ctx_lens,new_lens",
+    [([48], [48]), ([64, 100, 300], [64, 36, 44])],
+)
+def test_prefill_attention_paged_fp8kv(head_dim, ctx_lens, new_lens):
+    torch.manual_seed(7)
+    Hq, Hk, bs, D = 8, 2, 16, head_dim
+    S = len(ctx_lens)
+    Tn = sum(new_lens)
+    max_blocks = max((c + bs - 1) // bs for c in ctx_lens)
+    total_blocks = sum((c + bs - 1) // bs for c in ctx_lens) + 1
+    q = torch.randn(Tn, Hq, D, dtype=torch.bfloat16, device=DEV)
+    k_cache = _fp8_cache(
+        torch.randn(total_blocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV))
+    v_cache = _fp8_cache(
+        torch.randn(total_blocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV))
+    bt = torch.zeros(S, max_blocks, dtype=torch.int32, device=DEV)
+    nxt = 1
+    for s, c in enumerate(ctx_lens):
+        n = (c + bs - 1) // bs
+        bt[s, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    cu_q = torch.tensor(
+        [0] + list(torch.tensor(new_lens).cumsum(0)), dtype=torch.int32,
+        device=DEV
+    )
+    lens_k = torch.tensor(ctx_lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.prefill_attention_paged(q, k_cache, v_cache, bt, cu_q, lens_k,
+                                      scale)
+    expected = ref.prefill_attention_paged(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), cu_q.cpu(),
+        lens_k.cpu(), scale
+    )
+    assert_close_bf16(out.cpu(), expected, atol=3e-2, rtol=3e-2)
+
+
+def test_kv_block_copy_fp8_roundtrip():
+    Hk, D, bs, nblocks = 2, 128, 16, 20
+    k_cache = _fp8_cache(
+        torch.randn(nblocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV))
+    v_cache = _fp8_cache(
+        torch.randn(nblocks, Hk, bs, D, dtype=torch.bfloat16, device=DEV))
+    ids = torch.tensor([3, 7, 1, 19], dtype=torch.int32, device=DEV)
+    staging = ops.gather_kv_blocks(k_cache, v_cache, ids)
+    assert staging.dtype == torch.float8_e4m3fn
+    k2 = torch.zeros_like(k_cache)
+    v2 = torch.zeros_like(v_cache)
+    ids2 = torch.tensor([0, 2, 4, 6], dtype=torch.int32, device=DEV)
+    ops.scatter_kv_blocks(staging, k2, v2, ids2)
+    torch.testing.assert_close(
+        k2[ids2.long()].float().cpu(), k_cache[ids.long()].float().cpu())
+    torch.testing.assert_close(
+        v2[ids2.long()].float().cpu(), v_cache[ids.long()].float().cpu())
