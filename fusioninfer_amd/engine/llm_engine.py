@@ -80,21 +80,29 @@ class LLMEngine:
     def add_lora(self, name: str, rank: int = 16, alpha: float = 32.0,
                  seed: Optional[int] = None):
         """Register a LoRA adapter (random-init when seed given, zero-B
-        otherwise — zero-B adapters are numerically the base model)."""
-        from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
+        otherwise — zero-B adapters are numerically the base model).
 
-        # TP>1 would need the adapter registered on every rank before any
-        # batch references it (worker forwards must match the driver's) —
-        # adapter broadcast is round-2 work
-        assert self.cfg.parallel.tensor_parallel_size == 1, (
-            "LoRA with tensor parallelism needs rank-synchronized adapter "
-            "registration (not yet wired)"
-        )
+        TP > 1: the driver broadcasts the registration over the payload
+        channel so every worker holds its shard BEFORE any batch
+        references the adapter by name (adapters are generated from the
+        seed on the CPU identically on all ranks, then sharded by
+        tp_rank — same scheme as the base weights)."""
         # fp8 mode feeds (fp8, scale) tuples through the layers; LoRA's
         # bf16 shrink/expand GEMMs would need the bf16 activations
         assert self.cfg.model.quantization != "fp8", (
             "LoRA is not supported with --quantization fp8"
         )
+        if self.cfg.parallel.tensor_parallel_size > 1:
+            assert self.is_driver
+            self._ps.tp_broadcast_object(
+                {"kind": "add_lora", "name": name, "rank": rank,
+                 "alpha": alpha, "seed": seed}
+            )
+        return self._register_lora(name, rank, alpha, seed)
+
+    def _register_lora(self, name, rank, alpha, seed):
+        """Per-rank adapter construction (driver and TP workers)."""
+        from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
 
         if self.runner.lora_registry is None:
             self.runner.lora_registry = LoRARegistry()
@@ -333,7 +341,10 @@ class LLMEngine:
             payload = self._ps.tp_broadcast_object(None)
             if payload is None or payload.get("kind") == "stop":
                 return
-            if payload["kind"] == "mixed":
+            if payload["kind"] == "add_lora":
+                self._register_lora(payload["name"], payload["rank"],
+                                    payload["alpha"], payload["seed"])
+            elif payload["kind"] == "mixed":
                 self.runner.run_batch(payload)
             elif payload["kind"] == "prefill":
                 self.runner.run_prefill(payload)
