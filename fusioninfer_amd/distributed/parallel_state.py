@@ -18,6 +18,8 @@ import torch.distributed as dist
 _TP_GROUP: Optional[dist.ProcessGroup] = None
 _TP_RANK = 0
 _TP_WORLD = 1
+_PP_RANK = 0
+_PP_WORLD = 1
 _INITIALIZED = False
 
 
@@ -25,13 +27,17 @@ def init_distributed(
     tensor_parallel_size: int = 1,
     backend: str = "nccl",
     timeout_s: int = 600,
+    pipeline_parallel_size: int = 1,
 ) -> None:
     """Initialize the global process group and the TP subgroup.
 
     Reads RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT from env (torchrun), or
     LWS_LEADER_ADDRESS when launched by the control plane's LWS wrapper.
+
+    Pipeline parallelism composes with TP=1 for now: the whole world is
+    one pipeline (rank == stage); activations move with p2p send/recv.
     """
-    global _TP_GROUP, _TP_RANK, _TP_WORLD, _INITIALIZED
+    global _TP_GROUP, _TP_RANK, _TP_WORLD, _PP_RANK, _PP_WORLD, _INITIALIZED
     if _INITIALIZED:
         return
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -57,6 +63,12 @@ def init_distributed(
                 _TP_GROUP = grp
     _TP_RANK = rank % tp
     _TP_WORLD = tp
+    pp = pipeline_parallel_size
+    if pp > 1:
+        assert tp == 1, "PP currently composes with TP=1"
+        assert world_size == pp, (world_size, pp)
+        _PP_RANK = rank
+        _PP_WORLD = pp
     _INITIALIZED = True
 
 
@@ -80,14 +92,74 @@ def tp_group():
 
 
 def destroy() -> None:
-    global _TP_GROUP, _TP_RANK, _TP_WORLD, _INITIALIZED
+    global _TP_GROUP, _TP_RANK, _TP_WORLD, _PP_RANK, _PP_WORLD, _INITIALIZED
     if dist.is_initialized():
         dist.destroy_process_group()
     _TP_GROUP, _TP_RANK, _TP_WORLD, _INITIALIZED = None, 0, 1, False
+    _PP_RANK, _PP_WORLD = 0, 1
+
+
+# --------------------------------------------------------------- pipeline
+def pp_rank() -> int:
+    return _PP_RANK
+
+
+def pp_world_size() -> int:
+    return _PP_WORLD
+
+
+def pp_is_first() -> bool:
+    return _PP_RANK == 0
+
+
+def pp_is_last() -> bool:
+    return _PP_RANK == _PP_WORLD - 1
+
+
+def _gloo_safe_send(t: torch.Tensor, dst: int) -> None:
+    if t.device.type == "cpu" and t.dtype == torch.bfloat16:
+        dist.send(t.contiguous().view(torch.int16), dst)
+    else:
+        dist.send(t.contiguous(), dst)
+
+
+def _gloo_safe_recv(t: torch.Tensor, src: int) -> None:
+    if t.device.type == "cpu" and t.dtype == torch.bfloat16:
+        buf = torch.empty(t.shape, dtype=torch.int16)
+        dist.recv(buf, src)
+        t.copy_(buf.view(torch.bfloat16))
+    else:
+        dist.recv(t, src)
+
+
+def pp_send_next(t: torch.Tensor) -> None:
+    """Ship activations to the next pipeline stage (xGMI p2p under RCCL)."""
+    _gloo_safe_send(t, _PP_RANK + 1)
+
+
+def pp_recv_prev(shape, dtype, device) -> torch.Tensor:
+    t = torch.empty(shape, dtype=dtype, device=device)
+    _gloo_safe_recv(t, _PP_RANK - 1)
+    return t
+
+
+def pp_send_to(t: torch.Tensor, dst: int) -> None:
+    _gloo_safe_send(t, dst)
+
+
+def pp_recv_from(shape, dtype, device, src: int) -> torch.Tensor:
+    t = torch.empty(shape, dtype=dtype, device=device)
+    _gloo_safe_recv(t, src)
+    return t
 
 
 def tp_broadcast_object(obj=None):
-    """Broadcast a picklable object from the TP group's first rank."""
+    """Broadcast a picklable object from the execution group's first rank
+    (the TP group, or the whole pipeline when PP > 1)."""
+    if _PP_WORLD > 1:
+        lst = [obj]
+        dist.broadcast_object_list(lst, src=0)
+        return lst[0]
     if _TP_WORLD == 1:
         return obj
     lst = [obj]
@@ -97,7 +169,14 @@ def tp_broadcast_object(obj=None):
 
 
 def tp_all_reduce_min_int(value: int) -> int:
-    if _TP_WORLD == 1 or not dist.is_initialized():
+    """Min over the execution group (TP ranks, or all pipeline stages)."""
+    if not dist.is_initialized():
+        return value
+    if _PP_WORLD > 1:
+        t = torch.tensor([value], dtype=torch.int64)
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        return int(t.item())
+    if _TP_WORLD == 1:
         return value
     t = torch.tensor([value], dtype=torch.int64)
     dist.all_reduce(t, op=dist.ReduceOp.MIN, group=_TP_GROUP)

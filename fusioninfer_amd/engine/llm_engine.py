@@ -38,15 +38,17 @@ class LLMEngine:
     def __init__(self, cfg: EngineConfig, device: Optional[str] = None):
         from fusioninfer_amd.distributed import parallel_state as ps
 
-        if cfg.parallel.tensor_parallel_size > 1:
+        if (cfg.parallel.tensor_parallel_size > 1
+                or cfg.parallel.pipeline_parallel_size > 1):
             ps.init_distributed(
                 cfg.parallel.tensor_parallel_size,
                 backend=cfg.parallel.distributed_backend,
+                pipeline_parallel_size=cfg.parallel.pipeline_parallel_size,
             )
         else:
             ps.ensure_single_process()
         self._ps = ps
-        self.is_driver = ps.tp_rank() == 0
+        self.is_driver = ps.tp_rank() == 0 and ps.pp_rank() == 0
         self.cfg = cfg
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
@@ -92,7 +94,8 @@ class LLMEngine:
         assert self.cfg.model.quantization != "fp8", (
             "LoRA is not supported with --quantization fp8"
         )
-        if self.cfg.parallel.tensor_parallel_size > 1:
+        if (self.cfg.parallel.tensor_parallel_size
+                * self.cfg.parallel.pipeline_parallel_size) > 1:
             assert self.is_driver
             self._ps.tp_broadcast_object(
                 {"kind": "add_lora", "name": name, "rank": rank,
@@ -245,7 +248,8 @@ class LLMEngine:
         RCCL all-reduces line up."""
         assert self.is_driver, "only TP rank 0 steps; others run worker_loop"
         step_t0 = time.monotonic()
-        tp = self.cfg.parallel.tensor_parallel_size
+        tp = (self.cfg.parallel.tensor_parallel_size
+              * self.cfg.parallel.pipeline_parallel_size)
         batch = self.scheduler.schedule()
         t_sched = time.monotonic()
         if batch.is_empty:
@@ -352,7 +356,9 @@ class LLMEngine:
                 self.runner.run_decode(payload)
 
     def stop_workers(self) -> None:
-        if self.is_driver and self.cfg.parallel.tensor_parallel_size > 1:
+        exec_world = (self.cfg.parallel.tensor_parallel_size
+                      * self.cfg.parallel.pipeline_parallel_size)
+        if self.is_driver and exec_world > 1:
             self._ps.tp_broadcast_object({"kind": "stop"})
 
     # ------------------------------------------------------------ sync API

@@ -64,7 +64,8 @@ class ModelRunner:
 
         kv_heads = mc.num_kv_heads // ps.tp_world_size()
         esize = 1 if self.cfg.cache.kv_cache_dtype == "fp8" else 2
-        return mc.num_layers * 2 * kv_heads * self.block_size * mc.head_dim * esize
+        layers = self.model.num_local_layers  # PP stages hold a layer slice
+        return layers * 2 * kv_heads * self.block_size * mc.head_dim * esize
 
     def profile_num_blocks(self) -> int:
         import fusioninfer_amd.distributed.parallel_state as ps
@@ -101,7 +102,7 @@ class ModelRunner:
                 torch.zeros(shape, dtype=kv_dtype, device=self.device),
                 torch.zeros(shape, dtype=kv_dtype, device=self.device),
             )
-            for _ in range(mc.num_layers)
+            for _ in range(self.model.num_local_layers)
         ]
 
     # ------------------------------------------------------------ prefill
@@ -225,11 +226,54 @@ class ModelRunner:
             device=dev,
         )
         with torch.no_grad():
+            return self._forward_and_logits(ids, meta, logits_idx)
+
+    def _forward_and_logits(self, ids, meta, logits_idx):
+        """Run the model (all pipeline stages) and return logits on the
+        driver. PP: activations flow stage-to-stage with p2p send/recv
+        (xGMI under RCCL); the last stage computes logits over the sampled
+        rows and ships them back to rank 0. Middle stages return an empty
+        placeholder."""
+        import fusioninfer_amd.distributed.parallel_state as ps
+
+        V = self.cfg.model.vocab_size
+        if ps.pp_world_size() == 1:
             hidden = self.model(ids, meta, self.kv_caches)
             if logits_idx.numel() == 0:
-                return hidden.new_empty((0, self.cfg.model.vocab_size))
-            logits = self.model.compute_logits(hidden[logits_idx])
-        return logits
+                return hidden.new_empty((0, V))
+            return self.model.compute_logits(hidden[logits_idx])
+        T = ids.shape[0]
+        H = self.cfg.model.hidden_size
+        dt = torch.bfloat16
+        final = None
+        if ps.pp_is_first():
+            hidden, residual = self.model(ids, meta, self.kv_caches)
+            ps.pp_send_next(hidden)
+            ps.pp_send_next(residual)
+        else:
+            hidden = ps.pp_recv_prev((T, H), dt, self.device)
+            residual = ps.pp_recv_prev((T, H), dt, self.device)
+            out = self.model(None, meta, self.kv_caches,
+                             hidden=hidden, residual=residual)
+            if ps.pp_is_last():
+                final = out
+            else:
+                h2, r2 = out
+                ps.pp_send_next(h2)
+                ps.pp_send_next(r2)
+        last = ps.pp_world_size() - 1
+        S = int(logits_idx.numel())
+        if ps.pp_rank() == last:
+            if S == 0:
+                return torch.empty((0, V), dtype=dt, device=self.device)
+            logits = self.model.compute_logits(final[logits_idx])
+            ps.pp_send_to(logits, 0)
+            return logits
+        if ps.pp_rank() == 0:
+            if S == 0:
+                return torch.empty((0, V), dtype=dt, device=self.device)
+            return ps.pp_recv_from((S, V), dt, self.device, last)
+        return torch.empty((0, V), dtype=dt, device=self.device)
 
     def execute_prefill(self, seqs: List[Sequence], bm: BlockManager):
         return self.run_prefill(self.build_prefill_payload(seqs, bm))
@@ -313,8 +357,7 @@ class ModelRunner:
             pf_idx + list(range(np_, np_ + nd)), dtype=torch.long, device=dev
         )
         with torch.no_grad():
-            hidden = self.model(ids, meta, self.kv_caches)
-            return self.model.compute_logits(hidden[logits_idx])
+            return self._forward_and_logits(ids, meta, logits_idx)
 
     # ------------------------------------------------------------ decode
     def _alloc_static(self, max_bs: int) -> None:
@@ -340,12 +383,18 @@ class ModelRunner:
             block_tables=s["block_tables"][:bs],
             seq_lens=s["seq_lens"][:bs],
         )
-        hidden = self.model(s["ids"][:bs], meta, self.kv_caches)
-        return self.model.compute_logits(hidden)
+        return self._forward_and_logits(
+            s["ids"][:bs], meta,
+            torch.arange(bs, dtype=torch.long, device=self.device),
+        )
 
     def capture_decode_graphs(self) -> None:
+        import fusioninfer_amd.distributed.parallel_state as ps
+
         if not self.is_cuda or self.cfg.enforce_eager:
             return
+        if ps.pp_world_size() > 1:
+            return  # p2p sends inside capture: PP decode runs eager
         max_bs = min(self.cfg.scheduler.max_num_seqs, _DECODE_BUCKETS[-1])
         buckets = [b for b in _DECODE_BUCKETS if b <= max_bs]
         if not self._static:

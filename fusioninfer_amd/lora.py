@@ -48,8 +48,11 @@ class LoRAAdapter:
                 ("down", inter, H, False),
             ):
                 A = torch.randn(rank, inf, generator=gen, dtype=torch.float32) * 0.02
+                # seeded B std 0.1: synthetic adapters must move logits
+                # enough to flip greedy argmax on random-init base weights
+                # (0.02 was within sampler tie-break noise on some seeds)
                 B = torch.zeros(outf, rank) if seed is None else (
-                    torch.randn(outf, rank, generator=gen, dtype=torch.float32) * 0.02
+                    torch.randn(outf, rank, generator=gen, dtype=torch.float32) * 0.1
                 )
                 if col:  # column-parallel: shard B rows (per-segment is an
                     # approximation: qkv/gate_up segments are contiguous per

@@ -192,41 +192,82 @@ class DecoderLayer(nn.Module):
 
 
 class CausalLM(nn.Module):
-    """The full model: embedding -> N decoder layers -> final norm -> lm_head."""
+    """The full model: embedding -> N decoder layers -> final norm -> lm_head.
+
+    Pipeline parallelism: stage r holds a contiguous slice
+    of the layers; the first stage owns the embedding, the last the final
+    norm + lm_head. Random init draws every module from a per-module seed
+    derived from the caller's seed, so the SAME seed produces the SAME
+    weights regardless of the PP/TP layout (stages skip modules without
+    desyncing the stream; TP ranks still slice identical full matrices)."""
 
     def __init__(self, cfg: ModelConfig):
         super().__init__()
         self.cfg = cfg
-        self.embed_tokens = nn.Parameter(
-            torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16)
-            .normal_(0.0, 0.02),
-            requires_grad=False,
-        )
-        self.layers = nn.ModuleList(
-            [DecoderLayer(cfg, i) for i in range(cfg.num_layers)]
-        )
-        self.final_norm_weight = nn.Parameter(
-            torch.ones(cfg.hidden_size, dtype=torch.bfloat16), requires_grad=False
-        )
-        if cfg.tie_word_embeddings:
-            self.lm_head = None
+        pp = ps.pp_world_size()
+        ppr = ps.pp_rank()
+        self.pp_first = ppr == 0
+        self.pp_last = ppr == pp - 1
+        self.layer_start = cfg.num_layers * ppr // pp
+        self.layer_end = cfg.num_layers * (ppr + 1) // pp
+
+        base_seed = torch.initial_seed() % (2**62)
+
+        def seeded(i):
+            torch.manual_seed(base_seed + 1000003 * (i + 1))
+
+        if self.pp_first or (self.pp_last and cfg.tie_word_embeddings):
+            seeded(0)
+            self.embed_tokens = nn.Parameter(
+                torch.empty(cfg.vocab_size, cfg.hidden_size,
+                            dtype=torch.bfloat16).normal_(0.0, 0.02),
+                requires_grad=False,
+            )
         else:
-            self.lm_head = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size)
+            self.embed_tokens = None
+        layers = []
+        for i in range(self.layer_start, self.layer_end):
+            seeded(1 + i)
+            layers.append(DecoderLayer(cfg, i))
+        self.layers = nn.ModuleList(layers)
+        if self.pp_last:
+            self.final_norm_weight = nn.Parameter(
+                torch.ones(cfg.hidden_size, dtype=torch.bfloat16),
+                requires_grad=False,
+            )
+            if cfg.tie_word_embeddings:
+                self.lm_head = None
+            else:
+                seeded(1 + cfg.num_layers)
+                self.lm_head = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size)
+        else:
+            self.final_norm_weight = None
+            self.lm_head = None
+        torch.manual_seed(base_seed + 777)  # same post-init state on all ranks
         cos_sin = ops.compute_cos_sin_cache(
             cfg.head_dim, cfg.max_position_embeddings, cfg.rope_theta
         )
         self.register_buffer("cos_sin", cos_sin, persistent=False)
 
+    @property
+    def num_local_layers(self) -> int:
+        return self.layer_end - self.layer_start
+
     def forward(
         self,
-        input_ids: torch.Tensor,     # [T] int
+        input_ids,                   # [T] int (first stage; None otherwise)
         meta: AttnMetadata,
-        kv_caches,                   # list of (k_cache, v_cache) per layer
-    ) -> torch.Tensor:
-        hidden = self.embed_tokens[input_ids]
-        residual = None
+        kv_caches,                   # list of (k_cache, v_cache) per LOCAL layer
+        hidden: torch.Tensor = None,     # PP stages > 0: received activations
+        residual: torch.Tensor = None,
+    ):
+        if self.pp_first:
+            hidden = self.embed_tokens[input_ids]
+            residual = None
         for layer, kv in zip(self.layers, kv_caches):
             hidden, residual = layer(hidden, residual, meta, kv, self.cos_sin)
+        if not self.pp_last:
+            return hidden, residual
         hidden, _ = ops.fused_add_rms_norm(
             hidden, residual, self.final_norm_weight, self.cfg.rms_norm_eps
         )

@@ -156,9 +156,9 @@ def main(argv=None):
     if world <= 1:
         _rank_main(0, args, nproc)
         return
-    assert args.tensor_parallel_size == world, (
-        "multi-rank launch currently maps every rank into one TP group: "
-        f"--tensor-parallel-size {args.tensor_parallel_size} != {world}"
+    assert args.tensor_parallel_size * args.pipeline_parallel_size == world, (
+        "multi-rank launch maps every rank into one TP group or one "
+        f"pipeline: tp*pp != {world}"
     )
     import torch.multiprocessing as mp
 
