@@ -54,6 +54,10 @@ class CacheConfig:
     # ~2x decode-attention bandwidth and 2x cache capacity; mirrors the
     # reference engines' --kv-cache-dtype fp8 surface)
     kv_cache_dtype: str = "auto"
+    # CPU swap space for preemption (vLLM --swap-space surface): preempted
+    # sequences park their KV in host memory and resume without recompute.
+    # 0 = preemption-by-recompute only.
+    swap_space_gb: float = 0.0
 
 
 @dataclasses.dataclass

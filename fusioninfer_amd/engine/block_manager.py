@@ -82,6 +82,17 @@ class BlockManager:
         if self.enable_prefix_caching:
             self._register_hashes(seq)
 
+    def allocate_raw(self, seq: Sequence, num_blocks: int) -> None:
+        """Allocate exactly num_blocks fresh blocks, bypassing the prefix
+        cache (swap-in restore: the blocks will be overwritten with the
+        sequence's saved KV, so cache sharing would corrupt shared data)."""
+        assert not seq.block_ids
+        for _ in range(num_blocks):
+            blk = self._pop_free_block()
+            self.ref_count[blk] = 1
+            seq.block_ids.append(blk)
+        seq.num_cached_tokens = 0
+
     def _take(self, blk: int) -> None:
         if blk in self.cached_free:
             del self.cached_free[blk]

@@ -61,6 +61,13 @@ class LLMEngine:
             enable_prefix_caching=cfg.cache.enable_prefix_caching,
         )
         self.scheduler = Scheduler(cfg.scheduler, self.block_manager)
+        self._swapped: Dict[str, "torch.Tensor"] = {}
+        self._swap_bytes = 0
+        self.num_swap_outs = 0
+        if (cfg.cache.swap_space_gb > 0
+                and cfg.parallel.tensor_parallel_size == 1
+                and cfg.parallel.pipeline_parallel_size == 1):
+            self.scheduler.swap_out_fn = self._swap_out
         self.sampler = Sampler(cfg.seed, device)
         self.runner.capture_decode_graphs()
         self._req_counter = itertools.count()
@@ -150,6 +157,9 @@ class LLMEngine:
         seq = self.seqs.pop(request_id, None)
         if seq is None:
             return False
+        staging = self._swapped.pop(request_id, None)
+        if staging is not None:
+            self._swap_bytes -= staging.numel() * staging.element_size()
         if seq in self.scheduler.waiting:
             self.scheduler.waiting.remove(seq)
         if seq in self.scheduler.running:
@@ -159,6 +169,38 @@ class LLMEngine:
             self.block_manager.free(seq)
         seq.status = SeqStatus.FINISHED
         return True
+
+    # ----------------------------------------------------- swap preemption
+    def _swap_out(self, seq: Sequence) -> bool:
+        """Park a preempted sequence's KV blocks in host memory (the
+        vLLM --swap-space behavior). Called by the scheduler BEFORE the
+        victim's blocks are freed. Returns False when the swap budget is
+        exhausted (scheduler falls back to recompute)."""
+        from fusioninfer_amd.distributed.kv_transfer import pack_kv_blocks
+
+        cap = int(self.cfg.cache.swap_space_gb * (1 << 30))
+        ids = torch.tensor(seq.block_ids, dtype=torch.int32,
+                           device=self.runner.device)
+        staging = pack_kv_blocks(self.runner.kv_caches, ids).to(
+            "cpu", non_blocking=False
+        )
+        nbytes = staging.numel() * staging.element_size()
+        if self._swap_bytes + nbytes > cap:
+            return False
+        self._swapped[seq.seq_id] = staging
+        self._swap_bytes += nbytes
+        self.num_swap_outs += 1
+        return True
+
+    def _swap_in(self, seq: Sequence) -> None:
+        from fusioninfer_amd.distributed.kv_transfer import unpack_kv_blocks
+
+        staging = self._swapped.pop(seq.seq_id)
+        self._swap_bytes -= staging.numel() * staging.element_size()
+        ids = torch.tensor(seq.block_ids, dtype=torch.int32,
+                           device=self.runner.device)
+        unpack_kv_blocks(staging.to(self.runner.device), self.runner.kv_caches,
+                         ids)
 
     # ------------------------------------------------------- PD interfaces
     def prefill_export(self, prompt_token_ids: List[int]):
@@ -252,6 +294,8 @@ class LLMEngine:
               * self.cfg.parallel.pipeline_parallel_size)
         batch = self.scheduler.schedule()
         t_sched = time.monotonic()
+        for seq in batch.swap_in:
+            self._swap_in(seq)
         if batch.is_empty:
             return []
         payload = self.runner.build_batch_payload(

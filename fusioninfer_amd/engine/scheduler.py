@@ -33,6 +33,10 @@ class ScheduledBatch:
     prefill_chunks: List[int]          # tokens scheduled per prefill seq
     decode_seqs: List[Sequence]
     preempted: List[Sequence]
+    # swapped-out sequences re-admitted this step: the engine restores
+    # their KV from CPU swap into the freshly allocated blocks BEFORE the
+    # forward runs; they rejoin decode/prefill next step
+    swap_in: List[Sequence] = dataclasses.field(default_factory=list)
 
     @property
     def is_empty(self) -> bool:
@@ -49,6 +53,9 @@ class Scheduler:
         self.bm = block_manager
         self.waiting: Deque[Sequence] = deque()
         self.running: List[Sequence] = []
+        # engine-provided: callable(seq) -> bool, True = KV parked in CPU
+        # swap (preemption keeps computed state); False/None = recompute
+        self.swap_out_fn = None
 
     # ----------------------------------------------------------- queue ops
     def add(self, seq: Sequence) -> None:
@@ -104,6 +111,26 @@ class Scheduler:
             prefill.append(seq)
             chunks.append(chunk)
             budget -= chunk
+        # resume swapped-out sequences first: they are at the queue head,
+        # carry computed KV in host swap, and need neither token budget nor
+        # the admission hysteresis — just free blocks (+1 headroom so the
+        # next decode append does not instantly re-preempt them)
+        swap_in: List[Sequence] = []
+        while (
+            self.waiting
+            and self.waiting[0].status == SeqStatus.SWAPPED
+            and self.waiting[0] not in preempted
+            and len(self.running) < self.cfg.max_num_seqs
+        ):
+            seq = self.waiting[0]
+            if self.bm.num_free() < seq.swap_num_blocks + 1:
+                break
+            self.waiting.popleft()
+            self.bm.allocate_raw(seq, seq.swap_num_blocks)
+            seq.status = SeqStatus.RUNNING
+            swap_in.append(seq)
+            self.running.append(seq)
+
         # then admit waiting prompts — with hysteresis: when decodes are
         # running, hold admissions until enough prompt tokens queue up so
         # most steps stay pure-decode (hipGraph path)
@@ -138,10 +165,18 @@ class Scheduler:
             chunks.append(chunk)
             budget -= chunk
             self.running.append(seq)
-        return ScheduledBatch(prefill, chunks, decode, preempted)
+        return ScheduledBatch(prefill, chunks, decode, preempted, swap_in)
 
     def _preempt_newest(self) -> Sequence:
         victim = self.running.pop()  # newest
+        if self.swap_out_fn is not None and self.swap_out_fn(victim):
+            # KV parked in CPU swap: keep all computed state; resume
+            # restores the blocks instead of re-prefilling
+            victim.swap_num_blocks = len(victim.block_ids)
+            self.bm.free(victim)
+            victim.status = SeqStatus.SWAPPED
+            self.waiting.appendleft(victim)
+            return victim
         self.bm.free(victim)
         victim.status = SeqStatus.PREEMPTED
         # recompute: generated tokens become part of the prompt

@@ -11,6 +11,7 @@ class SeqStatus(enum.Enum):
     WAITING = "waiting"
     RUNNING = "running"
     PREEMPTED = "preempted"
+    SWAPPED = "swapped"      # KV saved to CPU swap space; resume skips recompute
     FINISHED = "finished"
 
 
@@ -63,6 +64,7 @@ class Sequence:
         self.num_computed_tokens = 0
         # PD producer: keep cache blocks alive after finish for KV export
         self.hold_blocks = False
+        self.swap_num_blocks = 0
         # LoRA adapter name (None = base model)
         self.lora_name = None
         # per-output-token logprob entries when sampling.logprobs is set:

@@ -35,6 +35,8 @@ def parse_args(argv=None):
     p.add_argument("--enforce-eager", action="store_true")
     p.add_argument("--quantization", choices=["fp8"], default=None)
     p.add_argument("--kv-cache-dtype", choices=["auto", "fp8"], default="auto")
+    p.add_argument("--swap-space", type=float, default=0.0,
+                   help="GiB of CPU swap for preempted sequences (0=recompute)")
     p.add_argument("--enable-lora", action="store_true")
     p.add_argument("--max-loras", type=int, default=8)
     p.add_argument("--lora-modules", nargs="*", default=[],
@@ -79,6 +81,7 @@ def build_engine_config(args):
             gpu_memory_utilization=args.gpu_memory_utilization,
             enable_prefix_caching=args.enable_prefix_caching,
             kv_cache_dtype=args.kv_cache_dtype,
+            swap_space_gb=args.swap_space,
         ),
         scheduler=SchedulerConfig(
             max_num_seqs=args.max_num_seqs,

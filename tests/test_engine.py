@@ -251,3 +251,47 @@ def test_admission_aging_escape_under_low_load(monkeypatch):
     eng.seqs[b].arrival_time = arrival - 1.0
     eng.step()
     assert eng.seqs[b].num_computed_tokens > 0  # admitted, not starved
+
+
+def test_swap_preemption_is_transparent():
+    """With CPU swap space, a preempted sequence parks its KV and resumes
+    with the EXACT same cache state — greedy outputs must be identical to
+    a run with a cache big enough to never preempt (recompute-preemption
+    cannot promise this: re-prefilling decode tokens changes op order and
+    can flip argmax). At least one swap must happen under the tiny cache."""
+    import torch
+
+    from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from fusioninfer_amd.models.registry import get_model_config
+
+    def run(swap_gb, num_blocks):
+        torch.manual_seed(9)
+        cfg = EngineConfig(
+            model=get_model_config("tiny-qwen3"),
+            # 12 blocks of 16 = 192 token slots: three 40-token prompts
+            # generating 40 tokens each cannot all stay resident
+            cache=CacheConfig(num_gpu_blocks=num_blocks,
+                              enable_prefix_caching=False,
+                              swap_space_gb=swap_gb),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=256, max_model_len=128
+            ),
+        )
+        eng = LLMEngine(cfg, device="cpu")
+        prompts = [
+            [(11 * i + j) % 500 for j in range(40)] for i in range(3)
+        ]
+        outs = eng.generate(
+            prompts,
+            SamplingParams(max_tokens=40, temperature=0.0, ignore_eos=True),
+        )
+        assert eng.block_manager.num_free() == eng.block_manager.num_blocks
+        return [o.output_token_ids for o in outs], eng.num_swap_outs
+
+    toks_unconstrained, s0 = run(0.0, 64)
+    assert s0 == 0
+    toks_swap, swaps = run(1.0, 12)
+    assert swaps > 0, "tiny cache never triggered a swap"
+    assert toks_swap == toks_unconstrained
