@@ -61,7 +61,11 @@ def bench_decode():
         lens = torch.full((batch,), ctx, dtype=torch.int32, device="cuda")
         t = timeit(lambda: ops.paged_attention_decode(q, k_cache, v_cache, bt, lens))
         gb = batch * ctx * Hk * D * 2 * 2 / 1e9  # K+V bytes actually needed
-        print(f"  b{batch} ctx{ctx}: {t*1e6:8.1f} us  {gb/t:7.0f} GB/s (KV)")
+        k8 = k_cache.to(torch.float8_e4m3fn)
+        v8 = v_cache.to(torch.float8_e4m3fn)
+        t8 = timeit(lambda: ops.paged_attention_decode(q, k8, v8, bt, lens))
+        print(f"  b{batch} ctx{ctx}: {t*1e6:8.1f} us  {gb/t:7.0f} GB/s (KV) | "
+              f"fp8kv {t8*1e6:8.1f} us  {gb/2/t8:7.0f} GB/s")
 
 
 def bench_elementwise():
