@@ -34,6 +34,15 @@ class ModelConfig:
     model_path: Optional[str] = None
     # None = bf16; "fp8" = OCP e4m3 weights + dynamic per-token activations
     quantization: Optional[str] = None
+    # MoE (Qwen3-MoE family): num_experts 0 = dense MLP
+    num_experts: int = 0
+    num_experts_per_tok: int = 8
+    moe_intermediate_size: int = 0
+    norm_topk_prob: bool = True
+
+    @property
+    def is_moe(self) -> bool:
+        return self.num_experts > 0
 
     @property
     def q_size(self) -> int:

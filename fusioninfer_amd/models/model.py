@@ -144,6 +144,81 @@ class MLP(nn.Module):
         return self.down_proj(act)
 
 
+class MoEMLP(nn.Module):
+    """Qwen3-MoE sparse MLP: softmax router over E experts, top-k with
+    normalized weights, SwiGLU experts (HF qwen3_moe semantics:
+    norm_topk_prob renormalizes the selected probabilities).
+
+    Expert parallelism over the TP group: experts are partitioned across
+    ranks (each rank holds E/tp complete experts); tokens routed to a
+    remote expert contribute nothing locally and the per-layer all-reduce
+    — the same one the dense row-parallel convention already requires —
+    sums expert outputs across ranks (allreduce-combine EP; all-to-all
+    dispatch is the multi-node variant). Expert weights
+    are drawn from per-(layer, expert) seeds so the same engine seed
+    yields identical experts under any EP layout."""
+
+    def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
+        super().__init__()
+        self.layer_idx = layer_idx
+        tp = ps.tp_world_size()
+        E = cfg.num_experts
+        assert E % tp == 0, "num_experts must divide TP size (EP sharding)"
+        self.num_experts = E
+        self.top_k = cfg.num_experts_per_tok
+        self.norm_topk = cfg.norm_topk_prob
+        self.e_start = ps.tp_rank() * (E // tp)
+        self.e_end = self.e_start + E // tp
+        H, inter = cfg.hidden_size, cfg.moe_intermediate_size
+        base = torch.initial_seed() % (2**62)
+        gen = torch.Generator(device="cpu")
+        gen.manual_seed(base)
+        self.router_weight = nn.Parameter(
+            (torch.randn(E, H, generator=gen) * 0.02).to(torch.bfloat16),
+            requires_grad=False,
+        )
+        gup, down = [], []
+        for e in range(self.e_start, self.e_end):
+            gen.manual_seed(base + 7919 * (e + 1))
+            gup.append(torch.randn(2 * inter, H, generator=gen) * 0.02)
+            down.append(torch.randn(H, inter, generator=gen) * 0.02)
+        self.gate_up = nn.Parameter(
+            torch.stack(gup).to(torch.bfloat16), requires_grad=False
+        )  # [E_local, 2*I, H]
+        self.down = nn.Parameter(
+            torch.stack(down).to(torch.bfloat16), requires_grad=False
+        )  # [E_local, H, I]
+
+    def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
+        T = x.shape[0]
+        logits = (x.float() @ self.router_weight.float().T)  # [T, E]
+        probs = torch.softmax(logits, dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)
+        if self.norm_topk:
+            topv = topv / topv.sum(dim=-1, keepdim=True)
+        # group token-slots by expert: one gather + two GEMMs per LIVE
+        # local expert (round-2: fused grouped MFMA GEMM)
+        flat_e = topi.reshape(-1)
+        order = torch.argsort(flat_e, stable=True)
+        counts = torch.bincount(flat_e, minlength=self.num_experts)
+        tok_of = order // self.top_k
+        w_of = topv.reshape(-1)[order]
+        out = torch.zeros(T, x.shape[1], dtype=torch.float32, device=x.device)
+        start = int(counts[: self.e_start].sum())
+        for le in range(self.e_end - self.e_start):
+            n = int(counts[self.e_start + le])
+            if n == 0:
+                continue
+            rows = tok_of[start : start + n]
+            xe = x[rows]
+            act = ops.silu_and_mul(xe @ self.gate_up[le].T)
+            ye = act @ self.down[le].T
+            out.index_add_(0, rows, ye.float() * w_of[start : start + n, None])
+            start += n
+        out = out.to(x.dtype)
+        return ps.tp_all_reduce(out)
+
+
 class DecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, layer_idx: int = 0):
         super().__init__()

@@ -47,6 +47,24 @@ register(
 )
 
 register(
+    "Qwen3-30B-A3B",
+    hidden_size=2048,
+    num_layers=48,
+    num_heads=32,
+    num_kv_heads=4,
+    head_dim=128,
+    intermediate_size=6144,          # dense fallback size (unused for MoE)
+    vocab_size=151936,
+    rope_theta=1_000_000.0,
+    rms_norm_eps=1e-6,
+    max_position_embeddings=40960,
+    qk_norm=True,
+    num_experts=128,
+    num_experts_per_tok=8,
+    moe_intermediate_size=768,
+)
+
+register(
     "Llama-3-8B",
     hidden_size=4096,
     num_layers=32,
@@ -90,6 +108,24 @@ register(
     rms_norm_eps=1e-6,
     max_position_embeddings=4096,
     qk_norm=True,
+)
+
+register(
+    "tiny-qwen3-moe",
+    hidden_size=256,
+    num_layers=2,
+    num_heads=4,
+    num_kv_heads=2,
+    head_dim=64,
+    intermediate_size=512,
+    vocab_size=1024,
+    rope_theta=10_000.0,
+    rms_norm_eps=1e-6,
+    max_position_embeddings=4096,
+    qk_norm=True,
+    num_experts=8,
+    num_experts_per_tok=2,
+    moe_intermediate_size=128,
 )
 
 
