@@ -39,6 +39,10 @@ class ModelRunner:
             from fusioninfer_amd.models.weight_loader import load_safetensors_dir
 
             load_safetensors_dir(self.model, mc.model_path)
+        if mc.quantization == "fp8" and mc.is_moe:
+            raise ValueError(
+                "fp8 quantization with MoE needs fp8 expert GEMMs (round 2)"
+            )
         if mc.quantization == "fp8":
             from fusioninfer_amd.quantization import convert_linear_to_fp8
 
@@ -395,6 +399,8 @@ class ModelRunner:
             return
         if ps.pp_world_size() > 1:
             return  # p2p sends inside capture: PP decode runs eager
+        if self.cfg.model.is_moe:
+            return  # MoE routing has data-dependent shapes: eager decode
         max_bs = min(self.cfg.scheduler.max_num_seqs, _DECODE_BUCKETS[-1])
         buckets = [b for b in _DECODE_BUCKETS if b <= max_bs]
         if not self._static:

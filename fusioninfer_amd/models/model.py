@@ -190,31 +190,43 @@ class MoEMLP(nn.Module):
         )  # [E_local, H, I]
 
     def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
-        T = x.shape[0]
+        T, H = x.shape
+        E_local = self.e_end - self.e_start
         logits = (x.float() @ self.router_weight.float().T)  # [T, E]
         probs = torch.softmax(logits, dim=-1)
         topv, topi = probs.topk(self.top_k, dim=-1)
         if self.norm_topk:
             topv = topv / topv.sum(dim=-1, keepdim=True)
-        # group token-slots by expert: one gather + two GEMMs per LIVE
-        # local expert (round-2: fused grouped MFMA GEMM)
+        # capacity-padded batched expert compute: token-slots sorted by
+        # expert, scattered into [E_local, cap, H], TWO bmm launches per
+        # layer regardless of E (a per-expert GEMM loop is launch-bound
+        # at 128 experts; fused grouped MFMA GEMM is the round-2 step up)
         flat_e = topi.reshape(-1)
         order = torch.argsort(flat_e, stable=True)
         counts = torch.bincount(flat_e, minlength=self.num_experts)
+        sorted_e = flat_e[order]
         tok_of = order // self.top_k
         w_of = topv.reshape(-1)[order]
-        out = torch.zeros(T, x.shape[1], dtype=torch.float32, device=x.device)
-        start = int(counts[: self.e_start].sum())
-        for le in range(self.e_end - self.e_start):
-            n = int(counts[self.e_start + le])
-            if n == 0:
-                continue
-            rows = tok_of[start : start + n]
-            xe = x[rows]
-            act = ops.silu_and_mul(xe @ self.gate_up[le].T)
-            ye = act @ self.down[le].T
-            out.index_add_(0, rows, ye.float() * w_of[start : start + n, None])
-            start += n
+        starts = counts.cumsum(0) - counts
+        pos_in_e = (
+            torch.arange(sorted_e.numel(), device=x.device) - starts[sorted_e]
+        )
+        local = (sorted_e >= self.e_start) & (sorted_e < self.e_end)
+        cap = int(counts[self.e_start : self.e_end].max())
+        out = torch.zeros(T, H, dtype=torch.float32, device=x.device)
+        if cap > 0:
+            dest = ((sorted_e - self.e_start) * cap + pos_in_e)[local]
+            rows = tok_of[local]
+            xpad = torch.zeros(E_local * cap, H, dtype=x.dtype, device=x.device)
+            xpad[dest] = x[rows]
+            gu = torch.bmm(
+                xpad.view(E_local, cap, H), self.gate_up.transpose(1, 2)
+            )
+            act = ops.silu_and_mul(gu.view(E_local * cap, -1))
+            y = torch.bmm(
+                act.view(E_local, cap, -1), self.down.transpose(1, 2)
+            ).view(E_local * cap, H)
+            out.index_add_(0, rows, y[dest].float() * w_of[local, None])
         out = out.to(x.dtype)
         return ps.tp_all_reduce(out)
 
@@ -229,7 +241,8 @@ class DecoderLayer(nn.Module):
             torch.ones(cfg.hidden_size, dtype=torch.bfloat16), requires_grad=False
         )
         self.self_attn = Attention(cfg, layer_idx)
-        self.mlp = MLP(cfg, layer_idx)
+        self.mlp = (MoEMLP(cfg, layer_idx) if cfg.is_moe
+                    else MLP(cfg, layer_idx))
         self.eps = cfg.rms_norm_eps
         self.fp8 = cfg.quantization == "fp8"
 

@@ -42,6 +42,7 @@ def load_hf_state_dict(
 
     # staging for fused weights: (layer, which) -> tensor
     pending: Dict[Tuple[int, str], torch.Tensor] = {}
+    moe_pending: Dict[Tuple[int, int, str], torch.Tensor] = {}
     loaded = 0
 
     def put(param: torch.nn.Parameter, value: torch.Tensor):
@@ -62,23 +63,45 @@ def load_hf_state_dict(
         if all(k in pending for k in keys):
             gw = _shard_rows(pending.pop(keys[0]), tp, rank)
             uw = _shard_rows(pending.pop(keys[1]), tp, rank)
-            put(model.layers[li].mlp.gate_up_proj.weight,
+            put(local(li).mlp.gate_up_proj.weight,
                 torch.cat([gw, uw], dim=0))
+
+    def local(li: int):
+        # PP: global layer index -> this stage's slice
+        return model.layers[li - model.layer_start]
+
+    def owns(li: int) -> bool:
+        return model.layer_start <= li < model.layer_end
+
+    def try_fuse_expert(li: int, e: int):
+        keys = [(li, e, "gate"), (li, e, "up")]
+        if all(k in moe_pending for k in keys):
+            mlp = local(li).mlp
+            le = e - mlp.e_start
+            gw = moe_pending.pop(keys[0])
+            uw = moe_pending.pop(keys[1])
+            mlp.gate_up.data[le].copy_(
+                torch.cat([gw, uw], dim=0).to(mlp.gate_up.dtype)
+            )
 
     for name, w in tensors:
         loaded += 1
         if name == "model.embed_tokens.weight":
-            put(model.embed_tokens, w)
+            if model.embed_tokens is not None:
+                put(model.embed_tokens, w)
         elif name == "model.norm.weight":
-            put(model.final_norm_weight, w)
+            if model.final_norm_weight is not None:
+                put(model.final_norm_weight, w)
         elif name == "lm_head.weight":
             if model.lm_head is not None:
                 put(model.lm_head.weight, w)
         elif name.startswith("model.layers."):
             parts = name.split(".")
             li = int(parts[2])
+            if not owns(li):
+                continue
             rest = ".".join(parts[3:])
-            layer = model.layers[li]
+            layer = local(li)
             if rest == "self_attn.q_proj.weight":
                 pending[(li, "q")] = w
                 try_fuse_qkv(li)
@@ -102,6 +125,19 @@ def load_hf_state_dict(
                 try_fuse_gate_up(li)
             elif rest == "mlp.down_proj.weight":
                 put(layer.mlp.down_proj.weight, _shard_cols(w, tp, rank))
+            elif rest == "mlp.gate.weight":  # MoE router (qwen3_moe)
+                put(layer.mlp.router_weight, w)
+            elif parts[3] == "mlp" and parts[4] == "experts":
+                e = int(parts[5])
+                mlp = layer.mlp
+                if not (mlp.e_start <= e < mlp.e_end):
+                    continue  # EP: expert lives on another rank
+                which = parts[6].replace("_proj", "")  # gate/up/down
+                if which == "down":
+                    mlp.down.data[e - mlp.e_start].copy_(w.to(mlp.down.dtype))
+                else:
+                    moe_pending[(li, e, which)] = w
+                    try_fuse_expert(li, e)
             elif rest == "input_layernorm.weight":
                 put(layer.input_norm_weight, w)
             elif rest == "post_attention_layernorm.weight":
@@ -111,6 +147,9 @@ def load_hf_state_dict(
         else:
             loaded -= 1
     assert not pending, f"unfused partial weights remain: {list(pending)}"
+    assert not moe_pending, (
+        f"unfused expert weights remain: {list(moe_pending)[:4]}"
+    )
     return loaded
 
 

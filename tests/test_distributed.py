@@ -340,3 +340,68 @@ def test_pp2_engine_matches_single_process():
             p.join(timeout=300)
             assert p.exitcode == 0, f"rank exited {p.exitcode}"
         assert results[0] == "ok" and results[1] == "ok"
+
+
+# ------------------------------------------------- EP (experts over TP group)
+def _ep_worker(rank, port, results):
+    from fusioninfer_amd.distributed import parallel_state as ps
+    from fusioninfer_amd.engine.block_manager import BlockManager
+    from fusioninfer_amd.engine.model_runner import ModelRunner
+    from fusioninfer_amd.engine.sequence import SamplingParams, Sequence
+    from fusioninfer_amd.models.registry import get_model_config
+
+    def moe_cfg():
+        cfg = _engine_cfg()
+        cfg.model = get_model_config("tiny-qwen3-moe")
+        return cfg
+
+    ps.ensure_single_process()
+    torch.manual_seed(77)
+    cfg1 = moe_cfg()
+    runner1 = ModelRunner(cfg1, "cpu")
+    runner1.allocate_kv_caches()
+    bm1 = BlockManager(64, 16)
+    seq = Sequence("s", PROMPT, SamplingParams())
+    bm1.allocate(seq)
+    ref_logits = runner1.execute_prefill([seq], bm1).float()
+    ps.destroy()
+
+    _init(rank, 2, port)
+    try:
+        ps.init_distributed(tensor_parallel_size=2, backend="gloo")
+        torch.manual_seed(77)
+        cfg2 = moe_cfg()
+        cfg2.parallel.tensor_parallel_size = 2
+        runner2 = ModelRunner(cfg2, "cpu")
+        # experts are sharded across the group (EP): 4 of 8 per rank
+        assert runner2.model.layers[0].mlp.gate_up.shape[0] == 4
+        runner2.allocate_kv_caches()
+        bm2 = BlockManager(64, 16)
+        seq2 = Sequence("s2", PROMPT, SamplingParams())
+        bm2.allocate(seq2)
+        tp_logits = runner2.execute_prefill([seq2], bm2).float()
+        rel = (tp_logits - ref_logits).norm() / ref_logits.norm()
+        assert rel.item() < 0.05, rel.item()
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+        ps.destroy()
+
+
+def test_ep2_moe_matches_single_rank():
+    """MoE experts sharded over a 2-rank group (allreduce-combine EP):
+    logits match the single-rank model (same per-expert seeds)."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29623
+        procs = [
+            ctx.Process(target=_ep_worker, args=(r, port, results))
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+            assert p.exitcode == 0, f"rank exited {p.exitcode}"
+        assert results[0] == "ok" and results[1] == "ok"

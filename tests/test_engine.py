@@ -295,3 +295,52 @@ def test_swap_preemption_is_transparent():
     toks_swap, swaps = run(1.0, 12)
     assert swaps > 0, "tiny cache never triggered a swap"
     assert toks_swap == toks_unconstrained
+
+
+def test_moe_engine_generates_and_routes():
+    """Qwen3-MoE family: tiny MoE model generates; the router actually
+    spreads tokens over multiple experts; MoE layer matches a hand
+    reference on one forward."""
+    import torch
+    import torch.nn.functional as F
+
+    from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from fusioninfer_amd.models.registry import get_model_config
+
+    torch.manual_seed(4)
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3-moe"),
+        cache=CacheConfig(num_gpu_blocks=64),
+        scheduler=SchedulerConfig(
+            max_num_seqs=4, max_num_batched_tokens=256, max_model_len=128
+        ),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    out = eng.generate(
+        [[5, 3, 1] * 10], SamplingParams(max_tokens=5, temperature=0.0)
+    )[0]
+    assert len(out.output_token_ids) == 5
+
+    # MoE layer oracle: dense per-expert compute, weighted-sum combine
+    moe = eng.runner.model.layers[0].mlp
+    x = torch.randn(7, cfg.model.hidden_size, dtype=torch.bfloat16)
+    got = moe(x).float()
+    logits = x.float() @ moe.router_weight.float().T
+    probs = torch.softmax(logits, -1)
+    topv, topi = probs.topk(moe.top_k, -1)
+    topv = topv / topv.sum(-1, keepdim=True)
+    exp = torch.zeros_like(x, dtype=torch.float32)
+    for t in range(x.shape[0]):
+        for k in range(moe.top_k):
+            e = int(topi[t, k])
+            gu = x[t].float() @ moe.gate_up[e].float().T
+            g, u = gu.chunk(2)
+            y = (F.silu(g) * u) @ moe.down[e].float().T
+            exp[t] += float(topv[t, k]) * y
+    # the layer computes in bf16 GEMMs; compare loosely
+    rel = (got - exp).norm() / exp.norm()
+    assert rel.item() < 0.05, rel.item()
+    # routing uses more than one expert across tokens
+    assert len(set(topi.flatten().tolist())) > 1

@@ -210,3 +210,50 @@ def test_gpu_fp8_engine_generates():
     eng = LLMEngine(cfg, device="cuda:0")
     out = eng.generate([[7, 8, 9] * 10], SamplingParams(max_tokens=4))[0]
     assert len(out.output_token_ids) == 4
+
+
+def test_gpu_moe_engine_generates():
+    """Qwen3-MoE on GPU: routing + capacity-padded bmm expert path over
+    the HIP kernels (eager decode: data-dependent shapes skip hipGraphs)."""
+    torch.manual_seed(0)
+    mc = get_model_config("tiny-qwen3-moe")
+    cfg = EngineConfig(
+        model=mc,
+        cache=CacheConfig(num_gpu_blocks=128),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=512, max_model_len=256
+        ),
+    )
+    eng = LLMEngine(cfg, device="cuda:0")
+    outs = eng.generate(
+        [[5, 3, 1] * 20, [9, 2] * 25], SamplingParams(max_tokens=6)
+    )
+    assert all(len(o.output_token_ids) == 6 for o in outs)
+
+    # GPU forward vs CPU reference engine: same weights (same seed),
+    # logits must agree
+    from fusioninfer_amd.engine.block_manager import BlockManager
+    from fusioninfer_amd.engine.sequence import Sequence
+
+    torch.manual_seed(0)
+    cfg2 = EngineConfig(
+        model=get_model_config("tiny-qwen3-moe"),
+        cache=CacheConfig(num_gpu_blocks=128),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=512, max_model_len=256
+        ),
+    )
+    eng_cpu = LLMEngine(cfg2, device="cpu")
+    prompt = list(range(40, 120))
+    for e, dev in ((eng, "cuda:0"), (eng_cpu, "cpu")):
+        pass
+    bm_g = BlockManager(eng.runner.num_gpu_blocks, 16)
+    s_g = Sequence("g", prompt, SamplingParams())
+    bm_g.allocate(s_g)
+    lg = eng.runner.execute_prefill([s_g], bm_g).float().cpu()
+    bm_c = BlockManager(eng_cpu.runner.num_gpu_blocks, 16)
+    s_c = Sequence("c", prompt, SamplingParams())
+    bm_c.allocate(s_c)
+    lc = eng_cpu.runner.execute_prefill([s_c], bm_c).float()
+    rel = (lg - lc).norm() / lc.norm()
+    assert rel.item() < 0.08, rel.item()

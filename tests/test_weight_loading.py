@@ -75,3 +75,44 @@ def test_safetensors_dir_loading(tmp_path):
         _forward_logits(dst, tokens).float(),
         _forward_logits(src, tokens).float(),
     )
+
+
+def test_moe_expert_weights_roundtrip():
+    """HF qwen3_moe expert names load into the stacked expert tensors and
+    change the model's output."""
+    import torch
+
+    from fusioninfer_amd.distributed import parallel_state as ps
+    from fusioninfer_amd.models.model import CausalLM
+    from fusioninfer_amd.models.registry import get_model_config
+    from fusioninfer_amd.models.weight_loader import load_hf_state_dict
+
+    ps.ensure_single_process()
+    torch.manual_seed(0)
+    cfg = get_model_config("tiny-qwen3-moe")
+    model = CausalLM(cfg)
+    mlp = model.layers[0].mlp
+    E, I, H = cfg.num_experts, cfg.moe_intermediate_size, cfg.hidden_size
+    tensors = []
+    torch.manual_seed(99)
+    want_gu = torch.randn(E, 2 * I, H) * 0.03
+    want_down = torch.randn(E, H, I) * 0.03
+    want_router = torch.randn(E, H) * 0.03
+    tensors.append(("model.layers.0.mlp.gate.weight", want_router))
+    for e in range(E):
+        tensors.append(
+            (f"model.layers.0.mlp.experts.{e}.gate_proj.weight", want_gu[e, :I])
+        )
+        tensors.append(
+            (f"model.layers.0.mlp.experts.{e}.up_proj.weight", want_gu[e, I:])
+        )
+        tensors.append(
+            (f"model.layers.0.mlp.experts.{e}.down_proj.weight", want_down[e])
+        )
+    n = load_hf_state_dict(model, tensors)
+    assert n == 1 + 3 * E
+    assert torch.allclose(mlp.gate_up.float(), want_gu.bfloat16().float())
+    assert torch.allclose(mlp.down.float(), want_down.bfloat16().float())
+    assert torch.allclose(
+        mlp.router_weight.float(), want_router.bfloat16().float()
+    )
