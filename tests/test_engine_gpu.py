@@ -245,8 +245,6 @@ def test_gpu_moe_engine_generates():
     )
     eng_cpu = LLMEngine(cfg2, device="cpu")
     prompt = list(range(40, 120))
-    for e, dev in ((eng, "cuda:0"), (eng_cpu, "cpu")):
-        pass
     bm_g = BlockManager(eng.runner.num_gpu_blocks, 16)
     s_g = Sequence("g", prompt, SamplingParams())
     bm_g.allocate(s_g)
