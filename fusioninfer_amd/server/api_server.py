@@ -109,15 +109,32 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             stop_token_ids=body.get("stop_token_ids") or [],
         )
 
-    async def _collect(q) -> List[int]:
+    def _parse_stops(body: Dict[str, Any]) -> List[str]:
+        stops = body.get("stop")
+        if stops is None:
+            return []
+        return [stops] if isinstance(stops, str) else list(stops)
+
+    async def _collect(q, stops=None, req_id=None):
+        """Drain a request's token stream. With OpenAI `stop` strings the
+        generated text is truncated BEFORE the first stop match and the
+        engine request is aborted. Returns (token_ids, text)."""
         loop = asyncio.get_event_loop()
         toks: List[int] = []
+        text = ""
         while True:
             tok, finished = await loop.run_in_executor(None, q.get)
             if tok is not None:
                 toks.append(tok)
+                if stops:
+                    text += decode_tokens([tok])
+                    for s in stops:
+                        i = text.find(s)
+                        if i >= 0:
+                            serving.abort(req_id)
+                            return toks, text[:i]
             if finished:
-                return toks
+                return toks, text if stops else decode_tokens(toks)
 
     async def _stream(q) -> AsyncGenerator:
         loop = asyncio.get_event_loop()
@@ -155,6 +172,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         lora = body.get("model")
         if lora not in serving.engine.active_loras():
             lora = None
+        stops = _parse_stops(body)
         pd_tag = request.headers.get("x-pd-tag")
         try:
             if pd_tag is not None:
@@ -204,7 +222,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
-        toks = await _collect(q)
+        toks, text = await _collect(q, stops=stops, req_id=req_id)
         return JSONResponse(
             {
                 "id": cid,
@@ -214,7 +232,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
                 "choices": [
                     {
                         "index": 0,
-                        "text": decode_tokens(toks),
+                        "text": text,
                         "token_ids": toks,
                         "finish_reason": "stop",
                     }
@@ -236,6 +254,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         )
         prompt_ids = encode_prompt(text, vocab)
         sampling = _sampling_from(body)
+        stops = _parse_stops(body)
         lora = body.get("model")
         if lora not in serving.engine.active_loras():
             lora = None
@@ -275,7 +294,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
-        toks = await _collect(q)
+        toks, out_text = await _collect(q, stops=stops, req_id=req_id)
         return JSONResponse(
             {
                 "id": cid,
@@ -287,7 +306,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
                         "index": 0,
                         "message": {
                             "role": "assistant",
-                            "content": decode_tokens(toks),
+                            "content": out_text,
                         },
                         "finish_reason": "stop",
                     }

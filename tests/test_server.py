@@ -151,3 +151,32 @@ def test_too_long_prompt_returns_400(app):
             assert "max_model_len" in r.json()["error"]["message"]
 
     asyncio.run(run())
+
+
+def test_stop_strings(app):
+    """OpenAI `stop`: generation halts at the stop string and the text is
+    truncated before it (byte-fallback tokenizer makes matching exact:
+    with temperature 0 on the tiny model we learn which byte repeats by
+    generating once, then stop on it)."""
+
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "abcabc", "max_tokens": 8,
+                      "temperature": 0, "ignore_eos": True},
+            )
+            full = r.json()["choices"][0]["text"]
+            assert len(full) > 1
+            stop_ch = full[1]
+            r2 = await c.post(
+                "/v1/completions",
+                json={"prompt": "abcabc", "max_tokens": 8, "stop": stop_ch,
+                      "temperature": 0, "ignore_eos": True},
+            )
+            body = r2.json()
+            text = body["choices"][0]["text"]
+            assert stop_ch not in text
+            assert text == full.split(stop_ch)[0]
+
+    asyncio.run(run())
