@@ -171,22 +171,28 @@ class MoEMLP(nn.Module):
         self.e_end = self.e_start + E // tp
         H, inter = cfg.hidden_size, cfg.moe_intermediate_size
         base = torch.initial_seed() % (2**62)
+        dev = torch.empty(0).device  # honor the ambient torch.device context
         gen = torch.Generator(device="cpu")
         gen.manual_seed(base)
+        # experts drawn on the CPU from per-(layer, expert) seeds so every
+        # rank generates identical experts regardless of the EP layout
         self.router_weight = nn.Parameter(
-            (torch.randn(E, H, generator=gen) * 0.02).to(torch.bfloat16),
+            (torch.randn(E, H, generator=gen, device="cpu") * 0.02)
+            .to(dtype=torch.bfloat16, device=dev),
             requires_grad=False,
         )
         gup, down = [], []
         for e in range(self.e_start, self.e_end):
             gen.manual_seed(base + 7919 * (e + 1))
-            gup.append(torch.randn(2 * inter, H, generator=gen) * 0.02)
-            down.append(torch.randn(H, inter, generator=gen) * 0.02)
+            gup.append(torch.randn(2 * inter, H, generator=gen, device="cpu") * 0.02)
+            down.append(torch.randn(H, inter, generator=gen, device="cpu") * 0.02)
         self.gate_up = nn.Parameter(
-            torch.stack(gup).to(torch.bfloat16), requires_grad=False
+            torch.stack(gup).to(dtype=torch.bfloat16, device=dev),
+            requires_grad=False,
         )  # [E_local, 2*I, H]
         self.down = nn.Parameter(
-            torch.stack(down).to(torch.bfloat16), requires_grad=False
+            torch.stack(down).to(dtype=torch.bfloat16, device=dev),
+            requires_grad=False,
         )  # [E_local, H, I]
 
     def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
