@@ -222,7 +222,22 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
+        # OpenAI `n`: extra choices run as sibling requests — prefix
+        # caching dedups the shared-prompt KV, so the n-1 extra prefills
+        # recompute at most one block each
+        n = max(int(body.get("n", 1)), 1)
+        extra = []
+        for i in range(1, n):
+            s_i = _sampling_from(body)
+            if s_i.seed is not None:
+                s_i.seed += i  # distinct choices under a fixed seed
+            extra.append(serving.submit(prompt_ids, s_i, lora_name=lora))
+        choices = []
         toks, text = await _collect(q, stops=stops, req_id=req_id)
+        choices.append((toks, text))
+        for rid_i, q_i in extra:
+            choices.append(await _collect(q_i, stops=stops, req_id=rid_i))
+        total_completion = sum(len(t) for t, _ in choices)
         return JSONResponse(
             {
                 "id": cid,
@@ -231,16 +246,17 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
                 "model": model_name,
                 "choices": [
                     {
-                        "index": 0,
-                        "text": text,
-                        "token_ids": toks,
+                        "index": i,
+                        "text": c_text,
+                        "token_ids": c_toks,
                         "finish_reason": "stop",
                     }
+                    for i, (c_toks, c_text) in enumerate(choices)
                 ],
                 "usage": {
                     "prompt_tokens": len(prompt_ids),
-                    "completion_tokens": len(toks),
-                    "total_tokens": len(prompt_ids) + len(toks),
+                    "completion_tokens": total_completion,
+                    "total_tokens": len(prompt_ids) + total_completion,
                 },
             }
         )

@@ -180,3 +180,23 @@ def test_stop_strings(app):
             assert text == full.split(stop_ch)[0]
 
     asyncio.run(run())
+
+
+def test_n_choices(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": [5, 6, 7] * 8, "max_tokens": 3, "n": 3,
+                      "temperature": 0.8, "seed": 42, "ignore_eos": True},
+            )
+            body = r.json()
+            assert len(body["choices"]) == 3
+            assert [ch["index"] for ch in body["choices"]] == [0, 1, 2]
+            assert all(len(ch["token_ids"]) == 3 for ch in body["choices"])
+            assert body["usage"]["completion_tokens"] == 9
+            # distinct seeds -> at least two distinct outputs (overwhelmingly)
+            outs = {tuple(ch["token_ids"]) for ch in body["choices"]}
+            assert len(outs) >= 2
+
+    asyncio.run(run())
