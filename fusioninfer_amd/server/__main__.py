@@ -39,6 +39,9 @@ def parse_args(argv=None):
                    help="GiB of CPU swap for preempted sequences (0=recompute)")
     p.add_argument("--enable-lora", action="store_true")
     p.add_argument("--max-loras", type=int, default=8)
+    p.add_argument("--max-cpu-loras", type=int, default=16,
+                   help="accepted for reference-surface parity; adapters "
+                        "are small enough to stay resident")
     p.add_argument("--lora-modules", nargs="*", default=[],
                    help="adapters to register at startup: name[=rank]")
     p.add_argument("--enable-prefix-caching", action="store_true")

@@ -214,7 +214,12 @@ def test_gpu_fp8_engine_generates():
 
 def test_gpu_moe_engine_generates():
     """Qwen3-MoE on GPU: routing + capacity-padded bmm expert path over
-    the HIP kernels (eager decode: data-dependent shapes skip hipGraphs)."""
+    the HIP kernels (eager decode: data-dependent shapes skip hipGraphs).
+    The MoE layer is checked against an on-DEVICE per-expert fp32 oracle
+    (cross-device comparisons are unstable: random-init routers have
+    near-uniform probabilities, so CPU/GPU rounding flips top-k picks)."""
+    import torch.nn.functional as F
+
     torch.manual_seed(0)
     mc = get_model_config("tiny-qwen3-moe")
     cfg = EngineConfig(
@@ -230,28 +235,21 @@ def test_gpu_moe_engine_generates():
     )
     assert all(len(o.output_token_ids) == 6 for o in outs)
 
-    # GPU forward vs CPU reference engine: same weights (same seed),
-    # logits must agree
-    from fusioninfer_amd.engine.block_manager import BlockManager
-    from fusioninfer_amd.engine.sequence import Sequence
-
-    torch.manual_seed(0)
-    cfg2 = EngineConfig(
-        model=get_model_config("tiny-qwen3-moe"),
-        cache=CacheConfig(num_gpu_blocks=128),
-        scheduler=SchedulerConfig(
-            max_num_seqs=8, max_num_batched_tokens=512, max_model_len=256
-        ),
-    )
-    eng_cpu = LLMEngine(cfg2, device="cpu")
-    prompt = list(range(40, 120))
-    bm_g = BlockManager(eng.runner.num_gpu_blocks, 16)
-    s_g = Sequence("g", prompt, SamplingParams())
-    bm_g.allocate(s_g)
-    lg = eng.runner.execute_prefill([s_g], bm_g).float().cpu()
-    bm_c = BlockManager(eng_cpu.runner.num_gpu_blocks, 16)
-    s_c = Sequence("c", prompt, SamplingParams())
-    bm_c.allocate(s_c)
-    lc = eng_cpu.runner.execute_prefill([s_c], bm_c).float()
-    rel = (lg - lc).norm() / lc.norm()
-    assert rel.item() < 0.08, rel.item()
+    moe = eng.runner.model.layers[0].mlp
+    x = torch.randn(9, cfg.model.hidden_size, dtype=torch.bfloat16,
+                    device="cuda:0")
+    got = moe(x).float()
+    logits = x.float() @ moe.router_weight.float().T
+    probs = torch.softmax(logits, -1)
+    topv, topi = probs.topk(moe.top_k, -1)
+    topv = topv / topv.sum(-1, keepdim=True)
+    exp = torch.zeros_like(x, dtype=torch.float32)
+    for t in range(x.shape[0]):
+        for k in range(moe.top_k):
+            e = int(topi[t, k])
+            gu = x[t].float() @ moe.gate_up[e].float().T
+            g, u = gu.chunk(2)
+            y = (F.silu(g) * u) @ moe.down[e].float().T
+            exp[t] += float(topv[t, k]) * y
+    rel = (got - exp).norm() / exp.norm()
+    assert rel.item() < 0.05, rel.item()
