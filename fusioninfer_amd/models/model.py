@@ -186,14 +186,21 @@ class MoEMLP(nn.Module):
             gen.manual_seed(base + 7919 * (e + 1))
             gup.append(torch.randn(2 * inter, H, generator=gen, device="cpu") * 0.02)
             down.append(torch.randn(H, inter, generator=gen, device="cpu") * 0.02)
-        self.gate_up = nn.Parameter(
-            torch.stack(gup).to(dtype=torch.bfloat16, device=dev),
+        # stored PRE-TRANSPOSED ([in, out] per expert) so both bmm B
+        # operands are contiguous: hipBLASLt/rocBLAS pick a custom
+        # Stream-K kernel for some transposed-view strided-batch shapes
+        # (e.g. [8,890,768]x[8,768,2048]^T) that memory-faults on gfx950
+        # — reproduced standalone, ROCm 7.2
+        self.gate_up_t = nn.Parameter(
+            torch.stack([w.T.contiguous() for w in gup]).to(
+                dtype=torch.bfloat16, device=dev),
             requires_grad=False,
-        )  # [E_local, 2*I, H]
-        self.down = nn.Parameter(
-            torch.stack(down).to(dtype=torch.bfloat16, device=dev),
+        )  # [E_local, H, 2*I]
+        self.down_t = nn.Parameter(
+            torch.stack([w.T.contiguous() for w in down]).to(
+                dtype=torch.bfloat16, device=dev),
             requires_grad=False,
-        )  # [E_local, H, I]
+        )  # [E_local, I, H]
 
     def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
         T, H = x.shape
@@ -225,12 +232,10 @@ class MoEMLP(nn.Module):
             rows = tok_of[local]
             xpad = torch.zeros(E_local * cap, H, dtype=x.dtype, device=x.device)
             xpad[dest] = x[rows]
-            gu = torch.bmm(
-                xpad.view(E_local, cap, H), self.gate_up.transpose(1, 2)
-            )
+            gu = torch.bmm(xpad.view(E_local, cap, H), self.gate_up_t)
             act = ops.silu_and_mul(gu.view(E_local * cap, -1))
             y = torch.bmm(
-                act.view(E_local, cap, -1), self.down.transpose(1, 2)
+                act.view(E_local, cap, -1), self.down_t
             ).view(E_local * cap, H)
             out.index_add_(0, rows, y[dest].float() * w_of[local, None])
         out = out.to(x.dtype)

@@ -80,8 +80,9 @@ def load_hf_state_dict(
             le = e - mlp.e_start
             gw = moe_pending.pop(keys[0])
             uw = moe_pending.pop(keys[1])
-            mlp.gate_up.data[le].copy_(
-                torch.cat([gw, uw], dim=0).to(mlp.gate_up.dtype)
+            # experts are stored pre-transposed ([H, 2I]) — see MoEMLP
+            mlp.gate_up_t.data[le].copy_(
+                torch.cat([gw, uw], dim=0).T.to(mlp.gate_up_t.dtype)
             )
 
     for name, w in tensors:
@@ -134,7 +135,9 @@ def load_hf_state_dict(
                     continue  # EP: expert lives on another rank
                 which = parts[6].replace("_proj", "")  # gate/up/down
                 if which == "down":
-                    mlp.down.data[e - mlp.e_start].copy_(w.to(mlp.down.dtype))
+                    mlp.down_t.data[e - mlp.e_start].copy_(
+                        w.T.to(mlp.down_t.dtype)
+                    )
                 else:
                     moe_pending[(li, e, which)] = w
                     try_fuse_expert(li, e)

@@ -374,7 +374,7 @@ def _ep_worker(rank, port, results):
         cfg2.parallel.tensor_parallel_size = 2
         runner2 = ModelRunner(cfg2, "cpu")
         # experts are sharded across the group (EP): 4 of 8 per rank
-        assert runner2.model.layers[0].mlp.gate_up.shape[0] == 4
+        assert runner2.model.layers[0].mlp.gate_up_t.shape[0] == 4
         runner2.allocate_kv_caches()
         bm2 = BlockManager(64, 16)
         seq2 = Sequence("s2", PROMPT, SamplingParams())

@@ -335,9 +335,9 @@ def test_moe_engine_generates_and_routes():
     for t in range(x.shape[0]):
         for k in range(moe.top_k):
             e = int(topi[t, k])
-            gu = x[t].float() @ moe.gate_up[e].float().T
+            gu = x[t].float() @ moe.gate_up_t[e].float()
             g, u = gu.chunk(2)
-            y = (F.silu(g) * u) @ moe.down[e].float().T
+            y = (F.silu(g) * u) @ moe.down_t[e].float()
             exp[t] += float(topv[t, k]) * y
     # the layer computes in bf16 GEMMs; compare loosely
     rel = (got - exp).norm() / exp.norm()

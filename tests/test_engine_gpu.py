@@ -247,9 +247,9 @@ def test_gpu_moe_engine_generates():
     for t in range(x.shape[0]):
         for k in range(moe.top_k):
             e = int(topi[t, k])
-            gu = x[t].float() @ moe.gate_up[e].float().T
+            gu = x[t].float() @ moe.gate_up_t[e].float()
             g, u = gu.chunk(2)
-            y = (F.silu(g) * u) @ moe.down[e].float().T
+            y = (F.silu(g) * u) @ moe.down_t[e].float()
             exp[t] += float(topv[t, k]) * y
     rel = (got - exp).norm() / exp.norm()
     assert rel.item() < 0.05, rel.item()

@@ -111,8 +111,12 @@ def test_moe_expert_weights_roundtrip():
         )
     n = load_hf_state_dict(model, tensors)
     assert n == 1 + 3 * E
-    assert torch.allclose(mlp.gate_up.float(), want_gu.bfloat16().float())
-    assert torch.allclose(mlp.down.float(), want_down.bfloat16().float())
+    assert torch.allclose(
+        mlp.gate_up_t.float(), want_gu.transpose(1, 2).bfloat16().float()
+    )
+    assert torch.allclose(
+        mlp.down_t.float(), want_down.transpose(1, 2).bfloat16().float()
+    )
     assert torch.allclose(
         mlp.router_weight.float(), want_router.bfloat16().float()
     )
