@@ -181,11 +181,24 @@ class MoEMLP(nn.Module):
             .to(dtype=torch.bfloat16, device=dev),
             requires_grad=False,
         )
+        # per-expert seeded draws; on GPU use the device RNG (Philox is
+        # seed-deterministic and rank-independent, so EP layouts still
+        # agree) — CPU-generating 128 experts x 48 layers took minutes
+        if dev.type == "cuda":
+            egen = torch.Generator(device=dev)
+            edev = dev
+        else:
+            egen = gen
+            edev = "cpu"
         gup, down = [], []
         for e in range(self.e_start, self.e_end):
-            gen.manual_seed(base + 7919 * (e + 1))
-            gup.append(torch.randn(2 * inter, H, generator=gen, device="cpu") * 0.02)
-            down.append(torch.randn(H, inter, generator=gen, device="cpu") * 0.02)
+            egen.manual_seed(base + 7919 * (e + 1))
+            gup.append(
+                torch.randn(2 * inter, H, generator=egen, device=edev) * 0.02
+            )
+            down.append(
+                torch.randn(H, inter, generator=egen, device=edev) * 0.02
+            )
         # stored PRE-TRANSPOSED ([in, out] per expert) so both bmm B
         # operands are contiguous: hipBLASLt/rocBLAS pick a custom
         # Stream-K kernel for some transposed-view strided-batch shapes
