@@ -154,6 +154,11 @@ class Scheduler:
             and len(self.running) < self.cfg.max_num_seqs
         ):
             seq = self.waiting[0]
+            if seq.status == SeqStatus.SWAPPED:
+                # the head must resume through the swap-in path above (it
+                # lacked blocks this step); admitting it as a fresh prompt
+                # would discard its computed state and leak its swap staging
+                break
             if not self.bm.can_allocate(seq.num_prompt_tokens):
                 break
             self.waiting.popleft()

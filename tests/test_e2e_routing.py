@@ -250,3 +250,49 @@ def test_metrics_scrape_parsing():
     m = parse_vllm_metrics(text)
     assert m["gpu_cache_usage_perc"] == 0.25
     assert m["num_requests_waiting"] == 3.0
+
+
+def test_pd_out_of_order_claims():
+    """Two PD requests; the decode claims arrive in reverse order — the
+    tag matching must pair each claim with its own KV batch."""
+    import torch
+
+    from fusioninfer_amd.distributed.kv_transfer import make_inmemory_pair
+    from fusioninfer_amd.engine.sequence import SamplingParams
+
+    def make_cfg():
+        return EngineConfig(
+            model=get_model_config("tiny-qwen3"),
+            cache=CacheConfig(num_gpu_blocks=64),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=256, max_model_len=128
+            ),
+        )
+
+    prod, cons = make_inmemory_pair()
+    torch.manual_seed(8)
+    pre = ServingEngine(make_cfg(), device="cpu", kv_connector=prod)
+    torch.manual_seed(8)
+    dec = ServingEngine(make_cfg(), device="cpu", kv_connector=cons)
+    try:
+        pa = [3, 1, 4] * 8
+        pb = [2, 7, 1, 8] * 6
+        tag_a, first_a = pre.prefill_via_pd(pa)
+        tag_b, first_b = pre.prefill_via_pd(pb)
+        # claim B first, then A
+        _, qb = dec.submit_imported(tag_b, SamplingParams(max_tokens=3,
+                                                          temperature=0.0))
+        _, qa = dec.submit_imported(tag_a, SamplingParams(max_tokens=3,
+                                                          temperature=0.0))
+        tb = qb.get(timeout=10)[0]
+        ta = qa.get(timeout=10)[0]
+        assert tb == first_b and ta == first_a
+        # drain
+        for q in (qa, qb):
+            while True:
+                _, fin = q.get(timeout=10)
+                if fin:
+                    break
+    finally:
+        pre.shutdown()
+        dec.shutdown()

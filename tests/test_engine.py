@@ -344,3 +344,72 @@ def test_moe_engine_generates_and_routes():
     assert rel.item() < 0.05, rel.item()
     # routing uses more than one expert across tokens
     assert len(set(topi.flatten().tolist())) > 1
+
+
+def test_swap_budget_exhaustion_falls_back_to_recompute():
+    """A swap budget too small for one sequence's KV forces the recompute
+    path; generation still completes and frees all blocks."""
+    import torch
+
+    from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from fusioninfer_amd.models.registry import get_model_config
+
+    torch.manual_seed(9)
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=12, enable_prefix_caching=False,
+                          swap_space_gb=1e-9),  # ~1 byte: never fits
+        scheduler=SchedulerConfig(
+            max_num_seqs=4, max_num_batched_tokens=256, max_model_len=128
+        ),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    prompts = [[(11 * i + j) % 500 for j in range(40)] for i in range(3)]
+    outs = eng.generate(
+        prompts, SamplingParams(max_tokens=40, temperature=0.0,
+                                ignore_eos=True)
+    )
+    assert all(len(o.output_token_ids) == 40 for o in outs)
+    assert eng.num_swap_outs == 0
+    assert eng.block_manager.num_free() == eng.block_manager.num_blocks
+
+
+def test_abort_swapped_request_frees_swap_space():
+    import torch
+
+    from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from fusioninfer_amd.models.registry import get_model_config
+
+    torch.manual_seed(9)
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=12, enable_prefix_caching=False,
+                          swap_space_gb=1.0),
+        scheduler=SchedulerConfig(
+            max_num_seqs=4, max_num_batched_tokens=256, max_model_len=128
+        ),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    ids = [
+        eng.add_request([(7 * i + j) % 500 for j in range(40)],
+                        SamplingParams(max_tokens=60, temperature=0.0,
+                                       ignore_eos=True))
+        for i in range(3)
+    ]
+    # run until something swaps, then abort the swapped request
+    for _ in range(200):
+        eng.step()
+        if eng._swapped:
+            break
+    assert eng._swapped, "no swap happened"
+    victim = next(iter(eng._swapped))
+    assert eng.abort_request(victim)
+    assert victim not in eng._swapped
+    while eng.has_unfinished():
+        eng.step()
+    assert eng._swap_bytes == 0
+    assert eng.block_manager.num_free() == eng.block_manager.num_blocks
