@@ -64,6 +64,7 @@ class LLMEngine:
         self._swapped: Dict[str, "torch.Tensor"] = {}
         self._swap_bytes = 0
         self.num_swap_outs = 0
+        self.num_preemptions = 0
         if (cfg.cache.swap_space_gb > 0
                 and cfg.parallel.tensor_parallel_size == 1
                 and cfg.parallel.pipeline_parallel_size == 1):
@@ -298,6 +299,7 @@ class LLMEngine:
             self._swap_in(seq)
         if batch.is_empty:
             return []
+        self.num_preemptions += len(batch.preempted)
         payload = self.runner.build_batch_payload(
             batch.prefill_seqs, batch.prefill_chunks, batch.decode_seqs,
             self.block_manager,

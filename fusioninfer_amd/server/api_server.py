@@ -87,6 +87,8 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             ("e2e_request_latency_seconds_count", "counter"),
             ("engine_step_seconds_sum", "counter"),
             ("engine_step_seconds_count", "counter"),
+            ("num_preemptions_total", "counter"),
+            ("num_swap_outs_total", "counter"),
         ]:
             lines.append(f"# TYPE vllm:{name} {mtype}")
             lines.append(
