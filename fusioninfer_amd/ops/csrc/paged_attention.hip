@@ -169,6 +169,10 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       kv_base = kv_base_of(chunk + kNWaves);
       load_k(kv_base, kraw);
     }
+    // fence: stop the scheduler from hoisting phase-B work (and its live
+    // ranges) above the softmax — cross-phase overlap doubled VGPRs
+    // (G=8 spills under the fences; its register file is already full)
+    if (G <= 4) __builtin_amdgcn_sched_barrier(0);
     // reduce over the 4 dim-quarters (lanes 4t..4t+3)
 #pragma unroll
     for (int g = 0; g < G; ++g) {
@@ -202,6 +206,7 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       l[g] = l[g] * alpha[g] + psum * 0.25f;
     }
 
+    if (G <= 4) __builtin_amdgcn_sched_barrier(0);
     // ---- phase B: PV accumulate; lane covers dims {2l, 2l+1} ----
     const int nvalid = min(ctx - chunk * kBlockSz, kBlockSz);
     const CT* v_rows = v_cache + kv_base_cur;
@@ -211,9 +216,32 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       acc[g][1] *= alpha[g];
     }
     if (lane < D / 2) {
-      for (int t = 0; t < nvalid; ++t) {
+      // G<=4: preload ALL 16 token rows' bits first (p is 0 for padding
+      // tokens — phase A masks every invalid position — so processing the
+      // full block is safe and the 16 loads overlap instead of serializing)
+      u32 vball[G <= 4 ? kBlockSz : 1];
+      if (G <= 4) {
+#pragma unroll
+        for (int t = 0; t < kBlockSz; ++t)
+          vball[t] = FP8
+              ? static_cast<u32>(*reinterpret_cast<const u16*>(
+                    v_rows + t * D + 2 * lane))
+              : *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
+      }
+      const int tmax = (G <= 4) ? kBlockSz : nvalid;
+      for (int t = 0; t < tmax; ++t) {
         float v0, v1;
-        if (FP8) {
+        if (G <= 4) {
+          if (FP8) {
+            float vf[4];
+            unpack_fp8x4(vball[t], vf);
+            v0 = vf[0];
+            v1 = vf[1];
+          } else {
+            v0 = bf16_to_f32(static_cast<u16>(vball[t] & 0xffff));
+            v1 = bf16_to_f32(static_cast<u16>(vball[t] >> 16));
+          }
+        } else if (FP8) {
           const u16 vb = *reinterpret_cast<const u16*>(v_rows + t * D + 2 * lane);
           float vf[4];
           unpack_fp8x4(vb, vf);
