@@ -198,10 +198,24 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         if body.get("stream"):
             async def sse():
                 done = False
+                text = ""
                 try:
                     async for tok, finished in _stream(q):
                         done = finished
                         delta = decode_tokens([tok]) if tok is not None else ""
+                        if stops and tok is not None:
+                            # hold back text that could extend into a stop
+                            # string; cut the stream at the first match
+                            text += delta
+                            hit = min(
+                                (i for i in (text.find(s) for s in stops)
+                                 if i >= 0),
+                                default=-1,
+                            )
+                            if hit >= 0:
+                                delta = text[len(text) - len(delta):hit] \
+                                    if hit >= len(text) - len(delta) else ""
+                                finished = True
                         chunk = {
                             "id": cid,
                             "object": "text_completion",
@@ -217,6 +231,10 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
                             ],
                         }
                         yield f"data: {json.dumps(chunk)}\n\n"
+                        if finished and not done:
+                            done = True
+                            serving.abort(req_id)
+                            break
                     yield "data: [DONE]\n\n"
                 finally:
                     if not done:  # client disconnected mid-stream

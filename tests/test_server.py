@@ -200,3 +200,34 @@ def test_n_choices(app):
             assert len(outs) >= 2
 
     asyncio.run(run())
+
+
+def test_stop_strings_streaming(app):
+    """Streaming requests honor `stop`: the stream ends at the match and
+    the concatenated deltas exclude the stop string."""
+
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "xyzxyz", "max_tokens": 8,
+                      "temperature": 0, "ignore_eos": True},
+            )
+            full = r.json()["choices"][0]["text"]
+            stop_ch = full[1]
+            text = ""
+            n_chunks = 0
+            async with c.stream("POST", "/v1/completions", json={
+                "prompt": "xyzxyz", "max_tokens": 8, "stream": True,
+                "stop": stop_ch, "temperature": 0, "ignore_eos": True,
+            }) as resp:
+                async for line in resp.aiter_lines():
+                    if line.startswith("data: ") and "[DONE]" not in line:
+                        chunk = json.loads(line[6:])
+                        text += chunk["choices"][0]["text"]
+                        n_chunks += 1
+            assert stop_ch not in text
+            assert text == full.split(stop_ch)[0]
+            assert n_chunks <= len(full)
+
+    asyncio.run(run())
