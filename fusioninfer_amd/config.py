@@ -113,5 +113,8 @@ class EngineConfig:
     scheduler: SchedulerConfig = dataclasses.field(default_factory=SchedulerConfig)
     parallel: ParallelConfig = dataclasses.field(default_factory=ParallelConfig)
     kv_transfer: KVTransferConfig = dataclasses.field(default_factory=KVTransferConfig)
+    # vLLM --speculative-config parity: None = off; set to a
+    # SpeculativeConfig (engine/spec_decode.py) for ngram drafting
+    speculative: Optional[object] = None
     seed: int = 0
     enforce_eager: bool = False            # True disables hipGraph decode capture

@@ -129,6 +129,18 @@ class BlockManager:
             self.ref_count[blk] = 1
             seq.block_ids.append(blk)
 
+    def extra_blocks_for(self, seq: Sequence, position: int) -> int:
+        """Blocks missing for `position` to have a slot (spec-decode draft
+        tails extend past the single decode slot the scheduler reserved)."""
+        need = position // self.block_size + 1
+        return max(need - len(seq.block_ids), 0)
+
+    def append_slots_upto(self, seq: Sequence, position: int) -> None:
+        for _ in range(self.extra_blocks_for(seq, position)):
+            blk = self._pop_free_block()
+            self.ref_count[blk] = 1
+            seq.block_ids.append(blk)
+
     def free(self, seq: Sequence) -> None:
         for blk in seq.block_ids:
             self.ref_count[blk] -= 1

@@ -23,10 +23,15 @@ from fusioninfer_amd.engine.sequence import SamplingParams, Sequence, SeqStatus
 
 
 class RequestOutput:
-    def __init__(self, seq: Sequence):
+    def __init__(self, seq: Sequence, new_token_ids=None):
         self.request_id = seq.seq_id
         self.prompt_token_ids = seq.prompt_token_ids
         self.output_token_ids = list(seq.output_token_ids)
+        # tokens produced THIS step: one for regular decode, up to k+1 for
+        # an accepted speculative draft (streaming pushes each)
+        self.new_token_ids = list(new_token_ids) if new_token_ids else (
+            self.output_token_ids[-1:] if self.output_token_ids else []
+        )
         self.finished = seq.is_finished()
         self.logprobs = list(seq.logprobs)
         self.ttft = seq.ttft
@@ -70,6 +75,11 @@ class LLMEngine:
                 and cfg.parallel.pipeline_parallel_size == 1):
             self.scheduler.swap_out_fn = self._swap_out
         self.sampler = Sampler(cfg.seed, device)
+        from fusioninfer_amd.engine.spec_decode import build_proposer
+
+        self.proposer = build_proposer(cfg.speculative)
+        self.num_spec_draft_tokens = 0
+        self.num_spec_accepted_tokens = 0
         self.runner.capture_decode_graphs()
         self._req_counter = itertools.count()
         self.seqs: Dict[str, Sequence] = {}
@@ -300,14 +310,27 @@ class LLMEngine:
         if batch.is_empty:
             return []
         self.num_preemptions += len(batch.preempted)
-        payload = self.runner.build_batch_payload(
-            batch.prefill_seqs, batch.prefill_chunks, batch.decode_seqs,
-            self.block_manager,
-        )
+        spec_drafts = None
+        if self.proposer is not None and batch.decode_seqs:
+            spec_drafts = self._propose_drafts(batch.decode_seqs)
+        if spec_drafts is not None:
+            payload = self.runner.build_spec_payload(
+                batch.prefill_seqs, batch.prefill_chunks, batch.decode_seqs,
+                spec_drafts, self.block_manager,
+            )
+        else:
+            payload = self.runner.build_batch_payload(
+                batch.prefill_seqs, batch.prefill_chunks, batch.decode_seqs,
+                self.block_manager,
+            )
         t_payload = time.monotonic()
         if tp > 1:
             self._ps.tp_broadcast_object(payload)
-        logits = self.runner.run_batch(payload)
+        logits = (
+            self.runner.run_prefill(payload)
+            if spec_drafts is not None
+            else self.runner.run_batch(payload)
+        )
         if self._timing:
             import torch as _t
 
@@ -326,15 +349,56 @@ class LLMEngine:
                 seq.num_computed_tokens or seq.num_cached_tokens
             ) + chunk
             self.num_prefilled_tokens += chunk
-        # logits rows: completing prefill chunks first, then decode rows
-        sample_seqs = [
-            s for s, smp in zip(batch.prefill_seqs, payload["sample"]) if smp
-        ] + batch.decode_seqs
-        if not sample_seqs:
-            return []
-        logits_f = logits.float()
-        next_tokens = self.sampler.sample(logits_f, sample_seqs)
-        next_tokens = next_tokens.tolist()
+        if spec_drafts is not None:
+            outputs = self._finish_spec_step(batch, payload, logits,
+                                             spec_drafts)
+        else:
+            # logits rows: completing prefill chunks first, then decode rows
+            sample_seqs = [
+                s for s, smp in zip(batch.prefill_seqs, payload["sample"])
+                if smp
+            ] + batch.decode_seqs
+            if not sample_seqs:
+                return []
+            outputs = self._sample_and_emit(sample_seqs, logits.float())
+        if self._timing and hasattr(self, "_t_fwd_mark"):
+            kind = "P" if batch.prefill_seqs else "D"
+            self._tacc[kind][3] += time.monotonic() - self._t_fwd_mark
+            n = self._tacc[kind][4]
+            if n % 50 == 0:
+                a = self._tacc[kind]
+                print(
+                    f"[timing {kind}] n={n} sched={a[0]/n*1e3:.2f}ms "
+                    f"payload={a[1]/n*1e3:.2f}ms fwd={a[2]/n*1e3:.2f}ms "
+                    f"sample+book={a[3]/n*1e3:.2f}ms", flush=True,
+                )
+        self.step_time_sum += time.monotonic() - step_t0
+        self.num_steps += 1
+        return outputs
+
+    # ------------------------------------------------- sampling / emission
+    def _emit_tokens(self, seq: Sequence, toks) -> RequestOutput:
+        """Append this step's tokens (one for regular decode, up to k+1
+        for an accepted speculative draft) with per-token stop checks."""
+        new = []
+        for tok in toks:
+            seq.append_token(int(tok))
+            new.append(int(tok))
+            self.num_generated_tokens += 1
+            if seq.check_stop():
+                seq.finish_time = time.monotonic()
+                self.scheduler.finish(seq)
+                self.num_finished += 1
+                if seq.ttft is not None:
+                    self.ttft_sum += seq.ttft
+                self.e2e_latency_sum += seq.finish_time - seq.arrival_time
+                del self.seqs[seq.seq_id]
+                break
+        return RequestOutput(seq, new_token_ids=new)
+
+    def _sample_and_emit(self, sample_seqs, logits_f) -> List[RequestOutput]:
+        """Regular sampling path: logits_f row i belongs to sample_seqs[i]."""
+        next_tokens = self.sampler.sample(logits_f, sample_seqs).tolist()
         # optional logprobs for requests that asked
         lp_rows = [
             i for i, s in enumerate(sample_seqs)
@@ -354,33 +418,66 @@ class LLMEngine:
                 s.logprobs.append(
                     (float(lp[j, next_tokens[i]]), entry_top)
                 )
+        return [
+            self._emit_tokens(seq, [tok])
+            for seq, tok in zip(sample_seqs, next_tokens)
+        ]
 
+    # ------------------------------------------------ speculative decoding
+    def _propose_drafts(self, decode_seqs):
+        """n-gram drafts per decode sequence; a draft is dropped when the
+        block pool cannot cover its tail. None = nothing to speculate
+        (step falls back to the fast pure-decode / hipGraph path)."""
+        bm = self.block_manager
+        drafts, any_d = [], False
+        for s in decode_seqs:
+            d = self.proposer.propose(s)
+            if d:
+                last_pos = s.num_tokens - 1 + len(d)
+                if bm.extra_blocks_for(s, last_pos) > bm.num_free():
+                    d = []
+                else:
+                    bm.append_slots_upto(s, last_pos)
+            drafts.append(list(d))
+            any_d = any_d or bool(d)
+        return drafts if any_d else None
+
+    def _finish_spec_step(self, batch, payload, logits, drafts):
+        """Greedy draft acceptance. Logits row layout (see
+        build_spec_payload): [completing-prefill rows | per decode seq,
+        1 + k_i rows]. Draft-less decode seqs go through the regular
+        sampler; drafted seqs accept the longest matching prefix plus the
+        corrected token — token-exact with non-speculative greedy."""
+        logits_f = logits.float()
+        n_prefill = len(batch.prefill_seqs)
+        pf_flags = payload["sample"][:n_prefill]
+        samp_seqs = [s for s, smp in zip(batch.prefill_seqs, pf_flags) if smp]
+        samp_rows = list(range(len(samp_seqs)))
+        spec = []  # (seq, draft, first logits row of its span)
+        r = len(samp_seqs)
+        for s, d in zip(batch.decode_seqs, drafts):
+            if d:
+                spec.append((s, d, r))
+            else:
+                samp_rows.append(r)
+                samp_seqs.append(s)
+            r += 1 + len(d)
         outputs: List[RequestOutput] = []
-        for seq, tok in zip(sample_seqs, next_tokens):
-            seq.append_token(int(tok))
-            self.num_generated_tokens += 1
-            if seq.check_stop():
-                seq.finish_time = time.monotonic()
-                self.scheduler.finish(seq)
-                self.num_finished += 1
-                if seq.ttft is not None:
-                    self.ttft_sum += seq.ttft
-                self.e2e_latency_sum += seq.finish_time - seq.arrival_time
-                del self.seqs[seq.seq_id]
-            outputs.append(RequestOutput(seq))
-        if self._timing and hasattr(self, "_t_fwd_mark"):
-            kind = "P" if batch.prefill_seqs else "D"
-            self._tacc[kind][3] += time.monotonic() - self._t_fwd_mark
-            n = self._tacc[kind][4]
-            if n % 50 == 0:
-                a = self._tacc[kind]
-                print(
-                    f"[timing {kind}] n={n} sched={a[0]/n*1e3:.2f}ms "
-                    f"payload={a[1]/n*1e3:.2f}ms fwd={a[2]/n*1e3:.2f}ms "
-                    f"sample+book={a[3]/n*1e3:.2f}ms", flush=True,
-                )
-        self.step_time_sum += time.monotonic() - step_t0
-        self.num_steps += 1
+        if samp_seqs:
+            outputs.extend(
+                self._sample_and_emit(samp_seqs, logits_f[samp_rows])
+            )
+        if spec:
+            greedy = logits_f.argmax(dim=-1).tolist()
+            for s, d, r0 in spec:
+                k = len(d)
+                g = greedy[r0: r0 + k + 1]
+                m = 0
+                while m < k and d[m] == g[m]:
+                    m += 1
+                self.num_spec_draft_tokens += k
+                self.num_spec_accepted_tokens += m
+                outputs.append(self._emit_tokens(s, d[:m] + [g[m]]))
         return outputs
 
     # --------------------------------------------------------- TP workers

@@ -223,12 +223,19 @@ class ModelRunner:
         )
         meta.lora = self._lora_batch(payload, None)
         ids = torch.tensor(payload["ids"], dtype=torch.long, device=dev)
-        sample = payload.get("sample") or [True] * (len(cu) - 1)
-        logits_idx = torch.tensor(
-            [c - 1 for c, smp in zip(cu[1:], sample) if smp],
-            dtype=torch.long,
-            device=dev,
-        )
+        if payload.get("logits_rows") is not None:
+            # spec-decode verification asks for explicit rows (every draft
+            # position, not just chunk ends)
+            logits_idx = torch.tensor(
+                payload["logits_rows"], dtype=torch.long, device=dev
+            )
+        else:
+            sample = payload.get("sample") or [True] * (len(cu) - 1)
+            logits_idx = torch.tensor(
+                [c - 1 for c, smp in zip(cu[1:], sample) if smp],
+                dtype=torch.long,
+                device=dev,
+            )
         with torch.no_grad():
             return self._forward_and_logits(ids, meta, logits_idx)
 
@@ -307,6 +314,77 @@ class ModelRunner:
         )
         payload["decode"] = d
         return payload
+
+    def build_spec_payload(
+        self,
+        prefill_seqs: List[Sequence],
+        chunks: List[int],
+        decode_seqs: List[Sequence],
+        drafts: List[List[int]],
+        bm: BlockManager,
+    ):
+        """One prefill-kind payload for a speculative step: real prefill
+        chunks first, then each decode sequence as a (1 + k_i)-token chunk
+        [last_sampled_token, draft_0 .. draft_{k_i-1}] verified through the
+        paged context-attention path. `logits_rows` lists the absolute
+        token rows the forward must score: the completing prefill chunks'
+        last rows, then EVERY row of each decode chunk (row j predicts the
+        token after input j — the acceptance test needs all of them).
+        Layout contract with LLMEngine._finish_spec_step: logits come back
+        as [prefill-sampled rows | seq0's 1+k_0 rows | seq1's ... ]."""
+        input_ids: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+        cu = [0]
+        new_lens: List[int] = []
+        total_lens: List[int] = []
+        sample: List[bool] = []
+        logits_rows: List[int] = []
+        bs = bm.block_size
+        all_seqs = list(prefill_seqs) + list(decode_seqs)
+
+        def _extend(seq, toks, C):
+            chunk = len(toks)
+            input_ids.extend(toks)
+            positions.extend(range(C, C + chunk))
+            pos = np.arange(C, C + chunk)
+            blocks = np.asarray(seq.block_ids, dtype=np.int64)
+            slots.extend((blocks[pos // bs] * bs + pos % bs).tolist())
+            cu.append(cu[-1] + chunk)
+            new_lens.append(chunk)
+            total_lens.append(C + chunk)
+
+        for seq, chunk in zip(prefill_seqs, chunks):
+            C = seq.num_computed_tokens or seq.num_cached_tokens
+            _extend(seq, seq.all_token_ids[C: C + chunk], C)
+            done = C + chunk >= seq.num_prompt_tokens
+            sample.append(done)
+            if done:
+                logits_rows.append(cu[-1] - 1)
+        for seq, draft in zip(decode_seqs, drafts):
+            row0 = cu[-1]
+            _extend(seq, [seq.all_token_ids[-1]] + list(draft),
+                    seq.num_tokens - 1)
+            sample.append(True)
+            logits_rows.extend(range(row0, cu[-1]))
+        max_blocks = max(len(s.block_ids) for s in all_seqs)
+        bt = [
+            s.block_ids + [0] * (max_blocks - len(s.block_ids))
+            for s in all_seqs
+        ]
+        return {
+            "kind": "prefill",
+            "ids": input_ids,
+            "positions": positions,
+            "slots": slots,
+            "cu": cu,
+            "new_lens": new_lens,
+            "total_lens": total_lens,
+            "bt": bt,
+            "sample": sample,
+            "logits_rows": logits_rows,
+            "lora_names": [s.lora_name for s in all_seqs],
+        }
 
     def run_batch(self, payload) -> torch.Tensor:
         """Mixed forward. Pure-decode payloads take the hipGraph path."""

@@ -45,6 +45,10 @@ def parse_args(argv=None):
     p.add_argument("--lora-modules", nargs="*", default=[],
                    help="adapters to register at startup: name[=rank]")
     p.add_argument("--enable-prefix-caching", action="store_true")
+    p.add_argument("--speculative-config", type=str, default=None,
+                   help='JSON: {"method": "ngram", '
+                        '"num_speculative_tokens": 4, '
+                        '"prompt_lookup_max": 4, "prompt_lookup_min": 2}')
     # multi-node rendezvous flags injected by the LWS wrapper
     p.add_argument("--nnodes", type=int, default=1)
     p.add_argument("--node-rank", type=int, default=0)
@@ -78,6 +82,17 @@ def build_engine_config(args):
     mc = get_model_config(args.model)
     mc.model_path = args.model_path
     mc.quantization = args.quantization
+    speculative = None
+    if args.speculative_config:
+        from fusioninfer_amd.engine.spec_decode import SpeculativeConfig
+
+        raw = json.loads(args.speculative_config)
+        speculative = SpeculativeConfig(
+            method=raw.get("method", "ngram"),
+            num_speculative_tokens=int(raw.get("num_speculative_tokens", 4)),
+            prompt_lookup_max=int(raw.get("prompt_lookup_max", 4)),
+            prompt_lookup_min=int(raw.get("prompt_lookup_min", 2)),
+        )
     return EngineConfig(
         model=mc,
         cache=CacheConfig(
@@ -97,6 +112,7 @@ def build_engine_config(args):
             data_parallel_size=args.data_parallel_size,
         ),
         kv_transfer=kvt,
+        speculative=speculative,
         enforce_eager=args.enforce_eager,
     )
 

@@ -89,6 +89,8 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             ("engine_step_seconds_count", "counter"),
             ("num_preemptions_total", "counter"),
             ("num_swap_outs_total", "counter"),
+            ("spec_decode_num_draft_tokens_total", "counter"),
+            ("spec_decode_num_accepted_tokens_total", "counter"),
         ]:
             lines.append(f"# TYPE vllm:{name} {mtype}")
             lines.append(

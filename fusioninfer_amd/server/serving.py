@@ -151,6 +151,12 @@ class ServingEngine:
             "engine_step_seconds_count": float(e.num_steps),
             "num_preemptions_total": float(e.num_preemptions),
             "num_swap_outs_total": float(e.num_swap_outs),
+            "spec_decode_num_draft_tokens_total": float(
+                e.num_spec_draft_tokens
+            ),
+            "spec_decode_num_accepted_tokens_total": float(
+                e.num_spec_accepted_tokens
+            ),
         }
 
     def abort(self, request_id: str) -> bool:
@@ -187,8 +193,11 @@ class ServingEngine:
                 q = self._streams.get(out.request_id)
                 if q is None:
                     continue
-                tok = out.output_token_ids[-1] if out.output_token_ids else None
-                q.put((tok, out.finished))
+                # speculative steps emit up to k+1 tokens at once; the
+                # finished flag rides on the last one
+                toks = out.new_token_ids or [None]
+                for i, tok in enumerate(toks):
+                    q.put((tok, out.finished and i == len(toks) - 1))
                 if out.finished:
                     with self._lock:
                         self._streams.pop(out.request_id, None)
