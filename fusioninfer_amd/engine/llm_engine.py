@@ -377,27 +377,53 @@ class LLMEngine:
         return outputs
 
     # ------------------------------------------------- sampling / emission
+    def _finish_seq(self, seq: Sequence) -> None:
+        seq.finish_time = time.monotonic()
+        self.scheduler.finish(seq)
+        self.num_finished += 1
+        if seq.ttft is not None:
+            self.ttft_sum += seq.ttft
+        self.e2e_latency_sum += seq.finish_time - seq.arrival_time
+        del self.seqs[seq.seq_id]
+
     def _emit_tokens(self, seq: Sequence, toks) -> RequestOutput:
         """Append this step's tokens (one for regular decode, up to k+1
         for an accepted speculative draft) with per-token stop checks."""
+        guided = seq.sampling.guided
         new = []
         for tok in toks:
             seq.append_token(int(tok))
             new.append(int(tok))
             self.num_generated_tokens += 1
-            if seq.check_stop():
-                seq.finish_time = time.monotonic()
-                self.scheduler.finish(seq)
-                self.num_finished += 1
-                if seq.ttft is not None:
-                    self.ttft_sum += seq.ttft
-                self.e2e_latency_sum += seq.finish_time - seq.arrival_time
-                del self.seqs[seq.seq_id]
+            if guided is not None:
+                guided.advance_token(int(tok))
+            if seq.check_stop() or (guided is not None
+                                    and guided.is_terminal()):
+                self._finish_seq(seq)
                 break
         return RequestOutput(seq, new_token_ids=new)
 
     def _sample_and_emit(self, sample_seqs, logits_f) -> List[RequestOutput]:
-        """Regular sampling path: logits_f row i belongs to sample_seqs[i]."""
+        """Regular sampling path: logits_f row i belongs to sample_seqs[i].
+        Guided sequences get their grammar mask applied first; a sequence
+        whose grammar admits no token is finished without emitting."""
+        outputs: List[RequestOutput] = []
+        keep = []
+        for i, s in enumerate(sample_seqs):
+            g = s.sampling.guided
+            if g is not None:
+                allowed = g.allowed_mask(logits_f.device)
+                if allowed is None:
+                    self._finish_seq(s)
+                    outputs.append(RequestOutput(s, new_token_ids=[]))
+                    continue
+                logits_f[i].masked_fill_(~allowed, float("-inf"))
+            keep.append(i)
+        if not keep:
+            return outputs
+        if len(keep) < len(sample_seqs):
+            sample_seqs = [sample_seqs[i] for i in keep]
+            logits_f = logits_f[keep]
         next_tokens = self.sampler.sample(logits_f, sample_seqs).tolist()
         # optional logprobs for requests that asked
         lp_rows = [
@@ -418,10 +444,11 @@ class LLMEngine:
                 s.logprobs.append(
                     (float(lp[j, next_tokens[i]]), entry_top)
                 )
-        return [
+        outputs.extend(
             self._emit_tokens(seq, [tok])
             for seq, tok in zip(sample_seqs, next_tokens)
-        ]
+        )
+        return outputs
 
     # ------------------------------------------------ speculative decoding
     def _propose_drafts(self, decode_seqs):

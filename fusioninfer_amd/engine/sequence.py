@@ -29,6 +29,7 @@ class SamplingParams:
         logprobs: Optional[int] = None,
         ignore_eos: bool = True,
         stop_token_ids: Optional[List[int]] = None,
+        guided=None,
     ):
         self.max_tokens = max_tokens
         self.temperature = temperature
@@ -41,6 +42,9 @@ class SamplingParams:
         self.logprobs = logprobs
         self.ignore_eos = ignore_eos
         self.stop_token_ids = stop_token_ids or []
+        # guided decoding: a guided.GuidedDecoder enforcing a grammar via
+        # logits masks (None = unconstrained)
+        self.guided = guided
 
 
 class Sequence:

@@ -1,0 +1,654 @@
+"""Guided decoding (structured outputs): grammar-constrained sampling.
+
+Capability parity with the reference engines' structured-output surface
+(vLLM guided decoding: OpenAI `response_format={"type": "json_object"}`
+and the `guided_json` / `guided_regex` / `guided_choice` request
+extensions). There is no network access for grammar libraries (outlines /
+xgrammar), so the machinery is self-contained:
+
+ * a lazy-DFA regex engine built on Brzozowski derivatives — each state
+   IS a (hash-consed) regex AST; stepping a character takes the
+   derivative. Supported syntax: literals, `.`, `[...]` classes (ranges,
+   negation), escapes (\\d \\w \\s \\D \\W \\S and literal escapes),
+   groups `(...)` / `(?:...)`, alternation `|`, quantifiers
+   `* + ? {m} {m,} {m,n}`. Anchored fullmatch semantics.
+ * a hand-written JSON pushdown grammar for `json_object` mode. Mask
+   caching stays finite at any nesting depth because the allowed-char
+   set depends only on (mode, extra, stack top, stack empty?).
+ * `guided_choice` compiles to an alternation of escaped literals;
+   `guided_json` (JSON-schema subset) compiles to a regex.
+
+Per-step cost: the token mask for a grammar state is computed once by
+walking the vocabulary trie with the grammar (dead prefixes prune whole
+subtrees) and cached per state signature — generation revisits a small
+set of DFA states, so steady-state masking is a dict hit + one
+masked_fill per guided sequence.
+"""
+
+from __future__ import annotations
+
+import json
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+# =============================================================== regex AST
+# Nodes are tuples (hashable, so derivative states can be memoized):
+#   ('empty',)            matches nothing (dead)
+#   ('eps',)              matches the empty string
+#   ('cls', frozenset, neg)  one char in (or not in) the set
+#   ('cat', r, s) ('alt', r, s) ('star', r)
+
+EMPTY = ("empty",)
+EPS = ("eps",)
+
+_D = frozenset("0123456789")
+_W = frozenset(
+    "abcdefghijklmnopqrstuvwxyzABCDEFGHIJKLMNOPQRSTUVWXYZ0123456789_"
+)
+_S = frozenset(" \t\n\r\f\v")
+
+
+def _cls(chars, neg=False):
+    return ("cls", frozenset(chars), neg)
+
+
+def _cat(r, s):
+    if r == EMPTY or s == EMPTY:
+        return EMPTY
+    if r == EPS:
+        return s
+    if s == EPS:
+        return r
+    return ("cat", r, s)
+
+
+def _alt(r, s):
+    if r == EMPTY:
+        return s
+    if s == EMPTY:
+        return r
+    if r == s:
+        return r
+    return ("alt", r, s)
+
+
+def _star(r):
+    if r in (EMPTY, EPS):
+        return EPS
+    if r[0] == "star":
+        return r
+    return ("star", r)
+
+
+def _nullable(r) -> bool:
+    tag = r[0]
+    if tag == "eps":
+        return True
+    if tag in ("empty", "cls"):
+        return False
+    if tag == "cat":
+        return _nullable(r[1]) and _nullable(r[2])
+    if tag == "alt":
+        return _nullable(r[1]) or _nullable(r[2])
+    return True  # star
+
+
+def _deriv(r, c, memo):
+    key = (r, c)
+    hit = memo.get(key)
+    if hit is not None:
+        return hit
+    tag = r[0]
+    if tag in ("empty", "eps"):
+        out = EMPTY
+    elif tag == "cls":
+        inside = c in r[1]
+        out = EPS if (inside != r[2]) else EMPTY
+    elif tag == "cat":
+        out = _cat(_deriv(r[1], c, memo), r[2])
+        if _nullable(r[1]):
+            out = _alt(out, _deriv(r[2], c, memo))
+    elif tag == "alt":
+        out = _alt(_deriv(r[1], c, memo), _deriv(r[2], c, memo))
+    else:  # star
+        out = _cat(_deriv(r[1], c, memo), r)
+    memo[key] = out
+    return out
+
+
+# ---------------------------------------------------------- regex parser
+class RegexError(ValueError):
+    pass
+
+
+_ESCAPES = {
+    "d": _cls(_D), "D": _cls(_D, True),
+    "w": _cls(_W), "W": _cls(_W, True),
+    "s": _cls(_S), "S": _cls(_S, True),
+    "n": _cls("\n"), "t": _cls("\t"), "r": _cls("\r"),
+}
+
+
+def parse_regex(pattern: str):
+    """Recursive-descent parse into the AST above (fullmatch semantics)."""
+    pos = [0]
+    n = len(pattern)
+
+    def peek():
+        return pattern[pos[0]] if pos[0] < n else None
+
+    def take():
+        c = pattern[pos[0]]
+        pos[0] += 1
+        return c
+
+    def parse_alt():
+        r = parse_cat()
+        while peek() == "|":
+            take()
+            r = _alt(r, parse_cat())
+        return r
+
+    def parse_cat():
+        r = EPS
+        while peek() is not None and peek() not in "|)":
+            r = _cat(r, parse_quant())
+        return r
+
+    def parse_quant():
+        r = parse_atom()
+        while True:
+            c = peek()
+            if c == "*":
+                take()
+                r = _star(r)
+            elif c == "+":
+                take()
+                r = _cat(r, _star(r))
+            elif c == "?":
+                take()
+                r = _alt(r, EPS)
+            elif c == "{":
+                save = pos[0]
+                take()
+                spec = ""
+                while peek() is not None and peek() != "}":
+                    spec += take()
+                if peek() != "}":
+                    pos[0] = save
+                    break
+                take()
+                r = _repeat(r, spec, pattern)
+            else:
+                break
+        return r
+
+    def _repeat(r, spec, pattern):
+        parts = spec.split(",")
+        try:
+            if len(parts) == 1:
+                m = x = int(parts[0])
+            elif parts[1] == "":
+                m, x = int(parts[0]), None
+            else:
+                m, x = int(parts[0]), int(parts[1])
+        except ValueError:
+            raise RegexError(f"bad repeat {{{spec}}} in {pattern!r}")
+        if x is not None and (x < m or x > 256) or m > 256:
+            raise RegexError(f"repeat bound too large in {pattern!r}")
+        out = EPS
+        for _ in range(m):
+            out = _cat(out, r)
+        if x is None:
+            out = _cat(out, _star(r))
+        else:
+            opt = _alt(r, EPS)
+            for _ in range(x - m):
+                out = _cat(out, opt)
+        return out
+
+    def parse_atom():
+        c = peek()
+        if c is None:
+            return EPS
+        if c == "(":
+            take()
+            if peek() == "?":
+                take()
+                if peek() != ":":
+                    raise RegexError("only (?:...) groups are supported")
+                take()
+            r = parse_alt()
+            if peek() != ")":
+                raise RegexError(f"unbalanced '(' in {pattern!r}")
+            take()
+            return r
+        if c == "[":
+            take()
+            return parse_class()
+        if c == ".":
+            take()
+            return _cls("\n", True)
+        if c == "\\":
+            take()
+            e = take()
+            return _ESCAPES.get(e, _cls(e))
+        if c in "*+?{":
+            raise RegexError(f"dangling quantifier in {pattern!r}")
+        return _cls(take())
+
+    def parse_class():
+        neg = False
+        if peek() == "^":
+            take()
+            neg = True
+        chars = set()
+        first = True
+        while True:
+            c = peek()
+            if c is None:
+                raise RegexError(f"unterminated [ in {pattern!r}")
+            if c == "]" and not first:
+                take()
+                break
+            first = False
+            c = take()
+            if c == "\\":
+                e = take()
+                esc = _ESCAPES.get(e)
+                if esc is not None and not esc[2]:
+                    chars |= esc[1]
+                    continue
+                c = {"n": "\n", "t": "\t", "r": "\r"}.get(e, e)
+            if peek() == "-" and pos[0] + 1 < n and pattern[pos[0] + 1] != "]":
+                take()
+                hi = take()
+                if hi == "\\":
+                    hi = take()
+                if ord(hi) < ord(c):
+                    raise RegexError(f"bad range {c}-{hi} in {pattern!r}")
+                chars |= {chr(x) for x in range(ord(c), ord(hi) + 1)}
+            else:
+                chars.add(c)
+        return _cls(chars, neg)
+
+    r = parse_alt()
+    if pos[0] != n:
+        raise RegexError(f"trailing {pattern[pos[0]:]!r} in {pattern!r}")
+    return r
+
+
+# ============================================================== grammars
+class RegexGrammar:
+    """Lazy DFA over Brzozowski derivatives; a state is a regex AST."""
+
+    def __init__(self, pattern: str):
+        self.pattern = pattern
+        self.root = parse_regex(pattern)
+        self._memo: Dict[Tuple, object] = {}
+
+    def initial(self):
+        return self.root
+
+    def step(self, state, ch):
+        out = _deriv(state, ch, self._memo)
+        return None if out == EMPTY else out
+
+    def is_complete(self, state) -> bool:
+        return _nullable(state)
+
+    def can_extend(self, state) -> bool:
+        return state != EPS
+
+    def signature(self, state):
+        return state
+
+
+class JsonGrammar:
+    """Pushdown grammar for well-formed JSON. State = (mode, extra, stack);
+    allowed characters depend only on (mode, extra, stack top, empty?), so
+    mask caching is finite at any depth. Compact form (no optional
+    whitespace at the top level, so termination is decidable; whitespace
+    IS allowed at structural positions inside containers).
+
+    root_object=True (OpenAI json_object mode) requires the value to be an
+    object; False accepts any JSON value.
+    """
+
+    WS = " \t\n\r"
+
+    def __init__(self, root_object: bool = True):
+        self.root_object = root_object
+
+    def initial(self):
+        return ("val", "root" if self.root_object else None, ())
+
+    # ------------------------------------------------------------- helpers
+    def _post(self, stack):
+        return ("post", None, stack)
+
+    def _post_step(self, stack, ch):
+        """Transitions out of 'a value just ended' — also used by number
+        states when a delimiter arrives."""
+        if not stack:
+            return None  # root value done: nothing may follow
+        if ch in self.WS:
+            return self._post(stack)
+        top = stack[-1]
+        if top == "{":
+            if ch == ",":
+                return ("key", None, stack)
+            if ch == "}":
+                return self._post(stack[:-1])
+        else:  # '['
+            if ch == ",":
+                return ("val", None, stack)
+            if ch == "]":
+                return self._post(stack[:-1])
+        return None
+
+    # ---------------------------------------------------------------- step
+    def step(self, state, ch):
+        mode, extra, stack = state
+        if mode == "val":
+            if ch in self.WS:
+                return state
+            if extra == "close" and ch == "]":
+                return self._post(stack[:-1])  # empty array
+            if extra == "root" and ch != "{":
+                return None  # json_object mode: root must be an object
+            if ch == "{":
+                return ("key", "first", stack + ("{",))
+            if ch == "[":
+                return ("val", "close", stack + ("[",))
+            if ch == '"':
+                return ("str", "val", stack)
+            if ch == "-":
+                return ("num", "sign", stack)
+            if ch == "0":
+                return ("num", "int0", stack)
+            if ch in "123456789":
+                return ("num", "int", stack)
+            if ch == "t":
+                return ("lit", "rue", stack)
+            if ch == "f":
+                return ("lit", "alse", stack)
+            if ch == "n":
+                return ("lit", "ull", stack)
+            return None
+        if mode == "key":
+            if ch in self.WS:
+                return (mode, extra, stack)
+            if ch == '"':
+                return ("str", "key", stack)
+            if extra == "first" and ch == "}":
+                return self._post(stack[:-1])  # empty object
+            return None
+        if mode == "colon":
+            if ch in self.WS:
+                return state
+            if ch == ":":
+                return ("val", None, stack)
+            return None
+        if mode == "str":
+            if ch == '"':
+                if extra == "key":
+                    return ("colon", None, stack)
+                return self._post(stack)
+            if ch == "\\":
+                return ("esc", extra, stack)
+            if ch in "\n\r":
+                return None
+            return state
+        if mode == "esc":
+            if ch == "u":
+                return ("hex", (extra, 4), stack)
+            if ch in '"\\/bfnrt':
+                return ("str", extra, stack)
+            return None
+        if mode == "hex":
+            ctx, left = extra
+            if ch in "0123456789abcdefABCDEF":
+                return ("str", ctx, stack) if left == 1 else \
+                    ("hex", (ctx, left - 1), stack)
+            return None
+        if mode == "lit":
+            if extra and ch == extra[0]:
+                rest = extra[1:]
+                return ("lit", rest, stack) if rest else self._post(stack)
+            return None
+        if mode == "num":
+            sub = extra
+            if sub == "sign":
+                if ch == "0":
+                    return ("num", "int0", stack)
+                if ch in "123456789":
+                    return ("num", "int", stack)
+                return None
+            accepting = sub in ("int", "int0", "frac", "exp")
+            if sub in ("int", "exp") and ch in _D:
+                return state
+            # strict JSON: no digits after a leading 0 — int0 digits fall
+            # through to _post_step below, which rejects them
+            if sub in ("int", "int0") and ch == ".":
+                return ("num", "frac0", stack)
+            if sub in ("int", "int0", "frac") and ch in "eE":
+                return ("num", "e0", stack)
+            if sub in ("frac0", "frac") and ch in _D:
+                return ("num", "frac", stack)
+            if sub == "e0":
+                if ch in "+-":
+                    return ("num", "esign", stack)
+                if ch in _D:
+                    return ("num", "exp", stack)
+                return None
+            if sub == "esign" and ch in _D:
+                return ("num", "exp", stack)
+            if accepting:
+                return self._post_step(stack, ch)
+            return None
+        # mode == "post"
+        return self._post_step(stack, ch)
+
+    def is_complete(self, state) -> bool:
+        mode, extra, stack = state
+        if stack:
+            return False
+        if mode == "post":
+            return True
+        return mode == "num" and extra in ("int", "int0", "frac", "exp")
+
+    def can_extend(self, state) -> bool:
+        mode, extra, stack = state
+        if mode == "post" and not stack:
+            return False
+        return True
+
+    def signature(self, state):
+        mode, extra, stack = state
+        return (mode, extra, stack[-1] if stack else None, bool(stack))
+
+
+# ==================================================== schema -> regex
+_RE_SPECIAL = set(".^$*+?{}[]\\|()")
+
+
+def _re_escape(s: str) -> str:
+    return "".join("\\" + c if c in _RE_SPECIAL else c for c in s)
+
+
+_STRING_RE = '"([^"\\\\\n\r]|\\\\.)*"'
+_INT_RE = "-?(0|[1-9][0-9]*)"
+_NUM_RE = "-?(0|[1-9][0-9]*)(\\.[0-9]+)?([eE][+-]?[0-9]+)?"
+
+
+def schema_to_regex(schema: dict, depth: int = 0) -> str:
+    """JSON-schema subset -> anchored regex (compact output, property
+    order fixed, all listed properties emitted). Unsupported constructs
+    raise ValueError — the API layer then falls back to generic JSON."""
+    if depth > 8:
+        raise ValueError("schema nesting too deep")
+    if not isinstance(schema, dict):
+        raise ValueError("schema must be an object")
+    if "enum" in schema:
+        opts = "|".join(
+            _re_escape(json.dumps(v, separators=(",", ":")))
+            for v in schema["enum"]
+        )
+        return f"({opts})"
+    t = schema.get("type")
+    if t == "string":
+        return _STRING_RE
+    if t == "integer":
+        return _INT_RE
+    if t == "number":
+        return _NUM_RE
+    if t == "boolean":
+        return "(true|false)"
+    if t == "null":
+        return "null"
+    if t == "array":
+        item = schema_to_regex(schema.get("items", {"type": "string"}),
+                               depth + 1)
+        return f"\\[({item}(,{item})*)?\\]"
+    if t == "object":
+        props = schema.get("properties") or {}
+        if not props:
+            raise ValueError("object schema needs properties")
+        parts = []
+        for name, sub in props.items():
+            key = _re_escape(json.dumps(name))
+            parts.append(f"{key}:{schema_to_regex(sub, depth + 1)}")
+        return "\\{" + ",".join(parts) + "\\}"
+    raise ValueError(f"unsupported schema type {t!r}")
+
+
+# ============================================================ vocabulary
+class Vocabulary:
+    """Token id -> decoded string, plus a trie over unique strings so a
+    grammar state's token mask is computed by pruned DFS instead of a
+    per-token scan."""
+
+    def __init__(self, vocab_size: int, decode_one):
+        self.vocab_size = vocab_size
+        by_string: Dict[str, List[int]] = {}
+        self.strings: List[str] = []
+        for t in range(vocab_size):
+            s = decode_one(t)
+            self.strings.append(s)
+            if s:  # tokens decoding to "" can never be grammar-checked
+                by_string.setdefault(s, []).append(t)
+        # trie node: (children: {ch: node}, ids ending here)
+        self.trie = ({}, [])
+        for s, ids in by_string.items():
+            node = self.trie
+            for ch in s:
+                node = node[0].setdefault(ch, ({}, []))
+            node[1].extend(ids)
+
+
+class GuidedMaskCache:
+    """Per-(grammar, vocab, device) cache of token masks by state sig."""
+
+    def __init__(self, grammar, vocab: Vocabulary):
+        self.grammar = grammar
+        self.vocab = vocab
+        self._masks: Dict[Tuple, torch.Tensor] = {}
+
+    def mask(self, state, device) -> torch.Tensor:
+        key = (self.grammar.signature(state), str(device))
+        m = self._masks.get(key)
+        if m is not None:
+            return m
+        allowed: List[int] = []
+        g = self.grammar
+
+        def dfs(node, st):
+            children, ids = node
+            if ids:
+                allowed.extend(ids)
+            for ch, child in children.items():
+                nxt = g.step(st, ch)
+                if nxt is not None:
+                    dfs(child, nxt)
+
+        # root children only: step once per distinct first char
+        for ch, child in self.vocab.trie[0].items():
+            nxt = g.step(state, ch)
+            if nxt is not None:
+                dfs(child, nxt)
+        m = torch.zeros(self.vocab.vocab_size, dtype=torch.bool)
+        if allowed:
+            m[torch.tensor(allowed, dtype=torch.long)] = True
+        m = m.to(device)
+        self._masks[key] = m
+        return m
+
+
+class GuidedDecoder:
+    """Per-request grammar cursor the engine consults at sampling time.
+
+    allowed_mask() -> bool[V] (None = no token fits: the engine finishes
+    the request — cleanly when the grammar is complete).
+    advance_token() steps the grammar over the emitted token's text;
+    is_terminal() tells the engine to stop (grammar complete and not
+    extendable)."""
+
+    def __init__(self, cache: GuidedMaskCache):
+        self.cache = cache
+        self.state = cache.grammar.initial()
+        self.dead = False
+
+    def allowed_mask(self, device) -> Optional[torch.Tensor]:
+        if self.dead:
+            return None
+        m = self.cache.mask(self.state, device)
+        return m if bool(m.any()) else None
+
+    def advance_token(self, token_id: int) -> None:
+        g = self.cache.grammar
+        s = self.state
+        for ch in self.cache.vocab.strings[token_id]:
+            s = g.step(s, ch)
+            if s is None:
+                self.dead = True
+                return
+        self.state = s
+
+    def is_terminal(self) -> bool:
+        if self.dead:
+            return True
+        g = self.cache.grammar
+        return g.is_complete(self.state) and not g.can_extend(self.state)
+
+
+# ============================================================== factory
+_CACHES: Dict[Tuple, GuidedMaskCache] = {}
+
+
+def build_guided(kind: str, spec, vocab: Vocabulary) -> GuidedDecoder:
+    """kind: 'json_object' | 'regex' | 'choice' | 'json_schema'.
+    Raises ValueError on unsupported specs (API layer maps it to 400)."""
+    if kind == "json_object":
+        key = ("json_object", id(vocab))
+        if key not in _CACHES:
+            _CACHES[key] = GuidedMaskCache(JsonGrammar(root_object=True),
+                                           vocab)
+        return GuidedDecoder(_CACHES[key])
+    if kind == "regex":
+        key = ("regex", spec, id(vocab))
+        if key not in _CACHES:
+            _CACHES[key] = GuidedMaskCache(RegexGrammar(spec), vocab)
+        return GuidedDecoder(_CACHES[key])
+    if kind == "choice":
+        pattern = "(" + "|".join(_re_escape(str(c)) for c in spec) + ")"
+        return build_guided("regex", pattern, vocab)
+    if kind == "json_schema":
+        try:
+            pattern = schema_to_regex(spec)
+        except ValueError:
+            # unsupported schema construct: enforce well-formed JSON
+            return build_guided("json_object", vocab=vocab, spec=None)
+        return build_guided("regex", pattern, vocab)
+    raise ValueError(f"unknown guided decoding kind {kind!r}")

@@ -98,6 +98,42 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             )
         return PlainTextResponse("\n".join(lines) + "\n")
 
+    _vocab_cache: Dict[str, Any] = {}
+
+    def _get_vocab():
+        if "v" not in _vocab_cache:
+            from fusioninfer_amd.guided import Vocabulary
+
+            _vocab_cache["v"] = Vocabulary(
+                vocab, lambda t: decode_tokens([t])
+            )
+        return _vocab_cache["v"]
+
+    def _guided_from(body: Dict[str, Any]):
+        """Structured outputs: OpenAI response_format plus the vLLM
+        guided_json / guided_regex / guided_choice extensions. Raises
+        ValueError for unsupported grammars (mapped to HTTP 400)."""
+        from fusioninfer_amd.guided import build_guided
+
+        if body.get("guided_regex"):
+            return build_guided("regex", str(body["guided_regex"]),
+                                _get_vocab())
+        if body.get("guided_choice"):
+            return build_guided("choice", list(body["guided_choice"]),
+                                _get_vocab())
+        if body.get("guided_json"):
+            return build_guided("json_schema", body["guided_json"],
+                                _get_vocab())
+        rf = body.get("response_format")
+        if isinstance(rf, dict):
+            t = rf.get("type")
+            if t == "json_object":
+                return build_guided("json_object", None, _get_vocab())
+            if t == "json_schema":
+                schema = (rf.get("json_schema") or {}).get("schema")
+                return build_guided("json_schema", schema, _get_vocab())
+        return None
+
     def _sampling_from(body: Dict[str, Any]) -> SamplingParams:
         return SamplingParams(
             max_tokens=int(body.get("max_tokens", 16)),
@@ -111,6 +147,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             logprobs=body.get("logprobs"),
             ignore_eos=bool(body.get("ignore_eos", False)),
             stop_token_ids=body.get("stop_token_ids") or [],
+            guided=_guided_from(body),
         )
 
     def _parse_stops(body: Dict[str, Any]) -> List[str]:
@@ -171,7 +208,12 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         body = await request.json()
         prompt = body.get("prompt", "")
         prompt_ids = encode_prompt(prompt, vocab)
-        sampling = _sampling_from(body)
+        try:
+            sampling = _sampling_from(body)
+        except ValueError as e:
+            return JSONResponse({"error": {"message": str(e),
+                                           "type": "invalid_request_error"}},
+                                400)
         # OpenAI-style adapter selection: model == a registered LoRA name
         lora = body.get("model")
         if lora not in serving.engine.active_loras():
@@ -291,7 +333,12 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             f"{m.get('role', 'user')}: {m.get('content', '')}" for m in messages
         )
         prompt_ids = encode_prompt(text, vocab)
-        sampling = _sampling_from(body)
+        try:
+            sampling = _sampling_from(body)
+        except ValueError as e:
+            return JSONResponse({"error": {"message": str(e),
+                                           "type": "invalid_request_error"}},
+                                400)
         stops = _parse_stops(body)
         lora = body.get("model")
         if lora not in serving.engine.active_loras():

@@ -231,3 +231,73 @@ def test_stop_strings_streaming(app):
             assert n_chunks <= len(full)
 
     asyncio.run(run())
+
+
+def test_guided_choice_and_json_object(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "pick one:", "max_tokens": 16,
+                      "temperature": 0.0,
+                      "guided_choice": ["yes", "no", "maybe"]},
+            )
+            assert r.status_code == 200
+            assert r.json()["choices"][0]["text"] in {"yes", "no", "maybe"}
+
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "as json:", "max_tokens": 48,
+                      "temperature": 0.0,
+                      "guided_json": {
+                          "type": "object",
+                          "properties": {"ok": {"type": "boolean"}},
+                      }},
+            )
+            assert r.status_code == 200
+            obj = json.loads(r.json()["choices"][0]["text"])
+            assert isinstance(obj.get("ok"), bool)
+
+            # OpenAI response_format on chat: output must stay a viable
+            # JSON prefix even if max_tokens truncates it
+            r = await c.post(
+                "/v1/chat/completions",
+                json={"messages": [{"role": "user", "content": "json pls"}],
+                      "max_tokens": 12, "temperature": 0.0,
+                      "response_format": {"type": "json_object"}},
+            )
+            assert r.status_code == 200
+            from fusioninfer_amd.guided import JsonGrammar
+
+            text = r.json()["choices"][0]["message"]["content"]
+            g = JsonGrammar(root_object=True)
+            st = g.initial()
+            for ch in text:
+                st = g.step(st, ch)
+                assert st is not None, text
+
+    asyncio.run(run())
+
+
+def test_guided_regex_and_bad_pattern(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "id:", "max_tokens": 20, "temperature": 0.0,
+                      "guided_regex": "[a-f]{4}-[0-9]{2}"},
+            )
+            assert r.status_code == 200
+            import re as _re
+
+            assert _re.fullmatch(r"[a-f]{4}-[0-9]{2}",
+                                 r.json()["choices"][0]["text"])
+
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "x", "max_tokens": 4,
+                      "guided_regex": "(unclosed"},
+            )
+            assert r.status_code == 400
+
+    asyncio.run(run())
