@@ -20,6 +20,10 @@ def parse_args(argv=None):
     p.add_argument("--model", default="Qwen3-8B")
     p.add_argument("--model-path", default=None,
                    help="HF-layout safetensors dir; default random init")
+    p.add_argument("--tokenizer", default=None,
+                   help="path to a vendored tokenizer.json (or a dir "
+                        "containing one); default: the checkpoint dir's, "
+                        "else the reversible byte-level fallback")
     p.add_argument("--host", default="0.0.0.0")
     p.add_argument("--port", type=int, default=8000)
     p.add_argument("--tensor-parallel-size", type=int, default=1)
@@ -154,7 +158,12 @@ def _rank_main(local_rank: int, args, nproc: int):
 
     import uvicorn
 
-    app = build_app(serving, args.model)
+    from fusioninfer_amd.tokenizer import get_tokenizer
+
+    tokenizer = get_tokenizer(
+        cfg.model.vocab_size, args.model_path, args.tokenizer
+    )
+    app = build_app(serving, args.model, tokenizer=tokenizer)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
     serving.engine.stop_workers()
 

@@ -46,9 +46,19 @@ def decode_tokens(token_ids: List[int]) -> str:
         return " ".join(str(t) for t in token_ids)
 
 
-def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
+def build_app(serving: ServingEngine, model_name: str,
+              tokenizer=None) -> FastAPI:
     app = FastAPI(title="fusioninfer-amd")
     vocab = serving.engine.cfg.model.vocab_size
+    if tokenizer is None:
+        from fusioninfer_amd.tokenizer import ByteTokenizer
+
+        tokenizer = ByteTokenizer(vocab)
+
+    def _encode(prompt: Union[str, List[int]]) -> List[int]:
+        if isinstance(prompt, list):
+            return [int(t) for t in prompt]
+        return tokenizer.encode(prompt)
 
     @app.get("/health")
     async def health():
@@ -105,7 +115,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             from fusioninfer_amd.guided import Vocabulary
 
             _vocab_cache["v"] = Vocabulary(
-                vocab, lambda t: decode_tokens([t])
+                vocab, tokenizer.decode_one
             )
         return _vocab_cache["v"]
 
@@ -135,6 +145,10 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         return None
 
     def _sampling_from(body: Dict[str, Any]) -> SamplingParams:
+        stop_ids = list(body.get("stop_token_ids") or [])
+        if tokenizer.eos_token_id is not None \
+                and tokenizer.eos_token_id not in stop_ids:
+            stop_ids.append(tokenizer.eos_token_id)
         return SamplingParams(
             max_tokens=int(body.get("max_tokens", 16)),
             temperature=float(body.get("temperature", 1.0)),
@@ -146,7 +160,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             seed=body.get("seed"),
             logprobs=body.get("logprobs"),
             ignore_eos=bool(body.get("ignore_eos", False)),
-            stop_token_ids=body.get("stop_token_ids") or [],
+            stop_token_ids=stop_ids,
             guided=_guided_from(body),
         )
 
@@ -168,14 +182,14 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
             if tok is not None:
                 toks.append(tok)
                 if stops:
-                    text += decode_tokens([tok])
+                    text += tokenizer.decode_one(tok)
                     for s in stops:
                         i = text.find(s)
                         if i >= 0:
                             serving.abort(req_id)
                             return toks, text[:i]
             if finished:
-                return toks, text if stops else decode_tokens(toks)
+                return toks, text if stops else tokenizer.decode(toks)
 
     async def _stream(q) -> AsyncGenerator:
         loop = asyncio.get_event_loop()
@@ -195,7 +209,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
                 {"error": {"message": "server is not a PD prefiller",
                            "type": "invalid_request_error"}}, 400)
         body = await request.json()
-        prompt_ids = encode_prompt(body.get("prompt", ""), vocab)
+        prompt_ids = _encode(body.get("prompt", ""))
         loop = asyncio.get_event_loop()
         tag, first_token = await loop.run_in_executor(
             None, serving.prefill_via_pd, prompt_ids
@@ -207,7 +221,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
     async def completions(request: Request):
         body = await request.json()
         prompt = body.get("prompt", "")
-        prompt_ids = encode_prompt(prompt, vocab)
+        prompt_ids = _encode(prompt)
         try:
             sampling = _sampling_from(body)
         except ValueError as e:
@@ -246,7 +260,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
                 try:
                     async for tok, finished in _stream(q):
                         done = finished
-                        delta = decode_tokens([tok]) if tok is not None else ""
+                        delta = tokenizer.decode_one(tok) if tok is not None else ""
                         if stops and tok is not None:
                             # hold back text that could extend into a stop
                             # string; cut the stream at the first match
@@ -332,7 +346,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
         text = "\n".join(
             f"{m.get('role', 'user')}: {m.get('content', '')}" for m in messages
         )
-        prompt_ids = encode_prompt(text, vocab)
+        prompt_ids = _encode(text)
         try:
             sampling = _sampling_from(body)
         except ValueError as e:
@@ -360,7 +374,7 @@ def build_app(serving: ServingEngine, model_name: str) -> FastAPI:
                         delta["role"] = "assistant"
                         first = False
                     if tok is not None:
-                        delta["content"] = decode_tokens([tok])
+                        delta["content"] = tokenizer.decode_one(tok)
                     chunk = {
                         "id": cid,
                         "object": "chat.completion.chunk",

@@ -1,0 +1,97 @@
+"""Tokenization for the OpenAI server.
+
+The environment has no network access, so tokenizer files can only come
+vendored with a checkpoint directory (`tokenizer.json`, the HF fast-
+tokenizer format, loaded with the offline `tokenizers` wheel). Without
+one, string prompts fall back to a reversible byte-level scheme — ids
+offset by 3 so they stay in-vocab — which the benchmark and EPP flows
+use (they submit token-id prompts anyway).
+
+Capability parity: the reference's engine containers mount HF tokenizer
+files next to the weights (SURVEY.md §2.3 `--model Qwen/Qwen3-8B`); this
+module gives the same behavior for local checkpoint dirs.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional, Union
+
+
+class ByteTokenizer:
+    """Reversible byte-level fallback (no vocab files needed)."""
+
+    def __init__(self, vocab_size: int):
+        self.vocab_size = vocab_size
+        self.eos_token_id: Optional[int] = None
+
+    def encode(self, text: str) -> List[int]:
+        return [min(b + 3, self.vocab_size - 1) for b in text.encode("utf-8")]
+
+    def decode(self, token_ids: List[int]) -> str:
+        return bytes(max(t - 3, 0) & 0xFF for t in token_ids).decode(
+            "utf-8", errors="replace"
+        )
+
+    def decode_one(self, token_id: int) -> str:
+        return bytes([max(token_id - 3, 0) & 0xFF]).decode(
+            "utf-8", errors="replace"
+        )
+
+
+class HFTokenizer:
+    """tokenizer.json (HF fast format) via the `tokenizers` library."""
+
+    def __init__(self, path: str, vocab_size: int):
+        from tokenizers import Tokenizer
+
+        self._tok = Tokenizer.from_file(path)
+        self.vocab_size = vocab_size
+        # model vocab may be padded past the tokenizer's (Qwen3 pads
+        # 151669 -> 151936); ids past the tokenizer's range decode to ""
+        self._n = self._tok.get_vocab_size(with_added_tokens=True)
+        self.eos_token_id = None
+        for name in ("</s>", "<|endoftext|>", "<|im_end|>", "<eos>"):
+            tid = self._tok.token_to_id(name)
+            if tid is not None:
+                self.eos_token_id = tid
+                break
+
+    def encode(self, text: str) -> List[int]:
+        return self._tok.encode(text, add_special_tokens=False).ids
+
+    def decode(self, token_ids: List[int]) -> str:
+        ids = [t for t in token_ids if 0 <= t < self._n]
+        return self._tok.decode(ids, skip_special_tokens=False)
+
+    def decode_one(self, token_id: int) -> str:
+        # per-token text for streaming deltas and guided-decoding masks
+        if not 0 <= token_id < self._n:
+            return ""
+        return self._tok.decode([token_id], skip_special_tokens=False)
+
+
+Tok = Union[ByteTokenizer, HFTokenizer]
+
+
+def get_tokenizer(
+    vocab_size: int,
+    model_path: Optional[str] = None,
+    tokenizer_path: Optional[str] = None,
+) -> Tok:
+    """tokenizer.json from --tokenizer, else from the checkpoint dir,
+    else the byte fallback."""
+    candidates = []
+    if tokenizer_path:
+        candidates.append(tokenizer_path)
+        candidates.append(os.path.join(tokenizer_path, "tokenizer.json"))
+    if model_path:
+        candidates.append(os.path.join(model_path, "tokenizer.json"))
+    for c in candidates:
+        if os.path.isfile(c):
+            return HFTokenizer(c, vocab_size)
+    if tokenizer_path:
+        raise FileNotFoundError(
+            f"--tokenizer {tokenizer_path!r}: no tokenizer.json found"
+        )
+    return ByteTokenizer(vocab_size)

@@ -1,0 +1,115 @@
+"""Tokenizer: vendored tokenizer.json (HF fast format) and byte fallback.
+
+The environment has no network, so the test builds a tiny BPE
+tokenizer.json programmatically with the offline `tokenizers` wheel —
+the same file format a vendored Qwen3 checkpoint dir would carry.
+"""
+
+import asyncio
+import json
+
+import httpx
+import pytest
+
+from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+from fusioninfer_amd.models.registry import get_model_config
+from fusioninfer_amd.tokenizer import ByteTokenizer, HFTokenizer, get_tokenizer
+
+
+@pytest.fixture(scope="module")
+def tok_file(tmp_path_factory):
+    from tokenizers import Tokenizer, models, pre_tokenizers, decoders
+
+    # byte-level BPE with no merges: every byte is a token — small but
+    # structurally identical to a real vendored tokenizer.json
+    bl = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    vocab = {ch: i for i, ch in enumerate(
+        sorted(pre_tokenizers.ByteLevel.alphabet())
+    )}
+    vocab["ab"] = len(vocab)          # one merge so encoding != raw bytes
+    vocab["<|endoftext|>"] = len(vocab)
+    tok = Tokenizer(models.BPE(vocab=vocab, merges=[("a", "b")]))
+    tok.pre_tokenizer = bl
+    tok.decoder = decoders.ByteLevel()
+    path = tmp_path_factory.mktemp("tok") / "tokenizer.json"
+    tok.save(str(path))
+    return str(path)
+
+
+def test_byte_fallback_roundtrip():
+    t = ByteTokenizer(1024)
+    assert t.decode(t.encode("hello world")) == "hello world"
+    assert t.eos_token_id is None
+
+
+def test_hf_tokenizer_roundtrip_and_eos(tok_file):
+    t = HFTokenizer(tok_file, vocab_size=1024)
+    ids = t.encode("hello, world!")
+    assert t.decode(ids) == "hello, world!"
+    assert all(isinstance(i, int) for i in ids)
+    assert t.eos_token_id is not None
+    # model-padded ids decode to empty, not an exception
+    assert t.decode_one(1000) == ""
+    assert t.decode(ids + [1000]) == "hello, world!"
+
+
+def test_get_tokenizer_resolution(tok_file, tmp_path):
+    import shutil
+    # explicit --tokenizer file
+    t = get_tokenizer(1024, tokenizer_path=tok_file)
+    assert isinstance(t, HFTokenizer)
+    # checkpoint dir containing tokenizer.json
+    d = tmp_path / "ckpt"
+    d.mkdir()
+    shutil.copy(tok_file, d / "tokenizer.json")
+    assert isinstance(get_tokenizer(1024, model_path=str(d)), HFTokenizer)
+    # nothing vendored -> byte fallback
+    assert isinstance(get_tokenizer(1024, model_path=str(tmp_path)),
+                      ByteTokenizer)
+    with pytest.raises(FileNotFoundError):
+        get_tokenizer(1024, tokenizer_path=str(tmp_path / "nope"))
+
+
+def test_server_with_hf_tokenizer(tok_file):
+    """End-to-end: server encodes string prompts with the vendored
+    tokenizer and guided decoding masks over its multi-char vocab."""
+    from fusioninfer_amd.server.api_server import build_app
+    from fusioninfer_amd.server.serving import ServingEngine
+
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=128),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=512, max_model_len=256
+        ),
+    )
+    serving = ServingEngine(cfg, device="cpu")
+    tok = HFTokenizer(tok_file, vocab_size=cfg.model.vocab_size)
+    app = build_app(serving, "tiny-qwen3", tokenizer=tok)
+
+    async def run():
+        async with httpx.AsyncClient(
+            transport=httpx.ASGITransport(app=app), base_url="http://t"
+        ) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "abc", "max_tokens": 4, "temperature": 0.0},
+            )
+            assert r.status_code == 200
+            # prompt went through the vendored BPE: "abc" -> ["ab", "c"]
+            # (the byte fallback would count 3)
+            assert r.json()["usage"]["prompt_tokens"] == 2
+
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "pick:", "max_tokens": 16,
+                      "temperature": 0.0,
+                      "guided_choice": ["yes", "no"]},
+            )
+            assert r.status_code == 200
+            assert r.json()["choices"][0]["text"] in {"yes", "no"}
+
+    try:
+        asyncio.run(run())
+    finally:
+        serving.shutdown()
