@@ -153,6 +153,56 @@ def pp_recv_from(shape, dtype, device, src: int) -> torch.Tensor:
     return t
 
 
+# ------------------------------------------- async p2p (microbatch overlap)
+class _PendingSend:
+    """isend handle; keeps the staging buffer alive until wait()."""
+
+    def __init__(self, work, buf):
+        self._work = work
+        self._buf = buf
+
+    def wait(self):
+        self._work.wait()
+
+
+class _PendingRecv:
+    """irecv handle; wait() returns the received tensor."""
+
+    def __init__(self, work, buf, bf16_view: bool):
+        self._work = work
+        self._buf = buf
+        self._bf16 = bf16_view
+
+    def wait(self) -> torch.Tensor:
+        self._work.wait()
+        return self._buf.view(torch.bfloat16) if self._bf16 else self._buf
+
+
+def _gloo_safe_isend(t: torch.Tensor, dst: int) -> _PendingSend:
+    if t.device.type == "cpu" and t.dtype == torch.bfloat16:
+        buf = t.contiguous().view(torch.int16)
+    else:
+        buf = t.contiguous()
+    return _PendingSend(dist.isend(buf, dst), buf)
+
+
+def pp_isend_next(t: torch.Tensor) -> _PendingSend:
+    return _gloo_safe_isend(t, _PP_RANK + 1)
+
+
+def pp_isend_to(t: torch.Tensor, dst: int) -> _PendingSend:
+    return _gloo_safe_isend(t, dst)
+
+
+def pp_irecv_from(shape, dtype, device, src: int) -> _PendingRecv:
+    """Post a receive NOW (pre-posting on the driver is what lets pipeline
+    microbatches drain without a send/recv deadlock) and collect later."""
+    bf16_cpu = dtype == torch.bfloat16 and torch.device(device).type == "cpu"
+    buf = torch.empty(shape, dtype=torch.int16 if bf16_cpu else dtype,
+                      device=device)
+    return _PendingRecv(dist.irecv(buf, src), buf, bf16_cpu)
+
+
 def tp_broadcast_object(obj=None):
     """Broadcast a picklable object from the execution group's first rank
     (the TP group, or the whole pipeline when PP > 1)."""

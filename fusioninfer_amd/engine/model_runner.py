@@ -197,6 +197,17 @@ class ModelRunner:
         )
 
     def run_prefill(self, payload) -> torch.Tensor:
+        import fusioninfer_amd.distributed.parallel_state as ps
+
+        if ps.pp_world_size() > 1:
+            subs = self._split_prefill_payload(payload)
+            if len(subs) > 1:
+                return self._run_prefill_pp_pipelined(subs)
+        ids, meta, logits_idx = self._prefill_inputs(payload)
+        with torch.no_grad():
+            return self._forward_and_logits(ids, meta, logits_idx)
+
+    def _prefill_inputs(self, payload):
         dev = self.device
         cu = payload["cu"]
         tile_seq, tile_row0 = ops_mod.build_prefill_tiles(
@@ -236,8 +247,101 @@ class ModelRunner:
                 dtype=torch.long,
                 device=dev,
             )
+        return ids, meta, logits_idx
+
+    # ------------------------------------------- PP microbatch pipelining
+    def _split_prefill_payload(self, payload, min_mb_tokens: int = 256):
+        """Split a prefill payload into per-stage microbatches (contiguous
+        sequence ranges, balanced by token count). Returns [payload] when
+        splitting is not worth it (small batches pay comm overhead without
+        filling the pipeline)."""
+        import fusioninfer_amd.distributed.parallel_state as ps
+
+        cu = payload["cu"]
+        n_seqs = len(cu) - 1
+        M = min(ps.pp_world_size(), n_seqs, cu[-1] // min_mb_tokens)
+        if M < 2:
+            return [payload]
+        target = cu[-1] / M
+        bounds = [0]
+        for i in range(1, M):
+            want = i * target
+            j = bounds[-1] + 1
+            while j < n_seqs - (M - 1 - i) and cu[j + 1] <= want:
+                j += 1
+            bounds.append(j)
+        bounds.append(n_seqs)
+        rows = payload.get("logits_rows")
+        subs = []
+        for a, b in zip(bounds, bounds[1:]):
+            lo, hi = cu[a], cu[b]
+            sp = {
+                "kind": "prefill",
+                "ids": payload["ids"][lo:hi],
+                "positions": payload["positions"][lo:hi],
+                "slots": payload["slots"][lo:hi],
+                "cu": [c - lo for c in cu[a: b + 1]],
+                "new_lens": payload["new_lens"][a:b],
+                "total_lens": payload["total_lens"][a:b],
+                "bt": payload["bt"][a:b],
+                "sample": payload["sample"][a:b],
+                "lora_names": (payload.get("lora_names") or [None] * n_seqs)[a:b],
+            }
+            if rows is not None:
+                sp["logits_rows"] = [r - lo for r in rows if lo <= r < hi]
+            subs.append(sp)
+        return subs
+
+    def _run_prefill_pp_pipelined(self, subs) -> torch.Tensor:
+        """Microbatch-overlapped pipeline prefill: every stage loops over
+        the microbatches in order, so stage s computes microbatch i while
+        stage s+1 computes i-1 (activation p2p over xGMI under RCCL; on
+        gloo this is a pure correctness path). The driver pre-posts every
+        logits irecv BEFORE its compute loop — sends stay isend — so the
+        pipeline drains without rendezvous deadlock."""
+        import fusioninfer_amd.distributed.parallel_state as ps
+
+        V = self.cfg.model.vocab_size
+        H = self.cfg.model.hidden_size
+        dt = torch.bfloat16
+        dev = self.device
+        last = ps.pp_world_size() - 1
+        inputs = [self._prefill_inputs(sp) for sp in subs]
+        sends = []
         with torch.no_grad():
-            return self._forward_and_logits(ids, meta, logits_idx)
+            if ps.pp_rank() == 0:
+                pending = [
+                    ps.pp_irecv_from((int(li.numel()), V), dt, dev, last)
+                    for _, _, li in inputs
+                    if li.numel() > 0
+                ]
+                for ids, meta, _ in inputs:
+                    hidden, residual = self.model(ids, meta, self.kv_caches)
+                    sends.append(ps.pp_isend_next(hidden))
+                    sends.append(ps.pp_isend_next(residual))
+                for s in sends:
+                    s.wait()
+                outs = [p.wait() for p in pending]
+                if not outs:
+                    return torch.empty((0, V), dtype=dt, device=dev)
+                return torch.cat(outs, dim=0)
+            for ids, meta, logits_idx in inputs:
+                T = int(meta.num_prefill_tokens)
+                hidden = ps.pp_recv_prev((T, H), dt, dev)
+                residual = ps.pp_recv_prev((T, H), dt, dev)
+                out = self.model(None, meta, self.kv_caches,
+                                 hidden=hidden, residual=residual)
+                if ps.pp_is_last():
+                    if logits_idx.numel() > 0:
+                        logits = self.model.compute_logits(out[logits_idx])
+                        sends.append(ps.pp_isend_to(logits, 0))
+                else:
+                    h2, r2 = out
+                    sends.append(ps.pp_isend_next(h2))
+                    sends.append(ps.pp_isend_next(r2))
+            for s in sends:
+                s.wait()
+        return torch.empty((0, V), dtype=dt, device=dev)
 
     def _forward_and_logits(self, ids, meta, logits_idx):
         """Run the model (all pipeline stages) and return logits on the

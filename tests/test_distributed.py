@@ -466,3 +466,89 @@ def test_tp2_spec_decode_matches_tp1():
             p.join(timeout=240)
             assert p.exitcode == 0, f"rank exited {p.exitcode}"
         assert results[0] == "ok" and results[1] == "ok"
+
+
+# ------------------------------------ PP microbatch-overlapped prefill
+def _pp_microbatch_worker(rank, port, results):
+    from fusioninfer_amd.config import CacheConfig, SchedulerConfig
+    from fusioninfer_amd.distributed import parallel_state as ps
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+
+    def big_cfg():
+        return EngineConfig(
+            model=get_model_config("tiny-qwen3"),
+            cache=CacheConfig(num_gpu_blocks=256),
+            scheduler=SchedulerConfig(
+                max_num_seqs=8, max_num_batched_tokens=2048,
+                max_model_len=512,
+            ),
+            seed=11,
+        )
+
+    # 4 prompts x ~150 tokens = one ~600-token admission step, which the
+    # splitter cuts into 2 microbatches (>= 256 tokens each)
+    prompts = [
+        [(7 * i + j) % 97 + 2 for j in range(150)] for i in range(4)
+    ]
+
+    ps.ensure_single_process()
+    torch.manual_seed(77)
+    eng1 = LLMEngine(big_cfg(), device="cpu")
+    ref = [
+        o.output_token_ids
+        for o in eng1.generate(prompts,
+                               SamplingParams(max_tokens=4, temperature=0.0))
+    ]
+    ps.destroy()
+
+    _init(rank, 2, port)
+    try:
+        torch.manual_seed(77)
+        cfg = big_cfg()
+        cfg.parallel.pipeline_parallel_size = 2
+        cfg.parallel.distributed_backend = "gloo"
+        engine = LLMEngine(cfg, device="cpu")
+        # the admission step really is split (assert on the driver side
+        # by probing the splitter with the same batch shape)
+        if engine.is_driver:
+            subs = engine.runner._split_prefill_payload({
+                "kind": "prefill",
+                "ids": [0] * 600, "positions": [0] * 600,
+                "slots": [0] * 600, "cu": [0, 150, 300, 450, 600],
+                "new_lens": [150] * 4, "total_lens": [150] * 4,
+                "bt": [[0]] * 4, "sample": [True] * 4,
+                "lora_names": [None] * 4,
+            })
+            assert len(subs) == 2
+            assert [len(s["ids"]) for s in subs] == [300, 300]
+            outs = engine.generate(
+                prompts, SamplingParams(max_tokens=4, temperature=0.0)
+            )
+            assert [o.output_token_ids for o in outs] == ref
+            engine.stop_workers()
+        else:
+            engine.worker_loop()
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+        ps.destroy()
+
+
+def test_pp2_microbatch_prefill_matches_single_process():
+    """Microbatch-overlapped PP prefill (pre-posted logits irecv, isend
+    activations): token-exact vs the single-process engine."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29637
+        procs = [
+            ctx.Process(target=_pp_microbatch_worker, args=(r, port, results))
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+            assert p.exitcode == 0, f"rank exited {p.exitcode}"
+        assert results[0] == "ok" and results[1] == "ok"
