@@ -39,11 +39,9 @@ class ModelRunner:
             from fusioninfer_amd.models.weight_loader import load_safetensors_dir
 
             load_safetensors_dir(self.model, mc.model_path)
-        if mc.quantization == "fp8" and mc.is_moe:
-            raise ValueError(
-                "fp8 quantization with MoE needs fp8 expert GEMMs (round 2)"
-            )
         if mc.quantization == "fp8":
+            # attention/shared-MLP projections convert here; MoE expert
+            # weights are born fp8 inside MoEMLP (per-expert _scaled_mm)
             from fusioninfer_amd.quantization import convert_linear_to_fp8
 
             n = convert_linear_to_fp8(self.model)

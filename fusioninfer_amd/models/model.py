@@ -169,6 +169,7 @@ class MoEMLP(nn.Module):
         self.norm_topk = cfg.norm_topk_prob
         self.e_start = ps.tp_rank() * (E // tp)
         self.e_end = self.e_start + E // tp
+        self.fp8 = cfg.quantization == "fp8"
         H, inter = cfg.hidden_size, cfg.moe_intermediate_size
         base = torch.initial_seed() % (2**62)
         dev = torch.empty(0).device  # honor the ambient torch.device context
@@ -190,6 +191,39 @@ class MoEMLP(nn.Module):
         else:
             egen = gen
             edev = "cpu"
+        if self.fp8:
+            # fp8 expert storage ([O, I] layout + per-expert per-output-
+            # channel scales): the fp8 path runs per-expert _scaled_mm
+            # (no bmm), halving expert-weight bytes. Same seeded draws as
+            # the bf16 path, so fp8-vs-bf16 numerics are comparable.
+            from fusioninfer_amd.quantization import quantize_weight_fp8
+
+            gu8, gus, dn8, dns = [], [], [], []
+            for e in range(self.e_start, self.e_end):
+                egen.manual_seed(base + 7919 * (e + 1))
+                g_w = torch.randn(2 * inter, H, generator=egen,
+                                  device=edev) * 0.02
+                d_w = torch.randn(H, inter, generator=egen,
+                                  device=edev) * 0.02
+                w8, s = quantize_weight_fp8(g_w)
+                gu8.append(w8)
+                gus.append(s)
+                w8, s = quantize_weight_fp8(d_w)
+                dn8.append(w8)
+                dns.append(s)
+            self.gate_up_fp8 = nn.Parameter(
+                torch.stack(gu8).to(dev), requires_grad=False
+            )  # [E_local, 2*I, H] e4m3
+            self.down_fp8 = nn.Parameter(
+                torch.stack(dn8).to(dev), requires_grad=False
+            )  # [E_local, H, I] e4m3
+            self.register_buffer(
+                "gate_up_scale", torch.stack(gus).to(dev), persistent=False
+            )
+            self.register_buffer(
+                "down_scale", torch.stack(dns).to(dev), persistent=False
+            )
+            return
         gup, down = [], []
         for e in range(self.e_start, self.e_end):
             egen.manual_seed(base + 7919 * (e + 1))
@@ -215,7 +249,60 @@ class MoEMLP(nn.Module):
             requires_grad=False,
         )  # [E_local, I, H]
 
+    def _forward_fp8(self, x) -> torch.Tensor:
+        """fp8 expert compute: per-expert torch._scaled_mm (hipBLASLt fp8)
+        with the fused silu_and_mul_fp8 epilogue between the two GEMMs.
+        Input is the (fp8, per-row scale) tuple the fused norm emits.
+        Launch-bound at high expert counts — the grouped MFMA GEMM is the
+        round-2 step up (ROADMAP.md) — but halves expert-weight bytes,
+        which is what decode is bound by."""
+        x8, xs = x
+        T, H = x8.shape
+        xd = x8.float() * xs.unsqueeze(1).float()
+        logits = xd @ self.router_weight.float().T  # [T, E]
+        probs = torch.softmax(logits, dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)
+        if self.norm_topk:
+            topv = topv / topv.sum(dim=-1, keepdim=True)
+        flat_e = topi.reshape(-1)
+        flat_w = topv.reshape(-1)
+        out = torch.zeros(T, H, dtype=torch.float32, device=x8.device)
+        cuda = x8.is_cuda
+        for el in range(self.e_end - self.e_start):
+            sel = (flat_e == self.e_start + el).nonzero(as_tuple=True)[0]
+            if sel.numel() == 0:
+                continue
+            rows = sel // self.top_k
+            if cuda:
+                gu = torch._scaled_mm(
+                    x8[rows],
+                    self.gate_up_fp8[el].t(),
+                    scale_a=xs[rows].unsqueeze(1),
+                    scale_b=self.gate_up_scale[el].unsqueeze(0),
+                    out_dtype=torch.bfloat16,
+                )
+                a8, a_s = ops.silu_and_mul_fp8(gu)
+                y = torch._scaled_mm(
+                    a8,
+                    self.down_fp8[el].t(),
+                    scale_a=a_s.unsqueeze(1),
+                    scale_b=self.down_scale[el].unsqueeze(0),
+                    out_dtype=torch.bfloat16,
+                )
+            else:  # CPU reference: dequantize + fp32 matmul
+                wg = self.gate_up_fp8[el].float() \
+                    * self.gate_up_scale[el].unsqueeze(1).float()
+                gu = (xd[rows] @ wg.t()).to(torch.bfloat16)
+                a8, a_s = ops.silu_and_mul_fp8(gu)
+                wd = self.down_fp8[el].float() \
+                    * self.down_scale[el].unsqueeze(1).float()
+                y = (a8.float() * a_s.unsqueeze(1).float()) @ wd.t()
+            out.index_add_(0, rows, y.float() * flat_w[sel].unsqueeze(1))
+        return ps.tp_all_reduce(out.to(torch.bfloat16))
+
     def forward(self, x: torch.Tensor, lora=None) -> torch.Tensor:
+        if self.fp8:
+            return self._forward_fp8(x)
         T, H = x.shape
         E_local = self.e_end - self.e_start
         logits = (x.float() @ self.router_weight.float().T)  # [T, E]

@@ -156,3 +156,46 @@ def test_fp8_scaled_mm_gpu():
     y = (x.float() @ w.float().t())
     rel = (y8 - y).norm() / y.norm()
     assert rel.item() < 0.05, rel.item()
+
+
+def test_fp8_moe_engine_close_to_bf16():
+    """fp8 MoE (per-expert _scaled_mm path; CPU reference dequant here):
+    same seeded experts as bf16, outputs close, engine generates."""
+    import torch
+
+    from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from fusioninfer_amd.models.registry import get_model_config
+
+    def build(quant):
+        torch.manual_seed(3)
+        mc = get_model_config("tiny-qwen3-moe")
+        mc.quantization = quant
+        cfg = EngineConfig(
+            model=mc,
+            cache=CacheConfig(num_gpu_blocks=64),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=512, max_model_len=128
+            ),
+            seed=3,
+        )
+        return LLMEngine(cfg, device="cpu")
+
+    prompt = [3, 1, 4, 1, 5, 9, 2, 6] * 3
+    bf16 = build(None)
+    ref = bf16.generate([prompt], SamplingParams(max_tokens=6))[0]
+    fp8 = build("fp8")
+    out = fp8.generate([prompt], SamplingParams(max_tokens=6))[0]
+    assert len(out.output_token_ids) == 6
+    # quantization perturbs logits; with a tiny random model greedy picks
+    # can drift, so compare the MoE layer output directly instead
+    layer_b = bf16.runner.model.layers[0].mlp
+    layer_f = fp8.runner.model.layers[0].mlp
+    x = torch.randn(5, bf16.cfg.model.hidden_size).to(torch.bfloat16)
+    y_b = layer_b(x).float()
+    from fusioninfer_amd.quantization import quantize_activation_fp8
+
+    y_f = layer_f(quantize_activation_fp8(x)).float()
+    rel = (y_b - y_f).norm() / y_b.norm().clamp(min=1e-6)
+    assert rel < 0.15, f"fp8 MoE relative error {rel:.3f}"
