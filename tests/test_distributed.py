@@ -552,3 +552,44 @@ def test_pp2_microbatch_prefill_matches_single_process():
             p.join(timeout=300)
             assert p.exitcode == 0, f"rank exited {p.exitcode}"
         assert results[0] == "ok" and results[1] == "ok"
+
+
+# ----------------------------------- one-shot all-reduce (small tensors)
+def _oneshot_worker(rank, port, results):
+    from fusioninfer_amd.distributed import parallel_state as ps
+
+    _init(rank, 2, port)
+    try:
+        ps.init_distributed(tensor_parallel_size=2, backend="gloo")
+        # small fp32 tensor -> one-shot (all-gather + local sum) path
+        small = torch.full((64,), float(rank + 1))
+        out = ps.tp_all_reduce(small.clone())
+        assert torch.equal(out, torch.full((64,), 3.0))
+        # large fp32 tensor -> ring all-reduce path; same result contract
+        big = torch.full((200_000,), float(rank + 1))
+        out = ps.tp_all_reduce(big.clone())
+        assert torch.equal(out, torch.full((200_000,), 3.0))
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+        ps.destroy()
+
+
+def test_tp2_oneshot_allreduce_matches_ring():
+    """Small latency-bound tensors take the one-shot (all-gather + sum)
+    path — 1 p2p step on near-fully-connected xGMI vs the ring's
+    2*(N-1); both paths must produce the exact sum."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29641
+        procs = [
+            ctx.Process(target=_oneshot_worker, args=(r, port, results))
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=120)
+            assert p.exitcode == 0, f"rank exited {p.exitcode}"
+        assert results[0] == "ok" and results[1] == "ok"
