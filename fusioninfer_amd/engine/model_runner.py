@@ -33,20 +33,40 @@ class ModelRunner:
         self.is_cuda = self.device.type == "cuda"
         mc = cfg.model
         torch.manual_seed(cfg.seed)
+        fp8_ckpt = False
+        if mc.model_path:
+            from fusioninfer_amd.models.weight_loader import checkpoint_is_fp8
+
+            fp8_ckpt = checkpoint_is_fp8(mc.model_path)
+            if fp8_ckpt and mc.quantization is None:
+                # fp8-native checkpoint implies fp8 serving (vLLM
+                # auto-detects this from quantization_config too)
+                mc.quantization = "fp8"
         with torch.device(self.device):
             self.model = CausalLM(mc).eval()
-        if mc.model_path:
+        if mc.quantization == "fp8" and fp8_ckpt:
+            # fp8-native checkpoint: convert modules FIRST (fp8 storage +
+            # scale buffers exist), then load e4m3 weights + weight_scale
             from fusioninfer_amd.models.weight_loader import load_safetensors_dir
-
-            load_safetensors_dir(self.model, mc.model_path)
-        if mc.quantization == "fp8":
-            # attention/shared-MLP projections convert here; MoE expert
-            # weights are born fp8 inside MoEMLP (per-expert _scaled_mm)
             from fusioninfer_amd.quantization import convert_linear_to_fp8
 
-            n = convert_linear_to_fp8(self.model)
-            assert n > 0
-        elif mc.quantization:
+            assert convert_linear_to_fp8(self.model) > 0
+            load_safetensors_dir(self.model, mc.model_path)
+        else:
+            if mc.model_path:
+                from fusioninfer_amd.models.weight_loader import (
+                    load_safetensors_dir,
+                )
+
+                load_safetensors_dir(self.model, mc.model_path)
+            if mc.quantization == "fp8":
+                # bf16 weights (checkpoint or random init) quantized here;
+                # MoE experts are born fp8 inside MoEMLP
+                from fusioninfer_amd.quantization import convert_linear_to_fp8
+
+                n = convert_linear_to_fp8(self.model)
+                assert n > 0
+        if mc.quantization not in (None, "fp8"):
             raise ValueError(f"unknown quantization {mc.quantization!r}")
         self.block_size = cfg.cache.block_size
         self.max_blocks_per_seq = (

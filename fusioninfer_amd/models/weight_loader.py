@@ -30,6 +30,33 @@ def _shard_cols(w: torch.Tensor, tp: int, rank: int) -> torch.Tensor:
     return w[:, rank * per : (rank + 1) * per]
 
 
+def _row_scale(s: torch.Tensor, rows: int) -> torch.Tensor:
+    """Normalize an fp8 weight_scale to per-output-row [rows] fp32:
+    checkpoints carry either a per-tensor scalar or per-channel rows."""
+    s = s.reshape(-1).float()
+    if s.numel() == 1:
+        return s.expand(rows).clone()
+    assert s.numel() == rows, (s.numel(), rows)
+    return s
+
+
+def checkpoint_is_fp8(path: str) -> bool:
+    """True when the safetensors checkpoint stores projection weights in
+    OCP e4m3 (HF fp8 checkpoints ship `weight` f8 + `weight_scale`)."""
+    from safetensors import safe_open
+
+    files = sorted(
+        f for f in os.listdir(path) if f.endswith(".safetensors")
+    )
+    if not files:
+        return False
+    with safe_open(os.path.join(path, files[0]), framework="pt") as sf:
+        for name in sf.keys():
+            if name.endswith("proj.weight"):
+                return "F8" in str(sf.get_slice(name).get_dtype()).upper()
+    return False
+
+
 def load_hf_state_dict(
     model, tensors: Iterable[Tuple[str, torch.Tensor]]
 ) -> int:
@@ -49,22 +76,61 @@ def load_hf_state_dict(
         assert param.data.shape == value.shape, (param.data.shape, value.shape)
         param.data.copy_(value.to(param.dtype))
 
+    def _is_fp8(w: torch.Tensor) -> bool:
+        return w.dtype == torch.float8_e4m3fn
+
     def try_fuse_qkv(li: int):
         keys = [(li, "q"), (li, "k"), (li, "v")]
-        if all(k in pending for k in keys):
-            qw = _shard_rows(pending.pop(keys[0]), tp, rank)
-            kw = _shard_rows(pending.pop(keys[1]), tp, rank)
-            vw = _shard_rows(pending.pop(keys[2]), tp, rank)
-            put(model.layers[li].self_attn.qkv_proj.weight,
-                torch.cat([qw, kw, vw], dim=0))
+        if not all(k in pending for k in keys):
+            return
+        fp8 = _is_fp8(pending[keys[0]])
+        skeys = [(li, "q_scale"), (li, "k_scale"), (li, "v_scale")]
+        if fp8 and not all(k in pending for k in skeys):
+            return  # wait for the weight_scale tensors
+        qw = pending.pop(keys[0])
+        kw = pending.pop(keys[1])
+        vw = pending.pop(keys[2])
+        attn = local(li).self_attn
+        w = torch.cat([
+            _shard_rows(qw, tp, rank),
+            _shard_rows(kw, tp, rank),
+            _shard_rows(vw, tp, rank),
+        ], dim=0)
+        if fp8:
+            sc = torch.cat([
+                _shard_rows(_row_scale(pending.pop(skeys[0]), qw.shape[0]),
+                            tp, rank),
+                _shard_rows(_row_scale(pending.pop(skeys[1]), kw.shape[0]),
+                            tp, rank),
+                _shard_rows(_row_scale(pending.pop(skeys[2]), vw.shape[0]),
+                            tp, rank),
+            ])
+            attn.qkv_proj.weight_scale.copy_(sc)
+        put(attn.qkv_proj.weight, w)
 
     def try_fuse_gate_up(li: int):
         keys = [(li, "gate"), (li, "up")]
-        if all(k in pending for k in keys):
-            gw = _shard_rows(pending.pop(keys[0]), tp, rank)
-            uw = _shard_rows(pending.pop(keys[1]), tp, rank)
-            put(local(li).mlp.gate_up_proj.weight,
-                torch.cat([gw, uw], dim=0))
+        if not all(k in pending for k in keys):
+            return
+        fp8 = _is_fp8(pending[keys[0]])
+        skeys = [(li, "gate_scale"), (li, "up_scale")]
+        if fp8 and not all(k in pending for k in skeys):
+            return
+        gw = pending.pop(keys[0])
+        uw = pending.pop(keys[1])
+        mlp = local(li).mlp
+        w = torch.cat([
+            _shard_rows(gw, tp, rank), _shard_rows(uw, tp, rank)
+        ], dim=0)
+        if fp8:
+            sc = torch.cat([
+                _shard_rows(_row_scale(pending.pop(skeys[0]), gw.shape[0]),
+                            tp, rank),
+                _shard_rows(_row_scale(pending.pop(skeys[1]), uw.shape[0]),
+                            tp, rank),
+            ])
+            mlp.gate_up_proj.weight_scale.copy_(sc)
+        put(mlp.gate_up_proj.weight, w)
 
     def local(li: int):
         # PP: global layer index -> this stage's slice
@@ -84,6 +150,46 @@ def load_hf_state_dict(
             mlp.gate_up_t.data[le].copy_(
                 torch.cat([gw, uw], dim=0).T.to(mlp.gate_up_t.dtype)
             )
+
+    def try_fuse_expert_fp8(li: int, e: int, le: int, mlp):
+        """fp8-mode expert slots: fp8-native checkpoint tensors load
+        directly (weight + weight_scale); bf16 checkpoint experts are
+        quantized on the way in."""
+        from fusioninfer_amd.quantization import quantize_weight_fp8
+
+        dk = (li, e, "down.weight")
+        if dk in moe_pending:
+            w = moe_pending[dk]
+            if w.dtype == torch.float8_e4m3fn:
+                sk = (li, e, "down.weight_scale")
+                if sk in moe_pending:
+                    moe_pending.pop(dk)
+                    s = _row_scale(moe_pending.pop(sk), w.shape[0])
+                    mlp.down_fp8.data[le].copy_(w)
+                    mlp.down_scale[le].copy_(s)
+            else:
+                moe_pending.pop(dk)
+                w8, s = quantize_weight_fp8(w)
+                mlp.down_fp8.data[le].copy_(w8)
+                mlp.down_scale[le].copy_(s)
+        gk, uk = (li, e, "gate.weight"), (li, e, "up.weight")
+        if gk in moe_pending and uk in moe_pending:
+            if moe_pending[gk].dtype == torch.float8_e4m3fn:
+                gsk = (li, e, "gate.weight_scale")
+                usk = (li, e, "up.weight_scale")
+                if gsk in moe_pending and usk in moe_pending:
+                    g = moe_pending.pop(gk)
+                    u = moe_pending.pop(uk)
+                    gs = _row_scale(moe_pending.pop(gsk), g.shape[0])
+                    us = _row_scale(moe_pending.pop(usk), u.shape[0])
+                    mlp.gate_up_fp8.data[le].copy_(torch.cat([g, u], dim=0))
+                    mlp.gate_up_scale[le].copy_(torch.cat([gs, us]))
+            else:
+                g = moe_pending.pop(gk)
+                u = moe_pending.pop(uk)
+                w8, s = quantize_weight_fp8(torch.cat([g, u], dim=0))
+                mlp.gate_up_fp8.data[le].copy_(w8)
+                mlp.gate_up_scale[le].copy_(s)
 
     for name, w in tensors:
         loaded += 1
@@ -112,8 +218,16 @@ def load_hf_state_dict(
             elif rest == "self_attn.v_proj.weight":
                 pending[(li, "v")] = w
                 try_fuse_qkv(li)
+            elif rest in ("self_attn.q_proj.weight_scale",
+                          "self_attn.k_proj.weight_scale",
+                          "self_attn.v_proj.weight_scale"):
+                pending[(li, rest.split(".")[1][0] + "_scale")] = w
+                try_fuse_qkv(li)
             elif rest == "self_attn.o_proj.weight":
                 put(layer.self_attn.o_proj.weight, _shard_cols(w, tp, rank))
+            elif rest == "self_attn.o_proj.weight_scale":
+                o = layer.self_attn.o_proj
+                o.weight_scale.copy_(_row_scale(w, o.weight.shape[0]))
             elif rest == "self_attn.q_norm.weight":
                 put(layer.self_attn.q_norm_weight, w)
             elif rest == "self_attn.k_norm.weight":
@@ -124,8 +238,17 @@ def load_hf_state_dict(
             elif rest == "mlp.up_proj.weight":
                 pending[(li, "up")] = w
                 try_fuse_gate_up(li)
+            elif rest == "mlp.gate_proj.weight_scale":
+                pending[(li, "gate_scale")] = w
+                try_fuse_gate_up(li)
+            elif rest == "mlp.up_proj.weight_scale":
+                pending[(li, "up_scale")] = w
+                try_fuse_gate_up(li)
             elif rest == "mlp.down_proj.weight":
                 put(layer.mlp.down_proj.weight, _shard_cols(w, tp, rank))
+            elif rest == "mlp.down_proj.weight_scale":
+                d = layer.mlp.down_proj
+                d.weight_scale.copy_(_row_scale(w, d.weight.shape[0]))
             elif rest == "mlp.gate.weight":  # MoE router (qwen3_moe)
                 put(layer.mlp.router_weight, w)
             elif parts[3] == "mlp" and parts[4] == "experts":
@@ -134,10 +257,13 @@ def load_hf_state_dict(
                 if not (mlp.e_start <= e < mlp.e_end):
                     continue  # EP: expert lives on another rank
                 which = parts[6].replace("_proj", "")  # gate/up/down
-                if which == "down":
-                    mlp.down_t.data[e - mlp.e_start].copy_(
-                        w.T.to(mlp.down_t.dtype)
-                    )
+                kind = parts[7] if len(parts) > 7 else "weight"
+                le = e - mlp.e_start
+                if mlp.fp8:
+                    moe_pending[(li, e, which + "." + kind)] = w
+                    try_fuse_expert_fp8(li, e, le, mlp)
+                elif which == "down":
+                    mlp.down_t.data[le].copy_(w.T.to(mlp.down_t.dtype))
                 else:
                     moe_pending[(li, e, which)] = w
                     try_fuse_expert(li, e)

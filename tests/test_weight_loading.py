@@ -120,3 +120,78 @@ def test_moe_expert_weights_roundtrip():
     assert torch.allclose(
         mlp.router_weight.float(), want_router.bfloat16().float()
     )
+
+
+def test_fp8_native_checkpoint_matches_quantize_on_load(tmp_path):
+    """An fp8-native checkpoint (e4m3 weights + weight_scale, the HF fp8
+    layout) must produce the SAME engine weights as loading the bf16
+    checkpoint with --quantization fp8 (both use per-output-channel
+    symmetric quantization), and fp8 mode must auto-enable."""
+    from safetensors.torch import save_file
+
+    from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from fusioninfer_amd.models.weight_loader import checkpoint_is_fp8
+    from fusioninfer_amd.quantization import quantize_weight_fp8
+
+    ps.ensure_single_process()
+    src = _model(5)
+    sd = export_hf_state_dict(src)
+
+    bf16_dir = tmp_path / "bf16"
+    bf16_dir.mkdir()
+    save_file({k: v.contiguous() for k, v in sd.items()},
+              str(bf16_dir / "model.safetensors"))
+
+    fp8_dir = tmp_path / "fp8"
+    fp8_dir.mkdir()
+    fp8_sd = {}
+    for k, v in sd.items():
+        if k.endswith("proj.weight"):
+            w8, s = quantize_weight_fp8(v)
+            fp8_sd[k] = w8.contiguous()
+            fp8_sd[k + "_scale"] = s.contiguous()
+        else:
+            fp8_sd[k] = v.contiguous()
+    save_file(fp8_sd, str(fp8_dir / "model.safetensors"))
+
+    assert checkpoint_is_fp8(str(fp8_dir))
+    assert not checkpoint_is_fp8(str(bf16_dir))
+
+    def engine(path, quant):
+        torch.manual_seed(9)
+        mc = get_model_config("tiny-qwen3")
+        mc.model_path = path
+        mc.quantization = quant
+        cfg = EngineConfig(
+            model=mc,
+            cache=CacheConfig(num_gpu_blocks=64),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=512, max_model_len=128
+            ),
+            seed=9,
+        )
+        return LLMEngine(cfg, device="cpu")
+
+    eng_a = engine(str(bf16_dir), "fp8")          # quantize-on-load
+    eng_b = engine(str(fp8_dir), None)            # fp8-native, auto-detect
+    assert eng_b.cfg.model.quantization == "fp8"
+
+    la = eng_a.runner.model.layers[0]
+    lb = eng_b.runner.model.layers[0]
+    assert torch.equal(
+        la.self_attn.qkv_proj.weight.data.view(torch.uint8),
+        lb.self_attn.qkv_proj.weight.data.view(torch.uint8),
+    )
+    torch.testing.assert_close(
+        la.self_attn.qkv_proj.weight_scale, lb.self_attn.qkv_proj.weight_scale
+    )
+    torch.testing.assert_close(
+        la.mlp.down_proj.weight_scale, lb.mlp.down_proj.weight_scale
+    )
+
+    prompt = [3, 1, 4, 1, 5, 9] * 4
+    out_a = eng_a.generate([prompt], SamplingParams(max_tokens=5))[0]
+    out_b = eng_b.generate([prompt], SamplingParams(max_tokens=5))[0]
+    assert out_a.output_token_ids == out_b.output_token_ids
