@@ -50,6 +50,11 @@ def parse_args():
                    help="shared-prefix workload (BASELINE config #4): each "
                         "prompt = one of 8 shared prefixes of this length + "
                         "a unique tail; exercises prefix caching")
+    p.add_argument("--spec-tokens", type=int, default=0,
+                   help="enable ngram speculative decoding with this draft "
+                        "length, and make prompts locally repetitive so the "
+                        "prompt-lookup proposer has matches (spec-decode "
+                        "workload mode; headline stays spec-off)")
     return p.parse_args()
 
 
@@ -57,7 +62,9 @@ class ClosedLoopLoad:
     """Keeps `concurrency` requests in flight; tracks completions + TTFT."""
 
     def __init__(self, engine, prompt_len: int, gen_len: int, concurrency: int,
-                 vocab: int, seed: int, shared_prefix_len: int = 0):
+                 vocab: int, seed: int, shared_prefix_len: int = 0,
+                 repetitive: bool = False):
+        self.repetitive = repetitive
         self.engine = engine
         self.gen_len = gen_len
         self.prompt_len = prompt_len
@@ -76,6 +83,14 @@ class ClosedLoopLoad:
             ]
 
     def _new_prompt(self):
+        if self.repetitive:
+            # period-P repeated random chunk: summarization/extraction-like
+            # texts where ngram prompt-lookup actually lands drafts
+            period = 32
+            chunk = torch.randint(0, self.vocab, (period,),
+                                  generator=self.rng).tolist()
+            reps = (self.prompt_len + period - 1) // period
+            return (chunk * reps)[: self.prompt_len]
         if self.prefixes:
             i = int(torch.randint(0, len(self.prefixes), (1,),
                                   generator=self.rng))
@@ -156,6 +171,12 @@ def main():
         seed=1234 + rank,
         enforce_eager=args.enforce_eager,
     )
+    if args.spec_tokens > 0:
+        from fusioninfer_amd.engine.spec_decode import SpeculativeConfig
+
+        cfg.speculative = SpeculativeConfig(
+            num_speculative_tokens=args.spec_tokens
+        )
     # TP: ranks form DP groups of size tp; only group leaders drive load.
     driver_group = None
     if distributed and args.tp > 1:
@@ -176,6 +197,7 @@ def main():
         engine, args.prompt_len, args.gen_len, args.concurrency,
         mc.vocab_size, seed=99 + rank,
         shared_prefix_len=args.shared_prefix_len,
+        repetitive=args.spec_tokens > 0,
     )
     load.top_up()
 
@@ -213,15 +235,26 @@ def main():
     ttfts = load.ttfts[ttft_mark:] or load.ttfts
 
     engine.stop_workers()
+    spec_stats = None
+    if args.spec_tokens > 0 and engine.num_spec_draft_tokens:
+        spec_stats = {
+            "draft_tokens": engine.num_spec_draft_tokens,
+            "accepted_tokens": engine.num_spec_accepted_tokens,
+            "accept_rate": round(
+                engine.num_spec_accepted_tokens
+                / max(engine.num_spec_draft_tokens, 1), 3
+            ),
+        }
     _aggregate_and_report(args, rank, world, distributed, device, use_cuda,
                           elapsed, completed, gen_tokens, ttfts,
-                          is_driver=True)
+                          is_driver=True, spec_stats=spec_stats)
     if distributed:
         dist.destroy_process_group()
 
 
 def _aggregate_and_report(args, rank, world, distributed, device, use_cuda,
-                          elapsed, completed, gen_tokens, ttfts, is_driver):
+                          elapsed, completed, gen_tokens, ttfts, is_driver,
+                          spec_stats=None):
     if distributed:
         dev = device if use_cuda else "cpu"
         t = torch.tensor([elapsed], dtype=torch.float64, device=dev)
@@ -257,6 +290,7 @@ def _aggregate_and_report(args, rank, world, distributed, device, use_cuda,
         + ("+kv_fp8" if args.kv_cache_dtype == "fp8" else ""),
         "data": "synthetic",
         "p50_ttft_ms": round(p50_ttft_ms, 1),
+        **({"spec_decode": spec_stats} if spec_stats else {}),
         "tokens_per_s": round(gen_tokens / elapsed, 1) if elapsed else 0.0,
         "config": {
             "model": args.model,

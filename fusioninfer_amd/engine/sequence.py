@@ -19,6 +19,7 @@ class SamplingParams:
     def __init__(
         self,
         max_tokens: int = 128,
+        min_tokens: int = 0,
         temperature: float = 0.0,
         top_p: float = 1.0,
         top_k: int = 0,
@@ -32,6 +33,7 @@ class SamplingParams:
         guided=None,
     ):
         self.max_tokens = max_tokens
+        self.min_tokens = min_tokens  # suppress stop tokens until reached
         self.temperature = temperature
         self.top_p = top_p
         self.top_k = top_k
@@ -102,6 +104,7 @@ class Sequence:
         if (
             not s.ignore_eos
             and self.output_token_ids
+            and len(self.output_token_ids) >= s.min_tokens
             and self.output_token_ids[-1] in s.stop_token_ids
         ):
             return True

@@ -151,6 +151,7 @@ def build_app(serving: ServingEngine, model_name: str,
             stop_ids.append(tokenizer.eos_token_id)
         return SamplingParams(
             max_tokens=int(body.get("max_tokens", 16)),
+            min_tokens=int(body.get("min_tokens", 0)),
             temperature=float(body.get("temperature", 1.0)),
             top_p=float(body.get("top_p", 1.0)),
             top_k=int(body.get("top_k", 0)),
@@ -198,6 +199,23 @@ def build_app(serving: ServingEngine, model_name: str,
             yield tok, finished
             if finished:
                 return
+
+    @app.post("/tokenize")
+    async def tokenize(request: Request):
+        body = await request.json()
+        ids = _encode(body.get("prompt", ""))
+        return {
+            "tokens": ids,
+            "count": len(ids),
+            "max_model_len": serving.engine.cfg.scheduler.max_model_len,
+        }
+
+    @app.post("/detokenize")
+    async def detokenize(request: Request):
+        body = await request.json()
+        return {"prompt": tokenizer.decode(
+            [int(t) for t in body.get("tokens", [])]
+        )}
 
     @app.post("/pd/prefill")
     async def pd_prefill(request: Request):

@@ -413,3 +413,18 @@ def test_abort_swapped_request_frees_swap_space():
         eng.step()
     assert eng._swap_bytes == 0
     assert eng.block_manager.num_free() == eng.block_manager.num_blocks
+
+
+def test_min_tokens_suppresses_stop():
+    """vLLM min_tokens parity: stop tokens ignored until min reached."""
+    from fusioninfer_amd.engine.sequence import SamplingParams, Sequence
+
+    sp = SamplingParams(max_tokens=8, min_tokens=3, ignore_eos=False,
+                        stop_token_ids=[7])
+    seq = Sequence("s", [1, 2], sp)
+    seq.append_token(7)
+    assert not seq.check_stop()      # 1 < min_tokens
+    seq.append_token(5)
+    assert not seq.check_stop()
+    seq.append_token(7)
+    assert seq.check_stop()          # 3 >= min_tokens and stop token
