@@ -253,3 +253,53 @@ def test_gpu_moe_engine_generates():
             exp[t] += float(topv[t, k]) * y
     rel = (got - exp).norm() / exp.norm()
     assert rel.item() < 0.05, rel.item()
+
+
+def test_gpu_spec_decode_token_exact_and_fires():
+    """ngram speculative decoding on the HIP path: token-exact with plain
+    greedy decode (hipGraph decode + paged-prefill verify), drafts fire."""
+    prompts = [[3, 1, 4, 1, 5, 9] * 8, [2, 7, 2, 7, 2, 7] * 8]
+    ref = [
+        o.output_token_ids
+        for o in make_engine(False).generate(
+            prompts, SamplingParams(max_tokens=24)
+        )
+    ]
+    from fusioninfer_amd.engine.spec_decode import SpeculativeConfig
+
+    mc = get_model_config("Qwen3-0.6B")
+    mc.num_layers = 4
+    cfg = EngineConfig(
+        model=mc,
+        cache=CacheConfig(num_gpu_blocks=512),
+        scheduler=SchedulerConfig(
+            max_num_seqs=16, max_num_batched_tokens=2048, max_model_len=512
+        ),
+        speculative=SpeculativeConfig(num_speculative_tokens=4),
+        seed=7,
+        enforce_eager=False,
+    )
+    eng = LLMEngine(cfg, device="cuda:0")
+    outs = eng.generate(prompts, SamplingParams(max_tokens=24))
+    assert [o.output_token_ids for o in outs] == ref
+    assert eng.num_spec_draft_tokens > 0
+
+
+def test_gpu_guided_decoding_masks_on_device():
+    """Grammar masks applied to device logits: choice output exact."""
+    from fusioninfer_amd.guided import Vocabulary, build_guided
+
+    eng = make_engine(True)
+    vocab = Vocabulary(
+        eng.cfg.model.vocab_size,
+        lambda t: bytes([max(t - 3, 0) & 0xFF]).decode("utf-8", "replace"),
+    )
+    guided = build_guided("choice", ["yes", "no", "maybe"], vocab)
+    outs = eng.generate(
+        [[5, 6, 7] * 6],
+        SamplingParams(max_tokens=16, temperature=0.0, guided=guided),
+    )
+    text = bytes(
+        max(t - 3, 0) & 0xFF for t in outs[0].output_token_ids
+    ).decode("utf-8", "replace")
+    assert text in {"yes", "no", "maybe"}
