@@ -410,6 +410,12 @@ class LLMEngine:
         outputs: List[RequestOutput] = []
         keep = []
         for i, s in enumerate(sample_seqs):
+            if s.sampling.logit_bias:
+                ids = list(s.sampling.logit_bias)
+                logits_f[i, ids] += torch.tensor(
+                    [s.sampling.logit_bias[t] for t in ids],
+                    dtype=logits_f.dtype, device=logits_f.device,
+                )
             g = s.sampling.guided
             if g is not None:
                 allowed = g.allowed_mask(logits_f.device)

@@ -31,6 +31,7 @@ class SamplingParams:
         ignore_eos: bool = True,
         stop_token_ids: Optional[List[int]] = None,
         guided=None,
+        logit_bias=None,
     ):
         self.max_tokens = max_tokens
         self.min_tokens = min_tokens  # suppress stop tokens until reached
@@ -47,6 +48,9 @@ class SamplingParams:
         # guided decoding: a guided.GuidedDecoder enforcing a grammar via
         # logits masks (None = unconstrained)
         self.guided = guided
+        # OpenAI logit_bias: {token_id: additive bias in [-100, 100]}
+        self.logit_bias = {int(k): float(v)
+                           for k, v in (logit_bias or {}).items()}
 
 
 class Sequence:

@@ -57,6 +57,7 @@ def seq_is_draftable(seq: Sequence) -> bool:
         and sp.frequency_penalty == 0.0
         and sp.logprobs is None
         and getattr(sp, "guided", None) is None
+        and not getattr(sp, "logit_bias", None)
     )
 
 

@@ -46,6 +46,39 @@ def decode_tokens(token_ids: List[int]) -> str:
         return " ".join(str(t) for t in token_ids)
 
 
+class StopStreamFilter:
+    """Streaming stop-string matcher: holds back the longest possible
+    stop prefix so a match that spans chunk boundaries is never partially
+    emitted. push() returns (text safe to emit now, matched?)."""
+
+    def __init__(self, stops: List[str]):
+        self.stops = stops
+        self.hold = max((len(s) for s in stops), default=1) - 1
+        self.text = ""
+        self.sent = 0
+
+    def push(self, piece: str):
+        self.text += piece
+        hit = min(
+            (i for i in (self.text.find(s) for s in self.stops) if i >= 0),
+            default=-1,
+        )
+        if hit >= 0:
+            emit = self.text[self.sent: hit]
+            self.sent = hit
+            return emit, True
+        safe = max(len(self.text) - self.hold, self.sent)
+        emit = self.text[self.sent: safe]
+        self.sent = safe
+        return emit, False
+
+    def flush(self) -> str:
+        """Emit the held-back tail (stream ended without a match)."""
+        emit = self.text[self.sent:]
+        self.sent = len(self.text)
+        return emit
+
+
 def build_app(serving: ServingEngine, model_name: str,
               tokenizer=None) -> FastAPI:
     app = FastAPI(title="fusioninfer-amd")
@@ -163,6 +196,7 @@ def build_app(serving: ServingEngine, model_name: str,
             ignore_eos=bool(body.get("ignore_eos", False)),
             stop_token_ids=stop_ids,
             guided=_guided_from(body),
+            logit_bias=body.get("logit_bias"),
         )
 
     def _parse_stops(body: Dict[str, Any]) -> List[str]:
@@ -274,24 +308,19 @@ def build_app(serving: ServingEngine, model_name: str,
         if body.get("stream"):
             async def sse():
                 done = False
-                text = ""
+                filt = StopStreamFilter(stops) if stops else None
                 try:
                     async for tok, finished in _stream(q):
                         done = finished
                         delta = tokenizer.decode_one(tok) if tok is not None else ""
-                        if stops and tok is not None:
+                        if filt is not None and tok is not None:
                             # hold back text that could extend into a stop
                             # string; cut the stream at the first match
-                            text += delta
-                            hit = min(
-                                (i for i in (text.find(s) for s in stops)
-                                 if i >= 0),
-                                default=-1,
-                            )
-                            if hit >= 0:
-                                delta = text[len(text) - len(delta):hit] \
-                                    if hit >= len(text) - len(delta) else ""
+                            delta, hit = filt.push(delta)
+                            if hit:
                                 finished = True
+                            elif finished:
+                                delta += filt.flush()
                         chunk = {
                             "id": cid,
                             "object": "text_completion",
@@ -386,28 +415,52 @@ def build_app(serving: ServingEngine, model_name: str,
         if body.get("stream"):
             async def sse():
                 first = True
-                async for tok, finished in _stream(q):
-                    delta: Dict[str, Any] = {}
-                    if first:
-                        delta["role"] = "assistant"
-                        first = False
-                    if tok is not None:
-                        delta["content"] = tokenizer.decode_one(tok)
-                    chunk = {
-                        "id": cid,
-                        "object": "chat.completion.chunk",
-                        "created": created,
-                        "model": model_name,
-                        "choices": [
-                            {
-                                "index": 0,
-                                "delta": delta,
-                                "finish_reason": "stop" if finished else None,
-                            }
-                        ],
-                    }
-                    yield f"data: {json.dumps(chunk)}\n\n"
-                yield "data: [DONE]\n\n"
+                done = False
+                filt = StopStreamFilter(stops) if stops else None
+                try:
+                    async for tok, finished in _stream(q):
+                        done = finished
+                        delta: Dict[str, Any] = {}
+                        if first:
+                            delta["role"] = "assistant"
+                            first = False
+                        piece = (
+                            tokenizer.decode_one(tok)
+                            if tok is not None else ""
+                        )
+                        if filt is not None and tok is not None:
+                            # cut the stream at the first stop-string match
+                            piece, hit = filt.push(piece)
+                            if hit:
+                                finished = True
+                            elif finished:
+                                piece += filt.flush()
+                        if piece:
+                            delta["content"] = piece
+                        chunk = {
+                            "id": cid,
+                            "object": "chat.completion.chunk",
+                            "created": created,
+                            "model": model_name,
+                            "choices": [
+                                {
+                                    "index": 0,
+                                    "delta": delta,
+                                    "finish_reason": (
+                                        "stop" if finished else None
+                                    ),
+                                }
+                            ],
+                        }
+                        yield f"data: {json.dumps(chunk)}\n\n"
+                        if finished and not done:
+                            done = True
+                            serving.abort(req_id)
+                            break
+                    yield "data: [DONE]\n\n"
+                finally:
+                    if not done:  # client disconnected mid-stream
+                        serving.abort(req_id)
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 

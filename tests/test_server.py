@@ -301,3 +301,59 @@ def test_guided_regex_and_bad_pattern(app):
             assert r.status_code == 400
 
     asyncio.run(run())
+
+
+def test_logit_bias_forces_token(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": [5, 6, 7], "max_tokens": 3,
+                      "temperature": 0.0,
+                      "logit_bias": {"42": 100.0}},
+            )
+            assert r.status_code == 200
+            # +100 on a random-init model's logits dominates every step
+            assert r.json()["choices"][0]["token_ids"] == [42, 42, 42]
+
+    asyncio.run(run())
+
+
+def test_chat_stream_stop_strings(app):
+    async def run():
+        async with _client(app) as c:
+            # unconstrained streamed reference (same per-token decode as
+            # the stop-matching path), then stop on a substring of it
+            async with c.stream(
+                "POST", "/v1/chat/completions",
+                json={"messages": [{"role": "user", "content": "hi"}],
+                      "max_tokens": 12, "temperature": 0.0, "stream": True},
+            ) as resp:
+                full = ""
+                async for line in resp.aiter_lines():
+                    if not line.startswith("data: ") or line == "data: [DONE]":
+                        continue
+                    chunk = json.loads(line[6:])
+                    full += chunk["choices"][0]["delta"].get("content", "")
+            stop = full[3:6]
+            assert stop
+            async with c.stream(
+                "POST", "/v1/chat/completions",
+                json={"messages": [{"role": "user", "content": "hi"}],
+                      "max_tokens": 12, "temperature": 0.0, "stream": True,
+                      "stop": [stop]},
+            ) as resp:
+                text = ""
+                finished = False
+                async for line in resp.aiter_lines():
+                    if not line.startswith("data: ") or line == "data: [DONE]":
+                        continue
+                    chunk = json.loads(line[6:])
+                    ch = chunk["choices"][0]
+                    text += ch["delta"].get("content", "")
+                    finished = finished or ch["finish_reason"] == "stop"
+            assert finished
+            assert stop not in text
+            assert text == full[: full.find(stop)]
+
+    asyncio.run(run())
