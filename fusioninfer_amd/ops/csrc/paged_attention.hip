@@ -38,7 +38,12 @@ constexpr float kNegInf = -1e30f;
 // full (4-wave measured ~5% faster there).
 // FP8: the cache holds OCP e4m3 at scale 1.0 — half the KV bytes of bf16
 // (this kernel is KV-bandwidth-bound); v_cvt_pk_f32_fp8 dequant on load.
-template <int D, int G, int kNWaves, bool FP8>
+// HS (head split): waves are partitioned into HS head-groups of G/HS
+// heads each; a wave streams chunks at stride kNWaves/HS for ITS head
+// slice only. G=8 runs HS=2 so the per-wave state is the G=4 footprint —
+// the monolithic G=8 variant spilled ~500 B/lane (W=8) or sat at
+// occupancy 1 (W=4).
+template <int D, int G, int kNWaves, bool FP8, int HS = 1>
 __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     u16* __restrict__ out,            // [S, Hq, D] (written when 1 partition)
     float* __restrict__ ml_ws,        // [S, Hq, P, 2] (multi-partition)
@@ -60,6 +65,11 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   const int tid = threadIdx.x;
   const int wave = tid / kWaveSize;
   const int lane = tid % kWaveSize;
+  // head-split decomposition: wave = cwave * HS + hsplit
+  constexpr int G2 = G / HS;          // heads this wave carries
+  constexpr int kCWaves = kNWaves / HS;  // waves streaming chunks together
+  const int hsplit = wave % HS;
+  const int cwave = wave / HS;
 
   const int ctx = seq_lens[seq];
   const int num_chunks = (ctx + kBlockSz - 1) / kBlockSz;
@@ -80,10 +90,10 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   }
 
   __shared__ float q_lds[G][D];
-  __shared__ float p_lds[kNWaves][kBlockSz][G];
-  __shared__ float merge_m[kNWaves][G];
-  __shared__ float merge_l[kNWaves][G];
-  __shared__ float merge_acc[kNWaves][G][D];
+  __shared__ float p_lds[kNWaves][kBlockSz][G2];
+  __shared__ float merge_m[kNWaves][G2];
+  __shared__ float merge_l[kNWaves][G2];
+  __shared__ float merge_acc[kNWaves][G2][D];
 
   // stage q (G heads) into LDS as fp32 (pre-scaled)
   {
@@ -95,9 +105,9 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   __syncthreads();
 
   // online-softmax state, wave-uniform (every lane holds the same copy)
-  float m[G], l[G], acc[G][2];
+  float m[G2], l[G2], acc[G2][2];
 #pragma unroll
-  for (int g = 0; g < G; ++g) {
+  for (int g = 0; g < G2; ++g) {
     m[g] = kNegInf;
     l[g] = 0.f;
     acc[g][0] = acc[g][1] = 0.f;
@@ -123,17 +133,17 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       dst[j8] = reinterpret_cast<const uint4*>(k_row)[j8];
   };
   int64_t kv_base = 0;
-  if (chunk_lo + wave < chunk_hi) {
-    kv_base = kv_base_of(chunk_lo + wave);
+  if (chunk_lo + cwave < chunk_hi) {
+    kv_base = kv_base_of(chunk_lo + cwave);
     load_k(kv_base, kraw);
   }
 
-  for (int chunk = chunk_lo + wave; chunk < chunk_hi; chunk += kNWaves) {
+  for (int chunk = chunk_lo + cwave; chunk < chunk_hi; chunk += kCWaves) {
     const int token_pos = chunk * kBlockSz + tok;
     const int64_t kv_base_cur = kv_base;
 
-    // ---- phase A: scores for 16 tokens x G heads ----
-    float s[G];
+    // ---- phase A: scores for 16 tokens x G2 heads ----
+    float s[G2];
     {
       float kf[DPQ];
       if (FP8) {
@@ -149,13 +159,13 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
         }
       }
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
+      for (int g = 0; g < G2; ++g) {
         float acc_s = 0.f;
 #pragma unroll
         for (int j4 = 0; j4 < DPQ / 4; ++j4) {
           // float4 -> ds_read_b128 (4x fewer LDS cycles than scalar reads)
           const float4 qv = *reinterpret_cast<const float4*>(
-              &q_lds[g][quad * DPQ + j4 * 4]);
+              &q_lds[hsplit * G2 + g][quad * DPQ + j4 * 4]);
           acc_s = fmaf(kf[j4 * 4 + 0], qv.x, acc_s);
           acc_s = fmaf(kf[j4 * 4 + 1], qv.y, acc_s);
           acc_s = fmaf(kf[j4 * 4 + 2], qv.z, acc_s);
@@ -165,26 +175,26 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       }
     }
     // issue next chunk's K now; it lands under softmax + PV
-    if (chunk + kNWaves < chunk_hi) {
-      kv_base = kv_base_of(chunk + kNWaves);
+    if (chunk + kCWaves < chunk_hi) {
+      kv_base = kv_base_of(chunk + kCWaves);
       load_k(kv_base, kraw);
     }
     // fence: stop the scheduler from hoisting phase-B work (and its live
     // ranges) above the softmax — cross-phase overlap doubled VGPRs
-    // (G=8 spills under the fences; its register file is already full)
-    if (G <= 4) __builtin_amdgcn_sched_barrier(0);
+    // (G2 is always <= 4: G=8 is head-split)
+    __builtin_amdgcn_sched_barrier(0);
     // reduce over the 4 dim-quarters (lanes 4t..4t+3)
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
+    for (int g = 0; g < G2; ++g) {
       s[g] += __shfl_xor(s[g], 1, 64);
       s[g] += __shfl_xor(s[g], 2, 64);
       if (token_pos >= ctx) s[g] = kNegInf;
     }
 
     // chunk max over tokens (xor 4..32 spans the 16 token groups)
-    float alpha[G];
+    float alpha[G2];
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
+    for (int g = 0; g < G2; ++g) {
       float cm = s[g];
 #pragma unroll
       for (int off = 4; off < 64; off <<= 1)
@@ -197,7 +207,7 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
 
     // probs + row-sum; every token is replicated on 4 lanes -> scale by 1/4
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
+    for (int g = 0; g < G2; ++g) {
       float p = (s[g] <= kNegInf) ? 0.f : __expf(s[g] - m[g]);
       if (quad == 0) p_lds[wave][tok][g] = p;
       float psum = p;
@@ -206,55 +216,38 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       l[g] = l[g] * alpha[g] + psum * 0.25f;
     }
 
-    if (G <= 4) __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_sched_barrier(0);
     // ---- phase B: PV accumulate; lane covers dims {2l, 2l+1} ----
-    const int nvalid = min(ctx - chunk * kBlockSz, kBlockSz);
     const CT* v_rows = v_cache + kv_base_cur;
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
+    for (int g = 0; g < G2; ++g) {
       acc[g][0] *= alpha[g];
       acc[g][1] *= alpha[g];
     }
     if (lane < D / 2) {
-      // G<=4: preload ALL 16 token rows' bits first (p is 0 for padding
+      // preload ALL 16 token rows' bits first (p is 0 for padding
       // tokens — phase A masks every invalid position — so processing the
       // full block is safe and the 16 loads overlap instead of serializing)
-      u32 vball[G <= 4 ? kBlockSz : 1];
-      if (G <= 4) {
+      u32 vball[kBlockSz];
 #pragma unroll
-        for (int t = 0; t < kBlockSz; ++t)
-          vball[t] = FP8
-              ? static_cast<u32>(*reinterpret_cast<const u16*>(
-                    v_rows + t * D + 2 * lane))
-              : *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
-      }
-      const int tmax = (G <= 4) ? kBlockSz : nvalid;
-      for (int t = 0; t < tmax; ++t) {
+      for (int t = 0; t < kBlockSz; ++t)
+        vball[t] = FP8
+            ? static_cast<u32>(*reinterpret_cast<const u16*>(
+                  v_rows + t * D + 2 * lane))
+            : *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
+      for (int t = 0; t < kBlockSz; ++t) {
         float v0, v1;
-        if (G <= 4) {
-          if (FP8) {
-            float vf[4];
-            unpack_fp8x4(vball[t], vf);
-            v0 = vf[0];
-            v1 = vf[1];
-          } else {
-            v0 = bf16_to_f32(static_cast<u16>(vball[t] & 0xffff));
-            v1 = bf16_to_f32(static_cast<u16>(vball[t] >> 16));
-          }
-        } else if (FP8) {
-          const u16 vb = *reinterpret_cast<const u16*>(v_rows + t * D + 2 * lane);
+        if (FP8) {
           float vf[4];
-          unpack_fp8x4(vb, vf);
+          unpack_fp8x4(vball[t], vf);
           v0 = vf[0];
           v1 = vf[1];
         } else {
-          const u32 vbits =
-              *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
-          v0 = bf16_to_f32(static_cast<u16>(vbits & 0xffff));
-          v1 = bf16_to_f32(static_cast<u16>(vbits >> 16));
+          v0 = bf16_to_f32(static_cast<u16>(vball[t] & 0xffff));
+          v1 = bf16_to_f32(static_cast<u16>(vball[t] >> 16));
         }
 #pragma unroll
-        for (int g = 0; g < G; ++g) {
+        for (int g = 0; g < G2; ++g) {
           const float p = p_lds[wave][t][g];
           acc[g][0] = fmaf(p, v0, acc[g][0]);
           acc[g][1] = fmaf(p, v1, acc[g][1]);
@@ -263,39 +256,39 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     }
   }
 
-  // ---- cross-wave flash merge ----
-  if (lane < G) {
+  // ---- cross-wave flash merge (within each head-group) ----
+  if (lane < G2) {
     merge_m[wave][lane] = m[lane];
     merge_l[wave][lane] = l[lane];
   }
   if (lane < D / 2) {
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
+    for (int g = 0; g < G2; ++g) {
       merge_acc[wave][g][2 * lane] = acc[g][0];
       merge_acc[wave][g][2 * lane + 1] = acc[g][1];
     }
   }
   __syncthreads();
 
-  // each wave merges a share of the G heads
-  constexpr int kGPerWave = (G + kNWaves - 1) / kNWaves;
+  // head-group hsplit's G2 heads are distributed over its kCWaves waves
 #pragma unroll
-  for (int gi = 0; gi < kGPerWave; ++gi) {
-    const int g = wave * kGPerWave + gi;
-    if (g >= G || lane >= D / 2) continue;
+  for (int g = 0; g < G2; ++g) {
+    if ((g % kCWaves) != cwave || lane >= D / 2) continue;
     float gm = kNegInf;
 #pragma unroll
-    for (int w = 0; w < kNWaves; ++w) gm = fmaxf(gm, merge_m[w][g]);
+    for (int c = 0; c < kCWaves; ++c)
+      gm = fmaxf(gm, merge_m[c * HS + hsplit][g]);
     float L = 0.f, o0 = 0.f, o1 = 0.f;
 #pragma unroll
-    for (int w = 0; w < kNWaves; ++w) {
+    for (int c = 0; c < kCWaves; ++c) {
+      const int w = c * HS + hsplit;
       const float mw = merge_m[w][g];
       const float f = (mw <= kNegInf) ? 0.f : __expf(mw - gm);
       L += merge_l[w][g] * f;
       o0 = fmaf(merge_acc[w][g][2 * lane], f, o0);
       o1 = fmaf(merge_acc[w][g][2 * lane + 1], f, o1);
     }
-    const int h = kv_head * G + g;
+    const int h = kv_head * G + hsplit * G2 + g;
     if (num_parts == 1) {
       const float inv = 1.f / L;
       u16* o_row = out + (static_cast<int64_t>(seq) * num_heads + h) * D;
@@ -361,18 +354,23 @@ void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
   const bool wide = num_seqs * num_kv_heads * num_parts < 256;
   const int nwaves = wide ? 8 : 4;
   dim3 grid(num_seqs, num_kv_heads, num_parts), block(nwaves * kWaveSize);
-#define FI_LAUNCH_1(DD, GG, NW, F8)                                           \
-  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG, NW, F8>), grid, block, \
-                     0, stream, out, ml_ws, acc_ws, q, k_cache, v_cache,      \
-                     block_tables, seq_lens, q_stride, max_blocks,            \
+#define FI_LAUNCH_1(DD, GG, NW, F8, HSP)                                      \
+  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG, NW, F8, HSP>), grid,   \
+                     block, 0, stream, out, ml_ws, acc_ws, q, k_cache,        \
+                     v_cache, block_tables, seq_lens, q_stride, max_blocks,   \
                      num_kv_heads, scale)
+// G=8 runs head-split (HS=2): two waves share each chunk position and
+// carry 4 heads each — the monolithic G=8 register file spilled. The
+// K/V lines the pair re-reads stay in the CU's L1 (same block, same
+// schedule), so HBM traffic is unchanged.
+#define FI_HS(GG) (GG == 8 ? 2 : 1)
 #define FI_LAUNCH(DD, GG)                                                     \
   if (fp8) {                                                                  \
-    if (wide) { FI_LAUNCH_1(DD, GG, 8, true); }                               \
-    else      { FI_LAUNCH_1(DD, GG, 4, true); }                               \
+    if (wide) { FI_LAUNCH_1(DD, GG, 8, true, FI_HS(GG)); }                    \
+    else      { FI_LAUNCH_1(DD, GG, 4, true, FI_HS(GG)); }                    \
   } else {                                                                    \
-    if (wide) { FI_LAUNCH_1(DD, GG, 8, false); }                              \
-    else      { FI_LAUNCH_1(DD, GG, 4, false); }                              \
+    if (wide) { FI_LAUNCH_1(DD, GG, 8, false, FI_HS(GG)); }                   \
+    else      { FI_LAUNCH_1(DD, GG, 4, false, FI_HS(GG)); }                   \
   }                                                                           \
   if (num_parts > 1) {                                                        \
     dim3 rgrid(num_seqs, num_kv_heads * GG), rblock(kWaveSize);               \

@@ -45,11 +45,12 @@ def bench_prefill():
         print(f"  {nseq}x{L}: {t*1e6:8.1f} us  {flops/t/1e12:7.1f} TF/s")
 
 
-def bench_decode():
-    print("== paged decode attention (GQA 32/8, D=128, bs=16) ==")
+def bench_decode(group=4):
+    Hq, Hk = 32, 32 // group
+    print(f"== paged decode attention (GQA 32/{Hk}, G={group}, D=128, bs=16) ==")
     for batch, ctx in [(1, 1024), (8, 1024), (64, 1024), (256, 1024),
                        (64, 4096), (256, 2048), (8, 8192)]:
-        Hq, Hk, D, bs = 32, 8, 128, 16
+        D, bs = 128, 16
         nblk = (ctx + bs - 1) // bs
         total_blocks = batch * nblk + 1
         q = torch.randn(batch, Hq, D, dtype=torch.bfloat16, device="cuda")
@@ -90,6 +91,7 @@ if __name__ == "__main__":
     if which in ("prefill", "all"):
         bench_prefill()
     if which in ("decode", "all"):
-        bench_decode()
+        bench_decode(group=4)
+        bench_decode(group=8)
     if which in ("elementwise", "all"):
         bench_elementwise()
