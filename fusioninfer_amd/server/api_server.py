@@ -390,9 +390,14 @@ def build_app(serving: ServingEngine, model_name: str,
     async def chat_completions(request: Request):
         body = await request.json()
         messages = body.get("messages", [])
-        text = "\n".join(
-            f"{m.get('role', 'user')}: {m.get('content', '')}" for m in messages
-        )
+        try:
+            # vendored chat template (tokenizer_config.json) when present;
+            # flat role-prefixed transcript otherwise
+            text = tokenizer.apply_chat_template(messages)
+        except ValueError as e:
+            return JSONResponse({"error": {"message": str(e),
+                                           "type": "invalid_request_error"}},
+                                400)
         prompt_ids = _encode(text)
         try:
             sampling = _sampling_from(body)

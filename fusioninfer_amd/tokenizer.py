@@ -18,12 +18,49 @@ import os
 from typing import List, Optional, Union
 
 
+def _render_chat_template(template: str, messages, bos: str, eos: str,
+                          add_generation_prompt: bool) -> str:
+    """HF-style chat template rendering (jinja2, sandboxed env with the
+    helpers templates expect: raise_exception, tojson)."""
+    from jinja2.sandbox import ImmutableSandboxedEnvironment
+
+    env = ImmutableSandboxedEnvironment(trim_blocks=True, lstrip_blocks=True)
+
+    def raise_exception(msg):
+        raise ValueError(f"chat template error: {msg}")
+
+    env.globals["raise_exception"] = raise_exception
+    return env.from_string(template).render(
+        messages=messages,
+        bos_token=bos,
+        eos_token=eos,
+        add_generation_prompt=add_generation_prompt,
+    )
+
+
+def default_chat_format(messages) -> str:
+    """Template-less fallback: the flat role-prefixed transcript."""
+    text = "\n".join(
+        f"{m.get('role', 'user')}: {m.get('content', '')}" for m in messages
+    )
+    return text
+
+
 class ByteTokenizer:
     """Reversible byte-level fallback (no vocab files needed)."""
 
     def __init__(self, vocab_size: int):
         self.vocab_size = vocab_size
         self.eos_token_id: Optional[int] = None
+        self.chat_template: Optional[str] = None
+
+    def apply_chat_template(self, messages,
+                            add_generation_prompt: bool = True) -> str:
+        if self.chat_template:
+            return _render_chat_template(
+                self.chat_template, messages, "", "", add_generation_prompt
+            )
+        return default_chat_format(messages)
 
     def encode(self, text: str) -> List[int]:
         return [min(b + 3, self.vocab_size - 1) for b in text.encode("utf-8")]
@@ -51,11 +88,50 @@ class HFTokenizer:
         # 151669 -> 151936); ids past the tokenizer's range decode to ""
         self._n = self._tok.get_vocab_size(with_added_tokens=True)
         self.eos_token_id = None
+        self.chat_template: Optional[str] = None
+        self._eos_token = ""
+        self._bos_token = ""
         for name in ("</s>", "<|endoftext|>", "<|im_end|>", "<eos>"):
             tid = self._tok.token_to_id(name)
             if tid is not None:
                 self.eos_token_id = tid
+                self._eos_token = name
                 break
+        # tokenizer_config.json next to tokenizer.json carries the chat
+        # template + canonical special tokens (HF layout)
+        cfg_path = os.path.join(os.path.dirname(path),
+                                "tokenizer_config.json")
+        if os.path.isfile(cfg_path):
+            import json as _json
+
+            try:
+                with open(cfg_path) as f:
+                    tc = _json.load(f)
+            except ValueError:
+                tc = {}
+            tmpl = tc.get("chat_template")
+            if isinstance(tmpl, str):
+                self.chat_template = tmpl
+            for key, attr in (("eos_token", "_eos_token"),
+                              ("bos_token", "_bos_token")):
+                v = tc.get(key)
+                if isinstance(v, dict):
+                    v = v.get("content")
+                if isinstance(v, str):
+                    setattr(self, attr, v)
+                    if key == "eos_token":
+                        tid = self._tok.token_to_id(v)
+                        if tid is not None:
+                            self.eos_token_id = tid
+
+    def apply_chat_template(self, messages,
+                            add_generation_prompt: bool = True) -> str:
+        if self.chat_template:
+            return _render_chat_template(
+                self.chat_template, messages, self._bos_token,
+                self._eos_token, add_generation_prompt,
+            )
+        return default_chat_format(messages)
 
     def encode(self, text: str) -> List[int]:
         return self._tok.encode(text, add_special_tokens=False).ids

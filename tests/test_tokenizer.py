@@ -113,3 +113,41 @@ def test_server_with_hf_tokenizer(tok_file):
         asyncio.run(run())
     finally:
         serving.shutdown()
+
+
+def test_chat_template_rendering(tok_file, tmp_path):
+    """tokenizer_config.json chat_template renders via jinja2; fallback
+    is the flat role-prefixed transcript."""
+    import shutil
+
+    d = tmp_path / "ckpt_tmpl"
+    d.mkdir()
+    shutil.copy(tok_file, d / "tokenizer.json")
+    (d / "tokenizer_config.json").write_text(json.dumps({
+        "chat_template": (
+            "{% for m in messages %}<|{{ m.role }}|>{{ m.content }}"
+            "{{ eos_token }}{% endfor %}"
+            "{% if add_generation_prompt %}<|assistant|>{% endif %}"
+        ),
+        "eos_token": "<|endoftext|>",
+    }))
+    t = get_tokenizer(1024, model_path=str(d))
+    out = t.apply_chat_template(
+        [{"role": "user", "content": "hi"},
+         {"role": "assistant", "content": "yo"}]
+    )
+    assert out == ("<|user|>hi<|endoftext|><|assistant|>yo<|endoftext|>"
+                   "<|assistant|>")
+    # eos from tokenizer_config resolves to a real id
+    assert t.eos_token_id is not None
+
+    # no template -> fallback
+    t2 = ByteTokenizer(1024)
+    assert t2.apply_chat_template([{"role": "user", "content": "x"}]) \
+        == "user: x"
+
+    # template errors surface as ValueError (mapped to HTTP 400)
+    t.chat_template = "{{ raise_exception('nope') }}"
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        t.apply_chat_template([])
