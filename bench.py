@@ -175,7 +175,9 @@ def main():
         from fusioninfer_amd.engine.spec_decode import SpeculativeConfig
 
         cfg.speculative = SpeculativeConfig(
-            num_speculative_tokens=args.spec_tokens
+            num_speculative_tokens=args.spec_tokens,
+            disable_by_batch_size=0,  # the bench measures spec as asked;
+                                      # pick the regime via --concurrency
         )
     # TP: ranks form DP groups of size tp; only group leaders drive load.
     driver_group = None

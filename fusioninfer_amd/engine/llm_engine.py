@@ -312,7 +312,9 @@ class LLMEngine:
         self.num_preemptions += len(batch.preempted)
         spec_drafts = None
         if self.proposer is not None and batch.decode_seqs:
-            spec_drafts = self._propose_drafts(batch.decode_seqs)
+            cap = self.proposer.cfg.disable_by_batch_size
+            if not cap or len(batch.decode_seqs) <= cap:
+                spec_drafts = self._propose_drafts(batch.decode_seqs)
         if spec_drafts is not None:
             payload = self.runner.build_spec_payload(
                 batch.prefill_seqs, batch.prefill_chunks, batch.decode_seqs,

@@ -43,6 +43,13 @@ class SpeculativeConfig:
     num_speculative_tokens: int = 4
     prompt_lookup_max: int = 4   # longest trailing n-gram to match
     prompt_lookup_min: int = 2   # shortest n-gram worth trusting
+    # vLLM speculative_disable_by_batch_size: above this many concurrent
+    # decode sequences, skip drafting — a spec step runs the WHOLE decode
+    # batch through the eager multi-token verify path (no hipGraph, no
+    # decode kernel), which loses at high batch (measured: 21.4 vs 44
+    # req/s at concurrency 256 with accept_rate 0.22). Speculation is a
+    # LATENCY feature for small batches. 0 disables the guard.
+    disable_by_batch_size: int = 32
 
 
 def seq_is_draftable(seq: Sequence) -> bool:

@@ -497,8 +497,26 @@ def schema_to_regex(schema: dict, depth: int = 0) -> str:
             for v in schema["enum"]
         )
         return f"({opts})"
+    if "const" in schema:
+        return _re_escape(json.dumps(schema["const"], separators=(",", ":")))
+    for alt_key in ("anyOf", "oneOf"):
+        if alt_key in schema:
+            opts = "|".join(
+                schema_to_regex(s, depth + 1) for s in schema[alt_key]
+            )
+            return f"({opts})"
     t = schema.get("type")
+    if isinstance(t, list):  # {"type": ["string", "null"]}
+        opts = "|".join(
+            schema_to_regex({**schema, "type": tt}, depth + 1) for tt in t
+        )
+        return f"({opts})"
     if t == "string":
+        pattern = schema.get("pattern")
+        if pattern:
+            body = pattern.lstrip("^").rstrip("$")
+            parse_regex(body)  # unsupported syntax raises -> fallback
+            return f'"({body})"'
         return _STRING_RE
     if t == "integer":
         return _INT_RE
@@ -511,9 +529,23 @@ def schema_to_regex(schema: dict, depth: int = 0) -> str:
     if t == "array":
         item = schema_to_regex(schema.get("items", {"type": "string"}),
                                depth + 1)
-        return f"\\[({item}(,{item})*)?\\]"
+        lo = int(schema.get("minItems", 0))
+        hi = schema.get("maxItems")
+        if lo == 0 and hi is None:
+            return f"\\[({item}(,{item})*)?\\]"
+        if hi is not None and (int(hi) < lo or int(hi) > 64):
+            raise ValueError("bad minItems/maxItems")
+        tail = (f"(,{item}){{{max(lo - 1, 0)},{int(hi) - 1}}}"
+                if hi is not None else f"(,{item}){{{max(lo - 1, 0)},}}")
+        body = f"{item}{tail}"
+        return f"\\[{body}\\]" if lo > 0 else f"\\[({body})?\\]"
     if t == "object":
         props = schema.get("properties") or {}
+        required = schema.get("required")
+        if required:
+            # optional properties would need order-permutation handling;
+            # emit the required subset in property order (documented)
+            props = {k: v for k, v in props.items() if k in set(required)}
         if not props:
             raise ValueError("object schema needs properties")
         parts = []

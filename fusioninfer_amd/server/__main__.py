@@ -96,6 +96,9 @@ def build_engine_config(args):
             num_speculative_tokens=int(raw.get("num_speculative_tokens", 4)),
             prompt_lookup_max=int(raw.get("prompt_lookup_max", 4)),
             prompt_lookup_min=int(raw.get("prompt_lookup_min", 2)),
+            disable_by_batch_size=int(
+                raw.get("disable_by_batch_size", 32)
+            ),
         )
     return EngineConfig(
         model=mc,

@@ -262,3 +262,40 @@ def test_engine_guided_regex_bounded():
     assert _re.fullmatch(r"[ab]{3}-\d{2}", text), text
     # grammar termination ended the request well before max_tokens
     assert len(outs[0].output_token_ids) == 6
+
+
+def test_schema_extensions():
+    """anyOf / const / type lists / string pattern / min-maxItems /
+    required-subset — the wider vLLM guided_json surface."""
+    g = RegexGrammar(schema_to_regex({
+        "anyOf": [{"type": "integer"}, {"type": "boolean"}],
+    }))
+    assert accepts(g, "42") and accepts(g, "true") and not accepts(g, '"x"')
+
+    g = RegexGrammar(schema_to_regex({"type": ["string", "null"]}))
+    assert accepts(g, '"hi"') and accepts(g, "null")
+
+    g = RegexGrammar(schema_to_regex({"const": "red"}))
+    assert accepts(g, '"red"') and not accepts(g, '"blue"')
+
+    g = RegexGrammar(schema_to_regex({
+        "type": "string", "pattern": "^[a-f]{2}-[0-9]+$",
+    }))
+    assert accepts(g, '"ab-42"') and not accepts(g, '"zz-1"')
+
+    g = RegexGrammar(schema_to_regex({
+        "type": "array", "items": {"type": "integer"},
+        "minItems": 2, "maxItems": 3,
+    }))
+    assert not accepts(g, "[1]")
+    assert accepts(g, "[1,2]") and accepts(g, "[1,2,3]")
+    assert not accepts(g, "[1,2,3,4]")
+
+    g = RegexGrammar(schema_to_regex({
+        "type": "object",
+        "properties": {"a": {"type": "integer"}, "b": {"type": "boolean"},
+                       "c": {"type": "string"}},
+        "required": ["a", "c"],
+    }))
+    assert accepts(g, '{"a":1,"c":"x"}')
+    assert not accepts(g, '{"a":1,"b":true,"c":"x"}')

@@ -188,3 +188,22 @@ def test_spec_under_tiny_block_pool():
     outs = eng.generate(prompts, SamplingParams(max_tokens=8))
     for o, exp in zip(outs, expected):
         assert o.output_token_ids == exp
+
+
+def test_disable_by_batch_size():
+    """vLLM speculative_disable_by_batch_size parity: above the cap the
+    step takes the plain decode path (no drafts)."""
+    torch.manual_seed(0)
+    spec = make_engine(
+        SpeculativeConfig(num_speculative_tokens=4, disable_by_batch_size=1)
+    )
+    outs = spec.generate(_repetitive_prompts()[:2],
+                         SamplingParams(max_tokens=12))
+    assert all(len(o.output_token_ids) == 12 for o in outs)
+    assert spec.num_spec_draft_tokens == 0  # 2 seqs > cap of 1
+
+    base = make_engine()
+    ref = [o.output_token_ids
+           for o in base.generate(_repetitive_prompts()[:2],
+                                  SamplingParams(max_tokens=12))]
+    assert [o.output_token_ids for o in outs] == ref
