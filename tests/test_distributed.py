@@ -593,3 +593,63 @@ def test_tp2_oneshot_allreduce_matches_ring():
             p.join(timeout=120)
             assert p.exitcode == 0, f"rank exited {p.exitcode}"
         assert results[0] == "ok" and results[1] == "ok"
+
+
+# --------------------------------------- speculative decoding under PP=2
+def _pp_spec_worker(rank, port, results):
+    from fusioninfer_amd.distributed import parallel_state as ps
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from fusioninfer_amd.engine.spec_decode import SpeculativeConfig
+
+    prompts = [[3, 1, 4, 1, 5, 9] * 5, [2, 7, 2, 7] * 8]
+
+    ps.ensure_single_process()
+    torch.manual_seed(13)
+    eng1 = LLMEngine(_engine_cfg(), device="cpu")
+    ref = [o.output_token_ids
+           for o in eng1.generate(prompts,
+                                  SamplingParams(max_tokens=12,
+                                                 temperature=0.0))]
+    ps.destroy()
+
+    _init(rank, 2, port)
+    try:
+        torch.manual_seed(13)
+        cfg = _engine_cfg()
+        cfg.parallel.pipeline_parallel_size = 2
+        cfg.parallel.distributed_backend = "gloo"
+        cfg.speculative = SpeculativeConfig(num_speculative_tokens=4)
+        engine = LLMEngine(cfg, device="cpu")
+        if engine.is_driver:
+            outs = engine.generate(
+                prompts, SamplingParams(max_tokens=12, temperature=0.0)
+            )
+            assert [o.output_token_ids for o in outs] == ref
+            assert engine.num_spec_draft_tokens > 0
+            engine.stop_workers()
+        else:
+            engine.worker_loop()
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+        ps.destroy()
+
+
+def test_pp2_spec_decode_matches_single_process():
+    """Spec verify payloads flow through the PP stages (and, when large
+    enough, the microbatch splitter rebases logits_rows): token-exact."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29643
+        procs = [
+            ctx.Process(target=_pp_spec_worker, args=(r, port, results))
+            for r in range(2)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+            assert p.exitcode == 0, f"rank exited {p.exitcode}"
+        assert results[0] == "ok" and results[1] == "ok"
