@@ -305,12 +305,30 @@ def build_app(serving: ServingEngine, model_name: str,
         created = int(time.time())
         cid = f"cmpl-{uuid.uuid4().hex[:16]}"
 
+        echo = bool(body.get("echo"))
+        include_usage = bool(
+            (body.get("stream_options") or {}).get("include_usage")
+        )
+
         if body.get("stream"):
             async def sse():
                 done = False
+                n_out = 0
                 filt = StopStreamFilter(stops) if stops else None
                 try:
+                    if echo:
+                        first = {
+                            "id": cid, "object": "text_completion",
+                            "created": created, "model": model_name,
+                            "choices": [{"index": 0,
+                                         "text": tokenizer.decode(prompt_ids),
+                                         "token_ids": [],
+                                         "finish_reason": None}],
+                        }
+                        yield f"data: {json.dumps(first)}\n\n"
                     async for tok, finished in _stream(q):
+                        if tok is not None:
+                            n_out += 1
                         done = finished
                         delta = tokenizer.decode_one(tok) if tok is not None else ""
                         if filt is not None and tok is not None:
@@ -340,6 +358,18 @@ def build_app(serving: ServingEngine, model_name: str,
                             done = True
                             serving.abort(req_id)
                             break
+                    if include_usage:
+                        usage = {
+                            "id": cid, "object": "text_completion",
+                            "created": created, "model": model_name,
+                            "choices": [],
+                            "usage": {
+                                "prompt_tokens": len(prompt_ids),
+                                "completion_tokens": n_out,
+                                "total_tokens": len(prompt_ids) + n_out,
+                            },
+                        }
+                        yield f"data: {json.dumps(usage)}\n\n"
                     yield "data: [DONE]\n\n"
                 finally:
                     if not done:  # client disconnected mid-stream
@@ -362,6 +392,9 @@ def build_app(serving: ServingEngine, model_name: str,
         choices.append((toks, text))
         for rid_i, q_i in extra:
             choices.append(await _collect(q_i, stops=stops, req_id=rid_i))
+        if echo:
+            prefix = tokenizer.decode(prompt_ids)
+            choices = [(t, prefix + x) for t, x in choices]
         total_completion = sum(len(t) for t, _ in choices)
         return JSONResponse(
             {

@@ -357,3 +357,31 @@ def test_chat_stream_stop_strings(app):
             assert text == full[: full.find(stop)]
 
     asyncio.run(run())
+
+
+def test_echo_and_stream_usage(app):
+    async def run():
+        async with _client(app) as c:
+            # echo: response text starts with the prompt
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "hello", "max_tokens": 4,
+                      "temperature": 0.0, "echo": True},
+            )
+            assert r.json()["choices"][0]["text"].startswith("hello")
+
+            # stream_options.include_usage: final chunk carries usage
+            async with c.stream(
+                "POST", "/v1/completions",
+                json={"prompt": "hi", "max_tokens": 5, "temperature": 0.0,
+                      "stream": True,
+                      "stream_options": {"include_usage": True}},
+            ) as resp:
+                chunks = []
+                async for line in resp.aiter_lines():
+                    if line.startswith("data: ") and line != "data: [DONE]":
+                        chunks.append(json.loads(line[6:]))
+            assert chunks[-1]["usage"]["completion_tokens"] == 5
+            assert chunks[-1]["choices"] == []
+
+    asyncio.run(run())
