@@ -77,7 +77,14 @@ class LLMEngine:
         self.sampler = Sampler(cfg.seed, device)
         from fusioninfer_amd.engine.spec_decode import build_proposer
 
-        self.proposer = build_proposer(cfg.speculative)
+        if (cfg.speculative is not None
+                and getattr(cfg.speculative, "method", "ngram") != "ngram"):
+            assert (cfg.parallel.tensor_parallel_size
+                    * cfg.parallel.pipeline_parallel_size) == 1, (
+                "draft-model speculation runs single-process; use the "
+                "ngram method under TP/PP"
+            )
+        self.proposer = build_proposer(cfg.speculative, cfg, device)
         self.num_spec_draft_tokens = 0
         self.num_spec_accepted_tokens = 0
         self.runner.capture_decode_graphs()
@@ -168,6 +175,8 @@ class LLMEngine:
         seq = self.seqs.pop(request_id, None)
         if seq is None:
             return False
+        if self.proposer is not None:
+            self.proposer.release(seq)
         staging = self._swapped.pop(request_id, None)
         if staging is not None:
             self._swap_bytes -= staging.numel() * staging.element_size()
@@ -380,6 +389,8 @@ class LLMEngine:
 
     # ------------------------------------------------- sampling / emission
     def _finish_seq(self, seq: Sequence) -> None:
+        if self.proposer is not None:
+            self.proposer.release(seq)
         seq.finish_time = time.monotonic()
         self.scheduler.finish(seq)
         self.num_finished += 1
@@ -460,20 +471,20 @@ class LLMEngine:
 
     # ------------------------------------------------ speculative decoding
     def _propose_drafts(self, decode_seqs):
-        """n-gram drafts per decode sequence; a draft is dropped when the
-        block pool cannot cover its tail. None = nothing to speculate
-        (step falls back to the fast pure-decode / hipGraph path)."""
+        """Drafts per decode sequence (ngram lookup or the draft model);
+        a draft is dropped when the block pool cannot cover its tail.
+        None = nothing to speculate (step falls back to the fast
+        pure-decode / hipGraph path)."""
         bm = self.block_manager
-        drafts, any_d = [], False
-        for s in decode_seqs:
-            d = self.proposer.propose(s)
+        drafts = [list(d) for d in self.proposer.propose_all(decode_seqs)]
+        any_d = False
+        for s, d in zip(decode_seqs, drafts):
             if d:
                 last_pos = s.num_tokens - 1 + len(d)
                 if bm.extra_blocks_for(s, last_pos) > bm.num_free():
-                    d = []
+                    d.clear()
                 else:
                     bm.append_slots_upto(s, last_pos)
-            drafts.append(list(d))
             any_d = any_d or bool(d)
         return drafts if any_d else None
 
@@ -513,6 +524,10 @@ class LLMEngine:
                 self.num_spec_draft_tokens += k
                 self.num_spec_accepted_tokens += m
                 outputs.append(self._emit_tokens(s, d[:m] + [g[m]]))
+                if not s.is_finished():
+                    # draft-model proposers roll their KV back to the
+                    # accepted prefix (no-op for ngram)
+                    self.proposer.commit(s, m)
         return outputs
 
     # --------------------------------------------------------- TP workers

@@ -99,6 +99,8 @@ def build_engine_config(args):
             disable_by_batch_size=int(
                 raw.get("disable_by_batch_size", 32)
             ),
+            model=raw.get("model"),
+            draft_gpu_blocks=raw.get("draft_gpu_blocks"),
         )
     return EngineConfig(
         model=mc,

@@ -207,3 +207,95 @@ def test_disable_by_batch_size():
            for o in base.generate(_repetitive_prompts()[:2],
                                   SamplingParams(max_tokens=12))]
     assert [o.output_token_ids for o in outs] == ref
+
+
+# --------------------------------------------------- draft-model method
+def make_draft_engine(draft_model="tiny-qwen3", k=4, blocks=None, seed=0):
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=256),
+        scheduler=SchedulerConfig(
+            max_num_seqs=8, max_num_batched_tokens=1024, max_model_len=256
+        ),
+        speculative=SpeculativeConfig(
+            method="draft_model", model=draft_model,
+            num_speculative_tokens=k, draft_gpu_blocks=blocks,
+        ),
+        seed=seed,
+    )
+    return LLMEngine(cfg, device="cpu")
+
+
+def test_draft_model_same_arch_always_accepts():
+    """Draft == target (same registry config + seed => identical random
+    weights): every draft token matches the target argmax, so each spec
+    step emits k+1 tokens and acceptance is 100%."""
+    torch.manual_seed(0)
+    base = make_engine()
+    prompts = [[3, 9, 27, 4] * 4, [8, 8, 1] * 5]
+    expected = [
+        o.output_token_ids
+        for o in base.generate(prompts, SamplingParams(max_tokens=21))
+    ]
+    eng = make_draft_engine()
+    outs = eng.generate(prompts, SamplingParams(max_tokens=21))
+    for o, exp in zip(outs, expected):
+        assert o.output_token_ids == exp
+    assert eng.num_spec_draft_tokens > 0
+    assert eng.num_spec_accepted_tokens == eng.num_spec_draft_tokens
+    # draft KV state is released when sequences finish
+    assert eng.proposer._state == {}
+
+
+def test_draft_model_exact_with_mixed_requests():
+    """Sampled/logprobs sequences ride draft-less; greedy stays exact."""
+    torch.manual_seed(0)
+    base = make_engine(seed=5)
+    prompts = [[3, 9, 27, 4] * 4, [1, 2, 3] * 6]
+    a = base.add_request(prompts[0], SamplingParams(max_tokens=12))
+    b = base.add_request(prompts[1],
+                         SamplingParams(max_tokens=12, temperature=0.7,
+                                        seed=77))
+    done = {}
+    while base.has_unfinished():
+        for o in base.step():
+            if o.finished:
+                done[o.request_id] = o
+    eng = make_draft_engine(seed=5)
+    a2 = eng.add_request(prompts[0], SamplingParams(max_tokens=12))
+    b2 = eng.add_request(prompts[1],
+                         SamplingParams(max_tokens=12, temperature=0.7,
+                                        seed=77))
+    done2 = {}
+    while eng.has_unfinished():
+        for o in eng.step():
+            if o.finished:
+                done2[o.request_id] = o
+    assert done2[a2].output_token_ids == done[a].output_token_ids
+    assert done2[b2].output_token_ids == done[b].output_token_ids
+
+
+def test_draft_model_tiny_pool_degrades_gracefully():
+    """Draft block pool exhaustion drops drafts, never breaks output."""
+    torch.manual_seed(0)
+    base = make_engine()
+    prompts = [[5, 6, 7] * 10, [9, 8] * 12]
+    expected = [
+        o.output_token_ids
+        for o in base.generate(prompts, SamplingParams(max_tokens=10))
+    ]
+    eng = make_draft_engine(blocks=2)  # room for ~1 sequence's draft KV
+    outs = eng.generate(prompts, SamplingParams(max_tokens=10))
+    for o, exp in zip(outs, expected):
+        assert o.output_token_ids == exp
+
+
+def test_draft_model_abort_releases_state():
+    torch.manual_seed(0)
+    eng = make_draft_engine()
+    rid = eng.add_request([3, 9, 27, 4] * 4, SamplingParams(max_tokens=50))
+    for _ in range(4):
+        eng.step()
+    assert rid in eng.seqs
+    assert eng.abort_request(rid)
+    assert eng.proposer._state == {}
