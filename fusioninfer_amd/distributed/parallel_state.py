@@ -20,6 +20,7 @@ _TP_RANK = 0
 _TP_WORLD = 1
 _PP_RANK = 0
 _PP_WORLD = 1
+_RANK = 0
 _INITIALIZED = False
 
 
@@ -34,10 +35,13 @@ def init_distributed(
     Reads RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT from env (torchrun), or
     LWS_LEADER_ADDRESS when launched by the control plane's LWS wrapper.
 
-    Pipeline parallelism composes with TP=1 for now: the whole world is
-    one pipeline (rank == stage); activations move with p2p send/recv.
+    TP x PP composition: world = pp * tp, stage = rank // tp (TP groups
+    are stage-contiguous), and pipeline p2p moves column-wise — stage s
+    tp-rank r talks to stage s+1 tp-rank r. PP-only (tp=1) reduces to
+    rank == stage.
     """
-    global _TP_GROUP, _TP_RANK, _TP_WORLD, _PP_RANK, _PP_WORLD, _INITIALIZED
+    global _TP_GROUP, _TP_RANK, _TP_WORLD, _PP_RANK, _PP_WORLD, _RANK, \
+        _INITIALIZED
     if _INITIALIZED:
         return
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -63,11 +67,11 @@ def init_distributed(
                 _TP_GROUP = grp
     _TP_RANK = rank % tp
     _TP_WORLD = tp
+    _RANK = rank
     pp = pipeline_parallel_size
     if pp > 1:
-        assert tp == 1, "PP currently composes with TP=1"
-        assert world_size == pp, (world_size, pp)
-        _PP_RANK = rank
+        assert world_size == tp * pp, (world_size, tp, pp)
+        _PP_RANK = rank // tp
         _PP_WORLD = pp
     _INITIALIZED = True
 
@@ -92,11 +96,12 @@ def tp_group():
 
 
 def destroy() -> None:
-    global _TP_GROUP, _TP_RANK, _TP_WORLD, _PP_RANK, _PP_WORLD, _INITIALIZED
+    global _TP_GROUP, _TP_RANK, _TP_WORLD, _PP_RANK, _PP_WORLD, _RANK, \
+        _INITIALIZED
     if dist.is_initialized():
         dist.destroy_process_group()
     _TP_GROUP, _TP_RANK, _TP_WORLD, _INITIALIZED = None, 0, 1, False
-    _PP_RANK, _PP_WORLD = 0, 1
+    _PP_RANK, _PP_WORLD, _RANK = 0, 1, 0
 
 
 # --------------------------------------------------------------- pipeline
@@ -132,24 +137,30 @@ def _gloo_safe_recv(t: torch.Tensor, src: int) -> None:
         dist.recv(t, src)
 
 
+def _stage_peer(stage: int) -> int:
+    """Global rank of `stage`'s member in THIS rank's TP column."""
+    return stage * _TP_WORLD + _TP_RANK
+
+
 def pp_send_next(t: torch.Tensor) -> None:
-    """Ship activations to the next pipeline stage (xGMI p2p under RCCL)."""
-    _gloo_safe_send(t, _PP_RANK + 1)
+    """Ship activations to the next pipeline stage (xGMI p2p under RCCL);
+    column-wise under TP x PP (same tp-rank in the next stage)."""
+    _gloo_safe_send(t, _RANK + _TP_WORLD)
 
 
 def pp_recv_prev(shape, dtype, device) -> torch.Tensor:
     t = torch.empty(shape, dtype=dtype, device=device)
-    _gloo_safe_recv(t, _PP_RANK - 1)
+    _gloo_safe_recv(t, _RANK - _TP_WORLD)
     return t
 
 
-def pp_send_to(t: torch.Tensor, dst: int) -> None:
-    _gloo_safe_send(t, dst)
+def pp_send_to(t: torch.Tensor, stage: int) -> None:
+    _gloo_safe_send(t, _stage_peer(stage))
 
 
-def pp_recv_from(shape, dtype, device, src: int) -> torch.Tensor:
+def pp_recv_from(shape, dtype, device, stage: int) -> torch.Tensor:
     t = torch.empty(shape, dtype=dtype, device=device)
-    _gloo_safe_recv(t, src)
+    _gloo_safe_recv(t, _stage_peer(stage))
     return t
 
 
@@ -187,20 +198,20 @@ def _gloo_safe_isend(t: torch.Tensor, dst: int) -> _PendingSend:
 
 
 def pp_isend_next(t: torch.Tensor) -> _PendingSend:
-    return _gloo_safe_isend(t, _PP_RANK + 1)
+    return _gloo_safe_isend(t, _RANK + _TP_WORLD)
 
 
-def pp_isend_to(t: torch.Tensor, dst: int) -> _PendingSend:
-    return _gloo_safe_isend(t, dst)
+def pp_isend_to(t: torch.Tensor, stage: int) -> _PendingSend:
+    return _gloo_safe_isend(t, _stage_peer(stage))
 
 
-def pp_irecv_from(shape, dtype, device, src: int) -> _PendingRecv:
+def pp_irecv_from(shape, dtype, device, stage: int) -> _PendingRecv:
     """Post a receive NOW (pre-posting on the driver is what lets pipeline
     microbatches drain without a send/recv deadlock) and collect later."""
     bf16_cpu = dtype == torch.bfloat16 and torch.device(device).type == "cpu"
     buf = torch.empty(shape, dtype=torch.int16 if bf16_cpu else dtype,
                       device=device)
-    return _PendingRecv(dist.irecv(buf, src), buf, bf16_cpu)
+    return _PendingRecv(dist.irecv(buf, _stage_peer(stage)), buf, bf16_cpu)
 
 
 def tp_broadcast_object(obj=None):

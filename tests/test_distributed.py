@@ -653,3 +653,65 @@ def test_pp2_spec_decode_matches_single_process():
             p.join(timeout=300)
             assert p.exitcode == 0, f"rank exited {p.exitcode}"
         assert results[0] == "ok" and results[1] == "ok"
+
+
+# ------------------------------------------------ TP x PP composition
+def _tp_pp_worker(rank, port, results):
+    from fusioninfer_amd.distributed import parallel_state as ps
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+
+    prompts = [PROMPT, [9, 9, 2] * 8]
+
+    ps.ensure_single_process()
+    torch.manual_seed(31)
+    eng1 = LLMEngine(_engine_cfg(), device="cpu")
+    ref = [o.output_token_ids
+           for o in eng1.generate(prompts,
+                                  SamplingParams(max_tokens=6,
+                                                 temperature=0.0))]
+    ps.destroy()
+
+    _init(rank, 4, port)
+    try:
+        torch.manual_seed(31)
+        cfg = _engine_cfg()
+        cfg.parallel.tensor_parallel_size = 2
+        cfg.parallel.pipeline_parallel_size = 2
+        cfg.parallel.distributed_backend = "gloo"
+        engine = LLMEngine(cfg, device="cpu")
+        # stage = rank // tp; each stage holds half the layers, each tp
+        # rank half the heads
+        assert engine.runner.model.num_local_layers * 2 == \
+            cfg.model.num_layers
+        if engine.is_driver:
+            outs = engine.generate(
+                prompts, SamplingParams(max_tokens=6, temperature=0.0)
+            )
+            assert [o.output_token_ids for o in outs] == ref
+            engine.stop_workers()
+        else:
+            engine.worker_loop()
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+        ps.destroy()
+
+
+def test_tp2_pp2_engine_matches_single_process():
+    """TP x PP (world 4, 2 stages x 2 shards over gloo): stage-contiguous
+    TP groups, column-wise activation p2p — token-exact vs 1 process."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29647
+        procs = [
+            ctx.Process(target=_tp_pp_worker, args=(r, port, results))
+            for r in range(4)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+            assert p.exitcode == 0, f"rank exited {p.exitcode}"
+        assert all(results[r] == "ok" for r in range(4))
