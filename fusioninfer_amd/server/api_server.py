@@ -205,10 +205,11 @@ def build_app(serving: ServingEngine, model_name: str,
             return []
         return [stops] if isinstance(stops, str) else list(stops)
 
-    async def _collect(q, stops=None, req_id=None):
+    async def _collect(q, stops=None, req_id=None, keep_stop=False):
         """Drain a request's token stream. With OpenAI `stop` strings the
-        generated text is truncated BEFORE the first stop match and the
-        engine request is aborted. Returns (token_ids, text)."""
+        generated text is truncated BEFORE the first stop match (AFTER it
+        with include_stop_str_in_output) and the engine request is
+        aborted. Returns (token_ids, text)."""
         loop = asyncio.get_event_loop()
         toks: List[int] = []
         text = ""
@@ -222,7 +223,8 @@ def build_app(serving: ServingEngine, model_name: str,
                         i = text.find(s)
                         if i >= 0:
                             serving.abort(req_id)
-                            return toks, text[:i]
+                            end = i + len(s) if keep_stop else i
+                            return toks, text[:end]
             if finished:
                 return toks, text if stops else tokenizer.decode(toks)
 
@@ -274,6 +276,9 @@ def build_app(serving: ServingEngine, model_name: str,
         body = await request.json()
         prompt = body.get("prompt", "")
         prompt_ids = _encode(prompt)
+        tpt = body.get("truncate_prompt_tokens")
+        if tpt:
+            prompt_ids = prompt_ids[-int(tpt):]
         try:
             sampling = _sampling_from(body)
         except ValueError as e:
@@ -387,11 +392,14 @@ def build_app(serving: ServingEngine, model_name: str,
             if s_i.seed is not None:
                 s_i.seed += i  # distinct choices under a fixed seed
             extra.append(serving.submit(prompt_ids, s_i, lora_name=lora))
+        keep_stop = bool(body.get("include_stop_str_in_output"))
         choices = []
-        toks, text = await _collect(q, stops=stops, req_id=req_id)
+        toks, text = await _collect(q, stops=stops, req_id=req_id,
+                                    keep_stop=keep_stop)
         choices.append((toks, text))
         for rid_i, q_i in extra:
-            choices.append(await _collect(q_i, stops=stops, req_id=rid_i))
+            choices.append(await _collect(q_i, stops=stops, req_id=rid_i,
+                                          keep_stop=keep_stop))
         if echo:
             prefix = tokenizer.decode(prompt_ids)
             choices = [(t, prefix + x) for t, x in choices]
@@ -502,7 +510,10 @@ def build_app(serving: ServingEngine, model_name: str,
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
-        toks, out_text = await _collect(q, stops=stops, req_id=req_id)
+        toks, out_text = await _collect(
+            q, stops=stops, req_id=req_id,
+            keep_stop=bool(body.get("include_stop_str_in_output")),
+        )
         return JSONResponse(
             {
                 "id": cid,

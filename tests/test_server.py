@@ -385,3 +385,35 @@ def test_echo_and_stream_usage(app):
             assert chunks[-1]["choices"] == []
 
     asyncio.run(run())
+
+
+def test_truncate_prompt_and_include_stop(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": [7] * 100, "max_tokens": 2,
+                      "truncate_prompt_tokens": 10, "temperature": 0.0,
+                      "ignore_eos": True},
+            )
+            assert r.json()["usage"]["prompt_tokens"] == 10
+
+            # include_stop_str_in_output keeps the matched stop string
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "mnomno", "max_tokens": 8,
+                      "temperature": 0, "ignore_eos": True},
+            )
+            full = r.json()["choices"][0]["text"]
+            stop_ch = full[1]
+            r2 = await c.post(
+                "/v1/completions",
+                json={"prompt": "mnomno", "max_tokens": 8, "stop": stop_ch,
+                      "include_stop_str_in_output": True,
+                      "temperature": 0, "ignore_eos": True},
+            )
+            text = r2.json()["choices"][0]["text"]
+            assert text.endswith(stop_ch)
+            assert text == full.split(stop_ch)[0] + stop_ch
+
+    asyncio.run(run())
