@@ -34,6 +34,7 @@ class RequestOutput:
         )
         self.finished = seq.is_finished()
         self.logprobs = list(seq.logprobs)
+        self.prompt_logprobs = list(seq.prompt_logprobs)
         self.ttft = seq.ttft
         self.arrival_time = seq.arrival_time
         self.finish_time = seq.finish_time
@@ -319,11 +320,18 @@ class LLMEngine:
         if batch.is_empty:
             return []
         self.num_preemptions += len(batch.preempted)
+        want_plp = any(
+            s.sampling.prompt_logprobs is not None for s in batch.prefill_seqs
+        )
         spec_drafts = None
-        if self.proposer is not None and batch.decode_seqs:
+        if self.proposer is not None and batch.decode_seqs and not want_plp:
             cap = self.proposer.cfg.disable_by_batch_size
             if not cap or len(batch.decode_seqs) <= cap:
                 spec_drafts = self._propose_drafts(batch.decode_seqs)
+        chunk_starts = [
+            s.num_computed_tokens or s.num_cached_tokens
+            for s in batch.prefill_seqs
+        ]
         if spec_drafts is not None:
             payload = self.runner.build_spec_payload(
                 batch.prefill_seqs, batch.prefill_chunks, batch.decode_seqs,
@@ -334,6 +342,17 @@ class LLMEngine:
                 batch.prefill_seqs, batch.prefill_chunks, batch.decode_seqs,
                 self.block_manager,
             )
+            if want_plp:
+                rows = set()
+                cu = payload["cu"]
+                for j, s in enumerate(batch.prefill_seqs):
+                    if s.sampling.prompt_logprobs is not None:
+                        rows.update(range(cu[j], cu[j + 1]))
+                    if payload["sample"][j]:
+                        rows.add(cu[j + 1] - 1)
+                nd = len(payload["decode"]["ids"])
+                rows.update(range(cu[-1], cu[-1] + nd))
+                payload["logits_rows"] = sorted(rows)
         t_payload = time.monotonic()
         if tp > 1:
             self._ps.tp_broadcast_object(payload)
@@ -369,9 +388,14 @@ class LLMEngine:
                 s for s, smp in zip(batch.prefill_seqs, payload["sample"])
                 if smp
             ] + batch.decode_seqs
+            logits_f = logits.float()
+            if payload.get("logits_rows") is not None:
+                logits_f = self._consume_prompt_logprobs(
+                    batch, payload, logits_f, chunk_starts
+                )
             if not sample_seqs:
                 return []
-            outputs = self._sample_and_emit(sample_seqs, logits.float())
+            outputs = self._sample_and_emit(sample_seqs, logits_f)
         if self._timing and hasattr(self, "_t_fwd_mark"):
             kind = "P" if batch.prefill_seqs else "D"
             self._tacc[kind][3] += time.monotonic() - self._t_fwd_mark
@@ -468,6 +492,57 @@ class LLMEngine:
             for seq, tok in zip(sample_seqs, next_tokens)
         )
         return outputs
+
+    def _consume_prompt_logprobs(self, batch, payload, logits_f,
+                                 chunk_starts):
+        """prompt_logprobs (vLLM parity): the payload carried explicit
+        logits_rows covering every prompt position of requesting
+        sequences; fill seq.prompt_logprobs (index 0 and prefix-cached
+        positions are None) and return the standard sampling-row subset
+        in the engine's expected order. log_softmax runs in row slices
+        so a full 8k-token admission never materializes a [T, V] fp32."""
+        rows = payload["logits_rows"]
+        pos_of = {r: i for i, r in enumerate(rows)}
+        cu = payload["cu"]
+        for j, seq in enumerate(batch.prefill_seqs):
+            k = seq.sampling.prompt_logprobs
+            if k is None:
+                continue
+            C = chunk_starts[j]
+            n = batch.prefill_chunks[j]
+            if not seq.prompt_logprobs:
+                # position 0 has no logprob; prefix-cached positions are
+                # skipped (their logits were never computed)
+                seq.prompt_logprobs.extend(
+                    [None] * min(C + 1, seq.num_prompt_tokens)
+                )
+            toks = seq.all_token_ids
+            idx = [pos_of[r] for r in range(cu[j], cu[j + 1])]
+            for lo in range(0, n, 256):
+                hi = min(lo + 256, n)
+                sub = torch.log_softmax(logits_f[idx[lo:hi]], dim=-1)
+                for i in range(lo, hi):
+                    pos_next = C + i + 1
+                    if pos_next >= seq.num_prompt_tokens:
+                        break
+                    row = sub[i - lo]
+                    entry_top = {}
+                    if k > 0:
+                        vals, tix = row.topk(k)
+                        entry_top = {
+                            int(t): float(v) for t, v in zip(tix, vals)
+                        }
+                    seq.prompt_logprobs.append(
+                        (float(row[toks[pos_next]]), entry_top)
+                    )
+        pf_rows = [
+            cu[j + 1] - 1 for j, smp in enumerate(payload["sample"]) if smp
+        ]
+        nd = len(payload["decode"]["ids"])
+        sel = [pos_of[r] for r in pf_rows] + [
+            pos_of[cu[-1] + i] for i in range(nd)
+        ]
+        return logits_f[sel]
 
     # ------------------------------------------------ speculative decoding
     def _propose_drafts(self, decode_seqs):

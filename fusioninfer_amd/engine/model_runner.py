@@ -553,13 +553,22 @@ class ModelRunner:
         ids = torch.tensor(
             payload["ids"] + d["ids"], dtype=torch.long, device=dev
         )
-        # logits rows: completing prefill chunks first, then every decode row
-        pf_idx = [
-            c - 1 for c, smp in zip(cu[1:], payload["sample"]) if smp
-        ]
-        logits_idx = torch.tensor(
-            pf_idx + list(range(np_, np_ + nd)), dtype=torch.long, device=dev
-        )
+        if payload.get("logits_rows") is not None:
+            # prompt_logprobs asks for explicit rows (prompt positions
+            # plus the standard sampling rows); the engine re-maps them
+            logits_idx = torch.tensor(
+                payload["logits_rows"], dtype=torch.long, device=dev
+            )
+        else:
+            # logits rows: completing prefill chunks first, then every
+            # decode row
+            pf_idx = [
+                c - 1 for c, smp in zip(cu[1:], payload["sample"]) if smp
+            ]
+            logits_idx = torch.tensor(
+                pf_idx + list(range(np_, np_ + nd)), dtype=torch.long,
+                device=dev,
+            )
         with torch.no_grad():
             return self._forward_and_logits(ids, meta, logits_idx)
 

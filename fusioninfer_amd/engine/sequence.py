@@ -28,6 +28,7 @@ class SamplingParams:
         frequency_penalty: float = 0.0,
         seed: Optional[int] = None,
         logprobs: Optional[int] = None,
+        prompt_logprobs: Optional[int] = None,
         ignore_eos: bool = True,
         stop_token_ids: Optional[List[int]] = None,
         guided=None,
@@ -43,6 +44,7 @@ class SamplingParams:
         self.frequency_penalty = frequency_penalty
         self.seed = seed
         self.logprobs = logprobs
+        self.prompt_logprobs = prompt_logprobs
         self.ignore_eos = ignore_eos
         self.stop_token_ids = stop_token_ids or []
         # guided decoding: a guided.GuidedDecoder enforcing a grammar via
@@ -80,6 +82,9 @@ class Sequence:
         # per-output-token logprob entries when sampling.logprobs is set:
         # [(logprob_of_sampled, {token_id: logprob, ...top-k}), ...]
         self.logprobs = []
+        # per-PROMPT-position entries when sampling.prompt_logprobs is
+        # set: index 0 and prefix-cached positions are None
+        self.prompt_logprobs = []
 
     @property
     def num_prompt_tokens(self) -> int:

@@ -193,6 +193,7 @@ def build_app(serving: ServingEngine, model_name: str,
             frequency_penalty=float(body.get("frequency_penalty", 0.0)),
             seed=body.get("seed"),
             logprobs=body.get("logprobs"),
+            prompt_logprobs=body.get("prompt_logprobs"),
             ignore_eos=bool(body.get("ignore_eos", False)),
             stop_token_ids=stop_ids,
             guided=_guided_from(body),
@@ -396,6 +397,7 @@ def build_app(serving: ServingEngine, model_name: str,
         choices = []
         toks, text = await _collect(q, stops=stops, req_id=req_id,
                                     keep_stop=keep_stop)
+        final = serving.take_final(req_id)
         choices.append((toks, text))
         for rid_i, q_i in extra:
             choices.append(await _collect(q_i, stops=stops, req_id=rid_i,
@@ -416,6 +418,12 @@ def build_app(serving: ServingEngine, model_name: str,
                         "text": c_text,
                         "token_ids": c_toks,
                         "finish_reason": "stop",
+                        **({"logprobs": final.logprobs,
+                            "prompt_logprobs": final.prompt_logprobs}
+                           if i == 0 and final is not None
+                           and (body.get("logprobs")
+                                or body.get("prompt_logprobs"))
+                           else {}),
                     }
                     for i, (c_toks, c_text) in enumerate(choices)
                 ],

@@ -27,6 +27,9 @@ class ServingEngine:
     ):
         self.engine = LLMEngine(cfg, device=device)
         self._streams: Dict[str, "queue.Queue[Tuple[Optional[int], bool]]"] = {}
+        # finished RequestOutputs kept briefly so the HTTP layer can
+        # attach logprobs / prompt_logprobs after draining the stream
+        self._final: Dict[str, object] = {}
         self._lock = threading.Lock()
         self._work = threading.Event()
         self._stop = False
@@ -159,6 +162,11 @@ class ServingEngine:
             ),
         }
 
+    def take_final(self, request_id: str):
+        """Pop the finished RequestOutput (None if aborted mid-flight)."""
+        with self._lock:
+            return self._final.pop(request_id, None)
+
     def abort(self, request_id: str) -> bool:
         with self._lock:
             self._streams.pop(request_id, None)
@@ -201,3 +209,6 @@ class ServingEngine:
                 if out.finished:
                     with self._lock:
                         self._streams.pop(out.request_id, None)
+                        self._final[out.request_id] = out
+                        if len(self._final) > 4096:  # belt-and-braces cap
+                            self._final.pop(next(iter(self._final)))

@@ -417,3 +417,23 @@ def test_truncate_prompt_and_include_stop(app):
             assert text == full.split(stop_ch)[0] + stop_ch
 
     asyncio.run(run())
+
+
+def test_prompt_logprobs_api(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": [5, 6, 7, 8] * 4, "max_tokens": 2,
+                      "temperature": 0.0, "prompt_logprobs": 1,
+                      "logprobs": 1, "ignore_eos": True},
+            )
+            body = r.json()
+            ch = body["choices"][0]
+            assert len(ch["prompt_logprobs"]) == 16
+            assert ch["prompt_logprobs"][0] is None
+            assert all(isinstance(e[0], float)
+                       for e in ch["prompt_logprobs"][1:])
+            assert len(ch["logprobs"]) == 2  # one entry per output token
+
+    asyncio.run(run())
