@@ -78,6 +78,10 @@ class SchedulerConfig:
     # are waiting (or nothing is decoding), so most steps stay pure-decode
     # and take the hipGraph path; 0 = admit eagerly every step
     prefill_admission_tokens: int = 8192
+    # "fcfs" (default) or "priority" (vLLM --scheduling-policy): priority
+    # orders admission by (priority, arrival) and preempts the
+    # lowest-priority running sequence first (lower value = higher prio)
+    policy: str = "fcfs"
 
 
 @dataclasses.dataclass

@@ -101,6 +101,13 @@ class LLMEngine:
         self.num_prefilled_tokens = 0
         self.ttft_sum = 0.0
         self.e2e_latency_sum = 0.0
+        # Prometheus-style histogram counts (vLLM bucket boundaries)
+        self.ttft_buckets = [0.01, 0.025, 0.05, 0.1, 0.25, 0.5, 1.0,
+                             2.5, 5.0, 10.0]
+        self.e2e_buckets = [0.3, 0.5, 0.8, 1.0, 1.5, 2.0, 2.5, 5.0,
+                            10.0, 15.0, 30.0, 60.0]
+        self.ttft_hist = [0] * (len(self.ttft_buckets) + 1)
+        self.e2e_hist = [0] * (len(self.e2e_buckets) + 1)
         self.step_time_sum = 0.0
         self.num_steps = 0
 
@@ -151,11 +158,13 @@ class LLMEngine:
         sampling: Optional[SamplingParams] = None,
         request_id: Optional[str] = None,
         lora_name: Optional[str] = None,
+        priority: int = 0,
     ) -> str:
         if request_id is None:
             request_id = f"req-{next(self._req_counter)}"
         seq = Sequence(request_id, prompt_token_ids, sampling or SamplingParams())
         seq.lora_name = lora_name
+        seq.priority = priority
         if seq.num_prompt_tokens > self.cfg.scheduler.max_model_len:
             raise ValueError(
                 f"prompt length {seq.num_prompt_tokens} exceeds "
@@ -412,6 +421,14 @@ class LLMEngine:
         return outputs
 
     # ------------------------------------------------- sampling / emission
+    @staticmethod
+    def _observe(hist, buckets, value) -> None:
+        for i, b in enumerate(buckets):
+            if value <= b:
+                hist[i] += 1
+                return
+        hist[-1] += 1
+
     def _finish_seq(self, seq: Sequence) -> None:
         if self.proposer is not None:
             self.proposer.release(seq)
@@ -420,7 +437,10 @@ class LLMEngine:
         self.num_finished += 1
         if seq.ttft is not None:
             self.ttft_sum += seq.ttft
-        self.e2e_latency_sum += seq.finish_time - seq.arrival_time
+            self._observe(self.ttft_hist, self.ttft_buckets, seq.ttft)
+        e2e = seq.finish_time - seq.arrival_time
+        self.e2e_latency_sum += e2e
+        self._observe(self.e2e_hist, self.e2e_buckets, e2e)
         del self.seqs[seq.seq_id]
 
     def _emit_tokens(self, seq: Sequence, toks) -> RequestOutput:

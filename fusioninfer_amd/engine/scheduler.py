@@ -81,6 +81,10 @@ class Scheduler:
         sequences never stall behind prefills."""
         preempted: List[Sequence] = []
         budget = self.cfg.max_num_batched_tokens
+        if self.cfg.policy == "priority" and len(self.waiting) > 1:
+            self.waiting = deque(sorted(
+                self.waiting, key=lambda s: (s.priority, s.arrival_time)
+            ))
 
         # decode: every fully-prefilled running sequence, one token each
         decode: List[Sequence] = []
@@ -173,7 +177,13 @@ class Scheduler:
         return ScheduledBatch(prefill, chunks, decode, preempted, swap_in)
 
     def _preempt_newest(self) -> Sequence:
-        victim = self.running.pop()  # newest
+        if self.cfg.policy == "priority":
+            # lowest-priority (then newest) sequence yields first
+            victim = max(self.running,
+                         key=lambda s: (s.priority, s.arrival_time))
+            self.running.remove(victim)
+        else:
+            victim = self.running.pop()  # newest
         if self.swap_out_fn is not None and self.swap_out_fn(victim):
             # KV parked in CPU swap: keep all computed state; resume
             # restores the blocks instead of re-prefilling

@@ -79,6 +79,8 @@ class Sequence:
         self.swap_num_blocks = 0
         # LoRA adapter name (None = base model)
         self.lora_name = None
+        # scheduling priority (lower = more urgent; "priority" policy)
+        self.priority = 0
         # per-output-token logprob entries when sampling.logprobs is set:
         # [(logprob_of_sampled, {token_id: logprob, ...top-k}), ...]
         self.logprobs = []

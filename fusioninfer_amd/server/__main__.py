@@ -49,6 +49,8 @@ def parse_args(argv=None):
     p.add_argument("--lora-modules", nargs="*", default=[],
                    help="adapters to register at startup: name[=rank]")
     p.add_argument("--enable-prefix-caching", action="store_true")
+    p.add_argument("--scheduling-policy", choices=["fcfs", "priority"],
+                   default="fcfs")
     p.add_argument("--speculative-config", type=str, default=None,
                    help='JSON: {"method": "ngram", '
                         '"num_speculative_tokens": 4, '
@@ -114,6 +116,7 @@ def build_engine_config(args):
             max_num_seqs=args.max_num_seqs,
             max_num_batched_tokens=args.max_num_batched_tokens,
             max_model_len=args.max_model_len,
+            policy=args.scheduling_policy,
         ),
         parallel=ParallelConfig(
             tensor_parallel_size=args.tensor_parallel_size,

@@ -139,6 +139,27 @@ def build_app(serving: ServingEngine, model_name: str,
             lines.append(
                 f'vllm:{name}{{model_name="{model_name}"}} {m[name]}'
             )
+        # vLLM-style histograms (cumulative buckets + +Inf)
+        e = serving.engine
+        for name, buckets, hist, total in [
+            ("time_to_first_token_seconds", e.ttft_buckets, e.ttft_hist,
+             e.ttft_sum),
+            ("e2e_request_latency_seconds", e.e2e_buckets, e.e2e_hist,
+             e.e2e_latency_sum),
+        ]:
+            lines.append(f"# TYPE vllm:{name} histogram")
+            cum = 0
+            for b, c in zip(buckets, hist):
+                cum += c
+                lines.append(
+                    f'vllm:{name}_bucket{{model_name="{model_name}",'
+                    f'le="{b}"}} {cum}'
+                )
+            cum += hist[-1]
+            lines.append(
+                f'vllm:{name}_bucket{{model_name="{model_name}",'
+                f'le="+Inf"}} {cum}'
+            )
         return PlainTextResponse("\n".join(lines) + "\n")
 
     _vocab_cache: Dict[str, Any] = {}
@@ -301,7 +322,10 @@ def build_app(serving: ServingEngine, model_name: str,
                     None, serving.submit_imported, int(pd_tag), sampling
                 )
             else:
-                req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
+                req_id, q = serving.submit(
+                    prompt_ids, sampling, lora_name=lora,
+                    priority=int(body.get("priority", 0)),
+                )
         except ValueError as e:
             return JSONResponse({"error": {"message": str(e),
                                            "type": "invalid_request_error"}}, 400)

@@ -49,12 +49,13 @@ class ServingEngine:
     # ------------------------------------------------------------- public
     def submit(
         self, prompt_token_ids: List[int], sampling: SamplingParams,
-        lora_name=None,
+        lora_name=None, priority: int = 0,
     ) -> Tuple[str, "queue.Queue[Tuple[Optional[int], bool]]"]:
         q: "queue.Queue[Tuple[Optional[int], bool]]" = queue.Queue()
         with self._lock:
             req_id = self.engine.add_request(
-                prompt_token_ids, sampling, lora_name=lora_name
+                prompt_token_ids, sampling, lora_name=lora_name,
+                priority=priority,
             )
             self._streams[req_id] = q
         self._work.set()

@@ -504,3 +504,55 @@ def test_prompt_logprobs_chunked_prefill():
     assert len(o1.prompt_logprobs) == len(o2.prompt_logprobs) == len(prompt)
     for a, b in zip(o1.prompt_logprobs[1:], o2.prompt_logprobs[1:]):
         assert abs(a[0] - b[0]) < 1e-3
+
+
+def test_priority_scheduling_admission_and_preemption():
+    """--scheduling-policy priority: lower value admits first and the
+    lowest-priority running sequence is preempted first."""
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=256),
+        scheduler=SchedulerConfig(
+            max_num_seqs=1, max_num_batched_tokens=1024, max_model_len=256,
+            policy="priority",
+        ),
+    )
+    torch.manual_seed(0)
+    eng = LLMEngine(cfg, device="cpu")
+    # low-priority arrives FIRST, high-priority second; with one slot the
+    # high-priority one must run first
+    lo = eng.add_request([5, 6, 7] * 4, SamplingParams(max_tokens=3),
+                         priority=10)
+    hi = eng.add_request([9, 8, 7] * 4, SamplingParams(max_tokens=3),
+                         priority=0)
+    order = []
+    while eng.has_unfinished():
+        for o in eng.step():
+            if o.finished:
+                order.append(o.request_id)
+    assert order == [hi, lo]
+
+    # preemption picks the lowest-priority victim (tiny pool forces it)
+    cfg2 = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=8),
+        scheduler=SchedulerConfig(
+            max_num_seqs=4, max_num_batched_tokens=1024, max_model_len=128,
+            policy="priority",
+        ),
+    )
+    torch.manual_seed(0)
+    eng2 = LLMEngine(cfg2, device="cpu")
+    a = eng2.add_request([4] * 40, SamplingParams(max_tokens=30), priority=0)
+    b = eng2.add_request([5] * 40, SamplingParams(max_tokens=30), priority=9)
+    done = []
+    steps = 0
+    while eng2.has_unfinished() and steps < 500:
+        steps += 1
+        for o in eng2.step():
+            if o.finished:
+                done.append(o.request_id)
+    assert set(done) == {a, b}
+    # the low-priority request was the preemption victim
+    assert eng2.num_preemptions > 0
+    assert done[0] == a
