@@ -28,6 +28,7 @@ class ModelConfig:
     rms_norm_eps: float = 1e-6
     max_position_embeddings: int = 40960
     qk_norm: bool = True              # Qwen3 per-head q/k RMSNorm
+    attention_bias: bool = False      # Qwen2.5-style qkv bias
     tie_word_embeddings: bool = False
     dtype: str = "bfloat16"
     # optional HF-layout safetensors checkpoint dir; None = random init

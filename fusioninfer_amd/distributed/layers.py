@@ -48,7 +48,8 @@ class MergedColumnParallelLinear(nn.Module):
     per-rank output layout is [seg0_shard | seg1_shard | ...] — slicing a
     merged matrix contiguously would give rank 0 all of segment 0."""
 
-    def __init__(self, in_features: int, out_sizes, dtype=torch.bfloat16):
+    def __init__(self, in_features: int, out_sizes, dtype=torch.bfloat16,
+                 bias: bool = False):
         super().__init__()
         tp = ps.tp_world_size()
         rank = ps.tp_rank()
@@ -63,9 +64,19 @@ class MergedColumnParallelLinear(nn.Module):
         self.weight = nn.Parameter(
             torch.cat(shards, dim=0).contiguous(), requires_grad=False
         )
+        # per-segment sharded bias (Qwen2.5 qkv); zero-init like HF does
+        # for missing bias, overwritten by the checkpoint loader
+        self.bias = (
+            nn.Parameter(
+                torch.zeros(sum(self.out_per_rank_sizes), dtype=dtype),
+                requires_grad=False,
+            )
+            if bias
+            else None
+        )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.linear(x, self.weight)
+        return F.linear(x, self.weight, self.bias)
 
 
 class RowParallelLinear(nn.Module):

@@ -50,6 +50,7 @@ class Attention(nn.Module):
                 cfg.num_kv_heads * cfg.head_dim,
                 cfg.num_kv_heads * cfg.head_dim,
             ],
+            bias=cfg.attention_bias,   # Qwen2.5 family
         )
         self.o_proj = RowParallelLinear(cfg.num_heads * cfg.head_dim, hidden)
         if cfg.qk_norm:

@@ -94,6 +94,38 @@ register(
     qk_norm=False,
 )
 
+register(
+    "Qwen2.5-7B",
+    hidden_size=3584,
+    num_layers=28,
+    num_heads=28,
+    num_kv_heads=4,
+    head_dim=128,
+    intermediate_size=18944,
+    vocab_size=152064,
+    rope_theta=1_000_000.0,
+    rms_norm_eps=1e-6,
+    max_position_embeddings=32768,
+    qk_norm=False,
+    attention_bias=True,
+)
+
+register(
+    "Qwen2.5-32B",
+    hidden_size=5120,
+    num_layers=64,
+    num_heads=40,
+    num_kv_heads=8,
+    head_dim=128,
+    intermediate_size=27648,
+    vocab_size=152064,
+    rope_theta=1_000_000.0,
+    rms_norm_eps=1e-6,
+    max_position_embeddings=32768,
+    qk_norm=False,
+    attention_bias=True,
+)
+
 # Tiny debug model: same code paths, CPU-testable sizes.
 register(
     "tiny-qwen3",

@@ -108,6 +108,20 @@ def load_hf_state_dict(
             attn.qkv_proj.weight_scale.copy_(sc)
         put(attn.qkv_proj.weight, w)
 
+    def try_fuse_qkv_bias(li: int):
+        keys = [(li, "q_bias"), (li, "k_bias"), (li, "v_bias")]
+        if not all(k in pending for k in keys):
+            return
+        attn = local(li).self_attn
+        b = torch.cat([
+            _shard_rows(pending.pop(k), tp, rank) for k in keys
+        ])
+        assert attn.qkv_proj.bias is not None, (
+            "checkpoint has qkv bias but the model config lacks "
+            "attention_bias=True"
+        )
+        put(attn.qkv_proj.bias, b)
+
     def try_fuse_gate_up(li: int):
         keys = [(li, "gate"), (li, "up")]
         if not all(k in pending for k in keys):
@@ -218,6 +232,11 @@ def load_hf_state_dict(
             elif rest == "self_attn.v_proj.weight":
                 pending[(li, "v")] = w
                 try_fuse_qkv(li)
+            elif rest in ("self_attn.q_proj.bias",
+                          "self_attn.k_proj.bias",
+                          "self_attn.v_proj.bias"):
+                pending[(li, rest.split(".")[1][0] + "_bias")] = w
+                try_fuse_qkv_bias(li)
             elif rest in ("self_attn.q_proj.weight_scale",
                           "self_attn.k_proj.weight_scale",
                           "self_attn.v_proj.weight_scale"):

@@ -91,9 +91,12 @@ def convert_linear_to_fp8(module: nn.Module) -> int:
 
 def _make_fp8_forward(m):
     is_row_parallel = hasattr(m, "in_per_rank")
+    bias = getattr(m, "bias", None)
 
     def fwd(x):
         out = fp8_linear(x, m.weight, m.weight_scale)
+        if bias is not None:
+            out = out + bias
         if is_row_parallel:
             out = ps.tp_all_reduce(out)
         return out

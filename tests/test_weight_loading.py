@@ -195,3 +195,40 @@ def test_fp8_native_checkpoint_matches_quantize_on_load(tmp_path):
     out_a = eng_a.generate([prompt], SamplingParams(max_tokens=5))[0]
     out_b = eng_b.generate([prompt], SamplingParams(max_tokens=5))[0]
     assert out_a.output_token_ids == out_b.output_token_ids
+
+
+def test_qwen25_attention_bias_loading():
+    """Qwen2.5-family qkv bias: zero-init bias is a no-op; loaded bias
+    changes the logits and q/k/v biases fuse + shard like their weights."""
+    ps.ensure_single_process()
+    torch.manual_seed(4)
+    mc = get_model_config("tiny-qwen3")
+    mc.attention_bias = True
+    mc.qk_norm = False
+    model = CausalLM(mc).eval()
+    tokens = [3, 1, 4, 1, 5] * 4
+    before = _forward_logits(model, tokens)
+
+    q_size = mc.num_heads * mc.head_dim
+    kv = mc.num_kv_heads * mc.head_dim
+    sd = {}
+    for li in range(mc.num_layers):
+        p = f"model.layers.{li}.self_attn."
+        sd[p + "q_proj.bias"] = torch.randn(q_size, dtype=torch.bfloat16) * 0.1
+        sd[p + "k_proj.bias"] = torch.randn(kv, dtype=torch.bfloat16) * 0.1
+        sd[p + "v_proj.bias"] = torch.randn(kv, dtype=torch.bfloat16) * 0.1
+    n = load_hf_state_dict(model, sd.items())
+    assert n == len(sd)
+    got = model.layers[0].self_attn.qkv_proj.bias.data
+    expect = torch.cat([
+        sd["model.layers.0.self_attn.q_proj.bias"],
+        sd["model.layers.0.self_attn.k_proj.bias"],
+        sd["model.layers.0.self_attn.v_proj.bias"],
+    ])
+    torch.testing.assert_close(got.float(), expect.float())
+    after = _forward_logits(model, tokens)
+    assert not torch.allclose(before.float(), after.float())
+
+    # real Qwen2.5 registry entries exist and build
+    assert get_model_config("Qwen2.5-7B").attention_bias
+    assert get_model_config("Qwen2.5-32B").attention_bias
