@@ -148,6 +148,30 @@ class LLMEngine:
         self.runner.lora_registry.add(adapter)
         return adapter
 
+    def add_lora_from_path(self, name: str, path: str):
+        """Load a PEFT adapter directory (vLLM /v1/load_lora_adapter).
+        TP > 1: the registration broadcasts; all ranks read the shared
+        path and keep their own shard."""
+        from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
+
+        if (self.cfg.parallel.tensor_parallel_size
+                * self.cfg.parallel.pipeline_parallel_size) > 1:
+            assert self.is_driver
+            self._ps.tp_broadcast_object(
+                {"kind": "add_lora_path", "name": name, "path": path}
+            )
+        if self.runner.lora_registry is None:
+            self.runner.lora_registry = LoRARegistry()
+        adapter = LoRAAdapter.from_safetensors(
+            name, path, self.cfg.model, device=self.device
+        )
+        self.runner.lora_registry.add(adapter)
+        return adapter
+
+    def remove_lora(self, name: str) -> bool:
+        reg = self.runner.lora_registry
+        return bool(reg and reg.remove(name))
+
     def active_loras(self) -> List[str]:
         reg = self.runner.lora_registry
         return reg.names() if reg else []
@@ -636,6 +660,15 @@ class LLMEngine:
             if payload["kind"] == "add_lora":
                 self._register_lora(payload["name"], payload["rank"],
                                     payload["alpha"], payload["seed"])
+            elif payload["kind"] == "add_lora_path":
+                from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
+
+                if self.runner.lora_registry is None:
+                    self.runner.lora_registry = LoRARegistry()
+                self.runner.lora_registry.add(LoRAAdapter.from_safetensors(
+                    payload["name"], payload["path"], self.cfg.model,
+                    device=self.device,
+                ))
             elif payload["kind"] == "mixed":
                 self.runner.run_batch(payload)
             elif payload["kind"] == "prefill":

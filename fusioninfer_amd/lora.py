@@ -94,6 +94,114 @@ class LoRAAdapter:
                 layer[tgt] = (A.to(device), B.to(device))
         return self
 
+    @classmethod
+    def from_safetensors(cls, name: str, path: str, model_cfg,
+                         dtype=torch.bfloat16, device="cpu"):
+        """Load a PEFT adapter directory (adapter_model.safetensors +
+        adapter_config.json). PEFT keeps one (A, B) pair per projection;
+        the merged targets (qkv, gate_up) are reconstructed as a
+        block-diagonal adapter of rank = sum of the member ranks:
+        A = [A_q; A_k; A_v], B with each B_proj in its segment's rows and
+        its own rank columns — mathematically identical to applying the
+        projections separately. Per-projection alpha/r scaling is folded
+        into B, so mixed configs compose; TP sharding then follows the
+        base-weight layout (B by column segments, A on row-parallel
+        inputs)."""
+        import json as _json
+        import os
+
+        from safetensors.torch import safe_open
+
+        cfg_path = os.path.join(path, "adapter_config.json")
+        alpha_over_r = 1.0
+        if os.path.isfile(cfg_path):
+            with open(cfg_path) as f:
+                pc = _json.load(f)
+            if pc.get("r"):
+                alpha_over_r = float(
+                    pc.get("lora_alpha", pc["r"])
+                ) / float(pc["r"])
+        st_path = os.path.join(path, "adapter_model.safetensors")
+        tensors: Dict[str, torch.Tensor] = {}
+        with safe_open(st_path, framework="pt") as sf:
+            for k in sf.keys():
+                tensors[k] = sf.get_tensor(k)
+
+        def pair(li: int, proj: str):
+            for prefix in (
+                f"base_model.model.model.layers.{li}.{proj}",
+                f"model.layers.{li}.{proj}",
+            ):
+                a = tensors.get(prefix + ".lora_A.weight")
+                b = tensors.get(prefix + ".lora_B.weight")
+                if a is not None and b is not None:
+                    return a.float(), b.float() * alpha_over_r
+            return None
+
+        tp = ps.tp_world_size()
+        tpr = ps.tp_rank()
+        H = model_cfg.hidden_size
+        inter = model_cfg.intermediate_size
+        self = cls.__new__(cls)
+        self.name = name
+        self.scaling = 1.0  # folded into B above
+        self.rank = 0
+        self.weights = []
+        merged = {
+            "qkv": (["self_attn.q_proj", "self_attn.k_proj",
+                     "self_attn.v_proj"],
+                    [model_cfg.num_heads * model_cfg.head_dim,
+                     model_cfg.num_kv_heads * model_cfg.head_dim,
+                     model_cfg.num_kv_heads * model_cfg.head_dim], H),
+            "gate_up": (["mlp.gate_proj", "mlp.up_proj"],
+                        [inter, inter], H),
+            "o": (["self_attn.o_proj"],
+                  [H], model_cfg.num_heads * model_cfg.head_dim),
+            "down": (["mlp.down_proj"], [H], inter),
+        }
+        for li in range(model_cfg.num_layers):
+            layer = {}
+            for tgt, (projs, seg_sizes, inf) in merged.items():
+                pairs = [pair(li, p) for p in projs]
+                if not any(p is not None for p in pairs):
+                    # adapter does not target this module: rank-1 zeros
+                    layer[tgt] = (
+                        torch.zeros(1, inf // (tp if tgt in ("o", "down")
+                                               else 1), dtype=dtype,
+                                    device=device),
+                        torch.zeros(
+                            sum(s // (tp if tgt in ("qkv", "gate_up")
+                                      else 1) for s in seg_sizes), 1,
+                            dtype=dtype, device=device),
+                    )
+                    continue
+                ranks = [0 if p is None else p[0].shape[0] for p in pairs]
+                R = sum(r for r in ranks) or 1
+                A = torch.zeros(R, inf)
+                out_total = sum(seg_sizes)
+                B = torch.zeros(out_total, R)
+                r_off = 0
+                seg_off = 0
+                for p, r, sz in zip(pairs, ranks, seg_sizes):
+                    if p is not None:
+                        a, b = p
+                        A[r_off: r_off + r] = a
+                        B[seg_off: seg_off + sz, r_off: r_off + r] = b
+                        r_off += r
+                    seg_off += sz
+                self.rank = max(self.rank, R)
+                if tgt in ("qkv", "gate_up"):
+                    B = LoRAAdapter._shard_col_segments(
+                        B, tgt, model_cfg, tp, tpr
+                    )
+                else:
+                    per = inf // tp
+                    A = A[:, tpr * per: (tpr + 1) * per]
+                layer[tgt] = (A.to(dtype=dtype, device=device),
+                              B.to(dtype=dtype, device=device))
+            self.weights.append(layer)
+        return self
+
 
 class LoRABatch:
     """Per-step grouping: [(adapter, row_index_tensor)]."""
@@ -129,3 +237,6 @@ class LoRARegistry:
 
     def names(self) -> List[str]:
         return sorted(self._adapters)
+
+    def remove(self, name: str) -> bool:
+        return self._adapters.pop(name, None) is not None

@@ -258,6 +258,36 @@ def build_app(serving: ServingEngine, model_name: str,
             if finished:
                 return
 
+    @app.post("/v1/load_lora_adapter")
+    async def load_lora(request: Request):
+        body = await request.json()
+        name = body.get("lora_name")
+        path = body.get("lora_path")
+        if not name or not path:
+            return JSONResponse({"error": {
+                "message": "lora_name and lora_path required",
+                "type": "invalid_request_error"}}, 400)
+        try:
+            loop = asyncio.get_event_loop()
+            await loop.run_in_executor(
+                None, serving.engine.add_lora_from_path, name, path
+            )
+        except Exception as e:
+            return JSONResponse({"error": {"message": repr(e),
+                                           "type": "invalid_request_error"}},
+                                400)
+        return PlainTextResponse(f"Success: LoRA adapter '{name}' added")
+
+    @app.post("/v1/unload_lora_adapter")
+    async def unload_lora(request: Request):
+        body = await request.json()
+        name = body.get("lora_name")
+        if not serving.engine.remove_lora(name or ""):
+            return JSONResponse({"error": {
+                "message": f"adapter {name!r} not found",
+                "type": "invalid_request_error"}}, 404)
+        return PlainTextResponse(f"Success: LoRA adapter '{name}' removed")
+
     @app.post("/tokenize")
     async def tokenize(request: Request):
         body = await request.json()

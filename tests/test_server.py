@@ -437,3 +437,34 @@ def test_prompt_logprobs_api(app):
             assert len(ch["logprobs"]) == 2  # one entry per output token
 
     asyncio.run(run())
+
+
+def test_dynamic_lora_endpoints(app, serving, tmp_path):
+    from tests.test_lora import _write_peft_adapter
+
+    _write_peft_adapter(tmp_path / "ad", serving.engine.cfg.model)
+
+    async def run():
+        async with _client(app) as c:
+            r = await c.post("/v1/load_lora_adapter",
+                             json={"lora_name": "dyn",
+                                   "lora_path": str(tmp_path / "ad")})
+            assert r.status_code == 200, r.text
+            r = await c.get("/v1/models")
+            assert any(m["id"] == "dyn" for m in r.json()["data"])
+            # request routed to the adapter by model name
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": [5, 6, 7] * 4, "max_tokens": 2,
+                      "model": "dyn", "temperature": 0.0,
+                      "ignore_eos": True},
+            )
+            assert r.status_code == 200
+            r = await c.post("/v1/unload_lora_adapter",
+                             json={"lora_name": "dyn"})
+            assert r.status_code == 200
+            r = await c.post("/v1/unload_lora_adapter",
+                             json={"lora_name": "dyn"})
+            assert r.status_code == 404
+
+    asyncio.run(run())
