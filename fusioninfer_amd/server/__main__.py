@@ -47,7 +47,9 @@ def parse_args(argv=None):
                    help="accepted for reference-surface parity; adapters "
                         "are small enough to stay resident")
     p.add_argument("--lora-modules", nargs="*", default=[],
-                   help="adapters to register at startup: name[=rank]")
+                   help="adapters to register at startup: name=/peft/dir "
+                        "(loads adapter_model.safetensors) or name[=rank] "
+                        "(seeded synthetic adapter)")
     p.add_argument("--enable-prefix-caching", action="store_true")
     p.add_argument("--scheduling-policy", choices=["fcfs", "priority"],
                    default="fcfs")
@@ -161,8 +163,11 @@ def _rank_main(local_rank: int, args, nproc: int):
     serving = ServingEngine(cfg, device=device, kv_connector=kv_connector)
     if args.enable_lora or args.lora_modules:
         for spec in args.lora_modules:
-            name, _, r = spec.partition("=")
-            serving.engine.add_lora(name, rank=int(r) if r else 16)
+            name, _, val = spec.partition("=")
+            if val and not val.isdigit():
+                serving.engine.add_lora_from_path(name, val)
+            else:
+                serving.engine.add_lora(name, rank=int(val) if val else 16)
 
     import uvicorn
 
