@@ -121,5 +121,9 @@ class EngineConfig:
     # vLLM --speculative-config parity: None = off; set to a
     # SpeculativeConfig (engine/spec_decode.py) for ngram drafting
     speculative: Optional[object] = None
+    # LoRA capacity tiering (vLLM --max-loras / --max-cpu-loras):
+    # device-resident cap and total (host-parked) cap
+    max_loras: int = 8
+    max_cpu_loras: int = 16
     seed: int = 0
     enforce_eager: bool = False            # True disables hipGraph decode capture

@@ -136,12 +136,22 @@ class LLMEngine:
             )
         return self._register_lora(name, rank, alpha, seed)
 
-    def _register_lora(self, name, rank, alpha, seed):
-        """Per-rank adapter construction (driver and TP workers)."""
-        from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
+    def _lora_registry(self):
+        from fusioninfer_amd.lora import LoRARegistry
 
         if self.runner.lora_registry is None:
-            self.runner.lora_registry = LoRARegistry()
+            self.runner.lora_registry = LoRARegistry(
+                max_loras=self.cfg.max_loras,
+                max_cpu_loras=self.cfg.max_cpu_loras,
+                device=self.device,
+            )
+        return self.runner.lora_registry
+
+    def _register_lora(self, name, rank, alpha, seed):
+        """Per-rank adapter construction (driver and TP workers)."""
+        from fusioninfer_amd.lora import LoRAAdapter
+
+        self._lora_registry()
         adapter = LoRAAdapter(
             name, rank, alpha, self.cfg.model, device=self.device, seed=seed
         )
@@ -152,7 +162,7 @@ class LLMEngine:
         """Load a PEFT adapter directory (vLLM /v1/load_lora_adapter).
         TP > 1: the registration broadcasts; all ranks read the shared
         path and keep their own shard."""
-        from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
+        from fusioninfer_amd.lora import LoRAAdapter
 
         if (self.cfg.parallel.tensor_parallel_size
                 * self.cfg.parallel.pipeline_parallel_size) > 1:
@@ -160,8 +170,7 @@ class LLMEngine:
             self._ps.tp_broadcast_object(
                 {"kind": "add_lora_path", "name": name, "path": path}
             )
-        if self.runner.lora_registry is None:
-            self.runner.lora_registry = LoRARegistry()
+        self._lora_registry()
         adapter = LoRAAdapter.from_safetensors(
             name, path, self.cfg.model, device=self.device
         )
@@ -661,11 +670,9 @@ class LLMEngine:
                 self._register_lora(payload["name"], payload["rank"],
                                     payload["alpha"], payload["seed"])
             elif payload["kind"] == "add_lora_path":
-                from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
+                from fusioninfer_amd.lora import LoRAAdapter
 
-                if self.runner.lora_registry is None:
-                    self.runner.lora_registry = LoRARegistry()
-                self.runner.lora_registry.add(LoRAAdapter.from_safetensors(
+                self._lora_registry().add(LoRAAdapter.from_safetensors(
                     payload["name"], payload["path"], self.cfg.model,
                     device=self.device,
                 ))

@@ -220,23 +220,57 @@ class LoRABatch:
 
 
 class LoRARegistry:
-    def __init__(self, max_loras: int = 8):
+    """Adapter store with vLLM's --max-loras / --max-cpu-loras tiering:
+    at most `max_loras` adapters stay device-resident; the LRU ones park
+    in host memory (up to `max_cpu_loras` total) and page back on use."""
+
+    def __init__(self, max_loras: int = 8, max_cpu_loras: int = 16,
+                 device: str = "cpu"):
         self.max_loras = max_loras
+        self.max_cpu_loras = max(max_cpu_loras, max_loras)
+        self.device = device
         self._adapters: Dict[str, LoRAAdapter] = {}
+        self._resident: Dict[str, int] = {}  # name -> last-use tick
+        self._tick = 0
+
+    def _evict_lru_if_needed(self) -> None:
+        while len(self._resident) > self.max_loras:
+            lru = min(self._resident, key=self._resident.get)
+            del self._resident[lru]
+            if self.device != "cpu":
+                self._adapters[lru].to("cpu")
 
     def add(self, adapter: LoRAAdapter) -> None:
-        if len(self._adapters) >= self.max_loras and adapter.name not in self._adapters:
-            raise RuntimeError(f"max_loras={self.max_loras} reached")
+        if (len(self._adapters) >= self.max_cpu_loras
+                and adapter.name not in self._adapters):
+            raise RuntimeError(f"max_cpu_loras={self.max_cpu_loras} reached")
         self._adapters[adapter.name] = adapter
+        self._tick += 1
+        self._resident[adapter.name] = self._tick
+        self._evict_lru_if_needed()
 
     def get(self, name: str) -> LoRAAdapter:
-        return self._adapters[name]
+        adapter = self._adapters[name]
+        self._tick += 1
+        if name not in self._resident:
+            # page back in from host memory
+            if self.device != "cpu":
+                adapter.to(self.device)
+            self._resident[name] = self._tick
+            self._evict_lru_if_needed()
+        else:
+            self._resident[name] = self._tick
+        return adapter
 
     def maybe_get(self, name: Optional[str]) -> Optional[LoRAAdapter]:
-        return self._adapters.get(name) if name else None
+        return self.get(name) if name and name in self._adapters else None
 
     def names(self) -> List[str]:
         return sorted(self._adapters)
 
+    def num_resident(self) -> int:
+        return len(self._resident)
+
     def remove(self, name: str) -> bool:
+        self._resident.pop(name, None)
         return self._adapters.pop(name, None) is not None

@@ -127,6 +127,8 @@ def build_engine_config(args):
         ),
         kv_transfer=kvt,
         speculative=speculative,
+        max_loras=args.max_loras,
+        max_cpu_loras=args.max_cpu_loras,
         enforce_eager=args.enforce_eager,
     )
 

@@ -136,3 +136,25 @@ def test_peft_adapter_from_safetensors(tmp_path):
     assert eng.remove_lora("tuned")
     again = eng.generate([prompt], SamplingParams(max_tokens=5))[0]
     assert again.output_token_ids == base.output_token_ids
+
+
+def test_registry_lru_tiering():
+    """--max-loras/--max-cpu-loras semantics: at most max_loras resident,
+    LRU pages out, paging back in on use; max_cpu_loras hard-caps."""
+    from fusioninfer_amd.lora import LoRAAdapter, LoRARegistry
+    from fusioninfer_amd.models.registry import get_model_config
+
+    mc = get_model_config("tiny-qwen3")
+    reg = LoRARegistry(max_loras=2, max_cpu_loras=3, device="cpu")
+    for n in ("a", "b", "c"):
+        reg.add(LoRAAdapter(n, 2, 4.0, mc, seed=1))
+    assert reg.num_resident() == 2          # 'a' was evicted (LRU)
+    assert set(reg.names()) == {"a", "b", "c"}
+    reg.get("a")                            # pages 'a' back in
+    assert reg.num_resident() == 2
+    import pytest as _p
+    with _p.raises(RuntimeError):
+        reg.add(LoRAAdapter("d", 2, 4.0, mc, seed=1))
+    assert reg.remove("c")
+    reg.add(LoRAAdapter("d", 2, 4.0, mc, seed=1))
+    assert set(reg.names()) == {"a", "b", "d"}
