@@ -119,7 +119,8 @@ class EngineConfig:
     parallel: ParallelConfig = dataclasses.field(default_factory=ParallelConfig)
     kv_transfer: KVTransferConfig = dataclasses.field(default_factory=KVTransferConfig)
     # vLLM --speculative-config parity: None = off; set to a
-    # SpeculativeConfig (engine/spec_decode.py) for ngram drafting
+    # SpeculativeConfig (engine/spec_decode.py) for ngram prompt-lookup
+    # or draft_model speculation
     speculative: Optional[object] = None
     # LoRA capacity tiering (vLLM --max-loras / --max-cpu-loras):
     # device-resident cap and total (host-parked) cap

@@ -8,9 +8,11 @@ and Prometheus /metrics with the vLLM-compatible gauge names the EPP
 scorers scrape (vllm:gpu_cache_usage_perc, vllm:num_requests_waiting —
 reference pkg/router/strategy.go:70-98 consumes these).
 
-Tokenization: there is no network access for tokenizer files, so string
-prompts use a reversible byte-level fallback; OpenAI's prompt-as-token-ids
-form is supported natively and is what the benchmark and EPP flows use.
+Tokenization: vendored tokenizer.json files load through the offline
+`tokenizers` wheel (--tokenizer / checkpoint-dir autodetect); without one,
+string prompts use a reversible byte-level fallback. OpenAI's
+prompt-as-token-ids form is first-class either way and is what the
+benchmark and EPP flows use.
 """
 
 from __future__ import annotations
@@ -449,13 +451,21 @@ def build_app(serving: ServingEngine, model_name: str,
             extra.append(serving.submit(prompt_ids, s_i, lora_name=lora))
         keep_stop = bool(body.get("include_stop_str_in_output"))
         choices = []
-        toks, text = await _collect(q, stops=stops, req_id=req_id,
-                                    keep_stop=keep_stop)
-        final = serving.take_final(req_id)
-        choices.append((toks, text))
-        for rid_i, q_i in extra:
-            choices.append(await _collect(q_i, stops=stops, req_id=rid_i,
-                                          keep_stop=keep_stop))
+        try:
+            toks, text = await _collect(q, stops=stops, req_id=req_id,
+                                        keep_stop=keep_stop)
+            final = serving.take_final(req_id)
+            choices.append((toks, text))
+            for rid_i, q_i in extra:
+                choices.append(await _collect(q_i, stops=stops,
+                                              req_id=rid_i,
+                                              keep_stop=keep_stop))
+        except asyncio.CancelledError:
+            # client disconnected: stop burning engine steps on it
+            serving.abort(req_id)
+            for rid_i, _ in extra:
+                serving.abort(rid_i)
+            raise
         if echo:
             prefix = tokenizer.decode(prompt_ids)
             choices = [(t, prefix + x) for t, x in choices]
@@ -572,10 +582,14 @@ def build_app(serving: ServingEngine, model_name: str,
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
-        toks, out_text = await _collect(
-            q, stops=stops, req_id=req_id,
-            keep_stop=bool(body.get("include_stop_str_in_output")),
-        )
+        try:
+            toks, out_text = await _collect(
+                q, stops=stops, req_id=req_id,
+                keep_stop=bool(body.get("include_stop_str_in_output")),
+            )
+        except asyncio.CancelledError:
+            serving.abort(req_id)
+            raise
         return JSONResponse(
             {
                 "id": cid,
