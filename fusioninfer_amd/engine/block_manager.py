@@ -31,6 +31,10 @@ class BlockManager:
         self.block_hash: Dict[int, int] = {}
         # blocks with ref 0 that still hold reusable content (LRU)
         self.cached_free: "OrderedDict[int, None]" = OrderedDict()
+        # prefix-cache effectiveness counters (EPP's prefix scorer assumes
+        # engine-side reuse; these make it observable)
+        self.cache_query_tokens = 0
+        self.cache_hit_tokens = 0
 
     # ------------------------------------------------------------ helpers
     def num_free(self) -> int:
@@ -75,6 +79,9 @@ class BlockManager:
                 seq.block_ids.append(blk)
                 cached += self.block_size
         seq.num_cached_tokens = cached
+        if self.enable_prefix_caching:
+            self.cache_query_tokens += seq.num_prompt_tokens
+            self.cache_hit_tokens += cached
         while len(seq.block_ids) < n:
             blk = self._pop_free_block()
             self.ref_count[blk] = 1

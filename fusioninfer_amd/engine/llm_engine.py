@@ -108,6 +108,9 @@ class LLMEngine:
                             10.0, 15.0, 30.0, 60.0]
         self.ttft_hist = [0] * (len(self.ttft_buckets) + 1)
         self.e2e_hist = [0] * (len(self.e2e_buckets) + 1)
+        self.tpot_buckets = [0.01, 0.025, 0.05, 0.075, 0.1, 0.15, 0.2,
+                             0.3, 0.4, 0.5, 0.75, 1.0]
+        self.tpot_hist = [0] * (len(self.tpot_buckets) + 1)
         self.step_time_sum = 0.0
         self.num_steps = 0
 
@@ -482,6 +485,12 @@ class LLMEngine:
         guided = seq.sampling.guided
         new = []
         for tok in toks:
+            now = time.monotonic()
+            if seq.last_token_time is not None:
+                # inter-token latency (vLLM time_per_output_token)
+                self._observe(self.tpot_hist, self.tpot_buckets,
+                              now - seq.last_token_time)
+            seq.last_token_time = now
             seq.append_token(int(tok))
             new.append(int(tok))
             self.num_generated_tokens += 1

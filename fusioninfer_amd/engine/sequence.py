@@ -68,6 +68,7 @@ class Sequence:
         self.block_ids: List[int] = []
         self.arrival_time = time.monotonic()
         self.first_token_time: Optional[float] = None
+        self.last_token_time: Optional[float] = None
         self.finish_time: Optional[float] = None
         # number of prompt tokens whose KV was satisfied by prefix cache
         self.num_cached_tokens = 0

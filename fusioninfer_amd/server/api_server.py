@@ -136,6 +136,8 @@ def build_app(serving: ServingEngine, model_name: str,
             ("num_swap_outs_total", "counter"),
             ("spec_decode_num_draft_tokens_total", "counter"),
             ("spec_decode_num_accepted_tokens_total", "counter"),
+            ("prefix_cache_queries_total", "counter"),
+            ("prefix_cache_hits_total", "counter"),
         ]:
             lines.append(f"# TYPE vllm:{name} {mtype}")
             lines.append(
@@ -148,6 +150,8 @@ def build_app(serving: ServingEngine, model_name: str,
              e.ttft_sum),
             ("e2e_request_latency_seconds", e.e2e_buckets, e.e2e_hist,
              e.e2e_latency_sum),
+            ("time_per_output_token_seconds", e.tpot_buckets, e.tpot_hist,
+             0.0),
         ]:
             lines.append(f"# TYPE vllm:{name} histogram")
             cum = 0

@@ -161,6 +161,12 @@ class ServingEngine:
             "spec_decode_num_accepted_tokens_total": float(
                 e.num_spec_accepted_tokens
             ),
+            "prefix_cache_queries_total": float(
+                e.block_manager.cache_query_tokens
+            ),
+            "prefix_cache_hits_total": float(
+                e.block_manager.cache_hit_tokens
+            ),
         }
 
     def take_final(self, request_id: str):
