@@ -205,6 +205,16 @@ def build_app(serving: ServingEngine, model_name: str,
         return None
 
     def _sampling_from(body: Dict[str, Any]) -> SamplingParams:
+        mt = int(body.get("max_tokens", 16))
+        if mt < 1:
+            raise ValueError("max_tokens must be >= 1")
+        if float(body.get("temperature", 1.0)) < 0:
+            raise ValueError("temperature must be >= 0")
+        tp_ = float(body.get("top_p", 1.0))
+        if not 0.0 < tp_ <= 1.0:
+            raise ValueError("top_p must be in (0, 1]")
+        if int(body.get("top_k", 0)) < 0:
+            raise ValueError("top_k must be >= 0")
         stop_ids = list(body.get("stop_token_ids") or [])
         if tokenizer.eos_token_id is not None \
                 and tokenizer.eos_token_id not in stop_ids:

@@ -468,3 +468,20 @@ def test_dynamic_lora_endpoints(app, serving, tmp_path):
             assert r.status_code == 404
 
     asyncio.run(run())
+
+
+def test_bad_sampling_params_400(app):
+    async def run():
+        async with _client(app) as c:
+            for bad in (
+                {"max_tokens": 0},
+                {"temperature": -1},
+                {"top_p": 0.0},
+                {"top_p": 1.5},
+                {"top_k": -2},
+            ):
+                r = await c.post("/v1/completions",
+                                 json={"prompt": "x", **bad})
+                assert r.status_code == 400, bad
+
+    asyncio.run(run())
