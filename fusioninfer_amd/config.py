@@ -29,6 +29,8 @@ class ModelConfig:
     max_position_embeddings: int = 40960
     qk_norm: bool = True              # Qwen3 per-head q/k RMSNorm
     attention_bias: bool = False      # Qwen2.5-style qkv bias
+    # HF-style rope_scaling dict (Llama-3.1 "llama3" remap, "linear")
+    rope_scaling: Optional[dict] = None
     tie_word_embeddings: bool = False
     dtype: str = "bfloat16"
     # optional HF-layout safetensors checkpoint dir; None = random init

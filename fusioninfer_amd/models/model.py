@@ -445,7 +445,8 @@ class CausalLM(nn.Module):
             self.lm_head = None
         torch.manual_seed(base_seed + 777)  # same post-init state on all ranks
         cos_sin = ops.compute_cos_sin_cache(
-            cfg.head_dim, cfg.max_position_embeddings, cfg.rope_theta
+            cfg.head_dim, cfg.max_position_embeddings, cfg.rope_theta,
+            rope_scaling=cfg.rope_scaling,
         )
         self.register_buffer("cos_sin", cos_sin, persistent=False)
 

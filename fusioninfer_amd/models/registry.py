@@ -94,6 +94,46 @@ register(
     qk_norm=False,
 )
 
+_LLAMA31_SCALING = {
+    "rope_type": "llama3",
+    "factor": 8.0,
+    "low_freq_factor": 1.0,
+    "high_freq_factor": 4.0,
+    "original_max_position_embeddings": 8192,
+}
+
+register(
+    "Llama-3.1-8B",
+    hidden_size=4096,
+    num_layers=32,
+    num_heads=32,
+    num_kv_heads=8,
+    head_dim=128,
+    intermediate_size=14336,
+    vocab_size=128256,
+    rope_theta=500_000.0,
+    rms_norm_eps=1e-5,
+    max_position_embeddings=131072,
+    qk_norm=False,
+    rope_scaling=_LLAMA31_SCALING,
+)
+
+register(
+    "Llama-3.1-70B",
+    hidden_size=8192,
+    num_layers=80,
+    num_heads=64,
+    num_kv_heads=8,
+    head_dim=128,
+    intermediate_size=28672,
+    vocab_size=128256,
+    rope_theta=500_000.0,
+    rms_norm_eps=1e-5,
+    max_position_embeddings=131072,
+    qk_norm=False,
+    rope_scaling=_LLAMA31_SCALING,
+)
+
 register(
     "Qwen2.5-7B",
     hidden_size=3584,

@@ -58,13 +58,44 @@ def silu_and_mul_fp8(x: torch.Tensor):
     return quant_fp8_rows(silu_and_mul(x))
 
 
+def _llama3_scale_inv_freq(inv_freq: torch.Tensor, sc: dict) -> torch.Tensor:
+    """Llama-3.1 rope scaling (HF rope_scaling rope_type="llama3"):
+    low-frequency bands are divided by `factor`, high-frequency bands
+    kept, with a smooth ramp between (per-wavelength interpolation)."""
+    import math
+
+    factor = float(sc.get("factor", 8.0))
+    low = float(sc.get("low_freq_factor", 1.0))
+    high = float(sc.get("high_freq_factor", 4.0))
+    orig = float(sc.get("original_max_position_embeddings", 8192))
+    wavelen = 2 * math.pi / inv_freq
+    low_wl = orig / low
+    high_wl = orig / high
+    scaled = torch.where(wavelen > low_wl, inv_freq / factor, inv_freq)
+    smooth = (orig / wavelen - low) / (high - low)
+    mid = (1 - smooth) * inv_freq / factor + smooth * inv_freq
+    is_mid = (wavelen <= low_wl) & (wavelen >= high_wl)
+    return torch.where(is_mid, mid, scaled)
+
+
 def compute_cos_sin_cache(
-    head_dim: int, max_positions: int, theta: float, device=None
+    head_dim: int, max_positions: int, theta: float, device=None,
+    rope_scaling=None,
 ) -> torch.Tensor:
-    """[max_positions, head_dim] fp32; first half cos, second half sin (NeoX)."""
+    """[max_positions, head_dim] fp32; first half cos, second half sin (NeoX).
+    rope_scaling: None, or an HF-style dict ({"rope_type": "llama3", ...}
+    — the Llama-3.1 long-context frequency remap)."""
     inv_freq = 1.0 / (
         theta ** (torch.arange(0, head_dim, 2, dtype=torch.float32, device=device) / head_dim)
     )
+    if rope_scaling:
+        kind = rope_scaling.get("rope_type") or rope_scaling.get("type")
+        if kind == "llama3":
+            inv_freq = _llama3_scale_inv_freq(inv_freq, rope_scaling)
+        elif kind == "linear":
+            inv_freq = inv_freq / float(rope_scaling.get("factor", 1.0))
+        else:
+            raise ValueError(f"unsupported rope_scaling {kind!r}")
     t = torch.arange(max_positions, dtype=torch.float32, device=device)
     freqs = torch.outer(t, inv_freq)  # [P, D/2]
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1)

@@ -71,3 +71,38 @@ def test_kv_pack_unpack_roundtrip():
     ref.scatter_kv_blocks(staging, k2, v2, ids)
     torch.testing.assert_close(k2[ids], k_cache[ids])
     torch.testing.assert_close(v2[ids], v_cache[ids])
+
+
+def test_llama31_rope_scaling_matches_transformers():
+    """Our llama3 rope remap equals transformers' reference math."""
+    import torch
+
+    from fusioninfer_amd.ops.reference import _llama3_scale_inv_freq
+
+    sc = {"rope_type": "llama3", "factor": 8.0, "low_freq_factor": 1.0,
+          "high_freq_factor": 4.0,
+          "original_max_position_embeddings": 8192}
+    head_dim, theta = 128, 500000.0
+    inv = 1.0 / (theta ** (torch.arange(0, head_dim, 2,
+                                        dtype=torch.float32) / head_dim))
+    ours = _llama3_scale_inv_freq(inv, sc)
+
+    try:
+        from transformers import LlamaConfig
+        from transformers.modeling_rope_utils import ROPE_INIT_FUNCTIONS
+
+        hf_cfg = LlamaConfig(
+            rope_theta=theta, head_dim=head_dim, hidden_size=head_dim * 32,
+            num_attention_heads=32, rope_scaling=dict(sc),
+            max_position_embeddings=131072,
+        )
+        theirs, att = ROPE_INIT_FUNCTIONS["llama3"](hf_cfg, device="cpu")
+        torch.testing.assert_close(ours, theirs.float(), rtol=1e-5,
+                                   atol=1e-7)
+        assert att == 1.0
+    except (ImportError, KeyError, AttributeError, TypeError):
+        # transformers helper interface moved: structural checks below
+        pass
+    # structural invariants hold either way
+    assert torch.allclose(ours[:4], inv[:4])          # high-freq kept
+    assert torch.allclose(ours[-4:], inv[-4:] / 8.0)  # low-freq / factor
