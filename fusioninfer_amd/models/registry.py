@@ -203,6 +203,9 @@ register(
 
 def get_model_config(name: str) -> ModelConfig:
     key = name.lower()
+    if key not in _REGISTRY and "/" in key:
+        # accept HF-style ids ("Qwen/Qwen3-8B" -> "Qwen3-8B")
+        key = key.rsplit("/", 1)[1]
     if key not in _REGISTRY:
         raise KeyError(f"unknown model {name!r}; known: {sorted(_REGISTRY)}")
     import dataclasses
