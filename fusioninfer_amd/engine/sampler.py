@@ -50,14 +50,16 @@ class Sampler:
     def sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> torch.Tensor:
         """logits: [S, V] fp32; returns [S] int64 token ids."""
         logits = self._apply_penalties(logits, seqs)
+        greedy = logits.argmax(dim=-1)
+        # host-side check: avoids a device sync on the (common) all-greedy
+        # decode step
+        if all(s.sampling.temperature == 0 for s in seqs):
+            return greedy
         temps = torch.tensor(
             [s.sampling.temperature for s in seqs],
             dtype=torch.float32,
             device=logits.device,
         )
-        greedy = logits.argmax(dim=-1)
-        if bool((temps == 0).all()):
-            return greedy
         scaled = logits / temps.clamp(min=1e-5).unsqueeze(1)
 
         # top-k then top-p filtering, batched over the rows that need it
