@@ -596,14 +596,28 @@ def build_app(serving: ServingEngine, model_name: str,
 
             return StreamingResponse(sse(), media_type="text/event-stream")
 
+        # OpenAI `n` (non-stream): extra choices as sibling requests —
+        # prefix caching dedups the shared-prompt KV
+        n = max(int(body.get("n", 1)), 1)
+        extra = []
+        for i in range(1, n):
+            s_i = _sampling_from(body)
+            if s_i.seed is not None:
+                s_i.seed += i
+            extra.append(serving.submit(prompt_ids, s_i, lora_name=lora))
+        keep = bool(body.get("include_stop_str_in_output"))
         try:
-            toks, out_text = await _collect(
-                q, stops=stops, req_id=req_id,
-                keep_stop=bool(body.get("include_stop_str_in_output")),
-            )
+            choices = [await _collect(q, stops=stops, req_id=req_id,
+                                      keep_stop=keep)]
+            for rid_i, q_i in extra:
+                choices.append(await _collect(q_i, stops=stops,
+                                              req_id=rid_i, keep_stop=keep))
         except asyncio.CancelledError:
             serving.abort(req_id)
+            for rid_i, _ in extra:
+                serving.abort(rid_i)
             raise
+        total = sum(len(t) for t, _ in choices)
         return JSONResponse(
             {
                 "id": cid,
@@ -612,18 +626,19 @@ def build_app(serving: ServingEngine, model_name: str,
                 "model": model_name,
                 "choices": [
                     {
-                        "index": 0,
+                        "index": i,
                         "message": {
                             "role": "assistant",
-                            "content": out_text,
+                            "content": c_text,
                         },
                         "finish_reason": "stop",
                     }
+                    for i, (_t, c_text) in enumerate(choices)
                 ],
                 "usage": {
                     "prompt_tokens": len(prompt_ids),
-                    "completion_tokens": len(toks),
-                    "total_tokens": len(prompt_ids) + len(toks),
+                    "completion_tokens": total,
+                    "total_tokens": len(prompt_ids) + total,
                 },
             }
         )

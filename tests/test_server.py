@@ -485,3 +485,19 @@ def test_bad_sampling_params_400(app):
                 assert r.status_code == 400, bad
 
     asyncio.run(run())
+
+
+def test_chat_n_choices(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/chat/completions",
+                json={"messages": [{"role": "user", "content": "pick"}],
+                      "max_tokens": 3, "n": 3, "temperature": 0.8,
+                      "seed": 7, "ignore_eos": True},
+            )
+            body = r.json()
+            assert [ch["index"] for ch in body["choices"]] == [0, 1, 2]
+            assert body["usage"]["completion_tokens"] == 9
+
+    asyncio.run(run())
