@@ -386,6 +386,86 @@ def build_app(serving: ServingEngine, model_name: str,
             (body.get("stream_options") or {}).get("include_usage")
         )
 
+        n_req = max(int(body.get("n", 1)), 1)
+        if body.get("stream") and n_req > 1:
+            # interleaved multi-choice stream: sibling requests advance in
+            # lockstep with the engine, so round-robin draining is fair
+            extra = []
+            for i in range(1, n_req):
+                s_i = _sampling_from(body)
+                if s_i.seed is not None:
+                    s_i.seed += i
+                extra.append(serving.submit(prompt_ids, s_i, lora_name=lora))
+            chans = [(0, req_id, q)] + [
+                (i + 1, rid_i, q_i) for i, (rid_i, q_i) in enumerate(extra)
+            ]
+
+            async def sse_multi():
+                loop = asyncio.get_event_loop()
+                live = {idx: (rid, qq) for idx, rid, qq in chans}
+                filts = {
+                    idx: (StopStreamFilter(stops) if stops else None)
+                    for idx, _, _ in chans
+                }
+                n_out = 0
+                try:
+                    while live:
+                        for idx in sorted(list(live)):
+                            rid, qq = live[idx]
+                            tok, finished = await loop.run_in_executor(
+                                None, qq.get
+                            )
+                            delta = (
+                                tokenizer.decode_one(tok)
+                                if tok is not None else ""
+                            )
+                            if tok is not None:
+                                n_out += 1
+                            f = filts[idx]
+                            if f is not None and tok is not None:
+                                delta, hit = f.push(delta)
+                                if hit:
+                                    finished = True
+                                    serving.abort(rid)
+                                elif finished:
+                                    delta += f.flush()
+                            chunk = {
+                                "id": cid,
+                                "object": "text_completion",
+                                "created": created,
+                                "model": model_name,
+                                "choices": [{
+                                    "index": idx,
+                                    "text": delta,
+                                    "token_ids":
+                                        [tok] if tok is not None else [],
+                                    "finish_reason":
+                                        "stop" if finished else None,
+                                }],
+                            }
+                            yield f"data: {json.dumps(chunk)}\n\n"
+                            if finished:
+                                del live[idx]
+                    if include_usage:
+                        yield "data: " + json.dumps({
+                            "id": cid, "object": "text_completion",
+                            "created": created, "model": model_name,
+                            "choices": [],
+                            "usage": {
+                                "prompt_tokens": len(prompt_ids) * n_req,
+                                "completion_tokens": n_out,
+                                "total_tokens":
+                                    len(prompt_ids) * n_req + n_out,
+                            },
+                        }) + "\n\n"
+                    yield "data: [DONE]\n\n"
+                finally:
+                    for idx in list(live):
+                        serving.abort(live[idx][0])
+
+            return StreamingResponse(sse_multi(),
+                                     media_type="text/event-stream")
+
         if body.get("stream"):
             async def sse():
                 done = False

@@ -501,3 +501,30 @@ def test_chat_n_choices(app):
             assert body["usage"]["completion_tokens"] == 9
 
     asyncio.run(run())
+
+
+def test_streaming_n_choices(app):
+    async def run():
+        async with _client(app) as c:
+            per_choice = {}
+            finish = {}
+            async with c.stream(
+                "POST", "/v1/completions",
+                json={"prompt": [5, 6, 7] * 6, "max_tokens": 3, "n": 2,
+                      "temperature": 0.8, "seed": 3, "stream": True,
+                      "ignore_eos": True},
+            ) as resp:
+                async for line in resp.aiter_lines():
+                    if not line.startswith("data: ") or line == "data: [DONE]":
+                        continue
+                    ch = json.loads(line[6:])["choices"][0]
+                    per_choice.setdefault(ch["index"], []).extend(
+                        ch["token_ids"]
+                    )
+                    if ch["finish_reason"] == "stop":
+                        finish[ch["index"]] = True
+            assert set(per_choice) == {0, 1}
+            assert all(len(v) == 3 for v in per_choice.values())
+            assert finish == {0: True, 1: True}
+
+    asyncio.run(run())
