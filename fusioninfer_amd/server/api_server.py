@@ -542,7 +542,10 @@ def build_app(serving: ServingEngine, model_name: str,
             s_i = _sampling_from(body)
             if s_i.seed is not None:
                 s_i.seed += i  # distinct choices under a fixed seed
-            extra.append(serving.submit(prompt_ids, s_i, lora_name=lora))
+            extra.append(serving.submit(
+                prompt_ids, s_i, lora_name=lora,
+                priority=int(body.get("priority", 0)),
+            ))
         keep_stop = bool(body.get("include_stop_str_in_output"))
         choices = []
         try:
