@@ -627,6 +627,77 @@ def build_app(serving: ServingEngine, model_name: str,
         created = int(time.time())
         cid = f"chatcmpl-{uuid.uuid4().hex[:16]}"
 
+        n_req = max(int(body.get("n", 1)), 1)
+        if body.get("stream") and n_req > 1:
+            extra = []
+            for i in range(1, n_req):
+                s_i = _sampling_from(body)
+                if s_i.seed is not None:
+                    s_i.seed += i
+                extra.append(serving.submit(
+                    prompt_ids, s_i, lora_name=lora,
+                    priority=int(body.get("priority", 0)),
+                ))
+            chans = [(0, req_id, q)] + [
+                (i + 1, rid_i, q_i) for i, (rid_i, q_i) in enumerate(extra)
+            ]
+
+            async def sse_multi():
+                loop = asyncio.get_event_loop()
+                live = {idx: (rid, qq) for idx, rid, qq in chans}
+                firsts = {idx: True for idx, _, _ in chans}
+                filts = {
+                    idx: (StopStreamFilter(stops) if stops else None)
+                    for idx, _, _ in chans
+                }
+                try:
+                    while live:
+                        for idx in sorted(list(live)):
+                            rid, qq = live[idx]
+                            tok, finished = await loop.run_in_executor(
+                                None, qq.get
+                            )
+                            delta: Dict[str, Any] = {}
+                            if firsts[idx]:
+                                delta["role"] = "assistant"
+                                firsts[idx] = False
+                            piece = (
+                                tokenizer.decode_one(tok)
+                                if tok is not None else ""
+                            )
+                            f = filts[idx]
+                            if f is not None and tok is not None:
+                                piece, hit = f.push(piece)
+                                if hit:
+                                    finished = True
+                                    serving.abort(rid)
+                                elif finished:
+                                    piece += f.flush()
+                            if piece:
+                                delta["content"] = piece
+                            chunk = {
+                                "id": cid,
+                                "object": "chat.completion.chunk",
+                                "created": created,
+                                "model": model_name,
+                                "choices": [{
+                                    "index": idx,
+                                    "delta": delta,
+                                    "finish_reason":
+                                        "stop" if finished else None,
+                                }],
+                            }
+                            yield f"data: {json.dumps(chunk)}\n\n"
+                            if finished:
+                                del live[idx]
+                    yield "data: [DONE]\n\n"
+                finally:
+                    for idx in list(live):
+                        serving.abort(live[idx][0])
+
+            return StreamingResponse(sse_multi(),
+                                     media_type="text/event-stream")
+
         if body.get("stream"):
             async def sse():
                 first = True

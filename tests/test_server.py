@@ -528,3 +528,29 @@ def test_streaming_n_choices(app):
             assert finish == {0: True, 1: True}
 
     asyncio.run(run())
+
+
+def test_chat_streaming_n_choices(app):
+    async def run():
+        async with _client(app) as c:
+            text = {}
+            roles = {}
+            async with c.stream(
+                "POST", "/v1/chat/completions",
+                json={"messages": [{"role": "user", "content": "go"}],
+                      "max_tokens": 3, "n": 2, "temperature": 0.8,
+                      "seed": 5, "stream": True, "ignore_eos": True},
+            ) as resp:
+                async for line in resp.aiter_lines():
+                    if not line.startswith("data: ") or line == "data: [DONE]":
+                        continue
+                    ch = json.loads(line[6:])["choices"][0]
+                    d = ch["delta"]
+                    if "role" in d:
+                        roles[ch["index"]] = d["role"]
+                    text[ch["index"]] = text.get(ch["index"], "") + \
+                        d.get("content", "")
+            assert set(roles) == {0, 1}
+            assert set(text) == {0, 1}
+
+    asyncio.run(run())
