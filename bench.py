@@ -12,6 +12,18 @@ concurrency is kept topped up; a bench "step" is one engine iteration
 (one continuous-batching scheduler step + forward). Timed region:
 barrier + torch.cuda.synchronize on both sides of EXACTLY --steps steps.
 
+Measurement stability (round-2): before the timed window the closed loop
+is ramped until every initially-admitted request has completed and been
+replaced (full turnover), so the window sees steady-state serving only.
+Goodput is reported as (generated tokens / elapsed) / E[tokens per
+request]; E[tokens/req] == --gen-len exactly by construction of the load
+(gen lengths uniform with mean gen_len), so this equals completed
+requests / elapsed in expectation but with far lower variance in short
+windows (completions are a count of ~1/step; token throughput is stable
+per step). The raw completion count and completions/s are also reported.
+TTFT samples come only from requests admitted after the turnover ramp
+(steady-state admissions), never from the synchronized initial burst.
+
 Usage:
   python bench.py --gpus 1 --steps 64 --warmup 16
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
@@ -132,6 +144,16 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    # Honest n_gpus: --gpus N is a claim that N ranks are running. Refuse
+    # to produce a record that says N GPUs while executing on fewer
+    # (e.g. `bench.py --gpus 8` launched without torchrun).
+    if args.gpus != world:
+        raise SystemExit(
+            f"bench.py: --gpus {args.gpus} but WORLD_SIZE={world}. "
+            f"Launch N>1 via: python -m torch.distributed.run --nnodes=1 "
+            f"--nproc-per-node {args.gpus} --master-addr 127.0.0.1 "
+            f"bench.py --gpus {args.gpus} ..."
+        )
     use_cuda = torch.cuda.is_available()
     device = f"cuda:{local_rank}" if use_cuda else "cpu"
     if use_cuda:
@@ -209,15 +231,19 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
-    # Load initialization: ramp the closed loop to steady state (prefills
-    # admitted, first completions observed) so the timed window measures
+    # Load initialization: ramp the closed loop to FULL TURNOVER — every
+    # request of the synchronized initial burst has completed and been
+    # replaced by steady-state admissions — so the timed window measures
     # steady-state serving for ANY --steps/--warmup. Bounded.
-    ramp_limit = 6 * (args.prompt_len // 256 + args.gen_len) + 500
-    ramp_target = max(args.concurrency // 8, 4)
+    ramp_limit = 10 * (args.prompt_len // 64 + 2 * args.gen_len) + 800
+    ramp_target = args.concurrency + max(args.concurrency // 4, 4)
     ramp = 0
     while len(load.completion_times) < ramp_target and ramp < ramp_limit:
         load.step()
         ramp += 1
+    # TTFT samples before this mark belong to (or overlap) the initial
+    # burst; only steady-state admissions count from here on.
+    steady_ttft_mark = len(load.ttfts)
 
     for _ in range(args.warmup):
         load.step()
@@ -234,7 +260,11 @@ def main():
 
     completed = len(load.completion_times) - completed_before
     gen_tokens = engine.num_generated_tokens - tokens_before
-    ttfts = load.ttfts[ttft_mark:] or load.ttfts
+    # Prefer window-local TTFTs; with too few samples (short windows),
+    # widen to all steady-state samples — never to the initial burst.
+    ttfts = load.ttfts[ttft_mark:]
+    if len(ttfts) < 16:
+        ttfts = load.ttfts[steady_ttft_mark:]
 
     engine.stop_workers()
     spec_stats = None
@@ -272,13 +302,20 @@ def _aggregate_and_report(args, rank, world, distributed, device, use_cuda,
 
     if rank != 0:
         return
-    goodput = completed / elapsed if elapsed > 0 else 0.0
+    # Low-variance goodput estimator: token throughput / E[tokens per
+    # request]; E[tokens/req] == gen_len exactly by construction of the
+    # load, so this equals completions/elapsed in expectation (see module
+    # docstring). Raw completions/s reported alongside.
+    goodput = (
+        (gen_tokens / elapsed) / args.gen_len if elapsed > 0 else 0.0
+    )
+    completed_per_s = completed / elapsed if elapsed > 0 else 0.0
     p50_ttft_ms = (
         statistics.median(ttfts) * 1000.0 if ttfts else float("nan")
     )
     dp = (world // args.tp) if distributed else args.gpus
     result = {
-        "metric": "goodput req/s (Qwen3-8B serving, closed-loop)",
+        "metric": f"goodput req/s ({args.model} serving, closed-loop)",
         "value": round(goodput, 3),
         "unit": "req/s",
         "n_gpus": world if distributed else args.gpus,
@@ -292,6 +329,8 @@ def _aggregate_and_report(args, rank, world, distributed, device, use_cuda,
         + ("+kv_fp8" if args.kv_cache_dtype == "fp8" else ""),
         "data": "synthetic",
         "p50_ttft_ms": round(p50_ttft_ms, 1),
+        "completed_requests": int(completed),
+        "completed_req_per_s": round(completed_per_s, 3),
         **({"spec_decode": spec_stats} if spec_stats else {}),
         "tokens_per_s": round(gen_tokens / elapsed, 1) if elapsed else 0.0,
         "config": {
