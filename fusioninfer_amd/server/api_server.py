@@ -48,26 +48,38 @@ def decode_tokens(token_ids: List[int]) -> str:
         return " ".join(str(t) for t in token_ids)
 
 
+def _find_stop(text: str, stops: List[str]):
+    """Earliest stop-string match (longest on index tie) or None."""
+    best = None
+    for s in stops:
+        i = text.find(s)
+        if i >= 0 and (best is None or i < best[0]
+                       or (i == best[0] and len(s) > len(best[1]))):
+            best = (i, s)
+    return best
+
+
 class StopStreamFilter:
     """Streaming stop-string matcher: holds back the longest possible
     stop prefix so a match that spans chunk boundaries is never partially
-    emitted. push() returns (text safe to emit now, matched?)."""
+    emitted. With keep=True (include_stop_str_in_output) the matched stop
+    string itself is emitted, same as the non-stream path. push() returns
+    (text safe to emit now, matched?)."""
 
-    def __init__(self, stops: List[str]):
+    def __init__(self, stops: List[str], keep: bool = False):
         self.stops = stops
+        self.keep = keep
         self.hold = max((len(s) for s in stops), default=1) - 1
         self.text = ""
         self.sent = 0
 
     def push(self, piece: str):
         self.text += piece
-        hit = min(
-            (i for i in (self.text.find(s) for s in self.stops) if i >= 0),
-            default=-1,
-        )
-        if hit >= 0:
-            emit = self.text[self.sent: hit]
-            self.sent = hit
+        best = _find_stop(self.text, self.stops)
+        if best is not None:
+            end = best[0] + len(best[1]) if self.keep else best[0]
+            emit = self.text[self.sent: end]
+            self.sent = end
             return emit, True
         safe = max(len(self.text) - self.hold, self.sent)
         emit = self.text[self.sent: safe]
@@ -247,24 +259,35 @@ def build_app(serving: ServingEngine, model_name: str,
         """Drain a request's token stream. With OpenAI `stop` strings the
         generated text is truncated BEFORE the first stop match (AFTER it
         with include_stop_str_in_output) and the engine request is
-        aborted. Returns (token_ids, text)."""
+        aborted. Text accumulates through incremental detokenization so
+        multi-byte characters split across BPE tokens decode correctly.
+        Returns (token_ids, text)."""
+        from fusioninfer_amd.tokenizer import IncrementalDetokenizer
+
         loop = asyncio.get_event_loop()
         toks: List[int] = []
         text = ""
+        detok = IncrementalDetokenizer(tokenizer) if stops else None
         while True:
             tok, finished = await loop.run_in_executor(None, q.get)
             if tok is not None:
                 toks.append(tok)
                 if stops:
-                    text += tokenizer.decode_one(tok)
-                    for s in stops:
-                        i = text.find(s)
-                        if i >= 0:
-                            serving.abort(req_id)
-                            end = i + len(s) if keep_stop else i
-                            return toks, text[:end]
+                    text += detok.push(tok)
+                    hit = _find_stop(text, stops)
+                    if hit is not None:
+                        serving.abort(req_id)
+                        end = hit[0] + len(hit[1]) if keep_stop else hit[0]
+                        return toks, text[:end]
             if finished:
-                return toks, text if stops else tokenizer.decode(toks)
+                if not stops:
+                    return toks, tokenizer.decode(toks)
+                text += detok.flush()
+                hit = _find_stop(text, stops)
+                if hit is not None:
+                    end = hit[0] + len(hit[1]) if keep_stop else hit[0]
+                    return toks, text[:end]
+                return toks, text
 
     async def _stream(q) -> AsyncGenerator:
         loop = asyncio.get_event_loop()
@@ -358,6 +381,7 @@ def build_app(serving: ServingEngine, model_name: str,
         if lora not in serving.engine.active_loras():
             lora = None
         stops = _parse_stops(body)
+        keep_stop = bool(body.get("include_stop_str_in_output"))
         pd_tag = request.headers.get("x-pd-tag")
         try:
             if pd_tag is not None:
@@ -395,16 +419,26 @@ def build_app(serving: ServingEngine, model_name: str,
                 s_i = _sampling_from(body)
                 if s_i.seed is not None:
                     s_i.seed += i
-                extra.append(serving.submit(prompt_ids, s_i, lora_name=lora))
+                extra.append(serving.submit(
+                    prompt_ids, s_i, lora_name=lora,
+                    priority=int(body.get("priority", 0)),
+                ))
             chans = [(0, req_id, q)] + [
                 (i + 1, rid_i, q_i) for i, (rid_i, q_i) in enumerate(extra)
             ]
 
             async def sse_multi():
+                from fusioninfer_amd.tokenizer import IncrementalDetokenizer
+
                 loop = asyncio.get_event_loop()
                 live = {idx: (rid, qq) for idx, rid, qq in chans}
                 filts = {
-                    idx: (StopStreamFilter(stops) if stops else None)
+                    idx: (StopStreamFilter(stops, keep_stop)
+                          if stops else None)
+                    for idx, _, _ in chans
+                }
+                detoks = {
+                    idx: IncrementalDetokenizer(tokenizer)
                     for idx, _, _ in chans
                 }
                 n_out = 0
@@ -416,13 +450,15 @@ def build_app(serving: ServingEngine, model_name: str,
                                 None, qq.get
                             )
                             delta = (
-                                tokenizer.decode_one(tok)
+                                detoks[idx].push(tok)
                                 if tok is not None else ""
                             )
+                            if finished:
+                                delta += detoks[idx].flush()
                             if tok is not None:
                                 n_out += 1
                             f = filts[idx]
-                            if f is not None and tok is not None:
+                            if f is not None:
                                 delta, hit = f.push(delta)
                                 if hit:
                                     finished = True
@@ -452,10 +488,10 @@ def build_app(serving: ServingEngine, model_name: str,
                             "created": created, "model": model_name,
                             "choices": [],
                             "usage": {
-                                "prompt_tokens": len(prompt_ids) * n_req,
+                                "prompt_tokens": len(prompt_ids),
                                 "completion_tokens": n_out,
                                 "total_tokens":
-                                    len(prompt_ids) * n_req + n_out,
+                                    len(prompt_ids) + n_out,
                             },
                         }) + "\n\n"
                     yield "data: [DONE]\n\n"
@@ -468,9 +504,12 @@ def build_app(serving: ServingEngine, model_name: str,
 
         if body.get("stream"):
             async def sse():
+                from fusioninfer_amd.tokenizer import IncrementalDetokenizer
+
                 done = False
                 n_out = 0
-                filt = StopStreamFilter(stops) if stops else None
+                filt = StopStreamFilter(stops, keep_stop) if stops else None
+                detok = IncrementalDetokenizer(tokenizer)
                 try:
                     if echo:
                         first = {
@@ -486,8 +525,10 @@ def build_app(serving: ServingEngine, model_name: str,
                         if tok is not None:
                             n_out += 1
                         done = finished
-                        delta = tokenizer.decode_one(tok) if tok is not None else ""
-                        if filt is not None and tok is not None:
+                        delta = detok.push(tok) if tok is not None else ""
+                        if finished:
+                            delta += detok.flush()
+                        if filt is not None:
                             # hold back text that could extend into a stop
                             # string; cut the stream at the first match
                             delta, hit = filt.push(delta)
@@ -546,7 +587,6 @@ def build_app(serving: ServingEngine, model_name: str,
                 prompt_ids, s_i, lora_name=lora,
                 priority=int(body.get("priority", 0)),
             ))
-        keep_stop = bool(body.get("include_stop_str_in_output"))
         choices = []
         try:
             toks, text = await _collect(q, stops=stops, req_id=req_id,
@@ -616,11 +656,15 @@ def build_app(serving: ServingEngine, model_name: str,
                                            "type": "invalid_request_error"}},
                                 400)
         stops = _parse_stops(body)
+        keep_stop = bool(body.get("include_stop_str_in_output"))
         lora = body.get("model")
         if lora not in serving.engine.active_loras():
             lora = None
         try:
-            req_id, q = serving.submit(prompt_ids, sampling, lora_name=lora)
+            req_id, q = serving.submit(
+                prompt_ids, sampling, lora_name=lora,
+                priority=int(body.get("priority", 0)),
+            )
         except ValueError as e:
             return JSONResponse({"error": {"message": str(e),
                                            "type": "invalid_request_error"}}, 400)
@@ -643,11 +687,18 @@ def build_app(serving: ServingEngine, model_name: str,
             ]
 
             async def sse_multi():
+                from fusioninfer_amd.tokenizer import IncrementalDetokenizer
+
                 loop = asyncio.get_event_loop()
                 live = {idx: (rid, qq) for idx, rid, qq in chans}
                 firsts = {idx: True for idx, _, _ in chans}
                 filts = {
-                    idx: (StopStreamFilter(stops) if stops else None)
+                    idx: (StopStreamFilter(stops, keep_stop)
+                          if stops else None)
+                    for idx, _, _ in chans
+                }
+                detoks = {
+                    idx: IncrementalDetokenizer(tokenizer)
                     for idx, _, _ in chans
                 }
                 try:
@@ -662,11 +713,13 @@ def build_app(serving: ServingEngine, model_name: str,
                                 delta["role"] = "assistant"
                                 firsts[idx] = False
                             piece = (
-                                tokenizer.decode_one(tok)
+                                detoks[idx].push(tok)
                                 if tok is not None else ""
                             )
+                            if finished:
+                                piece += detoks[idx].flush()
                             f = filts[idx]
-                            if f is not None and tok is not None:
+                            if f is not None:
                                 piece, hit = f.push(piece)
                                 if hit:
                                     finished = True
@@ -700,9 +753,12 @@ def build_app(serving: ServingEngine, model_name: str,
 
         if body.get("stream"):
             async def sse():
+                from fusioninfer_amd.tokenizer import IncrementalDetokenizer
+
                 first = True
                 done = False
-                filt = StopStreamFilter(stops) if stops else None
+                filt = StopStreamFilter(stops, keep_stop) if stops else None
+                detok = IncrementalDetokenizer(tokenizer)
                 try:
                     async for tok, finished in _stream(q):
                         done = finished
@@ -711,10 +767,12 @@ def build_app(serving: ServingEngine, model_name: str,
                             delta["role"] = "assistant"
                             first = False
                         piece = (
-                            tokenizer.decode_one(tok)
+                            detok.push(tok)
                             if tok is not None else ""
                         )
-                        if filt is not None and tok is not None:
+                        if finished:
+                            piece += detok.flush()
+                        if filt is not None:
                             # cut the stream at the first stop-string match
                             piece, hit = filt.push(piece)
                             if hit:
@@ -758,14 +816,17 @@ def build_app(serving: ServingEngine, model_name: str,
             s_i = _sampling_from(body)
             if s_i.seed is not None:
                 s_i.seed += i
-            extra.append(serving.submit(prompt_ids, s_i, lora_name=lora))
-        keep = bool(body.get("include_stop_str_in_output"))
+            extra.append(serving.submit(
+                prompt_ids, s_i, lora_name=lora,
+                priority=int(body.get("priority", 0)),
+            ))
         try:
             choices = [await _collect(q, stops=stops, req_id=req_id,
-                                      keep_stop=keep)]
+                                      keep_stop=keep_stop)]
             for rid_i, q_i in extra:
                 choices.append(await _collect(q_i, stops=stops,
-                                              req_id=rid_i, keep_stop=keep))
+                                              req_id=rid_i,
+                                              keep_stop=keep_stop))
         except asyncio.CancelledError:
             serving.abort(req_id)
             for rid_i, _ in extra:

@@ -147,6 +147,54 @@ class HFTokenizer:
         return self._tok.decode([token_id], skip_special_tokens=False)
 
 
+class IncrementalDetokenizer:
+    """Streaming detokenization that never splits a multi-byte UTF-8
+    character across deltas.
+
+    decode_one() decodes each token independently, so a character whose
+    bytes span two BPE tokens (CJK/emoji under byte-level BPE) comes out
+    as U+FFFD. This class instead decodes a sliding window of recent ids
+    and emits only the stable suffix: while the window's decode ends in
+    U+FFFD (an incomplete byte sequence) the text is held back until the
+    completing token arrives. Mirrors the reference engines' incremental
+    detokenization contract (the vLLM images the reference launches,
+    SURVEY.md §2.3).
+    """
+
+    #: tokens of context kept before the unread window (byte-level BPE
+    #: decoders are concatenative, but sentencepiece-style decoders join
+    #: with context; a few tokens of prefix keeps deltas exact)
+    _CTX = 6
+
+    def __init__(self, tok, initial_ids=()):
+        self.tok = tok
+        self.ids: List[int] = list(initial_ids)
+        self.prefix_offset = max(len(self.ids) - self._CTX, 0)
+        self.read_offset = len(self.ids)
+
+    def push(self, token_id: int) -> str:
+        """Add one generated token; return the newly-stable text ("" if
+        the tail is still an incomplete character)."""
+        self.ids.append(int(token_id))
+        prefix = self.tok.decode(self.ids[self.prefix_offset:self.read_offset])
+        full = self.tok.decode(self.ids[self.prefix_offset:])
+        if len(full) > len(prefix) and not full.endswith("�"):
+            delta = full[len(prefix):]
+            self.prefix_offset = max(len(self.ids) - self._CTX, 0)
+            self.read_offset = len(self.ids)
+            return delta
+        return ""
+
+    def flush(self) -> str:
+        """Emit any held-back tail (stream ended mid-character: the
+        replacement char is then the honest output)."""
+        prefix = self.tok.decode(self.ids[self.prefix_offset:self.read_offset])
+        full = self.tok.decode(self.ids[self.prefix_offset:])
+        self.read_offset = len(self.ids)
+        self.prefix_offset = max(len(self.ids) - self._CTX, 0)
+        return full[len(prefix):]
+
+
 Tok = Union[ByteTokenizer, HFTokenizer]
 
 

@@ -554,3 +554,75 @@ def test_chat_streaming_n_choices(app):
             assert set(text) == {0, 1}
 
     asyncio.run(run())
+
+
+def test_stop_strings_streaming_keep(app):
+    """include_stop_str_in_output applies to streams too: the
+    concatenated deltas END WITH the stop string (same request param,
+    same behavior as the non-stream path — ADVICE round-1)."""
+
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "xyzxyz", "max_tokens": 8,
+                      "temperature": 0, "ignore_eos": True},
+            )
+            full = r.json()["choices"][0]["text"]
+            stop_ch = full[1]
+            text = ""
+            async with c.stream("POST", "/v1/completions", json={
+                "prompt": "xyzxyz", "max_tokens": 8, "stream": True,
+                "stop": stop_ch, "temperature": 0, "ignore_eos": True,
+                "include_stop_str_in_output": True,
+            }) as resp:
+                async for line in resp.aiter_lines():
+                    if line.startswith("data: ") and "[DONE]" not in line:
+                        chunk = json.loads(line[6:])
+                        text += chunk["choices"][0]["text"]
+            assert text == full.split(stop_ch)[0] + stop_ch
+
+            # n>1 interleaved stream honors it per choice
+            texts = {0: "", 1: ""}
+            async with c.stream("POST", "/v1/completions", json={
+                "prompt": "xyzxyz", "max_tokens": 8, "stream": True, "n": 2,
+                "stop": stop_ch, "temperature": 0, "ignore_eos": True,
+                "include_stop_str_in_output": True,
+            }) as resp:
+                async for line in resp.aiter_lines():
+                    if line.startswith("data: ") and "[DONE]" not in line:
+                        chunk = json.loads(line[6:])
+                        ch = chunk["choices"][0]
+                        texts[ch["index"]] += ch["text"]
+            for v in texts.values():
+                assert v.endswith(stop_ch)
+                assert stop_ch not in v[:-1]
+
+    asyncio.run(run())
+
+
+def test_streaming_usage_not_inflated(app):
+    """Multi-choice stream usage counts the shared prompt ONCE
+    (OpenAI/vLLM semantics — ADVICE round-1)."""
+
+    async def run():
+        async with _client(app) as c:
+            usage = None
+            n_toks = 0
+            async with c.stream("POST", "/v1/completions", json={
+                "prompt": "hello", "max_tokens": 4, "stream": True, "n": 2,
+                "temperature": 0, "ignore_eos": True,
+                "stream_options": {"include_usage": True},
+            }) as resp:
+                async for line in resp.aiter_lines():
+                    if line.startswith("data: ") and "[DONE]" not in line:
+                        chunk = json.loads(line[6:])
+                        if chunk.get("usage"):
+                            usage = chunk["usage"]
+                        for ch in chunk.get("choices", []):
+                            n_toks += len(ch.get("token_ids", []))
+            assert usage is not None
+            assert usage["prompt_tokens"] == 5  # len("hello") bytes, once
+            assert usage["completion_tokens"] == n_toks == 8
+
+    asyncio.run(run())

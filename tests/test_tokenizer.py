@@ -151,3 +151,52 @@ def test_chat_template_rendering(tok_file, tmp_path):
     import pytest as _pytest
     with _pytest.raises(ValueError):
         t.apply_chat_template([])
+
+
+def test_incremental_detokenizer_multibyte(tok_file):
+    """A character whose UTF-8 bytes span several BPE tokens must stream
+    out whole, never as U+FFFD fragments (ADVICE round-1 medium)."""
+    from fusioninfer_amd.tokenizer import IncrementalDetokenizer
+
+    t = HFTokenizer(tok_file, vocab_size=1024)
+    text = "héllo 中文 🙂 ok"
+    ids = t.encode(text)
+    assert len(ids) > len(text)  # multi-byte chars really did split
+    detok = IncrementalDetokenizer(t)
+    out = ""
+    for i in ids:
+        piece = detok.push(i)
+        assert "�" not in piece
+        out += piece
+    out += detok.flush()
+    assert out == text
+    # per-token independent decode WOULD have mangled it
+    assert "�" in "".join(t.decode_one(i) for i in ids)
+
+
+def test_incremental_detokenizer_byte_fallback():
+    from fusioninfer_amd.tokenizer import IncrementalDetokenizer
+
+    t = ByteTokenizer(1024)
+    text = "中🙂a"
+    ids = t.encode(text)
+    detok = IncrementalDetokenizer(t)
+    out = ""
+    for i in ids:
+        piece = detok.push(i)
+        assert "�" not in piece
+        out += piece
+    out += detok.flush()
+    assert out == text
+
+
+def test_incremental_detokenizer_flush_mid_char():
+    """Stream ending inside a character: flush emits the honest
+    replacement char instead of dropping the tail."""
+    from fusioninfer_amd.tokenizer import IncrementalDetokenizer
+
+    t = ByteTokenizer(1024)
+    ids = t.encode("中")[:2]  # 2 of 3 bytes
+    detok = IncrementalDetokenizer(t)
+    assert "".join(detok.push(i) for i in ids) == ""
+    assert "�" in detok.flush()
