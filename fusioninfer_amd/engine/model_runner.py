@@ -608,8 +608,8 @@ class ModelRunner:
             return
         if ps.pp_world_size() > 1:
             return  # p2p sends inside capture: PP decode runs eager
-        if self.cfg.model.is_moe:
-            return  # MoE routing has data-dependent shapes: eager decode
+        # MoE decode captures too: the grouped-GEMM path is static-shaped
+        # (device-side block alignment, static grid + device tile count)
         max_bs = min(self.cfg.scheduler.max_num_seqs, _DECODE_BUCKETS[-1])
         buckets = [b for b in _DECODE_BUCKETS if b <= max_bs]
         if not self._static:

@@ -298,6 +298,11 @@ def load_hf_state_dict(
     assert not moe_pending, (
         f"unfused expert weights remain: {list(moe_pending)[:4]}"
     )
+    # .data mutations don't bump tensor._version — explicitly invalidate
+    # the MFMA-packed expert-weight cache of every MoE layer
+    for layer in model.layers:
+        if hasattr(layer.mlp, "invalidate_packed"):
+            layer.mlp.invalidate_packed()
     return loaded
 
 

@@ -302,4 +302,50 @@ def prefill_attention_paged(
     )
 
 
+# ---------------------------------------------------------------- MoE
+
+def pack_moe_weights(w: torch.Tensor) -> torch.Tensor:
+    """Pack per-expert weights [E, K, N] into MFMA B-fragment order
+    [E, K/32, N/16, 64, 8] for the grouped GEMM: lane l of a wave reads
+    its 8 contiguous bf16 B elements (B[k=(l>>4)*8+e][n=l&15] within a
+    32x16 k-by-n subtile) with one 16 B load — one fully-coalesced 1 KiB
+    transaction per wave per K-step."""
+    E, K, N = w.shape
+    assert K % 32 == 0 and N % 16 == 0, (K, N)
+    return (
+        w.view(E, K // 32, 4, 8, N // 16, 16)
+        .permute(0, 1, 4, 2, 5, 3)
+        .reshape(E, K // 32, N // 16, 64, 8)
+        .contiguous()
+    )
+
+
+def moe_gemm(out, a, b_packed, sorted_ids, expert_ids, n_valid,
+             block_m: int, gate_up: bool):
+    """Grouped GEMM over block-aligned expert segments.
+
+    out [PM, N] bf16; a = x [T, K] (gate_up, rows gathered via sorted_ids)
+    or act [PM, K]; b_packed from pack_moe_weights ([E, K/32, NB/16, 64, 8]
+    with NB = 2N when gate_up). n_valid: device int32 scalar tensor with
+    the real m-tile count (static grid, dynamic work — hipGraph-safe).
+    gate_up=True fuses the SwiGLU epilogue: out = silu(gate) * up."""
+    if out.is_cuda:
+        _require_native()
+        _C.moe_gemm(out, a, b_packed, sorted_ids, expert_ids, n_valid,
+                    block_m, gate_up)
+        return out
+    return ref.moe_gemm(out, a, b_packed, sorted_ids, expert_ids, n_valid,
+                        block_m, gate_up)
+
+
+def moe_combine(out, y, pos, w):
+    """out[t] = sum_k w[t,k] * y[pos[t,k]] (pos < 0 skipped). Deterministic
+    (no atomics) so token-exact tests stay reproducible."""
+    if out.is_cuda:
+        _require_native()
+        _C.moe_combine(out, y, pos, w)
+        return out
+    return ref.moe_combine(out, y, pos, w)
+
+
 compute_cos_sin_cache = ref.compute_cos_sin_cache

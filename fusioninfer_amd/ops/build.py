@@ -26,6 +26,7 @@ SOURCES = [
     "cache.hip",
     "paged_attention.hip",
     "prefill_attention.hip",
+    "moe_gemm.hip",
 ]
 
 

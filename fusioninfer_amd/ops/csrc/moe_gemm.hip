@@ -1,0 +1,202 @@
+// Grouped MFMA GEMM for MoE expert compute (gfx950).
+//
+// Replaces the round-1 capacity-padded torch.bmm experts (and their
+// per-layer `counts.max()` host sync) with a vLLM-style block-aligned
+// grouped GEMM: token assignments are sorted by expert and each expert's
+// segment is padded to a multiple of BLOCK_M on the DEVICE, so every
+// m-tile belongs to exactly one expert and all launch shapes are static
+// — the whole MoE decode step becomes hipGraph-capturable.
+// (Capability parity: expert serving the reference invokes via vLLM
+// images; reference publishes no kernels of its own.)
+//
+// Design (per /opt/skills/guides/cdna_hip_programming.md):
+//  * mfma_f32_16x16x32_bf16 per-wave tiles; expert weights are packed
+//    OFFLINE into MFMA B-fragment order [E][K/32][N/16][64][8] so each
+//    wave's per-K-step B load is ONE fully-coalesced 16 B/lane
+//    global_load_dwordx4 (1 KiB per wave per fragment).
+//  * No LDS, no barriers: weights stream once (the decode regime is
+//    weight-bandwidth-bound: every active expert's panel is read ~once
+//    per step), activations are L2-resident; per the guide's GEMV row,
+//    operands streamed once and not shared across waves go straight to
+//    VGPRs with a deep unroll.
+//  * GEMM1 fuses the SwiGLU epilogue: each wave accumulates the gate
+//    AND up columns for its 16-column slice and writes silu(g)*u —
+//    the [T, 2I] intermediate never round-trips HBM.
+//  * Blocks beyond the real (data-dependent) tile count read a device
+//    scalar and exit: grid size is static, work is dynamic.
+//
+// C-fragment map (16x16): row = (lane>>4)*4 + r, col = lane&15.
+// A-fragment: lane holds A[row = lane&15][k = (lane>>4)*8 + e], e<8.
+// B-fragment: lane holds B[k = (lane>>4)*8 + e][col = lane&15].
+
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+namespace fi {
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float floatx4;
+
+FI_DEV short8 load_b16x8(const u16* p) {
+  return *reinterpret_cast<const short8*>(p);
+}
+
+// WM x WN waves (4 total), MITER 16-row fragments per wave.
+// Block tile: rows BM = 16*WM*MITER, cols = 16*WN of the OUTPUT space.
+// GATE_UP: B is packed over 2N columns (gate at n, up at n + N);
+// epilogue writes silu(gate)*up. Otherwise a plain grouped GEMM.
+template <int WM, int WN, int MITER, bool GATE_UP>
+__global__ __launch_bounds__(256) void moe_gemm_kernel(
+    u16* __restrict__ out,              // [PM, N] bf16
+    const u16* __restrict__ a,          // GATE_UP: x [T, K]; else act [PM, K]
+    const u16* __restrict__ b,          // [E][K/32][NB/16][64][8] bf16
+    const int* __restrict__ sorted_ids, // [PM] token row per padded slot
+    const int* __restrict__ expert_ids, // [PM/BM] local expert per m-tile
+    const int* __restrict__ n_valid,    // device scalar: real m-tile count
+    const int K, const int N) {
+  constexpr int BM = 16 * WM * MITER;
+  const int mtile = blockIdx.x;
+  if (mtile >= *n_valid) return;
+
+  const int lane = threadIdx.x % kWaveSize;
+  const int wave = threadIdx.x / kWaveSize;
+  const int wm = wave / WN;
+  const int wn = wave % WN;
+  const int nt = blockIdx.y * WN + wn;  // 16-col fragment index in [0, N/16)
+  const int e = expert_ids[mtile];
+  const int row0 = mtile * BM + wm * (16 * MITER);
+
+  const int NB16 = (GATE_UP ? 2 * N : N) / 16;  // packed B n-fragments
+  const u16* b_e =
+      b + static_cast<int64_t>(e) * (K / 32) * NB16 * (64 * 8);
+  // per-K-step fragment pointers (advance by NB16*512 elements per kt)
+  const u16* bg_p = b_e + (static_cast<int64_t>(nt) * 64 + lane) * 8;
+  const u16* bu_p =
+      GATE_UP ? b_e + ((static_cast<int64_t>(nt) + N / 16) * 64 + lane) * 8
+              : nullptr;
+  const int64_t b_step = static_cast<int64_t>(NB16) * 64 * 8;
+
+  // A row pointers: gather through sorted_ids for GEMM1 (padding slots
+  // carry a valid dummy row — garbage rows are never read by combine)
+  const u16* a_p[MITER];
+#pragma unroll
+  for (int mi = 0; mi < MITER; ++mi) {
+    const int slot = row0 + mi * 16 + (lane & 15);
+    const int row = GATE_UP ? sorted_ids[slot] : slot;
+    a_p[mi] = a + static_cast<int64_t>(row) * K + (lane >> 4) * 8;
+  }
+
+  floatx4 acc_g[MITER];
+  floatx4 acc_u[GATE_UP ? MITER : 1];
+#pragma unroll
+  for (int mi = 0; mi < MITER; ++mi) {
+    acc_g[mi] = {0.f, 0.f, 0.f, 0.f};
+    if (GATE_UP) acc_u[mi] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int ksteps = K / 32;
+  for (int kt = 0; kt < ksteps; ++kt) {
+    const short8 bg = load_b16x8(bg_p + kt * b_step);
+    short8 bu;
+    if (GATE_UP) bu = load_b16x8(bu_p + kt * b_step);
+#pragma unroll
+    for (int mi = 0; mi < MITER; ++mi) {
+      const short8 av = load_b16x8(a_p[mi] + kt * 32);
+      acc_g[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          av, bg, acc_g[mi], 0, 0, 0);
+      if (GATE_UP)
+        acc_u[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            av, bu, acc_u[mi], 0, 0, 0);
+    }
+  }
+
+  const int col = nt * 16 + (lane & 15);
+#pragma unroll
+  for (int mi = 0; mi < MITER; ++mi) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int orow = row0 + mi * 16 + (lane >> 4) * 4 + r;
+      float v;
+      if (GATE_UP) {
+        const float g = acc_g[mi][r];
+        const float u = acc_u[mi][r];
+        v = (g / (1.f + __expf(-g))) * u;
+      } else {
+        v = acc_g[mi][r];
+      }
+      out[static_cast<int64_t>(orow) * N + col] = f32_to_bf16(v);
+    }
+  }
+}
+
+// out[t, :] = sum_k w[t,k] * y[pos[t,k], :]   (pos < 0 -> non-local expert
+// or padding: skipped). Deterministic — no atomics, so token-exact TP/PD
+// tests stay reproducible.
+__global__ void moe_combine_kernel(u16* __restrict__ out,
+                                   const u16* __restrict__ y,
+                                   const int* __restrict__ pos,
+                                   const float* __restrict__ w,
+                                   const int topk, const int H) {
+  const int t = blockIdx.y;
+  const int c0 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c0 + 8 > H) return;
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (int k = 0; k < topk; ++k) {
+    const int p = pos[t * topk + k];
+    if (p < 0) continue;
+    const float wk = w[t * topk + k];
+    const bf16x8 v = *reinterpret_cast<const bf16x8*>(
+        y + static_cast<int64_t>(p) * H + c0);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) acc[i] += wk * bf16_to_f32(v.h[i]);
+  }
+  bf16x8 o;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) o.h[i] = f32_to_bf16(acc[i]);
+  *reinterpret_cast<bf16x8*>(out + static_cast<int64_t>(t) * H + c0) = o;
+}
+
+}  // namespace
+
+// block_m 16: decode regime (few rows per expert; WM=1,WN=4,MITER=1).
+// block_m 128: prefill regime (WM=2,WN=2,MITER=4 — 4x in-register B reuse).
+void launch_moe_gemm(u16* out, const u16* a, const u16* b,
+                     const int* sorted_ids, const int* expert_ids,
+                     const int* n_valid, int max_mtiles, int K, int N,
+                     int block_m, bool gate_up, hipStream_t stream) {
+  if (block_m == 16) {
+    const dim3 grid(max_mtiles, N / 64);
+    if (gate_up)
+      hipLaunchKernelGGL((moe_gemm_kernel<1, 4, 1, true>), grid, dim3(256), 0,
+                         stream, out, a, b, sorted_ids, expert_ids, n_valid,
+                         K, N);
+    else
+      hipLaunchKernelGGL((moe_gemm_kernel<1, 4, 1, false>), grid, dim3(256),
+                         0, stream, out, a, b, sorted_ids, expert_ids,
+                         n_valid, K, N);
+  } else {  // block_m == 128
+    const dim3 grid(max_mtiles, N / 32);
+    if (gate_up)
+      hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, true>), grid, dim3(256), 0,
+                         stream, out, a, b, sorted_ids, expert_ids, n_valid,
+                         K, N);
+    else
+      hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, false>), grid, dim3(256),
+                         0, stream, out, a, b, sorted_ids, expert_ids,
+                         n_valid, K, N);
+  }
+}
+
+void launch_moe_combine(u16* out, const u16* y, const int* pos,
+                        const float* w, int tokens, int topk, int H,
+                        hipStream_t stream) {
+  const int threads = 64;
+  const dim3 grid(ceil_div(H / 8, threads), tokens);
+  hipLaunchKernelGGL(moe_combine_kernel, grid, dim3(threads), 0, stream, out,
+                     y, pos, w, topk, H);
+}
+
+}  // namespace fi

@@ -33,6 +33,10 @@ void launch_prefill_attn(u16*, const u16*, const void*, const void*,
                          const int*, const int*, const int*, const int*,
                          const int*, int, int, int64_t, int64_t, int64_t, int,
                          int, int, float, bool, hipStream_t);
+void launch_moe_gemm(u16*, const u16*, const u16*, const int*, const int*,
+                     const int*, int, int, int, int, bool, hipStream_t);
+void launch_moe_combine(u16*, const u16*, const int*, const float*, int, int,
+                        int, hipStream_t);
 
 }  // namespace fi
 
@@ -314,6 +318,53 @@ void prefill_attention_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       static_cast<float>(scale), fp8, current_stream());
 }
 
+void moe_gemm(at::Tensor out, at::Tensor a, at::Tensor b_packed,
+              at::Tensor sorted_ids, at::Tensor expert_ids,
+              at::Tensor n_valid, int64_t block_m, bool gate_up) {
+  CHECK_BF16_CUDA(out);
+  CHECK_BF16_CUDA(a);
+  CHECK_BF16_CUDA(b_packed);
+  TORCH_CHECK(out.is_contiguous() && a.is_contiguous() &&
+              b_packed.is_contiguous());
+  TORCH_CHECK(sorted_ids.scalar_type() == at::kInt &&
+              expert_ids.scalar_type() == at::kInt &&
+              n_valid.scalar_type() == at::kInt);
+  TORCH_CHECK(block_m == 16 || block_m == 128, "block_m must be 16 or 128");
+  const int K = a.size(1);
+  const int N = out.size(1);
+  const int PM = out.size(0);
+  TORCH_CHECK(PM % block_m == 0, "padded rows must be block_m-aligned");
+  TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32");
+  TORCH_CHECK(N % (block_m == 16 ? 64 : 32) == 0, "N tile misalignment");
+  // packed B: [E, K/32, NB/16, 64, 8]
+  TORCH_CHECK(b_packed.dim() == 5 && b_packed.size(1) == K / 32 &&
+              b_packed.size(2) == (gate_up ? 2 * N : N) / 16 &&
+              b_packed.size(3) == 64 && b_packed.size(4) == 8,
+              "b_packed layout mismatch");
+  TORCH_CHECK(expert_ids.size(0) >= PM / block_m);
+  if (gate_up) TORCH_CHECK(sorted_ids.size(0) >= PM);
+  fi::launch_moe_gemm(bf16_ptr(out), bf16_cptr(a),
+                      bf16_cptr(b_packed), sorted_ids.data_ptr<int>(),
+                      expert_ids.data_ptr<int>(), n_valid.data_ptr<int>(),
+                      PM / block_m, K, N, static_cast<int>(block_m), gate_up,
+                      current_stream());
+}
+
+void moe_combine(at::Tensor out, at::Tensor y, at::Tensor pos, at::Tensor w) {
+  CHECK_BF16_CUDA(out);
+  CHECK_BF16_CUDA(y);
+  TORCH_CHECK(out.is_contiguous() && y.is_contiguous());
+  TORCH_CHECK(pos.scalar_type() == at::kInt && pos.is_contiguous());
+  TORCH_CHECK(w.scalar_type() == at::kFloat && w.is_contiguous());
+  const int T = out.size(0);
+  const int H = out.size(1);
+  TORCH_CHECK(y.size(1) == H && H % 8 == 0);
+  const int topk = pos.numel() / T;
+  TORCH_CHECK(pos.numel() == T * topk && w.numel() == T * topk);
+  fi::launch_moe_combine(bf16_ptr(out), bf16_cptr(y), pos.data_ptr<int>(),
+                         w.data_ptr<float>(), T, topk, H, current_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -337,4 +388,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("prefill_attention", &prefill_attention, "varlen causal MFMA prefill");
   m.def("prefill_attention_paged", &prefill_attention_paged,
         "varlen causal MFMA prefill over the paged cache (context attention)");
+  m.def("moe_gemm", &moe_gemm,
+        "grouped MFMA GEMM over block-aligned expert segments "
+        "(gate_up=true fuses the SwiGLU epilogue)");
+  m.def("moe_combine", &moe_combine,
+        "weighted top-k combine of expert outputs (deterministic)");
 }

@@ -285,3 +285,54 @@ def scatter_kv_blocks(
     idx = block_ids.long()
     k_cache[idx] = staging[0].to(k_cache.dtype)
     v_cache[idx] = staging[1].to(v_cache.dtype)
+
+
+# ---------------------------------------------------------------- MoE
+
+def unpack_moe_weights(b_packed: torch.Tensor) -> torch.Tensor:
+    """Inverse of ops.pack_moe_weights: [E, K/32, N/16, 64, 8] -> [E, K, N]."""
+    E, K32, N16, _, _ = b_packed.shape
+    return (
+        b_packed.view(E, K32, N16, 4, 16, 8)
+        .permute(0, 1, 3, 5, 2, 4)
+        .reshape(E, K32 * 32, N16 * 16)
+        .contiguous()
+    )
+
+
+def moe_gemm(out, a, b_packed, sorted_ids, expert_ids, n_valid,
+             block_m: int, gate_up: bool):
+    """fp32 reference of the grouped GEMM (CPU; same semantics as the HIP
+    kernel incl. padding rows computed from the dummy token row)."""
+    w = unpack_moe_weights(b_packed).float()
+    n_tiles = int(n_valid.item())
+    N = out.shape[1]
+    for mt in range(n_tiles):
+        e = int(expert_ids[mt])
+        r0 = mt * block_m
+        rows = slice(r0, r0 + block_m)
+        if gate_up:
+            src = a[sorted_ids[rows].long()].float()
+        else:
+            src = a[rows].float()
+        acc = src @ w[e]
+        if gate_up:
+            g, u = acc[:, :N], acc[:, N:]
+            acc = torch.nn.functional.silu(g) * u
+        out[rows] = acc.to(out.dtype)
+    return out
+
+
+def moe_combine(out, y, pos, w):
+    T, H = out.shape
+    topk = pos.numel() // T
+    acc = torch.zeros(T, H, dtype=torch.float32)
+    yv = y.float()
+    for t in range(T):
+        for k in range(topk):
+            p = int(pos[t * topk + k])
+            if p < 0:
+                continue
+            acc[t] += float(w[t * topk + k]) * yv[p]
+    out.copy_(acc.to(out.dtype))
+    return out
