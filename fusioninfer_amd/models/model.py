@@ -338,7 +338,12 @@ class MoEMLP(nn.Module):
         key = torch.where(local, flat - self.e_start,
                           torch.full_like(flat, E_local))
         order = torch.argsort(key, stable=True)
-        counts = torch.bincount(key, minlength=E_local + 1)[:E_local]
+        # scatter_add, not bincount: bincount is hipGraph-capture-unsafe
+        # (hipErrorStreamCaptureUnsupported on ROCm 7.2)
+        counts = torch.zeros(E_local + 1, dtype=torch.long,
+                             device=dev).scatter_add_(
+            0, key, torch.ones_like(key)
+        )[:E_local]
         tiles = (counts + block_m - 1) // block_m
         pad_starts = (tiles.cumsum(0) - tiles) * block_m
         seg_starts = torch.cat(
