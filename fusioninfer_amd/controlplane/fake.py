@@ -140,7 +140,6 @@ class FakeClient:
                 )
             stored = copy.deepcopy(obj)
             md = stored["metadata"]
-            md["resourceVersion"] = str(next(self._rv))
             md.setdefault("uid", current["metadata"].get("uid"))
             # generation bumps only when spec changed (k8s semantics)
             old_gen = int(current["metadata"].get("generation", 1))
@@ -151,6 +150,13 @@ class FakeClient:
             # preserve status unless explicitly set
             if "status" not in stored and "status" in current:
                 stored["status"] = copy.deepcopy(current["status"])
+            # no-op writes keep resourceVersion and emit no event (k8s
+            # semantics — what lets the level-triggered watch loop
+            # converge instead of reconciling its own identical writes)
+            md["resourceVersion"] = current["metadata"]["resourceVersion"]
+            if stored == current:
+                return copy.deepcopy(current)
+            md["resourceVersion"] = str(next(self._rv))
             self._objects[key] = stored
             self._broadcast("MODIFIED", stored)
             return copy.deepcopy(stored)
@@ -169,7 +175,10 @@ class FakeClient:
                     f"{key}: stale resourceVersion {sent_rv} on status "
                     f"(current {current['metadata']['resourceVersion']})"
                 )
-            current["status"] = copy.deepcopy(obj.get("status", {}))
+            new_status = copy.deepcopy(obj.get("status", {}))
+            if current.get("status") == new_status:
+                return copy.deepcopy(current)  # no-op: no rv bump, no event
+            current["status"] = new_status
             current["metadata"]["resourceVersion"] = str(next(self._rv))
             self._broadcast("MODIFIED", current)
             return copy.deepcopy(current)

@@ -249,6 +249,11 @@ class MoEMLP(nn.Module):
                 dtype=torch.bfloat16, device=dev),
             requires_grad=False,
         )  # [E_local, I, H]
+        # pack for the grouped GEMM NOW, before the engine sizes the KV
+        # cache from free HBM — lazy packing after cache allocation OOMs
+        # (packed copies must be part of the model's memory footprint)
+        if dev.type == "cuda":
+            self._packed_weights()
 
     def _forward_fp8(self, x) -> torch.Tensor:
         """fp8 expert compute: per-expert torch._scaled_mm (hipBLASLt fp8)
@@ -303,8 +308,14 @@ class MoEMLP(nn.Module):
 
     def invalidate_packed(self):
         """Drop the packed-weight cache (the weight loader calls this:
-        mutations through Parameter.data don't bump tensor._version)."""
+        mutations through Parameter.data don't bump tensor._version).
+        Frees the stale packed copy immediately so the repack never
+        doubles up, then repacks eagerly on GPU (keeps the footprint
+        visible before any KV-cache sizing)."""
         self._pack_version = None
+        self._pack_cache = None
+        if not self.fp8 and self.gate_up_t.is_cuda:
+            self._packed_weights()
 
     def _packed_weights(self):
         """MFMA-fragment-packed expert weights for the grouped GEMM,
