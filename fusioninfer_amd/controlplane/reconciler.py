@@ -23,7 +23,12 @@ from fusioninfer_amd.controlplane.api import (
     InferenceService,
     Role,
 )
-from fusioninfer_amd.controlplane.fake import FakeClient
+from fusioninfer_amd.controlplane.fake import (
+    AlreadyExistsError,
+    ConflictError,
+    FakeClient,
+    NotFoundError,
+)
 from fusioninfer_amd.controlplane.workload import (
     LABEL_ROLE,
     LABEL_SERVICE,
@@ -44,7 +49,12 @@ class InferenceServiceReconciler:
             desired["kind"], md["name"], md.get("namespace", "default")
         )
         if existing is None:
-            self.client.create(desired)
+            try:
+                self.client.create(desired)
+            except AlreadyExistsError as e:
+                # raced with another writer: surface as a conflict so the
+                # manager's optimistic-retry loop re-reads and retries
+                raise ConflictError(str(e))
             return True
         old_hash = existing["metadata"].get("labels", {}).get(LABEL_SPEC_HASH)
         new_hash = md.get("labels", {}).get(LABEL_SPEC_HASH)
@@ -140,7 +150,10 @@ class InferenceServiceReconciler:
         for obj in existing:
             n = obj["metadata"]["name"]
             if n not in expected:
-                self.client.delete("LeaderWorkerSet", n, svc.namespace)
+                try:
+                    self.client.delete("LeaderWorkerSet", n, svc.namespace)
+                except NotFoundError:
+                    pass  # raced with GC/another deleter
 
     # -------------------------------------------------------------- router
     def _reconcile_router(self, svc: InferenceService, role: Role) -> None:
