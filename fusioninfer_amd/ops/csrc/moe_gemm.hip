@@ -44,11 +44,16 @@ FI_DEV short8 load_b16x8(const u16* p) {
   return *reinterpret_cast<const short8*>(p);
 }
 
-// WM x WN waves (4 total), MITER 16-row fragments per wave.
-// Block tile: rows BM = 16*WM*MITER, cols = 16*WN of the OUTPUT space.
+// WM x WN waves (4 total), MITER 16-row fragments per wave, NITER
+// 16-col fragments per wave. Block tile: rows BM = 16*WM*MITER, cols
+// BN = 16*WN*NITER of the OUTPUT space. NITER amortizes the A-panel
+// traffic: every n-tile block re-reads its m-tile's A rows, and at
+// 8192-token prefill chunks A is larger than L2 — the v1 profile
+// (profiles/r02_moe_grouped_v1.md) showed the block_m=128 GEMMs at 52%
+// of GPU time, A-re-read-bound.
 // GATE_UP: B is packed over 2N columns (gate at n, up at n + N);
 // epilogue writes silu(gate)*up. Otherwise a plain grouped GEMM.
-template <int WM, int WN, int MITER, bool GATE_UP>
+template <int WM, int WN, int MITER, int NITER, bool GATE_UP>
 __global__ __launch_bounds__(256) void moe_gemm_kernel(
     u16* __restrict__ out,              // [PM, N] bf16
     const u16* __restrict__ a,          // GATE_UP: x [T, K]; else act [PM, K]
@@ -65,7 +70,8 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
   const int wave = threadIdx.x / kWaveSize;
   const int wm = wave / WN;
   const int wn = wave % WN;
-  const int nt = blockIdx.y * WN + wn;  // 16-col fragment index in [0, N/16)
+  // wave's first 16-col fragment index in [0, N/16)
+  const int nt0 = (blockIdx.y * WN + wn) * NITER;
   const int e = expert_ids[mtile];
   const int row0 = mtile * BM + wm * (16 * MITER);
 
@@ -73,10 +79,15 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
   const u16* b_e =
       b + static_cast<int64_t>(e) * (K / 32) * NB16 * (64 * 8);
   // per-K-step fragment pointers (advance by NB16*512 elements per kt)
-  const u16* bg_p = b_e + (static_cast<int64_t>(nt) * 64 + lane) * 8;
-  const u16* bu_p =
-      GATE_UP ? b_e + ((static_cast<int64_t>(nt) + N / 16) * 64 + lane) * 8
-              : nullptr;
+  const u16* bg_p[NITER];
+  const u16* bu_p[NITER];
+#pragma unroll
+  for (int ni = 0; ni < NITER; ++ni) {
+    bg_p[ni] = b_e + (static_cast<int64_t>(nt0 + ni) * 64 + lane) * 8;
+    bu_p[ni] = GATE_UP
+        ? b_e + ((static_cast<int64_t>(nt0 + ni) + N / 16) * 64 + lane) * 8
+        : nullptr;
+  }
   const int64_t b_step = static_cast<int64_t>(NB16) * 64 * 8;
 
   // A row pointers: gather through sorted_ids for GEMM1 (padding slots
@@ -89,45 +100,56 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
     a_p[mi] = a + static_cast<int64_t>(row) * K + (lane >> 4) * 8;
   }
 
-  floatx4 acc_g[MITER];
-  floatx4 acc_u[GATE_UP ? MITER : 1];
+  floatx4 acc_g[MITER][NITER];
+  floatx4 acc_u[GATE_UP ? MITER : 1][NITER];
 #pragma unroll
-  for (int mi = 0; mi < MITER; ++mi) {
-    acc_g[mi] = {0.f, 0.f, 0.f, 0.f};
-    if (GATE_UP) acc_u[mi] = {0.f, 0.f, 0.f, 0.f};
-  }
+  for (int mi = 0; mi < MITER; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NITER; ++ni) {
+      acc_g[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+      if (GATE_UP) acc_u[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+    }
 
   const int ksteps = K / 32;
   for (int kt = 0; kt < ksteps; ++kt) {
-    const short8 bg = load_b16x8(bg_p + kt * b_step);
-    short8 bu;
-    if (GATE_UP) bu = load_b16x8(bu_p + kt * b_step);
+    short8 av[MITER];
 #pragma unroll
-    for (int mi = 0; mi < MITER; ++mi) {
-      const short8 av = load_b16x8(a_p[mi] + kt * 32);
-      acc_g[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          av, bg, acc_g[mi], 0, 0, 0);
-      if (GATE_UP)
-        acc_u[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            av, bu, acc_u[mi], 0, 0, 0);
+    for (int mi = 0; mi < MITER; ++mi)
+      av[mi] = load_b16x8(a_p[mi] + kt * 32);
+#pragma unroll
+    for (int ni = 0; ni < NITER; ++ni) {
+      const short8 bg = load_b16x8(bg_p[ni] + kt * b_step);
+      short8 bu;
+      if (GATE_UP) bu = load_b16x8(bu_p[ni] + kt * b_step);
+#pragma unroll
+      for (int mi = 0; mi < MITER; ++mi) {
+        acc_g[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            av[mi], bg, acc_g[mi][ni], 0, 0, 0);
+        if (GATE_UP)
+          acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              av[mi], bu, acc_u[mi][ni], 0, 0, 0);
+      }
     }
   }
 
-  const int col = nt * 16 + (lane & 15);
 #pragma unroll
-  for (int mi = 0; mi < MITER; ++mi) {
+  for (int ni = 0; ni < NITER; ++ni) {
+    const int col = (nt0 + ni) * 16 + (lane & 15);
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int orow = row0 + mi * 16 + (lane >> 4) * 4 + r;
-      float v;
-      if (GATE_UP) {
-        const float g = acc_g[mi][r];
-        const float u = acc_u[mi][r];
-        v = (g / (1.f + __expf(-g))) * u;
-      } else {
-        v = acc_g[mi][r];
+    for (int mi = 0; mi < MITER; ++mi) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int orow = row0 + mi * 16 + (lane >> 4) * 4 + r;
+        float v;
+        if (GATE_UP) {
+          const float g = acc_g[mi][ni][r];
+          const float u = acc_u[mi][ni][r];
+          v = (g / (1.f + __expf(-g))) * u;
+        } else {
+          v = acc_g[mi][ni][r];
+        }
+        out[static_cast<int64_t>(orow) * N + col] = f32_to_bf16(v);
       }
-      out[static_cast<int64_t>(orow) * N + col] = f32_to_bf16(v);
     }
   }
 }
@@ -161,8 +183,11 @@ __global__ void moe_combine_kernel(u16* __restrict__ out,
 
 }  // namespace
 
-// block_m 16: decode regime (few rows per expert; WM=1,WN=4,MITER=1).
-// block_m 128: prefill regime (WM=2,WN=2,MITER=4 — 4x in-register B reuse).
+// block_m 16: decode regime (few rows per expert; WM=1,WN=4,MITER=1 —
+// weight-streaming-bound, A is tiny/L2-resident).
+// block_m 128: prefill regime (WM=2,WN=2,MITER=4) with NITER 2 (gate_up,
+// BN=64) / 4 (down, BN=128) so the block's A-panel read amortizes over
+// 4x the output columns of v1.
 void launch_moe_gemm(u16* out, const u16* a, const u16* b,
                      const int* sorted_ids, const int* expert_ids,
                      const int* n_valid, int max_mtiles, int K, int N,
@@ -170,23 +195,25 @@ void launch_moe_gemm(u16* out, const u16* a, const u16* b,
   if (block_m == 16) {
     const dim3 grid(max_mtiles, N / 64);
     if (gate_up)
-      hipLaunchKernelGGL((moe_gemm_kernel<1, 4, 1, true>), grid, dim3(256), 0,
-                         stream, out, a, b, sorted_ids, expert_ids, n_valid,
-                         K, N);
-    else
-      hipLaunchKernelGGL((moe_gemm_kernel<1, 4, 1, false>), grid, dim3(256),
+      hipLaunchKernelGGL((moe_gemm_kernel<1, 4, 1, 1, true>), grid, dim3(256),
                          0, stream, out, a, b, sorted_ids, expert_ids,
                          n_valid, K, N);
+    else
+      hipLaunchKernelGGL((moe_gemm_kernel<1, 4, 1, 1, false>), grid,
+                         dim3(256), 0, stream, out, a, b, sorted_ids,
+                         expert_ids, n_valid, K, N);
   } else {  // block_m == 128
-    const dim3 grid(max_mtiles, N / 32);
-    if (gate_up)
-      hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, true>), grid, dim3(256), 0,
-                         stream, out, a, b, sorted_ids, expert_ids, n_valid,
-                         K, N);
-    else
-      hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, false>), grid, dim3(256),
+    if (gate_up) {
+      const dim3 grid(max_mtiles, N / 64);
+      hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 2, true>), grid, dim3(256),
                          0, stream, out, a, b, sorted_ids, expert_ids,
                          n_valid, K, N);
+    } else {
+      const dim3 grid(max_mtiles, N / 128);
+      hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 4, false>), grid,
+                         dim3(256), 0, stream, out, a, b, sorted_ids,
+                         expert_ids, n_valid, K, N);
+    }
   }
 }
 

@@ -335,7 +335,9 @@ void moe_gemm(at::Tensor out, at::Tensor a, at::Tensor b_packed,
   const int PM = out.size(0);
   TORCH_CHECK(PM % block_m == 0, "padded rows must be block_m-aligned");
   TORCH_CHECK(K % 32 == 0, "K must be a multiple of 32");
-  TORCH_CHECK(N % (block_m == 16 ? 64 : 32) == 0, "N tile misalignment");
+  TORCH_CHECK(
+      N % (block_m == 16 ? 64 : (gate_up ? 64 : 128)) == 0,
+      "N tile misalignment");
   // packed B: [E, K/32, NB/16, 64, 8]
   TORCH_CHECK(b_packed.dim() == 5 && b_packed.size(1) == K / 32 &&
               b_packed.size(2) == (gate_up ? 2 * N : N) / 16 &&
