@@ -150,6 +150,12 @@ def main(argv=None):
                    help="worker endpoints host:port[,component-type]")
     p.add_argument("--host", default="0.0.0.0")
     p.add_argument("--port", type=int, default=8080)
+    p.add_argument("--ext-proc", action="store_true",
+                   help="serve the Envoy ext-proc gRPC protocol (the "
+                        "reference's EPP wire contract) instead of the "
+                        "HTTP reverse proxy")
+    p.add_argument("--grpc-port", type=int, default=9002)
+    p.add_argument("--health-port", type=int, default=9003)
     args = p.parse_args(argv)
 
     with open(args.config) as f:
@@ -159,6 +165,22 @@ def main(argv=None):
         addr, _, ctype = spec.partition(",")
         labels = {"fusioninfer.io/component-type": ctype} if ctype else {}
         endpoints.append(Endpoint(addr, labels=labels))
+    if args.ext_proc:
+        import threading
+
+        from fusioninfer_amd.epp import extproc
+
+        server, health_server, addr, haddr = extproc.serve(
+            picker, lambda: endpoints, port=args.grpc_port,
+            health_port=args.health_port, host=args.host,
+        )
+        print(f"ext-proc gRPC on {addr}, health on {haddr}", flush=True)
+        try:
+            threading.Event().wait()
+        finally:
+            server.stop(0)
+            health_server.stop(0)
+        return
     app = build_router_app(picker, endpoints)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 
