@@ -99,7 +99,6 @@ class RcclKVConnector:
             int(header[0]), int(header[1]), int(header[2]), int(header[3])
         )
         block_ids = allocate_blocks(num_blocks)
-        assert len(block_ids) == num_blocks
         kc0, _ = kv_caches[0]
         layers = len(kv_caches)
         staging = torch.empty(
@@ -108,6 +107,12 @@ class RcclKVConnector:
             device=self.device,
         )
         dist.recv(staging, self.peer_rank, group=self.group)
+        if block_ids is None:
+            # backpressure overflow: the wire must stay consistent, so the
+            # payload is drained into scratch and DROPPED (caller rejects
+            # the request — HTTP 429)
+            return None, prompt_len, first_token, tag
+        assert len(block_ids) == num_blocks
         ids = torch.tensor(block_ids, dtype=torch.int32, device=self.device)
         unpack_kv_blocks(staging, kv_caches, ids)
         return block_ids, prompt_len, first_token, tag
@@ -137,6 +142,8 @@ class InMemoryKVConnector:
             timeout=timeout
         )
         block_ids = allocate_blocks(num_blocks)
+        if block_ids is None:  # backpressure overflow: drop (see Rccl form)
+            return None, prompt_len, first_token, tag
         ids = torch.tensor(block_ids, dtype=torch.int32,
                            device=kv_caches[0][0].device)
         unpack_kv_blocks(staging.to(kv_caches[0][0].device), kv_caches, ids)

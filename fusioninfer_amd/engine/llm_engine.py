@@ -22,6 +22,13 @@ from fusioninfer_amd.engine.scheduler import Scheduler
 from fusioninfer_amd.engine.sequence import SamplingParams, Sequence, SeqStatus
 
 
+class InsufficientBlocksError(RuntimeError):
+    """PD consumer has no room for an imported KV batch right now —
+    backpressure signal (NOT a crash): the recv loop waits for decode
+    completions to free blocks, and past its wait budget the transfer is
+    drained and the request rejected (HTTP 429)."""
+
+
 class RequestOutput:
     def __init__(self, seq: Sequence, new_token_ids=None):
         self.request_id = seq.seq_id
@@ -92,6 +99,9 @@ class LLMEngine:
         self._req_counter = itertools.count()
         self.seqs: Dict[str, Sequence] = {}
         self._held: Dict[str, Sequence] = {}
+        # PD: the KV connector (set by the server/pd wiring on every TP
+        # rank; worker ranks use it for pd_send / pd_recv commands)
+        self.kv_connector = None
         self._import_holder = None
         self._timing = os.environ.get("FI_STEP_TIMING") == "1"
         self._tacc = {}
@@ -290,11 +300,63 @@ class LLMEngine:
         seq = self._held.pop(req_id)
         self.block_manager.free(seq)
 
+    def add_export_request(self, prompt_token_ids: List[int],
+                           priority: int = 0) -> str:
+        """PD producer, ASYNC form: admit a prefill-only request whose
+        blocks are held for KV export. It batches with other prefills in
+        the normal serving loop; when its single token arrives, ship the
+        KV via pd_send_held() / release with release_held()."""
+        from fusioninfer_amd.engine.sequence import SamplingParams
+
+        req_id = self.add_request(
+            prompt_token_ids,
+            SamplingParams(max_tokens=1, temperature=0.0),
+            priority=priority,
+        )
+        seq = self.seqs[req_id]
+        seq.hold_blocks = True
+        self._held[req_id] = seq
+        return req_id
+
+    def pd_send_held(self, connector, req_id: str, prompt_len: int,
+                     first_token: int, tag: int) -> None:
+        """Ship a held export's KV (this rank's shard) and, under TP,
+        direct worker ranks to ship theirs (same block ids — block
+        layout is replicated across TP ranks)."""
+        seq = self._held[req_id]
+        if self.is_driver and self.exec_world_size() > 1:
+            self._ps.tp_broadcast_object({
+                "kind": "pd_send",
+                "block_ids": list(seq.block_ids),
+                "prompt_len": prompt_len,
+                "first_token": first_token,
+                "tag": tag,
+            })
+        connector.send_kv(self.runner.kv_caches, seq.block_ids, prompt_len,
+                          first_token, tag)
+
+    def pd_recv_broadcast(self, block_ids: List[int]) -> None:
+        """PD consumer driver under TP: direct worker ranks to post their
+        shard recvs into the same block ids."""
+        if self.is_driver and self.exec_world_size() > 1:
+            self._ps.tp_broadcast_object({
+                "kind": "pd_recv", "block_ids": list(block_ids),
+            })
+
+    def exec_world_size(self) -> int:
+        return (self.cfg.parallel.tensor_parallel_size
+                * self.cfg.parallel.pipeline_parallel_size)
+
     def allocate_import_blocks(self, num_blocks: int) -> List[int]:
-        """PD consumer: reserve blocks the connector will scatter KV into."""
+        """PD consumer: reserve blocks the connector will scatter KV into.
+        Raises InsufficientBlocksError (backpressure) when full."""
         from fusioninfer_amd.engine.sequence import SamplingParams, Sequence
 
-        assert self.block_manager.num_free() >= num_blocks
+        if self.block_manager.num_free() < num_blocks:
+            raise InsufficientBlocksError(
+                f"need {num_blocks} KV blocks, "
+                f"{self.block_manager.num_free()} free"
+            )
         holder = Sequence("_import", [0] * (num_blocks * self.cfg.cache.block_size),
                           SamplingParams())
         self.block_manager.allocate(holder)
@@ -685,6 +747,19 @@ class LLMEngine:
                     payload["name"], payload["path"], self.cfg.model,
                     device=self.device,
                 ))
+            elif payload["kind"] == "pd_send":
+                # ship this rank's KV shard for a PD export (same block
+                # ids on every TP rank)
+                self.kv_connector.send_kv(
+                    self.runner.kv_caches, payload["block_ids"],
+                    payload["prompt_len"], payload["first_token"],
+                    payload["tag"],
+                )
+            elif payload["kind"] == "pd_recv":
+                self.kv_connector.recv_kv(
+                    self.runner.kv_caches,
+                    lambda n: payload["block_ids"],
+                )
             elif payload["kind"] == "mixed":
                 self.runner.run_batch(payload)
             elif payload["kind"] == "prefill":

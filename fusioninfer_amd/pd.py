@@ -25,16 +25,21 @@ from fusioninfer_amd.engine.llm_engine import LLMEngine
 from fusioninfer_amd.engine.sequence import SamplingParams
 
 
-def build_pd_connector(kvt, device: str = "cpu"):
+def build_pd_connector(kvt, device: str = "cpu", tp: int = 1):
     """Create the KV connector for a PD server from its --kv-transfer-config
     (reference surface: '{"kv_connector":"...","kv_role":"kv_producer"}').
 
-    The PD pair forms a dedicated 2-rank process group — prefiller rank 0,
-    decoder rank 1 by convention — over RCCL on GPU (one xGMI p2p pair) or
-    gloo on CPU. MASTER_ADDR/MASTER_PORT name the rendezvous (the control
-    plane's workload renderer points both pods at the same service).
-    PD currently composes with TP=1 servers (the pair group is the default
-    process group)."""
+    TP=1: the PD pair forms a 2-rank process group — prefiller rank 0,
+    decoder rank 1 — over RCCL on GPU (one xGMI p2p pair) or gloo on CPU.
+
+    TP>1 (PD x TP composition): the global world is 2*tp — prefiller TP
+    ranks [0, tp), decoder [tp, 2tp); init_distributed has already built
+    the per-side TP groups (stage-contiguous blocks). Here every rank
+    joins a DEDICATED 2-rank pair group with its cross-side partner
+    (i <-> tp+i) and ships only its own KV shard — tp parallel xGMI
+    p2p channels, never the default group (VERDICT round-1 item 8).
+    MASTER_ADDR/MASTER_PORT name the rendezvous (the control plane's
+    workload renderer points both pods at the same service)."""
     if kvt is None or not kvt.kv_connector:
         return None
     import torch
@@ -46,6 +51,25 @@ def build_pd_connector(kvt, device: str = "cpu"):
         "nccl" if device.startswith("cuda") and torch.cuda.is_available()
         else "gloo"
     )
+    if tp > 1:
+        assert dist.is_initialized(), \
+            "PD x TP needs init_distributed first (world = 2*tp)"
+        world = dist.get_world_size()
+        rank = dist.get_rank()
+        assert world == 2 * tp, (world, tp)
+        side = rank // tp  # 0 = prefiller half, 1 = decoder half
+        assert (side == 0) == (role == KV_PRODUCER), \
+            f"rank {rank} is on side {side} but role is {role}"
+        pair_group = None
+        peer = None
+        for i in range(tp):
+            ranks = [i, tp + i]
+            grp = dist.new_group(ranks)  # collective: every rank calls all
+            if rank in ranks:
+                pair_group = grp
+                peer = ranks[1] if rank == ranks[0] else ranks[0]
+        return RcclKVConnector(role, peer_rank=peer, group=pair_group,
+                               device=device)
     if not dist.is_initialized():
         dist.init_process_group(
             backend=backend, rank=kvt.kv_rank, world_size=kvt.kv_world_size

@@ -153,15 +153,32 @@ def _rank_main(local_rank: int, args, nproc: int):
     from fusioninfer_amd.server.serving import ServingEngine
 
     cfg = build_engine_config(args)
+    tp_deg = cfg.parallel.tensor_parallel_size
     if rank != 0:
         engine = LLMEngine(cfg, device=device)
+        if cfg.kv_transfer.kv_connector and tp_deg > 1:
+            # PD x TP: worker ranks join their cross-side pair group and
+            # ship/receive their own KV shard on pd_send/pd_recv commands
+            from fusioninfer_amd.pd import build_pd_connector
+
+            engine.kv_connector = build_pd_connector(
+                cfg.kv_transfer, device, tp=tp_deg
+            )
         engine.worker_loop()
         return
     kv_connector = None
     if cfg.kv_transfer.kv_connector:
         from fusioninfer_amd.pd import build_pd_connector
 
-        kv_connector = build_pd_connector(cfg.kv_transfer, device)
+        if tp_deg > 1:
+            # pair groups need the global world up first (world = 2*tp)
+            from fusioninfer_amd.distributed import parallel_state as ps
+
+            ps.init_distributed(
+                tp_deg,
+                backend="nccl" if device.startswith("cuda") else "gloo",
+            )
+        kv_connector = build_pd_connector(cfg.kv_transfer, device, tp=tp_deg)
     serving = ServingEngine(cfg, device=device, kv_connector=kv_connector)
     if args.enable_lora or args.lora_modules:
         for spec in args.lora_modules:

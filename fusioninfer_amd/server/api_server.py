@@ -27,7 +27,7 @@ from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
 
 from fusioninfer_amd.engine.sequence import SamplingParams
-from fusioninfer_amd.server.serving import ServingEngine
+from fusioninfer_amd.server.serving import PDRejectedError, ServingEngine
 
 # ----------------------------------------------------------- tokenization
 
@@ -399,6 +399,11 @@ def build_app(serving: ServingEngine, model_name: str,
         except ValueError as e:
             return JSONResponse({"error": {"message": str(e),
                                            "type": "invalid_request_error"}}, 400)
+        except PDRejectedError as e:
+            # decoder out of KV blocks: tell the client/router to back off
+            return JSONResponse({"error": {"message": str(e),
+                                           "type": "rate_limit_exceeded"}},
+                                429)
         except TimeoutError as e:
             return JSONResponse({"error": {"message": str(e),
                                            "type": "server_error"}}, 504)

@@ -10,6 +10,7 @@ from __future__ import annotations
 import itertools
 import queue
 import threading
+import time
 from typing import Dict, List, Optional, Tuple
 
 from fusioninfer_amd.config import EngineConfig
@@ -18,12 +19,18 @@ from fusioninfer_amd.engine.llm_engine import LLMEngine
 from fusioninfer_amd.engine.sequence import SamplingParams
 
 
+class PDRejectedError(RuntimeError):
+    """The decoder could not admit an imported KV batch within its wait
+    budget (block exhaustion) — maps to HTTP 429."""
+
+
 class ServingEngine:
     def __init__(
         self,
         cfg: EngineConfig,
         device: Optional[str] = None,
         kv_connector=None,
+        import_block_wait_s: float = 30.0,
     ):
         self.engine = LLMEngine(cfg, device=device)
         self._streams: Dict[str, "queue.Queue[Tuple[Optional[int], bool]]"] = {}
@@ -38,9 +45,19 @@ class ServingEngine:
         # PD disaggregation: producer prefills+ships KV,
         # consumer receives KV on a background thread and admits on claim
         self.kv_connector = kv_connector
+        self.engine.kv_connector = kv_connector
+        self.import_block_wait_s = import_block_wait_s
         self._pd_tags = itertools.count(1)
         self._pending_imports: Dict[int, tuple] = {}
+        self._rejected_imports: set = set()
         self._pd_cond = threading.Condition()
+        # async export: finished prefills queue here; a dedicated sender
+        # thread ships KV so the engine loop keeps prefilling the next
+        # requests while transfers are in flight (SURVEY §7 stage-3
+        # overlap; VERDICT round-1 item 8)
+        self._pd_send_q: "queue.Queue" = queue.Queue()
+        if kv_connector is not None and kv_connector.role == KV_PRODUCER:
+            threading.Thread(target=self._pd_send_loop, daemon=True).start()
         if kv_connector is not None and kv_connector.role == KV_CONSUMER:
             threading.Thread(target=self._pd_recv_loop, daemon=True).start()
         self._thread = threading.Thread(target=self._loop, daemon=True)
@@ -62,39 +79,83 @@ class ServingEngine:
         return req_id, q
 
     # ---------------------------------------------------- PD disaggregation
-    def prefill_via_pd(self, prompt_token_ids: List[int]) -> Tuple[int, int]:
-        """Producer role: prefill a prompt, ship its KV to the decoder, and
-        return (pd_tag, first_token). The tag travels in the KV header and
-        back to the router, which hands it to the decode endpoint —
-        out-of-order decode claims still match their own KV."""
+    def prefill_via_pd(self, prompt_token_ids: List[int],
+                       priority: int = 0) -> Tuple[int, int]:
+        """Producer role: prefill a prompt (batched with other requests in
+        the normal engine loop — concurrent PD prefills do NOT serialize),
+        queue its KV for the sender thread, and return (pd_tag,
+        first_token) IMMEDIATELY: the transfer overlaps both the next
+        prefill and the router->decoder round trip. The tag travels in the
+        KV header; out-of-order decode claims still match their own KV."""
         assert self.kv_connector is not None and \
             self.kv_connector.role == KV_PRODUCER, "not a PD prefiller"
+        q: "queue.Queue[Tuple[Optional[int], bool]]" = queue.Queue()
         with self._lock:
-            req_id, first_token, block_ids = self.engine.prefill_export(
-                prompt_token_ids
-            )
-            tag = next(self._pd_tags)
-            self.kv_connector.send_kv(
-                self.engine.runner.kv_caches,
-                block_ids,
-                len(prompt_token_ids),
-                first_token,
-                tag,
-            )
-            self.engine.release_held(req_id)
+            req_id = self.engine.add_export_request(prompt_token_ids,
+                                                    priority=priority)
+            self._streams[req_id] = q
+        self._work.set()
+        first_token = None
+        finished = False
+        while not finished:
+            tok, finished = q.get(timeout=300.0)
+            if tok is not None:
+                first_token = tok
+        tag = next(self._pd_tags)
+        self._pd_send_q.put((req_id, tag, len(prompt_token_ids), first_token))
         return tag, first_token
 
+    def _pd_send_loop(self):
+        """Producer sender thread: ships held exports in tag order (one
+        connector, one ordered p2p channel)."""
+        while not self._stop:
+            try:
+                req_id, tag, prompt_len, first_token = self._pd_send_q.get(
+                    timeout=0.25
+                )
+            except queue.Empty:
+                continue
+            try:
+                with self._lock:
+                    self.engine.pd_send_held(
+                        self.kv_connector, req_id, prompt_len, first_token,
+                        tag,
+                    )
+                    self.engine.release_held(req_id)
+            except Exception as e:
+                if self._stop:
+                    return
+                self.healthy = False
+                self.last_error = repr(e)
+                return
+
     def _pd_recv_loop(self):
+        from fusioninfer_amd.engine.llm_engine import InsufficientBlocksError
+
         def alloc(n):
-            with self._lock:
-                return self.engine.allocate_import_blocks(n)
+            """Backpressure: wait for decode completions to free blocks;
+            past the budget return None — the connector drains the wire
+            and the request is rejected (429)."""
+            deadline = time.monotonic() + self.import_block_wait_s
+            while not self._stop:
+                with self._lock:
+                    try:
+                        ids = self.engine.allocate_import_blocks(n)
+                        self.engine.pd_recv_broadcast(ids)
+                        return ids
+                    except InsufficientBlocksError:
+                        pass
+                if time.monotonic() >= deadline:
+                    return None
+                time.sleep(0.01)
+            return None
 
         while not self._stop:
             try:
                 kwargs = {}
                 if hasattr(self.kv_connector, "channel"):  # in-memory: poll
                     kwargs["timeout"] = 0.25
-                _, prompt_len, first_token, tag = self.kv_connector.recv_kv(
+                ids, prompt_len, first_token, tag = self.kv_connector.recv_kv(
                     self.engine.runner.kv_caches, alloc, **kwargs
                 )
             except queue.Empty:
@@ -105,6 +166,11 @@ class ServingEngine:
                 self.healthy = False
                 self.last_error = repr(e)
                 return
+            if ids is None:  # rejected under block exhaustion
+                with self._pd_cond:
+                    self._rejected_imports.add(tag)
+                    self._pd_cond.notify_all()
+                continue
             with self._lock:
                 holder = self.engine.take_import_holder()
             with self._pd_cond:
@@ -119,6 +185,12 @@ class ServingEngine:
         into the decode loop."""
         with self._pd_cond:
             while pd_tag not in self._pending_imports:
+                if pd_tag in self._rejected_imports:
+                    self._rejected_imports.discard(pd_tag)
+                    raise PDRejectedError(
+                        f"KV for pd_tag={pd_tag} rejected: decoder out of "
+                        f"KV blocks (retry later)"
+                    )
                 if not self._pd_cond.wait(timeout=timeout):
                     raise TimeoutError(f"KV for pd_tag={pd_tag} never arrived")
             holder, prompt_len, first_token = self._pending_imports.pop(pd_tag)

@@ -296,3 +296,136 @@ def test_pd_out_of_order_claims():
     finally:
         pre.shutdown()
         dec.shutdown()
+
+
+def test_pd_concurrent_async_exports():
+    """Async prefill-export: concurrent prefill_via_pd calls batch in one
+    engine loop (no per-request engine spinning) and each decode claim
+    gets ITS OWN first token and completes (VERDICT round-1 item 8)."""
+    import threading
+
+    import torch
+
+    from fusioninfer_amd.distributed.kv_transfer import make_inmemory_pair
+    from fusioninfer_amd.engine.sequence import SamplingParams
+
+    def make_cfg():
+        return EngineConfig(
+            model=get_model_config("tiny-qwen3"),
+            cache=CacheConfig(num_gpu_blocks=64),
+            scheduler=SchedulerConfig(
+                max_num_seqs=8, max_num_batched_tokens=256, max_model_len=128
+            ),
+        )
+
+    prod, cons = make_inmemory_pair()
+    torch.manual_seed(5)
+    pre = ServingEngine(make_cfg(), device="cpu", kv_connector=prod)
+    torch.manual_seed(5)
+    dec = ServingEngine(make_cfg(), device="cpu", kv_connector=cons)
+    try:
+        prompts = [[3, 1, 4] * 8, [2, 7, 1, 8] * 6, [9, 9, 2] * 7]
+        results = {}
+
+        def one(i):
+            results[i] = pre.prefill_via_pd(prompts[i])
+
+        threads = [threading.Thread(target=one, args=(i,))
+                   for i in range(len(prompts))]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=60)
+        assert len(results) == len(prompts)
+        tags = {r[0] for r in results.values()}
+        assert len(tags) == len(prompts)  # unique tags
+        for i, (tag, first) in sorted(results.items()):
+            _, q = dec.submit_imported(
+                tag, SamplingParams(max_tokens=4, temperature=0.0)
+            )
+            tok, fin = q.get(timeout=20)
+            assert tok == first and not fin
+            n = 1
+            while not fin:
+                tok, fin = q.get(timeout=20)
+                if tok is not None:
+                    n += 1
+            assert n == 4
+        # exports released on the producer after shipping
+        assert pre.engine.gpu_cache_usage() == 0.0
+    finally:
+        pre.shutdown()
+        dec.shutdown()
+
+
+def test_pd_block_exhaustion_backpressures_then_429():
+    """Decoder out of KV blocks: the import first WAITS (backpressure),
+    and past the wait budget is drained + rejected — surfacing as
+    PDRejectedError / HTTP 429, never an assert crash."""
+    import asyncio
+
+    import httpx
+    import pytest
+    import torch
+
+    from fusioninfer_amd.distributed.kv_transfer import make_inmemory_pair
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from fusioninfer_amd.server.api_server import build_app
+    from fusioninfer_amd.server.serving import PDRejectedError
+
+    def make_cfg(blocks):
+        return EngineConfig(
+            model=get_model_config("tiny-qwen3"),
+            cache=CacheConfig(num_gpu_blocks=blocks),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=256, max_model_len=128
+            ),
+        )
+
+    prod, cons = make_inmemory_pair()
+    torch.manual_seed(6)
+    pre = ServingEngine(make_cfg(64), device="cpu", kv_connector=prod)
+    torch.manual_seed(6)
+    # decoder capacity: 4 blocks = 64 tokens total
+    dec = ServingEngine(make_cfg(4), device="cpu", kv_connector=cons,
+                        import_block_wait_s=0.4)
+    try:
+        big = [4, 2] * 40  # 80 tokens -> 5 blocks: can NEVER fit
+        tag1, _ = pre.prefill_via_pd(big)
+        with pytest.raises(PDRejectedError):
+            dec.submit_imported(
+                tag1, SamplingParams(max_tokens=4, temperature=0.0),
+                timeout=30.0,
+            )
+        assert dec.healthy  # the recv loop survived the overflow
+
+        # HTTP surface: a decode claim for a rejected tag -> 429
+        tag2, _ = pre.prefill_via_pd(big)
+        app = build_app(dec, "tiny-qwen3")
+
+        async def run():
+            async with httpx.AsyncClient(
+                transport=httpx.ASGITransport(app=app), base_url="http://t"
+            ) as c:
+                return await c.post(
+                    "/v1/completions",
+                    json={"prompt": "x", "max_tokens": 4},
+                    headers={"x-pd-tag": str(tag2)},
+                )
+
+        r = asyncio.run(run())
+        assert r.status_code == 429, r.text
+
+        # a transfer that FITS still goes through afterwards (the channel
+        # stayed consistent across the drained rejections)
+        tag3, first3 = pre.prefill_via_pd([7, 7] * 10)  # 20 tokens
+        _, q3 = dec.submit_imported(
+            tag3, SamplingParams(max_tokens=3, temperature=0.0)
+        )
+        tok, fin = q3.get(timeout=20)
+        assert tok == first3
+        while not fin:
+            _, fin = q3.get(timeout=20)
+    finally:
+        pre.shutdown()
+        dec.shutdown()
