@@ -23,6 +23,34 @@ bench:            ## flagship serving benchmark (1 GPU)
 bench-kernels:    ## kernel microbenchmarks (GPU)
 	$(PY) tools/bench_kernels.py all
 
+lint:             ## offline linter (reference: golangci-lint, 16 linters)
+	$(PY) tools/lint.py
+
+IMG_REGISTRY ?= fusioninfer-amd
+TAG ?= latest
+
+docker-build:     ## build engine / EPP / controller images (reference
+	## Makefile docker-build; needs a docker daemon)
+	docker build -f docker/Dockerfile.engine -t $(IMG_REGISTRY)/engine:$(TAG) .
+	docker build -f docker/Dockerfile.epp -t $(IMG_REGISTRY)/epp:$(TAG) .
+	docker build -f docker/Dockerfile.controller -t $(IMG_REGISTRY)/controller:$(TAG) .
+
+build-installer:  ## render dist/install.yaml from config/ (reference
+	## build-installer; kubectl kustomize when present, python fallback)
+	mkdir -p dist
+	@if command -v kubectl >/dev/null 2>&1; then \
+		kubectl kustomize config/default > dist/install.yaml; \
+	else \
+		$(PY) tools/render_installer.py > dist/install.yaml; \
+	fi
+	@echo wrote dist/install.yaml
+
+deploy:           ## apply the deploy stack to the current cluster
+	kubectl apply -f dist/install.yaml
+
+run:              ## run the manager locally against an in-process store
+	$(PY) -m fusioninfer_amd.controlplane run
+
 crd:              ## print the InferenceService CRD
 	$(PY) -m fusioninfer_amd.controlplane crd
 

@@ -212,8 +212,6 @@ class DraftModelProposer:
 
     # ------------------------------------------------------------- public
     def propose_all(self, seqs) -> List[List[int]]:
-        import torch
-
         k = self.cfg.num_speculative_tokens
         live = []   # (idx, seq, st, budget)
         drafts: List[List[int]] = [[] for _ in seqs]

@@ -38,7 +38,6 @@ adds the prefill handshake exactly like the HTTP router path.
 from __future__ import annotations
 
 import json
-import threading
 import urllib.request
 from typing import Callable, Dict, Iterable, List, Optional, Tuple
 

@@ -142,7 +142,6 @@ def main(argv=None):
     import argparse
 
     import uvicorn
-    import yaml
 
     p = argparse.ArgumentParser("fusioninfer-amd EPP router")
     p.add_argument("--config", required=True, help="EndpointPickerConfig YAML")

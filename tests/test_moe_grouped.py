@@ -9,7 +9,6 @@ runs entirely on-device with static shapes (hipGraph-capturable).
 
 import types
 
-import pytest
 import torch
 
 import fusioninfer_amd.ops as ops
