@@ -67,8 +67,13 @@ FI_DEV u32 pack_bf16x2(float lo, float hi) {
 // both K and V go through the V-style register-stage pipeline with a
 // v_cvt_pk_f32_fp8 dequant between load and ds_write — half the HBM bytes
 // per tile for ~8 extra VALU ops per 16 elements.
-template <int D, bool PAGED, bool FP8>
-__global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
+// WPS = target waves per SIMD (__launch_bounds__ 2nd arg, guide §1).
+// 2 = natural allocation (~200 unified regs, no spills); 3 caps at 168
+// VGPR with ~55 spill slots — whether the extra latency-hiding wave
+// beats the scratch traffic is measured on hardware (FI_PF_WPS env
+// selects at launch; the 64 KB LDS fits 2 blocks/CU either way).
+template <int D, bool PAGED, bool FP8, int WPS = 2>
+__global__ __launch_bounds__(kWaves * kWaveSize, WPS) void prefill_attn_kernel(
     u16* __restrict__ out,        // [T, Hq, D]
     const u16* __restrict__ q,    // [T] rows, stride q_stride
     const void* __restrict__ k_p, // dense rows OR k_cache when PAGED
@@ -87,10 +92,14 @@ __global__ __launch_bounds__(kWaves * kWaveSize) void prefill_attn_kernel(
   constexpr int KF = D / 16;           // QK^T 16-deep k-chunks over head dim
   constexpr int CB = D / 32;           // PV output 32-col blocks
   constexpr int kKRowB = D * 2;        // K row bytes (D=128 -> 256)
-  // V^T rows PADDED to 256 B (data is 128 B): lets the full (row&15)<<4
-  // swizzle fit in-row, making the PV B-fragment reads conflict-free
-  // (the 128-B row's (row&7) swizzle measured 2.6% bank-conflict cycles)
-  constexpr int kVTRowB = 256;
+  // V^T rows at their natural 128 B with the (row&7)<<4 in-row swizzle.
+  // Round 1 padded them to 256 B for the conflict-free (row&15)<<4
+  // swizzle — but that put the workgroup at 98 KB LDS = ONE block/CU
+  // (2 waves/SIMD), and PMC showed the kernel latency-bound there
+  // (MfmaUtil 11%, SQ_WAIT dominant). 64 KB -> 2 blocks/CU buys 2x the
+  // latency hiding for a measured 2.6% bank-conflict cost on the PV
+  // reads.
+  constexpr int kVTRowB = 128;
 
   const int head = blockIdx.y;
   const int kv_head = head / (num_q_heads / num_kv_heads);
@@ -455,11 +464,25 @@ void launch_prefill_attn(u16* out, const u16* q, const void* k, const void* v,
                          float scale, bool fp8, hipStream_t stream) {
   dim3 grid(ntiles, num_q_heads), block(kWaves * kWaveSize);
   const bool paged = block_tables != nullptr;
+  static const int wps = [] {
+    const char* e = getenv("FI_PF_WPS");
+    return (e && e[0] == '3') ? 3 : 2;
+  }();
 #define FI_PF_LAUNCH(DD, PP, F8)                                              \
-  hipLaunchKernelGGL((prefill_attn_kernel<DD, PP, F8>), grid, block, 0,       \
-                     stream, out, q, k, v, tile_seq, tile_row0, cu_seqlens,   \
-                     block_tables, seq_lens_k, max_blocks, q_stride,          \
-                     k_stride, v_stride, num_q_heads, num_kv_heads, scale)
+  do {                                                                        \
+    if (wps == 3)                                                             \
+      hipLaunchKernelGGL((prefill_attn_kernel<DD, PP, F8, 3>), grid, block,   \
+                         0, stream, out, q, k, v, tile_seq, tile_row0,        \
+                         cu_seqlens, block_tables, seq_lens_k, max_blocks,    \
+                         q_stride, k_stride, v_stride, num_q_heads,           \
+                         num_kv_heads, scale);                                \
+    else                                                                      \
+      hipLaunchKernelGGL((prefill_attn_kernel<DD, PP, F8, 2>), grid, block,   \
+                         0, stream, out, q, k, v, tile_seq, tile_row0,        \
+                         cu_seqlens, block_tables, seq_lens_k, max_blocks,    \
+                         q_stride, k_stride, v_stride, num_q_heads,           \
+                         num_kv_heads, scale);                                \
+  } while (0)
   if (fp8 && !paged) abort();  // fp8 KV is a cache format; dense is bf16
   if (head_dim == 128) {
     if (paged) {
