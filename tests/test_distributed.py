@@ -815,3 +815,62 @@ def test_pd_tp2_shard_handoff_token_exact():
         for p in procs:
             assert p.exitcode == 0, f"a rank exited {p.exitcode}"
         assert all(results.get(r) == "ok" for r in range(4)), dict(results)
+
+
+# --------------------------------------------------- TP=8 launch (world=8)
+def _tp8_worker(rank, port, results):
+    """TP=8 engine end-to-end on gloo (BASELINE config #5 launch shape:
+    Llama-70B-class TP=8 — exercised on a small config so the 8-way
+    sharding/collective paths are launch-robust before the driver's
+    8-GPU runs; VERDICT round-1 item 2)."""
+    from fusioninfer_amd.distributed import parallel_state as ps
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+
+    _init(rank, 8, port)
+    try:
+        from fusioninfer_amd.config import (
+            CacheConfig, EngineConfig, SchedulerConfig,
+        )
+        mc = get_model_config("Qwen3-0.6B")  # 16 heads / 8 kv: TP=8 divides
+        mc.num_layers = 2
+        cfg = EngineConfig(
+            model=mc,
+            cache=CacheConfig(num_gpu_blocks=32),
+            scheduler=SchedulerConfig(
+                max_num_seqs=2, max_num_batched_tokens=256, max_model_len=96
+            ),
+        )
+        cfg.parallel.tensor_parallel_size = 8
+        cfg.parallel.distributed_backend = "gloo"
+        engine = LLMEngine(cfg, device="cpu")
+        if engine.is_driver:
+            outs = engine.generate(
+                [PROMPT], SamplingParams(max_tokens=3, temperature=0.0)
+            )
+            assert len(outs[0].output_token_ids) == 3
+            engine.stop_workers()
+        else:
+            engine.worker_loop()
+        results[rank] = "ok"
+    finally:
+        dist.destroy_process_group()
+        ps.destroy()
+
+
+def test_tp8_engine_generates():
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29619
+        procs = [
+            ctx.Process(target=_tp8_worker, args=(r, port, results))
+            for r in range(8)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+        for p in procs:
+            assert p.exitcode == 0, f"a rank exited {p.exitcode}"
+        assert all(results.get(r) == "ok" for r in range(8)), dict(results)
