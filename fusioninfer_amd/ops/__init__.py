@@ -229,7 +229,13 @@ def paged_attention_decode(
     return ref.paged_attention_decode(q, k_cache, v_cache, block_tables, seq_lens, scale)
 
 
-PREFILL_TILE_ROWS = 256  # q rows per workgroup (8 waves x 32); kernel contract
+# q rows per prefill workgroup (NW waves x 32) — kernel contract; the
+# FI_PF_NW env selects the 4-wave variant on both sides (see
+# prefill_attention.hip launcher)
+import os as _os
+
+PREFILL_TILE_ROWS = (4 if _os.environ.get("FI_PF_NW", "8").startswith("4")
+                     else 8) * 32
 
 
 def build_prefill_tiles(seq_lens, device=None):

@@ -143,36 +143,53 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     const int64_t kv_base_cur = kv_base;
 
     // ---- phase A: scores for 16 tokens x G2 heads ----
+    // streaming convert+FMA: per 16-B K group, convert 8 elements and
+    // fold them into all G2 head accumulators immediately — kf[8] live
+    // instead of kf[32] (round-2 register diet: the G=4/D=128 variant
+    // was 218 VGPR -> 2 waves/SIMD; PMC showed WAIT-dominant)
     float s[G2];
+#pragma unroll
+    for (int g = 0; g < G2; ++g) s[g] = 0.f;
     {
-      float kf[DPQ];
-      if (FP8) {
-        const u32* kw = reinterpret_cast<const u32*>(kraw);
 #pragma unroll
-        for (int w = 0; w < DPQ / 4; ++w) unpack_fp8x4(kw[w], &kf[w * 4]);
-      } else {
+      for (int j8 = 0; j8 < KQ4; ++j8) {
+        float kf8[16 / sizeof(CT)];
+        if (FP8) {
+          const u32 w[4] = {kraw[j8].x, kraw[j8].y, kraw[j8].z, kraw[j8].w};
 #pragma unroll
-        for (int j8 = 0; j8 < KQ4; ++j8) {
+          for (int i = 0; i < 4; ++i) unpack_fp8x4(w[i], &kf8[i * 4]);
+        } else {
           const bf16x8 kv8 = __builtin_bit_cast(bf16x8, kraw[j8]);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) kf[j8 * 8 + j] = bf16_to_f32(kv8.h[j]);
+          for (int j = 0; j < 8; ++j) kf8[j] = bf16_to_f32(kv8.h[j]);
+        }
+        constexpr int EPG = 16 / sizeof(CT);  // elems per 16-B group
+#pragma unroll
+        for (int g = 0; g < G2; ++g) {
+#pragma unroll
+          for (int j4 = 0; j4 < EPG / 4; ++j4) {
+            // float4 -> ds_read_b128 (4x fewer LDS cycles than scalar)
+            const float4 qv = *reinterpret_cast<const float4*>(
+                &q_lds[hsplit * G2 + g][quad * DPQ + j8 * EPG + j4 * 4]);
+            s[g] = fmaf(kf8[j4 * 4 + 0], qv.x, s[g]);
+            s[g] = fmaf(kf8[j4 * 4 + 1], qv.y, s[g]);
+            s[g] = fmaf(kf8[j4 * 4 + 2], qv.z, s[g]);
+            s[g] = fmaf(kf8[j4 * 4 + 3], qv.w, s[g]);
+          }
         }
       }
+    }
+    // preload THIS chunk's V before the softmax so its latency hides
+    // under the reductions (it was serialized behind phase A before)
+    u32 vball[kBlockSz];
+    const CT* v_rows = v_cache + kv_base_cur;
+    if (lane < D / 2) {
 #pragma unroll
-      for (int g = 0; g < G2; ++g) {
-        float acc_s = 0.f;
-#pragma unroll
-        for (int j4 = 0; j4 < DPQ / 4; ++j4) {
-          // float4 -> ds_read_b128 (4x fewer LDS cycles than scalar reads)
-          const float4 qv = *reinterpret_cast<const float4*>(
-              &q_lds[hsplit * G2 + g][quad * DPQ + j4 * 4]);
-          acc_s = fmaf(kf[j4 * 4 + 0], qv.x, acc_s);
-          acc_s = fmaf(kf[j4 * 4 + 1], qv.y, acc_s);
-          acc_s = fmaf(kf[j4 * 4 + 2], qv.z, acc_s);
-          acc_s = fmaf(kf[j4 * 4 + 3], qv.w, acc_s);
-        }
-        s[g] = acc_s;
-      }
+      for (int t = 0; t < kBlockSz; ++t)
+        vball[t] = FP8
+            ? static_cast<u32>(*reinterpret_cast<const u16*>(
+                  v_rows + t * D + 2 * lane))
+            : *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
     }
     // issue next chunk's K now; it lands under softmax + PV
     if (chunk + kCWaves < chunk_hi) {
@@ -218,23 +235,14 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
 
     __builtin_amdgcn_sched_barrier(0);
     // ---- phase B: PV accumulate; lane covers dims {2l, 2l+1} ----
-    const CT* v_rows = v_cache + kv_base_cur;
+    // (vball preloaded before the softmax; p is 0 for padding tokens —
+    // phase A masks every invalid position — so the full block is safe)
 #pragma unroll
     for (int g = 0; g < G2; ++g) {
       acc[g][0] *= alpha[g];
       acc[g][1] *= alpha[g];
     }
     if (lane < D / 2) {
-      // preload ALL 16 token rows' bits first (p is 0 for padding
-      // tokens — phase A masks every invalid position — so processing the
-      // full block is safe and the 16 loads overlap instead of serializing)
-      u32 vball[kBlockSz];
-#pragma unroll
-      for (int t = 0; t < kBlockSz; ++t)
-        vball[t] = FP8
-            ? static_cast<u32>(*reinterpret_cast<const u16*>(
-                  v_rows + t * D + 2 * lane))
-            : *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
       for (int t = 0; t < kBlockSz; ++t) {
         float v0, v1;
         if (FP8) {
@@ -353,30 +361,40 @@ void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
                               hipStream_t stream) {
   const bool wide = num_seqs * num_kv_heads * num_parts < 256;
   const int nwaves = wide ? 8 : 4;
+  static const int hs8 = [] {
+    const char* e = getenv("FI_DEC_HS8");
+    return (e && e[0] == '2') ? 2 : 4;
+  }();
   dim3 grid(num_seqs, num_kv_heads, num_parts), block(nwaves * kWaveSize);
 #define FI_LAUNCH_1(DD, GG, NW, F8, HSP)                                      \
   hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG, NW, F8, HSP>), grid,   \
                      block, 0, stream, out, ml_ws, acc_ws, q, k_cache,        \
                      v_cache, block_tables, seq_lens, q_stride, max_blocks,   \
                      num_kv_heads, scale)
-// G=8 runs head-split (HS=2): two waves share each chunk position and
-// carry 4 heads each — the monolithic G=8 register file spilled. The
-// K/V lines the pair re-reads stay in the CU's L1 (same block, same
-// schedule), so HBM traffic is unchanged.
-#define FI_HS(GG) (GG == 8 ? 2 : 1)
-#define FI_LAUNCH(DD, GG)                                                     \
+// G>=4 runs head-split so per-wave softmax state is G2<=2 heads: the
+// monolithic G=8 file spilled, and the monolithic G=4 sat at 218 VGPR
+// = 2 waves/SIMD (PMC: ACTIVE 35% / WAIT 52%, latency-bound). With the
+// streaming-convert phase A, G2=2 compiles to 149 VGPR = 3 waves/SIMD,
+// no spills. The K/V lines an HS-group re-reads stay in the CU's L1
+// (same block, same schedule), so HBM traffic is unchanged.
+// FI_DEC_HS8=2 falls G=8 back to the round-1 HS=2 shape for A/B.
+#define FI_HS(GG) (GG == 8 ? 4 : (GG == 4 ? 2 : 1))
+#define FI_LAUNCH_HS(DD, GG, HSV)                                             \
   if (fp8) {                                                                  \
-    if (wide) { FI_LAUNCH_1(DD, GG, 8, true, FI_HS(GG)); }                    \
-    else      { FI_LAUNCH_1(DD, GG, 4, true, FI_HS(GG)); }                    \
+    if (wide) { FI_LAUNCH_1(DD, GG, 8, true, HSV); }                          \
+    else      { FI_LAUNCH_1(DD, GG, 4, true, HSV); }                          \
   } else {                                                                    \
-    if (wide) { FI_LAUNCH_1(DD, GG, 8, false, FI_HS(GG)); }                   \
-    else      { FI_LAUNCH_1(DD, GG, 4, false, FI_HS(GG)); }                   \
+    if (wide) { FI_LAUNCH_1(DD, GG, 8, false, HSV); }                         \
+    else      { FI_LAUNCH_1(DD, GG, 4, false, HSV); }                         \
   }                                                                           \
   if (num_parts > 1) {                                                        \
     dim3 rgrid(num_seqs, num_kv_heads * GG), rblock(kWaveSize);               \
     hipLaunchKernelGGL((paged_attn_reduce_kernel<DD>), rgrid, rblock, 0,      \
                        stream, out, ml_ws, acc_ws, num_parts);                \
   }
+#define FI_LAUNCH(DD, GG)                                                     \
+  if (GG == 8 && hs8 == 2) { FI_LAUNCH_HS(DD, 8, 2) }                         \
+  else { FI_LAUNCH_HS(DD, GG, FI_HS(GG)) }
   if (head_dim == 128) {
     switch (group) {
       case 1: FI_LAUNCH(128, 1); break;

@@ -68,12 +68,14 @@ FI_DEV u32 pack_bf16x2(float lo, float hi) {
 // v_cvt_pk_f32_fp8 dequant between load and ds_write — half the HBM bytes
 // per tile for ~8 extra VALU ops per 16 elements.
 // WPS = target waves per SIMD (__launch_bounds__ 2nd arg, guide §1).
-// 2 = natural allocation (~200 unified regs, no spills); 3 caps at 168
-// VGPR with ~55 spill slots — whether the extra latency-hiding wave
-// beats the scratch traffic is measured on hardware (FI_PF_WPS env
-// selects at launch; the 64 KB LDS fits 2 blocks/CU either way).
-template <int D, bool PAGED, bool FP8, int WPS = 2>
-__global__ __launch_bounds__(kWaves * kWaveSize, WPS) void prefill_attn_kernel(
+// Measured: WPS=3 spills ~55 slots and LOSES 40% — the kernel is
+// register-bound at 2 waves/SIMD; keep 2.
+// NW = waves per workgroup (q rows = NW*32). 8 = round-1 shape; 4 halves
+// the causal-raggedness idle (a wave's tiles stop at its own diagonal
+// but it still barriers with later waves) at the cost of re-reading K/V
+// tiles from L2 by 2x more workgroups — A/B'd on hardware (FI_PF_NW).
+template <int D, bool PAGED, bool FP8, int WPS = 2, int NW = kWaves>
+__global__ __launch_bounds__(NW * kWaveSize, WPS) void prefill_attn_kernel(
     u16* __restrict__ out,        // [T, Hq, D]
     const u16* __restrict__ q,    // [T] rows, stride q_stride
     const void* __restrict__ k_p, // dense rows OR k_cache when PAGED
@@ -158,18 +160,18 @@ __global__ __launch_bounds__(kWaves * kWaveSize, WPS) void prefill_attn_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) o_acc[cb][r] = 0.f;
 
-  const int wg_q_max = ctx_start + min(wg_row0 + kQBlock, seq_len) - 1;
+  const int wg_q_max = ctx_start + min(wg_row0 + NW * kQPerWave, seq_len) - 1;
   const int num_kv_tiles = wg_q_max / kKVTile + 1;
   const int my_q_max = ctx_start + min(row0 + kQPerWave, seq_len) - 1;
 
   // ---- staging (same pipeline as v2.1) --------------------------------
   constexpr int kKTileBytes = kKVTile * kKRowB;
-  constexpr int kGldsPerWave = kKTileBytes / (kWaves * kWaveSize * 16);
+  constexpr int kGldsPerWave = kKTileBytes / (NW * kWaveSize * 16);
   auto stage_k_glds = [&](int t, u16* kbuf) {
     const int kv0 = t * kKVTile;
 #pragma unroll
     for (int i = 0; i < kGldsPerWave; ++i) {
-      const int base_off = wave * (kKTileBytes / kWaves) + i * (kWaveSize * 16);
+      const int base_off = wave * (kKTileBytes / NW) + i * (kWaveSize * 16);
       const int X = base_off + lane * 16;
       const int row = X / kKRowB;
       const int sbyte = X % kKRowB;
@@ -185,8 +187,11 @@ __global__ __launch_bounds__(kWaves * kWaveSize, WPS) void prefill_attn_kernel(
     }
   };
   constexpr int kChunks = D / 16;
+  // each thread owns VITER (c16, token) staging positions so every NW
+  // covers the full 64 x D tile (NW=8/D=128: 1; NW=4/D=128: 2)
+  constexpr int VITER = ceil_div(kChunks, NW);
   const int v_kv = tid % kKVTile;
-  const int v_c16 = tid / kKVTile;
+  const int v_c0 = tid / kKVTile;  // first 16-dim chunk; step NW per iter
   // 16 fp8 bytes (one uint4) -> two bf16x8
   auto cvt_fp8x16 = [](uint4 raw, bf16x8& lo, bf16x8& hi) {
     const u32 w[4] = {raw.x, raw.y, raw.z, raw.w};
@@ -199,46 +204,64 @@ __global__ __launch_bounds__(kWaves * kWaveSize, WPS) void prefill_attn_kernel(
       hi.h[j] = f32_to_bf16(f[8 + j]);
     }
   };
-  bf16x8 vreg0, vreg1;
+  bf16x8 vreg0[VITER], vreg1[VITER];
   auto vload = [&](int t) {
-    if (v_c16 >= kChunks) return;
     const int src = min(t * kKVTile + v_kv, k_len - 1);
-    const CT* vrow = v + kv_off(src, v_stride) + v_c16 * 16;
-    if (FP8) {
-      cvt_fp8x16(*reinterpret_cast<const uint4*>(vrow), vreg0, vreg1);
-    } else {
-      vreg0 = *reinterpret_cast<const bf16x8*>(vrow);
-      vreg1 = *reinterpret_cast<const bf16x8*>(
-          reinterpret_cast<const u16*>(vrow) + 8);
+    const CT* vrow = v + kv_off(src, v_stride);
+#pragma unroll
+    for (int it = 0; it < VITER; ++it) {
+      const int c16 = v_c0 + it * NW;
+      if (c16 >= kChunks) break;
+      if (FP8) {
+        cvt_fp8x16(*reinterpret_cast<const uint4*>(vrow + c16 * 16),
+                   vreg0[it], vreg1[it]);
+      } else {
+        vreg0[it] = *reinterpret_cast<const bf16x8*>(vrow + c16 * 16);
+        vreg1[it] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const u16*>(vrow + c16 * 16) + 8);
+      }
     }
   };
   // FP8 K staging: same (row, 16-col chunk) ownership as V, but written
   // UNtransposed into the K tile layout ([64 rows][D] bf16, row swizzle);
   // each converted 16-elem group is one 16-B-aligned ds_write_b128 pair
-  bf16x8 kreg0, kreg1;
+  bf16x8 kreg0[VITER], kreg1[VITER];
   auto kload_fp8 = [&](int t) {
-    if (v_c16 >= kChunks) return;
     const int src = min(t * kKVTile + v_kv, k_len - 1);
-    const CT* krow = k + kv_off(src, k_stride) + v_c16 * 16;
-    cvt_fp8x16(*reinterpret_cast<const uint4*>(krow), kreg0, kreg1);
+    const CT* krow = k + kv_off(src, k_stride);
+#pragma unroll
+    for (int it = 0; it < VITER; ++it) {
+      const int c16 = v_c0 + it * NW;
+      if (c16 >= kChunks) break;
+      cvt_fp8x16(*reinterpret_cast<const uint4*>(krow + c16 * 16),
+                 kreg0[it], kreg1[it]);
+    }
   };
   auto kwrite_fp8 = [&](u16* kbuf) {
-    if (v_c16 >= kChunks) return;
     char* kbase = reinterpret_cast<char*>(kbuf);
-    *reinterpret_cast<bf16x8*>(
-        kbase + swz<kKRowB>(v_kv, (v_c16 * 16) * 2)) = kreg0;
-    *reinterpret_cast<bf16x8*>(
-        kbase + swz<kKRowB>(v_kv, (v_c16 * 16 + 8) * 2)) = kreg1;
+#pragma unroll
+    for (int it = 0; it < VITER; ++it) {
+      const int c16 = v_c0 + it * NW;
+      if (c16 >= kChunks) break;
+      *reinterpret_cast<bf16x8*>(
+          kbase + swz<kKRowB>(v_kv, (c16 * 16) * 2)) = kreg0[it];
+      *reinterpret_cast<bf16x8*>(
+          kbase + swz<kKRowB>(v_kv, (c16 * 16 + 8) * 2)) = kreg1[it];
+    }
   };
   auto vwrite = [&](u16* vt_buf) {
-    if (v_c16 >= kChunks) return;
     char* vbase = reinterpret_cast<char*>(vt_buf);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      *reinterpret_cast<u16*>(
-          vbase + swz<kVTRowB>(v_c16 * 16 + j, v_kv * 2)) = vreg0.h[j];
-      *reinterpret_cast<u16*>(
-          vbase + swz<kVTRowB>(v_c16 * 16 + j + 8, v_kv * 2)) = vreg1.h[j];
+    for (int it = 0; it < VITER; ++it) {
+      const int c16 = v_c0 + it * NW;
+      if (c16 >= kChunks) break;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        *reinterpret_cast<u16*>(
+            vbase + swz<kVTRowB>(c16 * 16 + j, v_kv * 2)) = vreg0[it].h[j];
+        *reinterpret_cast<u16*>(
+            vbase + swz<kVTRowB>(c16 * 16 + j + 8, v_kv * 2)) = vreg1[it].h[j];
+      }
     }
   };
 
@@ -462,26 +485,29 @@ void launch_prefill_attn(u16* out, const u16* q, const void* k, const void* v,
                          int64_t q_stride, int64_t k_stride, int64_t v_stride,
                          int num_q_heads, int num_kv_heads, int head_dim,
                          float scale, bool fp8, hipStream_t stream) {
-  dim3 grid(ntiles, num_q_heads), block(kWaves * kWaveSize);
   const bool paged = block_tables != nullptr;
-  static const int wps = [] {
-    const char* e = getenv("FI_PF_WPS");
-    return (e && e[0] == '3') ? 3 : 2;
+  // FI_PF_NW=4 runs 4-wave workgroups (128 q rows): the host tile table
+  // must be built with the matching PREFILL_TILE_ROWS (ops/__init__.py
+  // reads the same env)
+  static const int nw = [] {
+    const char* e = getenv("FI_PF_NW");
+    return (e && e[0] == '4') ? 4 : 8;
   }();
+  dim3 grid(ntiles, num_q_heads), block(nw * kWaveSize);
 #define FI_PF_LAUNCH(DD, PP, F8)                                              \
   do {                                                                        \
-    if (wps == 3)                                                             \
-      hipLaunchKernelGGL((prefill_attn_kernel<DD, PP, F8, 3>), grid, block,   \
-                         0, stream, out, q, k, v, tile_seq, tile_row0,        \
-                         cu_seqlens, block_tables, seq_lens_k, max_blocks,    \
-                         q_stride, k_stride, v_stride, num_q_heads,           \
-                         num_kv_heads, scale);                                \
+    if (nw == 4)                                                              \
+      hipLaunchKernelGGL((prefill_attn_kernel<DD, PP, F8, 2, 4>), grid,       \
+                         block, 0, stream, out, q, k, v, tile_seq,            \
+                         tile_row0, cu_seqlens, block_tables, seq_lens_k,     \
+                         max_blocks, q_stride, k_stride, v_stride,            \
+                         num_q_heads, num_kv_heads, scale);                   \
     else                                                                      \
-      hipLaunchKernelGGL((prefill_attn_kernel<DD, PP, F8, 2>), grid, block,   \
-                         0, stream, out, q, k, v, tile_seq, tile_row0,        \
-                         cu_seqlens, block_tables, seq_lens_k, max_blocks,    \
-                         q_stride, k_stride, v_stride, num_q_heads,           \
-                         num_kv_heads, scale);                                \
+      hipLaunchKernelGGL((prefill_attn_kernel<DD, PP, F8, 2, 8>), grid,       \
+                         block, 0, stream, out, q, k, v, tile_seq,            \
+                         tile_row0, cu_seqlens, block_tables, seq_lens_k,     \
+                         max_blocks, q_stride, k_stride, v_stride,            \
+                         num_q_heads, num_kv_heads, scale);                   \
   } while (0)
   if (fp8 && !paged) abort();  // fp8 KV is a cache format; dense is bf16
   if (head_dim == 128) {
