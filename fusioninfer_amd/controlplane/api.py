@@ -122,6 +122,9 @@ class InferenceService:
     annotations: Dict[str, str] = dataclasses.field(default_factory=dict)
     uid: str = ""
     generation: int = 1
+    #: resourceVersion at read time (optimistic concurrency through the
+    #: typed client; empty on fresh objects — the store assigns it)
+    resource_version: str = ""
     status: Dict[str, Any] = dataclasses.field(default_factory=dict)
 
     def worker_roles(self) -> List[Role]:
@@ -147,6 +150,8 @@ class InferenceService:
                 "annotations": dict(self.annotations),
                 "uid": self.uid,
                 "generation": self.generation,
+                **({"resourceVersion": self.resource_version}
+                   if self.resource_version else {}),
             },
             "spec": {"roles": [r.to_dict() for r in self.roles]},
             "status": copy.deepcopy(self.status),
@@ -163,6 +168,7 @@ class InferenceService:
             annotations=dict(md.get("annotations", {})),
             uid=md.get("uid", ""),
             generation=int(md.get("generation", 1)),
+            resource_version=md.get("resourceVersion", ""),
             status=copy.deepcopy(d.get("status", {})),
         )
 

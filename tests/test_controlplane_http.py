@@ -152,3 +152,44 @@ def test_manager_subprocess_kill_and_resume(http_store, tmp_path):
         stub.stop()
         proc.kill()
         proc.wait(timeout=10)
+
+
+def test_typed_client_crud_watch_and_conflict(http_store):
+    """Typed clientset parity (reference client-go/**): CRUD + informer-
+    style watch in InferenceService terms over the HTTP store, with
+    optimistic concurrency through the dataclass resourceVersion."""
+    import pytest as _pytest
+
+    from fusioninfer_amd.controlplane.client import InferenceServiceClient
+    from fusioninfer_amd.controlplane.fake import ConflictError as _Conflict
+
+    store, raw = http_store
+    c = InferenceServiceClient(raw)
+    created = c.create(monolithic_svc("t1"))
+    assert created.resource_version
+    assert c.get("t1").name == "t1"
+    assert [s.name for s in c.list()] == ["t1"]
+
+    w, poll = c.watch_typed(send_initial=True)
+    try:
+        ev = poll(timeout=3.0)
+        assert ev is not None and ev[0] == "ADDED" and ev[1].name == "t1"
+
+        fresh = c.get("t1")
+        stale = c.get("t1")
+        fresh.roles[0].replicas = 2
+        updated = c.update(fresh)
+        assert updated.roles[0].replicas == 2
+        with _pytest.raises(_Conflict):
+            stale.roles[0].replicas = 5
+            c.update(stale)
+        ev = poll(timeout=3.0)
+        assert ev is not None and ev[0] == "MODIFIED"
+        assert ev[1].roles[0].replicas == 2
+
+        c.delete("t1")
+        assert c.try_get("t1") is None
+        ev = poll(timeout=3.0)
+        assert ev is not None and ev[0] == "DELETED"
+    finally:
+        c.stop_watch(w)
