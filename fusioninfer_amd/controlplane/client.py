@@ -10,7 +10,7 @@ typed informer analog (watch_typed) for controllers/tools built on top.
 
 from __future__ import annotations
 
-from typing import Callable, Iterator, List, Optional, Tuple
+from typing import Callable, List, Optional, Tuple
 
 from fusioninfer_amd.controlplane.api import InferenceService
 
