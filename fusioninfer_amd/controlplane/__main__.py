@@ -56,6 +56,9 @@ def main(argv=None):
                        help="HTTP store URL (httpapi); default in-process")
         p.add_argument("--serve-apiserver", default=None,
                        help="expose the in-process store on this :PORT")
+        p.add_argument("--ports-file", default=None,
+                       help="write the bound probe/metrics/apiserver "
+                            "addresses here as JSON (e2e discovery)")
         p.add_argument("files", nargs="*")
         args = p.parse_args(argv[1:])
 
@@ -89,6 +92,16 @@ def main(argv=None):
             f"{mgr.metrics_addr}); Ctrl-C to stop",
             file=sys.stderr,
         )
+        if args.ports_file:
+            import json
+
+            with open(args.ports_file, "w") as f:
+                json.dump({
+                    "probe": list(mgr.probe_addr or ()),
+                    "metrics": list(mgr.metrics_addr or ()),
+                    "apiserver": list(api_srv.server_address)
+                    if api_srv is not None else None,
+                }, f)
         try:
             import signal
             import threading
