@@ -361,31 +361,27 @@ void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
                               hipStream_t stream) {
   const bool wide = num_seqs * num_kv_heads * num_parts < 256;
   const int nwaves = wide ? 8 : 4;
-  static const int hs8 = [] {
-    const char* e = getenv("FI_DEC_HS8");
-    return (e && e[0] == '2') ? 2 : 4;
-  }();
   dim3 grid(num_seqs, num_kv_heads, num_parts), block(nwaves * kWaveSize);
 #define FI_LAUNCH_1(DD, GG, NW, F8, HSP)                                      \
   hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG, NW, F8, HSP>), grid,   \
                      block, 0, stream, out, ml_ws, acc_ws, q, k_cache,        \
                      v_cache, block_tables, seq_lens, q_stride, max_blocks,   \
                      num_kv_heads, scale)
-// G>=4 runs head-split so per-wave softmax state is G2<=2 heads: the
-// monolithic G=8 file spilled, and the monolithic G=4 sat at 218 VGPR
-// = 2 waves/SIMD (PMC: ACTIVE 35% / WAIT 52%, latency-bound). With the
-// streaming-convert phase A, G2=2 compiles to 149 VGPR = 3 waves/SIMD,
-// no spills. The K/V lines an HS-group re-reads stay in the CU's L1
-// (same block, same schedule), so HBM traffic is unchanged.
-// FI_DEC_HS8=2 falls G=8 back to the round-1 HS=2 shape for A/B.
-#define FI_HS(GG) (GG == 8 ? 4 : (GG == 4 ? 2 : 1))
-#define FI_LAUNCH_HS(DD, GG, HSV)                                             \
+// Head-split policy, A/B'd on hardware (2026-09-14 kbench):
+// - G=8 always runs HS=2 (monolithic spilled; HS=4 measured a wash at
+//   full grids and -12% wide).
+// - G=4 runs HS=2 only at FULL grids (4-wave blocks): the G2=2 state
+//   compiles to 149 VGPR = 3 waves/SIMD, measured +2.5% bf16 / +8% fp8
+//   at b256; at WIDE grids (8-wave, small batches) the halved chunk
+//   parallelism lost 12-19%, so wide keeps the monolithic G=4 shape.
+// (The streaming-convert phase A + early V preload apply everywhere.)
+#define FI_LAUNCH_HS(DD, GG, HSW, HSN)                                        \
   if (fp8) {                                                                  \
-    if (wide) { FI_LAUNCH_1(DD, GG, 8, true, HSV); }                          \
-    else      { FI_LAUNCH_1(DD, GG, 4, true, HSV); }                          \
+    if (wide) { FI_LAUNCH_1(DD, GG, 8, true, HSW); }                          \
+    else      { FI_LAUNCH_1(DD, GG, 4, true, HSN); }                          \
   } else {                                                                    \
-    if (wide) { FI_LAUNCH_1(DD, GG, 8, false, HSV); }                         \
-    else      { FI_LAUNCH_1(DD, GG, 4, false, HSV); }                         \
+    if (wide) { FI_LAUNCH_1(DD, GG, 8, false, HSW); }                         \
+    else      { FI_LAUNCH_1(DD, GG, 4, false, HSN); }                         \
   }                                                                           \
   if (num_parts > 1) {                                                        \
     dim3 rgrid(num_seqs, num_kv_heads * GG), rblock(kWaveSize);               \
@@ -393,8 +389,9 @@ void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
                        stream, out, ml_ws, acc_ws, num_parts);                \
   }
 #define FI_LAUNCH(DD, GG)                                                     \
-  if (GG == 8 && hs8 == 2) { FI_LAUNCH_HS(DD, 8, 2) }                         \
-  else { FI_LAUNCH_HS(DD, GG, FI_HS(GG)) }
+  if (GG == 8)      { FI_LAUNCH_HS(DD, GG, 2, 2) }                            \
+  else if (GG == 4) { FI_LAUNCH_HS(DD, GG, 1, 2) }                            \
+  else              { FI_LAUNCH_HS(DD, GG, 1, 1) }
   if (head_dim == 128) {
     switch (group) {
       case 1: FI_LAUNCH(128, 1); break;
