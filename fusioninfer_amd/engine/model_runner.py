@@ -608,12 +608,9 @@ class ModelRunner:
             return
         if ps.pp_world_size() > 1:
             return  # p2p sends inside capture: PP decode runs eager
-        # MoE decode captures too: the grouped-GEMM path is static-shaped
-        # (device-side block alignment, static grid + device tile count).
-        # The fp8 MoE path still runs per-expert _scaled_mm with .nonzero()
-        # host syncs — not capture-safe; keep that combination eager.
-        if self.cfg.model.is_moe and self.cfg.model.quantization == "fp8":
-            return
+        # MoE decode captures in both dtypes: the grouped-GEMM paths
+        # (bf16 and fp8) are static-shaped — device-side block alignment,
+        # static grid + device tile count, no host syncs.
         max_bs = min(self.cfg.scheduler.max_num_seqs, _DECODE_BUCKETS[-1])
         buckets = [b for b in _DECODE_BUCKETS if b <= max_bs]
         if not self._static:

@@ -224,6 +224,8 @@ class MoEMLP(nn.Module):
             self.register_buffer(
                 "down_scale", torch.stack(dns).to(dev), persistent=False
             )
+            if dev.type == "cuda":
+                self._packed_weights_fp8()  # before KV-cache sizing
             return
         gup, down = [], []
         for e in range(self.e_start, self.e_end):
@@ -259,10 +261,12 @@ class MoEMLP(nn.Module):
         """fp8 expert compute: per-expert torch._scaled_mm (hipBLASLt fp8)
         with the fused silu_and_mul_fp8 epilogue between the two GEMMs.
         Input is the (fp8, per-row scale) tuple the fused norm emits.
-        Launch-bound at high expert counts — the grouped MFMA GEMM is the
-        round-2 step up (ROADMAP.md) — but halves expert-weight bytes,
-        which is what decode is bound by."""
+        Launch-bound at high expert counts — the round-2 GPU path is the
+        grouped fp8 MFMA GEMM (_forward_fp8_grouped); this per-expert
+        loop remains as the CPU reference."""
         x8, xs = x
+        if x8.is_cuda:
+            return self._forward_fp8_grouped(x8, xs)
         T, H = x8.shape
         xd = x8.float() * xs.unsqueeze(1).float()
         logits = xd @ self.router_weight.float().T  # [T, E]
@@ -314,8 +318,60 @@ class MoEMLP(nn.Module):
         visible before any KV-cache sizing)."""
         self._pack_version = None
         self._pack_cache = None
-        if not self.fp8 and self.gate_up_t.is_cuda:
+        if self.fp8:
+            if self.gate_up_fp8.is_cuda:
+                self._packed_weights_fp8()
+        elif self.gate_up_t.is_cuda:
             self._packed_weights()
+
+    def _packed_weights_fp8(self):
+        """MFMA-packed fp8 expert weights ([E, K, N] order: the stored
+        [O, I] weights transpose to K-major) + their per-out-channel
+        scales, cached like the bf16 form."""
+        v = (self.gate_up_fp8._version, self.down_fp8._version)
+        if getattr(self, "_pack_version", None) != v or \
+                getattr(self, "_pack_cache", None) is None:
+            gu = self.gate_up_fp8.data.transpose(1, 2).contiguous()  # [E,H,2I]
+            dn = self.down_fp8.data.transpose(1, 2).contiguous()     # [E,I,H]
+            self._pack_cache = (
+                ops.pack_moe_weights(gu.view(torch.int8))
+                .view(torch.float8_e4m3fn),
+                self.gate_up_scale.float().contiguous(),
+                ops.pack_moe_weights(dn.view(torch.int8))
+                .view(torch.float8_e4m3fn),
+                self.down_scale.float().contiguous(),
+            )
+            self._pack_version = v
+        return self._pack_cache
+
+    def _forward_fp8_grouped(self, x8, xs) -> torch.Tensor:
+        """GPU fp8 path: grouped e4m3 MFMA GEMMs with dequant epilogues —
+        half the expert-weight bytes of bf16 (decode is weight-BW-bound)
+        and no per-expert host loops, so fp8 MoE decode captures too."""
+        T, H = x8.shape
+        xd = x8.float() * xs.unsqueeze(1).float()
+        logits = xd @ self.router_weight.float().T
+        probs = torch.softmax(logits, dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)
+        if self.norm_topk:
+            topv = topv / topv.sum(dim=-1, keepdim=True)
+        E_local = self.e_end - self.e_start
+        block_m = 128 if T * self.top_k >= 64 * E_local else 16
+        sorted_ids, expert_ids, n_valid, pos, PM = self._moe_align(
+            topi, block_m
+        )
+        gu_p, gu_s, dn_p, dn_s = self._packed_weights_fp8()
+        inter = self.down_fp8.shape[2]
+        act = torch.empty(PM, inter, dtype=torch.bfloat16, device=x8.device)
+        ops.moe_gemm_fp8(act, x8, xs.float(), gu_p, gu_s, sorted_ids,
+                         expert_ids, n_valid, block_m, gate_up=True)
+        a8, a_s = ops.quant_fp8_rows(act)
+        y = torch.empty(PM, H, dtype=torch.bfloat16, device=x8.device)
+        ops.moe_gemm_fp8(y, a8, a_s, dn_p, dn_s, sorted_ids, expert_ids,
+                         n_valid, block_m, gate_up=False)
+        out = torch.empty(T, H, dtype=torch.bfloat16, device=x8.device)
+        ops.moe_combine(out, y, pos, topv.reshape(-1).float().contiguous())
+        return ps.tp_all_reduce(out)
 
     def _packed_weights(self):
         """MFMA-fragment-packed expert weights for the grouped GEMM,

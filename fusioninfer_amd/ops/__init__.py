@@ -344,6 +344,22 @@ def moe_gemm(out, a, b_packed, sorted_ids, expert_ids, n_valid,
                         block_m, gate_up)
 
 
+def moe_gemm_fp8(out, a, a_scales, b_packed, b_scales, sorted_ids,
+                 expert_ids, n_valid, block_m: int, gate_up: bool):
+    """Grouped fp8 (e4m3) GEMM with dequant epilogue: out[bf16] =
+    (A_fp8 @ B_fp8) * a_scale[row] * b_scale[expert, col]; gate_up fuses
+    SwiGLU. Same block-aligned layout as moe_gemm; b_packed from
+    pack_moe_weights over the [E, K, N] e4m3 weights."""
+    if out.is_cuda:
+        _require_native()
+        _C.moe_gemm_fp8(out, a, a_scales, b_packed, b_scales, sorted_ids,
+                        expert_ids, n_valid, block_m, gate_up)
+        return out
+    return ref.moe_gemm_fp8(out, a, a_scales, b_packed, b_scales,
+                            sorted_ids, expert_ids, n_valid, block_m,
+                            gate_up)
+
+
 def moe_combine(out, y, pos, w):
     """out[t] = sum_k w[t,k] * y[pos[t,k]] (pos < 0 skipped). Deterministic
     (no atomics) so token-exact tests stay reproducible."""

@@ -154,6 +154,122 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
   }
 }
 
+// fp8 grouped GEMM: A (e4m3, per-row scales from the fused-norm / row
+// quant) x B (e4m3, per-expert per-out-channel scales) on
+// mfma_f32_16x16x32_fp8_fp8 — HALF the weight bytes of bf16 in the
+// weight-bandwidth-bound decode regime, and no per-expert host loops:
+// the fp8 MoE path becomes hipGraph-capturable (round 1 ran per-expert
+// torch._scaled_mm with .nonzero() syncs). Same block-aligned layout
+// and epilogue structure as the bf16 kernel; dequant happens in the
+// epilogue: v = acc * a_scale[row] * b_scale[col].
+typedef long long i64;
+
+template <int WM, int WN, int MITER, int NITER, bool GATE_UP>
+__global__ __launch_bounds__(256) void moe_gemm_fp8_kernel(
+    u16* __restrict__ out,               // [PM, N] bf16
+    const unsigned char* __restrict__ a, // e4m3 [T, K] or act8 [PM, K]
+    const float* __restrict__ a_scales,  // [T] / [PM] per-row
+    const unsigned char* __restrict__ b, // [E][K/32][NB/16][64][8] e4m3
+    const float* __restrict__ b_scales,  // [E][NB] per out channel
+    const int* __restrict__ sorted_ids,
+    const int* __restrict__ expert_ids,
+    const int* __restrict__ n_valid,
+    const int K, const int N) {
+  constexpr int BM = 16 * WM * MITER;
+  const int mtile = blockIdx.x;
+  if (mtile >= *n_valid) return;
+
+  const int lane = threadIdx.x % kWaveSize;
+  const int wave = threadIdx.x / kWaveSize;
+  const int wm = wave / WN;
+  const int wn = wave % WN;
+  const int nt0 = (blockIdx.y * WN + wn) * NITER;
+  const int e = expert_ids[mtile];
+  const int row0 = mtile * BM + wm * (16 * MITER);
+
+  const int NB16 = (GATE_UP ? 2 * N : N) / 16;
+  const unsigned char* b_e =
+      b + static_cast<int64_t>(e) * (K / 32) * NB16 * (64 * 8);
+  const float* bs_e = b_scales + static_cast<int64_t>(e) * NB16 * 16;
+  const unsigned char* bg_p[NITER];
+  const unsigned char* bu_p[NITER];
+#pragma unroll
+  for (int ni = 0; ni < NITER; ++ni) {
+    bg_p[ni] = b_e + (static_cast<int64_t>(nt0 + ni) * 64 + lane) * 8;
+    bu_p[ni] = GATE_UP
+        ? b_e + ((static_cast<int64_t>(nt0 + ni) + N / 16) * 64 + lane) * 8
+        : nullptr;
+  }
+  const int64_t b_step = static_cast<int64_t>(NB16) * 64 * 8;
+
+  const unsigned char* a_p[MITER];
+  int a_row[MITER];
+#pragma unroll
+  for (int mi = 0; mi < MITER; ++mi) {
+    const int slot = row0 + mi * 16 + (lane & 15);
+    a_row[mi] = GATE_UP ? sorted_ids[slot] : slot;
+    a_p[mi] = a + static_cast<int64_t>(a_row[mi]) * K + (lane >> 4) * 8;
+  }
+
+  floatx4 acc_g[MITER][NITER];
+  floatx4 acc_u[GATE_UP ? MITER : 1][NITER];
+#pragma unroll
+  for (int mi = 0; mi < MITER; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NITER; ++ni) {
+      acc_g[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+      if (GATE_UP) acc_u[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+    }
+
+  const int ksteps = K / 32;
+  for (int kt = 0; kt < ksteps; ++kt) {
+    i64 av[MITER];
+#pragma unroll
+    for (int mi = 0; mi < MITER; ++mi)
+      av[mi] = *reinterpret_cast<const i64*>(a_p[mi] + kt * 32);
+#pragma unroll
+    for (int ni = 0; ni < NITER; ++ni) {
+      const i64 bg = *reinterpret_cast<const i64*>(bg_p[ni] + kt * b_step);
+      i64 bu = 0;
+      if (GATE_UP)
+        bu = *reinterpret_cast<const i64*>(bu_p[ni] + kt * b_step);
+#pragma unroll
+      for (int mi = 0; mi < MITER; ++mi) {
+        acc_g[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+            av[mi], bg, acc_g[mi][ni], 0, 0, 0);
+        if (GATE_UP)
+          acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              av[mi], bu, acc_u[mi][ni], 0, 0, 0);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int ni = 0; ni < NITER; ++ni) {
+    const int col = (nt0 + ni) * 16 + (lane & 15);
+    const float bs_g = bs_e[col];
+    const float bs_u = GATE_UP ? bs_e[N + col] : 0.f;
+#pragma unroll
+    for (int mi = 0; mi < MITER; ++mi) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int oslot = row0 + mi * 16 + (lane >> 4) * 4 + r;
+        const int arow = GATE_UP ? sorted_ids[oslot] : oslot;
+        const float as = a_scales[arow];
+        float v;
+        if (GATE_UP) {
+          const float g = acc_g[mi][ni][r] * as * bs_g;
+          const float u = acc_u[mi][ni][r] * as * bs_u;
+          v = (g / (1.f + __expf(-g))) * u;
+        } else {
+          v = acc_g[mi][ni][r] * as * bs_g;
+        }
+        out[static_cast<int64_t>(oslot) * N + col] = f32_to_bf16(v);
+      }
+    }
+  }
+}
+
 // out[t, :] = sum_k w[t,k] * y[pos[t,k], :]   (pos < 0 -> non-local expert
 // or padding: skipped). Deterministic — no atomics, so token-exact TP/PD
 // tests stay reproducible.
@@ -213,6 +329,37 @@ void launch_moe_gemm(u16* out, const u16* a, const u16* b,
       hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 4, false>), grid,
                          dim3(256), 0, stream, out, a, b, sorted_ids,
                          expert_ids, n_valid, K, N);
+    }
+  }
+}
+
+void launch_moe_gemm_fp8(u16* out, const unsigned char* a,
+                         const float* a_scales, const unsigned char* b,
+                         const float* b_scales, const int* sorted_ids,
+                         const int* expert_ids, const int* n_valid,
+                         int max_mtiles, int K, int N, int block_m,
+                         bool gate_up, hipStream_t stream) {
+  if (block_m == 16) {
+    const dim3 grid(max_mtiles, N / 64);
+    if (gate_up)
+      hipLaunchKernelGGL((moe_gemm_fp8_kernel<1, 4, 1, 1, true>), grid,
+                         dim3(256), 0, stream, out, a, a_scales, b, b_scales,
+                         sorted_ids, expert_ids, n_valid, K, N);
+    else
+      hipLaunchKernelGGL((moe_gemm_fp8_kernel<1, 4, 1, 1, false>), grid,
+                         dim3(256), 0, stream, out, a, a_scales, b, b_scales,
+                         sorted_ids, expert_ids, n_valid, K, N);
+  } else {
+    if (gate_up) {
+      const dim3 grid(max_mtiles, N / 64);
+      hipLaunchKernelGGL((moe_gemm_fp8_kernel<2, 2, 4, 2, true>), grid,
+                         dim3(256), 0, stream, out, a, a_scales, b, b_scales,
+                         sorted_ids, expert_ids, n_valid, K, N);
+    } else {
+      const dim3 grid(max_mtiles, N / 128);
+      hipLaunchKernelGGL((moe_gemm_fp8_kernel<2, 2, 4, 4, false>), grid,
+                         dim3(256), 0, stream, out, a, a_scales, b, b_scales,
+                         sorted_ids, expert_ids, n_valid, K, N);
     }
   }
 }

@@ -35,6 +35,10 @@ void launch_prefill_attn(u16*, const u16*, const void*, const void*,
                          int, int, float, bool, hipStream_t);
 void launch_moe_gemm(u16*, const u16*, const u16*, const int*, const int*,
                      const int*, int, int, int, int, bool, hipStream_t);
+void launch_moe_gemm_fp8(u16*, const unsigned char*, const float*,
+                         const unsigned char*, const float*, const int*,
+                         const int*, const int*, int, int, int, int, bool,
+                         hipStream_t);
 void launch_moe_combine(u16*, const u16*, const int*, const float*, int, int,
                         int, hipStream_t);
 
@@ -352,6 +356,41 @@ void moe_gemm(at::Tensor out, at::Tensor a, at::Tensor b_packed,
                       current_stream());
 }
 
+void moe_gemm_fp8(at::Tensor out, at::Tensor a, at::Tensor a_scales,
+                  at::Tensor b_packed, at::Tensor b_scales,
+                  at::Tensor sorted_ids, at::Tensor expert_ids,
+                  at::Tensor n_valid, int64_t block_m, bool gate_up) {
+  CHECK_BF16_CUDA(out);
+  TORCH_CHECK(a.scalar_type() == at::kFloat8_e4m3fn && a.is_cuda() &&
+              a.is_contiguous());
+  TORCH_CHECK(b_packed.scalar_type() == at::kFloat8_e4m3fn &&
+              b_packed.is_contiguous());
+  TORCH_CHECK(a_scales.scalar_type() == at::kFloat &&
+              b_scales.scalar_type() == at::kFloat &&
+              b_scales.is_contiguous());
+  TORCH_CHECK(sorted_ids.scalar_type() == at::kInt &&
+              expert_ids.scalar_type() == at::kInt &&
+              n_valid.scalar_type() == at::kInt);
+  TORCH_CHECK(block_m == 16 || block_m == 128);
+  const int K = a.size(1);
+  const int N = out.size(1);
+  const int PM = out.size(0);
+  TORCH_CHECK(PM % block_m == 0 && K % 32 == 0);
+  TORCH_CHECK(N % (block_m == 16 ? 64 : (gate_up ? 64 : 128)) == 0);
+  TORCH_CHECK(b_packed.dim() == 5 && b_packed.size(1) == K / 32 &&
+              b_packed.size(2) == (gate_up ? 2 * N : N) / 16 &&
+              b_packed.size(3) == 64 && b_packed.size(4) == 8);
+  TORCH_CHECK(b_scales.numel() ==
+              b_packed.size(0) * (gate_up ? 2 * N : N));
+  fi::launch_moe_gemm_fp8(
+      bf16_ptr(out), static_cast<const unsigned char*>(a.data_ptr()),
+      a_scales.data_ptr<float>(),
+      static_cast<const unsigned char*>(b_packed.data_ptr()),
+      b_scales.data_ptr<float>(), sorted_ids.data_ptr<int>(),
+      expert_ids.data_ptr<int>(), n_valid.data_ptr<int>(), PM / block_m, K,
+      N, static_cast<int>(block_m), gate_up, current_stream());
+}
+
 void moe_combine(at::Tensor out, at::Tensor y, at::Tensor pos, at::Tensor w) {
   CHECK_BF16_CUDA(out);
   CHECK_BF16_CUDA(y);
@@ -393,6 +432,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_gemm", &moe_gemm,
         "grouped MFMA GEMM over block-aligned expert segments "
         "(gate_up=true fuses the SwiGLU epilogue)");
+  m.def("moe_gemm_fp8", &moe_gemm_fp8,
+        "grouped fp8 (e4m3) MFMA GEMM with per-row/per-channel dequant "
+        "epilogue (gate_up=true fuses SwiGLU)");
   m.def("moe_combine", &moe_combine,
         "weighted top-k combine of expert outputs (deterministic)");
 }

@@ -65,11 +65,14 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   const int tid = threadIdx.x;
   const int wave = tid / kWaveSize;
   const int lane = tid % kWaveSize;
-  // head-split decomposition: wave = cwave * HS + hsplit
-  constexpr int G2 = G / HS;          // heads this wave carries
-  constexpr int kCWaves = kNWaves / HS;  // waves streaming chunks together
-  const int hsplit = wave % HS;
-  const int cwave = wave / HS;
+  // head-split decomposition: wave = cwave * HSE + hsplit
+  // (HSE clamps HS to G so dead template branches in the launcher's
+  // runtime dispatch still compile — e.g. G=1 never runs with HS=2)
+  constexpr int HSE = (HS > G) ? 1 : HS;
+  constexpr int G2 = G / HSE;         // heads this wave carries
+  constexpr int kCWaves = kNWaves / HSE;  // waves streaming chunks together
+  const int hsplit = wave % HSE;
+  const int cwave = wave / HSE;
 
   const int ctx = seq_lens[seq];
   const int num_chunks = (ctx + kBlockSz - 1) / kBlockSz;
@@ -285,11 +288,11 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     float gm = kNegInf;
 #pragma unroll
     for (int c = 0; c < kCWaves; ++c)
-      gm = fmaxf(gm, merge_m[c * HS + hsplit][g]);
+      gm = fmaxf(gm, merge_m[c * HSE + hsplit][g]);
     float L = 0.f, o0 = 0.f, o1 = 0.f;
 #pragma unroll
     for (int c = 0; c < kCWaves; ++c) {
-      const int w = c * HS + hsplit;
+      const int w = c * HSE + hsplit;
       const float mw = merge_m[w][g];
       const float f = (mw <= kNegInf) ? 0.f : __expf(mw - gm);
       L += merge_l[w][g] * f;

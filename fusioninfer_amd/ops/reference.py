@@ -336,3 +336,30 @@ def moe_combine(out, y, pos, w):
             acc[t] += float(w[t * topk + k]) * yv[p]
     out.copy_(acc.to(out.dtype))
     return out
+
+
+def moe_gemm_fp8(out, a, a_scales, b_packed, b_scales, sorted_ids,
+                 expert_ids, n_valid, block_m: int, gate_up: bool):
+    """fp32 reference of the grouped fp8 GEMM (dequant then matmul)."""
+    E = b_packed.shape[0]
+    w = unpack_moe_weights(b_packed.view(torch.int8)).view(
+        torch.float8_e4m3fn).float()          # [E, K, NB]
+    NB = w.shape[2]
+    ws = b_scales.view(E, NB).float()
+    n_tiles = int(n_valid.item())
+    N = out.shape[1]
+    af = a.float() * a_scales.unsqueeze(1).float()
+    for mt in range(n_tiles):
+        e = int(expert_ids[mt])
+        r0 = mt * block_m
+        rows = slice(r0, r0 + block_m)
+        if gate_up:
+            src = af[sorted_ids[rows].long()]
+        else:
+            src = af[rows]
+        acc = src @ (w[e] * ws[e].unsqueeze(0))
+        if gate_up:
+            g, u = acc[:, :N], acc[:, N:]
+            acc = torch.nn.functional.silu(g) * u
+        out[rows] = acc.to(out.dtype)
+    return out

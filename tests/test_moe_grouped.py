@@ -110,3 +110,24 @@ def test_packed_cache_invalidates_on_weight_mutation():
         mlp.gate_up_t[0].mul_(2.0)
     p3, _ = mlp._packed_weights()
     assert p3 is not p2
+
+
+def test_fp8_grouped_matches_per_expert_loop():
+    """fp8 grouped path (reference semantics on CPU) vs the per-expert
+    dequant loop, same quantized weights."""
+    torch.manual_seed(4)
+    cfg = get_model_config("tiny-qwen3-moe")
+    cfg.quantization = "fp8"
+    try:
+        mlp = MoEMLP(cfg, layer_idx=0)
+        for T in (3, 40):
+            x = torch.randn(T, cfg.hidden_size, dtype=torch.bfloat16) * 0.3
+            from fusioninfer_amd.ops import quant_fp8_rows
+
+            x8, xs = quant_fp8_rows(x)
+            want = mlp._forward_fp8((x8, xs)).float()
+            got = mlp._forward_fp8_grouped(x8, xs).float()
+            rel = (got - want).norm() / want.norm().clamp_min(1e-6)
+            assert rel.item() < 0.08, (T, rel.item())
+    finally:
+        cfg.quantization = None

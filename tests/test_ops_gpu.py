@@ -488,3 +488,49 @@ def test_moe_grouped_forward_matches_manual():
                 exp[t] += float(topv[t, k]) * y
         rel = (got - exp).norm() / exp.norm()
         assert rel.item() < 0.05, (T, rel.item())
+
+
+def test_moe_gemm_fp8_vs_reference():
+    """Grouped fp8 GEMM (dequant epilogue) vs the CPU fp32 reference."""
+    torch.manual_seed(9)
+    E, K, N, T = 4, 128, 128, 200
+    for gate_up in (True, False):
+        w = (torch.randn(E, K, (2 * N) if gate_up else N) * 0.1).to(
+            torch.float8_e4m3fn).cuda()
+        ws = torch.rand(E, (2 * N) if gate_up else N,
+                        dtype=torch.float32, device=DEV) * 0.2 + 0.01
+        b_packed = ops.pack_moe_weights(w.view(torch.int8)).view(
+            torch.float8_e4m3fn)
+        block_m = 16
+        counts = [5, 0, 2 * block_m, 9]
+        tiles = [-(-c // block_m) for c in counts if c > 0]
+        n_tiles = sum(tiles)
+        PM = (n_tiles + 1) * block_m
+        sorted_ids = torch.zeros(PM, dtype=torch.int32, device=DEV)
+        expert_ids = torch.zeros(PM // block_m, dtype=torch.int32, device=DEV)
+        p = t = 0
+        for e, c in enumerate(counts):
+            if c == 0:
+                continue
+            nt = -(-c // block_m)
+            for i in range(c):
+                sorted_ids[p + i] = t % T
+                t += 7
+            expert_ids[p // block_m: p // block_m + nt] = e
+            p += nt * block_m
+        n_valid = torch.tensor([n_tiles], dtype=torch.int32, device=DEV)
+        a = (torch.randn(max(T, PM), K) * 0.3).to(torch.float8_e4m3fn).cuda()
+        a_s = torch.rand(max(T, PM), dtype=torch.float32, device=DEV) + 0.1
+        out = torch.zeros(PM, N, dtype=torch.bfloat16, device=DEV)
+        ops.moe_gemm_fp8(out, a, a_s, b_packed, ws, sorted_ids, expert_ids,
+                         n_valid, block_m, gate_up)
+        want = torch.zeros(PM, N, dtype=torch.bfloat16)
+        from fusioninfer_amd.ops import reference as _r
+
+        _r.moe_gemm_fp8(want, a.cpu(), a_s.cpu(), b_packed.cpu(), ws.cpu(),
+                        sorted_ids.cpu(), expert_ids.cpu(), n_valid.cpu(),
+                        block_m, gate_up)
+        rows = n_tiles * block_m
+        torch.testing.assert_close(out[:rows].float().cpu(),
+                                   want[:rows].float(),
+                                   atol=5e-2, rtol=5e-2)
