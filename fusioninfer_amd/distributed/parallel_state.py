@@ -21,6 +21,8 @@ _TP_WORLD = 1
 _PP_RANK = 0
 _PP_WORLD = 1
 _RANK = 0
+_ENGINE_BASE = 0  # first global rank of this engine's tp*pp sub-world
+_EXEC_GROUP = None  # the engine's whole pipeline (PP>1 sub-world group)
 _INITIALIZED = False
 
 
@@ -41,7 +43,7 @@ def init_distributed(
     rank == stage.
     """
     global _TP_GROUP, _TP_RANK, _TP_WORLD, _PP_RANK, _PP_WORLD, _RANK, \
-        _INITIALIZED
+        _ENGINE_BASE, _INITIALIZED
     if _INITIALIZED:
         return
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -70,9 +72,23 @@ def init_distributed(
     _RANK = rank
     pp = pipeline_parallel_size
     if pp > 1:
-        assert world_size == tp * pp, (world_size, tp, pp)
-        _PP_RANK = rank // tp
+        # world may hold SEVERAL tp*pp engines (PD: prefiller + decoder
+        # halves); stage indices are sub-world-local
+        assert world_size % (tp * pp) == 0, (world_size, tp, pp)
+        _ENGINE_BASE = rank - rank % (tp * pp)
+        _PP_RANK = (rank % (tp * pp)) // tp
         _PP_WORLD = pp
+        # execution group = this engine's whole pipeline: broadcasts must
+        # NOT use the default group when the world holds several engines
+        # (PD composes prefiller + decoder sub-worlds)
+        global _EXEC_GROUP
+        if dist.is_initialized():
+            sub = tp * pp
+            for start in range(0, world_size, sub):
+                ranks = list(range(start, start + sub))
+                grp = dist.new_group(ranks)
+                if rank in ranks:
+                    _EXEC_GROUP = grp
     _INITIALIZED = True
 
 
@@ -97,11 +113,13 @@ def tp_group():
 
 def destroy() -> None:
     global _TP_GROUP, _TP_RANK, _TP_WORLD, _PP_RANK, _PP_WORLD, _RANK, \
-        _INITIALIZED
+        _ENGINE_BASE, _INITIALIZED
     if dist.is_initialized():
         dist.destroy_process_group()
     _TP_GROUP, _TP_RANK, _TP_WORLD, _INITIALIZED = None, 0, 1, False
-    _PP_RANK, _PP_WORLD, _RANK = 0, 1, 0
+    global _EXEC_GROUP
+    _PP_RANK, _PP_WORLD, _RANK, _ENGINE_BASE = 0, 1, 0, 0
+    _EXEC_GROUP = None
 
 
 # --------------------------------------------------------------- pipeline
@@ -138,8 +156,9 @@ def _gloo_safe_recv(t: torch.Tensor, src: int) -> None:
 
 
 def _stage_peer(stage: int) -> int:
-    """Global rank of `stage`'s member in THIS rank's TP column."""
-    return stage * _TP_WORLD + _TP_RANK
+    """Global rank of `stage`'s member in THIS rank's TP column
+    (within this engine's sub-world — PD composes two engines)."""
+    return _ENGINE_BASE + stage * _TP_WORLD + _TP_RANK
 
 
 def pp_send_next(t: torch.Tensor) -> None:
@@ -219,7 +238,7 @@ def tp_broadcast_object(obj=None):
     (the TP group, or the whole pipeline when PP > 1)."""
     if _PP_WORLD > 1:
         lst = [obj]
-        dist.broadcast_object_list(lst, src=0)
+        dist.broadcast_object_list(lst, src=_ENGINE_BASE, group=_EXEC_GROUP)
         return lst[0]
     if _TP_WORLD == 1:
         return obj
@@ -235,7 +254,7 @@ def tp_all_reduce_min_int(value: int) -> int:
         return value
     if _PP_WORLD > 1:
         t = torch.tensor([value], dtype=torch.int64)
-        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        dist.all_reduce(t, op=dist.ReduceOp.MIN, group=_EXEC_GROUP)
         return int(t.item())
     if _TP_WORLD == 1:
         return value

@@ -874,3 +874,103 @@ def test_tp8_engine_generates():
         for p in procs:
             assert p.exitcode == 0, f"a rank exited {p.exitcode}"
         assert all(results.get(r) == "ok" for r in range(8)), dict(results)
+
+
+# ------------------------------------------------ PD x PP (world=4, gloo)
+def _pd_pp_worker(rank, port, results):
+    """PP=2 prefiller (stages 0,1 = ranks 0,1) -> PP=2 decoder (ranks
+    2,3): each STAGE ships its own layers' KV shard over its pair group
+    (0<->2, 1<->3) — the PD x PP composition round 1 left unvalidated."""
+    import types
+
+    from fusioninfer_amd.distributed import parallel_state as ps
+    from fusioninfer_amd.engine.llm_engine import LLMEngine
+    from fusioninfer_amd.engine.sequence import SamplingParams
+    from fusioninfer_amd.pd import build_pd_connector
+
+    expected = None
+    if rank == 2:
+        ps.ensure_single_process()
+        mono = LLMEngine(_engine_cfg(), device="cpu")
+        outs = mono.generate(
+            [PROMPT], SamplingParams(max_tokens=N_TOKENS, temperature=0.0)
+        )
+        expected = outs[0].output_token_ids
+        ps.destroy()
+
+    _init(rank, 4, port)
+    try:
+        ps.init_distributed(tensor_parallel_size=1, backend="gloo",
+                            pipeline_parallel_size=2)
+        cfg = _engine_cfg()
+        cfg.parallel.pipeline_parallel_size = 2
+        cfg.parallel.distributed_backend = "gloo"
+        engine = LLMEngine(cfg, device="cpu")
+        kvt = types.SimpleNamespace(
+            kv_connector="RcclConnector",
+            kv_role="kv_producer" if rank < 2 else "kv_consumer",
+            kv_rank=rank // 2, kv_world_size=2,
+        )
+        # exec world per side = pp*tp = 2: pair groups (0,2) and (1,3)
+        conn = build_pd_connector(kvt, device="cpu", tp=2)
+        engine.kv_connector = conn
+
+        if rank in (1, 3):   # stage-1 workers
+            engine.worker_loop()
+            results[rank] = "ok"
+            return
+        if rank == 0:        # prefiller driver (stage 0)
+            req_id = engine.add_export_request(PROMPT)
+            while engine.has_unfinished():
+                engine.step()
+            seq = engine._held[req_id]
+            first = seq.output_token_ids[0]
+            engine.pd_send_held(conn, req_id, len(PROMPT), first, tag=9)
+            engine.release_held(req_id)
+            engine.stop_workers()
+            results[0] = "ok"
+            return
+
+        def alloc(n):
+            ids = engine.allocate_import_blocks(n)
+            engine.pd_recv_broadcast(ids)
+            return ids
+
+        _, prompt_len, first, tag = conn.recv_kv(
+            engine.runner.kv_caches, alloc
+        )
+        assert tag == 9 and prompt_len == len(PROMPT)
+        req_id = engine.add_imported_request(
+            prompt_len, first,
+            SamplingParams(max_tokens=N_TOKENS, temperature=0.0),
+        )
+        got = {}
+        while engine.has_unfinished():
+            for out in engine.step():
+                if out.finished:
+                    got[out.request_id] = out
+        toks = got[req_id].output_token_ids
+        assert toks == expected, (toks, expected)
+        engine.stop_workers()
+        results[2] = "ok"
+    finally:
+        dist.destroy_process_group()
+        ps.destroy()
+
+
+def test_pd_pp2_stage_shard_handoff_token_exact():
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        port = 29621
+        procs = [
+            ctx.Process(target=_pd_pp_worker, args=(r, port, results))
+            for r in range(4)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+        for p in procs:
+            assert p.exitcode == 0, f"a rank exited {p.exitcode}"
+        assert all(results.get(r) == "ok" for r in range(4)), dict(results)
