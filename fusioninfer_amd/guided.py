@@ -483,14 +483,54 @@ _INT_RE = "-?(0|[1-9][0-9]*)"
 _NUM_RE = "-?(0|[1-9][0-9]*)(\\.[0-9]+)?([eE][+-]?[0-9]+)?"
 
 
-def schema_to_regex(schema: dict, depth: int = 0) -> str:
+def _int_range_regex(lo, hi) -> str:
+    """Regex for integers in [lo, hi] (xgrammar-class bounds support).
+    Small ranges enumerate exactly; unbounded non-negative strips the
+    sign; anything else is unsupported (caller falls back)."""
+    if lo is not None and hi is not None:
+        lo, hi = int(lo), int(hi)
+        if hi < lo:
+            raise ValueError("maximum < minimum")
+        if hi - lo <= 512:
+            return "(" + "|".join(str(v) for v in range(lo, hi + 1)) + ")"
+        raise ValueError("integer range too wide to enumerate")
+    if lo is not None and int(lo) >= 0:
+        return "(0|[1-9][0-9]*)" if int(lo) == 0 else "[1-9][0-9]*"
+    if hi is not None and int(hi) <= 0:
+        return "(-[1-9][0-9]*|0)" if int(hi) == 0 else "-[1-9][0-9]*"
+    raise ValueError("unsupported integer bounds")
+
+
+def schema_to_regex(schema: dict, depth: int = 0, defs=None) -> str:
     """JSON-schema subset -> anchored regex (compact output, property
-    order fixed, all listed properties emitted). Unsupported constructs
-    raise ValueError — the API layer then falls back to generic JSON."""
+    order fixed, all listed properties emitted). Supported beyond the
+    basics: enum/const, anyOf/oneOf, allOf (single branch), type lists,
+    string pattern + minLength/maxLength, integer minimum/maximum
+    (enumerable or sign-determined), array minItems/maxItems, local
+    $defs/$ref (cycle-guarded by depth). Unsupported constructs raise
+    ValueError — the API layer then falls back to generic JSON."""
     if depth > 8:
         raise ValueError("schema nesting too deep")
     if not isinstance(schema, dict):
         raise ValueError("schema must be an object")
+    if defs is None:
+        defs = schema.get("$defs") or schema.get("definitions") or {}
+    if "$ref" in schema:
+        ref = schema["$ref"]
+        for prefix in ("#/$defs/", "#/definitions/"):
+            if ref.startswith(prefix):
+                name = ref[len(prefix):]
+                if name not in defs:
+                    raise ValueError(f"unresolved $ref {ref!r}")
+                return schema_to_regex(defs[name], depth + 1, defs)
+        raise ValueError(f"unsupported $ref {ref!r} (local #/$defs only)")
+    if "allOf" in schema:
+        branches = schema["allOf"]
+        if len(branches) != 1:
+            raise ValueError("allOf with multiple branches unsupported")
+        merged = {**branches[0],
+                  **{k: v for k, v in schema.items() if k != "allOf"}}
+        return schema_to_regex(merged, depth + 1, defs)
     if "enum" in schema:
         opts = "|".join(
             _re_escape(json.dumps(v, separators=(",", ":")))
@@ -502,13 +542,14 @@ def schema_to_regex(schema: dict, depth: int = 0) -> str:
     for alt_key in ("anyOf", "oneOf"):
         if alt_key in schema:
             opts = "|".join(
-                schema_to_regex(s, depth + 1) for s in schema[alt_key]
+                schema_to_regex(s, depth + 1, defs) for s in schema[alt_key]
             )
             return f"({opts})"
     t = schema.get("type")
     if isinstance(t, list):  # {"type": ["string", "null"]}
         opts = "|".join(
-            schema_to_regex({**schema, "type": tt}, depth + 1) for tt in t
+            schema_to_regex({**schema, "type": tt}, depth + 1, defs)
+            for tt in t
         )
         return f"({opts})"
     if t == "string":
@@ -517,8 +558,19 @@ def schema_to_regex(schema: dict, depth: int = 0) -> str:
             body = pattern.lstrip("^").rstrip("$")
             parse_regex(body)  # unsupported syntax raises -> fallback
             return f'"({body})"'
+        lo = schema.get("minLength")
+        hi = schema.get("maxLength")
+        if lo is not None or hi is not None:
+            lo = int(lo or 0)
+            if hi is not None and (int(hi) < lo or int(hi) > 4096):
+                raise ValueError("bad minLength/maxLength")
+            rep = f"{{{lo},{int(hi)}}}" if hi is not None else f"{{{lo},}}"
+            return f'"([^"\\\n\r]|\\.){rep}"'
         return _STRING_RE
     if t == "integer":
+        if "minimum" in schema or "maximum" in schema:
+            return _int_range_regex(schema.get("minimum"),
+                                    schema.get("maximum"))
         return _INT_RE
     if t == "number":
         return _NUM_RE
@@ -528,7 +580,7 @@ def schema_to_regex(schema: dict, depth: int = 0) -> str:
         return "null"
     if t == "array":
         item = schema_to_regex(schema.get("items", {"type": "string"}),
-                               depth + 1)
+                               depth + 1, defs)
         lo = int(schema.get("minItems", 0))
         hi = schema.get("maxItems")
         if lo == 0 and hi is None:
@@ -551,7 +603,7 @@ def schema_to_regex(schema: dict, depth: int = 0) -> str:
         parts = []
         for name, sub in props.items():
             key = _re_escape(json.dumps(name))
-            parts.append(f"{key}:{schema_to_regex(sub, depth + 1)}")
+            parts.append(f"{key}:{schema_to_regex(sub, depth + 1, defs)}")
         return "\\{" + ",".join(parts) + "\\}"
     raise ValueError(f"unsupported schema type {t!r}")
 

@@ -299,3 +299,58 @@ def test_schema_extensions():
     }))
     assert accepts(g, '{"a":1,"c":"x"}')
     assert not accepts(g, '{"a":1,"b":true,"c":"x"}')
+
+
+def test_schema_bounds_refs_and_allof():
+    """xgrammar-class schema coverage: $defs/$ref, string min/maxLength,
+    integer minimum/maximum, allOf(single), and the fallbacks."""
+    import pytest as _p
+
+    from fusioninfer_amd.guided import RegexGrammar, schema_to_regex
+
+    def accepts(rx, s):
+        g = RegexGrammar(rx)
+        st = g.initial()
+        for ch in s:
+            st = g.step(st, ch)
+            if st is None:
+                return False
+        return g.is_complete(st)
+
+    # $defs / $ref
+    rx = schema_to_regex({
+        "$defs": {"port": {"type": "integer", "minimum": 1, "maximum": 9}},
+        "type": "object",
+        "properties": {"p": {"$ref": "#/$defs/port"}},
+    })
+    assert accepts(rx, '{"p":7}')
+    assert not accepts(rx, '{"p":0}')
+
+    # integer bounds: enumerated range and sign-only
+    rx = schema_to_regex({"type": "integer", "minimum": 250, "maximum": 255})
+    assert accepts(rx, "255") and not accepts(rx, "249")
+    rx = schema_to_regex({"type": "integer", "minimum": 0})
+    assert accepts(rx, "0") and accepts(rx, "123") and not accepts(rx, "-3")
+    rx = schema_to_regex({"type": "integer", "maximum": -1})
+    assert accepts(rx, "-7") and not accepts(rx, "2")
+    with _p.raises(ValueError):
+        schema_to_regex({"type": "integer", "minimum": -5, "maximum": 10**9})
+
+    # string length bounds
+    rx = schema_to_regex({"type": "string", "minLength": 2, "maxLength": 3})
+    assert accepts(rx, '"ab"') and accepts(rx, '"abc"')
+    assert not accepts(rx, '"a"') and not accepts(rx, '"abcd"')
+
+    # allOf single-branch merge
+    rx = schema_to_regex({
+        "allOf": [{"type": "integer"}], "minimum": 1, "maximum": 3,
+    })
+    assert accepts(rx, "2") and not accepts(rx, "4")
+    with _p.raises(ValueError):
+        schema_to_regex({"allOf": [{"type": "integer"}, {"minimum": 1}]})
+
+    # unresolved / non-local refs fall back
+    with _p.raises(ValueError):
+        schema_to_regex({"$ref": "#/$defs/missing"})
+    with _p.raises(ValueError):
+        schema_to_regex({"$ref": "http://x/schema.json"})
