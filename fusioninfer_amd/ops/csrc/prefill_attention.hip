@@ -130,14 +130,15 @@ __global__ __launch_bounds__(NW * kWaveSize, WPS) void prefill_attn_kernel(
     return (seq_start + r) * dense_stride + static_cast<int64_t>(kv_head) * D;
   };
 
-  // LDS: double-buffered K (glds target) + DOUBLE-buffered V^T so the
-  // tile loop needs only ONE barrier (PV reads vt[t&1] while the next
-  // tile's V lands in vt[(t+1)&1]). ONE __shared__ object
-  // (guide §5 ".s-level traps" (a)).
-  __shared__ u16 smem[2 * kKVTile * D + 2 * D * (kVTRowB / 2)];
-  u16* k_lds0 = smem;
-  u16* k_lds1 = smem + kKVTile * D;
-  u16* vt_lds0 = smem + 2 * kKVTile * D;
+  // LDS: TRIPLE-buffered K (glds target, prefetch depth 2 so a K tile
+  // stays in flight ACROSS the tile barrier — round-2: with 2 buffers
+  // every barrier drained the glds queue, the guide's ~20% structural
+  // stall) + double-buffered V^T; ONE barrier per tile, raw s_barrier +
+  // counted vmcnt for bf16 (guide §5 "pipelining across barriers").
+  // ONE __shared__ object (guide §5 ".s-level traps" (a)).
+  __shared__ u16 smem[3 * kKVTile * D + 2 * D * (kVTRowB / 2)];
+  auto k_lds = [&](int i) -> u16* { return smem + i * (kKVTile * D); };
+  u16* vt_lds0 = smem + 3 * kKVTile * D;
   u16* vt_lds1 = vt_lds0 + D * (kVTRowB / 2);
 
   // ---- Q fragments (the QK^T B operand): lane holds
@@ -280,11 +281,23 @@ __global__ __launch_bounds__(NW * kWaveSize, WPS) void prefill_attn_kernel(
     for (int cb = 0; cb < CB; ++cb)
       v_off[kc][cb] = swz<kVTRowB>(cb * 32 + col, kc * 32 + hi1 * 16);
 
-  // prologue: K(0) in flight + V(0) staged and published, V(1) in flight
+  // outstanding-vmem count a tile barrier leaves in flight: the NEWEST
+  // K-tile glds (kGldsPerWave wave ops) + the newest V loads (2 per
+  // VITER) — everything older (incl. the K tile consumed next) drains.
+  // Staging is UNCONDITIONAL with clamped source rows, so the count is
+  // a compile-time constant on every iteration (no ragged tail).
+  constexpr int kAhead = kGldsPerWave + 2 * VITER;
+  auto tile_barrier = [&] {
+    asm volatile("s_waitcnt vmcnt(%0)" ::"n"(kAhead) : "memory");
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  };
+
+  // prologue: K(0)+K(1) and V(0)+V(1) in flight, vt[0] published
   if (FP8) {
     kload_fp8(0);
     vload(0);
-    kwrite_fp8(k_lds0);
+    kwrite_fp8(k_lds(0));
     vwrite(vt_lds0);
     if (num_kv_tiles > 1) {
       kload_fp8(1);
@@ -292,22 +305,27 @@ __global__ __launch_bounds__(NW * kWaveSize, WPS) void prefill_attn_kernel(
     }
     __syncthreads();        // publishes K(0) + vt[0]
   } else {
-    stage_k_glds(0, k_lds0);
+    stage_k_glds(0, k_lds(0));
     vload(0);
-    __syncthreads();        // drains K(0) glds
+    stage_k_glds(1, k_lds(1));
+    // drain K(0) (own portion; the barrier below publishes cross-wave)
+    asm volatile("s_waitcnt vmcnt(%0)" ::"n"(kAhead) : "memory");
     vwrite(vt_lds0);
-    if (num_kv_tiles > 1) vload(1);
-    __syncthreads();        // publishes vt[0]
+    vload(1);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();   // publishes K(0) + vt[0]
   }
 
   for (int t = 0; t < num_kv_tiles; ++t) {
     const int kv0 = t * kKVTile;
-    u16* kbuf = (t & 1) ? k_lds1 : k_lds0;
-    u16* kbuf_next = (t & 1) ? k_lds0 : k_lds1;
+    u16* kbuf = FP8 ? k_lds(t & 1) : k_lds(t % 3);
+    u16* kbuf_next = k_lds((t + 1) & 1);  // fp8 staging target
     const u16* vt_cur = (t & 1) ? vt_lds1 : vt_lds0;
     u16* vt_next = (t & 1) ? vt_lds0 : vt_lds1;
 
-    if (!FP8 && t + 1 < num_kv_tiles) stage_k_glds(t + 1, kbuf_next);
+    // depth-2 K prefetch (clamped rows: past-the-end tiles re-read the
+    // last row — a few hundred wasted bytes beat a ragged vmcnt count)
+    if (!FP8) stage_k_glds(t + 2, k_lds((t + 2) % 3));
 
     const bool compute = active && kv0 <= my_q_max;
     floatx16 s_acc[2];  // S^T for kv blocks [kv0, +32) and [kv0+32, +64)
@@ -447,18 +465,25 @@ __global__ __launch_bounds__(NW * kWaveSize, WPS) void prefill_attn_kernel(
       __builtin_amdgcn_s_setprio(0);
     }
     // stage K/V(t+1) (regs already resident) and prefetch (t+2)
-    if (t + 1 < num_kv_tiles) {
-      if (FP8) kwrite_fp8(kbuf_next);
-      vwrite(vt_next);
-      if (t + 2 < num_kv_tiles) {
-        if (FP8) kload_fp8(t + 2);
-        vload(t + 2);
+    if (FP8) {
+      if (t + 1 < num_kv_tiles) {
+        kwrite_fp8(kbuf_next);
+        vwrite(vt_next);
+        if (t + 2 < num_kv_tiles) {
+          kload_fp8(t + 2);
+          vload(t + 2);
+        }
       }
+      // fp8: register-staged K, no glds in flight — plain barrier
+      __syncthreads();
+    } else {
+      vwrite(vt_next);   // unconditional: garbage past the end, never read
+      vload(t + 2);      // clamped rows
+      // ONE raw barrier: publishes vt[(t+1)&1] (lgkmcnt) and drains
+      // K(t+1) (counted vmcnt) while K(t+2)+V(t+2) STAY IN FLIGHT
+      // across it — the whole point of the 3rd K buffer
+      tile_barrier();
     }
-    // ONE barrier: publishes vt[(t+1)&1]; all waves' PV(t) reads of
-    // vt[t&1] complete (it is rewritten only after the NEXT barrier);
-    // drains the K(t+1) glds that QK^T(t)+PV(t) covered
-    __syncthreads();
   }
 
   // ---- epilogue: normalize (l gathered per crow row) and store ----
