@@ -42,8 +42,9 @@ constexpr float kPNegInf = -1e30f;
 // defer-max threshold (guide T13): skip the O/l rescale while the tile max
 // grows by <= this much — P is then bounded by e^8, which f32 accumulation
 // absorbs (max-abs error ~3x vs THR=0 per the guide's measurement; the
-// spiked-scores test forces the rescale branch).
-constexpr float kDeferThr = 8.0f;
+// spiked-scores test forces the rescale branch). Scores are in BASE-2
+// units (q pre-scaled by log2e), so the e^8 bound is 8*log2e here.
+constexpr float kDeferThr = 11.5416f;
 
 // byte-address XOR swizzles, bijective within a ROWB-byte row (guide T2/G4)
 template <int ROWB>
@@ -143,14 +144,27 @@ __global__ __launch_bounds__(NW * kWaveSize, WPS) void prefill_attn_kernel(
 
   // ---- Q fragments (the QK^T B operand): lane holds
   // Q[row0+col][f*16 + hi1*8 .. +8] for each 16-deep k-chunk f
+  // Q pre-scaled by scale*log2e ONCE: the softmax then runs in base-2
+  // (exp2 of the raw MFMA scores) with no per-tile scale multiply and no
+  // exp argument multiply — ~64 fewer VALU ops per lane per tile.
+  // exp2(s*scale*log2e) == exp(s*scale) exactly as a function; the bf16
+  // re-round of q*scale adds ~2^-9 relative score noise (within the
+  // kernel's fp32-reference tolerance).
   short8 b_q[KF];
   if (active) {
     const int qr = min(row0 + col, seq_len - 1);
     const u16* qrow = q + (seq_start + qr) * q_stride +
                       static_cast<int64_t>(head) * D;
+    const float qs = scale * 1.4426950408889634f;  // log2(e)
 #pragma unroll
-    for (int f = 0; f < KF; ++f)
-      b_q[f] = *reinterpret_cast<const short8*>(qrow + f * 16 + hi1 * 8);
+    for (int f = 0; f < KF; ++f) {
+      short8 raw = *reinterpret_cast<const short8*>(qrow + f * 16 + hi1 * 8);
+      bf16x8 h = __builtin_bit_cast(bf16x8, raw);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        h.h[j] = f32_to_bf16(bf16_to_f32(h.h[j]) * qs);
+      b_q[f] = __builtin_bit_cast(short8, h);
+    }
   }
 
   // online-softmax state: ONE q row per lane (lanes j and j+32 share row j)
@@ -365,17 +379,10 @@ __global__ __launch_bounds__(NW * kWaveSize, WPS) void prefill_attn_kernel(
 #pragma unroll
           for (int r = 0; r < 16; ++r) {
             const int kv_pos = kv0 + 32 * b + crow(r, hi1);
-            float sv = s_acc[b][r] * scale;
             if (!row_valid || kv_pos > q_pos || kv_pos >= k_len)
-              sv = kPNegInf;
-            s_acc[b][r] = sv;
+              s_acc[b][r] = kPNegInf;
           }
-      } else {
-#pragma unroll
-        for (int b = 0; b < 2; ++b)
-#pragma unroll
-          for (int r = 0; r < 16; ++r) s_acc[b][r] *= scale;
-      }
+      }  // interior tiles: scores already scaled (q pre-scale) — no pass
       float rm = kPNegInf;
 #pragma unroll
       for (int b = 0; b < 2; ++b)
@@ -390,7 +397,7 @@ __global__ __launch_bounds__(NW * kWaveSize, WPS) void prefill_attn_kernel(
         if (m_new <= kPNegInf) {
           alpha = 0.f;  // row has seen no unmasked score yet (o is 0)
         } else {
-          alpha = (m_row <= kPNegInf) ? 0.f : __expf(m_row - m_new);
+          alpha = (m_row <= kPNegInf) ? 0.f : __builtin_amdgcn_exp2f(m_row - m_new);
           m_row = m_new;
         }
       }
@@ -401,7 +408,7 @@ __global__ __launch_bounds__(NW * kWaveSize, WPS) void prefill_attn_kernel(
         for (int r = 0; r < 16; ++r) {
           const float pv = (m_row <= kPNegInf || s_acc[b][r] <= kPNegInf)
                                ? 0.f
-                               : __expf(s_acc[b][r] - m_row);
+                               : __builtin_amdgcn_exp2f(s_acc[b][r] - m_row);
           s_acc[b][r] = pv;  // reuse as P (bounded by e^kDeferThr)
           rs += pv;
         }
