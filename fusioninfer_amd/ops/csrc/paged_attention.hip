@@ -103,7 +103,10 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     const u16* q_row = q + seq * q_stride +
                        static_cast<int64_t>(kv_head) * G * D;
     for (int e = tid; e < G * D; e += kNWaves * kWaveSize)
-      q_lds[e / D][e % D] = bf16_to_f32(q_row[e]) * scale;
+      // scale*log2e: scores land in base-2 units so the softmax
+      // uses bare v_exp_f32 (exp2) with no argument multiply
+      q_lds[e / D][e % D] =
+          bf16_to_f32(q_row[e]) * scale * 1.4426950408889634f;
   }
   __syncthreads();
 
@@ -220,7 +223,7 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       for (int off = 4; off < 64; off <<= 1)
         cm = fmaxf(cm, __shfl_xor(cm, off, 64));
       const float m_new = fmaxf(m[g], cm);
-      alpha[g] = __expf(m[g] - m_new);
+      alpha[g] = __builtin_amdgcn_exp2f(m[g] - m_new);
       if (m[g] <= kNegInf && m_new <= kNegInf) alpha[g] = 0.f;
       m[g] = m_new;
     }
@@ -228,7 +231,8 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     // probs + row-sum; every token is replicated on 4 lanes -> scale by 1/4
 #pragma unroll
     for (int g = 0; g < G2; ++g) {
-      float p = (s[g] <= kNegInf) ? 0.f : __expf(s[g] - m[g]);
+      float p = (s[g] <= kNegInf) ? 0.f
+                : __builtin_amdgcn_exp2f(s[g] - m[g]);
       if (quad == 0) p_lds[wave][tok][g] = p;
       float psum = p;
 #pragma unroll
@@ -294,7 +298,8 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     for (int c = 0; c < kCWaves; ++c) {
       const int w = c * HSE + hsplit;
       const float mw = merge_m[w][g];
-      const float f = (mw <= kNegInf) ? 0.f : __expf(mw - gm);
+      const float f = (mw <= kNegInf) ? 0.f
+                      : __builtin_amdgcn_exp2f(mw - gm);
       L += merge_l[w][g] * f;
       o0 = fmaf(merge_acc[w][g][2 * lane], f, o0);
       o1 = fmaf(merge_acc[w][g][2 * lane + 1], f, o1);
@@ -340,7 +345,7 @@ __global__ __launch_bounds__(kWaveSize) void paged_attn_reduce_kernel(
   for (int p = 0; p < num_parts; ++p) {
     const float mp = ml_ws[(base + p) * 2];
     if (mp <= kNegInf) continue;
-    const float f = __expf(mp - gm);
+    const float f = __builtin_amdgcn_exp2f(mp - gm);
     L += ml_ws[(base + p) * 2 + 1] * f;
     const float2 a =
         reinterpret_cast<const float2*>(acc_ws + (base + p) * D)[lane];
