@@ -49,15 +49,33 @@ OWNED_KINDS = [
 ]
 
 
+#: controller-runtime's reconcile-duration bucket boundaries (seconds)
+_TIME_BUCKETS = [0.005, 0.01, 0.025, 0.05, 0.1, 0.25, 0.5, 1.0, 2.5, 5.0,
+                 10.0]
+
+
 class _Metrics:
     def __init__(self):
         self.lock = threading.Lock()
         self.reconcile_total: Dict[str, int] = {}
         self.requeues = 0
+        self.time_hist = [0] * (len(_TIME_BUCKETS) + 1)
+        self.time_sum = 0.0
+        self.time_count = 0
 
     def inc(self, result: str) -> None:
         with self.lock:
             self.reconcile_total[result] = self.reconcile_total.get(result, 0) + 1
+
+    def observe(self, seconds: float) -> None:
+        with self.lock:
+            self.time_sum += seconds
+            self.time_count += 1
+            for i, b in enumerate(_TIME_BUCKETS):
+                if seconds <= b:
+                    self.time_hist[i] += 1
+                    return
+            self.time_hist[-1] += 1
 
     def render(self, queue_depth: int) -> str:
         with self.lock:
@@ -77,6 +95,29 @@ class _Metrics:
             lines.append(
                 "controller_runtime_reconcile_requeues{controller="
                 f'"inferenceservice"}} {self.requeues}'
+            )
+            lines.append(
+                "# TYPE controller_runtime_reconcile_time_seconds histogram"
+            )
+            cum = 0
+            for b, c in zip(_TIME_BUCKETS, self.time_hist):
+                cum += c
+                lines.append(
+                    "controller_runtime_reconcile_time_seconds_bucket"
+                    f'{{controller="inferenceservice",le="{b}"}} {cum}'
+                )
+            cum += self.time_hist[-1]
+            lines.append(
+                "controller_runtime_reconcile_time_seconds_bucket"
+                f'{{controller="inferenceservice",le="+Inf"}} {cum}'
+            )
+            lines.append(
+                "controller_runtime_reconcile_time_seconds_sum"
+                f'{{controller="inferenceservice"}} {self.time_sum}'
+            )
+            lines.append(
+                "controller_runtime_reconcile_time_seconds_count"
+                f'{{controller="inferenceservice"}} {self.time_count}'
             )
             return "\n".join(lines) + "\n"
 
@@ -173,6 +214,13 @@ class Manager:
         kind, ns, name = key
         if not self.is_leader:
             return
+        t0 = time.monotonic()
+        try:
+            self._process_inner(kind, ns, name)
+        finally:
+            self.metrics.observe(time.monotonic() - t0)
+
+    def _process_inner(self, kind: str, ns: str, name: str) -> None:
         try:
             if kind == "ModelLoader":
                 self.modelloader_reconciler.reconcile(name, ns)

@@ -283,6 +283,8 @@ def test_probe_and_metrics_endpoints():
         assert "controller_runtime_reconcile_total" in body
         assert 'result="success"' in body
         assert "workqueue_depth" in body
+        assert "controller_runtime_reconcile_time_seconds_bucket" in body
+        assert "controller_runtime_reconcile_time_seconds_count" in body
     finally:
         mgr.stop()
 
