@@ -43,12 +43,8 @@ constexpr float kNegInf = -1e30f;
 // slice only. G=8 runs HS=2 so the per-wave state is the G=4 footprint —
 // the monolithic G=8 variant spilled ~500 B/lane (W=8) or sat at
 // occupancy 1 (W=4).
-// G2<=2 variants target 3 waves/SIMD (<=170 VGPR); larger per-wave
-// state keeps the unconstrained allocation (2 waves).
 template <int D, int G, int kNWaves, bool FP8, int HS = 1>
-__global__ __launch_bounds__(
-    kNWaves * kWaveSize,
-    (G / ((HS > G) ? 1 : HS) <= 2) ? 3 : 1) void paged_attn_decode_kernel(
+__global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     u16* __restrict__ out,            // [S, Hq, D] (written when 1 partition)
     float* __restrict__ ml_ws,        // [S, Hq, P, 2] (multi-partition)
     float* __restrict__ acc_ws,       // [S, Hq, P, D]
@@ -135,37 +131,22 @@ __global__ __launch_bounds__(
     return ((static_cast<int64_t>(block_id) * num_kv_heads + kv_head) *
             kBlockSz) * D;
   };
-  // DEPTH-2 K prefetch: two kraw sets so a chunk's K is issued TWO
-  // iterations before its phase A (one iteration wasn't enough: PMC
-  // still showed WAIT-dominant; +16 VGPR stays within the 3-wave
-  // budget at G2<=2)
-  uint4 kraw[2][KQ4];
+  uint4 kraw[KQ4];
   auto load_k = [&](int64_t kv_base, uint4* dst) {
     const CT* k_row = k_cache + kv_base + tok * D + quad * DPQ;
 #pragma unroll
     for (int j8 = 0; j8 < KQ4; ++j8)
       dst[j8] = reinterpret_cast<const uint4*>(k_row)[j8];
   };
-  int64_t vbase_q[2] = {0, 0};  // V base for the chunk in kraw[i&1]
-  {
-    const int c0 = chunk_lo + cwave;
-    if (c0 < chunk_hi) {
-      vbase_q[0] = kv_base_of(c0);
-      load_k(vbase_q[0], kraw[0]);
-    }
-    if (c0 + kCWaves < chunk_hi) {
-      vbase_q[1] = kv_base_of(c0 + kCWaves);
-      load_k(vbase_q[1], kraw[1]);
-    }
+  int64_t kv_base = 0;
+  if (chunk_lo + cwave < chunk_hi) {
+    kv_base = kv_base_of(chunk_lo + cwave);
+    load_k(kv_base, kraw);
   }
 
-  // chunk body templated on the kraw/vbase SLOT (compile-time index —
-  // a runtime `it & 1` sent the prefetch buffers to scratch, guide
-  // rule 20: 144 B/lane)
-  auto chunk_body = [&](int chunk, auto slot_c) {
-    constexpr int SLOT = decltype(slot_c)::value;
+  for (int chunk = chunk_lo + cwave; chunk < chunk_hi; chunk += kCWaves) {
     const int token_pos = chunk * kBlockSz + tok;
-    const int64_t kv_base_cur = vbase_q[SLOT];
+    const int64_t kv_base_cur = kv_base;
 
     // ---- phase A: scores for 16 tokens x G2 heads ----
     // streaming convert+FMA: per 16-B K group, convert 8 elements and
@@ -180,12 +161,11 @@ __global__ __launch_bounds__(
       for (int j8 = 0; j8 < KQ4; ++j8) {
         float kf8[16 / sizeof(CT)];
         if (FP8) {
-          const uint4 kr = kraw[SLOT][j8];
-          const u32 w[4] = {kr.x, kr.y, kr.z, kr.w};
+          const u32 w[4] = {kraw[j8].x, kraw[j8].y, kraw[j8].z, kraw[j8].w};
 #pragma unroll
           for (int i = 0; i < 4; ++i) unpack_fp8x4(w[i], &kf8[i * 4]);
         } else {
-          const bf16x8 kv8 = __builtin_bit_cast(bf16x8, kraw[SLOT][j8]);
+          const bf16x8 kv8 = __builtin_bit_cast(bf16x8, kraw[j8]);
 #pragma unroll
           for (int j = 0; j < 8; ++j) kf8[j] = bf16_to_f32(kv8.h[j]);
         }
@@ -218,10 +198,9 @@ __global__ __launch_bounds__(
             : *reinterpret_cast<const u32*>(v_rows + t * D + 2 * lane);
     }
     // issue next chunk's K now; it lands under softmax + PV
-    if (chunk + 2 * kCWaves < chunk_hi) {
-      // refill the slot just consumed with the chunk TWO iterations out
-      vbase_q[SLOT] = kv_base_of(chunk + 2 * kCWaves);
-      load_k(vbase_q[SLOT], kraw[SLOT]);
+    if (chunk + kCWaves < chunk_hi) {
+      kv_base = kv_base_of(chunk + kCWaves);
+      load_k(kv_base, kraw);
     }
     // fence: stop the scheduler from hoisting phase-B work (and its live
     // ranges) above the softmax — cross-phase overlap doubled VGPRs
@@ -289,18 +268,6 @@ __global__ __launch_bounds__(
           acc[g][1] = fmaf(p, v1, acc[g][1]);
         }
       }
-    }
-  };
-
-  // ping-pong driver: even iterations consume slot 0, odd slot 1
-  {
-    int chunk = chunk_lo + cwave;
-    while (chunk < chunk_hi) {
-      chunk_body(chunk, std::integral_constant<int, 0>{});
-      chunk += kCWaves;
-      if (chunk >= chunk_hi) break;
-      chunk_body(chunk, std::integral_constant<int, 1>{});
-      chunk += kCWaves;
     }
   }
 
