@@ -52,6 +52,9 @@ def parse_args():
     p.add_argument("--gen-len", type=int, default=128)
     p.add_argument("--concurrency", type=int, default=256)
     p.add_argument("--max-batched-tokens", type=int, default=8192)
+    p.add_argument("--admission-ms", type=float, default=None,
+                   help="admission-hysteresis aging window (scheduler "
+                        "default when unset)")
     p.add_argument("--enforce-eager", action="store_true")
     p.add_argument("--kv-cache-dtype", choices=["auto", "fp8"], default="auto",
                    help="fp8 = e4m3 KV cache (halves KV bytes)")
@@ -186,6 +189,8 @@ def main():
             max_num_seqs=args.concurrency,
             max_num_batched_tokens=args.max_batched_tokens,
             max_model_len=max_len,
+            **({"prefill_admission_ms": args.admission_ms}
+               if args.admission_ms is not None else {}),
         ),
         parallel=ParallelConfig(
             tensor_parallel_size=args.tp, rank=rank, world_size=world

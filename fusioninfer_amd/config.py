@@ -81,6 +81,10 @@ class SchedulerConfig:
     # are waiting (or nothing is decoding), so most steps stay pure-decode
     # and take the hipGraph path; 0 = admit eagerly every step
     prefill_admission_tokens: int = 8192
+    # aging escape for the hysteresis: never hold a prompt longer than
+    # this (ms). At high load the queue reaches the token threshold
+    # first; at low load this bounds the TTFT cost of batching.
+    prefill_admission_ms: float = 50.0
     # "fcfs" (default) or "priority" (vLLM --scheduling-policy): priority
     # orders admission by (priority, arrival) and preempts the
     # lowest-priority running sequence first (lower value = higher prio)

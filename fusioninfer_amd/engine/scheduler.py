@@ -145,11 +145,13 @@ class Scheduler:
             admit = True
         else:
             admit = waiting_tokens >= threshold
-            # aging escape: at low load the batching threshold may never be
-            # reached — never hold a prompt longer than 50 ms
+            # aging escape: at low load the batching threshold may never
+            # be reached — never hold a prompt longer than the configured
+            # window (prefill_admission_ms)
             if not admit and self.waiting:
                 admit = (
-                    time.monotonic() - self.waiting[0].arrival_time > 0.05
+                    time.monotonic() - self.waiting[0].arrival_time
+                    > self.cfg.prefill_admission_ms / 1000.0
                 )
         while (
             admit
