@@ -96,6 +96,22 @@ class Sampler:
                 1, idx, sorted_logits
             )
 
+        # min-p (vLLM surface): drop tokens whose probability is below
+        # min_p * max-probability of the row
+        m_rows = [
+            i for i, s in enumerate(seqs)
+            if s.sampling.temperature != 0 and s.sampling.min_p > 0.0
+        ]
+        if m_rows:
+            mps = torch.tensor(
+                [seqs[i].sampling.min_p for i in m_rows],
+                device=logits.device,
+            ).unsqueeze(1)
+            sub = scaled[m_rows]
+            pr = torch.softmax(sub, dim=-1)
+            cut = pr < mps * pr.max(dim=-1, keepdim=True).values
+            scaled[m_rows] = sub.masked_fill(cut, float("-inf"))
+
         probs = torch.softmax(scaled, dim=-1)
         sampled = torch.empty_like(greedy)
         # group rows by generator: per-request seeds get their own draw

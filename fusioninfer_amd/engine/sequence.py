@@ -23,6 +23,7 @@ class SamplingParams:
         temperature: float = 0.0,
         top_p: float = 1.0,
         top_k: int = 0,
+        min_p: float = 0.0,
         repetition_penalty: float = 1.0,
         presence_penalty: float = 0.0,
         frequency_penalty: float = 0.0,
@@ -39,6 +40,7 @@ class SamplingParams:
         self.temperature = temperature
         self.top_p = top_p
         self.top_k = top_k
+        self.min_p = min_p  # drop tokens with p < min_p * max(p)
         self.repetition_penalty = repetition_penalty
         self.presence_penalty = presence_penalty
         self.frequency_penalty = frequency_penalty

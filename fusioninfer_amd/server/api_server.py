@@ -227,6 +227,8 @@ def build_app(serving: ServingEngine, model_name: str,
             raise ValueError("top_p must be in (0, 1]")
         if int(body.get("top_k", 0)) < 0:
             raise ValueError("top_k must be >= 0")
+        if not 0.0 <= float(body.get("min_p", 0.0)) <= 1.0:
+            raise ValueError("min_p must be in [0, 1]")
         stop_ids = list(body.get("stop_token_ids") or [])
         if tokenizer.eos_token_id is not None \
                 and tokenizer.eos_token_id not in stop_ids:
@@ -237,6 +239,7 @@ def build_app(serving: ServingEngine, model_name: str,
             temperature=float(body.get("temperature", 1.0)),
             top_p=float(body.get("top_p", 1.0)),
             top_k=int(body.get("top_k", 0)),
+            min_p=float(body.get("min_p", 0.0)),
             repetition_penalty=float(body.get("repetition_penalty", 1.0)),
             presence_penalty=float(body.get("presence_penalty", 0.0)),
             frequency_penalty=float(body.get("frequency_penalty", 0.0)),

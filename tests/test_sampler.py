@@ -60,3 +60,29 @@ def test_mixed_greedy_and_sampled_batch():
     seqs = [_seq(), _seq(temperature=1.0, seed=5), _seq(temperature=0.7)]
     out = Sampler().sample(logits.clone(), seqs)
     assert out[0].item() == int(logits[0].argmax())
+
+
+def test_min_p_filters_tail():
+    """min_p drops tokens below min_p * max-prob (vLLM surface)."""
+    import torch
+
+    from fusioninfer_amd.engine.sampler import Sampler
+    from fusioninfer_amd.engine.sequence import SamplingParams, Sequence
+
+    torch.manual_seed(0)
+    logits = torch.tensor([[5.0, 4.9, 0.0, -2.0, -50.0]])
+    sampler = Sampler()
+    picks = set()
+    for i in range(50):  # fresh seed per draw (seeded draws are keyed on
+        s = Sequence("a", [1, 2],  # seed + output length)
+                     SamplingParams(temperature=1.0, min_p=0.5, seed=i))
+        picks.add(int(sampler.sample(logits, [s])[0]))
+    # only tokens 0 and 1 survive (p1/p0 ~ 0.90; others < 0.5*max)
+    assert picks <= {0, 1} and picks
+    # min_p=0 keeps the tail reachable at high temperature
+    picks2 = set()
+    for i in range(200):
+        s2 = Sequence("b", [1, 2],
+                      SamplingParams(temperature=10.0, min_p=0.0, seed=i))
+        picks2.add(int(sampler.sample(logits, [s2])[0]))
+    assert len(picks2) >= 3
