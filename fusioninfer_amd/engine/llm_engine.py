@@ -40,6 +40,9 @@ class RequestOutput:
             self.output_token_ids[-1:] if self.output_token_ids else []
         )
         self.finished = seq.is_finished()
+        self.finish_reason = (
+            (seq.finish_reason or "stop") if self.finished else None
+        )
         self.logprobs = list(seq.logprobs)
         self.prompt_logprobs = list(seq.prompt_logprobs)
         self.ttft = seq.ttft
@@ -558,8 +561,11 @@ class LLMEngine:
             self.num_generated_tokens += 1
             if guided is not None:
                 guided.advance_token(int(tok))
-            if seq.check_stop() or (guided is not None
-                                    and guided.is_terminal()):
+            if seq.check_stop():
+                self._finish_seq(seq)
+                break
+            if guided is not None and guided.is_terminal():
+                seq.finish_reason = seq.finish_reason or "stop"
                 self._finish_seq(seq)
                 break
         return RequestOutput(seq, new_token_ids=new)
@@ -581,6 +587,7 @@ class LLMEngine:
             if g is not None:
                 allowed = g.allowed_mask(logits_f.device)
                 if allowed is None:
+                    s.finish_reason = s.finish_reason or "stop"
                     self._finish_seq(s)
                     outputs.append(RequestOutput(s, new_token_ids=[]))
                     continue

@@ -79,6 +79,9 @@ class Sequence:
         self.num_computed_tokens = 0
         # PD producer: keep cache blocks alive after finish for KV export
         self.hold_blocks = False
+        # OpenAI finish_reason: "length" (max_tokens) or "stop" (eos /
+        # stop token / grammar terminal); None while running
+        self.finish_reason = None
         self.swap_num_blocks = 0
         # LoRA adapter name (None = base model)
         self.lora_name = None
@@ -113,14 +116,16 @@ class Sequence:
 
     def check_stop(self) -> bool:
         s = self.sampling
-        if len(self.output_token_ids) >= s.max_tokens:
-            return True
         if (
             not s.ignore_eos
             and self.output_token_ids
             and len(self.output_token_ids) >= s.min_tokens
             and self.output_token_ids[-1] in s.stop_token_ids
         ):
+            self.finish_reason = "stop"
+            return True
+        if len(self.output_token_ids) >= s.max_tokens:
+            self.finish_reason = self.finish_reason or "length"
             return True
         return False
 

@@ -258,19 +258,32 @@ def build_app(serving: ServingEngine, model_name: str,
             return []
         return [stops] if isinstance(stops, str) else list(stops)
 
+    def _finish_reason(rid, by_stop_string):
+        """OpenAI finish_reason for a finished stream: stop-string cuts
+        are "stop"; otherwise the engine's reason ("length"/"stop")."""
+        if by_stop_string:
+            return "stop"
+        fin = serving.peek_final(rid)
+        return getattr(fin, "finish_reason", None) or "stop"
+
     async def _collect(q, stops=None, req_id=None, keep_stop=False):
         """Drain a request's token stream. With OpenAI `stop` strings the
         generated text is truncated BEFORE the first stop match (AFTER it
         with include_stop_str_in_output) and the engine request is
         aborted. Text accumulates through incremental detokenization so
         multi-byte characters split across BPE tokens decode correctly.
-        Returns (token_ids, text)."""
+        Returns (token_ids, text, finish_reason)."""
         from fusioninfer_amd.tokenizer import IncrementalDetokenizer
 
         loop = asyncio.get_event_loop()
         toks: List[int] = []
         text = ""
         detok = IncrementalDetokenizer(tokenizer) if stops else None
+
+        def engine_reason():
+            fin = serving.peek_final(req_id)
+            return getattr(fin, "finish_reason", None) or "stop"
+
         while True:
             tok, finished = await loop.run_in_executor(None, q.get)
             if tok is not None:
@@ -281,16 +294,16 @@ def build_app(serving: ServingEngine, model_name: str,
                     if hit is not None:
                         serving.abort(req_id)
                         end = hit[0] + len(hit[1]) if keep_stop else hit[0]
-                        return toks, text[:end]
+                        return toks, text[:end], "stop"
             if finished:
                 if not stops:
-                    return toks, tokenizer.decode(toks)
+                    return toks, tokenizer.decode(toks), engine_reason()
                 text += detok.flush()
                 hit = _find_stop(text, stops)
                 if hit is not None:
                     end = hit[0] + len(hit[1]) if keep_stop else hit[0]
-                    return toks, text[:end]
-                return toks, text
+                    return toks, text[:end], "stop"
+                return toks, text, engine_reason()
 
     async def _stream(q) -> AsyncGenerator:
         loop = asyncio.get_event_loop()
@@ -466,10 +479,12 @@ def build_app(serving: ServingEngine, model_name: str,
                             if tok is not None:
                                 n_out += 1
                             f = filts[idx]
+                            by_stop = False
                             if f is not None:
                                 delta, hit = f.push(delta)
                                 if hit:
                                     finished = True
+                                    by_stop = True
                                     serving.abort(rid)
                                 elif finished:
                                     delta += f.flush()
@@ -484,7 +499,8 @@ def build_app(serving: ServingEngine, model_name: str,
                                     "token_ids":
                                         [tok] if tok is not None else [],
                                     "finish_reason":
-                                        "stop" if finished else None,
+                                        _finish_reason(rid, by_stop)
+                                        if finished else None,
                                 }],
                             }
                             yield f"data: {json.dumps(chunk)}\n\n"
@@ -536,12 +552,14 @@ def build_app(serving: ServingEngine, model_name: str,
                         delta = detok.push(tok) if tok is not None else ""
                         if finished:
                             delta += detok.flush()
+                        by_stop = False
                         if filt is not None:
                             # hold back text that could extend into a stop
                             # string; cut the stream at the first match
                             delta, hit = filt.push(delta)
                             if hit:
                                 finished = True
+                                by_stop = True
                             elif finished:
                                 delta += filt.flush()
                         chunk = {
@@ -554,7 +572,9 @@ def build_app(serving: ServingEngine, model_name: str,
                                     "index": 0,
                                     "text": delta,
                                     "token_ids": [tok] if tok is not None else [],
-                                    "finish_reason": "stop" if finished else None,
+                                    "finish_reason":
+                                        _finish_reason(req_id, by_stop)
+                                        if finished else None,
                                 }
                             ],
                         }
@@ -597,10 +617,11 @@ def build_app(serving: ServingEngine, model_name: str,
             ))
         choices = []
         try:
-            toks, text = await _collect(q, stops=stops, req_id=req_id,
-                                        keep_stop=keep_stop)
+            toks, text, reason0 = await _collect(q, stops=stops,
+                                                 req_id=req_id,
+                                                 keep_stop=keep_stop)
             final = serving.take_final(req_id)
-            choices.append((toks, text))
+            choices.append((toks, text, reason0))
             for rid_i, q_i in extra:
                 choices.append(await _collect(q_i, stops=stops,
                                               req_id=rid_i,
@@ -613,8 +634,8 @@ def build_app(serving: ServingEngine, model_name: str,
             raise
         if echo:
             prefix = tokenizer.decode(prompt_ids)
-            choices = [(t, prefix + x) for t, x in choices]
-        total_completion = sum(len(t) for t, _ in choices)
+            choices = [(t, prefix + x, r) for t, x, r in choices]
+        total_completion = sum(len(t) for t, _, _ in choices)
         return JSONResponse(
             {
                 "id": cid,
@@ -626,7 +647,7 @@ def build_app(serving: ServingEngine, model_name: str,
                         "index": i,
                         "text": c_text,
                         "token_ids": c_toks,
-                        "finish_reason": "stop",
+                        "finish_reason": c_reason,
                         **({"logprobs": final.logprobs,
                             "prompt_logprobs": final.prompt_logprobs}
                            if i == 0 and final is not None
@@ -634,7 +655,7 @@ def build_app(serving: ServingEngine, model_name: str,
                                 or body.get("prompt_logprobs"))
                            else {}),
                     }
-                    for i, (c_toks, c_text) in enumerate(choices)
+                    for i, (c_toks, c_text, c_reason) in enumerate(choices)
                 ],
                 "usage": {
                     "prompt_tokens": len(prompt_ids),
@@ -727,10 +748,12 @@ def build_app(serving: ServingEngine, model_name: str,
                             if finished:
                                 piece += detoks[idx].flush()
                             f = filts[idx]
+                            by_stop = False
                             if f is not None:
                                 piece, hit = f.push(piece)
                                 if hit:
                                     finished = True
+                                    by_stop = True
                                     serving.abort(rid)
                                 elif finished:
                                     piece += f.flush()
@@ -745,7 +768,8 @@ def build_app(serving: ServingEngine, model_name: str,
                                     "index": idx,
                                     "delta": delta,
                                     "finish_reason":
-                                        "stop" if finished else None,
+                                        _finish_reason(rid, by_stop)
+                                        if finished else None,
                                 }],
                             }
                             yield f"data: {json.dumps(chunk)}\n\n"
@@ -780,11 +804,13 @@ def build_app(serving: ServingEngine, model_name: str,
                         )
                         if finished:
                             piece += detok.flush()
+                        by_stop = False
                         if filt is not None:
                             # cut the stream at the first stop-string match
                             piece, hit = filt.push(piece)
                             if hit:
                                 finished = True
+                                by_stop = True
                             elif finished:
                                 piece += filt.flush()
                         if piece:
@@ -799,7 +825,8 @@ def build_app(serving: ServingEngine, model_name: str,
                                     "index": 0,
                                     "delta": delta,
                                     "finish_reason": (
-                                        "stop" if finished else None
+                                        _finish_reason(req_id, by_stop)
+                                        if finished else None
                                     ),
                                 }
                             ],
@@ -840,7 +867,7 @@ def build_app(serving: ServingEngine, model_name: str,
             for rid_i, _ in extra:
                 serving.abort(rid_i)
             raise
-        total = sum(len(t) for t, _ in choices)
+        total = sum(len(t) for t, _, _ in choices)
         return JSONResponse(
             {
                 "id": cid,
@@ -854,9 +881,9 @@ def build_app(serving: ServingEngine, model_name: str,
                             "role": "assistant",
                             "content": c_text,
                         },
-                        "finish_reason": "stop",
+                        "finish_reason": c_reason,
                     }
-                    for i, (_t, c_text) in enumerate(choices)
+                    for i, (_t, c_text, c_reason) in enumerate(choices)
                 ],
                 "usage": {
                     "prompt_tokens": len(prompt_ids),

@@ -246,6 +246,13 @@ class ServingEngine:
         with self._lock:
             return self._final.pop(request_id, None)
 
+    def peek_final(self, request_id: str):
+        """Non-destructive read of the finished RequestOutput (finish
+        reason lookups must not steal the logprobs consumer's pop; the
+        _final dict stays bounded by the 4096-entry cap)."""
+        with self._lock:
+            return self._final.get(request_id)
+
     def abort(self, request_id: str) -> bool:
         with self._lock:
             self._streams.pop(request_id, None)
@@ -280,14 +287,16 @@ class ServingEngine:
                 q = self._streams.get(out.request_id)
                 if q is None:
                     continue
-                # speculative steps emit up to k+1 tokens at once; the
-                # finished flag rides on the last one
-                toks = out.new_token_ids or [None]
-                for i, tok in enumerate(toks):
-                    q.put((tok, out.finished and i == len(toks) - 1))
+                # record the final BEFORE pushing the finished flag so a
+                # consumer that wakes on it can take_final() immediately
                 if out.finished:
                     with self._lock:
                         self._streams.pop(out.request_id, None)
                         self._final[out.request_id] = out
                         if len(self._final) > 4096:  # belt-and-braces cap
                             self._final.pop(next(iter(self._final)))
+                # speculative steps emit up to k+1 tokens at once; the
+                # finished flag rides on the last one
+                toks = out.new_token_ids or [None]
+                for i, tok in enumerate(toks):
+                    q.put((tok, out.finished and i == len(toks) - 1))

@@ -521,7 +521,7 @@ def test_streaming_n_choices(app):
                     per_choice.setdefault(ch["index"], []).extend(
                         ch["token_ids"]
                     )
-                    if ch["finish_reason"] == "stop":
+                    if ch["finish_reason"] in ("stop", "length"):
                         finish[ch["index"]] = True
             assert set(per_choice) == {0, 1}
             assert all(len(v) == 3 for v in per_choice.values())
@@ -624,5 +624,45 @@ def test_streaming_usage_not_inflated(app):
             assert usage is not None
             assert usage["prompt_tokens"] == 5  # len("hello") bytes, once
             assert usage["completion_tokens"] == n_toks == 8
+
+    asyncio.run(run())
+
+
+def test_finish_reason_length_vs_stop(app):
+    """OpenAI semantics: max_tokens exhaustion -> "length"; stop-string
+    or stop-token -> "stop" (both transports)."""
+
+    async def run():
+        async with _client(app) as c:
+            r = await c.post("/v1/completions", json={
+                "prompt": "abcabc", "max_tokens": 4, "temperature": 0,
+                "ignore_eos": True,
+            })
+            assert r.json()["choices"][0]["finish_reason"] == "length"
+
+            full = (await c.post("/v1/completions", json={
+                "prompt": "abcabc", "max_tokens": 8, "temperature": 0,
+                "ignore_eos": True,
+            })).json()["choices"][0]["text"]
+            stop_ch = full[1]
+            r = await c.post("/v1/completions", json={
+                "prompt": "abcabc", "max_tokens": 8, "temperature": 0,
+                "ignore_eos": True, "stop": stop_ch,
+            })
+            assert r.json()["choices"][0]["finish_reason"] == "stop"
+
+            # streaming: the final chunk carries "length"
+            reasons = []
+            async with c.stream("POST", "/v1/completions", json={
+                "prompt": "xy", "max_tokens": 3, "stream": True,
+                "temperature": 0, "ignore_eos": True,
+            }) as resp:
+                async for line in resp.aiter_lines():
+                    if line.startswith("data: ") and "[DONE]" not in line:
+                        fr = json.loads(line[6:])["choices"][0][
+                            "finish_reason"]
+                        if fr:
+                            reasons.append(fr)
+            assert reasons == ["length"]
 
     asyncio.run(run())
