@@ -12,7 +12,6 @@ via the Envoy ext-proc protocol to the running worker."""
 import json
 import subprocess
 import sys
-import time
 import urllib.request
 
 import pytest
