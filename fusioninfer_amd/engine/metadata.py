@@ -22,6 +22,7 @@ class AttnMetadata:
     cu_seqlens: Optional[torch.Tensor] = None    # [P+1] int32, NEW tokens
     tile_seq: Optional[torch.Tensor] = None      # kernel tile table
     tile_row0: Optional[torch.Tensor] = None
+    tile_rows: int = 0                           # rows per tile (0 = default)
     prefill_block_tables: Optional[torch.Tensor] = None  # [P, max_blocks]
     prefill_seq_lens_k: Optional[torch.Tensor] = None    # [P] total ctx
     # decode segment

@@ -228,8 +228,9 @@ class ModelRunner:
     def _prefill_inputs(self, payload):
         dev = self.device
         cu = payload["cu"]
+        tile_rows = ops_mod.prefill_tile_rows(payload["new_lens"])
         tile_seq, tile_row0 = ops_mod.build_prefill_tiles(
-            payload["new_lens"], device=dev
+            payload["new_lens"], device=dev, rows=tile_rows
         )
         meta = AttnMetadata(
             num_prefill_tokens=cu[-1],
@@ -242,6 +243,7 @@ class ModelRunner:
             ),
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
             tile_seq=tile_seq,
+            tile_rows=tile_rows,
             tile_row0=tile_row0,
             prefill_block_tables=torch.tensor(
                 payload["bt"], dtype=torch.int32, device=dev
@@ -519,8 +521,9 @@ class ModelRunner:
         cu = payload["cu"]
         np_ = cu[-1]
         nd = len(d["ids"])
+        tile_rows = ops_mod.prefill_tile_rows(payload["new_lens"])
         tile_seq, tile_row0 = ops_mod.build_prefill_tiles(
-            payload["new_lens"], device=dev
+            payload["new_lens"], device=dev, rows=tile_rows
         )
         max_blocks = max(len(b) for b in d["bt"])
         dbt_np = np.zeros((nd, max_blocks), dtype=np.int32)
@@ -539,6 +542,7 @@ class ModelRunner:
             ),
             cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
             tile_seq=tile_seq,
+            tile_rows=tile_rows,
             tile_row0=tile_row0,
             prefill_block_tables=torch.tensor(
                 payload["bt"], dtype=torch.int32, device=dev

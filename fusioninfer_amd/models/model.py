@@ -101,6 +101,7 @@ class Attention(nn.Module):
                 self.scale,
                 tile_seq=meta.tile_seq,
                 tile_row0=meta.tile_row0,
+                tile_rows=meta.tile_rows or None,
             )
             out[:np_] = o.view(np_, self.q_size)
         if nd > 0:

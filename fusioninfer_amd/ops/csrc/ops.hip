@@ -32,7 +32,7 @@ void launch_paged_attn_decode(u16*, float*, float*, const u16*, const void*,
 void launch_prefill_attn(u16*, const u16*, const void*, const void*,
                          const int*, const int*, const int*, const int*,
                          const int*, int, int, int64_t, int64_t, int64_t, int,
-                         int, int, float, bool, hipStream_t);
+                         int, int, float, bool, int, hipStream_t);
 void launch_moe_gemm(u16*, const u16*, const u16*, const int*, const int*,
                      const int*, int, int, int, int, bool, hipStream_t);
 void launch_moe_gemm_fp8(u16*, const unsigned char*, const float*,
@@ -272,7 +272,10 @@ void paged_attention_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
 
 void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
                        at::Tensor v, at::Tensor tile_seq, at::Tensor tile_row0,
-                       at::Tensor cu_seqlens, double scale) {
+                       at::Tensor cu_seqlens, double scale,
+                       int64_t tile_rows) {
+  TORCH_CHECK(tile_rows == 128 || tile_rows == 256,
+              "tile_rows must be 128 or 256");
   CHECK_BF16_CUDA(out);
   CHECK_BF16_CUDA(q);
   TORCH_CHECK(tile_seq.scalar_type() == at::kInt &&
@@ -290,14 +293,17 @@ void prefill_attention(at::Tensor out, at::Tensor q, at::Tensor k,
       tile_seq.data_ptr<int>(), tile_row0.data_ptr<int>(),
       cu_seqlens.data_ptr<int>(), nullptr, nullptr, 0, tile_seq.size(0),
       q.stride(0), k.stride(0), v.stride(0), num_q_heads, num_kv_heads,
-      head_dim, static_cast<float>(scale), false, current_stream());
+      head_dim, static_cast<float>(scale), false,
+      static_cast<int>(tile_rows / 32), current_stream());
 }
 
 void prefill_attention_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                              at::Tensor v_cache, at::Tensor tile_seq,
                              at::Tensor tile_row0, at::Tensor cu_seqlens,
                              at::Tensor block_tables, at::Tensor seq_lens_k,
-                             double scale) {
+                             double scale, int64_t tile_rows) {
+  TORCH_CHECK(tile_rows == 128 || tile_rows == 256,
+              "tile_rows must be 128 or 256");
   CHECK_BF16_CUDA(out);
   CHECK_BF16_CUDA(q);
   const bool fp8 = cache_is_fp8(k_cache);
@@ -319,7 +325,8 @@ void prefill_attention_paged(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       cu_seqlens.data_ptr<int>(), block_tables.data_ptr<int>(),
       seq_lens_k.data_ptr<int>(), block_tables.size(1), tile_seq.size(0),
       q.stride(0), 0, 0, num_q_heads, num_kv_heads, head_dim,
-      static_cast<float>(scale), fp8, current_stream());
+      static_cast<float>(scale), fp8,
+      static_cast<int>(tile_rows / 32), current_stream());
 }
 
 void moe_gemm(at::Tensor out, at::Tensor a, at::Tensor b_packed,

@@ -516,15 +516,11 @@ void launch_prefill_attn(u16* out, const u16* q, const void* k, const void* v,
                          const int* seq_lens_k, int max_blocks, int ntiles,
                          int64_t q_stride, int64_t k_stride, int64_t v_stride,
                          int num_q_heads, int num_kv_heads, int head_dim,
-                         float scale, bool fp8, hipStream_t stream) {
+                         float scale, bool fp8, int nw, hipStream_t stream) {
   const bool paged = block_tables != nullptr;
-  // FI_PF_NW=4 runs 4-wave workgroups (128 q rows): the host tile table
-  // must be built with the matching PREFILL_TILE_ROWS (ops/__init__.py
-  // reads the same env)
-  static const int nw = [] {
-    const char* e = getenv("FI_PF_NW");
-    return (e && e[0] == '4') ? 4 : 8;
-  }();
+  // nw = waves per workgroup (4 or 8), chosen per batch by the host
+  // (ops.prefill_tile_rows: 128-row tiles for long sequences); the tile
+  // table must have been built with rows == nw*32
   dim3 grid(ntiles, num_q_heads), block(nw * kWaveSize);
 #define FI_PF_LAUNCH(DD, PP, F8)                                              \
   do {                                                                        \
