@@ -53,7 +53,13 @@ FI_DEV short8 load_b16x8(const u16* p) {
 // of GPU time, A-re-read-bound.
 // GATE_UP: B is packed over 2N columns (gate at n, up at n + N);
 // epilogue writes silu(gate)*up. Otherwise a plain grouped GEMM.
-template <int WM, int WN, int MITER, int NITER, bool GATE_UP>
+// STAGE_A (BM=128 variants): the block's A tile goes through LDS via
+// global_load_lds (double-buffered, T2 source-side swizzle, ONE barrier
+// per K-step — the guide's 2-phase grouped-GEMM shape), so the four
+// waves share one gather of the 128x32 panel instead of issuing
+// per-wave fragment loads against L2/L3.
+template <int WM, int WN, int MITER, int NITER, bool GATE_UP,
+          bool STAGE_A = false>
 __global__ __launch_bounds__(256) void moe_gemm_kernel(
     u16* __restrict__ out,              // [PM, N] bf16
     const u16* __restrict__ a,          // GATE_UP: x [T, K]; else act [PM, K]
@@ -63,6 +69,7 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
     const int* __restrict__ n_valid,    // device scalar: real m-tile count
     const int K, const int N) {
   constexpr int BM = 16 * WM * MITER;
+  static_assert(!STAGE_A || BM == 128, "STAGE_A is tuned for BM=128");
   const int mtile = blockIdx.x;
   if (mtile >= *n_valid) return;
 
@@ -100,6 +107,56 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
     a_p[mi] = a + static_cast<int64_t>(row) * K + (lane >> 4) * 8;
   }
 
+  // ---- STAGE_A machinery (BM=128): [128 rows][64 B] tile, rows
+  // swizzled byte^=(row&3)<<4 (T2; glds dest is lane-linear so the
+  // swizzle rides on the SOURCE address, rule 21)
+  constexpr int kARowB = 64;                    // 32 bf16 per row
+  constexpr int kATileB = 128 * kARowB;         // 8 KiB
+  __shared__ u16 a_lds[STAGE_A ? 2 * kATileB / 2 : 1];
+  const int tid = threadIdx.x;
+  // this thread's two 16-B staging pieces per K-step
+  int stage_src_off[2];   // element offset within the source row
+  int64_t stage_row_off[2];
+  int stage_dst[2];
+  if (STAGE_A) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int X = (i * 256 + tid) * 16;       // linear LDS byte
+      const int srow = X / kARowB;
+      const int sbyte = (X % kARowB) ^ ((srow & 3) << 4);
+      const int gr = GATE_UP ? sorted_ids[mtile * BM + srow]
+                             : mtile * BM + srow;
+      stage_row_off[i] = static_cast<int64_t>(gr) * K;
+      stage_src_off[i] = sbyte / 2;
+      stage_dst[i] = X / 2;
+    }
+  }
+  auto stage_a = [&](int kt, int buf) {
+    if (!STAGE_A) return;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const u16* src = a + stage_row_off[i] + kt * 32 + stage_src_off[i];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(
+              a_lds + buf * (kATileB / 2) + stage_dst[i]),
+          16, 0, 0);
+    }
+  };
+  // LDS read offsets for this wave's MITER A fragments (elements)
+  int a_lds_off[MITER];
+  if (STAGE_A) {
+#pragma unroll
+    for (int mi = 0; mi < MITER; ++mi) {
+      const int srow = wm * (16 * MITER) + mi * 16 + (lane & 15);
+      const int byte = ((lane >> 4) * 16) ^ ((srow & 3) << 4);
+      a_lds_off[mi] = (srow * kARowB + byte) / 2;
+    }
+    stage_a(0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+
   floatx4 acc_g[MITER][NITER];
   floatx4 acc_u[GATE_UP ? MITER : 1][NITER];
 #pragma unroll
@@ -112,10 +169,13 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
 
   const int ksteps = K / 32;
   for (int kt = 0; kt < ksteps; ++kt) {
+    if (STAGE_A && kt + 1 < ksteps) stage_a(kt + 1, (kt + 1) & 1);
     short8 av[MITER];
+    const u16* abuf = STAGE_A ? a_lds + (kt & 1) * (kATileB / 2) : nullptr;
 #pragma unroll
     for (int mi = 0; mi < MITER; ++mi)
-      av[mi] = load_b16x8(a_p[mi] + kt * 32);
+      av[mi] = STAGE_A ? load_b16x8(abuf + a_lds_off[mi])
+                       : load_b16x8(a_p[mi] + kt * 32);
 #pragma unroll
     for (int ni = 0; ni < NITER; ++ni) {
       const short8 bg = load_b16x8(bg_p[ni] + kt * b_step);
@@ -129,6 +189,12 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
           acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               av[mi], bu, acc_u[mi][ni], 0, 0, 0);
       }
+    }
+    if (STAGE_A) {
+      // 2-phase barrier: drains this step's prefetch glds (next tile
+      // becomes readable) and closes the LDS reads of the current one
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
     }
   }
 
@@ -319,16 +385,31 @@ void launch_moe_gemm(u16* out, const u16* a, const u16* b,
                          dim3(256), 0, stream, out, a, b, sorted_ids,
                          expert_ids, n_valid, K, N);
   } else {  // block_m == 128
+    // FI_MOE_STAGE_A=0 falls back to the register-gather form (A/B knob)
+    static const bool stage_a = [] {
+      const char* e = getenv("FI_MOE_STAGE_A");
+      return !(e && e[0] == '0');
+    }();
     if (gate_up) {
       const dim3 grid(max_mtiles, N / 64);
-      hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 2, true>), grid, dim3(256),
-                         0, stream, out, a, b, sorted_ids, expert_ids,
-                         n_valid, K, N);
+      if (stage_a)
+        hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 2, true, true>), grid,
+                           dim3(256), 0, stream, out, a, b, sorted_ids,
+                           expert_ids, n_valid, K, N);
+      else
+        hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 2, true, false>), grid,
+                           dim3(256), 0, stream, out, a, b, sorted_ids,
+                           expert_ids, n_valid, K, N);
     } else {
       const dim3 grid(max_mtiles, N / 128);
-      hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 4, false>), grid,
-                         dim3(256), 0, stream, out, a, b, sorted_ids,
-                         expert_ids, n_valid, K, N);
+      if (stage_a)
+        hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 4, false, true>), grid,
+                           dim3(256), 0, stream, out, a, b, sorted_ids,
+                           expert_ids, n_valid, K, N);
+      else
+        hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 4, false, false>), grid,
+                           dim3(256), 0, stream, out, a, b, sorted_ids,
+                           expert_ids, n_valid, K, N);
     }
   }
 }
