@@ -230,7 +230,8 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
 // epilogue: v = acc * a_scale[row] * b_scale[col].
 typedef long long i64;
 
-template <int WM, int WN, int MITER, int NITER, bool GATE_UP>
+template <int WM, int WN, int MITER, int NITER, bool GATE_UP,
+          bool STAGE_A = false>
 __global__ __launch_bounds__(256) void moe_gemm_fp8_kernel(
     u16* __restrict__ out,               // [PM, N] bf16
     const unsigned char* __restrict__ a, // e4m3 [T, K] or act8 [PM, K]
@@ -242,6 +243,7 @@ __global__ __launch_bounds__(256) void moe_gemm_fp8_kernel(
     const int* __restrict__ n_valid,
     const int K, const int N) {
   constexpr int BM = 16 * WM * MITER;
+  static_assert(!STAGE_A || BM == 128, "STAGE_A is tuned for BM=128");
   const int mtile = blockIdx.x;
   if (mtile >= *n_valid) return;
 
@@ -277,6 +279,46 @@ __global__ __launch_bounds__(256) void moe_gemm_fp8_kernel(
     a_p[mi] = a + static_cast<int64_t>(a_row[mi]) * K + (lane >> 4) * 8;
   }
 
+  // STAGE_A (BM=128): [128 rows][32 B] e4m3 tile through LDS via glds
+  // (2-phase; swizzle byte^=(row&1)<<4 on the SOURCE — rule 21)
+  constexpr int kARowB = 32;                    // 32 e4m3 per row
+  constexpr int kATileB = 128 * kARowB;         // 4 KiB
+  __shared__ unsigned char a8_lds[STAGE_A ? 2 * kATileB : 1];
+  const int tid = threadIdx.x;
+  int64_t stage_row_off = 0;
+  int stage_src_off = 0, stage_dst = 0;
+  if (STAGE_A) {
+    const int X = tid * 16;                     // one 16-B piece per thread
+    const int srow = X / kARowB;
+    const int sbyte = (X % kARowB) ^ ((srow & 1) << 4);
+    const int gr = GATE_UP ? sorted_ids[mtile * BM + srow]
+                           : mtile * BM + srow;
+    stage_row_off = static_cast<int64_t>(gr) * K;
+    stage_src_off = sbyte;
+    stage_dst = X;
+  }
+  auto stage_a = [&](int kt, int buf) {
+    if (!STAGE_A) return;
+    const unsigned char* src = a + stage_row_off + kt * 32 + stage_src_off;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(
+            a8_lds + buf * kATileB + stage_dst),
+        16, 0, 0);
+  };
+  int a_lds_off[MITER];
+  if (STAGE_A) {
+#pragma unroll
+    for (int mi = 0; mi < MITER; ++mi) {
+      const int srow = wm * (16 * MITER) + mi * 16 + (lane & 15);
+      const int byte = ((lane >> 4) * 8) ^ ((srow & 1) << 4);
+      a_lds_off[mi] = srow * kARowB + byte;
+    }
+    stage_a(0, 0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+
   floatx4 acc_g[MITER][NITER];
   floatx4 acc_u[GATE_UP ? MITER : 1][NITER];
 #pragma unroll
@@ -289,10 +331,15 @@ __global__ __launch_bounds__(256) void moe_gemm_fp8_kernel(
 
   const int ksteps = K / 32;
   for (int kt = 0; kt < ksteps; ++kt) {
+    if (STAGE_A && kt + 1 < ksteps) stage_a(kt + 1, (kt + 1) & 1);
+    const unsigned char* abuf =
+        STAGE_A ? a8_lds + (kt & 1) * kATileB : nullptr;
     i64 av[MITER];
 #pragma unroll
     for (int mi = 0; mi < MITER; ++mi)
-      av[mi] = *reinterpret_cast<const i64*>(a_p[mi] + kt * 32);
+      av[mi] = STAGE_A
+          ? *reinterpret_cast<const i64*>(abuf + a_lds_off[mi])
+          : *reinterpret_cast<const i64*>(a_p[mi] + kt * 32);
 #pragma unroll
     for (int ni = 0; ni < NITER; ++ni) {
       const i64 bg = *reinterpret_cast<const i64*>(bg_p[ni] + kt * b_step);
@@ -307,6 +354,10 @@ __global__ __launch_bounds__(256) void moe_gemm_fp8_kernel(
           acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
               av[mi], bu, acc_u[mi][ni], 0, 0, 0);
       }
+    }
+    if (STAGE_A) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
     }
   }
 
@@ -431,16 +482,30 @@ void launch_moe_gemm_fp8(u16* out, const unsigned char* a,
                          dim3(256), 0, stream, out, a, a_scales, b, b_scales,
                          sorted_ids, expert_ids, n_valid, K, N);
   } else {
+    static const bool stage_a8 = [] {
+      const char* e = getenv("FI_MOE_STAGE_A");
+      return !(e && e[0] == '0');
+    }();
     if (gate_up) {
       const dim3 grid(max_mtiles, N / 64);
-      hipLaunchKernelGGL((moe_gemm_fp8_kernel<2, 2, 4, 2, true>), grid,
-                         dim3(256), 0, stream, out, a, a_scales, b, b_scales,
-                         sorted_ids, expert_ids, n_valid, K, N);
+      if (stage_a8)
+        hipLaunchKernelGGL((moe_gemm_fp8_kernel<2, 2, 4, 2, true, true>),
+                           grid, dim3(256), 0, stream, out, a, a_scales, b,
+                           b_scales, sorted_ids, expert_ids, n_valid, K, N);
+      else
+        hipLaunchKernelGGL((moe_gemm_fp8_kernel<2, 2, 4, 2, true, false>),
+                           grid, dim3(256), 0, stream, out, a, a_scales, b,
+                           b_scales, sorted_ids, expert_ids, n_valid, K, N);
     } else {
       const dim3 grid(max_mtiles, N / 128);
-      hipLaunchKernelGGL((moe_gemm_fp8_kernel<2, 2, 4, 4, false>), grid,
-                         dim3(256), 0, stream, out, a, a_scales, b, b_scales,
-                         sorted_ids, expert_ids, n_valid, K, N);
+      if (stage_a8)
+        hipLaunchKernelGGL((moe_gemm_fp8_kernel<2, 2, 4, 4, false, true>),
+                           grid, dim3(256), 0, stream, out, a, a_scales, b,
+                           b_scales, sorted_ids, expert_ids, n_valid, K, N);
+      else
+        hipLaunchKernelGGL((moe_gemm_fp8_kernel<2, 2, 4, 4, false, false>),
+                           grid, dim3(256), 0, stream, out, a, a_scales, b,
+                           b_scales, sorted_ids, expert_ids, n_valid, K, N);
     }
   }
 }
