@@ -87,7 +87,10 @@ class SchedulerConfig:
     # 45.3 (+5.8%, TTFT 139ms), 300ms -> 44.9 (TTFT 146ms) — bigger
     # admission batches keep more steps on the pure-decode hipGraph
     # path and run the prefill GEMMs at larger M.
-    prefill_admission_ms: float = 150.0
+    # None (default) = ADAPTIVE: ~4 engine steps (EMA), clamped to
+    # [50, 250] ms — the same batching ratio across models whose step
+    # times differ 2x (8B bf16 ~45ms, fp8 ~33ms, 30B MoE ~65ms).
+    prefill_admission_ms: Optional[float] = None
     # "fcfs" (default) or "priority" (vLLM --scheduling-policy): priority
     # orders admission by (priority, arrival) and preempts the
     # lowest-priority running sequence first (lower value = higher prio)

@@ -517,7 +517,12 @@ class LLMEngine:
                     f"payload={a[1]/n*1e3:.2f}ms fwd={a[2]/n*1e3:.2f}ms "
                     f"sample+book={a[3]/n*1e3:.2f}ms", flush=True,
                 )
-        self.step_time_sum += time.monotonic() - step_t0
+        elapsed_step = time.monotonic() - step_t0
+        self.step_time_sum += elapsed_step
+        # feed the scheduler's adaptive admission window (EMA)
+        self.scheduler._step_ema_s = (
+            0.9 * self.scheduler._step_ema_s + 0.1 * elapsed_step
+        )
         self.num_steps += 1
         return outputs
 

@@ -48,9 +48,13 @@ class ScheduledBatch:
 
 
 class Scheduler:
+    #: adaptive admission window = this many steps of EMA step time
+    ADMISSION_STEPS = 4.0
+
     def __init__(self, cfg: SchedulerConfig, block_manager: BlockManager):
         self.cfg = cfg
         self.bm = block_manager
+        self._step_ema_s = 0.05  # updated by the engine after each step
         self.waiting: Deque[Sequence] = deque()
         self.running: List[Sequence] = []
         # engine-provided: callable(seq) -> bool, True = KV parked in CPU
@@ -146,12 +150,18 @@ class Scheduler:
         else:
             admit = waiting_tokens >= threshold
             # aging escape: at low load the batching threshold may never
-            # be reached — never hold a prompt longer than the configured
-            # window (prefill_admission_ms)
+            # be reached — never hold a prompt longer than the admission
+            # window (explicit prefill_admission_ms, or adaptive:
+            # ADMISSION_STEPS x the EMA step time, clamped [50, 250] ms)
             if not admit and self.waiting:
+                if self.cfg.prefill_admission_ms is not None:
+                    window_s = self.cfg.prefill_admission_ms / 1000.0
+                else:
+                    window_s = min(max(
+                        self.ADMISSION_STEPS * self._step_ema_s, 0.05), 0.25)
                 admit = (
                     time.monotonic() - self.waiting[0].arrival_time
-                    > self.cfg.prefill_admission_ms / 1000.0
+                    > window_s
                 )
         while (
             admit
