@@ -68,6 +68,14 @@ class ModelRunner:
                 assert n > 0
         if mc.quantization not in (None, "fp8"):
             raise ValueError(f"unknown quantization {mc.quantization!r}")
+        if self.is_cuda:
+            # weights are final: keep only the MFMA-packed expert layout
+            # resident — the freed HBM is picked up by the KV-cache
+            # sizing (profile_num_blocks reads mem_get_info after this)
+            for m in self.model.modules():
+                if hasattr(m, "release_unpacked"):
+                    m.release_unpacked()
+            torch.cuda.empty_cache()
         self.block_size = cfg.cache.block_size
         self.max_blocks_per_seq = (
             cfg.scheduler.max_model_len + self.block_size - 1

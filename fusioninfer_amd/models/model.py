@@ -317,6 +317,12 @@ class MoEMLP(nn.Module):
         Frees the stale packed copy immediately so the repack never
         doubles up, then repacks eagerly on GPU (keeps the footprint
         visible before any KV-cache sizing)."""
+        stored = self.gate_up_fp8 if self.fp8 else self.gate_up_t
+        if stored.numel() == 0 and self.e_end > self.e_start:
+            raise RuntimeError(
+                "unpacked MoE expert weights were released; call "
+                "ensure_unpacked() before mutating them"
+            )
         self._pack_version = None
         self._pack_cache = None
         if self.fp8:
@@ -324,6 +330,70 @@ class MoEMLP(nn.Module):
                 self._packed_weights_fp8()
         elif self.gate_up_t.is_cuda:
             self._packed_weights()
+
+    def release_unpacked(self):
+        """Free the unpacked expert weights: GPU serving reads ONLY the
+        MFMA-packed layout, so keeping both resident doubles expert HBM
+        (round-1 debt). Zero-expert stand-ins preserve shape[1:]/dtype/
+        device (forward paths read .shape[2] only); ensure_unpacked()
+        re-materializes exact contents from the packed cache if the
+        loader needs to write again. The runner calls this after weight
+        load and BEFORE KV-cache sizing, so the freed bytes become KV
+        blocks. CPU keeps the unpacked form (it IS the compute path)."""
+        if self.e_end == self.e_start:
+            return
+        if self.fp8:
+            if not self.gate_up_fp8.is_cuda or self.gate_up_fp8.numel() == 0:
+                return
+            self._packed_weights_fp8()
+            self.gate_up_fp8.data = self.gate_up_fp8.data.new_empty(
+                (0,) + tuple(self.gate_up_fp8.shape[1:]))
+            self.down_fp8.data = self.down_fp8.data.new_empty(
+                (0,) + tuple(self.down_fp8.shape[1:]))
+            self._pack_version = (self.gate_up_fp8._version,
+                                  self.down_fp8._version)
+        else:
+            if not self.gate_up_t.is_cuda or self.gate_up_t.numel() == 0:
+                return
+            self._packed_weights()
+            self.gate_up_t.data = self.gate_up_t.data.new_empty(
+                (0,) + tuple(self.gate_up_t.shape[1:]))
+            self.down_t.data = self.down_t.data.new_empty(
+                (0,) + tuple(self.down_t.shape[1:]))
+            self._pack_version = (self.gate_up_t._version,
+                                  self.down_t._version)
+
+    def ensure_unpacked(self):
+        """Inverse of release_unpacked: re-materialize the unpacked
+        params EXACTLY from the packed cache (unpack_moe_weights is pure
+        view/permute tensor ops, so it runs on-device). No-op when the
+        unpacked form is already resident."""
+        if self.e_end == self.e_start:
+            return
+        from fusioninfer_amd.ops.reference import unpack_moe_weights
+
+        if self.fp8:
+            if self.gate_up_fp8.numel():
+                return
+            gu_p, _, dn_p, _ = self._pack_cache
+            self.gate_up_fp8.data = (
+                unpack_moe_weights(gu_p.view(torch.int8))
+                .view(torch.float8_e4m3fn).transpose(1, 2).contiguous()
+            )
+            self.down_fp8.data = (
+                unpack_moe_weights(dn_p.view(torch.int8))
+                .view(torch.float8_e4m3fn).transpose(1, 2).contiguous()
+            )
+            self._pack_version = (self.gate_up_fp8._version,
+                                  self.down_fp8._version)
+        else:
+            if self.gate_up_t.numel():
+                return
+            gu_p, dn_p = self._pack_cache
+            self.gate_up_t.data = unpack_moe_weights(gu_p)
+            self.down_t.data = unpack_moe_weights(dn_p)
+            self._pack_version = (self.gate_up_t._version,
+                                  self.down_t._version)
 
     def _packed_weights_fp8(self):
         """MFMA-packed fp8 expert weights ([E, K, N] order: the stored
@@ -377,9 +447,9 @@ class MoEMLP(nn.Module):
     def _packed_weights(self):
         """MFMA-fragment-packed expert weights for the grouped GEMM,
         cached; invalidated explicitly by the weight loader and on any
-        tracked in-place mutation (tensor._version). Both layouts stay
-        resident — a deliberate 288 GB-HBM tradeoff so the unpacked
-        params remain the loader/test-visible surface."""
+        tracked in-place mutation (tensor._version). After weight load
+        the runner drops the unpacked copy (release_unpacked), leaving
+        only this layout resident on GPU."""
         v = (self.gate_up_t._version, self.down_t._version)
         if getattr(self, "_pack_version", None) != v:
             self._pack_cache = (

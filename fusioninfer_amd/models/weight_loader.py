@@ -275,6 +275,8 @@ def load_hf_state_dict(
                 mlp = layer.mlp
                 if not (mlp.e_start <= e < mlp.e_end):
                     continue  # EP: expert lives on another rank
+                if hasattr(mlp, "ensure_unpacked"):
+                    mlp.ensure_unpacked()  # no-op unless released
                 which = parts[6].replace("_proj", "")  # gate/up/down
                 kind = parts[7] if len(parts) > 7 else "weight"
                 le = e - mlp.e_start

@@ -9,6 +9,7 @@ runs entirely on-device with static shapes (hipGraph-capturable).
 
 import types
 
+import pytest
 import torch
 
 import fusioninfer_amd.ops as ops
@@ -110,6 +111,78 @@ def test_packed_cache_invalidates_on_weight_mutation():
         mlp.gate_up_t[0].mul_(2.0)
     p3, _ = mlp._packed_weights()
     assert p3 is not p2
+
+
+def test_release_unpacked_cpu_noop():
+    """CPU never releases (the unpacked form IS the CPU compute path)."""
+    cfg = get_model_config("tiny-qwen3-moe")
+    mlp = MoEMLP(cfg, layer_idx=0)
+    n = mlp.gate_up_t.numel()
+    mlp.release_unpacked()
+    assert mlp.gate_up_t.numel() == n
+    x = torch.randn(4, cfg.hidden_size, dtype=torch.bfloat16)
+    mlp.forward(x)
+
+
+def test_ensure_unpacked_roundtrip():
+    """ensure_unpacked re-materializes EXACT contents from the packed
+    cache (release itself is CUDA-gated; emulate the released state)."""
+    cfg = get_model_config("tiny-qwen3-moe")
+    mlp = MoEMLP(cfg, layer_idx=2)
+    gu0 = mlp.gate_up_t.data.clone()
+    dn0 = mlp.down_t.data.clone()
+    mlp._packed_weights()
+    mlp.gate_up_t.data = gu0.new_empty((0,) + tuple(gu0.shape[1:]))
+    mlp.down_t.data = dn0.new_empty((0,) + tuple(dn0.shape[1:]))
+    # mutating released weights without re-materializing is a bug
+    with pytest.raises(RuntimeError, match="released"):
+        mlp.invalidate_packed()
+    mlp.ensure_unpacked()
+    assert torch.equal(mlp.gate_up_t.data, gu0)
+    assert torch.equal(mlp.down_t.data, dn0)
+    mlp.ensure_unpacked()  # idempotent
+    assert torch.equal(mlp.gate_up_t.data, gu0)
+
+
+def test_ensure_unpacked_roundtrip_fp8():
+    """fp8 variant: the packed form stores transposed int8-viewed bytes —
+    the round-trip must restore the [E, O, I] e4m3 layout bit-exactly."""
+    cfg = get_model_config("tiny-qwen3-moe")
+    cfg.quantization = "fp8"
+    try:
+        mlp = MoEMLP(cfg, layer_idx=0)
+        gu0 = mlp.gate_up_fp8.data.clone()
+        dn0 = mlp.down_fp8.data.clone()
+        mlp._packed_weights_fp8()
+        mlp.gate_up_fp8.data = gu0.new_empty((0,) + tuple(gu0.shape[1:]))
+        mlp.down_fp8.data = dn0.new_empty((0,) + tuple(dn0.shape[1:]))
+        mlp.ensure_unpacked()
+        assert torch.equal(mlp.gate_up_fp8.data.view(torch.int8),
+                           gu0.view(torch.int8))
+        assert torch.equal(mlp.down_fp8.data.view(torch.int8),
+                           dn0.view(torch.int8))
+    finally:
+        cfg.quantization = None
+
+
+@pytest.mark.gpu
+def test_release_unpacked_gpu_forward_parity():
+    """On GPU, releasing the unpacked copy must not change the grouped
+    forward (it reads only the packed cache + shapes)."""
+    torch.manual_seed(2)
+    cfg = get_model_config("tiny-qwen3-moe")
+    with torch.device("cuda"):
+        mlp = MoEMLP(cfg, layer_idx=0)
+    x = torch.randn(17, cfg.hidden_size, dtype=torch.bfloat16,
+                    device="cuda")
+    want = mlp.forward(x)
+    gu0 = mlp.gate_up_t.data.clone()
+    mlp.release_unpacked()
+    assert mlp.gate_up_t.numel() == 0
+    torch.testing.assert_close(mlp.forward(x), want)
+    mlp.ensure_unpacked()
+    assert torch.equal(mlp.gate_up_t.data, gu0)
+    torch.testing.assert_close(mlp.forward(x), want)
 
 
 def test_fp8_grouped_matches_per_expert_loop():
