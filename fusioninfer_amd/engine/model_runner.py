@@ -11,6 +11,7 @@ host lists so the TP driver can broadcast them to worker ranks.
 
 from __future__ import annotations
 
+import os
 from typing import Dict, List, Optional, Tuple
 
 import numpy as np
@@ -68,7 +69,8 @@ class ModelRunner:
                 assert n > 0
         if mc.quantization not in (None, "fp8"):
             raise ValueError(f"unknown quantization {mc.quantization!r}")
-        if self.is_cuda:
+        if self.is_cuda and os.environ.get(
+                "FI_MOE_KEEP_UNPACKED", "0") != "1":
             # weights are final: keep only the MFMA-packed expert layout
             # resident — the freed HBM is picked up by the KV-cache
             # sizing (profile_num_blocks reads mem_get_info after this)
