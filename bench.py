@@ -189,7 +189,8 @@ def main():
             max_num_seqs=args.concurrency,
             max_num_batched_tokens=args.max_batched_tokens,
             max_model_len=max_len,
-            **({"prefill_admission_ms": args.admission_ms}
+            **({"prefill_admission_ms": (
+                None if args.admission_ms < 0 else args.admission_ms)}
                if args.admission_ms is not None else {}),
         ),
         parallel=ParallelConfig(
