@@ -503,12 +503,13 @@ def _int_range_regex(lo, hi) -> str:
 
 def schema_to_regex(schema: dict, depth: int = 0, defs=None) -> str:
     """JSON-schema subset -> anchored regex (compact output, property
-    order fixed, all listed properties emitted). Supported beyond the
-    basics: enum/const, anyOf/oneOf, allOf (single branch), type lists,
-    string pattern + minLength/maxLength, integer minimum/maximum
-    (enumerable or sign-determined), array minItems/maxItems, local
-    $defs/$ref (cycle-guarded by depth). Unsupported constructs raise
-    ValueError — the API layer then falls back to generic JSON."""
+    order fixed; properties outside `required` are optional — emitted
+    in order or skipped). Supported beyond the basics: enum/const,
+    anyOf/oneOf, allOf (single branch), type lists, string pattern +
+    minLength/maxLength, integer minimum/maximum (enumerable or
+    sign-determined), array minItems/maxItems, local $defs/$ref
+    (cycle-guarded by depth). Unsupported constructs raise ValueError —
+    the API layer then falls back to generic JSON."""
     if depth > 8:
         raise ValueError("schema nesting too deep")
     if not isinstance(schema, dict):
@@ -593,18 +594,44 @@ def schema_to_regex(schema: dict, depth: int = 0, defs=None) -> str:
         return f"\\[{body}\\]" if lo > 0 else f"\\[({body})?\\]"
     if t == "object":
         props = schema.get("properties") or {}
-        required = schema.get("required")
-        if required:
-            # optional properties would need order-permutation handling;
-            # emit the required subset in property order (documented)
-            props = {k: v for k, v in props.items() if k in set(required)}
         if not props:
             raise ValueError("object schema needs properties")
-        parts = []
-        for name, sub in props.items():
-            key = _re_escape(json.dumps(name))
-            parts.append(f"{key}:{schema_to_regex(sub, depth + 1, defs)}")
-        return "\\{" + ",".join(parts) + "\\}"
+        # JSON-schema semantics: properties outside `required` are
+        # optional. Deviation (documented): with NO `required` key every
+        # listed property is emitted — the useful default for extraction
+        # prompts, and the historical behavior. Property order is fixed
+        # to the schema's listing order either way.
+        req_list = schema.get("required")
+        required = set(props.keys() if req_list is None else req_list)
+        rendered = [
+            (f"{_re_escape(json.dumps(name))}:"
+             f"{schema_to_regex(sub, depth + 1, defs)}",
+             name in required)
+            for name, sub in props.items()
+        ]
+        if sum(1 for _, req in rendered if not req) > 12:
+            raise ValueError("too many optional properties")
+        # tails[i]: properties i.. when some property was already
+        # emitted (each present one is comma-prefixed; optional ones
+        # are skippable groups)
+        tails = [""]
+        for part, req in reversed(rendered):
+            tails.append(f"(,{part}){'' if req else '?'}{tails[-1]}")
+        tails.reverse()
+        # head: alternation over which property appears FIRST (no
+        # comma). Only properties up to and including the first
+        # required one can be first.
+        k = next((i for i, (_, req) in enumerate(rendered) if req),
+                 len(rendered))
+        starts = [f"{part}{tails[i + 1]}"
+                  for i, (part, _) in enumerate(rendered[:k + 1])]
+        if k == len(rendered):  # no required property: {} is valid too
+            body = "(" + "|".join(starts) + ")?"
+        elif len(starts) == 1:
+            body = starts[0]
+        else:
+            body = "(" + "|".join(starts) + ")"
+        return "\\{" + body + "\\}"
     raise ValueError(f"unsupported schema type {t!r}")
 
 

@@ -152,6 +152,45 @@ def test_schema_to_regex_object():
     assert not accepts(g, '{"name":"bo","age":4.5,"tags":[],"ok":true}')
 
 
+def test_schema_object_optional_properties():
+    """Properties outside `required` may be skipped (order stays fixed,
+    commas only between present properties)."""
+    pattern = schema_to_regex({
+        "type": "object",
+        "properties": {
+            "pre": {"type": "boolean"},       # optional, before required
+            "name": {"type": "string"},       # required
+            "age": {"type": "integer"},       # optional
+            "ok": {"type": "boolean"},        # required
+            "tag": {"type": "string"},        # optional, trailing
+        },
+        "required": ["name", "ok"],
+    })
+    g = RegexGrammar(pattern)
+    assert accepts(g, '{"pre":true,"name":"bo","age":4,"ok":true,"tag":"x"}')
+    assert accepts(g, '{"name":"bo","ok":false}')          # optionals skipped
+    assert accepts(g, '{"name":"bo","age":7,"ok":true}')
+    assert accepts(g, '{"pre":false,"name":"bo","ok":true,"tag":"y"}')
+    assert not accepts(g, '{"age":7,"ok":true}')           # missing required
+    assert not accepts(g, '{"name":"bo"}')                 # missing required
+    assert not accepts(g, '{"age":7,"name":"bo","ok":true}')  # order fixed
+    assert not accepts(g, '{"name":"bo","ok":true,}')      # dangling comma
+
+
+def test_schema_object_all_optional():
+    g = RegexGrammar(schema_to_regex({
+        "type": "object",
+        "properties": {"a": {"type": "integer"}, "b": {"type": "boolean"}},
+        "required": [],
+    }))
+    assert accepts(g, "{}")
+    assert accepts(g, '{"a":1}')
+    assert accepts(g, '{"b":true}')
+    assert accepts(g, '{"a":1,"b":false}')
+    assert not accepts(g, '{"b":true,"a":1}')  # order fixed
+    assert not accepts(g, '{,}')
+
+
 def test_schema_enum_and_unsupported():
     g = RegexGrammar(schema_to_regex({"enum": ["red", "green", 3]}))
     assert accepts(g, '"red"') and accepts(g, "3") and not accepts(g, '"blue"')
@@ -266,7 +305,7 @@ def test_engine_guided_regex_bounded():
 
 def test_schema_extensions():
     """anyOf / const / type lists / string pattern / min-maxItems /
-    required-subset — the wider vLLM guided_json surface."""
+    optional properties — the wider vLLM guided_json surface."""
     g = RegexGrammar(schema_to_regex({
         "anyOf": [{"type": "integer"}, {"type": "boolean"}],
     }))
@@ -298,7 +337,8 @@ def test_schema_extensions():
         "required": ["a", "c"],
     }))
     assert accepts(g, '{"a":1,"c":"x"}')
-    assert not accepts(g, '{"a":1,"b":true,"c":"x"}')
+    assert accepts(g, '{"a":1,"b":true,"c":"x"}')  # optional b allowed
+    assert not accepts(g, '{"b":true,"c":"x"}')    # required a missing
 
 
 def test_schema_bounds_refs_and_allof():
