@@ -41,6 +41,12 @@ class Attention(nn.Module):
         self.q_size = self.num_heads * self.head_dim
         self.kv_size = self.num_kv_heads * self.head_dim
         self.scale = 1.0 / math.sqrt(self.head_dim)
+        # fp8-KV per-layer static dequant scales (vLLM k_scale/v_scale;
+        # set by the weight loader from calibrated checkpoints, 1.0
+        # default). Plain python floats: K folds into the softmax scalar
+        # and V into a post-multiply, so no device reads — graph-safe.
+        self.k_scale = 1.0
+        self.v_scale = 1.0
         self.eps = cfg.rms_norm_eps
         hidden = cfg.hidden_size
         self.qkv_proj = MergedColumnParallelLinear(
@@ -83,7 +89,15 @@ class Attention(nn.Module):
             self.q_norm_weight, self.k_norm_weight, self.eps,
         )
         k_cache, v_cache = kv_cache
-        ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
+        fp8_kv = k_cache.dtype == torch.float8_e4m3fn
+        ks = self.k_scale if fp8_kv else 1.0
+        vs = self.v_scale if fp8_kv else 1.0
+        ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping,
+                              1.0 / ks, 1.0 / vs)
+        # K's dequant scale folds EXACTLY into the softmax scalar:
+        # scale*(Q . ks*K8) == (scale*ks)*(Q . K8); V's is a linear
+        # post-multiply on the attention output (applied below).
+        scale = self.scale * ks
 
         T = qkv.shape[0]
         np_, nd = meta.num_prefill_tokens, meta.num_decode_tokens
@@ -98,7 +112,7 @@ class Attention(nn.Module):
                 meta.prefill_block_tables,
                 meta.cu_seqlens,
                 meta.prefill_seq_lens_k,
-                self.scale,
+                scale,
                 tile_seq=meta.tile_seq,
                 tile_row0=meta.tile_row0,
                 tile_rows=meta.tile_rows or None,
@@ -111,9 +125,11 @@ class Attention(nn.Module):
                 v_cache,
                 meta.block_tables,
                 meta.seq_lens,
-                self.scale,
+                scale,
             )
             out[np_:] = o.view(nd, self.q_size)
+        if vs != 1.0:
+            out *= vs
         if meta.lora is not None:
             return self.o_proj.forward_with_lora(out, meta.lora,
                                                  self.layer_idx, "o")

@@ -247,6 +247,15 @@ def load_hf_state_dict(
             elif rest == "self_attn.o_proj.weight_scale":
                 o = layer.self_attn.o_proj
                 o.weight_scale.copy_(_row_scale(w, o.weight.shape[0]))
+            elif rest in ("self_attn.attn.k_scale", "self_attn.k_scale"):
+                # calibrated fp8-KV dequant scale (vLLM checkpoint keys)
+                layer.self_attn.k_scale = float(w.reshape(-1)[0])
+            elif rest in ("self_attn.attn.v_scale", "self_attn.v_scale"):
+                layer.self_attn.v_scale = float(w.reshape(-1)[0])
+            elif rest in ("self_attn.attn.kv_scale", "self_attn.kv_scale"):
+                # legacy combined K/V scale
+                layer.self_attn.k_scale = float(w.reshape(-1)[0])
+                layer.self_attn.v_scale = float(w.reshape(-1)[0])
             elif rest == "self_attn.q_norm.weight":
                 put(layer.self_attn.q_norm_weight, w)
             elif rest == "self_attn.k_norm.weight":

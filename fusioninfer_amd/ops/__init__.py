@@ -156,14 +156,21 @@ def rope_qk_norm_(
 
 # ---------------------------------------------------------------- kv cache
 
-def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping):
+def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping,
+                      k_inv_scale: float = 1.0, v_inv_scale: float = 1.0):
+    """Scatter K/V rows into the paged cache. For fp8 caches the rows are
+    multiplied by the INVERSE per-layer static scales before e4m3
+    conversion (vLLM k_scale/v_scale); bf16 caches ignore the scales
+    (the read side compensates only on the fp8 path)."""
     if k.is_cuda:
         _require_native()
-        _C.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        _C.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping,
+                             k_inv_scale, v_inv_scale)
         return
     H, D = k_cache.shape[1], k_cache.shape[3]
     ref.reshape_and_cache(
-        k.view(-1, H, D), v.view(-1, H, D), k_cache, v_cache, slot_mapping.long()
+        k.view(-1, H, D), v.view(-1, H, D), k_cache, v_cache,
+        slot_mapping.long(), k_inv_scale, v_inv_scale,
     )
 
 

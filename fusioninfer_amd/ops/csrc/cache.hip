@@ -11,9 +11,12 @@
 
 namespace fi {
 
-// FP8: the cache stores OCP e4m3 at scale 1.0 (K rows are qk-normed and
-// O(1); saturation at |448| is the only loss) — halves decode KV bytes
-// and doubles cache capacity. Conversion uses v_cvt_pk_fp8_f32.
+// FP8: the cache stores OCP e4m3 at a per-layer static scale (default
+// 1.0: K rows are qk-normed and O(1); saturation at |448| is the only
+// loss) — halves decode KV bytes and doubles cache capacity. The write
+// multiplies by the INVERSE scale; the read side needs no kernel work
+// (K's scale folds into the softmax scalar, V's into a post-multiply on
+// the attention output). Conversion uses v_cvt_pk_fp8_f32.
 template <bool FP8>
 __global__ void reshape_and_cache_kernel(
     const u16* __restrict__ k,   // [T] rows of Hk*D, stride k_stride (bf16)
@@ -23,7 +26,7 @@ __global__ void reshape_and_cache_kernel(
     const int* __restrict__ slot_mapping,  // [T]
     const int64_t k_stride, const int64_t v_stride,
     const int num_tokens, const int kv_heads, const int block_size,
-    const int head_dim) {
+    const int head_dim, const float k_inv_scale, const float v_inv_scale) {
   const int vec_per_tok = kv_heads * head_dim / 8;
   const int64_t total = static_cast<int64_t>(num_tokens) * vec_per_tok;
   for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < total;
@@ -45,8 +48,8 @@ __global__ void reshape_and_cache_kernel(
       float kf[8], vf[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        kf[j] = bf16_to_f32(kv.h[j]);
-        vf[j] = bf16_to_f32(vv.h[j]);
+        kf[j] = bf16_to_f32(kv.h[j]) * k_inv_scale;
+        vf[j] = bf16_to_f32(vv.h[j]) * v_inv_scale;
       }
       u32 kp[2], vp[2];
       pack_fp8x8(kf, kp);
@@ -66,7 +69,8 @@ void launch_reshape_and_cache(const u16* k, const u16* v, void* k_cache,
                               void* v_cache, const int* slot_mapping,
                               int64_t k_stride, int64_t v_stride, int tokens,
                               int kv_heads, int block_size, int head_dim,
-                              bool fp8, hipStream_t stream) {
+                              bool fp8, float k_inv_scale, float v_inv_scale,
+                              hipStream_t stream) {
   const int64_t total = static_cast<int64_t>(tokens) * kv_heads * head_dim / 8;
   const int block = 256;
   const int grid = static_cast<int>(std::min<int64_t>((total + block - 1) / block, (int64_t)2048));
@@ -74,12 +78,12 @@ void launch_reshape_and_cache(const u16* k, const u16* v, void* k_cache,
     hipLaunchKernelGGL((reshape_and_cache_kernel<true>), dim3(grid),
                        dim3(block), 0, stream, k, v, k_cache, v_cache,
                        slot_mapping, k_stride, v_stride, tokens, kv_heads,
-                       block_size, head_dim);
+                       block_size, head_dim, k_inv_scale, v_inv_scale);
   } else {
     hipLaunchKernelGGL((reshape_and_cache_kernel<false>), dim3(grid),
                        dim3(block), 0, stream, k, v, k_cache, v_cache,
                        slot_mapping, k_stride, v_stride, tokens, kv_heads,
-                       block_size, head_dim);
+                       block_size, head_dim, 1.0f, 1.0f);
   }
 }
 

@@ -21,7 +21,7 @@ void launch_rope_qk_norm(u16*, u16*, int64_t, int64_t, const u16*, const u16*,
                          hipStream_t);
 void launch_reshape_and_cache(const u16*, const u16*, void*, void*,
                               const int*, int64_t, int64_t, int, int, int,
-                              int, bool, hipStream_t);
+                              int, bool, float, float, hipStream_t);
 template <bool GATHER>
 void launch_kv_block_copy(u16*, u16*, u16*, const int*, int, int64_t,
                           hipStream_t);
@@ -188,7 +188,8 @@ void rope_qk_norm(at::Tensor q, at::Tensor k,
 }
 
 void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
-                       at::Tensor v_cache, at::Tensor slot_mapping) {
+                       at::Tensor v_cache, at::Tensor slot_mapping,
+                       double k_inv_scale, double v_inv_scale) {
   CHECK_BF16_CUDA(k);
   CHECK_BF16_CUDA(v);
   const bool fp8 = cache_is_fp8(k_cache);
@@ -201,7 +202,9 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
   fi::launch_reshape_and_cache(
       bf16_cptr(k), bf16_cptr(v), k_cache.data_ptr(), v_cache.data_ptr(),
       slot_mapping.data_ptr<int>(), k.stride(0), v.stride(0), k.size(0),
-      kv_heads, block_size, head_dim, fp8, current_stream());
+      kv_heads, block_size, head_dim, fp8,
+      static_cast<float>(k_inv_scale), static_cast<float>(v_inv_scale),
+      current_stream());
 }
 
 void gather_kv_blocks(at::Tensor staging, at::Tensor k_cache,

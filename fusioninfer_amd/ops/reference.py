@@ -143,13 +143,25 @@ def reshape_and_cache(
     k_cache: torch.Tensor,
     v_cache: torch.Tensor,
     slot_mapping: torch.Tensor,
+    k_inv_scale: float = 1.0,
+    v_inv_scale: float = 1.0,
 ):
-    """k/v: [T, Hk, D]; caches: [num_blocks, Hk, block_size, D]; slots: [T]."""
+    """k/v: [T, Hk, D]; caches: [num_blocks, Hk, block_size, D]; slots: [T].
+    fp8 caches quantize at the inverse per-layer static scale (matching
+    the HIP kernel); bf16 caches store unscaled."""
     block_size = k_cache.shape[2]
     blk = torch.div(slot_mapping, block_size, rounding_mode="floor")
     off = slot_mapping % block_size
-    k_cache[blk, :, off] = k.to(k_cache.dtype)
-    v_cache[blk, :, off] = v.to(v_cache.dtype)
+    if k_cache.dtype == torch.float8_e4m3fn:
+        # clamp to +-448: v_cvt_pk_fp8_f32 SATURATES out-of-range values
+        # where torch's .to(e4m3) would produce NaN
+        k_cache[blk, :, off] = (k.float() * k_inv_scale).clamp(
+            -448.0, 448.0).to(k_cache.dtype)
+        v_cache[blk, :, off] = (v.float() * v_inv_scale).clamp(
+            -448.0, 448.0).to(v_cache.dtype)
+    else:
+        k_cache[blk, :, off] = k.to(k_cache.dtype)
+        v_cache[blk, :, off] = v.to(v_cache.dtype)
 
 
 def paged_attention_decode(

@@ -307,6 +307,33 @@ def test_reshape_and_cache_fp8():
     )
 
 
+def test_reshape_and_cache_fp8_inv_scale():
+    """Per-layer static KV scales: the write multiplies by the inverse
+    scale before e4m3 conversion; out-of-range inputs saturate (matching
+    the reference's clamp) instead of NaN-ing."""
+    T, Hk, D, bs, nblocks = 9, 2, 64, 16, 4
+    k = torch.randn(T, Hk * D, dtype=torch.bfloat16, device=DEV) * 3.0
+    v = torch.randn(T, Hk * D, dtype=torch.bfloat16, device=DEV) * 600.0
+    k_cache = torch.zeros(nblocks, Hk, bs, D, dtype=torch.float8_e4m3fn,
+                          device=DEV)
+    v_cache = torch.zeros_like(k_cache)
+    slots = torch.arange(T, device=DEV, dtype=torch.int32)
+    k_ref = torch.zeros(nblocks, Hk, bs, D, dtype=torch.float8_e4m3fn)
+    v_ref = torch.zeros_like(k_ref)
+    ref.reshape_and_cache(
+        k.cpu().view(T, Hk, D), v.cpu().view(T, Hk, D), k_ref, v_ref,
+        slots.cpu().long(), 1.0 / 3.0, 1.0 / 4.0
+    )
+    ops.reshape_and_cache(k, v, k_cache, v_cache, slots, 1.0 / 3.0, 1.0 / 4.0)
+    assert torch.isfinite(v_cache.float()).all()
+    torch.testing.assert_close(
+        k_cache.float().cpu(), k_ref.float(), atol=0.07, rtol=0.07
+    )
+    torch.testing.assert_close(
+        v_cache.float().cpu(), v_ref.float(), atol=12.0, rtol=0.07
+    )
+
+
 @pytest.mark.parametrize("group", [1, 4])
 @pytest.mark.parametrize("seq_lens", [[1], [1, 5, 16, 17, 255, 1023]])
 def test_paged_attention_decode_fp8kv(group, seq_lens):

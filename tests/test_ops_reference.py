@@ -62,6 +62,44 @@ def test_decode_consistent_with_prefill_last_row():
     torch.testing.assert_close(dec[0], full[-1], atol=1e-4, rtol=1e-4)
 
 
+def test_fp8_kv_static_scale_recovers_large_values():
+    """Per-layer static k/v scales: writes quantize at the inverse scale;
+    the read side folds k_scale into the softmax scalar and v_scale into
+    a post-multiply. With |V| well beyond e4m3's 448 max, scale-1.0
+    clips badly while a calibrated scale recovers the fp32 answer."""
+    torch.manual_seed(3)
+    L, Hq, Hk, D, bs = 24, 2, 1, 32, 16
+    scale = 1.0 / math.sqrt(D)
+    q = torch.randn(L, Hq, D)
+    k = torch.randn(L, Hk, D) * 3.0
+    v = torch.randn(L, Hk, D) * 600.0   # saturates e4m3 at scale 1.0
+    cu = torch.tensor([0, L], dtype=torch.int32)
+    want = ref.prefill_attention(q, k, v, cu, scale)[-1]
+
+    nblk = (L + bs - 1) // bs
+    bt = torch.arange(nblk, dtype=torch.int32).unsqueeze(0)
+    lens = torch.tensor([L], dtype=torch.int32)
+    slots = torch.arange(L)
+
+    def run(k_scale, v_scale):
+        k_cache = torch.zeros(nblk, Hk, bs, D, dtype=torch.float8_e4m3fn)
+        v_cache = torch.zeros_like(k_cache)
+        ref.reshape_and_cache(k, v, k_cache, v_cache, slots,
+                              1.0 / k_scale, 1.0 / v_scale)
+        out = ref.paged_attention_decode(
+            q[-1:], k_cache, v_cache, bt, lens, scale * k_scale
+        )[0]
+        return out * v_scale
+
+    err_unscaled = (run(1.0, 1.0) - want).norm() / want.norm()
+    err_scaled = (run(3.0, 4.0) - want).norm() / want.norm()
+    assert err_unscaled > 0.2          # saturation destroys the answer
+    # e4m3 carries ~6% per-element quantization noise (3 mantissa bits);
+    # the calibrated scale must recover to that noise floor
+    assert err_scaled < 0.12, float(err_scaled)
+    assert err_scaled < err_unscaled / 3
+
+
 def test_kv_pack_unpack_roundtrip():
     k_cache = torch.randn(8, 2, 16, 32)
     v_cache = torch.randn(8, 2, 16, 32)

@@ -77,6 +77,25 @@ def test_safetensors_dir_loading(tmp_path):
     )
 
 
+def test_fp8_kv_scale_keys_load():
+    """Calibrated fp8-KV dequant scales land on the attention module
+    (vLLM checkpoint key spellings, incl. the legacy combined kv_scale)."""
+    m = _model(5)
+    n = load_hf_state_dict(m, [
+        ("model.layers.0.self_attn.attn.k_scale", torch.tensor(2.5)),
+        ("model.layers.0.self_attn.attn.v_scale", torch.tensor([1.75])),
+        ("model.layers.1.self_attn.kv_scale", torch.tensor(3.0)),
+    ])
+    assert n == 3
+    assert m.layers[0].self_attn.k_scale == 2.5
+    assert m.layers[0].self_attn.v_scale == 1.75
+    assert m.layers[1].self_attn.k_scale == 3.0
+    assert m.layers[1].self_attn.v_scale == 3.0
+    # untouched layers keep the 1.0 default
+    if len(m.layers) > 2:
+        assert m.layers[2].self_attn.k_scale == 1.0
+
+
 def test_moe_expert_weights_roundtrip():
     """HF qwen3_moe expert names load into the stacked expert tensors and
     change the model's output."""
