@@ -48,8 +48,13 @@ __global__ void reshape_and_cache_kernel(
       float kf[8], vf[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        kf[j] = bf16_to_f32(kv.h[j]) * k_inv_scale;
-        vf[j] = bf16_to_f32(vv.h[j]) * v_inv_scale;
+        // saturate at e4m3 max: v_cvt_pk_fp8_f32 NaNs out-of-range
+        // values (no hardware clamp), and a NaN in the cache poisons
+        // every later attention read of the block
+        kf[j] = fminf(fmaxf(bf16_to_f32(kv.h[j]) * k_inv_scale, -448.f),
+                      448.f);
+        vf[j] = fminf(fmaxf(bf16_to_f32(vv.h[j]) * v_inv_scale, -448.f),
+                      448.f);
       }
       u32 kp[2], vp[2];
       pack_fp8x8(kf, kp);
