@@ -236,6 +236,7 @@ def test_gpu_moe_engine_generates():
     assert all(len(o.output_token_ids) == 6 for o in outs)
 
     moe = eng.runner.model.layers[0].mlp
+    moe.ensure_unpacked()  # the runner released the unpacked copy
     x = torch.randn(9, cfg.model.hidden_size, dtype=torch.bfloat16,
                     device="cuda:0")
     got = moe(x).float()
