@@ -108,6 +108,16 @@ class LLMEngine:
         self._import_holder = None
         self._timing = os.environ.get("FI_STEP_TIMING") == "1"
         self._tacc = {}
+        # pipelined decode: the last launched-but-unprocessed pure-decode
+        # step (seqs, on-device sampled ids, pinned host copy, event).
+        # Host emission of step N overlaps the GPU running step N+1; the
+        # next step's input ids are fed device-to-device, so the chain
+        # never waits on a host sync to launch. FI_ASYNC_DECODE=0 opts
+        # out (every step then drains synchronously, round-1 behavior).
+        self._pending = None
+        self._async_decode = os.environ.get("FI_ASYNC_DECODE", "1") == "1"
+        self._drained_backlog: List[RequestOutput] = []
+        self.num_async_steps = 0  # pipelined continuations taken
         # running stats for metrics / EPP scorers
         self.num_finished = 0
         self.num_generated_tokens = 0
@@ -230,6 +240,13 @@ class LLMEngine:
         """Cancel a request wherever it is (waiting / running); frees its
         cache blocks. Returns True if it was found."""
         from fusioninfer_amd.engine.sequence import SeqStatus
+
+        if self._pending is not None:
+            # flush the pipelined in-flight step: removing a seq from
+            # running mid-chain would desync the device-id row mapping.
+            # The drained outputs belong to OTHER requests — surface
+            # them through the next step() call
+            self._drained_backlog.extend(self._drain_pending())
 
         seq = self.seqs.pop(request_id, None)
         if seq is None:
@@ -420,6 +437,14 @@ class LLMEngine:
         ranks (in worker_loop) execute the same forward so the per-layer
         RCCL all-reduces line up."""
         assert self.is_driver, "only TP rank 0 steps; others run worker_loop"
+        pre: List[RequestOutput] = self._drained_backlog
+        if pre:
+            self._drained_backlog = []
+        if self._pending is not None:
+            outs = self._try_continue_async()
+            if outs is not None:
+                return pre + outs
+            pre = pre + self._drain_pending()
         step_t0 = time.monotonic()
         tp = (self.cfg.parallel.tensor_parallel_size
               * self.cfg.parallel.pipeline_parallel_size)
@@ -428,7 +453,7 @@ class LLMEngine:
         for seq in batch.swap_in:
             self._swap_in(seq)
         if batch.is_empty:
-            return []
+            return pre
         self.num_preemptions += len(batch.preempted)
         want_plp = any(
             s.sampling.prompt_logprobs is not None for s in batch.prefill_seqs
@@ -489,6 +514,28 @@ class LLMEngine:
                 seq.num_computed_tokens or seq.num_cached_tokens
             ) + chunk
             self.num_prefilled_tokens += chunk
+        if (
+            self._async_decode
+            and self.proposer is None  # spec engines keep the sync path
+            and spec_drafts is None
+            and not batch.prefill_seqs
+            and batch.decode_seqs
+            and tp == 1
+            and payload.get("logits_rows") is None
+            and all(self._seq_async_ok(s) for s in batch.decode_seqs)
+        ):
+            # enter the pipelined-decode chain: sample WITHOUT a host
+            # sync and defer emission to the next step() call (which
+            # launches step N+1 first, then processes these tokens
+            # while the GPU runs)
+            self._stash_pending(batch.decode_seqs, logits)
+            elapsed_step = time.monotonic() - step_t0
+            self.step_time_sum += elapsed_step
+            self.scheduler._step_ema_s = (
+                0.9 * self.scheduler._step_ema_s + 0.1 * elapsed_step
+            )
+            self.num_steps += 1
+            return pre
         if spec_drafts is not None:
             outputs = self._finish_spec_step(batch, payload, logits,
                                              spec_drafts)
@@ -504,7 +551,7 @@ class LLMEngine:
                     batch, payload, logits_f, chunk_starts
                 )
             if not sample_seqs:
-                return []
+                return pre
             outputs = self._sample_and_emit(sample_seqs, logits_f)
         if self._timing and hasattr(self, "_t_fwd_mark"):
             kind = "P" if batch.prefill_seqs else "D"
@@ -524,7 +571,7 @@ class LLMEngine:
             0.9 * self.scheduler._step_ema_s + 0.1 * elapsed_step
         )
         self.num_steps += 1
-        return outputs
+        return pre + outputs
 
     # ------------------------------------------------- sampling / emission
     @staticmethod
@@ -574,6 +621,130 @@ class LLMEngine:
                 self._finish_seq(seq)
                 break
         return RequestOutput(seq, new_token_ids=new)
+
+    # ------------------------------------------------- pipelined decode
+    @staticmethod
+    def _seq_async_ok(s: Sequence) -> bool:
+        """Pipeline-eligible: nothing about next-step sampling may depend
+        on THIS step's token value on the host (guided masks, penalties
+        over emitted tokens) and nothing extra is read back (logprobs)."""
+        sp = s.sampling
+        return (
+            sp.guided is None
+            and not sp.logit_bias
+            and sp.logprobs is None
+            and sp.prompt_logprobs is None
+            and sp.repetition_penalty == 1.0
+            and sp.presence_penalty == 0.0
+            and sp.frequency_penalty == 0.0
+            and s.lora_name is None
+            # seeded draws key on len(output) — one short pre-drain
+            and (sp.temperature == 0 or sp.seed is None)
+        )
+
+    def _stash_pending(self, seqs, logits) -> None:
+        """Sample on-device and record the in-flight step: the sampled
+        ids tensor feeds the next launch device-to-device; the pinned
+        host copy (awaited via the event at drain time) feeds emission."""
+        logits_f = logits.float()
+        sampled = self.sampler.sample(logits_f, seqs)
+        if sampled.is_cuda:
+            host = torch.empty(
+                sampled.shape, dtype=sampled.dtype, pin_memory=True
+            )
+            host.copy_(sampled, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record()
+        else:
+            host, ev = sampled, None
+        self._pending = {
+            "seqs": list(seqs), "sampled": sampled, "host": host, "event": ev,
+        }
+
+    def _drain_pending(self) -> List[RequestOutput]:
+        """Emit the in-flight step's tokens (stop checks, block frees,
+        stats). Safe against seqs that finished since the stash: their
+        late token is discarded."""
+        p = self._pending
+        if p is None:
+            return []
+        self._pending = None
+        if p["event"] is not None:
+            p["event"].synchronize()
+        toks = p["host"].tolist()
+        return [
+            self._emit_tokens(s, [t])
+            for s, t in zip(p["seqs"], toks)
+            if not s.is_finished()
+        ]
+
+    def _try_continue_async(self):
+        """Continue the pipelined pure-decode chain: launch step N+1
+        from step N's ON-DEVICE sampled ids, then overlap N's host-side
+        emission with the GPU. Returns N's outputs, or None when the
+        chain must break to the sync path (admission due, composition
+        changed, block pressure, or every seq deterministically
+        finishes). Deterministic finishers (in-flight token completes
+        max_tokens) are dropped from the continued batch up front; data-
+        dependent stops (eos/stop tokens) cost one wasted decode row and
+        break the chain at the next composition check."""
+        p = self._pending
+        sch = self.scheduler
+        if not self._async_decode or not sch.decode_only_next():
+            return None
+        if len(sch.running) != len(p["seqs"]) or any(
+            a is not b for a, b in zip(sch.running, p["seqs"])
+        ):
+            return None
+        bm = self.block_manager
+        max_len = self.cfg.scheduler.max_model_len
+        keep: List[Sequence] = []
+        keep_rows: List[int] = []
+        for i, s in enumerate(p["seqs"]):
+            out_after = s.num_tokens + 1 - s.num_prompt_tokens
+            if (
+                out_after >= s.sampling.max_tokens
+                or s.num_tokens + 1 >= max_len
+                or not self._seq_async_ok(s)
+            ):
+                continue
+            keep.append(s)
+            keep_rows.append(i)
+        if not keep:
+            return None
+        need = sum(bm.extra_blocks_for(s, s.num_tokens) for s in keep)
+        if bm.num_free() < need:
+            return None  # sync path can preempt
+        step_t0 = time.monotonic()
+        for s in keep:
+            bm.append_slots_upto(s, s.num_tokens)
+        payload = {
+            "kind": "decode",
+            "ids": None,  # fed on-device below
+            "positions": [s.num_tokens for s in keep],
+            "slots": [bm.slot_for(s, s.num_tokens) for s in keep],
+            "lens": [s.num_tokens + 1 for s in keep],
+            "bt": [list(s.block_ids) for s in keep],
+            "lora_names": [None] * len(keep),
+        }
+        sampled = p["sampled"]
+        if len(keep) != len(p["seqs"]):
+            idx = torch.tensor(
+                keep_rows, dtype=torch.long, device=sampled.device
+            )
+            sampled = sampled.index_select(0, idx)
+        logits = self.runner.run_decode_device_ids(payload, sampled)
+        # N+1 is on the GPU; process N on the host in the shadow
+        outputs = self._drain_pending()
+        self._stash_pending(keep, logits)
+        self.num_async_steps += 1
+        elapsed_step = time.monotonic() - step_t0
+        self.step_time_sum += elapsed_step
+        self.scheduler._step_ema_s = (
+            0.9 * self.scheduler._step_ema_s + 0.1 * elapsed_step
+        )
+        self.num_steps += 1
+        return outputs
 
     def _sample_and_emit(self, sample_seqs, logits_f) -> List[RequestOutput]:
         """Regular sampling path: logits_f row i belongs to sample_seqs[i].

@@ -657,11 +657,13 @@ class ModelRunner:
         }
 
     def _fill_decode_inputs(self, payload, bs: int) -> None:
-        n = len(payload["ids"])
+        n = len(payload["lens"])
         st = self._static
-        st["ids"][:n].copy_(
-            torch.tensor(payload["ids"], dtype=torch.long), non_blocking=True
-        )
+        if payload["ids"] is not None:
+            st["ids"][:n].copy_(
+                torch.tensor(payload["ids"], dtype=torch.long),
+                non_blocking=True,
+            )
         st["positions"][:n].copy_(
             torch.tensor(payload["positions"], dtype=torch.int32),
             non_blocking=True,
@@ -687,6 +689,29 @@ class ModelRunner:
             st["slots"][n:bs].fill_(-1)
             st["seq_lens"][n:bs].fill_(1)
             st["block_tables"][n:bs].zero_()
+
+    def run_decode_device_ids(self, payload, ids_dev: torch.Tensor):
+        """Graph-replay decode with ON-DEVICE input ids: the pipelined
+        engine feeds the previous step's sampled ids straight into the
+        static graph buffer (device-to-device copy, no host sync).
+        payload["ids"] is None; the batch size comes from "lens"."""
+        n = len(payload["lens"])
+        if not self._static:
+            self._alloc_static(max(self.cfg.scheduler.max_num_seqs, n))
+        st = self._static
+        st["ids"][:n].copy_(ids_dev[:n])
+        bucket = next((b for b in _DECODE_BUCKETS if b >= n), None)
+        use_graph = (
+            bucket is not None and bucket in self._graphs and self.is_cuda
+        )
+        bs = bucket if use_graph else n
+        self._fill_decode_inputs(payload, bs)
+        if use_graph:
+            g, out = self._graphs[bucket]
+            g.replay()
+            return out[:n]
+        with torch.no_grad():
+            return self._decode_forward(n)[:n]
 
     def run_decode(self, payload) -> torch.Tensor:
         n = len(payload["ids"])

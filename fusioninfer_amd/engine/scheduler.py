@@ -77,6 +77,38 @@ class Scheduler:
         return bool(self.waiting or self.running)
 
     # ----------------------------------------------------------- schedule
+    def decode_only_next(self) -> bool:
+        """Read-only preview: would schedule() produce a full-composition
+        pure-decode batch (no admissions, swap-ins, prefills, or
+        preemptions by composition)? The engine's pipelined decode path
+        uses this to continue a graph-replay chain WITHOUT invoking
+        schedule()'s side effects (slot appends). Mirrors the admission
+        hysteresis below exactly; conservative on swapped heads."""
+        if not self.running:
+            return False
+        for s in self.running:
+            if (s.status != SeqStatus.RUNNING
+                    or s.num_computed_tokens < s.num_prompt_tokens):
+                return False
+        if len(self.running) >= self.cfg.max_num_seqs:
+            # full: neither the swap-in loop nor the admission loop can
+            # run (both gate on len(running) < max_num_seqs)
+            return True
+        if not self.waiting:
+            return True
+        if self.waiting[0].status == SeqStatus.SWAPPED:
+            return False
+        budget = self.cfg.max_num_batched_tokens
+        threshold = min(self.cfg.prefill_admission_tokens, budget)
+        if sum(s.num_prompt_tokens for s in self.waiting) >= threshold:
+            return False
+        if self.cfg.prefill_admission_ms is not None:
+            window_s = self.cfg.prefill_admission_ms / 1000.0
+        else:
+            window_s = min(max(
+                self.ADMISSION_STEPS * self._step_ema_s, 0.05), 0.25)
+        return time.monotonic() - self.waiting[0].arrival_time <= window_s
+
     def schedule(self) -> ScheduledBatch:
         """One MIXED batch per step (vLLM-v1-style unified scheduling):
         decode tokens for every fully-prefilled running sequence PLUS

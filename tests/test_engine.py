@@ -556,3 +556,83 @@ def test_priority_scheduling_admission_and_preemption():
     # the low-priority request was the preemption victim
     assert eng2.num_preemptions > 0
     assert done[0] == a
+
+
+# ---------------------------------------------------- pipelined decode
+def test_async_decode_matches_sync():
+    """The pipelined decode chain (deferred emission, device-fed ids)
+    must produce byte-identical greedy streams to the sync path, across
+    ragged max_tokens (deterministic finishers shrink the chain)."""
+    torch.manual_seed(0)
+    prompts = [[3, 1, 4, 1, 5] * 4, [2, 7] * 9, [11, 12, 13] * 7]
+    lens = [6, 17, 11]
+    sync = make_engine()
+    sync._async_decode = False
+    a = [
+        sync.generate([p], SamplingParams(max_tokens=n))[0].output_token_ids
+        for p, n in zip(prompts, lens)
+    ]
+    # batched, async (default): all three in flight together
+    eng = make_engine()
+    assert eng._async_decode
+    ids = [
+        eng.add_request(p, SamplingParams(max_tokens=n))
+        for p, n in zip(prompts, lens)
+    ]
+    done = {}
+    while eng.has_unfinished():
+        for out in eng.step():
+            if out.finished:
+                done[out.request_id] = out
+    assert eng.num_async_steps > 0  # the pipeline actually engaged
+    for rid, exp in zip(ids, a):
+        assert done[rid].output_token_ids == exp
+        assert done[rid].finish_reason == "length"
+
+
+def test_async_decode_abort_mid_chain():
+    """Aborting while a step is in flight drains the pipeline; other
+    requests' tokens survive via the backlog and finish correctly."""
+    torch.manual_seed(0)
+    eng = make_engine()
+    keep_exp = None
+    sync = make_engine()
+    sync._async_decode = False
+    keep_exp = sync.generate(
+        [[5, 6, 7] * 5], SamplingParams(max_tokens=12)
+    )[0].output_token_ids
+    r1 = eng.add_request([5, 6, 7] * 5, SamplingParams(max_tokens=12))
+    r2 = eng.add_request([9, 9, 2] * 5, SamplingParams(max_tokens=40))
+    for _ in range(6):
+        eng.step()
+    assert eng._pending is not None
+    assert eng.abort_request(r2)
+    assert eng._pending is None  # chain flushed
+    done = {}
+    while eng.has_unfinished():
+        for out in eng.step():
+            if out.finished:
+                done[out.request_id] = out
+    assert done[r1].output_token_ids == keep_exp
+
+
+def test_async_decode_mixed_admission_breaks_chain():
+    """A prompt arriving mid-chain breaks the pipeline at the admission
+    window and still prefills + finishes correctly."""
+    torch.manual_seed(0)
+    eng = make_engine()
+    r1 = eng.add_request([1, 2, 3] * 6, SamplingParams(max_tokens=30))
+    for _ in range(5):
+        eng.step()
+    assert eng._pending is not None
+    r2 = eng.add_request([4, 5] * 8, SamplingParams(max_tokens=5))
+    done = {}
+    steps = 0
+    while eng.has_unfinished() and steps < 200:
+        steps += 1
+        for out in eng.step():
+            if out.finished:
+                done[out.request_id] = out
+    assert set(done) == {r1, r2}
+    assert len(done[r2].output_token_ids) == 5
+    assert len(done[r1].output_token_ids) == 30

@@ -361,3 +361,31 @@ def test_gpu_attention_bias_engine_generates():
         layer.self_attn.qkv_proj.bias.data.normal_(0, 0.05)
     outs = eng.generate([[5, 3, 1] * 8], SamplingParams(max_tokens=6))
     assert len(outs[0].output_token_ids) == 6
+
+
+def test_async_decode_matches_sync_gpu():
+    """Pipelined decode on the REAL path (hipGraph replay, device-fed
+    ids, event-synced pinned readback) must match the sync engine's
+    greedy streams exactly; ragged max_tokens shrink the chain."""
+    torch.manual_seed(3)
+    prompts = [[3, 1, 4, 1, 5] * 8, [2, 7] * 15, [11, 12, 13] * 9]
+    lens = [9, 21, 14]
+    sync = make_engine(enforce_eager=False)
+    sync._async_decode = False
+    exp = [
+        sync.generate([p], SamplingParams(max_tokens=n))[0].output_token_ids
+        for p, n in zip(prompts, lens)
+    ]
+    eng = make_engine(enforce_eager=False)
+    ids = [
+        eng.add_request(p, SamplingParams(max_tokens=n))
+        for p, n in zip(prompts, lens)
+    ]
+    done = {}
+    while eng.has_unfinished():
+        for out in eng.step():
+            if out.finished:
+                done[out.request_id] = out
+    assert eng.num_async_steps > 0
+    for rid, e in zip(ids, exp):
+        assert done[rid].output_token_ids == e
