@@ -484,6 +484,11 @@ class MoEMLP(nn.Module):
         - n_valid [1] int32: real tile count (device scalar)
         - pos [T*topk] int32: padded slot per assignment (-1 = non-local)
         """
+        if topi.is_cuda:
+            # one fused kernel instead of this ~12-op torch composition
+            # (same contract; within-expert slot order is arrival order,
+            # which is output-invariant — see ops.moe_align)
+            return ops.moe_align(topi, self.e_start, self.e_end, block_m)
         T, k = topi.shape
         dev = topi.device
         E_local = self.e_end - self.e_start

@@ -389,6 +389,29 @@ def moe_gemm_fp8(out, a, a_scales, b_packed, b_scales, sorted_ids,
                             gate_up)
 
 
+def moe_align(topi: torch.Tensor, e_start: int, e_end: int, block_m: int):
+    """Single-kernel device-side block alignment of expert assignments
+    (GPU only; MoEMLP._moe_align's torch composition is the CPU
+    reference and the semantic oracle). Within-expert slot order is
+    arrival order rather than stable order — downstream results are
+    bit-identical either way (each row's K-loop and the pos-gathered
+    combine don't depend on the slot). Returns
+    (sorted_ids [PM], expert_ids [PM/bm], n_valid [1], pos [T*k], PM)."""
+    _require_native()
+    T, k = topi.shape
+    n = T * k
+    E_local = e_end - e_start
+    PM = ((n + block_m - 1) // block_m + E_local) * block_m
+    dev = topi.device
+    sorted_ids = torch.empty(PM, dtype=torch.int32, device=dev)
+    expert_ids = torch.empty(PM // block_m, dtype=torch.int32, device=dev)
+    n_valid = torch.empty(1, dtype=torch.int32, device=dev)
+    pos = torch.empty(n, dtype=torch.int32, device=dev)
+    _C.moe_align(topi.reshape(-1).int().contiguous(), sorted_ids,
+                 expert_ids, n_valid, pos, k, e_start, e_end, block_m)
+    return sorted_ids, expert_ids, n_valid, pos, PM
+
+
 def moe_combine(out, y, pos, w):
     """out[t] = sum_k w[t,k] * y[pos[t,k]] (pos < 0 skipped). Deterministic
     (no atomics) so token-exact tests stay reproducible."""

@@ -39,6 +39,8 @@ void launch_moe_gemm_fp8(u16*, const unsigned char*, const float*,
                          const unsigned char*, const float*, const int*,
                          const int*, const int*, int, int, int, int, bool,
                          hipStream_t);
+void launch_moe_align(const int*, int*, int*, int*, int*, int, int, int,
+                      int, int, int, hipStream_t);
 void launch_moe_combine(u16*, const u16*, const int*, const float*, int, int,
                         int, hipStream_t);
 
@@ -416,6 +418,26 @@ void moe_combine(at::Tensor out, at::Tensor y, at::Tensor pos, at::Tensor w) {
                          w.data_ptr<float>(), T, topk, H, current_stream());
 }
 
+void moe_align(at::Tensor topi, at::Tensor sorted_ids, at::Tensor expert_ids,
+               at::Tensor n_valid, at::Tensor pos, int64_t topk,
+               int64_t e_start, int64_t e_end, int64_t block_m) {
+  TORCH_CHECK(topi.is_cuda() && topi.scalar_type() == at::kInt &&
+              topi.is_contiguous());
+  TORCH_CHECK(sorted_ids.scalar_type() == at::kInt &&
+              expert_ids.scalar_type() == at::kInt &&
+              n_valid.scalar_type() == at::kInt &&
+              pos.scalar_type() == at::kInt);
+  const int n = topi.numel();
+  const int PM = sorted_ids.numel();
+  TORCH_CHECK(e_end - e_start <= 128, "moe_align: E_local > 128");
+  TORCH_CHECK(PM % block_m == 0 && expert_ids.numel() == PM / block_m);
+  TORCH_CHECK(pos.numel() == n);
+  fi::launch_moe_align(topi.data_ptr<int>(), sorted_ids.data_ptr<int>(),
+                       expert_ids.data_ptr<int>(), n_valid.data_ptr<int>(),
+                       pos.data_ptr<int>(), n, (int)topk, (int)e_start,
+                       (int)e_end, (int)block_m, PM, current_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -447,4 +469,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "epilogue (gate_up=true fuses SwiGLU)");
   m.def("moe_combine", &moe_combine,
         "weighted top-k combine of expert outputs (deterministic)");
+  m.def("moe_align", &moe_align,
+        "single-kernel block alignment of expert assignments "
+        "(vLLM moe_align_block_size analog)");
 }

@@ -561,3 +561,46 @@ def test_moe_gemm_fp8_vs_reference():
         torch.testing.assert_close(out[:rows].float().cpu(),
                                    want[:rows].float(),
                                    atol=5e-2, rtol=5e-2)
+
+
+def test_moe_align_kernel_matches_python():
+    """The fused alignment kernel vs the torch composition (the CPU/
+    semantic oracle, which also runs on GPU tensors): tile counts and
+    expert_ids must be EQUAL; slot assignments may permute within an
+    expert's segment (arrival vs stable order) but must satisfy the
+    same invariants."""
+    import types
+
+    from fusioninfer_amd.models.model import MoEMLP
+
+    torch.manual_seed(5)
+    cases = [
+        (13, 2, 8, 0, 8, 16),     # small, all local
+        (300, 8, 16, 4, 12, 16),  # EP slice: half the experts non-local
+        (256, 8, 8, 0, 8, 128),   # prefill block_m
+        (1, 8, 128, 0, 128, 16),  # decode single token, many experts
+    ]
+    for T, k, E, e0, e1, bm in cases:
+        topi = torch.randint(0, E, (T, k), device=DEV)
+        fake = types.SimpleNamespace(e_start=e0, e_end=e1)
+        p_sid, p_eid, p_nv, p_pos, p_PM = MoEMLP._moe_align(fake, topi, bm)
+        sid, eid, nv, pos, PM = ops.moe_align(topi, e0, e1, bm)
+        assert PM == p_PM
+        assert int(nv.item()) == int(p_nv.item())
+        assert torch.equal(eid, p_eid)
+        flat = topi.reshape(-1).cpu()
+        pos_c, sid_c = pos.cpu(), sid.cpu()
+        local = [(int(e) >= e0) and (int(e) < e1) for e in flat]
+        seen = set()
+        for i, e in enumerate(flat.tolist()):
+            p = int(pos_c[i])
+            if not local[i]:
+                assert p == -1
+                continue
+            assert 0 <= p < PM and p not in seen
+            seen.add(p)
+            assert int(sid_c[p]) == i // k          # row gather correct
+            assert int(eid[p // bm]) == e - e0      # right expert tile
+        # padding rows are zeroed (the GEMM's dummy-row contract)
+        pad = [i for i in range(PM) if i not in seen]
+        assert all(int(sid_c[i]) == 0 for i in pad)
