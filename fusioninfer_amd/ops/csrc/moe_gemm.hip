@@ -168,33 +168,77 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
     }
 
   const int ksteps = K / 32;
-  for (int kt = 0; kt < ksteps; ++kt) {
-    if (STAGE_A && kt + 1 < ksteps) stage_a(kt + 1, (kt + 1) & 1);
-    short8 av[MITER];
-    const u16* abuf = STAGE_A ? a_lds + (kt & 1) * (kATileB / 2) : nullptr;
+  if (STAGE_A) {
+    for (int kt = 0; kt < ksteps; ++kt) {
+      if (kt + 1 < ksteps) stage_a(kt + 1, (kt + 1) & 1);
+      short8 av[MITER];
+      const u16* abuf = a_lds + (kt & 1) * (kATileB / 2);
 #pragma unroll
-    for (int mi = 0; mi < MITER; ++mi)
-      av[mi] = STAGE_A ? load_b16x8(abuf + a_lds_off[mi])
-                       : load_b16x8(a_p[mi] + kt * 32);
+      for (int mi = 0; mi < MITER; ++mi)
+        av[mi] = load_b16x8(abuf + a_lds_off[mi]);
 #pragma unroll
-    for (int ni = 0; ni < NITER; ++ni) {
-      const short8 bg = load_b16x8(bg_p[ni] + kt * b_step);
-      short8 bu;
-      if (GATE_UP) bu = load_b16x8(bu_p[ni] + kt * b_step);
+      for (int ni = 0; ni < NITER; ++ni) {
+        const short8 bg = load_b16x8(bg_p[ni] + kt * b_step);
+        short8 bu;
+        if (GATE_UP) bu = load_b16x8(bu_p[ni] + kt * b_step);
 #pragma unroll
-      for (int mi = 0; mi < MITER; ++mi) {
-        acc_g[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            av[mi], bg, acc_g[mi][ni], 0, 0, 0);
-        if (GATE_UP)
-          acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              av[mi], bu, acc_u[mi][ni], 0, 0, 0);
+        for (int mi = 0; mi < MITER; ++mi) {
+          acc_g[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              av[mi], bg, acc_g[mi][ni], 0, 0, 0);
+          if (GATE_UP)
+            acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                av[mi], bu, acc_u[mi][ni], 0, 0, 0);
+        }
       }
-    }
-    if (STAGE_A) {
       // 2-phase barrier: drains this step's prefetch glds (next tile
       // becomes readable) and closes the LDS reads of the current one
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
+    }
+  } else {
+    // register double-buffered software pipeline: the NEXT K-step's
+    // A/B fragments stay in flight while this step's MFMAs issue. The
+    // decode-regime variants (BM=16, no LDS) were LATENCY-bound at
+    // ~50% of the weight-BW roofline with the single-buffer loop (one
+    // load batch outstanding per iteration). Manual unroll-by-2 keeps
+    // the buffer index compile-time (runtime-indexed register arrays
+    // spill to scratch — guide rule 20).
+    constexpr int NU = GATE_UP ? NITER : 1;
+    short8 av0[MITER], av1[MITER];
+    short8 bg0[NITER], bg1[NITER];
+    short8 bu0[NU], bu1[NU];
+    auto issue = [&](short8 (&avd)[MITER], short8 (&bgd)[NITER],
+                     short8 (&bud)[NU], int kt) {
+#pragma unroll
+      for (int mi = 0; mi < MITER; ++mi)
+        avd[mi] = load_b16x8(a_p[mi] + kt * 32);
+#pragma unroll
+      for (int ni = 0; ni < NITER; ++ni) {
+        bgd[ni] = load_b16x8(bg_p[ni] + kt * b_step);
+        if (GATE_UP) bud[ni] = load_b16x8(bu_p[ni] + kt * b_step);
+      }
+    };
+    auto consume = [&](short8 (&avv)[MITER], short8 (&bgv)[NITER],
+                       short8 (&buv)[NU]) {
+#pragma unroll
+      for (int ni = 0; ni < NITER; ++ni)
+#pragma unroll
+        for (int mi = 0; mi < MITER; ++mi) {
+          acc_g[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              avv[mi], bgv[ni], acc_g[mi][ni], 0, 0, 0);
+          if (GATE_UP)
+            acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                avv[mi], buv[ni % NU], acc_u[mi][ni], 0, 0, 0);
+        }
+    };
+    issue(av0, bg0, bu0, 0);
+    for (int kt = 0; kt < ksteps; kt += 2) {
+      if (kt + 1 < ksteps) issue(av1, bg1, bu1, kt + 1);
+      consume(av0, bg0, bu0);
+      if (kt + 1 < ksteps) {
+        if (kt + 2 < ksteps) issue(av0, bg0, bu0, kt + 2);
+        consume(av1, bg1, bu1);
+      }
     }
   }
 
@@ -330,34 +374,71 @@ __global__ __launch_bounds__(256) void moe_gemm_fp8_kernel(
     }
 
   const int ksteps = K / 32;
-  for (int kt = 0; kt < ksteps; ++kt) {
-    if (STAGE_A && kt + 1 < ksteps) stage_a(kt + 1, (kt + 1) & 1);
-    const unsigned char* abuf =
-        STAGE_A ? a8_lds + (kt & 1) * kATileB : nullptr;
-    i64 av[MITER];
+  if (STAGE_A) {
+    for (int kt = 0; kt < ksteps; ++kt) {
+      if (kt + 1 < ksteps) stage_a(kt + 1, (kt + 1) & 1);
+      const unsigned char* abuf = a8_lds + (kt & 1) * kATileB;
+      i64 av[MITER];
 #pragma unroll
-    for (int mi = 0; mi < MITER; ++mi)
-      av[mi] = STAGE_A
-          ? *reinterpret_cast<const i64*>(abuf + a_lds_off[mi])
-          : *reinterpret_cast<const i64*>(a_p[mi] + kt * 32);
+      for (int mi = 0; mi < MITER; ++mi)
+        av[mi] = *reinterpret_cast<const i64*>(abuf + a_lds_off[mi]);
 #pragma unroll
-    for (int ni = 0; ni < NITER; ++ni) {
-      const i64 bg = *reinterpret_cast<const i64*>(bg_p[ni] + kt * b_step);
-      i64 bu = 0;
-      if (GATE_UP)
-        bu = *reinterpret_cast<const i64*>(bu_p[ni] + kt * b_step);
-#pragma unroll
-      for (int mi = 0; mi < MITER; ++mi) {
-        acc_g[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-            av[mi], bg, acc_g[mi][ni], 0, 0, 0);
+      for (int ni = 0; ni < NITER; ++ni) {
+        const i64 bg = *reinterpret_cast<const i64*>(bg_p[ni] + kt * b_step);
+        i64 bu = 0;
         if (GATE_UP)
-          acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-              av[mi], bu, acc_u[mi][ni], 0, 0, 0);
+          bu = *reinterpret_cast<const i64*>(bu_p[ni] + kt * b_step);
+#pragma unroll
+        for (int mi = 0; mi < MITER; ++mi) {
+          acc_g[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              av[mi], bg, acc_g[mi][ni], 0, 0, 0);
+          if (GATE_UP)
+            acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                av[mi], bu, acc_u[mi][ni], 0, 0, 0);
+        }
       }
-    }
-    if (STAGE_A) {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
+    }
+  } else {
+    // register double-buffered pipeline (see the bf16 kernel: the
+    // BM=16 decode variants are latency-bound without it; unroll-by-2
+    // keeps buffer indices compile-time — rule 20)
+    constexpr int NU = GATE_UP ? NITER : 1;
+    i64 av0[MITER], av1[MITER], bg0[NITER], bg1[NITER], bu0[NU], bu1[NU];
+    auto issue = [&](i64 (&avd)[MITER], i64 (&bgd)[NITER],
+                     i64 (&bud)[NU], int kt) {
+#pragma unroll
+      for (int mi = 0; mi < MITER; ++mi)
+        avd[mi] = *reinterpret_cast<const i64*>(a_p[mi] + kt * 32);
+#pragma unroll
+      for (int ni = 0; ni < NITER; ++ni) {
+        bgd[ni] = *reinterpret_cast<const i64*>(bg_p[ni] + kt * b_step);
+        if (GATE_UP)
+          bud[ni] = *reinterpret_cast<const i64*>(bu_p[ni] + kt * b_step);
+      }
+    };
+    auto consume = [&](i64 (&avv)[MITER], i64 (&bgv)[NITER],
+                       i64 (&buv)[NU]) {
+#pragma unroll
+      for (int ni = 0; ni < NITER; ++ni)
+#pragma unroll
+        for (int mi = 0; mi < MITER; ++mi) {
+          acc_g[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              avv[mi], bgv[ni], acc_g[mi][ni], 0, 0, 0);
+          if (GATE_UP)
+            acc_u[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                avv[mi], buv[ni % NU], acc_u[mi][ni], 0, 0, 0);
+        }
+    };
+    issue(av0, bg0, bu0, 0);
+    for (int kt = 0; kt < ksteps; kt += 2) {
+      if (kt + 1 < ksteps) issue(av1, bg1, bu1, kt + 1);
+      consume(av0, bg0, bu0);
+      if (kt + 1 < ksteps) {
+        if (kt + 2 < ksteps) issue(av0, bg0, bu0, kt + 2);
+        consume(av1, bg1, bu1);
+      }
     }
   }
 
