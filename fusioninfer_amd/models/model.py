@@ -438,10 +438,7 @@ class MoEMLP(nn.Module):
         T, H = x8.shape
         xd = x8.float() * xs.unsqueeze(1).float()
         logits = xd @ self.router_weight.float().T
-        probs = torch.softmax(logits, dim=-1)
-        topv, topi = probs.topk(self.top_k, dim=-1)
-        if self.norm_topk:
-            topv = topv / topv.sum(dim=-1, keepdim=True)
+        topv, topi = self._router_topk(logits)
         E_local = self.e_end - self.e_start
         block_m = 128 if T * self.top_k >= 64 * E_local else 16
         sorted_ids, expert_ids, n_valid, pos, PM = self._moe_align(
@@ -474,6 +471,18 @@ class MoEMLP(nn.Module):
             )
             self._pack_version = v
         return self._pack_cache
+
+    def _router_topk(self, logits: torch.Tensor):
+        """Router tail: softmax + top-k + optional renormalize. GPU uses
+        the fused wave-per-token kernel; the torch composition is the
+        CPU path and the semantic reference."""
+        if logits.is_cuda:
+            return ops.moe_router_topk(logits, self.top_k, self.norm_topk)
+        probs = torch.softmax(logits, dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)
+        if self.norm_topk:
+            topv = topv / topv.sum(dim=-1, keepdim=True)
+        return topv, topi
 
     def _moe_align(self, topi: torch.Tensor, block_m: int):
         """Device-side block alignment (no host sync, static shapes —
@@ -536,10 +545,7 @@ class MoEMLP(nn.Module):
         T, H = x.shape
         E_local = self.e_end - self.e_start
         logits = x.float() @ self.router_weight.float().T
-        probs = torch.softmax(logits, dim=-1)
-        topv, topi = probs.topk(self.top_k, dim=-1)
-        if self.norm_topk:
-            topv = topv / topv.sum(dim=-1, keepdim=True)
+        topv, topi = self._router_topk(logits)
         # block_m: 16 in the few-rows-per-expert (decode) regime, 128 when
         # segments are long enough for the 4x in-register B-reuse variant
         block_m = 128 if T * self.top_k >= 64 * E_local else 16

@@ -412,6 +412,19 @@ def moe_align(topi: torch.Tensor, e_start: int, e_end: int, block_m: int):
     return sorted_ids, expert_ids, n_valid, pos, PM
 
 
+def moe_router_topk(logits: torch.Tensor, k: int, renorm: bool):
+    """Fused router tail (GPU): softmax + top-k + optional top-k
+    renormalization in one wave-per-token kernel. logits [T, E] fp32;
+    returns (topv [T,k] fp32 weights, topi [T,k] int32 expert ids,
+    descending; ties pick the smaller index)."""
+    _require_native()
+    T, E = logits.shape
+    topv = torch.empty(T, k, dtype=torch.float32, device=logits.device)
+    topi = torch.empty(T, k, dtype=torch.int32, device=logits.device)
+    _C.moe_router_topk(logits.contiguous(), topv, topi, k, renorm)
+    return topv, topi
+
+
 def moe_combine(out, y, pos, w):
     """out[t] = sum_k w[t,k] * y[pos[t,k]] (pos < 0 skipped). Deterministic
     (no atomics) so token-exact tests stay reproducible."""

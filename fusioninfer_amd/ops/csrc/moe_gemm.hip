@@ -674,4 +674,66 @@ void launch_moe_align(const int* topi, int* sorted_ids, int* expert_ids,
                      e_end, block_m, PM);
 }
 
+// Fused router tail: softmax + top-k + (optional) renormalize in ONE
+// wave-per-token kernel (replaces torch softmax/gatherTopK/bitonic-
+// sort/div — ~3.5% of 30B MoE GPU time and 6 launches per layer).
+// With renorm (norm_topk_prob), weights = softmax over the top-k
+// LOGITS (the global partition cancels); without it, global softmax
+// probabilities. Ties pick the smaller expert index.
+namespace {
+__global__ void moe_router_topk_kernel(
+    const float* __restrict__ logits,  // [T, E] fp32
+    float* __restrict__ topv,          // [T, k]
+    int* __restrict__ topi,            // [T, k]
+    const int T, const int E, const int k, const bool renorm) {
+  const int t = blockIdx.x * (blockDim.x / kWaveSize)
+      + threadIdx.x / kWaveSize;
+  if (t >= T) return;
+  const int lane = threadIdx.x % kWaveSize;
+  const float* row = logits + static_cast<int64_t>(t) * E;
+  // two logit slots per lane (E <= 128)
+  float l0 = lane < E ? row[lane] : -INFINITY;
+  float l1 = lane + 64 < E ? row[lane + 64] : -INFINITY;
+  float m = fmaxf(l0, l1);
+#pragma unroll
+  for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_xor(m, off));
+  float s = (lane < E ? __expf(l0 - m) : 0.f) +
+            (lane + 64 < E ? __expf(l1 - m) : 0.f);
+#pragma unroll
+  for (int off = 32; off; off >>= 1) s += __shfl_xor(s, off);
+  float a0 = l0, a1 = l1;  // remaining candidates
+  float ksum = 0.f;
+  for (int it = 0; it < k; ++it) {
+    float v = a0 >= a1 ? a0 : a1;
+    int idx = a0 >= a1 ? lane : lane + 64;
+#pragma unroll
+    for (int off = 32; off; off >>= 1) {
+      const float ov = __shfl_xor(v, off);
+      const int oi = __shfl_xor(idx, off);
+      if (ov > v || (ov == v && oi < idx)) { v = ov; idx = oi; }
+    }
+    const float e = __expf(v - m);
+    ksum += e;
+    if (lane == 0) {
+      topi[t * k + it] = idx;
+      topv[t * k + it] = renorm ? e : e / s;
+    }
+    if (idx == lane) a0 = -INFINITY;
+    if (idx == lane + 64) a1 = -INFINITY;
+  }
+  if (renorm && lane == 0) {
+    for (int it = 0; it < k; ++it) topv[t * k + it] /= ksum;
+  }
+}
+}  // namespace
+
+void launch_moe_router_topk(const float* logits, float* topv, int* topi,
+                            int T, int E, int k, bool renorm,
+                            hipStream_t stream) {
+  const int waves = 4;  // 4 tokens per 256-thread block
+  hipLaunchKernelGGL(moe_router_topk_kernel,
+                     dim3(ceil_div(T, waves)), dim3(waves * kWaveSize), 0,
+                     stream, logits, topv, topi, T, E, k, renorm);
+}
+
 }  // namespace fi

@@ -41,6 +41,8 @@ void launch_moe_gemm_fp8(u16*, const unsigned char*, const float*,
                          hipStream_t);
 void launch_moe_align(const int*, int*, int*, int*, int*, int, int, int,
                       int, int, int, hipStream_t);
+void launch_moe_router_topk(const float*, float*, int*, int, int, int,
+                            bool, hipStream_t);
 void launch_moe_combine(u16*, const u16*, const int*, const float*, int, int,
                         int, hipStream_t);
 
@@ -438,6 +440,21 @@ void moe_align(at::Tensor topi, at::Tensor sorted_ids, at::Tensor expert_ids,
                        (int)e_end, (int)block_m, PM, current_stream());
 }
 
+void moe_router_topk(at::Tensor logits, at::Tensor topv, at::Tensor topi,
+                     int64_t k, bool renorm) {
+  TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == at::kFloat &&
+              logits.is_contiguous());
+  TORCH_CHECK(topv.scalar_type() == at::kFloat &&
+              topi.scalar_type() == at::kInt);
+  const int T = logits.size(0);
+  const int E = logits.size(1);
+  TORCH_CHECK(E <= 128 && k <= E);
+  TORCH_CHECK(topv.numel() == T * k && topi.numel() == T * k);
+  fi::launch_moe_router_topk(logits.data_ptr<float>(),
+                             topv.data_ptr<float>(), topi.data_ptr<int>(),
+                             T, E, (int)k, renorm, current_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -472,4 +489,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_align", &moe_align,
         "single-kernel block alignment of expert assignments "
         "(vLLM moe_align_block_size analog)");
+  m.def("moe_router_topk", &moe_router_topk,
+        "fused router tail: softmax + top-k + optional renormalize");
 }

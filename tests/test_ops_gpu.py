@@ -604,3 +604,20 @@ def test_moe_align_kernel_matches_python():
         # padding rows are zeroed (the GEMM's dummy-row contract)
         pad = [i for i in range(PM) if i not in seen]
         assert all(int(sid_c[i]) == 0 for i in pad)
+
+
+def test_moe_router_topk_matches_torch():
+    """Fused router tail vs the torch composition: same experts, same
+    weights, same (descending) order. Random fp32 logits are distinct
+    almost surely, so exact order comparison is stable."""
+    torch.manual_seed(9)
+    for T, E, k, renorm in [(7, 8, 2, True), (256, 128, 8, True),
+                            (33, 64, 4, False), (1, 128, 8, True)]:
+        logits = torch.randn(T, E, device=DEV, dtype=torch.float32)
+        tv, ti = ops.moe_router_topk(logits, k, renorm)
+        probs = torch.softmax(logits, -1)
+        etv, eti = probs.topk(k, -1)
+        if renorm:
+            etv = etv / etv.sum(-1, keepdim=True)
+        assert torch.equal(ti.long(), eti), (T, E, k)
+        torch.testing.assert_close(tv, etv, atol=2e-6, rtol=2e-6)
