@@ -258,6 +258,28 @@ def test_guided_choice_and_json_object(app):
             obj = json.loads(r.json()["choices"][0]["text"])
             assert isinstance(obj.get("ok"), bool)
 
+            # optional properties: required subset must be present, any
+            # extra emitted key must be a declared optional in order
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "as json:", "max_tokens": 64,
+                      "temperature": 0.0,
+                      "guided_json": {
+                          "type": "object",
+                          "properties": {
+                              "name": {"enum": ["a", "b"]},
+                              "age": {"type": "integer",
+                                      "minimum": 0, "maximum": 9},
+                              "ok": {"type": "boolean"},
+                          },
+                          "required": ["name", "ok"],
+                      }},
+            )
+            assert r.status_code == 200
+            obj = json.loads(r.json()["choices"][0]["text"])
+            assert obj["name"] in {"a", "b"} and isinstance(obj["ok"], bool)
+            assert set(obj) <= {"name", "age", "ok"}
+
             # OpenAI response_format on chat: output must stay a viable
             # JSON prefix even if max_tokens truncates it
             r = await c.post(
