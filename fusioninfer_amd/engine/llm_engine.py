@@ -236,6 +236,35 @@ class LLMEngine:
     def has_unfinished(self) -> bool:
         return self.scheduler.has_work()
 
+    def embed(self, prompts: List[List[int]], pooling: str = "last"):
+        """Embeddings (vLLM embed task / OpenAI /v1/embeddings): one
+        prefill forward per prompt, pooled final hidden state
+        (pre-lm_head), L2-normalized. Temporary blocks come from the
+        engine's pool and are freed immediately. Driver-only single-rank
+        path (TP/PP embedding is not wired through the worker loop)."""
+        tp = (self.cfg.parallel.tensor_parallel_size
+              * self.cfg.parallel.pipeline_parallel_size)
+        if tp != 1:
+            raise ValueError("embeddings unsupported with TP/PP")
+        outs = []
+        for p in prompts:
+            seq = Sequence(f"embed-{next(self._req_counter)}", p,
+                           SamplingParams())
+            self.block_manager.allocate(seq)
+            # force full recompute: a prefix-cache hit would shrink the
+            # payload to the uncached suffix (empty for a full hit) and
+            # the pooled hidden must cover the whole prompt
+            seq.num_cached_tokens = 0
+            try:
+                h = self.runner.execute_prefill_hidden(
+                    [seq], self.block_manager, pooling=pooling
+                )[0]
+            finally:
+                self.block_manager.free(seq)
+            v = h / h.norm().clamp_min(1e-12)
+            outs.append(v.cpu().tolist())
+        return outs
+
     def abort_request(self, request_id: str) -> bool:
         """Cancel a request wherever it is (waiting / running); frees its
         cache blocks. Returns True if it was found."""

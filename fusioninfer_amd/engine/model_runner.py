@@ -423,6 +423,28 @@ class ModelRunner:
     def execute_prefill(self, seqs: List[Sequence], bm: BlockManager):
         return self.run_prefill(self.build_prefill_payload(seqs, bm))
 
+    def execute_prefill_hidden(self, seqs: List[Sequence], bm: BlockManager,
+                               pooling: str = "last") -> torch.Tensor:
+        """Embeddings path (vLLM embed-task analog): one full prefill
+        forward, return the POOLED final hidden state per sequence
+        (pre-lm_head, [len(seqs), H]). pooling: "last" (decoder-LM
+        default) or "mean". PP unsupported (hidden lives on the last
+        stage only)."""
+        import fusioninfer_amd.distributed.parallel_state as ps
+
+        assert ps.pp_world_size() == 1, "embeddings: PP unsupported"
+        payload = self.build_prefill_payload(seqs, bm)
+        ids, meta, _ = self._prefill_inputs(payload)
+        with torch.no_grad():
+            hidden = self.model(ids, meta, self.kv_caches)
+        cu = payload["cu"]
+        outs = []
+        for j in range(len(seqs)):
+            h = hidden[cu[j]:cu[j + 1]]
+            outs.append(h.float().mean(0) if pooling == "mean"
+                        else h[-1].float())
+        return torch.stack(outs)
+
     # ------------------------------------------------------------- mixed
     def build_batch_payload(
         self,

@@ -360,6 +360,53 @@ def build_app(serving: ServingEngine, model_name: str,
             [int(t) for t in body.get("tokens", [])]
         )}
 
+    @app.post("/v1/embeddings")
+    async def embeddings(request: Request):
+        """OpenAI embeddings: last-token pooled, L2-normalized hidden
+        state of a decoder-LM prefill (vLLM embed-task semantics).
+        input: str | [str] | [int] | [[int]]."""
+        import asyncio
+
+        body = await request.json()
+        inp = body.get("input", "")
+        if isinstance(inp, str):
+            items = [inp]
+        elif isinstance(inp, list) and inp and isinstance(inp[0], int):
+            items = [inp]
+        elif isinstance(inp, list):
+            items = inp
+        else:
+            return JSONResponse(
+                {"error": {"message": "invalid input",
+                           "type": "invalid_request_error"}}, 400)
+        max_len = serving.engine.cfg.scheduler.max_model_len
+        pooling = body.get("pooling", "last")
+        data = []
+        n_prompt = 0
+        for i, item in enumerate(items):
+            ids = _encode(item) if isinstance(item, str) else \
+                [int(t) for t in item]
+            if not ids or len(ids) > max_len:
+                return JSONResponse(
+                    {"error": {"message": f"input {i}: length {len(ids)} "
+                               f"not in [1, {max_len}]",
+                               "type": "invalid_request_error"}}, 400)
+            try:
+                vec = await asyncio.to_thread(serving.embed, ids, pooling)
+            except ValueError as e:
+                return JSONResponse(
+                    {"error": {"message": str(e),
+                               "type": "invalid_request_error"}}, 400)
+            n_prompt += len(ids)
+            data.append({"object": "embedding", "index": i,
+                         "embedding": vec})
+        return {
+            "object": "list",
+            "data": data,
+            "model": body.get("model") or model_name,
+            "usage": {"prompt_tokens": n_prompt, "total_tokens": n_prompt},
+        }
+
     @app.post("/pd/prefill")
     async def pd_prefill(request: Request):
         """PD producer: prefill the prompt, ship its KV to the decoder, and

@@ -210,6 +210,12 @@ class ServingEngine:
         self._work.set()
         return req_id, q
 
+    def embed(self, prompt_token_ids, pooling: str = "last"):
+        """One pooled-embedding forward under the engine lock (runs
+        between engine steps, like the PD sender)."""
+        with self._lock:
+            return self.engine.embed([prompt_token_ids], pooling=pooling)[0]
+
     def metrics(self) -> Dict[str, float]:
         e = self.engine
         return {

@@ -389,3 +389,12 @@ def test_async_decode_matches_sync_gpu():
     assert eng.num_async_steps > 0
     for rid, e in zip(ids, exp):
         assert done[rid].output_token_ids == e
+
+
+def test_gpu_embeddings():
+    """Pooled-embedding forward on the real kernels: unit norm,
+    input-dependent."""
+    eng = make_engine(enforce_eager=False)
+    vs = eng.embed([[5, 6, 7] * 10, [9, 8] * 12])
+    assert abs(sum(x * x for x in vs[0]) - 1.0) < 1e-2
+    assert vs[0] != vs[1]

@@ -688,3 +688,35 @@ def test_finish_reason_length_vs_stop(app):
             assert reasons == ["length"]
 
     asyncio.run(run())
+
+
+def test_embeddings_endpoint(app):
+    """OpenAI /v1/embeddings: unit-norm pooled vectors, deterministic
+    per input, distinct across inputs; token-id input and length
+    validation."""
+    import math as _math
+
+    async def run():
+        async with _client(app) as c:
+            r = await c.post("/v1/embeddings",
+                             json={"input": ["hello world", "other text"]})
+            assert r.status_code == 200
+            out = r.json()
+            assert out["object"] == "list" and len(out["data"]) == 2
+            v0 = out["data"][0]["embedding"]
+            v1 = out["data"][1]["embedding"]
+            assert abs(_math.fsum(x * x for x in v0) - 1.0) < 1e-3
+            assert v0 != v1
+            assert out["usage"]["prompt_tokens"] > 0
+
+            r2 = await c.post("/v1/embeddings",
+                              json={"input": "hello world"})
+            assert r2.json()["data"][0]["embedding"] == v0  # deterministic
+
+            r3 = await c.post("/v1/embeddings", json={"input": [5, 6, 7]})
+            assert r3.status_code == 200
+            r4 = await c.post("/v1/embeddings",
+                              json={"input": list(range(500))})
+            assert r4.status_code == 400
+
+    asyncio.run(run())
