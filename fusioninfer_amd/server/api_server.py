@@ -109,7 +109,9 @@ def build_app(serving: ServingEngine, model_name: str,
 
     @app.get("/health")
     async def health():
-        if not serving.healthy:
+        # check_health also trips on a HUNG engine step (watchdog) so a
+        # wedged collective fails the k8s liveness probe -> pod restart
+        if not serving.check_health():
             return JSONResponse(
                 {"status": "unhealthy", "error": serving.last_error}, 503
             )

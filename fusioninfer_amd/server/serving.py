@@ -8,6 +8,7 @@ per-request token deltas through thread-safe queues.
 from __future__ import annotations
 
 import itertools
+import os
 import queue
 import threading
 import time
@@ -42,6 +43,14 @@ class ServingEngine:
         self._stop = False
         self.healthy = True
         self.last_error: str = ""
+        # engine-hang watchdog (failure detection is
+        # Kubernetes-native — a wedged collective / kernel turns /health
+        # 503 so the liveness probe restarts the pod). A step that has
+        # been running longer than this is considered hung.
+        self.step_timeout_s: float = float(
+            os.environ.get("FI_STEP_TIMEOUT_S", "120")
+        )
+        self._step_started: Optional[float] = None
         # PD disaggregation: producer prefills+ships KV,
         # consumer receives KV on a background thread and admits on claim
         self.kv_connector = kv_connector
@@ -210,6 +219,21 @@ class ServingEngine:
         self._work.set()
         return req_id, q
 
+    def check_health(self) -> bool:
+        """Liveness: False once the engine loop died on an exception OR
+        the current step has been stuck past step_timeout_s (wedged
+        collective, hung kernel). One-way: a hung engine cannot recover
+        in-process — the pod restart is the recovery path."""
+        started = self._step_started
+        if (self.healthy and started is not None
+                and time.monotonic() - started > self.step_timeout_s):
+            self.healthy = False
+            self.last_error = (
+                f"engine step exceeded {self.step_timeout_s:.0f}s "
+                "(hung collective or kernel)"
+            )
+        return self.healthy
+
     def embed(self, prompt_token_ids, pooling: str = "last"):
         """One pooled-embedding forward under the engine lock (runs
         between engine steps, like the PD sender)."""
@@ -275,12 +299,15 @@ class ServingEngine:
             with self._lock:
                 has_work = self.engine.has_unfinished()
             if not has_work:
+                self._step_started = None
                 self._work.clear()
                 self._work.wait(timeout=0.25)
                 continue
             try:
+                self._step_started = time.monotonic()
                 with self._lock:
                     outputs = self.engine.step()
+                self._step_started = None
             except Exception as e:  # engine fault: fail requests, go unhealthy
                 self.healthy = False
                 self.last_error = repr(e)

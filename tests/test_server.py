@@ -720,3 +720,28 @@ def test_embeddings_endpoint(app):
             assert r4.status_code == 400
 
     asyncio.run(run())
+
+
+def test_health_watchdog_detects_hung_step(app, serving):
+    """A step stuck past step_timeout_s flips /health to 503 (k8s
+    liveness -> pod restart is the recovery path)."""
+    import time as _time
+
+    async def run():
+        async with _client(app) as c:
+            assert (await c.get("/health")).status_code == 200
+            old = serving.step_timeout_s
+            serving.step_timeout_s = 0.01
+            serving._step_started = _time.monotonic() - 1.0  # fake a hang
+            try:
+                r = await c.get("/health")
+                assert r.status_code == 503
+                assert "exceeded" in r.json()["error"]
+            finally:
+                # one-way latch: restore for other tests in this module
+                serving.step_timeout_s = old
+                serving.healthy = True
+                serving.last_error = ""
+                serving._step_started = None
+
+    asyncio.run(run())
