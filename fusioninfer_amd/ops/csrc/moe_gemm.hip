@@ -523,19 +523,6 @@ void launch_moe_gemm(u16* out, const u16* a, const u16* b,
       return !(e && e[0] == '0');
     }();
     if (gate_up) {
-      // FI_MOE_GU_N4=1: NITER=4 experiment — halves per-m-tile A
-      // stagings (BN 64 -> 128) at double the accumulator registers
-      static const bool gu_n4 = [] {
-        const char* e = getenv("FI_MOE_GU_N4");
-        return e && e[0] == '1';
-      }();
-      if (gu_n4 && stage_a && N % 128 == 0) {
-        const dim3 grid4(max_mtiles, N / 128);
-        hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 4, true, true>), grid4,
-                           dim3(256), 0, stream, out, a, b, sorted_ids,
-                           expert_ids, n_valid, K, N);
-        return;
-      }
       const dim3 grid(max_mtiles, N / 64);
       if (stage_a)
         hipLaunchKernelGGL((moe_gemm_kernel<2, 2, 4, 2, true, true>), grid,
