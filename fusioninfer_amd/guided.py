@@ -697,6 +697,278 @@ class GuidedMaskCache:
         return m
 
 
+# ============================================================ GBNF grammar
+class _GbnfParser:
+    """GBNF (llama.cpp grammar format) parser. Produces
+    {rule: [alt, ...]} where alt is a tuple of symbols and a symbol is
+    ("t", ranges, negated) — a char-class terminal with ranges a tuple
+    of (lo, hi) ordinals — or ("r", name). Groups and repetition
+    suffixes desugar into synthetic right-recursive rules (so * + ?
+    never introduce left recursion)."""
+
+    def __init__(self, text: str):
+        self.s = text
+        self.i = 0
+        self.rules: Dict[str, list] = {}
+        self._gen = 0
+
+    def err(self, msg: str):
+        line = self.s.count("\n", 0, self.i) + 1
+        raise ValueError(f"GBNF parse error (line {line}): {msg}")
+
+    def ws(self):
+        while self.i < len(self.s):
+            c = self.s[self.i]
+            if c == "#":  # comment to end of line
+                while self.i < len(self.s) and self.s[self.i] != "\n":
+                    self.i += 1
+            elif c in " \t\r\n":
+                self.i += 1
+            else:
+                return
+
+    def _at_rule_start(self) -> bool:
+        j = self.i
+        while j < len(self.s) and (self.s[j].isalnum()
+                                   or self.s[j] in "_-"):
+            j += 1
+        if j == self.i:
+            return False
+        while j < len(self.s) and self.s[j] in " \t":
+            j += 1
+        return self.s[j:j + 3] == "::=" or self.s[j:j + 1] == ":"
+
+    def ident(self) -> str:
+        j = self.i
+        while j < len(self.s) and (self.s[j].isalnum() or self.s[j] in "_-"):
+            j += 1
+        if j == self.i:
+            self.err("expected rule name")
+        name, self.i = self.s[self.i:j], j
+        return name
+
+    def parse(self) -> Dict[str, list]:
+        while True:
+            self.ws()
+            if self.i >= len(self.s):
+                break
+            name = self.ident()
+            self.ws()
+            if self.s.startswith("::=", self.i):
+                self.i += 3
+            elif self.s.startswith(":", self.i):
+                self.i += 1
+            else:
+                self.err(f"expected '::=' after {name!r}")
+            self.rules[name] = self.alternates()
+        return self.rules
+
+    def fresh(self, alts) -> tuple:
+        self._gen += 1
+        name = f"%g{self._gen}"
+        self.rules[name] = alts
+        return ("r", name)
+
+    def alternates(self) -> list:
+        alts = [self.sequence()]
+        while True:
+            self.ws()
+            if self.i < len(self.s) and self.s[self.i] == "|":
+                self.i += 1
+                alts.append(self.sequence())
+            else:
+                return alts
+
+    def sequence(self) -> tuple:
+        syms: list = []
+        while True:
+            self.ws()
+            if self.i >= len(self.s):
+                return tuple(syms)
+            c = self.s[self.i]
+            if c in "|)" or self._at_rule_start():
+                return tuple(syms)
+            item = self.item()  # list of symbols
+            self.ws()
+            rep = self.s[self.i] if self.i < len(self.s) else ""
+            lo = hi = None
+            if rep and rep in "*+?":
+                self.i += 1
+                lo, hi = {"*": (0, None), "+": (1, None),
+                          "?": (0, 1)}[rep]
+            elif rep == "{":
+                j = self.s.index("}", self.i)
+                body = self.s[self.i + 1:j]
+                self.i = j + 1
+                parts = body.split(",")
+                lo = int(parts[0])
+                hi = (lo if len(parts) == 1 else
+                      (None if parts[1].strip() == "" else int(parts[1])))
+                if hi is not None and (hi < lo or hi > 64):
+                    self.err("bad {m,n} bounds")
+            if lo is None:
+                syms.extend(item)
+                continue
+            unit = (item[0] if len(item) == 1
+                    else self.fresh([tuple(item)]))
+            out = [unit] * lo
+            if hi is None:  # X{lo,} -> lo copies + X*
+                star = self.fresh([])
+                self.rules[star[1]] = [(unit, star), ()]
+                out.append(star)
+            else:
+                opt = self.fresh([(unit,), ()])
+                out.extend([opt] * (hi - lo))
+            syms.extend(out)
+
+    def item(self) -> list:
+        c = self.s[self.i]
+        if c == '"':
+            return [("t", ((o, o),), False) for o in self.literal()]
+        if c == "[":
+            return [self.char_class()]
+        if c == "(":
+            self.i += 1
+            alts = self.alternates()
+            self.ws()
+            if self.i >= len(self.s) or self.s[self.i] != ")":
+                self.err("expected ')'")
+            self.i += 1
+            return [self.fresh(alts)]
+        if c.isalnum() or c in "_-":
+            return [("r", self.ident())]
+        self.err(f"unexpected {c!r}")
+
+    def _escape(self) -> int:
+        c = self.s[self.i]
+        self.i += 1
+        if c != "\\":
+            return ord(c)
+        e = self.s[self.i]
+        self.i += 1
+        if e == "x":
+            v = int(self.s[self.i:self.i + 2], 16)
+            self.i += 2
+            return v
+        if e == "u":
+            v = int(self.s[self.i:self.i + 4], 16)
+            self.i += 4
+            return v
+        return ord({"n": "\n", "t": "\t", "r": "\r"}.get(e, e))
+
+    def literal(self) -> list:
+        self.i += 1  # opening quote
+        out = []
+        while self.i < len(self.s) and self.s[self.i] != '"':
+            out.append(self._escape())
+        if self.i >= len(self.s):
+            self.err("unterminated string")
+        self.i += 1
+        if not out:
+            self.err("empty literal")
+        return out
+
+    def char_class(self):
+        self.i += 1  # [
+        neg = self.s[self.i] == "^"
+        if neg:
+            self.i += 1
+        ranges = []
+        while self.i < len(self.s) and self.s[self.i] != "]":
+            lo = self._escape()
+            if (self.s[self.i] == "-" and self.i + 1 < len(self.s)
+                    and self.s[self.i + 1] != "]"):
+                self.i += 1
+                hi = self._escape()
+            else:
+                hi = lo
+            ranges.append((lo, hi))
+        if self.i >= len(self.s):
+            self.err("unterminated char class")
+        self.i += 1
+        if not ranges:
+            self.err("empty char class")
+        return ("t", tuple(ranges), neg)
+
+
+class GbnfGrammar:
+    """GBNF (llama.cpp grammar format) engine — vLLM guided_grammar
+    parity. State = frozenset of nondeterministic parse stacks
+    (llama.cpp's scheme): a stack is a tuple of frames (alt, pos) with
+    the top frame last, normalized so the top's current symbol is a
+    terminal; the empty stack () means the root completed. Left
+    recursion is rejected (expansion-depth bound), the same limitation
+    llama.cpp documents."""
+
+    _MAX_STACKS = 4096
+
+    def __init__(self, text: str, root: str = "root"):
+        self.rules = _GbnfParser(text).parse()
+        if root not in self.rules:
+            raise ValueError(f"GBNF grammar has no {root!r} rule")
+        for alts in self.rules.values():
+            for alt in alts:
+                for sym in alt:
+                    if sym[0] == "r" and sym[1] not in self.rules:
+                        raise ValueError(f"undefined rule {sym[1]!r}")
+        out: set = set()
+        self._norm((((("r", root),), 0),), out)
+        self._initial = frozenset(out)
+
+    def _norm(self, stack, out, depth=0):
+        if depth > 256:
+            raise ValueError(
+                "GBNF expansion too deep (left recursion is unsupported)"
+            )
+        while True:
+            if not stack:
+                out.add(())
+                return
+            alt, pos = stack[-1]
+            if pos >= len(alt):
+                stack = stack[:-1]
+                if stack:
+                    palt, ppos = stack[-1]
+                    stack = stack[:-1] + ((palt, ppos + 1),)
+                continue
+            sym = alt[pos]
+            if sym[0] == "t":
+                out.add(stack)
+                return
+            for a in self.rules[sym[1]]:
+                self._norm(stack + ((a, 0),), out, depth + 1)
+            return
+
+    def initial(self):
+        return self._initial
+
+    def step(self, state, ch):
+        o = ord(ch)
+        nxt: set = set()
+        for stack in state:
+            if not stack:
+                continue  # completed root: nothing may follow
+            alt, pos = stack[-1]
+            _, ranges, neg = alt[pos]
+            hit = any(lo <= o <= hi for lo, hi in ranges)
+            if hit != neg:
+                self._norm(stack[:-1] + ((alt, pos + 1),), nxt)
+        if not nxt:
+            return None
+        if len(nxt) > self._MAX_STACKS:
+            raise ValueError("GBNF state explosion")
+        return frozenset(nxt)
+
+    def is_complete(self, state) -> bool:
+        return () in state
+
+    def can_extend(self, state) -> bool:
+        return any(stack for stack in state)
+
+    def signature(self, state):
+        return state
+
+
 class GuidedDecoder:
     """Per-request grammar cursor the engine consults at sampling time.
 
@@ -762,4 +1034,9 @@ def build_guided(kind: str, spec, vocab: Vocabulary) -> GuidedDecoder:
             # unsupported schema construct: enforce well-formed JSON
             return build_guided("json_object", vocab=vocab, spec=None)
         return build_guided("regex", pattern, vocab)
+    if kind == "grammar":
+        key = ("grammar", spec, id(vocab))
+        if key not in _CACHES:
+            _CACHES[key] = GuidedMaskCache(GbnfGrammar(spec), vocab)
+        return GuidedDecoder(_CACHES[key])
     raise ValueError(f"unknown guided decoding kind {kind!r}")

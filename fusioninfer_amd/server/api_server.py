@@ -208,6 +208,9 @@ def build_app(serving: ServingEngine, model_name: str,
         if body.get("guided_json"):
             return build_guided("json_schema", body["guided_json"],
                                 _get_vocab())
+        if body.get("guided_grammar"):
+            return build_guided("grammar", str(body["guided_grammar"]),
+                                _get_vocab())
         rf = body.get("response_format")
         if isinstance(rf, dict):
             t = rf.get("type")

@@ -394,3 +394,83 @@ def test_schema_bounds_refs_and_allof():
         schema_to_regex({"$ref": "#/$defs/missing"})
     with _p.raises(ValueError):
         schema_to_regex({"$ref": "http://x/schema.json"})
+
+
+# ------------------------------------------------------------ GBNF grammar
+def test_gbnf_basics():
+    from fusioninfer_amd.guided import GbnfGrammar
+
+    g = GbnfGrammar('root ::= "yes" | "no" | "maybe"')
+    assert accepts(g, "yes") and accepts(g, "no") and accepts(g, "maybe")
+    assert not accepts(g, "ye") and viable(g, "ye")
+    assert not accepts(g, "yesno")
+    assert not viable(g, "z")
+
+
+def test_gbnf_repetition_classes_groups():
+    from fusioninfer_amd.guided import GbnfGrammar
+
+    g = GbnfGrammar('root ::= [a-z]+ ("-" [0-9]{2,3})?')
+    assert accepts(g, "abc")
+    assert accepts(g, "abc-42") and accepts(g, "abc-123")
+    assert not accepts(g, "abc-1")        # {2,3} lower bound
+    assert not accepts(g, "abc-1234")     # upper bound
+    assert not accepts(g, "ABC")
+    g2 = GbnfGrammar('root ::= ("ab")* "!"')
+    assert accepts(g2, "!") and accepts(g2, "abab!")
+    assert not accepts(g2, "aba!")
+    g3 = GbnfGrammar('root ::= [^x]+')
+    assert accepts(g3, "abc") and not viable(g3, "x")
+
+
+def test_gbnf_recursive_rules():
+    """Nested/recursive rules — the llama.cpp arithmetic-expression
+    shape (right recursion + nesting through parens)."""
+    from fusioninfer_amd.guided import GbnfGrammar
+
+    g = GbnfGrammar('''
+# expression grammar
+root  ::= expr
+expr  ::= term (("+" | "-") term)*
+term  ::= num | "(" expr ")"
+num   ::= [0-9]+
+''')
+    assert accepts(g, "1")
+    assert accepts(g, "1+2-3")
+    assert accepts(g, "(1+2)-(3+(4-5))")
+    assert not accepts(g, "1+")
+    assert viable(g, "((")
+    assert not viable(g, ")")
+
+
+def test_gbnf_escapes_and_errors():
+    from fusioninfer_amd.guided import GbnfGrammar
+
+    g = GbnfGrammar(r'root ::= "a\nb" | "\x41"')
+    assert accepts(g, "a\nb") and accepts(g, "A")
+    with pytest.raises(ValueError, match="no 'root'"):
+        GbnfGrammar('start ::= "x"')
+    with pytest.raises(ValueError, match="undefined rule"):
+        GbnfGrammar('root ::= missing')
+    with pytest.raises(ValueError, match="left recursion"):
+        GbnfGrammar('root ::= root "x" | "y"')
+
+
+def test_gbnf_engine_masked_generation():
+    """End-to-end: guided_grammar constrains engine output (byte
+    vocab)."""
+    from fusioninfer_amd.guided import GbnfGrammar, GuidedMaskCache
+
+    vocab = byte_vocab()
+    cache = GuidedMaskCache(
+        GbnfGrammar('root ::= ("ab" | "cd")+'), vocab
+    )
+    g = cache.grammar
+    st = g.initial()
+    m = cache.mask(st, "cpu")
+    allowed = {vocab.strings[i] for i in torch.nonzero(m).flatten().tolist()}
+    assert allowed == {"a", "c"}
+    st = g.step(st, "a")
+    m = cache.mask(st, "cpu")
+    allowed = {vocab.strings[i] for i in torch.nonzero(m).flatten().tolist()}
+    assert allowed == {"b"}

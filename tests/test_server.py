@@ -690,6 +690,32 @@ def test_finish_reason_length_vs_stop(app):
     asyncio.run(run())
 
 
+def test_guided_grammar(app):
+    """vLLM guided_grammar (GBNF): output constrained to the grammar."""
+
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "pick:", "max_tokens": 16,
+                      "temperature": 0.0,
+                      "guided_grammar":
+                          'root ::= "left" | "right" | "straight"'},
+            )
+            assert r.status_code == 200
+            assert r.json()["choices"][0]["text"] in {
+                "left", "right", "straight"}
+
+            r = await c.post(  # malformed grammar -> 400
+                "/v1/completions",
+                json={"prompt": "x", "max_tokens": 4,
+                      "guided_grammar": 'root ::= root "x"'},
+            )
+            assert r.status_code == 400
+
+    asyncio.run(run())
+
+
 def test_embeddings_endpoint(app):
     """OpenAI /v1/embeddings: unit-norm pooled vectors, deterministic
     per input, distinct across inputs; token-id input and length
