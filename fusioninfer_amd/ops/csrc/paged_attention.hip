@@ -43,7 +43,21 @@ constexpr float kNegInf = -1e30f;
 // slice only. G=8 runs HS=2 so the per-wave state is the G=4 footprint —
 // the monolithic G=8 variant spilled ~500 B/lane (W=8) or sat at
 // occupancy 1 (W=4).
-template <int D, int G, int kNWaves, bool FP8, int HS = 1>
+typedef __attribute__((ext_vector_type(8))) short short8;
+typedef __attribute__((ext_vector_type(4))) float floatx4;
+
+// MFA (bf16 only): phase A runs QK^T on the MATRIX CORES instead of
+// per-lane FMAs + 6-7-deep shfl_xor reduction chains (PMC showed the
+// old phase A WAIT-dominant on those serial permlanes). One
+// mfma_f32_16x16x32_bf16 per 32-dim slice: A = Q fragment with the G2
+// head rows REPLICATED over all 16 rows (so every 16-lane col group
+// holds a full copy of the scores and the softmax reductions are
+// 4-deep within a group, wave-uniform by construction — alpha needs no
+// broadcast), B = K fragment (lane reads K[tok=l&15][(l>>4)*8..+8]:
+// same bytes as the scalar path, different pattern, block stays
+// L2-resident). The K-dim reduction happens INSIDE the MFMA.
+template <int D, int G, int kNWaves, bool FP8, int HS = 1,
+          bool MFA = false>
 __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     u16* __restrict__ out,            // [S, Hq, D] (written when 1 partition)
     float* __restrict__ ml_ws,        // [S, Hq, P, 2] (multi-partition)
@@ -55,6 +69,7 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     const int* __restrict__ seq_lens,      // [S]
     const int64_t q_stride, const int max_blocks, const int num_kv_heads,
     const float scale) {
+  static_assert(!(MFA && FP8), "MFA phase A is bf16-only (for now)");
   using CT = typename std::conditional<FP8, unsigned char, u16>::type;
   const CT* k_cache = static_cast<const CT*>(k_cache_p);
   const CT* v_cache = static_cast<const CT*>(v_cache_p);
@@ -92,14 +107,33 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     return;
   }
 
-  __shared__ float q_lds[G][D];
+  __shared__ float q_lds[MFA ? 1 : G][MFA ? 1 : D];
   __shared__ float p_lds[kNWaves][kBlockSz][G2];
   __shared__ float merge_m[kNWaves][G2];
   __shared__ float merge_l[kNWaves][G2];
   __shared__ float merge_acc[kNWaves][G2][D];
 
-  // stage q (G heads) into LDS as fp32 (pre-scaled)
-  {
+  // MFA: per-wave replicated-row Q fragments, built once (Q is fixed
+  // for the whole kernel). Row i of A carries head i % G2, pre-scaled
+  // by scale*log2e and re-rounded to bf16 (same trick as the prefill
+  // kernel's pre-scaled Q).
+  short8 av_q[MFA ? D / 32 : 1];
+  if constexpr (MFA) {
+    const u16* q_row = q + seq * q_stride +
+                       static_cast<int64_t>(kv_head) * G * D;
+    const int head = hsplit * G2 + ((lane & 15) % G2);
+#pragma unroll
+    for (int kk = 0; kk < D / 32; ++kk) {
+      const int d0 = kk * 32 + (lane >> 4) * 8;
+      u16 tmp[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        tmp[e] = f32_to_bf16(bf16_to_f32(q_row[head * D + d0 + e]) *
+                             scale * 1.4426950408889634f);
+      av_q[kk] = *reinterpret_cast<short8*>(tmp);
+    }
+  } else {
+    // stage q (G heads) into LDS as fp32 (pre-scaled)
     const u16* q_row = q + seq * q_stride +
                        static_cast<int64_t>(kv_head) * G * D;
     for (int e = tid; e < G * D; e += kNWaves * kWaveSize)
@@ -107,8 +141,8 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       // uses bare v_exp_f32 (exp2) with no argument multiply
       q_lds[e / D][e % D] =
           bf16_to_f32(q_row[e]) * scale * 1.4426950408889634f;
+    __syncthreads();
   }
-  __syncthreads();
 
   // online-softmax state, wave-uniform (every lane holds the same copy)
   float m[G2], l[G2], acc[G2][2];
@@ -133,10 +167,19 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   };
   uint4 kraw[KQ4];
   auto load_k = [&](int64_t kv_base, uint4* dst) {
-    const CT* k_row = k_cache + kv_base + tok * D + quad * DPQ;
+    if constexpr (MFA) {
+      // B-fragment pattern: lane supplies K[tok = l&15][(l>>4)*8 + e]
+      const CT* k_row = k_cache + kv_base + (lane & 15) * D;
 #pragma unroll
-    for (int j8 = 0; j8 < KQ4; ++j8)
-      dst[j8] = reinterpret_cast<const uint4*>(k_row)[j8];
+      for (int kk = 0; kk < KQ4; ++kk)
+        dst[kk] = *reinterpret_cast<const uint4*>(
+            k_row + kk * 32 + (lane >> 4) * 8);
+    } else {
+      const CT* k_row = k_cache + kv_base + tok * D + quad * DPQ;
+#pragma unroll
+      for (int j8 = 0; j8 < KQ4; ++j8)
+        dst[j8] = reinterpret_cast<const uint4*>(k_row)[j8];
+    }
   };
   int64_t kv_base = 0;
   if (chunk_lo + cwave < chunk_hi) {
@@ -156,7 +199,17 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     float s[G2];
 #pragma unroll
     for (int g = 0; g < G2; ++g) s[g] = 0.f;
-    {
+    if constexpr (MFA) {
+      // QK^T on the matrix cores: K-dim reduction inside the MFMA, no
+      // cross-lane score reduction needed. c[row=g][col=lane&15].
+      floatx4 c = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < KQ4; ++kk)
+        c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            av_q[kk], __builtin_bit_cast(short8, kraw[kk]), c, 0, 0, 0);
+#pragma unroll
+      for (int g = 0; g < G2; ++g) s[g] = c[g];
+    } else {
 #pragma unroll
       for (int j8 = 0; j8 < KQ4; ++j8) {
         float kf8[16 / sizeof(CT)];
@@ -206,38 +259,64 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     // ranges) above the softmax — cross-phase overlap doubled VGPRs
     // (G2 is always <= 4: G=8 is head-split)
     __builtin_amdgcn_sched_barrier(0);
-    // reduce over the 4 dim-quarters (lanes 4t..4t+3)
+    if constexpr (MFA) {
+      // this lane's score column is token l&15 (replicated per group)
+      const int col_pos = chunk * kBlockSz + (lane & 15);
 #pragma unroll
-    for (int g = 0; g < G2; ++g) {
-      s[g] += __shfl_xor(s[g], 1, 64);
-      s[g] += __shfl_xor(s[g], 2, 64);
-      if (token_pos >= ctx) s[g] = kNegInf;
+      for (int g = 0; g < G2; ++g)
+        if (col_pos >= ctx) s[g] = kNegInf;
+    } else {
+      // reduce over the 4 dim-quarters (lanes 4t..4t+3)
+#pragma unroll
+      for (int g = 0; g < G2; ++g) {
+        s[g] += __shfl_xor(s[g], 1, 64);
+        s[g] += __shfl_xor(s[g], 2, 64);
+        if (token_pos >= ctx) s[g] = kNegInf;
+      }
     }
 
-    // chunk max over tokens (xor 4..32 spans the 16 token groups)
+    // chunk max over tokens (MFA: 4-deep within the replicated 16-lane
+    // col group; scalar path: xor 4..32 spans the 16 token groups)
     float alpha[G2];
 #pragma unroll
     for (int g = 0; g < G2; ++g) {
       float cm = s[g];
+      if constexpr (MFA) {
 #pragma unroll
-      for (int off = 4; off < 64; off <<= 1)
-        cm = fmaxf(cm, __shfl_xor(cm, off, 64));
+        for (int off = 1; off < 16; off <<= 1)
+          cm = fmaxf(cm, __shfl_xor(cm, off, 64));
+      } else {
+#pragma unroll
+        for (int off = 4; off < 64; off <<= 1)
+          cm = fmaxf(cm, __shfl_xor(cm, off, 64));
+      }
       const float m_new = fmaxf(m[g], cm);
       alpha[g] = __builtin_amdgcn_exp2f(m[g] - m_new);
       if (m[g] <= kNegInf && m_new <= kNegInf) alpha[g] = 0.f;
       m[g] = m_new;
     }
 
-    // probs + row-sum; every token is replicated on 4 lanes -> scale by 1/4
+    // probs + row-sum
 #pragma unroll
     for (int g = 0; g < G2; ++g) {
       float p = (s[g] <= kNegInf) ? 0.f
                 : __builtin_amdgcn_exp2f(s[g] - m[g]);
-      if (quad == 0) p_lds[wave][tok][g] = p;
-      float psum = p;
+      if constexpr (MFA) {
+        if (lane < 16) p_lds[wave][lane][g] = p;
+        float psum = p;
 #pragma unroll
-      for (int off = 1; off < 64; off <<= 1) psum += __shfl_xor(psum, off, 64);
-      l[g] = l[g] * alpha[g] + psum * 0.25f;
+        for (int off = 1; off < 16; off <<= 1)
+          psum += __shfl_xor(psum, off, 64);
+        l[g] = l[g] * alpha[g] + psum;  // each token counted once
+      } else {
+        // every token is replicated on 4 lanes -> scale by 1/4
+        if (quad == 0) p_lds[wave][tok][g] = p;
+        float psum = p;
+#pragma unroll
+        for (int off = 1; off < 64; off <<= 1)
+          psum += __shfl_xor(psum, off, 64);
+        l[g] = l[g] * alpha[g] + psum * 0.25f;
+      }
     }
 
     __builtin_amdgcn_sched_barrier(0);
@@ -370,11 +449,24 @@ void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
   const bool wide = num_seqs * num_kv_heads * num_parts < 256;
   const int nwaves = wide ? 8 : 4;
   dim3 grid(num_seqs, num_kv_heads, num_parts), block(nwaves * kWaveSize);
+  // FI_DECODE_MFMA=0 falls back to the scalar-FMA phase A (bf16 only;
+  // fp8 always runs the scalar path until the fp8 MFA variant lands)
+  static const bool mfa = [] {
+    const char* e = getenv("FI_DECODE_MFMA");
+    return !(e && e[0] == '0');
+  }();
 #define FI_LAUNCH_1(DD, GG, NW, F8, HSP)                                      \
-  hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG, NW, F8, HSP>), grid,   \
-                     block, 0, stream, out, ml_ws, acc_ws, q, k_cache,        \
-                     v_cache, block_tables, seq_lens, q_stride, max_blocks,   \
-                     num_kv_heads, scale)
+  if (!F8 && mfa) {                                                           \
+    hipLaunchKernelGGL(                                                       \
+        (paged_attn_decode_kernel<DD, GG, NW, false, HSP, true>), grid,       \
+        block, 0, stream, out, ml_ws, acc_ws, q, k_cache, v_cache,            \
+        block_tables, seq_lens, q_stride, max_blocks, num_kv_heads, scale);   \
+  } else {                                                                    \
+    hipLaunchKernelGGL((paged_attn_decode_kernel<DD, GG, NW, F8, HSP>), grid, \
+                       block, 0, stream, out, ml_ws, acc_ws, q, k_cache,      \
+                       v_cache, block_tables, seq_lens, q_stride, max_blocks, \
+                       num_kv_heads, scale);                                  \
+  }
 // Head-split policy, A/B'd on hardware (2026-09-14 kbench):
 // - G=8 always runs HS=2 (monolithic spilled; HS=4 measured a wash at
 //   full grids and -12% wide).
