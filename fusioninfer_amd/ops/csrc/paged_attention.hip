@@ -69,7 +69,6 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
     const int* __restrict__ seq_lens,      // [S]
     const int64_t q_stride, const int max_blocks, const int num_kv_heads,
     const float scale) {
-  static_assert(!(MFA && FP8), "MFA phase A is bf16-only (for now)");
   using CT = typename std::conditional<FP8, unsigned char, u16>::type;
   const CT* k_cache = static_cast<const CT*>(k_cache_p);
   const CT* v_cache = static_cast<const CT*>(v_cache_p);
@@ -117,8 +116,11 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   // for the whole kernel). Row i of A carries head i % G2, pre-scaled
   // by scale*log2e and re-rounded to bf16 (same trick as the prefill
   // kernel's pre-scaled Q).
-  short8 av_q[MFA ? D / 32 : 1];
-  if constexpr (MFA) {
+  short8 av_q[(MFA && !FP8) ? D / 32 : 1];
+  long long av8_q[(MFA && FP8) ? D / 32 : 1];
+  __shared__ float qsc_lds[(MFA && FP8) ? G : 1];
+  float qsc_r[G2];  // fp8 MFA: per-head dequant*softmax scalar
+  if constexpr (MFA && !FP8) {
     const u16* q_row = q + seq * q_stride +
                        static_cast<int64_t>(kv_head) * G * D;
     const int head = hsplit * G2 + ((lane & 15) % G2);
@@ -132,6 +134,36 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
                              scale * 1.4426950408889634f);
       av_q[kk] = *reinterpret_cast<short8*>(tmp);
     }
+  } else if constexpr (MFA && FP8) {
+    // Q quantized to e4m3 per HEAD; the dequant scale folds into the
+    // post-MFMA per-head scalar together with scale*log2e (the model
+    // already folded the cache's k_scale into `scale`)
+    const u16* q_row = q + seq * q_stride +
+                       static_cast<int64_t>(kv_head) * G * D;
+    if (tid < G) {
+      float mx = 1e-8f;
+      for (int d = 0; d < D; ++d)
+        mx = fmaxf(mx, fabsf(bf16_to_f32(q_row[tid * D + d])));
+      qsc_lds[tid] = mx / 448.f;
+    }
+    __syncthreads();
+    const int head = hsplit * G2 + ((lane & 15) % G2);
+    const float inv_qs = 1.f / qsc_lds[head];
+#pragma unroll
+    for (int kk = 0; kk < D / 32; ++kk) {
+      const int d0 = kk * 32 + (lane >> 4) * 8;
+      float f[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        f[e] = bf16_to_f32(q_row[head * D + d0 + e]) * inv_qs;
+      u32 w[2];
+      pack_fp8x8(f, w);
+      av8_q[kk] = static_cast<long long>(
+          (static_cast<unsigned long long>(w[1]) << 32) | w[0]);
+    }
+#pragma unroll
+    for (int g = 0; g < G2; ++g)
+      qsc_r[g] = qsc_lds[hsplit * G2 + g] * scale * 1.4426950408889634f;
   } else {
     // stage q (G heads) into LDS as fp32 (pre-scaled)
     const u16* q_row = q + seq * q_stride +
@@ -167,7 +199,15 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
   };
   uint4 kraw[KQ4];
   auto load_k = [&](int64_t kv_base, uint4* dst) {
-    if constexpr (MFA) {
+    if constexpr (MFA && FP8) {
+      // fp8 B-fragment: 8 e4m3 bytes per lane per 32-dim slice
+      const CT* k_row = k_cache + kv_base + (lane & 15) * D;
+#pragma unroll
+      for (int kk = 0; kk < D / 32; ++kk)
+        reinterpret_cast<unsigned long long*>(dst)[kk] =
+            *reinterpret_cast<const unsigned long long*>(
+                k_row + kk * 32 + (lane >> 4) * 8);
+    } else if constexpr (MFA) {
       // B-fragment pattern: lane supplies K[tok = l&15][(l>>4)*8 + e]
       const CT* k_row = k_cache + kv_base + (lane & 15) * D;
 #pragma unroll
@@ -203,12 +243,24 @@ __global__ __launch_bounds__(kNWaves * kWaveSize) void paged_attn_decode_kernel(
       // QK^T on the matrix cores: K-dim reduction inside the MFMA, no
       // cross-lane score reduction needed. c[row=g][col=lane&15].
       floatx4 c = {0.f, 0.f, 0.f, 0.f};
+      if constexpr (FP8) {
 #pragma unroll
-      for (int kk = 0; kk < KQ4; ++kk)
-        c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            av_q[kk], __builtin_bit_cast(short8, kraw[kk]), c, 0, 0, 0);
+        for (int kk = 0; kk < D / 32; ++kk)
+          c = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              av8_q[kk],
+              static_cast<long long>(
+                  reinterpret_cast<const unsigned long long*>(kraw)[kk]),
+              c, 0, 0, 0);
 #pragma unroll
-      for (int g = 0; g < G2; ++g) s[g] = c[g];
+        for (int g = 0; g < G2; ++g) s[g] = c[g] * qsc_r[g];
+      } else {
+#pragma unroll
+        for (int kk = 0; kk < KQ4; ++kk)
+          c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              av_q[kk], __builtin_bit_cast(short8, kraw[kk]), c, 0, 0, 0);
+#pragma unroll
+        for (int g = 0; g < G2; ++g) s[g] = c[g];
+      }
     } else {
 #pragma unroll
       for (int j8 = 0; j8 < KQ4; ++j8) {
@@ -449,16 +501,16 @@ void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
   const bool wide = num_seqs * num_kv_heads * num_parts < 256;
   const int nwaves = wide ? 8 : 4;
   dim3 grid(num_seqs, num_kv_heads, num_parts), block(nwaves * kWaveSize);
-  // FI_DECODE_MFMA=0 falls back to the scalar-FMA phase A (bf16 only;
-  // fp8 always runs the scalar path until the fp8 MFA variant lands)
+  // FI_DECODE_MFMA=0 falls back to the scalar-FMA phase A (both dtypes;
+  // the fp8 MFA variant quantizes Q to e4m3 per head)
   static const bool mfa = [] {
     const char* e = getenv("FI_DECODE_MFMA");
     return !(e && e[0] == '0');
   }();
 #define FI_LAUNCH_1(DD, GG, NW, F8, HSP)                                      \
-  if (!F8 && mfa) {                                                           \
+  if (mfa) {                                                                  \
     hipLaunchKernelGGL(                                                       \
-        (paged_attn_decode_kernel<DD, GG, NW, false, HSP, true>), grid,       \
+        (paged_attn_decode_kernel<DD, GG, NW, F8, HSP, true>), grid,          \
         block, 0, stream, out, ml_ws, acc_ws, q, k_cache, v_cache,            \
         block_tables, seq_lens, q_stride, max_blocks, num_kv_heads, scale);   \
   } else {                                                                    \
