@@ -507,8 +507,11 @@ void launch_paged_attn_decode(u16* out, float* ml_ws, float* acc_ws,
     const char* e = getenv("FI_DECODE_MFMA");
     return !(e && e[0] == '0');
   }();
+// fp8 MFA dispatch is G=8-only: kbench showed +41-43% there (G2=4
+// after head-split) but -21% at G=4 fp8, where the scalar HS=2 path
+// with halved KV bytes was already latency-optimal.
 #define FI_LAUNCH_1(DD, GG, NW, F8, HSP)                                      \
-  if (mfa) {                                                                  \
+  if (mfa && (!F8 || GG == 8)) {                                              \
     hipLaunchKernelGGL(                                                       \
         (paged_attn_decode_kernel<DD, GG, NW, F8, HSP, true>), grid,          \
         block, 0, stream, out, ml_ws, acc_ws, q, k_cache, v_cache,            \
