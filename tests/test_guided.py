@@ -474,3 +474,29 @@ def test_gbnf_engine_masked_generation():
     m = cache.mask(st, "cpu")
     allowed = {vocab.strings[i] for i in torch.nonzero(m).flatten().tolist()}
     assert allowed == {"b"}
+
+
+def test_gbnf_edge_cases():
+    """Parser edges: multiline rules, nested groups with alternates,
+    escaped chars in classes, comments, {n} exact repetition."""
+    from fusioninfer_amd.guided import GbnfGrammar
+
+    g = GbnfGrammar('''
+root ::= kv ("," kv)*        # object-ish list
+kv   ::= key "=" val
+key  ::= [a-zA-Z_] [a-zA-Z0-9_]*
+val  ::= num | "'" [^']* "'"
+num  ::= "-"? [0-9]{1,3}
+''')
+    assert accepts(g, "a=1")
+    assert accepts(g, "key_1=-42,b='x y',c=999")
+    assert not accepts(g, "1a=2")          # key can't start with digit
+    assert not accepts(g, "a=1234")        # {1,3}
+    assert not accepts(g, "a=1,")          # trailing comma
+    g2 = GbnfGrammar(r'root ::= ("a" | ("b" "c")+) [\-\]x]{2}')
+    assert accepts(g2, "a-]")
+    assert accepts(g2, "bcbcxx")
+    assert not accepts(g2, "a-")           # exactly 2 from the class
+    g3 = GbnfGrammar('root ::= "a"{3}')
+    assert accepts(g3, "aaa")
+    assert not accepts(g3, "aa") and not accepts(g3, "aaaa")
