@@ -57,6 +57,11 @@ def parse_args(argv=None):
                    help='JSON: {"method": "ngram", '
                         '"num_speculative_tokens": 4, '
                         '"prompt_lookup_max": 4, "prompt_lookup_min": 2}')
+    p.add_argument("--api-key", default=None,
+                   help="bearer token required on every endpoint except "
+                        "/health and /metrics (vLLM --api-key)")
+    p.add_argument("--served-model-name", default=None,
+                   help="name reported by /v1/models (vLLM parity)")
     # multi-node rendezvous flags injected by the LWS wrapper
     p.add_argument("--nnodes", type=int, default=1)
     p.add_argument("--node-rank", type=int, default=0)
@@ -195,7 +200,8 @@ def _rank_main(local_rank: int, args, nproc: int):
     tokenizer = get_tokenizer(
         cfg.model.vocab_size, args.model_path, args.tokenizer
     )
-    app = build_app(serving, args.model, tokenizer=tokenizer)
+    app = build_app(serving, args.served_model_name or args.model,
+                    tokenizer=tokenizer, api_key=args.api_key)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
     serving.engine.stop_workers()
 

@@ -21,7 +21,7 @@ import asyncio
 import json
 import time
 import uuid
-from typing import Any, AsyncGenerator, Dict, List, Union
+from typing import Any, AsyncGenerator, Dict, List, Optional, Union
 
 from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
@@ -94,13 +94,26 @@ class StopStreamFilter:
 
 
 def build_app(serving: ServingEngine, model_name: str,
-              tokenizer=None) -> FastAPI:
+              tokenizer=None, api_key: Optional[str] = None) -> FastAPI:
     app = FastAPI(title="fusioninfer-amd")
     vocab = serving.engine.cfg.model.vocab_size
     if tokenizer is None:
         from fusioninfer_amd.tokenizer import ByteTokenizer
 
         tokenizer = ByteTokenizer(vocab)
+
+    if api_key:
+        # vLLM --api-key parity: bearer auth on everything except the
+        # probes/metrics the platform (kubelet, Prometheus, EPP) scrapes
+        @app.middleware("http")
+        async def _auth(request: Request, call_next):
+            if request.url.path not in ("/health", "/metrics"):
+                if request.headers.get(
+                        "Authorization") != f"Bearer {api_key}":
+                    return JSONResponse(
+                        {"error": {"message": "invalid API key",
+                                   "type": "authentication_error"}}, 401)
+            return await call_next(request)
 
     def _encode(prompt: Union[str, List[int]]) -> List[int]:
         if isinstance(prompt, list):

@@ -771,3 +771,28 @@ def test_health_watchdog_detects_hung_step(app, serving):
                 serving._step_started = None
 
     asyncio.run(run())
+
+
+def test_api_key_auth(serving):
+    """--api-key: bearer auth on API endpoints; /health and /metrics
+    stay open for the kubelet/Prometheus/EPP scrapers."""
+
+    async def run():
+        app2 = build_app(serving, "tiny-qwen3", api_key="sk-test")
+        async with _client(app2) as c:
+            assert (await c.get("/health")).status_code == 200
+            assert (await c.get("/metrics")).status_code == 200
+            r = await c.post("/v1/completions",
+                             json={"prompt": "x", "max_tokens": 1})
+            assert r.status_code == 401
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "x", "max_tokens": 1, "temperature": 0,
+                      "ignore_eos": True},
+                headers={"Authorization": "Bearer sk-test"})
+            assert r.status_code == 200
+            r = await c.get("/v1/models",
+                            headers={"Authorization": "Bearer wrong"})
+            assert r.status_code == 401
+
+    asyncio.run(run())
