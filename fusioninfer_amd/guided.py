@@ -935,8 +935,15 @@ class GbnfGrammar:
             if sym[0] == "t":
                 out.add(stack)
                 return
+            # tail-call elision: a ref in FINAL position drops its frame
+            # before expanding (the child's completion continues with the
+            # grandparent directly). Without this, right-recursive
+            # desugared repetitions (X* -> R := X R | eps) would grow the
+            # stack by one frame per consumed repetition and trip the
+            # depth guard after ~256 characters.
+            base = stack[:-1] if pos + 1 == len(alt) else stack
             for a in self.rules[sym[1]]:
-                self._norm(stack + ((a, 0),), out, depth + 1)
+                self._norm(base + ((a, 0),), out, depth + 1)
             return
 
     def initial(self):

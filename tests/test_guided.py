@@ -500,3 +500,25 @@ num  ::= "-"? [0-9]{1,3}
     g3 = GbnfGrammar('root ::= "a"{3}')
     assert accepts(g3, "aaa")
     assert not accepts(g3, "aa") and not accepts(g3, "aaaa")
+
+
+def test_gbnf_long_repetition_bounded_state():
+    """Tail-call elision: unbounded repetition must not grow the parse
+    stacks (X* desugars right-recursively; without the elision a ~256
+    char input tripped the expansion-depth guard)."""
+    from fusioninfer_amd.guided import GbnfGrammar
+
+    g = GbnfGrammar('root ::= [a-z]* "!"')
+    st = g.initial()
+    for _ in range(2000):
+        st = g.step(st, "a")
+        assert st is not None
+        assert max(len(s) for s in st) <= 4  # stacks stay shallow
+    st = g.step(st, "!")
+    assert g.is_complete(st)
+    g2 = GbnfGrammar('root ::= ("ab")+')
+    st = g2.initial()
+    for _ in range(500):
+        st = g2.step(st, "a")
+        st = g2.step(st, "b")
+    assert g2.is_complete(st)
