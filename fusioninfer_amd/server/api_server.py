@@ -159,6 +159,7 @@ def build_app(serving: ServingEngine, model_name: str,
             ("e2e_request_latency_seconds_count", "counter"),
             ("engine_step_seconds_sum", "counter"),
             ("engine_step_seconds_count", "counter"),
+            ("pipelined_decode_steps_total", "counter"),
             ("num_preemptions_total", "counter"),
             ("num_swap_outs_total", "counter"),
             ("spec_decode_num_draft_tokens_total", "counter"),

@@ -255,6 +255,7 @@ class ServingEngine:
             "e2e_request_latency_seconds_count": float(e.num_finished),
             "engine_step_seconds_sum": e.step_time_sum,
             "engine_step_seconds_count": float(e.num_steps),
+            "pipelined_decode_steps_total": float(e.num_async_steps),
             "num_preemptions_total": float(e.num_preemptions),
             "num_swap_outs_total": float(e.num_swap_outs),
             "spec_decode_num_draft_tokens_total": float(
