@@ -744,6 +744,9 @@ def build_app(serving: ServingEngine, model_name: str,
                                            "type": "invalid_request_error"}},
                                 400)
         prompt_ids = _encode(text)
+        tpt = body.get("truncate_prompt_tokens")
+        if tpt:
+            prompt_ids = prompt_ids[-int(tpt):]
         try:
             sampling = _sampling_from(body)
         except ValueError as e:
