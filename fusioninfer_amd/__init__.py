@@ -13,4 +13,4 @@ see SURVEY.md). This framework is both halves, MI355X-first:
   ``amd.com/gpu`` nodes.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"

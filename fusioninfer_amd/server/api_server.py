@@ -130,6 +130,12 @@ def build_app(serving: ServingEngine, model_name: str,
             )
         return {"status": "ok"}
 
+    @app.get("/version")
+    async def version():
+        from fusioninfer_amd import __version__
+
+        return {"version": __version__}
+
     @app.get("/v1/models")
     async def models():
         data = [
