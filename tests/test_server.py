@@ -796,3 +796,34 @@ def test_api_key_auth(serving):
             assert r.status_code == 401
 
     asyncio.run(run())
+
+
+def test_moe_model_serves_http():
+    """MoE model end-to-end through the HTTP surface (CPU bmm expert
+    path): completions + embeddings against tiny-qwen3-moe."""
+
+    async def run():
+        cfg = EngineConfig(
+            model=get_model_config("tiny-qwen3-moe"),
+            cache=CacheConfig(num_gpu_blocks=128),
+            scheduler=SchedulerConfig(
+                max_num_seqs=4, max_num_batched_tokens=256, max_model_len=128
+            ),
+        )
+        s = ServingEngine(cfg, device="cpu")
+        try:
+            app2 = build_app(s, "tiny-qwen3-moe")
+            async with _client(app2) as c:
+                r = await c.post(
+                    "/v1/completions",
+                    json={"prompt": [7, 8, 9] * 8, "max_tokens": 4,
+                          "temperature": 0, "ignore_eos": True},
+                )
+                assert r.status_code == 200
+                assert len(r.json()["choices"][0]["token_ids"]) == 4
+                r = await c.post("/v1/embeddings", json={"input": [3, 4, 5]})
+                assert r.status_code == 200
+        finally:
+            s.shutdown()
+
+    asyncio.run(run())
