@@ -107,7 +107,12 @@ def build_app(serving: ServingEngine, model_name: str,
         # probes/metrics the platform (kubelet, Prometheus, EPP) scrapes
         @app.middleware("http")
         async def _auth(request: Request, call_next):
-            if request.url.path not in ("/health", "/metrics"):
+            path = request.url.path
+            # exempt: platform scrapers (kubelet probes, Prometheus,
+            # EPP metrics) and the cluster-internal PD handshake the
+            # EPP issues without client credentials
+            if (path not in ("/health", "/metrics")
+                    and not path.startswith("/pd/")):
                 if request.headers.get(
                         "Authorization") != f"Bearer {api_key}":
                     return JSONResponse(
