@@ -334,7 +334,7 @@ def test_reshape_and_cache_fp8_inv_scale():
     )
 
 
-@pytest.mark.parametrize("group", [1, 4])
+@pytest.mark.parametrize("group", [1, 4, 8])
 @pytest.mark.parametrize("seq_lens", [[1], [1, 5, 16, 17, 255, 1023]])
 def test_paged_attention_decode_fp8kv(group, seq_lens):
     """Decode over an fp8 (e4m3, scale-1) cache: the kernel and the fp32
