@@ -362,7 +362,11 @@ def test_paged_attention_decode_fp8kv(group, seq_lens):
     expected = ref.paged_attention_decode(
         q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), lens.cpu(), scale
     )
-    assert_close_bf16(out.cpu(), expected, atol=3e-2, rtol=3e-2)
+    # G=8 dispatches the fp8 MFA variant, which quantizes Q to e4m3
+    # per head (the reference keeps fp32 Q) — one extra quantization
+    # step of score error
+    tol = 6e-2 if group == 8 else 3e-2
+    assert_close_bf16(out.cpu(), expected, atol=tol, rtol=tol)
 
 
 @pytest.mark.parametrize("head_dim", [64, 128])
