@@ -636,3 +636,27 @@ def test_async_decode_mixed_admission_breaks_chain():
     assert set(done) == {r1, r2}
     assert len(done[r2].output_token_ids) == 5
     assert len(done[r1].output_token_ids) == 30
+
+
+def test_async_decode_under_block_pressure():
+    """Block exhaustion mid-chain: the pipeline refuses continuation
+    (capacity check), drains, and the sync path preempts/recomputes —
+    all requests still complete with the right lengths."""
+    torch.manual_seed(0)
+    # tiny pool: 3 seqs x (prompt 32 + gen 40) over 16-token blocks
+    # cannot all stay resident
+    eng = make_engine(num_blocks=10, max_seqs=4, max_len=128)
+    assert eng._async_decode
+    ids = [eng.add_request([i + 1, i + 2, i + 3] * 6,
+                           SamplingParams(max_tokens=40))
+           for i in range(3)]
+    done = {}
+    steps = 0
+    while eng.has_unfinished() and steps < 500:
+        steps += 1
+        for out in eng.step():
+            if out.finished:
+                done[out.request_id] = out
+    assert set(done) == set(ids)
+    assert all(len(done[r].output_token_ids) == 40 for r in ids)
+    assert eng.num_preemptions > 0  # the pressure actually materialized
