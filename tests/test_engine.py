@@ -660,3 +660,56 @@ def test_async_decode_under_block_pressure():
     assert set(done) == set(ids)
     assert all(len(done[r].output_token_ids) == 40 for r in ids)
     assert eng.num_preemptions > 0  # the pressure actually materialized
+
+
+def test_decode_only_next_is_read_only():
+    """The pipeline's scheduler preview must not mutate state (no slot
+    appends, no admissions) — schedule() afterwards sees a clean view."""
+    torch.manual_seed(0)
+    eng = make_engine()
+    eng.add_request([1, 2, 3] * 6, SamplingParams(max_tokens=8))
+    eng.step()  # prefill
+    free0 = eng.block_manager.num_free()
+    blocks0 = {
+        s.seq_id: list(s.block_ids) for s in eng.scheduler.running
+    }
+    for _ in range(3):
+        eng.scheduler.decode_only_next()
+    assert eng.block_manager.num_free() == free0
+    assert {
+        s.seq_id: list(s.block_ids) for s in eng.scheduler.running
+    } == blocks0
+
+
+def test_engine_fault_marks_unhealthy():
+    """An exception inside engine.step flips the serving loop unhealthy
+    and fails outstanding streams (complements the hang watchdog)."""
+    from fusioninfer_amd.config import CacheConfig, EngineConfig, SchedulerConfig
+    from fusioninfer_amd.server.serving import ServingEngine
+
+    cfg = EngineConfig(
+        model=get_model_config("tiny-qwen3"),
+        cache=CacheConfig(num_gpu_blocks=64),
+        scheduler=SchedulerConfig(
+            max_num_seqs=2, max_num_batched_tokens=256, max_model_len=64
+        ),
+    )
+    s = ServingEngine(cfg, device="cpu")
+    try:
+        def boom():
+            raise RuntimeError("injected fault")
+
+        s.engine.step = boom
+        _, q = s.submit([5, 6, 7], SamplingParams(max_tokens=4))
+        tok, finished = q.get(timeout=10)
+        assert finished and tok is None  # stream failed, not hung
+        import time as _t
+
+        for _ in range(100):
+            if not s.healthy:
+                break
+            _t.sleep(0.02)
+        assert not s.healthy
+        assert "injected fault" in s.last_error
+    finally:
+        s.shutdown()
