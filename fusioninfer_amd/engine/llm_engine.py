@@ -641,7 +641,13 @@ class LLMEngine:
             new.append(int(tok))
             self.num_generated_tokens += 1
             if guided is not None:
-                guided.advance_token(int(tok))
+                try:
+                    guided.advance_token(int(tok))
+                except ValueError:
+                    # grammar blow-up: finish THIS request gracefully
+                    seq.finish_reason = seq.finish_reason or "stop"
+                    self._finish_seq(seq)
+                    break
             if seq.check_stop():
                 self._finish_seq(seq)
                 break
@@ -790,7 +796,12 @@ class LLMEngine:
                 )
             g = s.sampling.guided
             if g is not None:
-                allowed = g.allowed_mask(logits_f.device)
+                try:
+                    allowed = g.allowed_mask(logits_f.device)
+                except ValueError:
+                    # grammar blow-up (state-explosion cap, expansion
+                    # bound): fail THIS request, not the engine loop
+                    allowed = None
                 if allowed is None:
                     s.finish_reason = s.finish_reason or "stop"
                     self._finish_seq(s)
