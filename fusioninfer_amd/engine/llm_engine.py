@@ -789,11 +789,15 @@ class LLMEngine:
         keep = []
         for i, s in enumerate(sample_seqs):
             if s.sampling.logit_bias:
-                ids = list(s.sampling.logit_bias)
-                logits_f[i, ids] += torch.tensor(
-                    [s.sampling.logit_bias[t] for t in ids],
-                    dtype=logits_f.dtype, device=logits_f.device,
-                )
+                V = logits_f.shape[1]
+                # drop out-of-vocab ids: a bad request must not raise a
+                # device-side index fault in the engine loop
+                ids = [t for t in s.sampling.logit_bias if 0 <= t < V]
+                if ids:
+                    logits_f[i, ids] += torch.tensor(
+                        [s.sampling.logit_bias[t] for t in ids],
+                        dtype=logits_f.dtype, device=logits_f.device,
+                    )
             g = s.sampling.guided
             if g is not None:
                 try:

@@ -827,3 +827,23 @@ def test_moe_model_serves_http():
             s.shutdown()
 
     asyncio.run(run())
+
+
+def test_out_of_vocab_logit_bias_is_safe(app):
+    """logit_bias keys outside the vocab must not crash the engine loop
+    (device-side index fault); they are dropped."""
+
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": [5, 6, 7] * 4, "max_tokens": 3,
+                      "temperature": 0, "ignore_eos": True,
+                      "logit_bias": {"999999999": 50, "7": 10}},
+            )
+            assert r.status_code == 200
+            assert len(r.json()["choices"][0]["token_ids"]) == 3
+            # engine stayed healthy
+            assert (await c.get("/health")).status_code == 200
+
+    asyncio.run(run())
