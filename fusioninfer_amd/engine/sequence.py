@@ -44,7 +44,9 @@ class SamplingParams:
         self.repetition_penalty = repetition_penalty
         self.presence_penalty = presence_penalty
         self.frequency_penalty = frequency_penalty
-        self.seed = seed
+        # normalize into torch.Generator.manual_seed's accepted range —
+        # a huge JSON seed must not raise inside the engine loop
+        self.seed = None if seed is None else int(seed) % (2**63)
         self.logprobs = logprobs
         self.prompt_logprobs = prompt_logprobs
         self.ignore_eos = ignore_eos
