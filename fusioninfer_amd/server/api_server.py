@@ -250,8 +250,18 @@ def build_app(serving: ServingEngine, model_name: str,
         mt = int(body.get("max_tokens", 16))
         if mt < 1:
             raise ValueError("max_tokens must be >= 1")
-        if float(body.get("temperature", 1.0)) < 0:
-            raise ValueError("temperature must be >= 0")
+        import math as _math
+
+        temp = float(body.get("temperature", 1.0))
+        # NaN compares False everywhere: an explicit finite check keeps
+        # a JSON "NaN" literal from reaching torch.multinomial
+        if not _math.isfinite(temp) or temp < 0:
+            raise ValueError("temperature must be finite and >= 0")
+        for pk in ("presence_penalty", "frequency_penalty",
+                   "repetition_penalty"):
+            v = body.get(pk)
+            if v is not None and not _math.isfinite(float(v)):
+                raise ValueError(f"{pk} must be finite")
         tp_ = float(body.get("top_p", 1.0))
         if not 0.0 < tp_ <= 1.0:
             raise ValueError("top_p must be in (0, 1]")

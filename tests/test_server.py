@@ -847,3 +847,24 @@ def test_out_of_vocab_logit_bias_is_safe(app):
             assert (await c.get("/health")).status_code == 200
 
     asyncio.run(run())
+
+
+def test_nan_sampling_params_rejected(app):
+    """JSON NaN literals in sampling params must 400, not NaN the
+    engine's multinomial."""
+
+    async def run():
+        async with _client(app) as c:
+            for payload in (
+                '{"prompt": "x", "max_tokens": 2, "temperature": NaN}',
+                '{"prompt": "x", "max_tokens": 2, "temperature": 0.5, '
+                '"presence_penalty": NaN}',
+                '{"prompt": "x", "max_tokens": 2, "top_p": NaN}',
+            ):
+                r = await c.post(
+                    "/v1/completions", content=payload,
+                    headers={"Content-Type": "application/json"})
+                assert r.status_code == 400, payload
+            assert (await c.get("/health")).status_code == 200
+
+    asyncio.run(run())
