@@ -519,6 +519,10 @@ def build_app(serving: ServingEngine, model_name: str,
         )
 
         n_req = max(int(body.get("n", 1)), 1)
+        if n_req > 64:
+            return JSONResponse({"error": {"message": "n must be <= 64",
+                                           "type": "invalid_request_error"}},
+                                400)
         if body.get("stream") and n_req > 1:
             # interleaved multi-choice stream: sibling requests advance in
             # lockstep with the engine, so round-robin draining is fair
@@ -693,6 +697,10 @@ def build_app(serving: ServingEngine, model_name: str,
         # caching dedups the shared-prompt KV, so the n-1 extra prefills
         # recompute at most one block each
         n = max(int(body.get("n", 1)), 1)
+        if n > 64:
+            return JSONResponse({"error": {"message": "n must be <= 64",
+                                           "type": "invalid_request_error"}},
+                                400)
         extra = []
         for i in range(1, n):
             s_i = _sampling_from(body)
@@ -791,6 +799,10 @@ def build_app(serving: ServingEngine, model_name: str,
         cid = f"chatcmpl-{uuid.uuid4().hex[:16]}"
 
         n_req = max(int(body.get("n", 1)), 1)
+        if n_req > 64:
+            return JSONResponse({"error": {"message": "n must be <= 64",
+                                           "type": "invalid_request_error"}},
+                                400)
         if body.get("stream") and n_req > 1:
             extra = []
             for i in range(1, n_req):
@@ -936,6 +948,10 @@ def build_app(serving: ServingEngine, model_name: str,
         # OpenAI `n` (non-stream): extra choices as sibling requests —
         # prefix caching dedups the shared-prompt KV
         n = max(int(body.get("n", 1)), 1)
+        if n > 64:
+            return JSONResponse({"error": {"message": "n must be <= 64",
+                                           "type": "invalid_request_error"}},
+                                400)
         extra = []
         for i in range(1, n):
             s_i = _sampling_from(body)

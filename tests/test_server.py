@@ -868,3 +868,15 @@ def test_nan_sampling_params_rejected(app):
             assert (await c.get("/health")).status_code == 200
 
     asyncio.run(run())
+
+
+def test_excessive_n_rejected(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "x", "max_tokens": 1, "n": 100000},
+            )
+            assert r.status_code == 400
+
+    asyncio.run(run())
