@@ -121,9 +121,23 @@ def build_app(serving: ServingEngine, model_name: str,
             return await call_next(request)
 
     def _encode(prompt: Union[str, List[int]]) -> List[int]:
-        if isinstance(prompt, list):
-            return [int(t) for t in prompt]
-        return tokenizer.encode(prompt)
+        try:
+            if isinstance(prompt, list):
+                return [int(t) for t in prompt]
+            if not isinstance(prompt, str):
+                raise TypeError(f"prompt must be a string or token-id "
+                                f"list, got {type(prompt).__name__}")
+            return tokenizer.encode(prompt)
+        except (TypeError, ValueError, AttributeError) as e:
+            raise ValueError(f"malformed prompt: {e}") from e
+
+    @app.exception_handler(ValueError)
+    async def _bad_request(request: Request, exc: ValueError):
+        # malformed client input surfacing anywhere in a handler is a
+        # 400, never a 500 stack trace
+        return JSONResponse(
+            {"error": {"message": str(exc),
+                       "type": "invalid_request_error"}}, 400)
 
     @app.get("/health")
     async def health():

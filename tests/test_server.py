@@ -880,3 +880,20 @@ def test_excessive_n_rejected(app):
             assert r.status_code == 400
 
     asyncio.run(run())
+
+
+def test_malformed_prompt_is_400(app):
+    async def run():
+        async with _client(app) as c:
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": [["nested"]], "max_tokens": 1},
+            )
+            assert r.status_code == 400
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": {"not": "a prompt"}, "max_tokens": 1},
+            )
+            assert r.status_code == 400
+
+    asyncio.run(run())
