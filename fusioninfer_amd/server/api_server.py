@@ -782,7 +782,7 @@ def build_app(serving: ServingEngine, model_name: str,
             # vendored chat template (tokenizer_config.json) when present;
             # flat role-prefixed transcript otherwise
             text = tokenizer.apply_chat_template(messages)
-        except ValueError as e:
+        except (ValueError, TypeError, KeyError, AttributeError) as e:
             return JSONResponse({"error": {"message": str(e),
                                            "type": "invalid_request_error"}},
                                 400)

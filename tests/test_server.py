@@ -897,3 +897,23 @@ def test_malformed_prompt_is_400(app):
             assert r.status_code == 400
 
     asyncio.run(run())
+
+
+def test_malformed_chat_messages_400(app):
+    async def run():
+        async with _client(app) as c:
+            for messages in ("not-a-list", [{"role": "user"}, 42]):
+                r = await c.post(
+                    "/v1/chat/completions",
+                    json={"messages": messages, "max_tokens": 1},
+                )
+                assert r.status_code == 400, messages
+            # lenient cases must not 500 either
+            r = await c.post(
+                "/v1/chat/completions",
+                json={"messages": [{"content": 3.14}], "max_tokens": 1,
+                      "temperature": 0, "ignore_eos": True},
+            )
+            assert r.status_code in (200, 400)
+
+    asyncio.run(run())
