@@ -713,3 +713,12 @@ def test_engine_fault_marks_unhealthy():
         assert "injected fault" in s.last_error
     finally:
         s.shutdown()
+
+
+def test_async_decode_env_kill_switch(monkeypatch):
+    monkeypatch.setenv("FI_ASYNC_DECODE", "0")
+    eng = make_engine()
+    assert not eng._async_decode
+    out = eng.generate([[1, 2, 3] * 5], SamplingParams(max_tokens=5))[0]
+    assert len(out.output_token_ids) == 5
+    assert eng.num_async_steps == 0

@@ -522,3 +522,25 @@ def test_gbnf_long_repetition_bounded_state():
         st = g2.step(st, "a")
         st = g2.step(st, "b")
     assert g2.is_complete(st)
+
+
+def test_gbnf_mini_json_grammar():
+    """A practical GBNF grammar (mini-JSON: nested arrays/objects of
+    ints and short strings) — the shape users actually ship."""
+    from fusioninfer_amd.guided import GbnfGrammar
+
+    g = GbnfGrammar(r'''
+root   ::= value
+value  ::= object | array | number | string
+object ::= "{" (pair ("," pair)*)? "}"
+pair   ::= string ":" value
+array  ::= "[" (value ("," value)*)? "]"
+number ::= "-"? [0-9]+
+string ::= "\"" [a-z0-9_ ]* "\""
+''')
+    for ok in ('{}', '[]', '[1,2,[3,-4]]',
+               '{"a":1,"b":{"c":[{"d":"x y"},-7]}}'):
+        assert accepts(g, ok), ok
+    for bad in ('{', '[1,]', '{"a" 1}', '{"a":}', '[1 2]', '"A"'):
+        assert not accepts(g, bad), bad
+    assert viable(g, '{"deep":[[[[[')
