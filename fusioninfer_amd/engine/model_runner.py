@@ -7,6 +7,10 @@ overhead from the latency-critical decode loop (guide: "capture
 launch-bound inner loops in hipGraphs"). Prefill and mixed steps run
 eager; batches with live LoRA adapters also run eager. Payloads are plain
 host lists so the TP driver can broadcast them to worker ranks.
+
+Capability parity: the execution layer of the continuous-batching
+engine the reference delegates to its vLLM containers (SURVEY.md §2.3
+"Continuous-batching scheduler + paged KV cache").
 """
 
 from __future__ import annotations

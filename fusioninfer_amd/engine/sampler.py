@@ -1,4 +1,6 @@
-"""Token sampling: penalties, temperature, top-k/top-p, per-request seeds."""
+"""Token sampling: penalties, temperature, top-k/top-p/min-p, per-request
+seeds — the vLLM SamplingParams surface the reference's delegated engine
+exposes."""
 
 from __future__ import annotations
 
