@@ -544,3 +544,27 @@ string ::= "\"" [a-z0-9_ ]* "\""
     for bad in ('{', '[1,]', '{"a" 1}', '{"a":}', '[1 2]', '"A"'):
         assert not accepts(g, bad), bad
     assert viable(g, '{"deep":[[[[[')
+
+
+def test_gbnf_regex_cross_check():
+    """Oracle test: grammars expressible in both engines must agree on
+    random strings (acceptance AND viability)."""
+    import random
+
+    from fusioninfer_amd.guided import GbnfGrammar, RegexGrammar
+
+    pairs = [
+        ('root ::= "a" [b-d]* "e"?', "a[b-d]*e?"),
+        ('root ::= ("x" | "yz")+', "(x|yz)+"),
+        ('root ::= [0-9]{2,4} ("-" [a-c])?', "[0-9]{2,4}(-[a-c])?"),
+    ]
+    rng = random.Random(7)
+    alphabet = "abcdexyz0123456789-"
+    for gb_text, rx in pairs:
+        gb = GbnfGrammar(gb_text)
+        rg = RegexGrammar(rx)
+        for _ in range(300):
+            s = "".join(rng.choice(alphabet)
+                        for _ in range(rng.randrange(0, 8)))
+            assert accepts(gb, s) == accepts(rg, s), (gb_text, s)
+            assert viable(gb, s) == viable(rg, s), (gb_text, s)
