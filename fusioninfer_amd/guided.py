@@ -1017,6 +1017,16 @@ class GuidedDecoder:
 _CACHES: Dict[Tuple, GuidedMaskCache] = {}
 
 
+def _byteize(text: str) -> str:
+    """UTF-8 bytes of `text` as a latin-1 string (ASCII fixed-point).
+    Grammar specs are matched against BYTE-level token strings
+    (Vocabulary is built from decode_one_bytes), so non-ASCII literal
+    chars in user regex/choice/GBNF specs must expand to their UTF-8
+    byte sequence. Limitation: character-class RANGES with non-ASCII
+    endpoints are not byte-expandable and will mismatch."""
+    return text.encode("utf-8").decode("latin-1")
+
+
 def build_guided(kind: str, spec, vocab: Vocabulary) -> GuidedDecoder:
     """kind: 'json_object' | 'regex' | 'choice' | 'json_schema'.
     Raises ValueError on unsupported specs (API layer maps it to 400)."""
@@ -1027,6 +1037,7 @@ def build_guided(kind: str, spec, vocab: Vocabulary) -> GuidedDecoder:
                                            vocab)
         return GuidedDecoder(_CACHES[key])
     if kind == "regex":
+        spec = _byteize(spec)
         key = ("regex", spec, id(vocab))
         if key not in _CACHES:
             _CACHES[key] = GuidedMaskCache(RegexGrammar(spec), vocab)
@@ -1042,6 +1053,7 @@ def build_guided(kind: str, spec, vocab: Vocabulary) -> GuidedDecoder:
             return build_guided("json_object", vocab=vocab, spec=None)
         return build_guided("regex", pattern, vocab)
     if kind == "grammar":
+        spec = _byteize(spec)
         key = ("grammar", spec, id(vocab))
         if key not in _CACHES:
             _CACHES[key] = GuidedMaskCache(GbnfGrammar(spec), vocab)

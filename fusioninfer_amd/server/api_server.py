@@ -227,9 +227,11 @@ def build_app(serving: ServingEngine, model_name: str,
         if "v" not in _vocab_cache:
             from fusioninfer_amd.guided import Vocabulary
 
-            _vocab_cache["v"] = Vocabulary(
-                vocab, tokenizer.decode_one
-            )
+            # byte-level decode keeps multi-byte UTF-8 representable in
+            # grammar masks (decode_one collapses split chars to U+FFFD)
+            decode = getattr(tokenizer, "decode_one_bytes",
+                             tokenizer.decode_one)
+            _vocab_cache["v"] = Vocabulary(vocab, decode)
         return _vocab_cache["v"]
 
     def _guided_from(body: Dict[str, Any]):

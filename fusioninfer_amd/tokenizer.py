@@ -46,6 +46,27 @@ def default_chat_format(messages) -> str:
     return text
 
 
+_BYTE_DECODER: Optional[dict] = None
+
+
+def _byte_level_decoder() -> dict:
+    """char -> byte map of the byte-level BPE alphabet (the GPT-2 table
+    used by Qwen/Llama tokenizer.json ByteLevel pre-tokenizers)."""
+    global _BYTE_DECODER
+    if _BYTE_DECODER is None:
+        bs = (list(range(33, 127)) + list(range(161, 173))
+              + list(range(174, 256)))
+        cs = bs[:]
+        n = 0
+        for b in range(256):
+            if b not in bs:
+                bs.append(b)
+                cs.append(256 + n)
+                n += 1
+        _BYTE_DECODER = {chr(c): b for b, c in zip(bs, cs)}
+    return _BYTE_DECODER
+
+
 class ByteTokenizer:
     """Reversible byte-level fallback (no vocab files needed)."""
 
@@ -74,6 +95,14 @@ class ByteTokenizer:
         return bytes([max(token_id - 3, 0) & 0xFF]).decode(
             "utf-8", errors="replace"
         )
+
+    def decode_one_bytes(self, token_id: int) -> str:
+        """Per-token RAW byte as a latin-1 char (byte<->char bijection).
+        Guided-decoding masks step grammars at byte level with this, so
+        bytes >= 0x80 stay representable (decode_one turns them into
+        U+FFFD, which silently masked non-ASCII out of grammars —
+        round-1 advisor finding, second half)."""
+        return chr(max(token_id - 3, 0) & 0xFF)
 
 
 class HFTokenizer:
@@ -145,6 +174,27 @@ class HFTokenizer:
         if not 0 <= token_id < self._n:
             return ""
         return self._tok.decode([token_id], skip_special_tokens=False)
+
+    def decode_one_bytes(self, token_id: int) -> str:
+        """Per-token RAW BYTES as a latin-1 string: a byte-level BPE
+        token that carries part of a multi-byte UTF-8 character decodes
+        to U+FFFD through decode_one, which excluded every such token
+        from guided-decoding masks (no CJK/emoji under guided JSON).
+        Mapping token text back through the byte-level alphabet keeps
+        the exact bytes; grammars then run at byte level (the engine's
+        detokenizer reassembles characters on output)."""
+        if not 0 <= token_id < self._n:
+            return ""
+        tok = self._tok.id_to_token(token_id)
+        if tok is None:
+            return ""
+        inv = _byte_level_decoder()
+        try:
+            return bytes(inv[c] for c in tok).decode("latin-1")
+        except KeyError:
+            # non-byte-level token (e.g. an added special token): its
+            # literal text is the right grammar-visible form
+            return self.decode_one(token_id)
 
 
 class IncrementalDetokenizer:

@@ -200,3 +200,52 @@ def test_incremental_detokenizer_flush_mid_char():
     detok = IncrementalDetokenizer(t)
     assert "".join(detok.push(i) for i in ids) == ""
     assert "�" in detok.flush()
+
+
+def test_guided_masks_allow_multibyte_utf8(tok_file):
+    """Byte-level guided decoding (round-1 advisor, second half): BPE
+    tokens carrying PART of a multi-byte UTF-8 char decode to U+FFFD
+    through decode_one, which made grammar literals like "café"
+    unreachable. With decode_one_bytes + byteized specs the byte
+    tokens step the grammar exactly."""
+    from fusioninfer_amd.guided import Vocabulary, build_guided
+
+    t = HFTokenizer(tok_file, vocab_size=1024)
+    vocab = Vocabulary(1024, t.decode_one_bytes)
+    dec = build_guided("choice", ["café", "thé"], vocab)
+    ids = t.encode("café")
+    assert len(ids) == 5  # c a f + the two é bytes (no-merge tokenizer)
+    for tid in ids:
+        m = dec.allowed_mask("cpu")
+        assert bool(m[tid]), tid
+        dec.advance_token(tid)
+    assert dec.is_terminal()
+
+    # the OLD char-level vocabulary dead-ends after "caf": the é byte
+    # tokens decode to U+FFFD and never match the literal
+    vocab_old = Vocabulary(1024, t.decode_one)
+    dec_old = build_guided("choice", ["café", "thé"], vocab_old)
+    for tid in ids[:3]:
+        dec_old.advance_token(tid)
+    m = dec_old.allowed_mask("cpu")
+    assert m is None or not bool(m[ids[3]])
+
+
+def test_guided_json_multibyte_string_content(tok_file):
+    """json_object mode: CJK bytes are legal JSON-string content at the
+    byte level (three continuation bytes step the pushdown grammar)."""
+    from fusioninfer_amd.guided import (
+        GuidedMaskCache, JsonGrammar, Vocabulary,
+    )
+
+    t = HFTokenizer(tok_file, vocab_size=1024)
+    vocab = Vocabulary(1024, t.decode_one_bytes)
+    g = JsonGrammar(root_object=False)
+    cache = GuidedMaskCache(g, vocab)
+    st = g.initial()
+    for tid in t.encode('"中文"'):
+        m = cache.mask(st, "cpu")
+        assert bool(m[tid]), tid
+        for ch in vocab.strings[tid]:
+            st = g.step(st, ch)
+    assert g.is_complete(st)
