@@ -109,6 +109,17 @@ def test_server_with_hf_tokenizer(tok_file):
             assert r.status_code == 200
             assert r.json()["choices"][0]["text"] in {"yes", "no"}
 
+            # non-ASCII choices: byte-level masks make the multi-byte
+            # option reachable and the detokenizer reassembles it
+            r = await c.post(
+                "/v1/completions",
+                json={"prompt": "drink:", "max_tokens": 16,
+                      "temperature": 0.0,
+                      "guided_choice": ["café", "thé"]},
+            )
+            assert r.status_code == 200
+            assert r.json()["choices"][0]["text"] in {"café", "thé"}
+
     try:
         asyncio.run(run())
     finally:
